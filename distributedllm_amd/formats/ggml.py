@@ -1,0 +1,316 @@
+"""GGJT-v3 model / slice file reader-writer.
+
+The on-disk formats this framework is byte-compatible with:
+
+* **model files** — standard GGJT v3: magic ``'ggjt'`` + version 3, then a
+  **7-field** hparams block ``n_vocab, n_embd, n_mult, n_head, n_layer,
+  n_rot, ftype`` (reference reads it at slice_model.cpp:167-175), then the
+  vocab (``len:u32 ‖ bytes ‖ score:f32`` × n_vocab), then 32-byte-aligned
+  tensor records.
+* **slice files / extra_layers.bin** — the reference's extended format: an
+  extra ``first_layer:u32`` inserted between ``n_rot`` and ``ftype``
+  (**8 fields**; written at slice_model.cpp:253-263, read at
+  tensor_processor.cpp:179-188).  ``extra_layers.bin`` has ``n_layer=0`` and
+  ``first_layer=0xFFFFFFFF`` (slice_model.cpp:380-383).
+* tensor record: ``n_dims:u32 ‖ name_len:u32 ‖ type:u32 ‖ ne[n_dims]:u32 ‖
+  name`` then zero padding to the next 32-byte file offset, then raw data.
+  ``ne[0]`` is the contiguous (row/input) dimension.
+* slice tensor names keep their ORIGINAL layer indices; loaders re-base
+  with ``first_layer`` (tensor_processor.cpp:1340).
+
+Nothing is copied from the reference — this is a clean-room implementation
+of the byte layout described above.
+"""
+from __future__ import annotations
+
+import struct
+from dataclasses import dataclass, field
+from typing import BinaryIO, Dict, List, Optional, Tuple
+
+import numpy as np
+
+from . import q4
+
+GGJT_MAGIC = 0x67676A74  # bytes 'tjgg' little-endian == "ggjt"
+GGJT_VERSION = 3
+
+EXTRA_LAYERS_FIRST_LAYER = 0xFFFFFFFF
+
+# ggml tensor dtypes (on-disk ids)
+GGML_TYPE_F32 = 0
+GGML_TYPE_F16 = 1
+GGML_TYPE_Q4_0 = 2
+GGML_TYPE_Q4_1 = 3
+
+# model-level ftype
+FTYPE_ALL_F32 = 0
+FTYPE_MOSTLY_F16 = 1
+FTYPE_MOSTLY_Q4_0 = 2
+FTYPE_MOSTLY_Q4_1 = 3
+
+_FTYPE_TO_GGML = {
+    FTYPE_ALL_F32: GGML_TYPE_F32,
+    FTYPE_MOSTLY_F16: GGML_TYPE_F16,
+    FTYPE_MOSTLY_Q4_0: GGML_TYPE_Q4_0,
+    FTYPE_MOSTLY_Q4_1: GGML_TYPE_Q4_1,
+}
+
+TYPE_NAMES = {
+    GGML_TYPE_F32: "f32",
+    GGML_TYPE_F16: "f16",
+    GGML_TYPE_Q4_0: "q4_0",
+    GGML_TYPE_Q4_1: "q4_1",
+}
+
+
+def tensor_nbytes(gtype: int, ne: Tuple[int, ...]) -> int:
+    """Byte size of a tensor's data given its ggml type and dims."""
+    n0 = ne[0]
+    rows = 1
+    for d in ne[1:]:
+        rows *= d
+    if gtype == GGML_TYPE_F32:
+        return 4 * n0 * rows
+    if gtype == GGML_TYPE_F16:
+        return 2 * n0 * rows
+    if gtype == GGML_TYPE_Q4_0:
+        if n0 % q4.QK4:
+            raise ValueError(f"q4_0 row length {n0} not a multiple of 32")
+        return (n0 // q4.QK4) * q4.Q4_0_BLOCK_BYTES * rows
+    if gtype == GGML_TYPE_Q4_1:
+        if n0 % q4.QK4:
+            raise ValueError(f"q4_1 row length {n0} not a multiple of 32")
+        return (n0 // q4.QK4) * q4.Q4_1_BLOCK_BYTES * rows
+    raise ValueError(f"unsupported ggml type {gtype}")
+
+
+@dataclass
+class Hparams:
+    n_vocab: int
+    n_embd: int
+    n_mult: int
+    n_head: int
+    n_layer: int
+    n_rot: int
+    ftype: int
+    first_layer: Optional[int] = None  # present only in the extended format
+
+    @property
+    def n_ff(self) -> int:
+        """FFN width, the reference's formula (tensor_processor.cpp:1250)."""
+        return ((2 * (4 * self.n_embd) // 3 + self.n_mult - 1)
+                // self.n_mult) * self.n_mult
+
+    @property
+    def head_dim(self) -> int:
+        return self.n_embd // self.n_head
+
+
+@dataclass
+class GGMLTensor:
+    name: str
+    ne: Tuple[int, ...]          # ne[0] = contiguous/input dim
+    gtype: int
+    raw: bytes                   # on-disk bytes
+
+    @property
+    def nbytes(self) -> int:
+        return tensor_nbytes(self.gtype, self.ne)
+
+    @property
+    def shape_rows_cols(self) -> Tuple[int, int]:
+        """(rows, cols) with cols = ne[0] (contiguous)."""
+        rows = 1
+        for d in self.ne[1:]:
+            rows *= d
+        return rows, self.ne[0]
+
+    def to_f32(self) -> np.ndarray:
+        """Dequantize/convert to a float32 array of shape [rows, ne0]
+        (or [ne0] for 1-D tensors)."""
+        rows, cols = self.shape_rows_cols
+        if self.gtype == GGML_TYPE_F32:
+            a = np.frombuffer(self.raw, dtype=np.float32).astype(np.float32)
+        elif self.gtype == GGML_TYPE_F16:
+            a = np.frombuffer(self.raw, dtype=np.float16).astype(np.float32)
+        elif self.gtype == GGML_TYPE_Q4_0:
+            u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
+            a = q4.dequantize_q4_0(u, cols)
+        elif self.gtype == GGML_TYPE_Q4_1:
+            u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
+            a = q4.dequantize_q4_1(u, cols)
+        else:
+            raise ValueError(f"unsupported ggml type {self.gtype}")
+        if len(self.ne) == 1:
+            return a.reshape(cols)
+        return a.reshape(rows, cols)
+
+    @classmethod
+    def from_f32(cls, name: str, a: np.ndarray, gtype: int) -> "GGMLTensor":
+        """Quantize/convert a float array into a tensor record."""
+        a = np.asarray(a, dtype=np.float32)
+        if a.ndim == 1:
+            ne: Tuple[int, ...] = (a.shape[0],)
+        elif a.ndim == 2:
+            ne = (a.shape[1], a.shape[0])  # ne[0] = cols (contiguous)
+        else:
+            raise ValueError("only 1-D/2-D tensors are supported")
+        if gtype == GGML_TYPE_F32:
+            raw = a.astype(np.float32).tobytes()
+        elif gtype == GGML_TYPE_F16:
+            raw = a.astype(np.float16).tobytes()
+        elif gtype == GGML_TYPE_Q4_0:
+            raw = q4.quantize_q4_0(a).tobytes()
+        elif gtype == GGML_TYPE_Q4_1:
+            raw = q4.quantize_q4_1(a).tobytes()
+        else:
+            raise ValueError(f"unsupported ggml type {gtype}")
+        return cls(name=name, ne=ne, gtype=gtype, raw=raw)
+
+
+@dataclass
+class GGMLFile:
+    hparams: Hparams
+    vocab: List[Tuple[bytes, float]] = field(default_factory=list)
+    tensors: List[GGMLTensor] = field(default_factory=list)
+
+    @property
+    def is_extended(self) -> bool:
+        return self.hparams.first_layer is not None
+
+    def tensor_map(self) -> Dict[str, GGMLTensor]:
+        return {t.name: t for t in self.tensors}
+
+    # ---------------- writing ----------------
+
+    def save(self, path: str) -> None:
+        with open(path, "wb") as f:
+            self._write(f)
+
+    def _write(self, f: BinaryIO) -> None:
+        hp = self.hparams
+        f.write(struct.pack("<II", GGJT_MAGIC, GGJT_VERSION))
+        fields = [hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head, hp.n_layer,
+                  hp.n_rot]
+        if hp.first_layer is not None:
+            fields.append(hp.first_layer)
+        fields.append(hp.ftype)
+        f.write(struct.pack("<%dI" % len(fields), *fields))
+        if len(self.vocab) != hp.n_vocab:
+            raise ValueError(
+                f"vocab has {len(self.vocab)} entries, hparams say {hp.n_vocab}")
+        for word, score in self.vocab:
+            f.write(struct.pack("<I", len(word)))
+            f.write(word)
+            f.write(struct.pack("<f", score))
+        for t in self.tensors:
+            name_b = t.name.encode("utf-8")
+            f.write(struct.pack("<III", len(t.ne), len(name_b), t.gtype))
+            f.write(struct.pack("<%dI" % len(t.ne), *t.ne))
+            f.write(name_b)
+            pad = -f.tell() & 31
+            f.write(b"\x00" * pad)
+            if len(t.raw) != t.nbytes:
+                raise ValueError(
+                    f"tensor {t.name}: raw {len(t.raw)} B != expected {t.nbytes} B")
+            f.write(t.raw)
+
+    # ---------------- reading ----------------
+
+    @classmethod
+    def load(cls, path: str, extended: bool,
+             with_data: bool = True) -> "GGMLFile":
+        """Read a GGJT v3 file.
+
+        ``extended=True`` for slice/extra_layers files (8-field hparams),
+        ``False`` for original model files (7-field).
+        """
+        with open(path, "rb") as f:
+            data = f.read()
+        off = 0
+
+        def u32() -> int:
+            nonlocal off
+            (v,) = struct.unpack_from("<I", data, off)
+            off += 4
+            return v
+
+        magic = u32()
+        if magic != GGJT_MAGIC:
+            raise ValueError(f"bad magic 0x{magic:08x}; not a GGJT file")
+        version = u32()
+        if version != GGJT_VERSION:
+            raise ValueError(f"unsupported GGJT version {version}")
+        n_vocab = u32()
+        n_embd = u32()
+        n_mult = u32()
+        n_head = u32()
+        n_layer = u32()
+        n_rot = u32()
+        first_layer = u32() if extended else None
+        ftype = u32()
+        hp = Hparams(n_vocab=n_vocab, n_embd=n_embd, n_mult=n_mult,
+                     n_head=n_head, n_layer=n_layer, n_rot=n_rot,
+                     ftype=ftype, first_layer=first_layer)
+
+        vocab: List[Tuple[bytes, float]] = []
+        for _ in range(n_vocab):
+            ln = u32()
+            word = data[off:off + ln]
+            off += ln
+            (score,) = struct.unpack_from("<f", data, off)
+            off += 4
+            vocab.append((word, score))
+
+        tensors: List[GGMLTensor] = []
+        total = len(data)
+        while off < total:
+            n_dims = u32()
+            name_len = u32()
+            gtype = u32()
+            if n_dims < 1 or n_dims > 2:
+                raise ValueError(f"tensor with {n_dims} dims")
+            ne = struct.unpack_from("<%dI" % n_dims, data, off)
+            off += 4 * n_dims
+            name = data[off:off + name_len].decode("utf-8")
+            off += name_len
+            off += -off & 31
+            size = tensor_nbytes(gtype, ne)
+            raw = data[off:off + size] if with_data else b""
+            if with_data and len(raw) != size:
+                raise ValueError(f"truncated tensor data for {name}")
+            off += size
+            tensors.append(GGMLTensor(name=name, ne=tuple(ne), gtype=gtype,
+                                      raw=bytes(raw)))
+        return cls(hparams=hp, vocab=vocab, tensors=tensors)
+
+
+def sniff_extended(path: str) -> bool:
+    """Heuristic: is this file in the extended (8-field) header format?
+
+    The extended header is valid iff interpreting field 7 as ``first_layer``
+    and field 8 as ``ftype`` yields a known ftype AND the resulting vocab
+    walk lands exactly on a tensor record. We use the cheap test: ftype
+    value plausibility in both interpretations, preferring the explicit one
+    when unambiguous.
+    """
+    with open(path, "rb") as f:
+        head = f.read(4 * 10)
+    vals = struct.unpack_from("<10I", head, 0)
+    if vals[0] != GGJT_MAGIC:
+        raise ValueError("not a GGJT file")
+    # vals[2:8] = 7-field hparams; vals[8] would be first tensor field.
+    f7_ftype = vals[8 - 1]    # 7-field: ftype at index 8-1=7? careful below
+    # indices: 0 magic, 1 version, 2 n_vocab, 3 n_embd, 4 n_mult, 5 n_head,
+    # 6 n_layer, 7 n_rot, then [8]=ftype (7-field) or first_layer (8-field),
+    # [9]=ftype (8-field).
+    seven_ok = vals[8] in _FTYPE_TO_GGML
+    eight_ok = vals[9] in _FTYPE_TO_GGML and (
+        vals[8] == EXTRA_LAYERS_FIRST_LAYER or vals[8] < 4096)
+    if eight_ok and not seven_ok:
+        return True
+    if seven_ok and not eight_ok:
+        return False
+    # Ambiguous (both parse): treat files with the extra_layers sentinel or a
+    # small first_layer AND a valid trailing ftype as extended.
+    return eight_ok

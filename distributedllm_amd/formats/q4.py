@@ -1,0 +1,117 @@
+"""q4_0 / q4_1 block quantization codecs (GGJT v3 block layout).
+
+Bit-exact with the GGJT v3 layouts the reference engine consumes
+(/root/reference/distllm/tensor_processor.cpp:846-855 lists the ftypes; the
+vendored ``quantize`` binary produced the blocks — SURVEY.md §2.2 N4).
+
+Block layout, 32 weights per block:
+
+* q4_0: ``d`` (f16 scale) ‖ 16 bytes of nibbles.  Byte ``j`` holds weight
+  ``j`` in its low nibble and weight ``j+16`` in its high nibble.
+  Dequant: ``x[j] = d * (q[j] - 8)``.
+  Quant: ``amax``-signed scaling — let ``m`` be the element with the largest
+  absolute value; ``d = m / -8``; ``q = clamp(round(x/d) + 8, 0, 15)``.
+* q4_1: ``d`` (f16) ‖ ``m`` (f16) ‖ 16 nibble bytes.
+  Dequant: ``x = d*q + m``; quant: ``d=(max-min)/15``, ``m=min``.
+
+All functions are vectorized numpy; used by provisioning (offline tooling)
+and by tests as the ground truth the HIP dequant kernels must match.
+"""
+from __future__ import annotations
+
+import numpy as np
+
+QK4 = 32  # weights per block
+Q4_0_BLOCK_BYTES = 2 + 16
+Q4_1_BLOCK_BYTES = 4 + 16
+
+
+def _check_shape(n: int) -> int:
+    if n % QK4 != 0:
+        raise ValueError(f"row length {n} is not a multiple of {QK4}")
+    return n // QK4
+
+
+def quantize_q4_0(x: np.ndarray) -> np.ndarray:
+    """Quantize a float array (last dim multiple of 32) to q4_0 bytes.
+
+    Returns a uint8 array of shape ``(*x.shape[:-1], nblocks*18)``.
+    """
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    lead = x.shape[:-1]
+    nb = _check_shape(x.shape[-1])
+    b = x.reshape(-1, nb, QK4)
+
+    # signed amax: the element with the largest |value|, keeping its sign
+    idx = np.argmax(np.abs(b), axis=-1)
+    m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
+    d = (m / -8.0).astype(np.float16)
+    df = d.astype(np.float32)
+    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    q = np.clip(np.rint(b * inv[..., None]) + 8, 0, 15).astype(np.uint8)
+    lo, hi = q[..., :16], q[..., 16:]
+    packed = (lo | (hi << 4)).astype(np.uint8)
+
+    out = np.empty(b.shape[:2] + (Q4_0_BLOCK_BYTES,), dtype=np.uint8)
+    out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
+    out[..., 2:] = packed
+    return out.reshape(lead + (nb * Q4_0_BLOCK_BYTES,))
+
+
+def dequantize_q4_0(raw: np.ndarray, n: int) -> np.ndarray:
+    """Dequantize q4_0 bytes back to f32. ``n`` = row length in weights."""
+    raw = np.ascontiguousarray(raw, dtype=np.uint8)
+    nb = _check_shape(n)
+    lead = raw.shape[:-1]
+    if raw.shape[-1] != nb * Q4_0_BLOCK_BYTES:
+        raise ValueError(
+            f"raw length {raw.shape[-1]} != {nb * Q4_0_BLOCK_BYTES} for n={n}")
+    b = raw.reshape(-1, nb, Q4_0_BLOCK_BYTES)
+    d = b[..., 0:2].copy().view(np.float16)[..., 0].astype(np.float32)
+    qs = b[..., 2:]
+    lo = (qs & 0x0F).astype(np.int8) - 8
+    hi = (qs >> 4).astype(np.int8) - 8
+    q = np.concatenate([lo, hi], axis=-1).astype(np.float32)
+    return (q * d[..., None]).reshape(lead + (n,))
+
+
+def quantize_q4_1(x: np.ndarray) -> np.ndarray:
+    """Quantize to q4_1 (min/max affine)."""
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    lead = x.shape[:-1]
+    nb = _check_shape(x.shape[-1])
+    b = x.reshape(-1, nb, QK4)
+
+    mn = b.min(axis=-1)
+    mx = b.max(axis=-1)
+    d = ((mx - mn) / 15.0).astype(np.float16)
+    m = mn.astype(np.float16)
+    df = d.astype(np.float32)
+    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) * inv[..., None]),
+                0, 15).astype(np.uint8)
+    lo, hi = q[..., :16], q[..., 16:]
+    packed = (lo | (hi << 4)).astype(np.uint8)
+
+    out = np.empty(b.shape[:2] + (Q4_1_BLOCK_BYTES,), dtype=np.uint8)
+    out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
+    out[..., 2:4] = m[..., None].view(np.uint8).reshape(m.shape + (2,))
+    out[..., 4:] = packed
+    return out.reshape(lead + (nb * Q4_1_BLOCK_BYTES,))
+
+
+def dequantize_q4_1(raw: np.ndarray, n: int) -> np.ndarray:
+    raw = np.ascontiguousarray(raw, dtype=np.uint8)
+    nb = _check_shape(n)
+    lead = raw.shape[:-1]
+    if raw.shape[-1] != nb * Q4_1_BLOCK_BYTES:
+        raise ValueError(
+            f"raw length {raw.shape[-1]} != {nb * Q4_1_BLOCK_BYTES} for n={n}")
+    b = raw.reshape(-1, nb, Q4_1_BLOCK_BYTES)
+    d = b[..., 0:2].copy().view(np.float16)[..., 0].astype(np.float32)
+    m = b[..., 2:4].copy().view(np.float16)[..., 0].astype(np.float32)
+    qs = b[..., 4:]
+    lo = (qs & 0x0F).astype(np.float32)
+    hi = (qs >> 4).astype(np.float32)
+    q = np.concatenate([lo, hi], axis=-1)
+    return (q * d[..., None] + m[..., None]).reshape(lead + (n,))

@@ -1,0 +1,153 @@
+"""Formats layer: q4 codecs, GGJT v3 reader/writer, slicer, synthetic gen."""
+import struct
+
+import numpy as np
+import pytest
+
+from distributedllm_amd.formats import ggml, q4, slicer, synthetic
+from distributedllm_amd.models.llama import PRESETS
+
+
+class TestQ4:
+    def test_q4_0_roundtrip_error(self):
+        rng = np.random.default_rng(0)
+        x = rng.standard_normal((8, 256)).astype(np.float32)
+        raw = q4.quantize_q4_0(x)
+        y = q4.dequantize_q4_0(raw, 256)
+        # q4_0 error bound: quantization step is |d| = amax/8; values on the
+        # clipped (+7d..+8d) side can be off by up to one full step.
+        scales = np.abs(x.reshape(8, -1, 32)).max(axis=-1) / 8.0
+        err = np.abs(x - y).reshape(8, -1, 32).max(axis=-1)
+        assert np.all(err <= scales * 1.01 + 1e-6)
+
+    def test_q4_0_block_layout(self):
+        # one block with known values: weight j -> low nibble of byte j,
+        # weight j+16 -> high nibble (GGJT v3 layout)
+        x = np.zeros(32, dtype=np.float32)
+        x[0] = -8.0   # amax element, signed -> d = -8/-8 = 1.0
+        x[1] = 3.0
+        x[17] = -2.0
+        raw = q4.quantize_q4_0(x)
+        assert raw.shape == (18,)
+        d = np.frombuffer(raw[:2].tobytes(), dtype=np.float16)[0]
+        assert d == np.float16(1.0)
+        qs = raw[2:]
+        assert qs[0] & 0x0F == 0        # -8 -> q=0
+        assert qs[1] & 0x0F == 11       # 3 -> 11
+        assert qs[1] >> 4 == 6          # x[17] = -2 -> 6
+        assert qs[2] >> 4 == 8          # 0 -> 8
+
+    def test_q4_0_exact_grid(self):
+        # values on the q4_0 grid reconstruct exactly
+        d = 0.5
+        q = np.arange(32) % 16
+        x = ((q - 8) * d).astype(np.float32)
+        x[0] = -8 * d  # ensure amax yields d
+        y = q4.dequantize_q4_0(q4.quantize_q4_0(x), 32)
+        assert np.allclose(x, y, atol=1e-3)
+
+    def test_q4_1_roundtrip(self):
+        rng = np.random.default_rng(1)
+        x = rng.uniform(-1, 3, size=(4, 64)).astype(np.float32)
+        y = q4.dequantize_q4_1(q4.quantize_q4_1(x), 64)
+        scales = (x.reshape(4, -1, 32).max(-1) - x.reshape(4, -1, 32).min(-1)) / 15
+        err = np.abs(x - y).reshape(4, -1, 32).max(axis=-1)
+        assert np.all(err <= scales * 0.51 + 5e-3)
+
+
+class TestGGMLFile:
+    def _tiny_file(self, ftype=ggml.FTYPE_MOSTLY_Q4_0):
+        return synthetic.build_model("tiny", ftype=ftype, seed=3)
+
+    def test_roundtrip(self, tmp_path):
+        f = self._tiny_file()
+        p = str(tmp_path / "m.bin")
+        f.save(p)
+        g = ggml.GGMLFile.load(p, extended=False)
+        assert g.hparams == f.hparams
+        assert len(g.vocab) == f.hparams.n_vocab
+        assert [t.name for t in g.tensors] == [t.name for t in f.tensors]
+        for a, b in zip(f.tensors, g.tensors):
+            assert a.ne == b.ne and a.gtype == b.gtype and a.raw == b.raw
+
+    def test_header_layout(self, tmp_path):
+        f = self._tiny_file()
+        p = str(tmp_path / "m.bin")
+        f.save(p)
+        head = open(p, "rb").read(4 * 9)
+        vals = struct.unpack("<9I", head)
+        hp = f.hparams
+        assert vals[0] == 0x67676A74 and vals[1] == 3
+        assert vals[2:9] == (hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head,
+                             hp.n_layer, hp.n_rot, hp.ftype)
+
+    def test_tensor_alignment(self, tmp_path):
+        f = self._tiny_file()
+        p = str(tmp_path / "m.bin")
+        f.save(p)
+        raw = open(p, "rb").read()
+        # find first tensor record: after header + vocab
+        off = 4 * 9
+        for w, _s in f.vocab:
+            off += 4 + len(w) + 4
+        n_dims, name_len, gtype = struct.unpack_from("<III", raw, off)
+        off += 12 + 4 * n_dims + name_len
+        data_off = off + (-off & 31)
+        assert data_off % 32 == 0
+
+    def test_extended_header(self, tmp_path):
+        f = self._tiny_file()
+        sl = slicer.make_slice(f, 1, 2)
+        p = str(tmp_path / "s.bin")
+        sl.save(p)
+        g = ggml.GGMLFile.load(p, extended=True)
+        assert g.hparams.first_layer == 1
+        assert g.hparams.n_layer == 2
+        vals = struct.unpack("<10I", open(p, "rb").read(40))
+        assert vals[8] == 1  # first_layer between n_rot and ftype
+        assert vals[9] == f.hparams.ftype
+        assert ggml.sniff_extended(p) is True
+
+    def test_nff_formula(self):
+        assert PRESETS["open_llama_3b"].n_ff == 8640
+        assert PRESETS["llama_7b"].n_ff == 11008
+        assert PRESETS["llama_13b"].n_ff == 13824
+        assert PRESETS["llama_30b"].n_ff == 17920
+        assert PRESETS["llama_65b"].n_ff == 22016
+
+
+class TestSlicer:
+    def test_slice_selection(self):
+        f = synthetic.build_model("tiny", seed=0)
+        sl = slicer.make_slice(f, 0, 1)
+        names = {t.name for t in sl.tensors}
+        assert all(n.startswith(("layers.0.", "layers.1.")) for n in names)
+        assert len(names) == 18
+        # original indices preserved
+        assert "layers.1.attention.wq.weight" in names
+
+    def test_extra_layers(self):
+        f = synthetic.build_model("tiny", seed=0)
+        ex = slicer.make_extra_layers(f)
+        assert {t.name for t in ex.tensors} == {
+            "tok_embeddings.weight", "norm.weight", "output.weight"}
+        assert ex.hparams.n_layer == 0
+        assert ex.hparams.first_layer == ggml.EXTRA_LAYERS_FIRST_LAYER
+
+    def test_raw_bytes_preserved(self, tmp_path):
+        f = synthetic.build_model("tiny", seed=0)
+        src = str(tmp_path / "m.bin")
+        out = str(tmp_path / "s.bin")
+        f.save(src)
+        slicer.slice_model_file(src, 1, 2, out)
+        sl = ggml.GGMLFile.load(out, extended=True)
+        orig = f.tensor_map()
+        for t in sl.tensors:
+            assert t.raw == orig[t.name].raw
+
+    def test_bad_range(self):
+        f = synthetic.build_model("tiny", seed=0)
+        with pytest.raises(ValueError):
+            slicer.make_slice(f, 2, 1)
+        with pytest.raises(ValueError):
+            slicer.make_slice(f, 0, 99)
