@@ -1,0 +1,290 @@
+"""Slice engines: the HIP/CDNA4 production engine and a CPU torch twin.
+
+Both expose the same *stateless* interface (positions and sequence ids are
+explicit per-token arrays), which is what makes the decode step
+graph-capturable and the pipeline testable on CPU:
+
+    forward(x[T,E] f32, pos[T] i32, seq[T] i32) -> x'   (KV appended)
+    embed(tokens[T]) -> x[T,E]
+    logits(x[T,E], all_logits=False) -> [T|1, V]
+
+``clear_context`` in the reference (tensor_processor.cpp:1512-1521 destroys
+and recreates the llama_context) is here simply "start writing positions
+from 0 again" — cache slots are overwritten, nothing to reset.
+
+The HIP engine repacks on-disk q4 AoS blocks into the SoA layout the
+kernels stream (scales ‖ nibbles, see csrc/kernels.hip); repacking happens
+once at load on the host, uploads via torch pinned copies.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..formats import ggml, q4
+from ..models.llama import RMS_EPS, ROPE_BASE, rms_norm, rope_interleaved
+
+
+def _repack_q4(t: ggml.GGMLTensor):
+    rows, cols = t.shape_rows_cols
+    nb = cols // 32
+    if t.gtype == ggml.GGML_TYPE_Q4_0:
+        a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, 18)
+        scales = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
+        scales = scales.reshape(rows, nb)
+        qs = np.ascontiguousarray(a[:, :, 2:]).reshape(rows, nb * 16)
+    else:  # q4_1: 20-byte blocks, (d, m) f16 pairs
+        a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, 20)
+        scales = np.ascontiguousarray(a[:, :, :4]).view(np.float16)
+        scales = scales.reshape(rows, nb * 2)
+        qs = np.ascontiguousarray(a[:, :, 4:]).reshape(rows, nb * 16)
+    return scales, qs
+
+
+def _upload_mat(t: ggml.GGMLTensor, device: str):
+    """-> (data, scales, wtype) device tensors in the kernel layout."""
+    rows, cols = t.shape_rows_cols
+    if t.gtype in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+        scales, qs = _repack_q4(t)
+        d = torch.from_numpy(qs).to(device)
+        s = torch.from_numpy(scales.copy()).to(device)
+        return d, s, t.gtype
+    if t.gtype == ggml.GGML_TYPE_F16:
+        a = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
+        return torch.from_numpy(a.copy()).to(device), torch.empty(0), t.gtype
+    a = np.frombuffer(t.raw, np.float32).reshape(rows, cols)
+    return torch.from_numpy(a.copy()).to(device), torch.empty(0), t.gtype
+
+
+class HIPSliceEngine:
+    """Production engine: CDNA4 kernels, weights resident in HBM3E."""
+
+    def __init__(self, hp: ggml.Hparams, n_layers: int, first_layer: int,
+                 n_ctx: int = 2048, max_batch: int = 16):
+        from .. import ops
+        core = ops.core()
+        self.hp = hp
+        self.first_layer = first_layer
+        self.n_layers = n_layers
+        self.n_ctx = n_ctx
+        self.max_batch = max_batch
+        self._eng = core.SliceEngine(
+            n_embd=hp.n_embd, n_head=hp.n_head, n_layers=n_layers,
+            n_ff=hp.n_ff, n_ctx=n_ctx, max_batch=max_batch,
+            eps=RMS_EPS, rope_base=ROPE_BASE)
+        self.device = "cuda"
+        self.has_extra = False
+
+    @classmethod
+    def from_ggml(cls, f: ggml.GGMLFile, n_ctx: int = 2048,
+                  max_batch: int = 16) -> "HIPSliceEngine":
+        hp = f.hparams
+        first = hp.first_layer if hp.first_layer is not None else 0
+        eng = cls(hp, hp.n_layer, first, n_ctx, max_batch)
+        eng.load_layers(f)
+        return eng
+
+    @classmethod
+    def random(cls, hp: ggml.Hparams, n_layers: int, first_layer: int = 0,
+               n_ctx: int = 2048, max_batch: int = 16, seed: int = 0,
+               with_extra: bool = True) -> "HIPSliceEngine":
+        """Random-init engine straight on the GPU (synthetic benchmarking).
+
+        Generates weights directly in the repacked kernel layout — identical
+        compute and HBM traffic to a real checkpoint of this architecture,
+        without materializing a multi-GB GGML file on disk.
+        """
+        eng = cls(hp, n_layers, first_layer, n_ctx, max_batch)
+        g = torch.Generator(device="cuda")
+        g.manual_seed(seed)
+        E, F, V = hp.n_embd, hp.n_ff, hp.n_vocab
+        wt = ggml._FTYPE_TO_GGML[hp.ftype]
+
+        def mat(rows: int, cols: int):
+            if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+                nb = cols // 32
+                per = 2 if wt == ggml.GGML_TYPE_Q4_1 else 1
+                data = torch.randint(0, 256, (rows, nb * 16),
+                                     dtype=torch.uint8, device="cuda",
+                                     generator=g)
+                scales = ((torch.rand(rows, nb * per, device="cuda",
+                                      generator=g) * 0.5 + 0.75) *
+                          0.003).to(torch.float16)
+                return data, scales, wt
+            dt = torch.float16 if wt == ggml.GGML_TYPE_F16 else torch.float32
+            data = (torch.randn(rows, cols, device="cuda", generator=g,
+                                dtype=torch.float32) * 0.02).to(dt)
+            return data.contiguous(), torch.empty(0), wt
+
+        def norm_w(n: int):
+            return (1.0 + torch.randn(n, device="cuda", generator=g) *
+                    0.01).contiguous()
+
+        shapes = [(E, E), (E, E), (E, E), (E, E), (F, E), (E, F), (F, E)]
+        for li in range(n_layers):
+            mats = [mat(r, c) for r, c in shapes]
+            eng._eng.set_layer(li, norm_w(E), norm_w(E), mats)
+        if with_extra:
+            tok_d, tok_s, tok_t = mat(V, E)
+            out_d, out_s, out_t = mat(V, E)
+            eng._eng.set_extra(tok_d, tok_s, tok_t, norm_w(E), out_d, out_s,
+                               out_t, V)
+            eng.has_extra = True
+        return eng
+
+    def load_layers(self, f: ggml.GGMLFile) -> None:
+        tm = f.tensor_map()
+        for li in range(self.n_layers):
+            gi = li + self.first_layer
+            pre = f"layers.{gi}."
+            attn_norm = torch.from_numpy(
+                tm[pre + "attention_norm.weight"].to_f32()).to(self.device)
+            ffn_norm = torch.from_numpy(
+                tm[pre + "ffn_norm.weight"].to_f32()).to(self.device)
+            mats = []
+            for nm in ("attention.wq.weight", "attention.wk.weight",
+                       "attention.wv.weight", "attention.wo.weight",
+                       "feed_forward.w1.weight", "feed_forward.w2.weight",
+                       "feed_forward.w3.weight"):
+                mats.append(_upload_mat(tm[pre + nm], self.device))
+            self._eng.set_layer(li, attn_norm, ffn_norm, mats)
+
+    def attach_extra(self, f: ggml.GGMLFile) -> None:
+        tm = f.tensor_map()
+        tok_d, tok_s, tok_t = _upload_mat(tm["tok_embeddings.weight"],
+                                          self.device)
+        out_d, out_s, out_t = _upload_mat(tm["output.weight"], self.device)
+        norm = torch.from_numpy(tm["norm.weight"].to_f32()).to(self.device)
+        self._eng.set_extra(tok_d, tok_s, tok_t, norm, out_d, out_s, out_t,
+                            self.hp.n_vocab)
+        self.has_extra = True
+
+    def forward(self, x: torch.Tensor, pos: torch.Tensor,
+                seq: torch.Tensor) -> torch.Tensor:
+        T = x.shape[0]
+        mt = self._eng.max_tokens
+        if T <= mt:
+            return self._eng.forward(x, pos, seq)
+        # token-tile larger inputs (prefill); KV order is preserved because
+        # tile i's cache rows are written before tile i+1 attends.
+        outs = []
+        for t0 in range(0, T, mt):
+            t1 = min(T, t0 + mt)
+            outs.append(self._eng.forward(
+                x[t0:t1].contiguous(), pos[t0:t1].contiguous(),
+                seq[t0:t1].contiguous()))
+        return torch.cat(outs, dim=0)
+
+    def embed(self, tokens: torch.Tensor) -> torch.Tensor:
+        return self._eng.embed(tokens.to(self.device, torch.int32))
+
+    def logits(self, x: torch.Tensor, all_logits: bool = False) -> torch.Tensor:
+        if not all_logits:
+            return self._eng.logits(x, False)
+        mt = self._eng.max_tokens
+        if x.shape[0] <= mt:
+            return self._eng.logits(x, True)
+        return torch.cat([self._eng.logits(x[t0:t0 + mt].contiguous(), True)
+                          for t0 in range(0, x.shape[0], mt)], dim=0)
+
+    def argmax(self, lg: torch.Tensor) -> torch.Tensor:
+        return self._eng.argmax(lg)
+
+
+class TorchSliceEngine:
+    """CPU twin with the identical stateless interface (fp32 torch math).
+
+    Used for CPU tests, the gloo pipeline path, and machines without a GPU.
+    Never selected automatically when CUDA is available — the HIP engine is
+    the only GPU path.
+    """
+
+    def __init__(self, hp: ggml.Hparams, weights: Dict[str, torch.Tensor],
+                 n_layers: int, first_layer: int, n_ctx: int = 512,
+                 max_batch: int = 4, device: str = "cpu"):
+        self.hp = hp
+        self.w = weights
+        self.first_layer = first_layer
+        self.n_layers = n_layers
+        self.n_ctx = n_ctx
+        self.max_batch = max_batch
+        self.device = device
+        h, d = hp.n_head, hp.head_dim
+        self.k_cache = torch.zeros(n_layers, max_batch, n_ctx, h, d)
+        self.v_cache = torch.zeros(n_layers, max_batch, n_ctx, h, d)
+        self.has_extra = all(
+            k in weights for k in
+            ("tok_embeddings.weight", "norm.weight", "output.weight"))
+
+    @classmethod
+    def from_ggml(cls, f: ggml.GGMLFile, n_ctx: int = 512,
+                  max_batch: int = 4) -> "TorchSliceEngine":
+        from ..models.llama import weights_from_ggml
+        hp = f.hparams
+        first = hp.first_layer if hp.first_layer is not None else 0
+        return cls(hp, weights_from_ggml(f), hp.n_layer, first, n_ctx,
+                   max_batch)
+
+    def attach_extra(self, f: ggml.GGMLFile) -> None:
+        from ..models.llama import weights_from_ggml
+        self.w.update(weights_from_ggml(f))
+        self.has_extra = True
+
+    def forward(self, x: torch.Tensor, pos: torch.Tensor,
+                seq: torch.Tensor) -> torch.Tensor:
+        hp = self.hp
+        T, E = x.shape
+        h, d = hp.n_head, hp.head_dim
+        pos_l = pos.tolist()
+        seq_l = seq.tolist()
+        for li in range(self.n_layers):
+            gi = li + self.first_layer
+            pre = f"layers.{gi}."
+            a = rms_norm(x) * self.w[pre + "attention_norm.weight"]
+            q = a @ self.w[pre + "attention.wq.weight"].T
+            k = a @ self.w[pre + "attention.wk.weight"].T
+            v = a @ self.w[pre + "attention.wv.weight"].T
+            o = torch.empty_like(x)
+            for t in range(T):  # per-row positions (prefill/decode unified)
+                p, b = pos_l[t], seq_l[t]
+                qt = rope_interleaved(q[t:t + 1].view(1, h, d), p)[0]
+                kt = rope_interleaved(k[t:t + 1].view(1, h, d), p)[0]
+                self.k_cache[li, b, p] = kt
+                self.v_cache[li, b, p] = v[t].view(h, d)
+                keys = self.k_cache[li, b, :p + 1]
+                vals = self.v_cache[li, b, :p + 1]
+                att = torch.einsum("hd,jhd->hj", qt, keys) / math.sqrt(d)
+                prob = torch.softmax(att, dim=-1)
+                o[t] = torch.einsum("hj,jhd->hd", prob, vals).reshape(E)
+            x = x + o @ self.w[pre + "attention.wo.weight"].T
+            f = rms_norm(x) * self.w[pre + "ffn_norm.weight"]
+            g = torch.nn.functional.silu(
+                f @ self.w[pre + "feed_forward.w1.weight"].T)
+            u = f @ self.w[pre + "feed_forward.w3.weight"].T
+            x = x + (g * u) @ self.w[pre + "feed_forward.w2.weight"].T
+        return x
+
+    def embed(self, tokens: torch.Tensor) -> torch.Tensor:
+        idx = tokens.to(torch.long)
+        return self.w["tok_embeddings.weight"][idx]
+
+    def logits(self, x: torch.Tensor, all_logits: bool = False) -> torch.Tensor:
+        y = rms_norm(x) * self.w["norm.weight"]
+        lg = y @ self.w["output.weight"].T
+        return lg if all_logits else lg[-1:]
+
+    def argmax(self, lg: torch.Tensor) -> torch.Tensor:
+        return lg.argmax(dim=-1).to(torch.int32)
+
+
+def engine_for_slice(f: ggml.GGMLFile, n_ctx: int, max_batch: int,
+                     device: Optional[str] = None):
+    """Pick the engine for this machine: HIP when a GPU is present."""
+    if device is None:
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+    if device == "cuda":
+        return HIPSliceEngine.from_ggml(f, n_ctx=n_ctx, max_batch=max_batch)
+    return TorchSliceEngine.from_ggml(f, n_ctx=n_ctx, max_batch=max_batch)

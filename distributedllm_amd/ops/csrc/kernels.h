@@ -1,0 +1,44 @@
+// Launch API of the CDNA4 kernel library (kernels.hip).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <stdint.h>
+
+enum WType { W_F32 = 0, W_F16 = 1, W_Q4_0 = 2, W_Q4_1 = 3 };
+
+// One weight matrix resident in HBM, repacked SoA (see kernels.hip header).
+struct WMat {
+    const void* data;    // f16/f32 values, or u8 nibbles for q4_*
+    const void* scales;  // q4_0: f16[rows][nb]; q4_1: f16[rows][nb*2] (d,m)
+    int rows;
+    int cols;
+    int wtype;  // WType
+};
+
+void launch_rmsnorm(hipStream_t s, const float* x, const float* w, float* y,
+                    int T, int E, float eps);
+
+void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
+                            const WMat& wv, const float* xn, float* q_buf,
+                            __half* k_cache_layer, __half* v_cache_layer,
+                            const int* pos, const int* seq,
+                            const float* inv_freq, int E, int D, int n_ctx,
+                            int T);
+
+void launch_attention(hipStream_t s, const float* q_buf,
+                      const __half* k_cache_layer,
+                      const __half* v_cache_layer, float* out, const int* pos,
+                      const int* seq, int T, int H, int E, int D, int n_ctx);
+
+void launch_gemv(hipStream_t s, const WMat& w, const float* x,
+                 const float* res, float* y, int T);
+
+void launch_ffn_gate(hipStream_t s, const WMat& w1, const WMat& w3,
+                     const float* xn, float* g, int T);
+
+void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
+                  float* out, int T, int E);
+
+void launch_argmax(hipStream_t s, const float* logits, int* out, int T,
+                   int V);

@@ -1,0 +1,703 @@
+// CDNA4 (gfx950) kernels for the layer-sliced LLaMA inference engine.
+//
+// Replaces the vendored ggml CPU kernels of the reference
+// (/root/reference/distllm/tensor_processor.cpp:474-809 builds the graph this
+// file executes natively; SURVEY.md §2.5 maps each op K1-K15 to a kernel
+// here). Decode (N small) is HBM-bandwidth-bound on the q4_0 weight stream,
+// so the GEMV family streams repacked (SoA) q4_0 blocks with coalesced
+// 16-byte nibble loads and fuses dequant + per-block f16 scale in-register;
+// RoPE + KV-cache append are fused into the QKV projection epilogue;
+// attention is a streaming online-softmax kernel (no [P+N,N,H] score tensor).
+//
+// Weight layout in HBM (repacked at load, not the on-disk AoS layout):
+//   q4_0 : scales f16[rows][nb]  +  qs u8[rows][nb*16]   (nb = cols/32)
+//   q4_1 : scales f16[rows][nb*2] (d,m pairs) + qs as above
+//   f16  : data [rows][cols]
+//   f32  : data [rows][cols]
+//
+// Wavefront = 64 (CDNA); blocks of 256 threads = 4 waves; one wave per
+// output row in the GEMV family, lanes striding the row's q4 blocks so that
+// scale loads (2 B x 64 lanes) and nibble loads (16 B x 64 lanes) are fully
+// coalesced 128 B / 1 KiB wave transactions.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_fp16.h>
+#include <math.h>
+#include <stdint.h>
+
+#include "kernels.h"
+
+#define WAVE 64
+#define NWAVES 4
+#define BLOCK 256
+
+// ---------------------------------------------------------------- reductions
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off);
+    return v;  // valid in lane 0
+}
+
+__device__ __forceinline__ float wave_reduce_max(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off));
+    return v;
+}
+
+// --------------------------------------------------------------- row x vec
+// One WAVE computes dot(row r of W, x_t) for T tokens (T <= TMAX).
+// x: [T][cols] f32 (row-major, ld = cols). Results valid in lane 0.
+
+template <int TMAX>
+__device__ __forceinline__ void wave_row_dot_q4_0(
+    const __half* __restrict__ scales, const uint8_t* __restrict__ qs,
+    int row, int cols, const float* __restrict__ x, int T,
+    float* __restrict__ acc /* [TMAX] */) {
+    const int nb = cols >> 5;  // 32 weights per block
+    const __half* srow = scales + (size_t)row * nb;
+    const uint8_t* qrow = qs + (size_t)row * nb * 16;
+    const int lane = threadIdx.x & (WAVE - 1);
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
+    for (int b = lane; b < nb; b += WAVE) {
+        const float d = __half2float(srow[b]);
+        const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
+        const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            const float* xb = x + (size_t)t * cols + (b << 5);
+            float s = 0.0f;
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const uint32_t q = w[i];
+                // byte j of the block: low nibble = weight j,
+                // high nibble = weight j+16 (GGJT v3 q4_0 layout)
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int lo = (int)((q >> (8 * j)) & 0xF) - 8;
+                    const int hi = (int)((q >> (8 * j + 4)) & 0xF) - 8;
+                    s = fmaf((float)lo, xb[i * 4 + j], s);
+                    s = fmaf((float)hi, xb[16 + i * 4 + j], s);
+                }
+            }
+            acc[t] = fmaf(d, s, acc[t]);
+        }
+    }
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t)
+        if (t < T) acc[t] = wave_reduce_sum(acc[t]);
+}
+
+template <int TMAX>
+__device__ __forceinline__ void wave_row_dot_q4_1(
+    const __half* __restrict__ scales, const uint8_t* __restrict__ qs,
+    int row, int cols, const float* __restrict__ x, int T,
+    float* __restrict__ acc) {
+    const int nb = cols >> 5;
+    const __half* srow = scales + (size_t)row * nb * 2;
+    const uint8_t* qrow = qs + (size_t)row * nb * 16;
+    const int lane = threadIdx.x & (WAVE - 1);
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
+    for (int b = lane; b < nb; b += WAVE) {
+        const __half2 dm = *reinterpret_cast<const __half2*>(srow + b * 2);
+        const float d = __half2float(__low2half(dm));
+        const float m = __half2float(__high2half(dm));
+        const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
+        const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            const float* xb = x + (size_t)t * cols + (b << 5);
+            float s = 0.0f;   // sum q*x
+            float sx = 0.0f;  // sum x
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+                const uint32_t q = w[i];
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const float x0 = xb[i * 4 + j];
+                    const float x1 = xb[16 + i * 4 + j];
+                    s = fmaf((float)((q >> (8 * j)) & 0xF), x0, s);
+                    s = fmaf((float)((q >> (8 * j + 4)) & 0xF), x1, s);
+                    sx += x0 + x1;
+                }
+            }
+            acc[t] += d * s + m * sx;
+        }
+    }
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t)
+        if (t < T) acc[t] = wave_reduce_sum(acc[t]);
+}
+
+template <int TMAX>
+__device__ __forceinline__ void wave_row_dot_f16(
+    const __half* __restrict__ data, int row, int cols,
+    const float* __restrict__ x, int T, float* __restrict__ acc) {
+    const __half* wrow = data + (size_t)row * cols;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int nch = cols >> 3;  // 8 halves (16 B) per chunk
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
+    for (int c = lane; c < nch; c += WAVE) {
+        const uint4 packed = *reinterpret_cast<const uint4*>(wrow + c * 8);
+        const __half2* h2 = reinterpret_cast<const __half2*>(&packed);
+        float wv[8];
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+            const float2 f = __half22float2(h2[i]);
+            wv[2 * i] = f.x;
+            wv[2 * i + 1] = f.y;
+        }
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            const float* xb = x + (size_t)t * cols + c * 8;
+            float s = 0.0f;
+#pragma unroll
+            for (int i = 0; i < 8; ++i) s = fmaf(wv[i], xb[i], s);
+            acc[t] += s;
+        }
+    }
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t)
+        if (t < T) acc[t] = wave_reduce_sum(acc[t]);
+}
+
+template <int TMAX>
+__device__ __forceinline__ void wave_row_dot_f32(
+    const float* __restrict__ data, int row, int cols,
+    const float* __restrict__ x, int T, float* __restrict__ acc) {
+    const float* wrow = data + (size_t)row * cols;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int nch = cols >> 2;  // float4 chunks
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
+    for (int c = lane; c < nch; c += WAVE) {
+        const float4 w4 = *reinterpret_cast<const float4*>(wrow + c * 4);
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            const float* xb = x + (size_t)t * cols + c * 4;
+            float s = fmaf(w4.x, xb[0], 0.0f);
+            s = fmaf(w4.y, xb[1], s);
+            s = fmaf(w4.z, xb[2], s);
+            s = fmaf(w4.w, xb[3], s);
+            acc[t] += s;
+        }
+    }
+#pragma unroll
+    for (int t = 0; t < TMAX; ++t)
+        if (t < T) acc[t] = wave_reduce_sum(acc[t]);
+}
+
+template <int WT, int TMAX>
+__device__ __forceinline__ void wave_row_dot(
+    const WMat& w, int row, const float* __restrict__ x, int T,
+    float* __restrict__ acc) {
+    if (WT == W_Q4_0)
+        wave_row_dot_q4_0<TMAX>((const __half*)w.scales, (const uint8_t*)w.data,
+                                row, w.cols, x, T, acc);
+    else if (WT == W_Q4_1)
+        wave_row_dot_q4_1<TMAX>((const __half*)w.scales, (const uint8_t*)w.data,
+                                row, w.cols, x, T, acc);
+    else if (WT == W_F16)
+        wave_row_dot_f16<TMAX>((const __half*)w.data, row, w.cols, x, T, acc);
+    else
+        wave_row_dot_f32<TMAX>((const float*)w.data, row, w.cols, x, T, acc);
+}
+
+// ------------------------------------------------------------------ RMSNorm
+// y[t] = x[t] * rsqrt(mean(x[t]^2) + eps) * w     (K2 of SURVEY §2.5;
+// reference: ggml_rms_norm + ggml_mul, tensor_processor.cpp:555-570)
+
+__global__ void k_rmsnorm(const float* __restrict__ x,
+                          const float* __restrict__ w,
+                          float* __restrict__ y, int E, float eps) {
+    const int t = blockIdx.x;
+    const float* xt = x + (size_t)t * E;
+    float* yt = y + (size_t)t * E;
+    __shared__ float red[NWAVES];
+
+    float ss = 0.0f;
+    for (int i = threadIdx.x; i < (E >> 2); i += BLOCK) {
+        const float4 v = reinterpret_cast<const float4*>(xt)[i];
+        ss += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
+    }
+    ss = wave_reduce_sum(ss);
+    const int wid = threadIdx.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[wid] = ss;
+    __syncthreads();
+    float total = 0.0f;
+#pragma unroll
+    for (int i = 0; i < NWAVES; ++i) total += red[i];
+    const float scale = rsqrtf(total / (float)E + eps);
+
+    for (int i = threadIdx.x; i < (E >> 2); i += BLOCK) {
+        const float4 v = reinterpret_cast<const float4*>(xt)[i];
+        const float4 g = reinterpret_cast<const float4*>(w)[i];
+        float4 o;
+        o.x = v.x * scale * g.x;
+        o.y = v.y * scale * g.y;
+        o.z = v.z * scale * g.z;
+        o.w = v.w * scale * g.w;
+        reinterpret_cast<float4*>(yt)[i] = o;
+    }
+}
+
+// ------------------------------------------------- QKV + RoPE + KV append
+// Fuses K3+K4+K5 of SURVEY §2.5 (reference: ggml_mul_mat wq/wk/wv +
+// ggml_rope_inplace + ggml_cpy into cache, tensor_processor.cpp:579-622).
+// Grid: (3E/4) blocks; block computes 4 consecutive rows of one of
+// {wq, wk, wv}. RoPE (GGML mode 0: rotate adjacent pairs (2i, 2i+1) within
+// each head, theta = pos * base^(-2i/D)) is applied in the epilogue to q and
+// k rows; k/v rows are converted to f16 and appended to the cache at
+// [seq[t]][pos[t]].
+
+template <int WT, int TMAX>
+__global__ void k_qkv_rope_append(
+    WMat wq, WMat wk, WMat wv, const float* __restrict__ xn,
+    float* __restrict__ q_buf, __half* __restrict__ k_cache,
+    __half* __restrict__ v_cache, const int* __restrict__ pos,
+    const int* __restrict__ seq, const float* __restrict__ inv_freq,
+    int E, int D, int n_ctx, int T) {
+    const int r_global = blockIdx.x * 4;  // in [0, 3E)
+    const int mat = r_global / E;         // 0=q 1=k 2=v
+    const int row0 = r_global % E;
+    const int wid = threadIdx.x / WAVE;
+    const int row = row0 + wid;
+
+    float acc[TMAX];
+    const WMat& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
+    wave_row_dot<WT, TMAX>(w, row, xn, T, acc);
+
+    __shared__ float ybuf[4][TMAX];
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t)
+            if (t < T) ybuf[wid][t] = acc[t];
+    }
+    __syncthreads();
+
+    if (mat == 2) {
+        // v rows: straight f16 append, thread handles (row r, token t)
+        const int idx = threadIdx.x;
+        if (idx < 4 * T) {
+            const int rr = idx / T, t = idx % T;
+            const int e = row0 + rr;
+            __half* dst = v_cache +
+                ((size_t)seq[t] * n_ctx + pos[t]) * E + e;
+            *dst = __float2half(ybuf[rr][t]);
+        }
+        return;
+    }
+    // q/k rows: rope on pairs (row0+2p, row0+2p+1)
+    const int idx = threadIdx.x;
+    if (idx < 2 * T) {
+        const int p = idx / T, t = idx % T;
+        const int e = row0 + 2 * p;
+        const int d = e % D;
+        const float theta = (float)pos[t] * inv_freq[d >> 1];
+        float c, s;
+        __sincosf(theta, &s, &c);
+        const float x0 = ybuf[2 * p][t], x1 = ybuf[2 * p + 1][t];
+        const float o0 = x0 * c - x1 * s;
+        const float o1 = x0 * s + x1 * c;
+        if (mat == 0) {
+            q_buf[(size_t)t * E + e] = o0;
+            q_buf[(size_t)t * E + e + 1] = o1;
+        } else {
+            __half* dst = k_cache +
+                ((size_t)seq[t] * n_ctx + pos[t]) * E + e;
+            dst[0] = __float2half(o0);
+            dst[1] = __float2half(o1);
+        }
+    }
+}
+
+// ------------------------------------------------------ streaming attention
+// Fuses K6+K7+K8 of SURVEY §2.5 (reference: ggml_mul_mat KxQ + scale +
+// diag_mask_inf + soft_max + VxP, tensor_processor.cpp:645-700) into one
+// online-softmax kernel: O(1) memory in context length, never materializes
+// the [P+N, N, H] score tensor. Grid: (T, H); token t attends over cache
+// entries [0, pos[t]] of sequence seq[t] — causality for prefill rows comes
+// from their per-row positions.
+
+__global__ void k_attention(
+    const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
+    const __half* __restrict__ v_cache, float* __restrict__ out,
+    const int* __restrict__ pos, const int* __restrict__ seq,
+    int E, int D, int n_ctx) {
+    const int t = blockIdx.x;
+    const int h = blockIdx.y;
+    const int J = pos[t] + 1;
+    const size_t base = (size_t)seq[t] * n_ctx * E + h * D;
+    const float inv_sqrt_d = rsqrtf((float)D);
+
+    extern __shared__ float smem[];
+    float* lds_q = smem;          // [D]
+    float* lds_p = smem + D;      // [BLOCK]
+    float* lds_red = lds_p + BLOCK;  // [NWAVES]
+
+    for (int d = threadIdx.x; d < D; d += BLOCK)
+        lds_q[d] = q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d;
+    __syncthreads();
+
+    float m = -INFINITY;  // running max (block-uniform)
+    float l = 0.0f;       // running denom (block-uniform)
+    float o = 0.0f;       // thread d accumulator (threads >= D idle)
+
+    for (int j0 = 0; j0 < J; j0 += BLOCK) {
+        const int jj = j0 + threadIdx.x;
+        float s = -INFINITY;
+        if (jj < J) {
+            const __half* krow = k_cache + base + (size_t)jj * E;
+            float acc = 0.0f;
+            const int nh2 = D >> 2;  // 4 halves per float2 load
+            const float2* k2 = reinterpret_cast<const float2*>(krow);
+            for (int c = 0; c < nh2; ++c) {
+                const float2 raw = k2[c];
+                const __half2* hh = reinterpret_cast<const __half2*>(&raw);
+                const float2 a = __half22float2(hh[0]);
+                const float2 b = __half22float2(hh[1]);
+                acc = fmaf(a.x, lds_q[c * 4 + 0], acc);
+                acc = fmaf(a.y, lds_q[c * 4 + 1], acc);
+                acc = fmaf(b.x, lds_q[c * 4 + 2], acc);
+                acc = fmaf(b.y, lds_q[c * 4 + 3], acc);
+            }
+            s = acc;
+        }
+        // block max
+        float wm = wave_reduce_max(s);
+        const int wid = threadIdx.x / WAVE;
+        if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = wm;
+        __syncthreads();
+        float cm = fmaxf(fmaxf(lds_red[0], lds_red[1]),
+                         fmaxf(lds_red[2], lds_red[3]));
+        const float m_new = fmaxf(m, cm);
+        const float alpha = (m == -INFINITY) ? 0.0f : __expf(m - m_new);
+        const float p = (jj < J) ? __expf(s - m_new) : 0.0f;
+        __syncthreads();  // lds_p reuse across chunks
+        lds_p[threadIdx.x] = p;
+        // block sum of p
+        float ws = wave_reduce_sum(p);
+        if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = ws;
+        __syncthreads();
+        const float sum_p = lds_red[0] + lds_red[1] + lds_red[2] + lds_red[3];
+        l = l * alpha + sum_p;
+        m = m_new;
+        // V accumulation: thread d sums p_j * V[j][d]
+        if (threadIdx.x < D) {
+            o *= alpha;
+            const int jmax = min(BLOCK, J - j0);
+            const __half* vcol = v_cache + base + (size_t)j0 * E + threadIdx.x;
+            int jc = 0;
+            for (; jc + 4 <= jmax; jc += 4) {
+                const float p0 = lds_p[jc], p1 = lds_p[jc + 1];
+                const float p2 = lds_p[jc + 2], p3 = lds_p[jc + 3];
+                const float v0 = __half2float(vcol[(size_t)jc * E]);
+                const float v1 = __half2float(vcol[(size_t)(jc + 1) * E]);
+                const float v2 = __half2float(vcol[(size_t)(jc + 2) * E]);
+                const float v3 = __half2float(vcol[(size_t)(jc + 3) * E]);
+                o = fmaf(p0, v0, o);
+                o = fmaf(p1, v1, o);
+                o = fmaf(p2, v2, o);
+                o = fmaf(p3, v3, o);
+            }
+            for (; jc < jmax; ++jc)
+                o = fmaf(lds_p[jc], __half2float(vcol[(size_t)jc * E]), o);
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x < D)
+        out[(size_t)t * E + h * D + threadIdx.x] = o / l;
+}
+
+// --------------------------------------------------------------- GEMV(+res)
+// K9/K10/K13/K14 of SURVEY §2.5: output projection / FFN down / lm_head,
+// with the residual add fused into the epilogue.
+
+template <int WT, int TMAX, bool RES>
+__global__ void k_gemv(WMat w, const float* __restrict__ x,
+                       const float* __restrict__ res,
+                       float* __restrict__ y, int T) {
+    const int row = blockIdx.x * 4 + threadIdx.x / WAVE;
+    float acc[TMAX];
+    wave_row_dot<WT, TMAX>(w, row, x, T, acc);
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            float v = acc[t];
+            if (RES) v += res[(size_t)t * w.rows + row];
+            y[(size_t)t * w.rows + row] = v;
+        }
+    }
+}
+
+// ------------------------------------------------------------ w1/w3 SwiGLU
+// K11+K12 of SURVEY §2.5 (reference: ggml_mul_mat w1/w3 + ggml_silu +
+// ggml_mul, tensor_processor.cpp:732-751): g = silu(w1·x) * (w3·x).
+
+template <int WT, int TMAX>
+__global__ void k_ffn_gate(WMat w1, WMat w3, const float* __restrict__ xn,
+                           float* __restrict__ g, int T) {
+    const int row = blockIdx.x * 4 + threadIdx.x / WAVE;
+    float a1[TMAX], a3[TMAX];
+    wave_row_dot<WT, TMAX>(w1, row, xn, T, a1);
+    wave_row_dot<WT, TMAX>(w3, row, xn, T, a3);
+    if ((threadIdx.x & (WAVE - 1)) == 0) {
+#pragma unroll
+        for (int t = 0; t < TMAX; ++t) {
+            if (t >= T) continue;
+            const float v1 = a1[t];
+            const float silu = v1 / (1.0f + __expf(-v1));
+            g[(size_t)t * w1.rows + row] = silu * a3[t];
+        }
+    }
+}
+
+// -------------------------------------------------------- embedding gather
+// K1 of SURVEY §2.5 (ggml_get_rows, tensor_processor.cpp:1767): dequantize
+// row tok[t] of the embedding table into f32 activations.
+
+template <int WT>
+__global__ void k_embed(WMat tab, const int* __restrict__ tokens,
+                        float* __restrict__ out, int E) {
+    const int t = blockIdx.x;
+    const int row = tokens[t];
+    float* dst = out + (size_t)t * E;
+    if (WT == W_Q4_0) {
+        const int nb = E >> 5;
+        const __half* srow = (const __half*)tab.scales + (size_t)row * nb;
+        const uint8_t* qrow = (const uint8_t*)tab.data + (size_t)row * nb * 16;
+        for (int b = threadIdx.x; b < nb; b += BLOCK) {
+            const float d = __half2float(srow[b]);
+            const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
+            const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int lo = (int)((w[i] >> (8 * j)) & 0xF) - 8;
+                    const int hi = (int)((w[i] >> (8 * j + 4)) & 0xF) - 8;
+                    dst[(b << 5) + i * 4 + j] = d * (float)lo;
+                    dst[(b << 5) + 16 + i * 4 + j] = d * (float)hi;
+                }
+            }
+        }
+    } else if (WT == W_Q4_1) {
+        const int nb = E >> 5;
+        const __half* srow = (const __half*)tab.scales + (size_t)row * nb * 2;
+        const uint8_t* qrow = (const uint8_t*)tab.data + (size_t)row * nb * 16;
+        for (int b = threadIdx.x; b < nb; b += BLOCK) {
+            const float d = __half2float(srow[b * 2]);
+            const float mm = __half2float(srow[b * 2 + 1]);
+            const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
+            const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+#pragma unroll
+            for (int i = 0; i < 4; ++i) {
+#pragma unroll
+                for (int j = 0; j < 4; ++j) {
+                    const int lo = (int)((w[i] >> (8 * j)) & 0xF);
+                    const int hi = (int)((w[i] >> (8 * j + 4)) & 0xF);
+                    dst[(b << 5) + i * 4 + j] = d * (float)lo + mm;
+                    dst[(b << 5) + 16 + i * 4 + j] = d * (float)hi + mm;
+                }
+            }
+        }
+    } else if (WT == W_F16) {
+        const __half* wrow = (const __half*)tab.data + (size_t)row * E;
+        for (int i = threadIdx.x; i < E; i += BLOCK)
+            dst[i] = __half2float(wrow[i]);
+    } else {
+        const float* wrow = (const float*)tab.data + (size_t)row * E;
+        for (int i = threadIdx.x; i < E; i += BLOCK) dst[i] = wrow[i];
+    }
+}
+
+// ------------------------------------------------------------------- argmax
+// K15 of SURVEY §2.5 (sample_next_token greedy argmax,
+// tensor_processor.cpp:1894-1908), on-device so greedy decode never copies
+// the [V] logits to host.
+
+__global__ void k_argmax(const float* __restrict__ logits,
+                         int* __restrict__ out, int V) {
+    const int t = blockIdx.x;
+    const float* lt = logits + (size_t)t * V;
+    float best = -INFINITY;
+    int bi = 0;
+    for (int i = threadIdx.x; i < V; i += BLOCK) {
+        const float v = lt[i];
+        if (v > best || (v == best && i < bi)) {
+            best = v;
+            bi = i;
+        }
+    }
+    __shared__ float sv[BLOCK];
+    __shared__ int si[BLOCK];
+    sv[threadIdx.x] = best;
+    si[threadIdx.x] = bi;
+    __syncthreads();
+    for (int stride = BLOCK / 2; stride > 0; stride >>= 1) {
+        if (threadIdx.x < stride) {
+            const float ov = sv[threadIdx.x + stride];
+            const int oi = si[threadIdx.x + stride];
+            if (ov > sv[threadIdx.x] ||
+                (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
+                sv[threadIdx.x] = ov;
+                si[threadIdx.x] = oi;
+            }
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) out[t] = si[0];
+}
+
+// ============================================================== launchers
+
+static inline int pick_tmax(int T) {
+    if (T <= 1) return 1;
+    if (T <= 2) return 2;
+    if (T <= 4) return 4;
+    if (T <= 8) return 8;
+    return 16;
+}
+
+void launch_rmsnorm(hipStream_t s, const float* x, const float* w, float* y,
+                    int T, int E, float eps) {
+    hipLaunchKernelGGL(k_rmsnorm, dim3(T), dim3(BLOCK), 0, s, x, w, y, E, eps);
+}
+
+#define DISPATCH_WT(WTV, ...)                    \
+    switch (WTV) {                               \
+        case W_Q4_0: {                           \
+            constexpr int WTc = W_Q4_0;          \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        case W_Q4_1: {                           \
+            constexpr int WTc = W_Q4_1;          \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        case W_F16: {                            \
+            constexpr int WTc = W_F16;           \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        default: {                               \
+            constexpr int WTc = W_F32;           \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+    }
+
+#define DISPATCH_TMAX(TMAXV, ...)        \
+    switch (TMAXV) {                     \
+        case 1: {                        \
+            constexpr int TMc = 1;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        case 2: {                        \
+            constexpr int TMc = 2;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        case 4: {                        \
+            constexpr int TMc = 4;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        case 8: {                        \
+            constexpr int TMc = 8;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        default: {                       \
+            constexpr int TMc = 16;      \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+    }
+
+void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
+                            const WMat& wv, const float* xn, float* q_buf,
+                            __half* k_cache_layer, __half* v_cache_layer,
+                            const int* pos, const int* seq,
+                            const float* inv_freq, int E, int D, int n_ctx,
+                            int T) {
+    const int tmax = pick_tmax(T);
+    const dim3 grid(3 * E / 4);
+    DISPATCH_WT(wq.wtype, DISPATCH_TMAX(
+        tmax, hipLaunchKernelGGL((k_qkv_rope_append<WTc, TMc>), grid,
+                                 dim3(BLOCK), 0, s, wq, wk, wv, xn, q_buf,
+                                 k_cache_layer, v_cache_layer, pos, seq,
+                                 inv_freq, E, D, n_ctx, T)));
+}
+
+void launch_attention(hipStream_t s, const float* q_buf,
+                      const __half* k_cache_layer,
+                      const __half* v_cache_layer, float* out, const int* pos,
+                      const int* seq, int T, int H, int E, int D, int n_ctx) {
+    const dim3 grid(T, H);
+    const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
+    hipLaunchKernelGGL(k_attention, grid, dim3(BLOCK), lds, s, q_buf,
+                       k_cache_layer, v_cache_layer, out, pos, seq, E, D,
+                       n_ctx);
+}
+
+void launch_gemv(hipStream_t s, const WMat& w, const float* x,
+                 const float* res, float* y, int T) {
+    const int tmax = pick_tmax(T);
+    const dim3 grid(w.rows / 4);
+    if (res != nullptr) {
+        DISPATCH_WT(w.wtype, DISPATCH_TMAX(
+            tmax, hipLaunchKernelGGL((k_gemv<WTc, TMc, true>), grid,
+                                     dim3(BLOCK), 0, s, w, x, res, y, T)));
+    } else {
+        DISPATCH_WT(w.wtype, DISPATCH_TMAX(
+            tmax, hipLaunchKernelGGL((k_gemv<WTc, TMc, false>), grid,
+                                     dim3(BLOCK), 0, s, w, x, res, y, T)));
+    }
+}
+
+void launch_ffn_gate(hipStream_t s, const WMat& w1, const WMat& w3,
+                     const float* xn, float* g, int T) {
+    const int tmax = pick_tmax(T);
+    const dim3 grid(w1.rows / 4);
+    DISPATCH_WT(w1.wtype, DISPATCH_TMAX(
+        tmax, hipLaunchKernelGGL((k_ffn_gate<WTc, TMc>), grid, dim3(BLOCK),
+                                 0, s, w1, w3, xn, g, T)));
+}
+
+void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
+                  float* out, int T, int E) {
+    switch (tab.wtype) {
+        case W_Q4_0:
+            hipLaunchKernelGGL((k_embed<W_Q4_0>), dim3(T), dim3(BLOCK), 0, s,
+                               tab, tokens, out, E);
+            break;
+        case W_Q4_1:
+            hipLaunchKernelGGL((k_embed<W_Q4_1>), dim3(T), dim3(BLOCK), 0, s,
+                               tab, tokens, out, E);
+            break;
+        case W_F16:
+            hipLaunchKernelGGL((k_embed<W_F16>), dim3(T), dim3(BLOCK), 0, s,
+                               tab, tokens, out, E);
+            break;
+        default:
+            hipLaunchKernelGGL((k_embed<W_F32>), dim3(T), dim3(BLOCK), 0, s,
+                               tab, tokens, out, E);
+            break;
+    }
+}
+
+void launch_argmax(hipStream_t s, const float* logits, int* out, int T,
+                   int V) {
+    hipLaunchKernelGGL(k_argmax, dim3(T), dim3(BLOCK), 0, s, logits, out, V);
+}
