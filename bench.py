@@ -1,0 +1,140 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: OpenLLaMA-3B q4_0 batched decode, layer-sliced
+across N MI355X GPUs (RCCL pipeline stages over xGMI).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W`; for N>1
+launched via torch.distributed.run with one rank per GPU. Rank 0 prints one
+JSON line with the whole-job tokens/sec.
+
+Metric: tokens generated per second across the whole node (every sequence
+advances one token per step; tokens/step = global_batch). Weak scaling:
+global batch = 4 sequences per pipeline stage, so per-GPU work per step is
+constant as N grows (each stage holds ~L/N layers but serves N micro-
+batches per step). Synthetic setup: random-init weights of the real
+architecture generated directly on-device in the engine's q4_0 layout
+(identical compute + HBM traffic to a real checkpoint; no network for real
+weights — BASELINE.md).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+import torch
+
+from distributedllm_amd.formats import ggml
+from distributedllm_amd.models.llama import PRESETS
+from distributedllm_amd.parallel.pipeline import (
+    DecodePipeline, PipelineConfig, partition_layers, timed_decode)
+
+FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
+          "f16": ggml.FTYPE_MOSTLY_F16, "f32": ggml.FTYPE_ALL_F32}
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=16)
+    ap.add_argument("--model", default="open_llama_3b")
+    ap.add_argument("--ftype", default="q4_0", choices=list(FTYPES))
+    ap.add_argument("--mbs", type=int, default=4,
+                    help="sequences per micro-batch")
+    ap.add_argument("--ctx", type=int, default=2048)
+    ap.add_argument("--seed", type=int, default=0)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(args.gpus, world)
+
+    import torch.distributed as dist
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    backend = "nccl" if device == "cuda" else "gloo"
+    if world > 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    if device == "cuda":
+        torch.cuda.set_device(local_rank)
+
+    preset = PRESETS[args.model]
+    hp = preset.hparams(FTYPES[args.ftype])
+    parts = partition_layers(preset.n_layer, world)
+    first, count = parts[rank]
+
+    # micro-batches in flight = number of stages (keeps every stage busy);
+    # weak scaling: global batch grows with N while per-GPU work is fixed.
+    n_mb = world
+    cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
+
+    if device == "cuda":
+        from distributedllm_amd.engine import HIPSliceEngine
+        eng = HIPSliceEngine.random(hp, n_layers=count, first_layer=first,
+                                    n_ctx=args.ctx,
+                                    max_batch=cfg.global_batch,
+                                    seed=args.seed, with_extra=True)
+    else:  # CPU fallback so the contract is testable without a GPU
+        from distributedllm_amd.engine import TorchSliceEngine
+        from distributedllm_amd.formats import synthetic
+        from distributedllm_amd.models.llama import weights_from_ggml, LlamaPreset
+        tiny = PRESETS["tiny"]
+        hp = tiny.hparams(FTYPES[args.ftype])
+        parts = partition_layers(tiny.n_layer, world)
+        first, count = parts[rank]
+        f = synthetic.build_model(tiny, seed=args.seed,
+                                  ftype=ggml.FTYPE_MOSTLY_F16)
+        w = weights_from_ggml(f)
+        eng = TorchSliceEngine(hp, w, n_layers=count, first_layer=first,
+                               n_ctx=args.ctx, max_batch=cfg.global_batch)
+        preset = tiny
+
+    pipe = DecodePipeline(eng, cfg, rank=rank, world=world)
+    elapsed = timed_decode(pipe, args.steps, args.warmup, device)
+
+    # MAX over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if backend == "nccl" else "cpu")
+    if world > 1:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    tokens = args.steps * cfg.global_batch
+    toks_per_s = tokens / elapsed
+    if rank == 0:
+        result = {
+            "metric": "tokens/sec (whole node), "
+                      f"{preset.name} {args.ftype} sliced across "
+                      f"{n_gpus} MI355X",
+            "value": round(toks_per_s, 2),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": args.ftype,
+            "data": "synthetic (random-init weights, random prompt ids)",
+            "config": {
+                "model": preset.name,
+                "global_batch": cfg.global_batch,
+                "seq_len": args.warmup + args.steps,
+                "n_ctx": args.ctx,
+                "parallelism": f"pp{n_gpus}",
+                "micro_batches": n_mb,
+                "mbs": args.mbs,
+                "device": device,
+            },
+        }
+        print(json.dumps(result), flush=True)
+    if world > 1:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

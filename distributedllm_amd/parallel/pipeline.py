@@ -1,0 +1,127 @@
+"""RCCL pipeline runner: layer-sliced decode across GPUs over xGMI.
+
+The MI355X-native replacement of the reference's activation hop
+(client-mediated TCP round-trip per node per token,
+/root/reference/distllm/cli_api/common.py:148-154 +
+control_center.py:224-244): here each pipeline stage is one process per GPU
+and the hop is a direct ``torch.distributed`` P2P send/recv (backend "nccl"
+== RCCL on ROCm) of the [mbs, E] activation block between adjacent ranks;
+the sampled token ids ride a tiny P2P message from the last stage back to
+rank 0. Micro-batches keep every stage busy (the reference pipeline is
+fully serialized — SURVEY §2.3).
+
+Also runs on CPU with the gloo backend (world_size > 1) for tests.
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.distributed as dist
+
+
+def partition_layers(n_layers: int, world: int) -> List[Tuple[int, int]]:
+    """Contiguous (first_layer, count) per rank; remainder to early ranks."""
+    base = n_layers // world
+    rem = n_layers % world
+    out = []
+    first = 0
+    for r in range(world):
+        cnt = base + (1 if r < rem else 0)
+        out.append((first, cnt))
+        first += cnt
+    return out
+
+
+@dataclass
+class PipelineConfig:
+    mbs: int = 4          # sequences per micro-batch
+    n_mb: int = 1         # micro-batches in flight
+    device: str = "cuda"
+
+    @property
+    def global_batch(self) -> int:
+        return self.mbs * self.n_mb
+
+
+class DecodePipeline:
+    """Synchronized batched decode across pipeline stages.
+
+    Every sequence advances one token per step. Stage r owns a contiguous
+    layer range; rank 0 embeds, the last rank samples (greedy argmax on
+    device) and feeds token ids back to rank 0.
+    """
+
+    def __init__(self, engine, cfg: PipelineConfig,
+                 rank: int = 0, world: int = 1):
+        self.engine = engine
+        self.cfg = cfg
+        self.rank = rank
+        self.world = world
+        self.is_first = rank == 0
+        self.is_last = rank == world - 1
+        dev = cfg.device
+        E = engine.hp.n_embd
+        M, mbs = cfg.n_mb, cfg.mbs
+        self.pos = [torch.zeros(mbs, dtype=torch.int32, device=dev)
+                    for _ in range(M)]
+        self.seq = [torch.arange(m * mbs, (m + 1) * mbs, dtype=torch.int32,
+                                 device=dev) for m in range(M)]
+        self.x_recv = [torch.empty(mbs, E, dtype=torch.float32, device=dev)
+                       for _ in range(M)]
+        self.tok = [torch.randint(3, engine.hp.n_vocab, (mbs,),
+                                  dtype=torch.int32, device=dev)
+                    for _ in range(M)]
+
+    def _advance_mb(self, m: int) -> None:
+        """Push micro-batch m one token forward through this stage."""
+        eng, cfg = self.engine, self.cfg
+        if self.is_first:
+            x = eng.embed(self.tok[m])
+        else:
+            dist.recv(self.x_recv[m], src=self.rank - 1)
+            x = self.x_recv[m]
+        y = eng.forward(x, self.pos[m], self.seq[m])
+        if not self.is_last:
+            dist.send(y.contiguous(), dst=self.rank + 1)
+        else:
+            lg = eng.logits(y, all_logits=True)
+            nxt = eng.argmax(lg)
+            if self.world > 1:
+                dist.send(nxt, dst=0)
+            else:
+                self.tok[m] = nxt
+        if self.is_first and self.world > 1:
+            # token ids for this micro-batch's next step come from the
+            # last stage; ordered FIFO per rank pair, so recv here pairs
+            # with the send above.
+            dist.recv(self.tok[m], src=self.world - 1)
+        self.pos[m] += 1
+
+    def run_steps(self, steps: int) -> None:
+        for _ in range(steps):
+            for m in range(self.cfg.n_mb):
+                self._advance_mb(m)
+
+    def current_tokens(self) -> torch.Tensor:
+        return torch.stack(self.tok)
+
+
+def timed_decode(pipe: DecodePipeline, steps: int, warmup: int,
+                 device: str) -> float:
+    """Barrier-bracketed timing of exactly `steps` steps; returns seconds
+    (this rank's wall time — reduce MAX across ranks for the job time)."""
+    pipe.run_steps(warmup)
+    if dist.is_initialized():
+        dist.barrier()
+    if device == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    pipe.run_steps(steps)
+    if dist.is_initialized():
+        dist.barrier()
+    if device == "cuda":
+        torch.cuda.synchronize()
+    return time.perf_counter() - t0
