@@ -48,6 +48,14 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
 // --------------------------------------------------------------- row x vec
 // One WAVE computes dot(row r of W, x_t) for T tokens (T <= TMAX).
 // x: [T][cols] f32 (row-major, ld = cols). Results valid in lane 0.
+//
+// Coalescing is the whole game here (decode GEMV = HBM-bound weight
+// stream): every lane covers 4 CONSECUTIVE weights, so the wave's x loads
+// are one contiguous 1 KiB float4 transaction, the q4 nibble loads one
+// contiguous 128 B u32 transaction, and the f16 scale loads one 16 B span.
+// (The naive block-per-lane layout put each lane's x on its own cache line
+// — 64 transactions per wave instruction — and measured 25x off the
+// bandwidth roofline; rocprof evidence in profiles/.)
 
 template <int TMAX>
 __device__ __forceinline__ void wave_row_dot_q4_0(
@@ -58,30 +66,37 @@ __device__ __forceinline__ void wave_row_dot_q4_0(
     const __half* srow = scales + (size_t)row * nb;
     const uint8_t* qrow = qs + (size_t)row * nb * 16;
     const int lane = threadIdx.x & (WAVE - 1);
+    const int bofs = lane >> 3;        // block within the 8-block group
+    const int p = (lane & 7) * 4;      // weight span [p, p+4) in the block
+    const int qoff = p & 15;           // byte offset of the span's u32
+    const int shift = (p >= 16) ? 4 : 0;  // high or low nibbles
 #pragma unroll
     for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
-    for (int b = lane; b < nb; b += WAVE) {
-        const float d = __half2float(srow[b]);
-        const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
-        const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+    const int ngroups = (nb + 7) >> 3;
+    for (int g = 0; g < ngroups; ++g) {
+        const int b = g * 8 + bofs;
+        float d = 0.0f;
+        uint32_t q = 0x88888888u;  // dequantizes to 0 with the -8 bias
+        if (b < nb) {
+            d = __half2float(srow[b]);
+            q = *reinterpret_cast<const uint32_t*>(qrow + b * 16 + qoff);
+        }
+        const float w0 = (float)((int)((q >> (shift + 0)) & 0xF) - 8);
+        const float w1 = (float)((int)((q >> (shift + 8)) & 0xF) - 8);
+        const float w2 = (float)((int)((q >> (shift + 16)) & 0xF) - 8);
+        const float w3 = (float)((int)((q >> (shift + 24)) & 0xF) - 8);
+        const int xoff = b * 32 + p;
 #pragma unroll
         for (int t = 0; t < TMAX; ++t) {
             if (t >= T) continue;
-            const float* xb = x + (size_t)t * cols + (b << 5);
-            float s = 0.0f;
-#pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const uint32_t q = w[i];
-                // byte j of the block: low nibble = weight j,
-                // high nibble = weight j+16 (GGJT v3 q4_0 layout)
-#pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const int lo = (int)((q >> (8 * j)) & 0xF) - 8;
-                    const int hi = (int)((q >> (8 * j + 4)) & 0xF) - 8;
-                    s = fmaf((float)lo, xb[i * 4 + j], s);
-                    s = fmaf((float)hi, xb[16 + i * 4 + j], s);
-                }
-            }
+            float4 xv = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (b < nb)
+                xv = *reinterpret_cast<const float4*>(
+                    x + (size_t)t * cols + xoff);
+            float s = fmaf(w0, xv.x, 0.0f);
+            s = fmaf(w1, xv.y, s);
+            s = fmaf(w2, xv.z, s);
+            s = fmaf(w3, xv.w, s);
             acc[t] = fmaf(d, s, acc[t]);
         }
     }
@@ -96,35 +111,44 @@ __device__ __forceinline__ void wave_row_dot_q4_1(
     int row, int cols, const float* __restrict__ x, int T,
     float* __restrict__ acc) {
     const int nb = cols >> 5;
-    const __half* srow = scales + (size_t)row * nb * 2;
+    const __half* srow = scales + (size_t)row * nb * 2;  // (d, m) pairs
     const uint8_t* qrow = qs + (size_t)row * nb * 16;
     const int lane = threadIdx.x & (WAVE - 1);
+    const int bofs = lane >> 3;
+    const int p = (lane & 7) * 4;
+    const int qoff = p & 15;
+    const int shift = (p >= 16) ? 4 : 0;
 #pragma unroll
     for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
-    for (int b = lane; b < nb; b += WAVE) {
-        const __half2 dm = *reinterpret_cast<const __half2*>(srow + b * 2);
-        const float d = __half2float(__low2half(dm));
-        const float m = __half2float(__high2half(dm));
-        const uint4 packed = *reinterpret_cast<const uint4*>(qrow + b * 16);
-        const uint32_t w[4] = {packed.x, packed.y, packed.z, packed.w};
+    const int ngroups = (nb + 7) >> 3;
+    for (int g = 0; g < ngroups; ++g) {
+        const int b = g * 8 + bofs;
+        float d = 0.0f, m = 0.0f;
+        uint32_t q = 0;
+        if (b < nb) {
+            const __half2 dm =
+                *reinterpret_cast<const __half2*>(srow + b * 2);
+            d = __half2float(__low2half(dm));
+            m = __half2float(__high2half(dm));
+            q = *reinterpret_cast<const uint32_t*>(qrow + b * 16 + qoff);
+        }
+        const float w0 = (float)((q >> (shift + 0)) & 0xF);
+        const float w1 = (float)((q >> (shift + 8)) & 0xF);
+        const float w2 = (float)((q >> (shift + 16)) & 0xF);
+        const float w3 = (float)((q >> (shift + 24)) & 0xF);
+        const int xoff = b * 32 + p;
 #pragma unroll
         for (int t = 0; t < TMAX; ++t) {
             if (t >= T) continue;
-            const float* xb = x + (size_t)t * cols + (b << 5);
-            float s = 0.0f;   // sum q*x
-            float sx = 0.0f;  // sum x
-#pragma unroll
-            for (int i = 0; i < 4; ++i) {
-                const uint32_t q = w[i];
-#pragma unroll
-                for (int j = 0; j < 4; ++j) {
-                    const float x0 = xb[i * 4 + j];
-                    const float x1 = xb[16 + i * 4 + j];
-                    s = fmaf((float)((q >> (8 * j)) & 0xF), x0, s);
-                    s = fmaf((float)((q >> (8 * j + 4)) & 0xF), x1, s);
-                    sx += x0 + x1;
-                }
-            }
+            float4 xv = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (b < nb)
+                xv = *reinterpret_cast<const float4*>(
+                    x + (size_t)t * cols + xoff);
+            float s = fmaf(w0, xv.x, 0.0f);
+            s = fmaf(w1, xv.y, s);
+            s = fmaf(w2, xv.z, s);
+            s = fmaf(w3, xv.w, s);
+            const float sx = (xv.x + xv.y) + (xv.z + xv.w);
             acc[t] += d * s + m * sx;
         }
     }
@@ -139,26 +163,32 @@ __device__ __forceinline__ void wave_row_dot_f16(
     const float* __restrict__ x, int T, float* __restrict__ acc) {
     const __half* wrow = data + (size_t)row * cols;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int nch = cols >> 3;  // 8 halves (16 B) per chunk
 #pragma unroll
     for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
-    for (int c = lane; c < nch; c += WAVE) {
-        const uint4 packed = *reinterpret_cast<const uint4*>(wrow + c * 8);
-        const __half2* h2 = reinterpret_cast<const __half2*>(&packed);
-        float wv[8];
-#pragma unroll
-        for (int i = 0; i < 4; ++i) {
-            const float2 f = __half22float2(h2[i]);
-            wv[2 * i] = f.x;
-            wv[2 * i + 1] = f.y;
+    // 4 consecutive halves per lane -> 256 weights per wave iteration;
+    // w loads 8 B/lane (512 B contiguous), x loads 16 B/lane (1 KiB).
+    const int niter = (cols + 255) >> 8;
+    for (int g = 0; g < niter; ++g) {
+        const int c = g * 256 + lane * 4;
+        float w0 = 0.f, w1 = 0.f, w2 = 0.f, w3 = 0.f;
+        if (c < cols) {
+            const uint2 raw = *reinterpret_cast<const uint2*>(wrow + c);
+            const __half2* h2 = reinterpret_cast<const __half2*>(&raw);
+            const float2 a = __half22float2(h2[0]);
+            const float2 b = __half22float2(h2[1]);
+            w0 = a.x; w1 = a.y; w2 = b.x; w3 = b.y;
         }
 #pragma unroll
         for (int t = 0; t < TMAX; ++t) {
             if (t >= T) continue;
-            const float* xb = x + (size_t)t * cols + c * 8;
-            float s = 0.0f;
-#pragma unroll
-            for (int i = 0; i < 8; ++i) s = fmaf(wv[i], xb[i], s);
+            float4 xv = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (c < cols)
+                xv = *reinterpret_cast<const float4*>(
+                    x + (size_t)t * cols + c);
+            float s = fmaf(w0, xv.x, 0.0f);
+            s = fmaf(w1, xv.y, s);
+            s = fmaf(w2, xv.z, s);
+            s = fmaf(w3, xv.w, s);
             acc[t] += s;
         }
     }
@@ -173,19 +203,25 @@ __device__ __forceinline__ void wave_row_dot_f32(
     const float* __restrict__ x, int T, float* __restrict__ acc) {
     const float* wrow = data + (size_t)row * cols;
     const int lane = threadIdx.x & (WAVE - 1);
-    const int nch = cols >> 2;  // float4 chunks
 #pragma unroll
     for (int t = 0; t < TMAX; ++t) acc[t] = 0.0f;
-    for (int c = lane; c < nch; c += WAVE) {
-        const float4 w4 = *reinterpret_cast<const float4*>(wrow + c * 4);
+    const int niter = (cols + 255) >> 8;
+    for (int g = 0; g < niter; ++g) {
+        const int c = g * 256 + lane * 4;
+        float4 w4 = make_float4(0.f, 0.f, 0.f, 0.f);
+        if (c < cols)
+            w4 = *reinterpret_cast<const float4*>(wrow + c);
 #pragma unroll
         for (int t = 0; t < TMAX; ++t) {
             if (t >= T) continue;
-            const float* xb = x + (size_t)t * cols + c * 4;
-            float s = fmaf(w4.x, xb[0], 0.0f);
-            s = fmaf(w4.y, xb[1], s);
-            s = fmaf(w4.z, xb[2], s);
-            s = fmaf(w4.w, xb[3], s);
+            float4 xv = make_float4(0.f, 0.f, 0.f, 0.f);
+            if (c < cols)
+                xv = *reinterpret_cast<const float4*>(
+                    x + (size_t)t * cols + c);
+            float s = fmaf(w4.x, xv.x, 0.0f);
+            s = fmaf(w4.y, xv.y, s);
+            s = fmaf(w4.z, xv.z, s);
+            s = fmaf(w4.w, xv.w, s);
             acc[t] += s;
         }
     }
