@@ -28,6 +28,73 @@ from ..formats import ggml, q4
 from ..models.llama import RMS_EPS, ROPE_BASE, rms_norm, rope_interleaved
 
 
+def _f32_to_bf16_u16(a: np.ndarray) -> np.ndarray:
+    """Round-to-nearest-even f32 -> bf16 bit pattern (u16)."""
+    u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
+    rounded = u + 0x7FFF + ((u >> 16) & 1)
+    return (rounded >> 16).astype(np.uint16)
+
+
+def _nibbles(t: ggml.GGMLTensor) -> np.ndarray:
+    """q4 tensor -> raw nibble values [rows, nb, 32] (u8, order = weights)."""
+    rows, cols = t.shape_rows_cols
+    nb = cols // 32
+    bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
+    a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
+    qs = a[:, :, bs - 16:]
+    lo = qs & 0x0F
+    hi = qs >> 4
+    return np.concatenate([lo, hi], axis=-1)  # weight j / j+16 layout
+
+
+def repack_mfma(t: ggml.GGMLTensor, device: str):
+    """On-disk tensor -> (data, scales, wtype) in the MFMA tile layout.
+
+    q4_0/q4_1: data u32[R][nb][4 words][16 rows] where word ws of a block
+    packs its 8 weights at bit positions (j%2)*16 + (j//2)*4 — the exact
+    arrangement the kernel's three OR/AND/SHR turn into a bf16x8 A-fragment
+    (kernels.hip a_frag_q4). scales f32[R][nb][16] (q4_1: (d,m) pairs).
+    f16: weights converted to bf16 tiles [R][cols/8][16 rows][8].
+    """
+    rows, cols = t.shape_rows_cols
+    assert rows % 16 == 0 and cols % 32 == 0, (t.name, rows, cols)
+    R = rows // 16
+    if t.gtype in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+        nb = cols // 32
+        n = _nibbles(t).astype(np.uint32).reshape(rows, nb, 4, 8)
+        qword = np.zeros((rows, nb, 4), dtype=np.uint32)
+        for j in range(8):
+            qword |= n[:, :, :, j] << ((j % 2) * 16 + (j // 2) * 4)
+        qs2 = np.ascontiguousarray(
+            qword.reshape(R, 16, nb, 4).transpose(0, 2, 3, 1))
+        bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
+        a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
+        if t.gtype == ggml.GGML_TYPE_Q4_0:
+            d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
+            d = d.reshape(rows, nb).astype(np.float32)
+            scales = np.ascontiguousarray(
+                d.reshape(R, 16, nb).transpose(0, 2, 1))
+        else:
+            dm = np.ascontiguousarray(a[:, :, :4]).view(np.float16)
+            dm = dm.reshape(rows, nb, 2).astype(np.float32)
+            scales = np.ascontiguousarray(
+                dm.reshape(R, 16, nb, 2).transpose(0, 2, 1, 3))
+        data = torch.from_numpy(qs2.view(np.int32)).to(device)
+        sc = torch.from_numpy(scales).to(device)
+        return data, sc, t.gtype
+    if t.gtype == ggml.GGML_TYPE_F16:
+        w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
+        bf = _f32_to_bf16_u16(w.astype(np.float32))
+        tile = np.ascontiguousarray(
+            bf.reshape(R, 16, cols // 8, 8).transpose(0, 2, 1, 3))
+        data = torch.from_numpy(tile.view(np.int16)).to(device)
+        return data, torch.empty(0), ggml.GGML_TYPE_F16
+    # f32: legacy scalar path, plain [rows, cols]
+    a = np.frombuffer(t.raw, np.float32).reshape(rows, cols)
+    return (torch.from_numpy(a.copy()).to(device), torch.empty(0),
+            ggml.GGML_TYPE_F32)
+
+
 def _repack_q4(t: ggml.GGMLTensor):
     rows, cols = t.shape_rows_cols
     nb = cols // 32
@@ -104,19 +171,25 @@ class HIPSliceEngine:
         wt = ggml._FTYPE_TO_GGML[hp.ftype]
 
         def mat(rows: int, cols: int):
+            # random weights directly in the MFMA tile layouts — any random
+            # bit pattern is a valid q4 nibble word, so this is byte-for-
+            # byte the same compute/HBM traffic as a real checkpoint
             if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
-                nb = cols // 32
+                R, nb = rows // 16, cols // 32
                 per = 2 if wt == ggml.GGML_TYPE_Q4_1 else 1
-                data = torch.randint(0, 256, (rows, nb * 16),
-                                     dtype=torch.uint8, device="cuda",
+                data = torch.randint(-2**31, 2**31 - 1, (R * nb * 64,),
+                                     dtype=torch.int32, device="cuda",
                                      generator=g)
-                scales = ((torch.rand(rows, nb * per, device="cuda",
-                                      generator=g) * 0.5 + 0.75) *
-                          0.003).to(torch.float16)
-                return data, scales, wt
-            dt = torch.float16 if wt == ggml.GGML_TYPE_F16 else torch.float32
+                scales = ((torch.rand(R * nb * 16 * per, device="cuda",
+                                      generator=g) * 0.5 + 0.75) * 0.003)
+                return data, scales.contiguous(), wt
+            if wt == ggml.GGML_TYPE_F16:
+                data = (torch.randn(rows * cols, device="cuda", generator=g,
+                                    dtype=torch.float32) * 0.02)
+                return (data.to(torch.bfloat16).view(torch.int16),
+                        torch.empty(0), wt)
             data = (torch.randn(rows, cols, device="cuda", generator=g,
-                                dtype=torch.float32) * 0.02).to(dt)
+                                dtype=torch.float32) * 0.02)
             return data.contiguous(), torch.empty(0), wt
 
         def norm_w(n: int):
@@ -128,7 +201,25 @@ class HIPSliceEngine:
             mats = [mat(r, c) for r, c in shapes]
             eng._eng.set_layer(li, norm_w(E), norm_w(E), mats)
         if with_extra:
-            tok_d, tok_s, tok_t = mat(V, E)
+            # the embedding table uses the legacy SoA layout (gather kernel)
+            if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+                nb = E // 32
+                per = 2 if wt == ggml.GGML_TYPE_Q4_1 else 1
+                tok_d = torch.randint(0, 256, (V, nb * 16),
+                                      dtype=torch.uint8, device="cuda",
+                                      generator=g)
+                tok_s = ((torch.rand(V, nb * per, device="cuda",
+                                     generator=g) * 0.5 + 0.75) *
+                         0.003).to(torch.float16)
+                tok_t = wt
+            elif wt == ggml.GGML_TYPE_F16:
+                tok_d = (torch.randn(V, E, device="cuda", generator=g) *
+                         0.02).to(torch.float16)
+                tok_s, tok_t = torch.empty(0), wt
+            else:
+                tok_d = (torch.randn(V, E, device="cuda", generator=g) *
+                         0.02)
+                tok_s, tok_t = torch.empty(0), wt
             out_d, out_s, out_t = mat(V, E)
             eng._eng.set_extra(tok_d, tok_s, tok_t, norm_w(E), out_d, out_s,
                                out_t, V)
@@ -149,14 +240,16 @@ class HIPSliceEngine:
                        "attention.wv.weight", "attention.wo.weight",
                        "feed_forward.w1.weight", "feed_forward.w2.weight",
                        "feed_forward.w3.weight"):
-                mats.append(_upload_mat(tm[pre + nm], self.device))
+                mats.append(repack_mfma(tm[pre + nm], self.device))
             self._eng.set_layer(li, attn_norm, ffn_norm, mats)
 
     def attach_extra(self, f: ggml.GGMLFile) -> None:
         tm = f.tensor_map()
+        # embedding table: legacy SoA layout (row-gather kernel);
+        # lm_head: MFMA tiles (it is a GEMV like any other)
         tok_d, tok_s, tok_t = _upload_mat(tm["tok_embeddings.weight"],
                                           self.device)
-        out_d, out_s, out_t = _upload_mat(tm["output.weight"], self.device)
+        out_d, out_s, out_t = repack_mfma(tm["output.weight"], self.device)
         norm = torch.from_numpy(tm["norm.weight"].to_f32()).to(self.device)
         self._eng.set_extra(tok_d, tok_s, tok_t, norm, out_d, out_s, out_t,
                             self.hp.n_vocab)

@@ -4,15 +4,17 @@
 // (/root/reference/distllm/tensor_processor.cpp:2238-2260 exposed 9 functions
 // around a global TransformerSlice). This engine instead:
 //   * is an explicit object (no global mutable slice, SURVEY.md §5.2 hazard),
-//   * keeps weights resident in HBM3E in a repacked SoA layout,
+//   * keeps weights resident in HBM3E in MFMA-tiled layouts (kernels.hip),
 //   * exchanges activations as torch tensors (DLPack-compatible device
 //     buffers), never per-element Python lists,
 //   * is stateless w.r.t. generation: positions/sequence ids are explicit
 //     device tensors, so the decode step is hipGraph-capturable and the KV
 //     "clear_context" is a host-side position reset.
 //
-// The forward pass launches the CDNA4 kernels of kernels.hip per layer on
-// the current torch HIP stream.
+// Decode layer = 5 fused kernels (QKV+RoPE+KV-append, attention, WO+res+
+// sumsq, FFN-gate+SwiGLU, W2+res+sumsq) — RMSNorm rides the sumsq
+// side-channel into the consumers' B-fragment builds. f32-weight models
+// (test configs) take a legacy scalar path with explicit RMSNorm kernels.
 
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
@@ -25,48 +27,71 @@
 
 namespace {
 
-struct DevMat {
-    torch::Tensor data;    // u8 nibbles (q4_*) or f16/f32 values
-    torch::Tensor scales;  // f16 scales (q4_*), undefined otherwise
+constexpr int kMaxTokens = 16;  // per-forward token-tile cap (MFMA N)
+
+struct DevMat {  // legacy scalar-path matrix (W_F32 models)
+    torch::Tensor data, scales;
     WMat w{};
 };
 
-struct Layer {
-    torch::Tensor attn_norm, ffn_norm;  // f32 [E]
-    DevMat wq, wk, wv, wo, w1, w2, w3;
+struct DevMat2 {  // MFMA-tiled matrix
+    torch::Tensor data, scales;
+    WMat2 w{};
 };
 
-constexpr int kMaxTokens = 16;  // per-forward token-tile cap (TMAX ceiling)
+struct Layer {
+    torch::Tensor attn_norm, ffn_norm;           // f32 [E] (legacy path)
+    torch::Tensor attn_normprep, ffn_normprep;   // bf16 [E] (MFMA path)
+    // legacy (f32 models)
+    DevMat wq, wk, wv, wo, w1, w2, w3;
+    // MFMA path
+    DevMat2 mq, mk, mv, mo, m1, m2, m3;
+    bool mfma = false;
+};
 
-DevMat make_devmat(torch::Tensor data, torch::Tensor scales, int64_t wtype,
-                   int64_t rows, int64_t cols) {
+DevMat make_devmat_f32(torch::Tensor data, int64_t rows, int64_t cols) {
+    TORCH_CHECK(data.is_cuda() && data.is_contiguous() &&
+                    data.scalar_type() == torch::kFloat32 &&
+                    data.numel() == rows * cols,
+                "f32 weight tensor mismatch");
+    DevMat m;
+    m.data = data;
+    m.w = WMat{data.data_ptr(), nullptr, (int)rows, (int)cols, W_F32};
+    return m;
+}
+
+DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
+                     int64_t rows, int64_t cols) {
     TORCH_CHECK(data.is_cuda() && data.is_contiguous(),
                 "weight data must be contiguous on device");
-    DevMat m;
+    TORCH_CHECK(rows % 16 == 0, "rows must be a multiple of 16");
+    TORCH_CHECK(cols % 32 == 0, "cols must be a multiple of 32");
+    const int64_t R = rows / 16, nb = cols / 32;
+    DevMat2 m;
     m.data = data;
     m.w.rows = (int)rows;
     m.w.cols = (int)cols;
     m.w.wtype = (int)wtype;
     m.w.data = data.data_ptr();
     if (wtype == W_Q4_0 || wtype == W_Q4_1) {
+        TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
+                        data.numel() == R * nb * 64,
+                    "q4 tiled data must be u32[R][nb][4][16]");
+        const int64_t per = (wtype == W_Q4_1) ? 2 : 1;
         TORCH_CHECK(scales.defined() && scales.is_cuda() &&
                         scales.is_contiguous() &&
-                        scales.scalar_type() == torch::kFloat16,
-                    "q4 weights need f16 scales on device");
-        const int nb = (int)(cols / 32);
-        const int64_t per_block = (wtype == W_Q4_1) ? 2 : 1;
-        TORCH_CHECK(scales.numel() == rows * nb * per_block,
-                    "scales size mismatch");
-        TORCH_CHECK(data.numel() == rows * nb * 16 &&
-                        data.scalar_type() == torch::kUInt8,
-                    "q4 nibble data size mismatch");
+                        scales.scalar_type() == torch::kFloat32 &&
+                        scales.numel() == R * nb * 16 * per,
+                    "q4 tiled scales must be f32[R][nb][16]",
+                    (wtype == W_Q4_1 ? "x2" : ""));
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {
-        const auto want =
-            (wtype == W_F16) ? torch::kFloat16 : torch::kFloat32;
-        TORCH_CHECK(data.scalar_type() == want, "weight dtype mismatch");
-        TORCH_CHECK(data.numel() == rows * cols, "weight size mismatch");
+        TORCH_CHECK(wtype == W_F16, "unsupported tiled wtype");
+        TORCH_CHECK(data.scalar_type() == torch::kInt16 ||
+                        data.scalar_type() == torch::kBFloat16,
+                    "f16 weights must arrive as bf16 tiles");
+        TORCH_CHECK(data.numel() == rows * cols, "bf16 tile size mismatch");
         m.w.scales = nullptr;
     }
     return m;
@@ -86,6 +111,10 @@ torch::Tensor check_i32(const torch::Tensor& t, const char* name) {
     return t;
 }
 
+unsigned short* u16p(torch::Tensor& t) {
+    return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
 class SliceEngine {
  public:
     SliceEngine(int64_t n_embd, int64_t n_head, int64_t n_layers,
@@ -102,25 +131,33 @@ class SliceEngine {
         TORCH_CHECK(E_ % H_ == 0, "n_embd not divisible by n_head");
         TORCH_CHECK(D_ % 2 == 0, "head_dim must be even for RoPE pairs");
         TORCH_CHECK(D_ <= 256, "head_dim > 256 unsupported");
+        TORCH_CHECK(E_ % 16 == 0 && F_ % 16 == 0, "E/F must be 16-aligned");
         layers_.resize(L_);
         loaded_.assign(L_, false);
         auto dev = torch::TensorOptions().device(torch::kCUDA);
         auto f32 = dev.dtype(torch::kFloat32);
         auto f16 = dev.dtype(torch::kFloat16);
+        auto u16 = dev.dtype(torch::kInt16);
         // KV cache: [L, B, ctx, E] f16 — the HBM3E-resident analog of the
         // reference's kv_cache_init (tensor_processor.cpp:1089-1132).
         k_cache_ = torch::zeros({L_, B_, ctx_, E_}, f16);
         v_cache_ = torch::zeros({L_, B_, ctx_, E_}, f16);
         // RoPE pair frequencies: theta_i = pos * base^(-2i/D)
-        auto freqs = torch::pow(
+        inv_freq_ = torch::pow(
             (float)rope_base,
-            torch::arange(0, D_ / 2, dev.dtype(torch::kFloat32)) *
-                (-2.0f / (float)D_));
-        inv_freq_ = freqs.contiguous();
+            torch::arange(0, D_ / 2, f32) * (-2.0f / (float)D_))
+            .contiguous();
         xn_ = torch::empty({kMaxTokens, E_}, f32);
         qb_ = torch::empty({kMaxTokens, E_}, f32);
         ab_ = torch::empty({kMaxTokens, E_}, f32);
         ffb_ = torch::empty({kMaxTokens, F_}, f32);
+        // MFMA-path side channels
+        xprep_ = torch::empty({(int64_t)E_ * kMaxTokens}, u16);
+        aprep_ = torch::empty({(int64_t)E_ * kMaxTokens}, u16);
+        gprep_ = torch::empty({(int64_t)F_ * kMaxTokens}, u16);
+        ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * kMaxTokens}, f32);
+        ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
+        ss_tmp_ = torch::zeros({kMaxTokens}, f32);
     }
 
     void set_layer(int64_t li, torch::Tensor attn_norm,
@@ -130,14 +167,28 @@ class SliceEngine {
         Layer& l = layers_[li];
         l.attn_norm = check_f32(attn_norm, "attn_norm");
         l.ffn_norm = check_f32(ffn_norm, "ffn_norm");
-        DevMat* slots[7] = {&l.wq, &l.wk, &l.wv, &l.wo, &l.w1, &l.w2, &l.w3};
+        l.attn_normprep = attn_norm.to(torch::kBFloat16).contiguous();
+        l.ffn_normprep = ffn_norm.to(torch::kBFloat16).contiguous();
         const int64_t rows[7] = {E_, E_, E_, E_, F_, E_, F_};
         const int64_t cols[7] = {E_, E_, E_, E_, E_, F_, E_};
+        auto first = mats[0].cast<py::tuple>();
+        const int64_t wt0 = first[2].cast<int64_t>();
+        l.mfma = (wt0 != W_F32);
+        DevMat* slots[7] = {&l.wq, &l.wk, &l.wv, &l.wo, &l.w1, &l.w2, &l.w3};
+        DevMat2* slots2[7] = {&l.mq, &l.mk, &l.mv, &l.mo, &l.m1, &l.m2,
+                              &l.m3};
         for (size_t i = 0; i < 7; ++i) {
             auto tup = mats[i].cast<py::tuple>();
-            *slots[i] = make_devmat(tup[0].cast<torch::Tensor>(),
-                                    tup[1].cast<torch::Tensor>(),
-                                    tup[2].cast<int64_t>(), rows[i], cols[i]);
+            const int64_t wt = tup[2].cast<int64_t>();
+            TORCH_CHECK((wt == W_F32) == !l.mfma,
+                        "all matrices of a layer must share the path");
+            if (l.mfma)
+                *slots2[i] = make_devmat2(tup[0].cast<torch::Tensor>(),
+                                          tup[1].cast<torch::Tensor>(), wt,
+                                          rows[i], cols[i]);
+            else
+                *slots[i] = make_devmat_f32(tup[0].cast<torch::Tensor>(),
+                                            rows[i], cols[i]);
         }
         loaded_[li] = true;
     }
@@ -147,9 +198,25 @@ class SliceEngine {
                    torch::Tensor out_data, torch::Tensor out_scales,
                    int64_t out_wtype, int64_t n_vocab) {
         V_ = (int)n_vocab;
-        tok_ = make_devmat(tok_data, tok_scales, tok_wtype, V_, E_);
-        out_ = make_devmat(out_data, out_scales, out_wtype, V_, E_);
         final_norm_ = check_f32(norm_w, "norm_w");
+        final_normprep_ = norm_w.to(torch::kBFloat16).contiguous();
+        // the embedding table always arrives in the legacy SoA layout
+        // (gather kernel), the lm_head in the layout its path needs
+        tok_ = DevMat{};
+        tok_.data = tok_data;
+        tok_.w.rows = V_;
+        tok_.w.cols = E_;
+        tok_.w.wtype = (int)tok_wtype;
+        tok_.w.data = tok_data.data_ptr();
+        if (tok_wtype == W_Q4_0 || tok_wtype == W_Q4_1) {
+            tok_.scales = tok_scales;
+            tok_.w.scales = tok_scales.data_ptr();
+        }
+        out_mfma_ = (out_wtype != W_F32);
+        if (out_mfma_)
+            mout_ = make_devmat2(out_data, out_scales, out_wtype, V_, E_);
+        else
+            out_ = make_devmat_f32(out_data, V_, E_);
         has_extra_ = true;
     }
 
@@ -174,25 +241,38 @@ class SliceEngine {
         const size_t layer_stride = (size_t)B_ * ctx_ * E_;
         __half* kbase = reinterpret_cast<__half*>(k_cache_.data_ptr());
         __half* vbase = reinterpret_cast<__half*>(v_cache_.data_ptr());
-        float* xn = xn_.data_ptr<float>();
+        const float* ifr = inv_freq_.data_ptr<float>();
         float* qb = qb_.data_ptr<float>();
         float* ab = ab_.data_ptr<float>();
-        float* ffb = ffb_.data_ptr<float>();
-        const float* ifr = inv_freq_.data_ptr<float>();
+
+        if (!layers_[0].mfma) {
+            forward_legacy(s, xp, pp, sp, T, kbase, vbase, ifr, layer_stride);
+            return x;
+        }
+        unsigned short* xprep = u16p(xprep_);
+        unsigned short* aprep = u16p(aprep_);
+        unsigned short* gprep = u16p(gprep_);
+        float* ssa = ss_attn_.data_ptr<float>();
+        float* ssf = ss_ffn_.data_ptr<float>();
+        // zero the atomic sumsq slots, then stage x into the side channel
+        hipMemsetAsync(ssa, 0, sizeof(float) * (L_ + 1) * kMaxTokens, s);
+        hipMemsetAsync(ssf, 0, sizeof(float) * L_ * kMaxTokens, s);
+        launch_prep_x(s, xp, xprep, ssa, E_, T);
         for (int li = 0; li < L_; ++li) {
             Layer& l = layers_[li];
             __half* kc = kbase + (size_t)li * layer_stride;
             __half* vc = vbase + (size_t)li * layer_stride;
-            launch_rmsnorm(s, xp, l.attn_norm.data_ptr<float>(), xn, T, E_,
-                           eps_);
-            launch_qkv_rope_append(s, l.wq.w, l.wk.w, l.wv.w, xn, qb, kc, vc,
-                                   pp, sp, ifr, E_, D_, ctx_, T);
-            launch_attention(s, qb, kc, vc, ab, pp, sp, T, H_, E_, D_, ctx_);
-            launch_gemv(s, l.wo.w, ab, /*res=*/xp, xp, T);
-            launch_rmsnorm(s, xp, l.ffn_norm.data_ptr<float>(), xn, T, E_,
-                           eps_);
-            launch_ffn_gate(s, l.w1.w, l.w3.w, xn, ffb, T);
-            launch_gemv(s, l.w2.w, ffb, /*res=*/xp, xp, T);
+            launch_qkv16(s, l.mq.w, l.mk.w, l.mv.w, xprep,
+                         u16p(l.attn_normprep), ssa + li * kMaxTokens, eps_,
+                         qb, kc, vc, pp, sp, ifr, E_, D_, ctx_, T);
+            launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_, D_,
+                             ctx_);
+            launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
+                          xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
+            launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
+                         ssf + li * kMaxTokens, eps_, gprep, T);
+            launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
+                          xprep, ssa + (li + 1) * kMaxTokens, T, GM_RES_SQ);
         }
         return x;
     }
@@ -222,14 +302,23 @@ class SliceEngine {
         const int T = (int)xin.size(0);
         TORCH_CHECK(T <= kMaxTokens, "logits: too many rows per call");
         hipStream_t s = c10::hip::getCurrentHIPStream().stream();
-        launch_rmsnorm(s, xin.data_ptr<float>(),
-                       final_norm_.data_ptr<float>(), xn_.data_ptr<float>(),
-                       T, E_, eps_);
         auto lg = torch::empty(
             {T, V_},
             torch::TensorOptions().device(torch::kCUDA).dtype(torch::kFloat32));
-        launch_gemv(s, out_.w, xn_.data_ptr<float>(), nullptr,
-                    lg.data_ptr<float>(), T);
+        if (out_mfma_) {
+            launch_prep_x(s, xin.data_ptr<float>(), u16p(xprep_),
+                          ss_tmp_.data_ptr<float>(), E_, T);
+            launch_gemm16(s, mout_.w, u16p(xprep_), u16p(final_normprep_),
+                          ss_tmp_.data_ptr<float>(), eps_,
+                          lg.data_ptr<float>(), nullptr, nullptr, T,
+                          GM_NORM_PLAIN);
+        } else {
+            launch_rmsnorm(s, xin.data_ptr<float>(),
+                           final_norm_.data_ptr<float>(),
+                           xn_.data_ptr<float>(), T, E_, eps_);
+            launch_gemv(s, out_.w, xn_.data_ptr<float>(), nullptr,
+                        lg.data_ptr<float>(), T);
+        }
         return lg;
     }
 
@@ -250,6 +339,31 @@ class SliceEngine {
     int64_t max_batch() const { return B_; }
 
  private:
+    void forward_legacy(hipStream_t s, float* xp, const int* pp,
+                        const int* sp, int T, __half* kbase, __half* vbase,
+                        const float* ifr, size_t layer_stride) {
+        float* xn = xn_.data_ptr<float>();
+        float* qb = qb_.data_ptr<float>();
+        float* ab = ab_.data_ptr<float>();
+        float* ffb = ffb_.data_ptr<float>();
+        for (int li = 0; li < L_; ++li) {
+            Layer& l = layers_[li];
+            __half* kc = kbase + (size_t)li * layer_stride;
+            __half* vc = vbase + (size_t)li * layer_stride;
+            launch_rmsnorm(s, xp, l.attn_norm.data_ptr<float>(), xn, T, E_,
+                           eps_);
+            launch_qkv_rope_append(s, l.wq.w, l.wk.w, l.wv.w, xn, qb, kc, vc,
+                                   pp, sp, ifr, E_, D_, ctx_, T);
+            launch_attention(s, qb, kc, vc, ab, nullptr, pp, sp, T, H_, E_,
+                             D_, ctx_);
+            launch_gemv(s, l.wo.w, ab, /*res=*/xp, xp, T);
+            launch_rmsnorm(s, xp, l.ffn_norm.data_ptr<float>(), xn, T, E_,
+                           eps_);
+            launch_ffn_gate(s, l.w1.w, l.w3.w, xn, ffb, T);
+            launch_gemv(s, l.w2.w, ffb, /*res=*/xp, xp, T);
+        }
+    }
+
     int E_, H_, D_, F_, L_, ctx_, B_;
     int V_ = 0;
     float eps_;
@@ -257,9 +371,12 @@ class SliceEngine {
     std::vector<bool> loaded_;
     torch::Tensor k_cache_, v_cache_, inv_freq_;
     torch::Tensor xn_, qb_, ab_, ffb_;
+    torch::Tensor xprep_, aprep_, gprep_, ss_attn_, ss_ffn_, ss_tmp_;
     bool has_extra_ = false;
+    bool out_mfma_ = false;
     DevMat tok_, out_;
-    torch::Tensor final_norm_;
+    DevMat2 mout_;
+    torch::Tensor final_norm_, final_normprep_;
 };
 
 }  // namespace

@@ -16,6 +16,43 @@ struct WMat {
     int wtype;  // WType
 };
 
+// MFMA-tiled weight matrix (the production decode path). Layouts, with
+// R = rows/16 row-tiles and nb = cols/32 q4 blocks:
+//   q4_0/q4_1: data   = u32[R][nb][4][16]   (nibble words, see kernels.hip)
+//              scales = f32[R][nb][16]      (q4_1: float2 (d, m))
+//   W_F16    : data   = bf16[R][cols/8][16][8]  (weights converted to bf16)
+struct WMat2 {
+    const void* data;
+    const void* scales;
+    int rows;
+    int cols;
+    int wtype;  // WType (W_F32 never appears here — legacy path)
+};
+
+// k_gemm16 fused-epilogue modes
+enum GemmMode { GM_PLAIN = 0, GM_RES_SQ = 1, GM_NORM_PLAIN = 2 };
+
+void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
+                   float* ss, int cols, int T);
+
+void launch_gemm16(hipStream_t s, const WMat2& w,
+                   const unsigned short* bprep,
+                   const unsigned short* normprep, const float* ss_in,
+                   float eps, float* y, unsigned short* xprep_out,
+                   float* ss_out, int T, int mode);
+
+void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                  const WMat2& wv, const unsigned short* xprep,
+                  const unsigned short* normprep, const float* ss_in,
+                  float eps, float* q_buf, __half* k_cache_layer,
+                  __half* v_cache_layer, const int* pos, const int* seq,
+                  const float* inv_freq, int E, int D, int n_ctx, int T);
+
+void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
+                  const unsigned short* xprep,
+                  const unsigned short* normprep, const float* ss_in,
+                  float eps, unsigned short* gprep, int T);
+
 void launch_rmsnorm(hipStream_t s, const float* x, const float* w, float* y,
                     int T, int E, float eps);
 
@@ -28,7 +65,8 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
 
 void launch_attention(hipStream_t s, const float* q_buf,
                       const __half* k_cache_layer,
-                      const __half* v_cache_layer, float* out, const int* pos,
+                      const __half* v_cache_layer, float* out,
+                      unsigned short* out_prep, const int* pos,
                       const int* seq, int T, int H, int E, int D, int n_ctx);
 
 void launch_gemv(hipStream_t s, const WMat& w, const float* x,

@@ -22,6 +22,7 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
+#include <hip/hip_bf16.h>
 #include <math.h>
 #include <stdint.h>
 
@@ -365,8 +366,8 @@ __global__ void k_qkv_rope_append(
 __global__ void k_attention(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
     const __half* __restrict__ v_cache, float* __restrict__ out,
-    const int* __restrict__ pos, const int* __restrict__ seq,
-    int E, int D, int n_ctx) {
+    unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
+    const int* __restrict__ seq, int E, int D, int n_ctx) {
     const int t = blockIdx.x;
     const int h = blockIdx.y;
     const int J = pos[t] + 1;
@@ -448,8 +449,18 @@ __global__ void k_attention(
         }
         __syncthreads();
     }
-    if (threadIdx.x < D)
-        out[(size_t)t * E + h * D + threadIdx.x] = o / l;
+    if (threadIdx.x < D) {
+        const float v = o / l;
+        const int e = h * D + threadIdx.x;
+        out[(size_t)t * E + e] = v;
+        if (out_prep != nullptr) {
+            // bf16 B-layout side-channel for the wo MFMA consumer:
+            // element (k=e, j=t) at [(e>>3)*16 + t]*8 + (e&7)
+            union { __hip_bfloat16 b; unsigned short u; } c;
+            c.b = __float2bfloat16(v);
+            out_prep[((size_t)(e >> 3) * 16 + t) * 8 + (e & 7)] = c.u;
+        }
+    }
 }
 
 // --------------------------------------------------------------- GEMV(+res)
@@ -593,6 +604,447 @@ __global__ void k_argmax(const float* __restrict__ logits,
     if (threadIdx.x == 0) out[t] = si[0];
 }
 
+// ======================================================================
+// MFMA dequant-GEMM family — the production decode path.
+//
+// Matrix cores do the FLOPs; the VALU only unpacks. One
+// mfma_f32_16x16x32_bf16 consumes exactly one q4_0 block column (K=32) of
+// a 16-row tile, with the 4-bit dequantization folded in two ways:
+//   * nibbles become exact bf16 integers by OR-ing into the 0x4300
+//     exponent pattern (bf16(0x4300 | n) == 128 + n), so A carries (n+128)
+//     and the per-block scale/bias is applied to the 16x16 MFMA result:
+//     w = d*(n-8)  =>  d*(D - 136*sum(B_col));
+//   * the repacked nibble word interleaves weights so the three OR+AND+SHR
+//     per word produce the bf16x8 A-fragment directly (see repack in
+//     engine/slice_engine.py).
+// Activations ride a bf16 side-channel ("xprep", layout [cols/8][16][8] =
+// exactly the B-fragment gather) written by the producing kernel's
+// epilogue; the f32 residual stream stays f32. RMSNorm is fused into the
+// consumer's B-fragment build (sumsq side-channel written by the previous
+// residual epilogue), so a decode layer is 5 kernels total.
+//
+// Block = 4 waves; each wave owns a quarter of the K blocks of ONE 16-row
+// tile; partial accumulators combine through LDS; wave 0 runs the fused
+// epilogue (residual add + sumsq atomics + xprep/rope/cache/silu writes).
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+union ABFrag {
+    uint32_t u[4];
+    bf16x8 v;
+};
+
+__device__ __forceinline__ float bflo(uint32_t w) {
+    union { uint32_t u; float f; } c;
+    c.u = w << 16;
+    return c.f;
+}
+__device__ __forceinline__ float bfhi(uint32_t w) {
+    union { uint32_t u; float f; } c;
+    c.u = w & 0xFFFF0000u;
+    return c.f;
+}
+__device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
+    union { __hip_bfloat162 b; uint32_t u; } c;
+    c.b = __hip_bfloat162(__float2bfloat16(lo), __float2bfloat16(hi));
+    return c.u;
+}
+
+// Build the A fragment from one repacked nibble word (8 weights as n+128).
+__device__ __forceinline__ void a_frag_q4(uint32_t q, ABFrag& a) {
+    a.u[0] = 0x43004300u | (q & 0x000F000Fu);
+    a.u[1] = 0x43004300u | ((q >> 4) & 0x000F000Fu);
+    a.u[2] = 0x43004300u | ((q >> 8) & 0x000F000Fu);
+    a.u[3] = 0x43004300u | ((q >> 12) & 0x000F000Fu);
+}
+
+// Per-wave state for the K loop: lane (i = l&15 row, ks = l>>4 k-span).
+struct KLoop {
+    int lane, i, ks;
+    int kb0, kb1;  // this wave's K-block range
+    __device__ void init(int nb) {
+        lane = threadIdx.x & (WAVE - 1);
+        i = lane & 15;
+        ks = lane >> 4;
+        const int wid = threadIdx.x / WAVE;
+        const int per = (nb + 3) >> 2;
+        kb0 = wid * per;
+        kb1 = min(nb, kb0 + per);
+    }
+};
+
+// Load + (for the norm path) scale the B fragment for block kb.
+// Returns sumB (sum over the full K=32 of this lane's column) via xor-shfl.
+template <bool NORM>
+__device__ __forceinline__ float b_frag(
+    const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep, float scale, int kb,
+    int i /*col j = i*/, int ks, ABFrag& b, bool need_sum) {
+    const int kc = kb * 4 + ks;
+    const uint4 xb = *reinterpret_cast<const uint4*>(xprep + (kc * 16 + i) * 8);
+    float sum = 0.0f;
+    if (NORM) {
+        const uint4 nb_ = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+        const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
+        const uint32_t nw[4] = {nb_.x, nb_.y, nb_.z, nb_.w};
+#pragma unroll
+        for (int w = 0; w < 4; ++w) {
+            const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
+            const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
+            b.u[w] = pack_bf16(f0, f1);
+            sum += f0 + f1;
+        }
+    } else {
+        b.u[0] = xb.x;
+        b.u[1] = xb.y;
+        b.u[2] = xb.z;
+        b.u[3] = xb.w;
+        if (need_sum) {
+#pragma unroll
+            for (int w = 0; w < 4; ++w)
+                sum += bflo(b.u[w]) + bfhi(b.u[w]);
+        }
+    }
+    if (need_sum) {
+        // reduce over the 4 k-spans (lanes l, l^16, l^32, l^48 share col i)
+        sum += __shfl_xor(sum, 16);
+        sum += __shfl_xor(sum, 32);
+    }
+    return sum;
+}
+
+// One wave's K loop for one 16-row tile of W against the 16-col B panel.
+// acc[jj] accumulates rows (l>>4)*4 + jj, col l&15 of the output.
+template <int WT, bool NORM>
+__device__ __forceinline__ void wave_tile_kloop(
+    const WMat2& w, int tile_row /* R index */,
+    const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep, float scale,
+    float acc[4]) {
+    KLoop kl;
+    const int nb = w.cols >> 5;
+    kl.init(nb);
+    acc[0] = acc[1] = acc[2] = acc[3] = 0.0f;
+    const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
+    if (WT == W_F16) {
+        // bf16 tile: data [R][cols/8][16][8]; pure load + chained MFMA.
+        const unsigned short* base =
+            (const unsigned short*)w.data +
+            ((size_t)tile_row * (w.cols >> 3)) * 128;  // 16*8 per kc
+        f32x4 c = zero;
+        for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
+            const int kc = kb * 4 + kl.ks;
+            ABFrag a, b;
+            const uint4 aw = *reinterpret_cast<const uint4*>(
+                base + ((size_t)kc * 16 + kl.i) * 8);
+            a.u[0] = aw.x; a.u[1] = aw.y; a.u[2] = aw.z; a.u[3] = aw.w;
+            b_frag<NORM>(xprep, normprep, scale, kb, kl.i, kl.ks, b, false);
+            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c, 0, 0, 0);
+        }
+        acc[0] = c[0]; acc[1] = c[1]; acc[2] = c[2]; acc[3] = c[3];
+        return;
+    }
+    // q4_0 / q4_1: per-block scale => C starts at 0 each block, the scaled
+    // result folds into acc on the VALU.
+    const uint32_t* qbase =
+        (const uint32_t*)w.data + ((size_t)tile_row * nb) * 64;
+    for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
+        ABFrag a, b;
+        const uint32_t q = __builtin_nontemporal_load(
+            qbase + (size_t)kb * 64 + kl.ks * 16 + kl.i);
+        a_frag_q4(q, a);
+        const float sumB = b_frag<NORM>(xprep, normprep, scale, kb, kl.i,
+                                        kl.ks, b, true);
+        f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, zero,
+                                                          0, 0, 0);
+        if (WT == W_Q4_0) {
+            const float4 sc = *reinterpret_cast<const float4*>(
+                (const float*)w.scales +
+                ((size_t)tile_row * nb + kb) * 16 + (kl.lane >> 4) * 4);
+            const float s4[4] = {sc.x, sc.y, sc.z, sc.w};
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj)
+                acc[jj] = fmaf(s4[jj], fmaf(-136.0f, sumB, d[jj]), acc[jj]);
+        } else {  // q4_1: scales hold (d, m) float2 per row
+            const float* sp = (const float*)w.scales +
+                (((size_t)tile_row * nb + kb) * 16 + (kl.lane >> 4) * 4) * 2;
+            const float4 s01 = *reinterpret_cast<const float4*>(sp);
+            const float4 s23 = *reinterpret_cast<const float4*>(sp + 4);
+            const float dd[4] = {s01.x, s01.z, s23.x, s23.z};
+            const float mm[4] = {s01.y, s01.w, s23.y, s23.w};
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                acc[jj] = fmaf(dd[jj], fmaf(-128.0f, sumB, d[jj]), acc[jj]);
+                acc[jj] = fmaf(mm[jj], sumB, acc[jj]);
+            }
+        }
+    }
+}
+
+// LDS combine of the 4 waves' partial accumulators; wave 0 ends with the
+// full 16x16 tile (4 rows x 1 col per lane). NACC = accumulator sets.
+template <int NACC>
+__device__ __forceinline__ void combine_acc(float acc[NACC][4],
+                                            float* lds /* 3*64*4*NACC */) {
+    const int wid = threadIdx.x / WAVE;
+    const int lane = threadIdx.x & (WAVE - 1);
+    if (wid > 0) {
+        float* dst = lds + (((wid - 1) * 64 + lane) * 4) * NACC;
+#pragma unroll
+        for (int n = 0; n < NACC; ++n)
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) dst[n * 4 + jj] = acc[n][jj];
+    }
+    __syncthreads();
+    if (wid == 0) {
+#pragma unroll
+        for (int w = 0; w < 3; ++w) {
+            const float* src = lds + ((w * 64 + lane) * 4) * NACC;
+#pragma unroll
+            for (int n = 0; n < NACC; ++n)
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj) acc[n][jj] += src[n * 4 + jj];
+        }
+    }
+}
+
+__device__ __forceinline__ float norm_scale(const float* ss, int j, int cols,
+                                            float eps) {
+    return rsqrtf(ss[j] / (float)cols + eps);
+}
+
+// ------------------------------------------------------------- k_prep_x
+// sumsq + bf16 xprep of an incoming f32 activation block (forward entry,
+// logits entry). Grid: T blocks.
+__global__ void k_prep_x(const float* __restrict__ x,
+                         unsigned short* __restrict__ xprep,
+                         float* __restrict__ ss, int cols) {
+    const int t = blockIdx.x;
+    const float* xt = x + (size_t)t * cols;
+    __shared__ float red[NWAVES];
+    float sum = 0.0f;
+    const int nkc = cols >> 3;
+    for (int kc = threadIdx.x; kc < nkc; kc += BLOCK) {
+        const float4 a = *reinterpret_cast<const float4*>(xt + kc * 8);
+        const float4 b = *reinterpret_cast<const float4*>(xt + kc * 8 + 4);
+        sum += a.x * a.x + a.y * a.y + a.z * a.z + a.w * a.w;
+        sum += b.x * b.x + b.y * b.y + b.z * b.z + b.w * b.w;
+        uint4 o;
+        o.x = pack_bf16(a.x, a.y);
+        o.y = pack_bf16(a.z, a.w);
+        o.z = pack_bf16(b.x, b.y);
+        o.w = pack_bf16(b.z, b.w);
+        *reinterpret_cast<uint4*>(xprep + ((size_t)kc * 16 + t) * 8) = o;
+    }
+    sum = wave_reduce_sum(sum);
+    const int wid = threadIdx.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[wid] = sum;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        ss[t] = red[0] + red[1] + red[2] + red[3];
+}
+
+// ------------------------------------------------------------- k_gemm16
+template <int WT, int MODE>
+__global__ __launch_bounds__(BLOCK) void k_gemm16(
+    WMat2 w, const unsigned short* __restrict__ bprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps, float* __restrict__ y,
+    unsigned short* __restrict__ xprep_out, float* __restrict__ ss_out,
+    int T) {
+    constexpr bool NORM = (MODE == GM_NORM_PLAIN);
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float scale = 1.0f;
+    if (NORM) scale = norm_scale(ss_in, j < T ? j : 0, w.cols, eps);
+    float acc[1][4];
+    wave_tile_kloop<WT, NORM>(w, blockIdx.x, bprep, normprep, scale, acc[0]);
+    __shared__ float lds[3 * 64 * 4];
+    combine_acc<1>(acc, lds);
+    if (threadIdx.x >= WAVE) return;
+    // wave 0 epilogue: rows r = blockIdx.x*16 + (lane>>4)*4 + jj, col j.
+    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
+    float sq = 0.0f;
+#pragma unroll
+    for (int jj = 0; jj < 4; ++jj) {
+        const int row = r0 + jj;
+        float v = acc[0][jj];
+        if (MODE == GM_RES_SQ) {
+            if (j < T) {
+                v += y[(size_t)j * w.rows + row];
+                y[(size_t)j * w.rows + row] = v;
+                sq += v * v;
+            }
+        } else {
+            if (j < T) y[(size_t)j * w.rows + row] = v;
+        }
+        acc[0][jj] = v;
+    }
+    if (MODE == GM_RES_SQ && xprep_out != nullptr && j < T) {
+        // 4 consecutive rows -> one aligned 8 B bf16x4 chunk of xprep
+        uint2 o;
+        o.x = pack_bf16(acc[0][0], acc[0][1]);
+        o.y = pack_bf16(acc[0][2], acc[0][3]);
+        *reinterpret_cast<uint2*>(
+            xprep_out + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
+    }
+    if (MODE == GM_RES_SQ && ss_out != nullptr) {
+        sq += __shfl_xor(sq, 16);
+        sq += __shfl_xor(sq, 32);
+        if (lane < 16 && lane < T) atomicAdd(ss_out + lane, sq);
+    }
+}
+
+// ------------------------------------------------------------- k_qkv16
+// QKV projections on MFMA + fused input RMSNorm + RoPE + KV append.
+template <int WT>
+__global__ __launch_bounds__(BLOCK) void k_qkv16(
+    WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps, float* __restrict__ q_buf,
+    __half* __restrict__ k_cache, __half* __restrict__ v_cache,
+    const int* __restrict__ pos, const int* __restrict__ seq,
+    const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T) {
+    const int tiles_per_mat = E >> 4;
+    const int mat = blockIdx.x / tiles_per_mat;
+    const int tile = blockIdx.x % tiles_per_mat;
+    const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    const float scale = norm_scale(ss_in, j < T ? j : 0, E, eps);
+    float acc[1][4];
+    wave_tile_kloop<WT, true>(w, tile, xprep, normprep, scale, acc[0]);
+    __shared__ float lds[3 * 64 * 4];
+    combine_acc<1>(acc, lds);
+    if (threadIdx.x >= WAVE || j >= T) return;
+    const int r0 = tile * 16 + (lane >> 4) * 4;
+    const int p = pos[j];
+    if (mat == 2) {  // V rows: straight f16 cache append
+        __half* dst = v_cache + ((size_t)seq[j] * n_ctx + p) * E + r0;
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) dst[jj] = __float2half(acc[0][jj]);
+        return;
+    }
+    // q/k: RoPE on in-lane pairs (rows r0+2q, r0+2q+1)
+#pragma unroll
+    for (int q2 = 0; q2 < 2; ++q2) {
+        const int e = r0 + 2 * q2;
+        const int d = e % D;
+        const float theta = (float)p * inv_freq[d >> 1];
+        float sn, cs;
+        __sincosf(theta, &sn, &cs);
+        const float x0 = acc[0][2 * q2], x1 = acc[0][2 * q2 + 1];
+        const float o0 = x0 * cs - x1 * sn;
+        const float o1 = x0 * sn + x1 * cs;
+        if (mat == 0) {
+            q_buf[(size_t)j * E + e] = o0;
+            q_buf[(size_t)j * E + e + 1] = o1;
+        } else {
+            __half* dst = k_cache + ((size_t)seq[j] * n_ctx + p) * E + e;
+            dst[0] = __float2half(o0);
+            dst[1] = __float2half(o1);
+        }
+    }
+}
+
+// ------------------------------------------------------------- k_ffn16
+// w1 + w3 against the same B panel + fused input RMSNorm + SwiGLU; emits
+// the gate product straight into gprep (bf16 B-layout over F).
+template <int WT>
+__global__ __launch_bounds__(BLOCK) void k_ffn16(
+    WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps,
+    unsigned short* __restrict__ gprep, int T) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    const float scale = norm_scale(ss_in, j < T ? j : 0, w1.cols, eps);
+    float acc[2][4];
+    {
+        KLoop kl;
+        const int nb = w1.cols >> 5;
+        kl.init(nb);
+        acc[0][0] = acc[0][1] = acc[0][2] = acc[0][3] = 0.0f;
+        acc[1][0] = acc[1][1] = acc[1][2] = acc[1][3] = 0.0f;
+        const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
+        for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
+            ABFrag b;
+            const float sumB = b_frag<true>(xprep, normprep, scale, kb,
+                                            kl.i, kl.ks, b, true);
+#pragma unroll
+            for (int m = 0; m < 2; ++m) {
+                const WMat2& w = m ? w3 : w1;
+                ABFrag a;
+                if (WT == W_F16) {
+                    const unsigned short* base =
+                        (const unsigned short*)w.data +
+                        ((size_t)blockIdx.x * (w.cols >> 3)) * 128;
+                    const int kc = kb * 4 + kl.ks;
+                    const uint4 aw = *reinterpret_cast<const uint4*>(
+                        base + ((size_t)kc * 16 + kl.i) * 8);
+                    a.u[0] = aw.x; a.u[1] = aw.y;
+                    a.u[2] = aw.z; a.u[3] = aw.w;
+                    f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                        a.v, b.v, zero, 0, 0, 0);
+#pragma unroll
+                    for (int jj = 0; jj < 4; ++jj) acc[m][jj] += d[jj];
+                    continue;
+                }
+                const uint32_t* qbase =
+                    (const uint32_t*)w.data + ((size_t)blockIdx.x * nb) * 64;
+                const uint32_t q = __builtin_nontemporal_load(
+                    qbase + (size_t)kb * 64 + kl.ks * 16 + kl.i);
+                a_frag_q4(q, a);
+                f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a.v, b.v, zero, 0, 0, 0);
+                if (WT == W_Q4_0) {
+                    const float4 sc = *reinterpret_cast<const float4*>(
+                        (const float*)w.scales +
+                        ((size_t)blockIdx.x * nb + kb) * 16 +
+                        (kl.lane >> 4) * 4);
+                    const float s4[4] = {sc.x, sc.y, sc.z, sc.w};
+#pragma unroll
+                    for (int jj = 0; jj < 4; ++jj)
+                        acc[m][jj] = fmaf(
+                            s4[jj], fmaf(-136.0f, sumB, d[jj]), acc[m][jj]);
+                } else {
+                    const float* sp = (const float*)w.scales +
+                        (((size_t)blockIdx.x * nb + kb) * 16 +
+                         (kl.lane >> 4) * 4) * 2;
+                    const float4 s01 = *reinterpret_cast<const float4*>(sp);
+                    const float4 s23 =
+                        *reinterpret_cast<const float4*>(sp + 4);
+                    const float dd[4] = {s01.x, s01.z, s23.x, s23.z};
+                    const float mm[4] = {s01.y, s01.w, s23.y, s23.w};
+#pragma unroll
+                    for (int jj = 0; jj < 4; ++jj) {
+                        acc[m][jj] = fmaf(
+                            dd[jj], fmaf(-128.0f, sumB, d[jj]), acc[m][jj]);
+                        acc[m][jj] = fmaf(mm[jj], sumB, acc[m][jj]);
+                    }
+                }
+            }
+        }
+    }
+    __shared__ float lds[3 * 64 * 4 * 2];
+    combine_acc<2>(acc, lds);
+    if (threadIdx.x >= WAVE || j >= T) return;
+    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
+    float g[4];
+#pragma unroll
+    for (int jj = 0; jj < 4; ++jj) {
+        const float v1 = acc[0][jj];
+        const float silu = v1 / (1.0f + __expf(-v1));
+        g[jj] = silu * acc[1][jj];
+    }
+    uint2 o;
+    o.x = pack_bf16(g[0], g[1]);
+    o.y = pack_bf16(g[2], g[3]);
+    *reinterpret_cast<uint2*>(
+        gprep + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
+}
+
 // ============================================================== launchers
 
 static inline int pick_tmax(int T) {
@@ -678,13 +1130,86 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
 
 void launch_attention(hipStream_t s, const float* q_buf,
                       const __half* k_cache_layer,
-                      const __half* v_cache_layer, float* out, const int* pos,
+                      const __half* v_cache_layer, float* out,
+                      unsigned short* out_prep, const int* pos,
                       const int* seq, int T, int H, int E, int D, int n_ctx) {
     const dim3 grid(T, H);
     const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
     hipLaunchKernelGGL(k_attention, grid, dim3(BLOCK), lds, s, q_buf,
-                       k_cache_layer, v_cache_layer, out, pos, seq, E, D,
-                       n_ctx);
+                       k_cache_layer, v_cache_layer, out, out_prep, pos, seq,
+                       E, D, n_ctx);
+}
+
+// ------------------------------------------------- MFMA-path launchers
+
+void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
+                   float* ss, int cols, int T) {
+    hipLaunchKernelGGL(k_prep_x, dim3(T), dim3(BLOCK), 0, s, x, xprep, ss,
+                       cols);
+}
+
+#define DISPATCH_WT2(WTV, ...)                   \
+    switch (WTV) {                               \
+        case W_Q4_0: {                           \
+            constexpr int WTc = W_Q4_0;          \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        case W_Q4_1: {                           \
+            constexpr int WTc = W_Q4_1;          \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        default: {                               \
+            constexpr int WTc = W_F16;           \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+    }
+
+void launch_gemm16(hipStream_t s, const WMat2& w,
+                   const unsigned short* bprep,
+                   const unsigned short* normprep, const float* ss_in,
+                   float eps, float* y, unsigned short* xprep_out,
+                   float* ss_out, int T, int mode) {
+    const dim3 grid(w.rows / 16);
+    DISPATCH_WT2(w.wtype, {
+        if (mode == GM_RES_SQ)
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_RES_SQ>), grid, dim3(BLOCK),
+                               0, s, w, bprep, normprep, ss_in, eps, y,
+                               xprep_out, ss_out, T);
+        else if (mode == GM_NORM_PLAIN)
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN>), grid,
+                               dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
+                               eps, y, xprep_out, ss_out, T);
+        else
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_PLAIN>), grid, dim3(BLOCK),
+                               0, s, w, bprep, normprep, ss_in, eps, y,
+                               xprep_out, ss_out, T);
+    });
+}
+
+void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                  const WMat2& wv, const unsigned short* xprep,
+                  const unsigned short* normprep, const float* ss_in,
+                  float eps, float* q_buf, __half* k_cache_layer,
+                  __half* v_cache_layer, const int* pos, const int* seq,
+                  const float* inv_freq, int E, int D, int n_ctx, int T) {
+    const dim3 grid(3 * (E >> 4));
+    DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
+        (k_qkv16<WTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep, normprep,
+        ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq, inv_freq,
+        E, D, n_ctx, T));
+}
+
+void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
+                  const unsigned short* xprep,
+                  const unsigned short* normprep, const float* ss_in,
+                  float eps, unsigned short* gprep, int T) {
+    const dim3 grid(w1.rows / 16);
+    DISPATCH_WT2(w1.wtype, hipLaunchKernelGGL(
+        (k_ffn16<WTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep, normprep,
+        ss_in, eps, gprep, T));
 }
 
 void launch_gemv(hipStream_t s, const WMat& w, const float* x,
