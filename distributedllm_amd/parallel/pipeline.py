@@ -74,31 +74,75 @@ class DecodePipeline:
         self.tok = [torch.randint(3, engine.hp.n_vocab, (mbs,),
                                   dtype=torch.int32, device=dev)
                     for _ in range(M)]
+        self._graphs = None      # per-mb captured hipGraphs
+        self._graph_out = None   # per-mb tensors the graph writes
 
-    def _advance_mb(self, m: int) -> None:
-        """Push micro-batch m one token forward through this stage."""
-        eng, cfg = self.engine, self.cfg
+    # ------------------------------------------------------ hipGraph mode
+
+    def _mb_compute(self, m: int):
+        """The capturable (comm-free) compute of micro-batch m.
+
+        Returns the tensor the next stage needs: activations for mid
+        stages, sampled token ids for the last stage (None for single-GPU,
+        where tok[m] is updated in place)."""
+        eng = self.engine
         if self.is_first:
             x = eng.embed(self.tok[m])
         else:
-            dist.recv(self.x_recv[m], src=self.rank - 1)
             x = self.x_recv[m]
         y = eng.forward(x, self.pos[m], self.seq[m])
-        if not self.is_last:
-            dist.send(y.contiguous(), dst=self.rank + 1)
-        else:
+        out = y
+        if self.is_last:
             lg = eng.logits(y, all_logits=True)
             nxt = eng.argmax(lg)
-            if self.world > 1:
-                dist.send(nxt, dst=0)
+            if self.world == 1:
+                self.tok[m].copy_(nxt)
+                out = None
             else:
-                self.tok[m] = nxt
+                out = nxt
+        self.pos[m] += 1
+        return out
+
+    def capture_graphs(self, warmup_steps: int = 2) -> None:
+        """Capture each micro-batch's compute into a hipGraph (decode
+        steady state). Comm (RCCL send/recv) stays eager between replays.
+        The warmup steps run for real (they advance positions/KV)."""
+        assert self.cfg.device == "cuda"
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup_steps):
+                for m in range(self.cfg.n_mb):
+                    self._mb_compute(m)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        self._graphs = []
+        self._graph_out = []
+        for m in range(self.cfg.n_mb):
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                out = self._mb_compute(m)
+            self._graphs.append(g)
+            self._graph_out.append(out)
+
+    def _advance_mb(self, m: int) -> None:
+        """Push micro-batch m one token forward through this stage."""
+        if not self.is_first:
+            dist.recv(self.x_recv[m], src=self.rank - 1)
+        if self._graphs is not None:
+            self._graphs[m].replay()
+            out = self._graph_out[m]
+        else:
+            out = self._mb_compute(m)
+        if not self.is_last:
+            dist.send(out, dst=self.rank + 1)
+        elif self.world > 1:
+            dist.send(out, dst=0)
         if self.is_first and self.world > 1:
             # token ids for this micro-batch's next step come from the
             # last stage; ordered FIFO per rank pair, so recv here pairs
             # with the send above.
             dist.recv(self.tok[m], src=self.world - 1)
-        self.pos[m] += 1
 
     def run_steps(self, steps: int) -> None:
         for _ in range(steps):
@@ -110,9 +154,11 @@ class DecodePipeline:
 
 
 def timed_decode(pipe: DecodePipeline, steps: int, warmup: int,
-                 device: str) -> float:
+                 device: str, use_graphs: bool = True) -> float:
     """Barrier-bracketed timing of exactly `steps` steps; returns seconds
     (this rank's wall time — reduce MAX across ranks for the job time)."""
+    if use_graphs and device == "cuda":
+        pipe.capture_graphs(warmup_steps=2)
     pipe.run_steps(warmup)
     if dist.is_initialized():
         dist.barrier()
