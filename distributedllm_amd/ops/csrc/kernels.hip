@@ -714,71 +714,172 @@ __device__ __forceinline__ float b_frag(
     return sum;
 }
 
-// One wave's K loop for one 16-row tile of W against the 16-col B panel.
-// acc[jj] accumulates rows (l>>4)*4 + jj, col l&15 of the output.
-template <int WT, bool NORM>
-__device__ __forceinline__ void wave_tile_kloop(
-    const WMat2& w, int tile_row /* R index */,
-    const unsigned short* __restrict__ xprep,
-    const unsigned short* __restrict__ normprep, float scale,
-    float acc[4]) {
-    KLoop kl;
-    const int nb = w.cols >> 5;
-    kl.init(nb);
-    acc[0] = acc[1] = acc[2] = acc[3] = 0.0f;
-    const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-    if (WT == W_F16) {
-        // bf16 tile: data [R][cols/8][16][8]; pure load + chained MFMA.
-        const unsigned short* base =
-            (const unsigned short*)w.data +
-            ((size_t)tile_row * (w.cols >> 3)) * 128;  // 16*8 per kc
-        f32x4 c = zero;
-        for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
-            const int kc = kb * 4 + kl.ks;
-            ABFrag a, b;
-            const uint4 aw = *reinterpret_cast<const uint4*>(
-                base + ((size_t)kc * 16 + kl.i) * 8);
-            a.u[0] = aw.x; a.u[1] = aw.y; a.u[2] = aw.z; a.u[3] = aw.w;
-            b_frag<NORM>(xprep, normprep, scale, kb, kl.i, kl.ks, b, false);
-            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c, 0, 0, 0);
-        }
-        acc[0] = c[0]; acc[1] = c[1]; acc[2] = c[2]; acc[3] = c[3];
-        return;
-    }
-    // q4_0 / q4_1: per-block scale => C starts at 0 each block, the scaled
-    // result folds into acc on the VALU.
-    const uint32_t* qbase =
-        (const uint32_t*)w.data + ((size_t)tile_row * nb) * 64;
-    for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
-        ABFrag a, b;
-        const uint32_t q = __builtin_nontemporal_load(
-            qbase + (size_t)kb * 64 + kl.ks * 16 + kl.i);
-        a_frag_q4(q, a);
-        const float sumB = b_frag<NORM>(xprep, normprep, scale, kb, kl.i,
-                                        kl.ks, b, true);
-        f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, zero,
-                                                          0, 0, 0);
+// Per-block scale record + its application to the 16x16 MFMA result.
+template <int WT>
+struct AScale {
+    float4 a, b;  // q4_0: a only; q4_1: (d,m) pairs across a and b
+    __device__ __forceinline__ void load(const float* scales, int nb,
+                                         size_t tile_row, int kb, int grp) {
         if (WT == W_Q4_0) {
-            const float4 sc = *reinterpret_cast<const float4*>(
-                (const float*)w.scales +
-                ((size_t)tile_row * nb + kb) * 16 + (kl.lane >> 4) * 4);
-            const float s4[4] = {sc.x, sc.y, sc.z, sc.w};
+            a = *reinterpret_cast<const float4*>(
+                scales + (tile_row * nb + kb) * 16 + grp * 4);
+        } else if (WT == W_Q4_1) {
+            const float* sp =
+                scales + ((tile_row * nb + kb) * 16 + grp * 4) * 2;
+            a = *reinterpret_cast<const float4*>(sp);
+            b = *reinterpret_cast<const float4*>(sp + 4);
+        }
+    }
+    __device__ __forceinline__ void apply(const f32x4& d, float sumB,
+                                          float acc[4]) const {
+        if (WT == W_Q4_0) {
+            const float s4[4] = {a.x, a.y, a.z, a.w};
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
                 acc[jj] = fmaf(s4[jj], fmaf(-136.0f, sumB, d[jj]), acc[jj]);
-        } else {  // q4_1: scales hold (d, m) float2 per row
-            const float* sp = (const float*)w.scales +
-                (((size_t)tile_row * nb + kb) * 16 + (kl.lane >> 4) * 4) * 2;
-            const float4 s01 = *reinterpret_cast<const float4*>(sp);
-            const float4 s23 = *reinterpret_cast<const float4*>(sp + 4);
-            const float dd[4] = {s01.x, s01.z, s23.x, s23.z};
-            const float mm[4] = {s01.y, s01.w, s23.y, s23.w};
+        } else if (WT == W_Q4_1) {
+            const float dd[4] = {a.x, a.z, b.x, b.z};
+            const float mm[4] = {a.y, a.w, b.y, b.w};
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj) {
                 acc[jj] = fmaf(dd[jj], fmaf(-128.0f, sumB, d[jj]), acc[jj]);
                 acc[jj] = fmaf(mm[jj], sumB, acc[jj]);
             }
         }
+    }
+};
+
+// One wave's software-pipelined K loop over NM matrices sharing the B
+// panel (NM=2 for the FFN's w1/w3 — halves B traffic and doubles the
+// MFMA work per load batch). PF blocks' loads are issued together before
+// any compute so ~PF*NM*256B of weight stream stays in flight per wave —
+// the un-pipelined form was HBM-latency-bound at <1 TB/s.
+// acc[n][jj] accumulates rows (l>>4)*4 + jj, col l&15 of output n.
+template <int WT, bool NORM, int NM, int PF = 4>
+__device__ __forceinline__ void wave_tile_kloop(
+    const WMat2* const* ws, int tile_row,
+    const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep, float scale,
+    float acc[NM][4]) {
+    KLoop kl;
+    const int nb = ws[0]->cols >> 5;
+    kl.init(nb);
+#pragma unroll
+    for (int n = 0; n < NM; ++n)
+        acc[n][0] = acc[n][1] = acc[n][2] = acc[n][3] = 0.0f;
+    const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
+    const int grp = kl.lane >> 4;
+
+    const uint32_t* qbase[NM];
+    const unsigned short* tbase[NM];
+    const float* scbase[NM];
+#pragma unroll
+    for (int n = 0; n < NM; ++n) {
+        qbase[n] = (const uint32_t*)ws[n]->data + ((size_t)tile_row * nb) * 64;
+        tbase[n] = (const unsigned short*)ws[n]->data +
+                   ((size_t)tile_row * (ws[n]->cols >> 3)) * 128;
+        scbase[n] = (const float*)ws[n]->scales;
+    }
+
+    auto compute_one = [&](const uint32_t q[NM], const uint4 aw[NM],
+                           const uint4& xb, const uint4& nbv,
+                           const AScale<WT> sc[NM]) {
+        ABFrag b;
+        float sumB = 0.0f;
+        if (NORM) {
+            const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
+            const uint32_t nw[4] = {nbv.x, nbv.y, nbv.z, nbv.w};
+#pragma unroll
+            for (int w = 0; w < 4; ++w) {
+                const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
+                const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
+                b.u[w] = pack_bf16(f0, f1);
+                sumB += f0 + f1;
+            }
+        } else {
+            b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
+            if (WT != W_F16) {
+#pragma unroll
+                for (int w = 0; w < 4; ++w)
+                    sumB += bflo(b.u[w]) + bfhi(b.u[w]);
+            }
+        }
+        if (WT != W_F16) {
+            sumB += __shfl_xor(sumB, 16);
+            sumB += __shfl_xor(sumB, 32);
+        }
+#pragma unroll
+        for (int n = 0; n < NM; ++n) {
+            ABFrag a;
+            if (WT == W_F16) {
+                a.u[0] = aw[n].x; a.u[1] = aw[n].y;
+                a.u[2] = aw[n].z; a.u[3] = aw[n].w;
+                f32x4 c = {acc[n][0], acc[n][1], acc[n][2], acc[n][3]};
+                c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c,
+                                                            0, 0, 0);
+                acc[n][0] = c[0]; acc[n][1] = c[1];
+                acc[n][2] = c[2]; acc[n][3] = c[3];
+            } else {
+                a_frag_q4(q[n], a);
+                const f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    a.v, b.v, zero, 0, 0, 0);
+                sc[n].apply(d, sumB, acc[n]);
+            }
+        }
+    };
+
+    int g = kl.kb0;
+    for (; g + PF <= kl.kb1; g += PF) {
+        uint32_t q[PF][NM];
+        uint4 aw[PF][NM];
+        uint4 xb[PF], nbv[PF];
+        AScale<WT> sc[PF][NM];
+#pragma unroll
+        for (int u = 0; u < PF; ++u) {
+            const int kb = g + u;
+            const int kc = kb * 4 + kl.ks;
+            xb[u] = *reinterpret_cast<const uint4*>(
+                xprep + ((size_t)kc * 16 + kl.i) * 8);
+            if (NORM)
+                nbv[u] = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+#pragma unroll
+            for (int n = 0; n < NM; ++n) {
+                if (WT == W_F16) {
+                    aw[u][n] = *reinterpret_cast<const uint4*>(
+                        tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
+                } else {
+                    q[u][n] = __builtin_nontemporal_load(
+                        qbase[n] + (size_t)kb * 64 + kl.ks * 16 + kl.i);
+                    sc[u][n].load(scbase[n], nb, tile_row, kb, grp);
+                }
+            }
+        }
+#pragma unroll
+        for (int u = 0; u < PF; ++u)
+            compute_one(q[u], aw[u], xb[u], nbv[u], sc[u]);
+    }
+    for (; g < kl.kb1; ++g) {
+        uint32_t q[NM];
+        uint4 aw[NM];
+        uint4 xb, nbv;
+        AScale<WT> sc[NM];
+        const int kc = g * 4 + kl.ks;
+        xb = *reinterpret_cast<const uint4*>(
+            xprep + ((size_t)kc * 16 + kl.i) * 8);
+        if (NORM)
+            nbv = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+#pragma unroll
+        for (int n = 0; n < NM; ++n) {
+            if (WT == W_F16) {
+                aw[n] = *reinterpret_cast<const uint4*>(
+                    tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
+            } else {
+                q[n] = __builtin_nontemporal_load(
+                    qbase[n] + (size_t)g * 64 + kl.ks * 16 + kl.i);
+                sc[n].load(scbase[n], nb, tile_row, g, grp);
+            }
+        }
+        compute_one(q, aw, xb, nbv, sc);
     }
 }
 
@@ -859,7 +960,8 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     float scale = 1.0f;
     if (NORM) scale = norm_scale(ss_in, j < T ? j : 0, w.cols, eps);
     float acc[1][4];
-    wave_tile_kloop<WT, NORM>(w, blockIdx.x, bprep, normprep, scale, acc[0]);
+    const WMat2* ws[1] = {&w};
+    wave_tile_kloop<WT, NORM, 1>(ws, blockIdx.x, bprep, normprep, scale, acc);
     __shared__ float lds[3 * 64 * 4];
     combine_acc<1>(acc, lds);
     if (threadIdx.x >= WAVE) return;
@@ -914,7 +1016,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const int j = lane & 15;
     const float scale = norm_scale(ss_in, j < T ? j : 0, E, eps);
     float acc[1][4];
-    wave_tile_kloop<WT, true>(w, tile, xprep, normprep, scale, acc[0]);
+    const WMat2* ws[1] = {&w};
+    wave_tile_kloop<WT, true, 1>(ws, tile, xprep, normprep, scale, acc);
     __shared__ float lds[3 * 64 * 4];
     combine_acc<1>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;
@@ -961,72 +1064,9 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     const int j = lane & 15;
     const float scale = norm_scale(ss_in, j < T ? j : 0, w1.cols, eps);
     float acc[2][4];
-    {
-        KLoop kl;
-        const int nb = w1.cols >> 5;
-        kl.init(nb);
-        acc[0][0] = acc[0][1] = acc[0][2] = acc[0][3] = 0.0f;
-        acc[1][0] = acc[1][1] = acc[1][2] = acc[1][3] = 0.0f;
-        const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-        for (int kb = kl.kb0; kb < kl.kb1; ++kb) {
-            ABFrag b;
-            const float sumB = b_frag<true>(xprep, normprep, scale, kb,
-                                            kl.i, kl.ks, b, true);
-#pragma unroll
-            for (int m = 0; m < 2; ++m) {
-                const WMat2& w = m ? w3 : w1;
-                ABFrag a;
-                if (WT == W_F16) {
-                    const unsigned short* base =
-                        (const unsigned short*)w.data +
-                        ((size_t)blockIdx.x * (w.cols >> 3)) * 128;
-                    const int kc = kb * 4 + kl.ks;
-                    const uint4 aw = *reinterpret_cast<const uint4*>(
-                        base + ((size_t)kc * 16 + kl.i) * 8);
-                    a.u[0] = aw.x; a.u[1] = aw.y;
-                    a.u[2] = aw.z; a.u[3] = aw.w;
-                    f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                        a.v, b.v, zero, 0, 0, 0);
-#pragma unroll
-                    for (int jj = 0; jj < 4; ++jj) acc[m][jj] += d[jj];
-                    continue;
-                }
-                const uint32_t* qbase =
-                    (const uint32_t*)w.data + ((size_t)blockIdx.x * nb) * 64;
-                const uint32_t q = __builtin_nontemporal_load(
-                    qbase + (size_t)kb * 64 + kl.ks * 16 + kl.i);
-                a_frag_q4(q, a);
-                f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a.v, b.v, zero, 0, 0, 0);
-                if (WT == W_Q4_0) {
-                    const float4 sc = *reinterpret_cast<const float4*>(
-                        (const float*)w.scales +
-                        ((size_t)blockIdx.x * nb + kb) * 16 +
-                        (kl.lane >> 4) * 4);
-                    const float s4[4] = {sc.x, sc.y, sc.z, sc.w};
-#pragma unroll
-                    for (int jj = 0; jj < 4; ++jj)
-                        acc[m][jj] = fmaf(
-                            s4[jj], fmaf(-136.0f, sumB, d[jj]), acc[m][jj]);
-                } else {
-                    const float* sp = (const float*)w.scales +
-                        (((size_t)blockIdx.x * nb + kb) * 16 +
-                         (kl.lane >> 4) * 4) * 2;
-                    const float4 s01 = *reinterpret_cast<const float4*>(sp);
-                    const float4 s23 =
-                        *reinterpret_cast<const float4*>(sp + 4);
-                    const float dd[4] = {s01.x, s01.z, s23.x, s23.z};
-                    const float mm[4] = {s01.y, s01.w, s23.y, s23.w};
-#pragma unroll
-                    for (int jj = 0; jj < 4; ++jj) {
-                        acc[m][jj] = fmaf(
-                            dd[jj], fmaf(-128.0f, sumB, d[jj]), acc[m][jj]);
-                        acc[m][jj] = fmaf(mm[jj], sumB, acc[m][jj]);
-                    }
-                }
-            }
-        }
-    }
+    const WMat2* ws[2] = {&w1, &w3};
+    wave_tile_kloop<WT, true, 2>(ws, blockIdx.x, xprep, normprep, scale,
+                                 acc);
     __shared__ float lds[3 * 64 * 4 * 2];
     combine_acc<2>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;

@@ -1,0 +1,99 @@
+"""Tokenizer greedy-BPE behavior + sampler behavioral parity."""
+import numpy as np
+import pytest
+
+from distributedllm_amd.engine.sampler import Sampler, softmax
+from distributedllm_amd.engine.tokenizer import (
+    BOS_ID, SPM_SPACE, Tokenizer)
+from distributedllm_amd.formats.synthetic import synthetic_vocab
+
+
+def _vocab(extra=None):
+    v = [(b"<unk>", 0.0), (b"<s>", 0.0), (b"</s>", 0.0)]
+    v += [(f"<0x{b:02X}>".encode(), 0.0) for b in range(256)]
+    for i, (tok, score) in enumerate(extra or []):
+        v.append((tok, score))
+    return v
+
+
+class TestTokenizer:
+    def test_byte_fallback(self):
+        tk = Tokenizer(_vocab())
+        ids = tk.encode("ab", bos=False)
+        assert ids == [ord("a") + 3, ord("b") + 3]
+
+    def test_bos(self):
+        tk = Tokenizer(_vocab())
+        assert tk.encode("a")[0] == BOS_ID
+        assert tk.encode("", bos=True) == [BOS_ID]
+        assert tk.encode("", bos=False) == []
+
+    def test_greedy_merge_prefers_higher_score(self):
+        # "ab" and "bc" both in vocab; "ab" scores higher -> a|b merge wins
+        tk = Tokenizer(_vocab([(b"ab", -1.0), (b"bc", -5.0)]))
+        ids = tk.encode("abc", bos=False)
+        ab = tk.token_to_id[b"ab"]
+        assert ids == [ab, ord("c") + 3]
+
+    def test_recursive_merges(self):
+        tk = Tokenizer(_vocab([(b"he", -1.0), (b"ll", -2.0), (b"hell", -3.0),
+                               (b"hello", -4.0), (b"o", -0.5)]))
+        ids = tk.encode("hello", bos=False)
+        assert ids == [tk.token_to_id[b"hello"]]
+
+    def test_multibyte_utf8(self):
+        # greedy BPE needs the intermediate merges in-vocab (just like the
+        # reference: only existing-pair merges are enqueued)
+        word = SPM_SPACE + "the"
+        inter = [((SPM_SPACE + "t").encode(), -9.0),
+                 ((SPM_SPACE + "th").encode(), -5.0),
+                 (word.encode(), -1.0)]
+        tk = Tokenizer(_vocab(inter))
+        ids = tk.encode(SPM_SPACE + "the", bos=False)
+        assert ids == [tk.token_to_id[word.encode()]]
+
+    def test_decode_roundtrip_bytes(self):
+        tk = Tokenizer(_vocab())
+        ids = tk.encode("hi there", bos=True)
+        assert tk.decode(ids) == "hi there"
+
+    def test_decode_spm_space(self):
+        tk = Tokenizer(synthetic_vocab(400))
+        wid = tk.token_to_id[("▁the").encode()]
+        assert tk.decode_token(wid) == " the"
+
+
+class TestSampler:
+    def test_greedy(self):
+        s = Sampler(greedy=True)
+        logits = np.array([0.1, 2.0, -1.0])
+        assert s(logits) == 1
+        assert s.previous_ids == [1]
+
+    def test_penalty_divides_logits(self):
+        # reference semantics incl. the negative-logit quirk: after token 0
+        # is sampled, its (negative) logit gets divided -> boosted
+        s = Sampler(temperature=1.0, repeat_penalty=2.0, seed=0)
+        s.previous_ids = [0]
+        logits = np.array([-4.0, 0.0, 0.0])
+        mask = np.isin(np.arange(3), s.previous_ids)
+        penalties = (mask * 2.0 + ~mask) * (1.0 + s.EPS)
+        expected = softmax(logits / penalties)
+        assert expected[0] > softmax(logits / (1.0 + s.EPS))[0]
+
+    def test_distribution_matches_reference_formula(self):
+        rng_draws = []
+        s = Sampler(temperature=0.5, repeat_penalty=1.3, seed=42)
+        logits = np.linspace(-1, 1, 16)
+        for _ in range(200):
+            rng_draws.append(s(logits))
+        assert set(rng_draws) <= set(range(16))
+        # temperature sharpening: top logit dominates draws
+        assert np.bincount(rng_draws, minlength=16)[15] > 50
+
+    def test_deterministic_with_seed(self):
+        a = Sampler(seed=7)
+        b = Sampler(seed=7)
+        logits = np.linspace(-1, 1, 32)
+        assert [a(logits) for _ in range(10)] == \
+               [b(logits) for _ in range(10)]
