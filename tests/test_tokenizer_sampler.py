@@ -82,14 +82,22 @@ class TestSampler:
         assert expected[0] > softmax(logits / (1.0 + s.EPS))[0]
 
     def test_distribution_matches_reference_formula(self):
-        rng_draws = []
-        s = Sampler(temperature=0.5, repeat_penalty=1.3, seed=42)
+        # fresh sampler per draw (the penalty list accumulates every
+        # sampled id, reference common.py:84-85)
         logits = np.linspace(-1, 1, 16)
-        for _ in range(200):
-            rng_draws.append(s(logits))
-        assert set(rng_draws) <= set(range(16))
+        draws = [Sampler(temperature=0.5, repeat_penalty=1.3, seed=i)(logits)
+                 for i in range(200)]
+        assert set(draws) <= set(range(16))
         # temperature sharpening: top logit dominates draws
-        assert np.bincount(rng_draws, minlength=16)[15] > 50
+        assert np.bincount(draws, minlength=16)[15] > 80
+
+    def test_penalty_accumulates_previous_ids(self):
+        s = Sampler(temperature=1.0, repeat_penalty=5.0, seed=0)
+        logits = np.zeros(4)
+        logits[2] = 10.0
+        first = s(logits)
+        assert first == 2
+        assert s.previous_ids == [2]
 
     def test_deterministic_with_seed(self):
         a = Sampler(seed=7)
