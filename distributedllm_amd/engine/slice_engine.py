@@ -67,18 +67,22 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             qword |= n[:, :, :, j] << ((j % 2) * 16 + (j // 2) * 4)
         qs2 = np.ascontiguousarray(
             qword.reshape(R, 16, nb, 4).transpose(0, 2, 3, 1))
+        # per-(row, block) prescale pair: w = alpha*(n+128) + beta
+        # (q4_0: alpha=d, beta=-136d; q4_1: alpha=d, beta=m-128d)
         bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
         a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
         if t.gtype == ggml.GGML_TYPE_Q4_0:
             d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
             d = d.reshape(rows, nb).astype(np.float32)
-            scales = np.ascontiguousarray(
-                d.reshape(R, 16, nb).transpose(0, 2, 1))
+            alpha, beta = d, -136.0 * d
         else:
             dm = np.ascontiguousarray(a[:, :, :4]).view(np.float16)
             dm = dm.reshape(rows, nb, 2).astype(np.float32)
-            scales = np.ascontiguousarray(
-                dm.reshape(R, 16, nb, 2).transpose(0, 2, 1, 3))
+            alpha = dm[:, :, 0]
+            beta = dm[:, :, 1] - 128.0 * dm[:, :, 0]
+        ab = np.stack([alpha, beta], axis=-1).astype(np.float16)
+        scales = np.ascontiguousarray(
+            ab.reshape(R, 16, nb, 2).transpose(0, 2, 1, 3))
         data = torch.from_numpy(qs2.view(np.int32)).to(device)
         sc = torch.from_numpy(scales).to(device)
         return data, sc, t.gtype
@@ -176,13 +180,13 @@ class HIPSliceEngine:
             # byte the same compute/HBM traffic as a real checkpoint
             if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
                 R, nb = rows // 16, cols // 32
-                per = 2 if wt == ggml.GGML_TYPE_Q4_1 else 1
                 data = torch.randint(-2**31, 2**31 - 1, (R * nb * 64,),
                                      dtype=torch.int32, device="cuda",
                                      generator=g)
-                scales = ((torch.rand(R * nb * 16 * per, device="cuda",
-                                      generator=g) * 0.5 + 0.75) * 0.003)
-                return data, scales.contiguous(), wt
+                alpha = ((torch.rand(R * nb * 16, 1, device="cuda",
+                                     generator=g) * 0.5 + 0.75) * 0.003)
+                ab = torch.cat([alpha, -136.0 * alpha], dim=1)
+                return data, ab.to(torch.float16).contiguous(), wt
             if wt == ggml.GGML_TYPE_F16:
                 data = (torch.randn(rows * cols, device="cuda", generator=g,
                                     dtype=torch.float32) * 0.02)

@@ -77,13 +77,11 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
         TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
                         data.numel() == R * nb * 64,
                     "q4 tiled data must be u32[R][nb][4][16]");
-        const int64_t per = (wtype == W_Q4_1) ? 2 : 1;
         TORCH_CHECK(scales.defined() && scales.is_cuda() &&
                         scales.is_contiguous() &&
-                        scales.scalar_type() == torch::kFloat32 &&
-                        scales.numel() == R * nb * 16 * per,
-                    "q4 tiled scales must be f32[R][nb][16]",
-                    (wtype == W_Q4_1 ? "x2" : ""));
+                        scales.scalar_type() == torch::kFloat16 &&
+                        scales.numel() == R * nb * 16 * 2,
+                    "q4 tiled scales must be f16 (alpha,beta)[R][nb][16]");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {

@@ -651,14 +651,6 @@ __device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
     return c.u;
 }
 
-// Build the A fragment from one repacked nibble word (8 weights as n+128).
-__device__ __forceinline__ void a_frag_q4(uint32_t q, ABFrag& a) {
-    a.u[0] = 0x43004300u | (q & 0x000F000Fu);
-    a.u[1] = 0x43004300u | ((q >> 4) & 0x000F000Fu);
-    a.u[2] = 0x43004300u | ((q >> 8) & 0x000F000Fu);
-    a.u[3] = 0x43004300u | ((q >> 12) & 0x000F000Fu);
-}
-
 // Per-wave state for the K loop: lane (i = l&15 row, ks = l>>4 k-span).
 struct KLoop {
     int lane, i, ks;
@@ -674,16 +666,14 @@ struct KLoop {
     }
 };
 
-// Load + (for the norm path) scale the B fragment for block kb.
-// Returns sumB (sum over the full K=32 of this lane's column) via xor-shfl.
+// Load + (for the norm path) RMS-normalize the B fragment for block kb.
 template <bool NORM>
-__device__ __forceinline__ float b_frag(
+__device__ __forceinline__ void b_frag(
     const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep, float scale, int kb,
-    int i /*col j = i*/, int ks, ABFrag& b, bool need_sum) {
+    int i /*col j = i*/, int ks, ABFrag& b) {
     const int kc = kb * 4 + ks;
     const uint4 xb = *reinterpret_cast<const uint4*>(xprep + (kc * 16 + i) * 8);
-    float sum = 0.0f;
     if (NORM) {
         const uint4 nb_ = *reinterpret_cast<const uint4*>(normprep + kc * 8);
         const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
@@ -693,68 +683,45 @@ __device__ __forceinline__ float b_frag(
             const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
             const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
             b.u[w] = pack_bf16(f0, f1);
-            sum += f0 + f1;
         }
     } else {
-        b.u[0] = xb.x;
-        b.u[1] = xb.y;
-        b.u[2] = xb.z;
-        b.u[3] = xb.w;
-        if (need_sum) {
-#pragma unroll
-            for (int w = 0; w < 4; ++w)
-                sum += bflo(b.u[w]) + bfhi(b.u[w]);
-        }
+        b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
     }
-    if (need_sum) {
-        // reduce over the 4 k-spans (lanes l, l^16, l^32, l^48 share col i)
-        sum += __shfl_xor(sum, 16);
-        sum += __shfl_xor(sum, 32);
-    }
-    return sum;
 }
 
-// Per-block scale record + its application to the 16x16 MFMA result.
-template <int WT>
-struct AScale {
-    float4 a, b;  // q4_0: a only; q4_1: (d,m) pairs across a and b
-    __device__ __forceinline__ void load(const float* scales, int nb,
-                                         size_t tile_row, int kb, int grp) {
-        if (WT == W_Q4_0) {
-            a = *reinterpret_cast<const float4*>(
-                scales + (tile_row * nb + kb) * 16 + grp * 4);
-        } else if (WT == W_Q4_1) {
-            const float* sp =
-                scales + ((tile_row * nb + kb) * 16 + grp * 4) * 2;
-            a = *reinterpret_cast<const float4*>(sp);
-            b = *reinterpret_cast<const float4*>(sp + 4);
-        }
-    }
-    __device__ __forceinline__ void apply(const f32x4& d, float sumB,
-                                          float acc[4]) const {
-        if (WT == W_Q4_0) {
-            const float s4[4] = {a.x, a.y, a.z, a.w};
+// Dequantize one repacked nibble word into a prescaled bf16 A fragment:
+// w = alpha * (n+128) + beta, with the per-(row, block) (alpha, beta) f16
+// pair precomputed at repack time (q4_0: alpha=d, beta=-136d; q4_1:
+// alpha=d, beta=m-128d). Prescaling on the A side keeps the MFMA
+// C-chainable (AGPR accumulate, no per-block VALU fixup, no cross-lane
+// bias reduction) and unifies q4_0/q4_1 into one kernel path.
+__device__ __forceinline__ void a_frag_q4_scaled(uint32_t q, uint32_t ab,
+                                                 ABFrag& a) {
+    // ab = (beta_f16 << 16) | alpha_f16
+    union { __half2 h; uint32_t u; } c;
+    c.u = ab;
+    const float al = __half2float(__low2half(c.h));
+    const float be = __half2float(__high2half(c.h));
+    uint32_t w[4];
+    w[0] = 0x43004300u | (q & 0x000F000Fu);
+    w[1] = 0x43004300u | ((q >> 4) & 0x000F000Fu);
+    w[2] = 0x43004300u | ((q >> 8) & 0x000F000Fu);
+    w[3] = 0x43004300u | ((q >> 12) & 0x000F000Fu);
 #pragma unroll
-            for (int jj = 0; jj < 4; ++jj)
-                acc[jj] = fmaf(s4[jj], fmaf(-136.0f, sumB, d[jj]), acc[jj]);
-        } else if (WT == W_Q4_1) {
-            const float dd[4] = {a.x, a.z, b.x, b.z};
-            const float mm[4] = {a.y, a.w, b.y, b.w};
-#pragma unroll
-            for (int jj = 0; jj < 4; ++jj) {
-                acc[jj] = fmaf(dd[jj], fmaf(-128.0f, sumB, d[jj]), acc[jj]);
-                acc[jj] = fmaf(mm[jj], sumB, acc[jj]);
-            }
-        }
+    for (int i = 0; i < 4; ++i) {
+        const float f0 = fmaf(al, bflo(w[i]), be);
+        const float f1 = fmaf(al, bfhi(w[i]), be);
+        a.u[i] = pack_bf16(f0, f1);
     }
-};
+}
 
 // One wave's software-pipelined K loop over NM matrices sharing the B
-// panel (NM=2 for the FFN's w1/w3 — halves B traffic and doubles the
-// MFMA work per load batch). PF blocks' loads are issued together before
-// any compute so ~PF*NM*256B of weight stream stays in flight per wave —
-// the un-pipelined form was HBM-latency-bound at <1 TB/s.
-// acc[n][jj] accumulates rows (l>>4)*4 + jj, col l&15 of output n.
+// panel (NM=2 for the FFN's w1/w3 — halves B traffic and doubles the MFMA
+// work per load batch). PF blocks' loads are issued together before any
+// compute so ~PF*NM*256B of weight stream stays in flight per wave.
+// Accumulation is MFMA C-chained over two alternating accumulators per
+// matrix (covers the dependent-accumulator latency).
+// acc[n][jj] ends with rows (l>>4)*4 + jj, col l&15 of output n.
 template <int WT, bool NORM, int NM, int PF = 4>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
@@ -764,28 +731,29 @@ __device__ __forceinline__ void wave_tile_kloop(
     KLoop kl;
     const int nb = ws[0]->cols >> 5;
     kl.init(nb);
-#pragma unroll
-    for (int n = 0; n < NM; ++n)
-        acc[n][0] = acc[n][1] = acc[n][2] = acc[n][3] = 0.0f;
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-    const int grp = kl.lane >> 4;
+    f32x4 c0[NM], c1[NM];
+#pragma unroll
+    for (int n = 0; n < NM; ++n) {
+        c0[n] = zero;
+        c1[n] = zero;
+    }
 
     const uint32_t* qbase[NM];
     const unsigned short* tbase[NM];
-    const float* scbase[NM];
+    const uint32_t* abbase[NM];
 #pragma unroll
     for (int n = 0; n < NM; ++n) {
         qbase[n] = (const uint32_t*)ws[n]->data + ((size_t)tile_row * nb) * 64;
         tbase[n] = (const unsigned short*)ws[n]->data +
                    ((size_t)tile_row * (ws[n]->cols >> 3)) * 128;
-        scbase[n] = (const float*)ws[n]->scales;
+        abbase[n] = (const uint32_t*)ws[n]->scales + (size_t)tile_row * nb * 16;
     }
 
-    auto compute_one = [&](const uint32_t q[NM], const uint4 aw[NM],
-                           const uint4& xb, const uint4& nbv,
-                           const AScale<WT> sc[NM]) {
+    auto compute_one = [&](int parity, const uint32_t q[NM],
+                           const uint32_t ab[NM], const uint4 aw[NM],
+                           const uint4& xb, const uint4& nbv) {
         ABFrag b;
-        float sumB = 0.0f;
         if (NORM) {
             const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
             const uint32_t nw[4] = {nbv.x, nbv.y, nbv.z, nbv.w};
@@ -794,19 +762,9 @@ __device__ __forceinline__ void wave_tile_kloop(
                 const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
                 const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
                 b.u[w] = pack_bf16(f0, f1);
-                sumB += f0 + f1;
             }
         } else {
             b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
-            if (WT != W_F16) {
-#pragma unroll
-                for (int w = 0; w < 4; ++w)
-                    sumB += bflo(b.u[w]) + bfhi(b.u[w]);
-            }
-        }
-        if (WT != W_F16) {
-            sumB += __shfl_xor(sumB, 16);
-            sumB += __shfl_xor(sumB, 32);
         }
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
@@ -814,26 +772,19 @@ __device__ __forceinline__ void wave_tile_kloop(
             if (WT == W_F16) {
                 a.u[0] = aw[n].x; a.u[1] = aw[n].y;
                 a.u[2] = aw[n].z; a.u[3] = aw[n].w;
-                f32x4 c = {acc[n][0], acc[n][1], acc[n][2], acc[n][3]};
-                c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c,
-                                                            0, 0, 0);
-                acc[n][0] = c[0]; acc[n][1] = c[1];
-                acc[n][2] = c[2]; acc[n][3] = c[3];
             } else {
-                a_frag_q4(q[n], a);
-                const f32x4 d = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    a.v, b.v, zero, 0, 0, 0);
-                sc[n].apply(d, sumB, acc[n]);
+                a_frag_q4_scaled(q[n], ab[n], a);
             }
+            f32x4& c = parity ? c1[n] : c0[n];
+            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c, 0, 0, 0);
         }
     };
 
     int g = kl.kb0;
     for (; g + PF <= kl.kb1; g += PF) {
-        uint32_t q[PF][NM];
+        uint32_t q[PF][NM], ab[PF][NM];
         uint4 aw[PF][NM];
         uint4 xb[PF], nbv[PF];
-        AScale<WT> sc[PF][NM];
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
             const int kb = g + u;
@@ -850,19 +801,19 @@ __device__ __forceinline__ void wave_tile_kloop(
                 } else {
                     q[u][n] = __builtin_nontemporal_load(
                         qbase[n] + (size_t)kb * 64 + kl.ks * 16 + kl.i);
-                    sc[u][n].load(scbase[n], nb, tile_row, kb, grp);
+                    ab[u][n] = __builtin_nontemporal_load(
+                        abbase[n] + (size_t)kb * 16 + kl.i);
                 }
             }
         }
 #pragma unroll
         for (int u = 0; u < PF; ++u)
-            compute_one(q[u], aw[u], xb[u], nbv[u], sc[u]);
+            compute_one(u & 1, q[u], ab[u], aw[u], xb[u], nbv[u]);
     }
     for (; g < kl.kb1; ++g) {
-        uint32_t q[NM];
+        uint32_t q[NM], ab[NM];
         uint4 aw[NM];
         uint4 xb, nbv;
-        AScale<WT> sc[NM];
         const int kc = g * 4 + kl.ks;
         xb = *reinterpret_cast<const uint4*>(
             xprep + ((size_t)kc * 16 + kl.i) * 8);
@@ -876,11 +827,17 @@ __device__ __forceinline__ void wave_tile_kloop(
             } else {
                 q[n] = __builtin_nontemporal_load(
                     qbase[n] + (size_t)g * 64 + kl.ks * 16 + kl.i);
-                sc[n].load(scbase[n], nb, tile_row, g, grp);
+                ab[n] = __builtin_nontemporal_load(
+                    abbase[n] + (size_t)g * 16 + kl.i);
             }
         }
-        compute_one(q, aw, xb, nbv, sc);
+        compute_one(g & 1, q, ab, aw, xb, nbv);
     }
+#pragma unroll
+    for (int n = 0; n < NM; ++n)
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj)
+            acc[n][jj] = c0[n][jj] + c1[n][jj];
 }
 
 // LDS combine of the 4 waves' partial accumulators; wave 0 ends with the
