@@ -1,0 +1,226 @@
+"""All CLI commands.
+
+Command-for-command parity with the reference CLI
+(/root/reference/distllm/cli_api/__init__.py:9-24 and the per-command
+files): provision, run_node, status, push_slice, load_slice, list_slices,
+generate_text, perplexity, run_proxy.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import threading
+
+from .base import Command
+
+
+def _progress(label: str):
+    def cb(sent: int, total: int) -> None:
+        pct = 100.0 * sent / max(total, 1)
+        print(f"\r[{label}] {sent}/{total} bytes ({pct:5.1f}%)",
+              end="", flush=True)
+        if sent >= total:
+            print()
+    return cb
+
+
+class ProvisionCommand(Command):
+    name = "provision"
+    help = ("Prepare a model (convert/requantize), slice it per the "
+            "config's nodes_map, record it in the registry, push slices "
+            "to the nodes")
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("config", help="cluster+model JSON config")
+        p.add_argument("--root", default=".",
+                       help="working root (models/ + models_registry/)")
+        p.add_argument("--no-push", action="store_true",
+                       help="slice + register only, do not upload to nodes")
+
+    def __call__(self, args) -> int:
+        from ..cluster.provision import provision
+        entry = provision(args.config, root=args.root, push=not args.no_push,
+                          progress=_progress("push"))
+        print(f"provisioned {entry.model_id}: "
+              f"{len(entry.slices)} slice(s) in {entry.model_dir}")
+        return 0
+
+
+class RunNodeCommand(Command):
+    name = "run_node"
+    help = "Start a compute node (TCP server, or reverse-connect to a proxy)"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("--host", default="0.0.0.0")
+        p.add_argument("--port", type=int, default=9998)
+        p.add_argument("--uploads_dir", default="uploads")
+        p.add_argument("--device", default=None,
+                       help="force engine device (cuda/cpu); default auto")
+        p.add_argument("--n-ctx", type=int, default=2048)
+        p.add_argument("--reverse", action="store_true",
+                       help="dial out to a proxy instead of listening")
+        p.add_argument("--proxy-host", default="127.0.0.1")
+        p.add_argument("--proxy-port", type=int, default=9999)
+        p.add_argument("--name", default="node")
+
+    def __call__(self, args) -> int:
+        from ..cluster.node import NodeState, run_server
+        if args.reverse:
+            from ..cluster.proxy import connect_then_serve
+            state = NodeState(args.uploads_dir, device=args.device,
+                              n_ctx=args.n_ctx)
+            print(f"[node] reverse-connecting to "
+                  f"{args.proxy_host}:{args.proxy_port}")
+            connect_then_serve(args.proxy_host, args.proxy_port, state,
+                               name=args.name)
+            return 0
+        run_server(args.host, args.port, args.uploads_dir,
+                   device=args.device, n_ctx=args.n_ctx)
+        return 0
+
+
+class _NodeCommand(Command):
+    """Base for commands addressing one node as host:port."""
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("address", help="node address host:port")
+
+    def _conn(self, args):
+        from ..cluster.client import Connection, parse_address
+        host, port = parse_address(args.address)
+        return Connection(host, port)
+
+
+class StatusCommand(_NodeCommand):
+    name = "status"
+    help = "Report a node's status (loaded slice, device)"
+
+    def __call__(self, args) -> int:
+        conn = self._conn(args)
+        s = conn.get_status()
+        print(json.dumps({"status": s.status, "model": s.model,
+                          "first_layer": s.first_layer,
+                          "n_layers": s.n_layers, "device": s.device},
+                         indent=1))
+        conn.close()
+        return 0
+
+
+class PushSliceCommand(_NodeCommand):
+    name = "push_slice"
+    help = "Upload a slice file to a node (chunked, checksummed)"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        super().configure(p)
+        p.add_argument("path", help="slice file to upload")
+        p.add_argument("--metadata", default="{}",
+                       help="JSON metadata (e.g. '{\"format\": \"test\"}')")
+
+    def __call__(self, args) -> int:
+        conn = self._conn(args)
+        resp = conn.push_slice(args.path, json.loads(args.metadata),
+                               progress=_progress("push"))
+        print(f"uploaded {resp.name} ({resp.total_size} bytes)")
+        conn.close()
+        return 0
+
+
+class LoadSliceCommand(_NodeCommand):
+    name = "load_slice"
+    help = "Load an uploaded slice into the node's engine"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        super().configure(p)
+        p.add_argument("slice_name")
+
+    def __call__(self, args) -> int:
+        conn = self._conn(args)
+        resp = conn.load_slice(args.slice_name)
+        print(f"loaded {resp.name}: layers "
+              f"[{resp.first_layer}, {resp.first_layer + resp.n_layers - 1}]")
+        conn.close()
+        return 0
+
+
+class ListSlicesCommand(_NodeCommand):
+    name = "list_slices"
+    help = "List slices uploaded to a node"
+
+    def __call__(self, args) -> int:
+        conn = self._conn(args)
+        for s in conn.list_slices():
+            print(json.dumps(s))
+        conn.close()
+        return 0
+
+
+class GenerateTextCommand(Command):
+    name = "generate_text"
+    help = "Generate text with a provisioned distributed model"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("config", help="cluster+model JSON config")
+        p.add_argument("--prompt", required=True)
+        p.add_argument("--num-tokens", type=int, default=50)
+        p.add_argument("--temp", type=float, default=0.7)
+        p.add_argument("--rp", type=float, default=1.1,
+                       help="repetition penalty")
+        p.add_argument("--greedy", action="store_true")
+        p.add_argument("--seed", type=int, default=None)
+        p.add_argument("--root", default=".")
+
+    def __call__(self, args) -> int:
+        from ..cluster.llm_client import get_llm
+        llm = get_llm(args.config, root=args.root)
+        print(args.prompt, end="", flush=True)
+        for piece in llm.generate(args.prompt, max_steps=args.num_tokens,
+                                  temperature=args.temp,
+                                  repeat_penalty=args.rp,
+                                  greedy=args.greedy, seed=args.seed):
+            print(piece, end="", flush=True)
+        print()
+        return 0
+
+
+class PerplexityCommand(Command):
+    name = "perplexity"
+    help = "Perplexity of a text under a provisioned distributed model"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("config")
+        p.add_argument("--prompt", default=None)
+        p.add_argument("--file", default=None,
+                       help="read the text from a file instead")
+        p.add_argument("--root", default=".")
+
+    def __call__(self, args) -> int:
+        if (args.prompt is None) == (args.file is None):
+            print("provide exactly one of --prompt / --file",
+                  file=sys.stderr)
+            return 2
+        text = args.prompt
+        if args.file:
+            with open(args.file) as f:
+                text = f.read()
+        from ..cluster.llm_client import get_llm
+        llm = get_llm(args.config, root=args.root)
+        ppl = llm.perplexity(text)
+        print(f"perplexity: {ppl:.4f}")
+        return 0
+
+
+class RunProxyCommand(Command):
+    name = "run_proxy"
+    help = "Run the NAT-traversal proxy (bridges clients to a reverse node)"
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("--host", default="0.0.0.0")
+        p.add_argument("--client-port", type=int, default=9997)
+        p.add_argument("--node-port", type=int, default=9999)
+
+    def __call__(self, args) -> int:
+        from ..cluster.proxy import run_proxy
+        run_proxy(args.host, args.client_port, args.node_port)
+        threading.Event().wait()  # serve until killed
+        return 0

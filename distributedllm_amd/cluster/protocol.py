@@ -189,7 +189,18 @@ class RequestClearContext(Message):
     pass
 
 
+@dataclass
+class RequestGreeting(Message):
+    """Reverse-connect handshake: a node announcing itself to a proxy
+    (reference serve.py:49-56)."""
+    name: str = "node"
+
+
 # responses
+@dataclass
+class ResponseGreeting(Message):
+    status: str = "ok"
+
 @dataclass
 class ResponseStatus(Message):
     status: str = "up"

@@ -140,9 +140,10 @@ def provision(config_path: str, root: str = ".", push: bool = True,
             existing = {e["name"] for e in conn.list_slices()}
             if name not in existing:
                 conn.push_slice(s.path,
-                                metadata={"name": name, "model": model_id,
+                                metadata={**meta, "name": name,
+                                          "model": model_id,
                                           "a": s.a, "b": s.b,
-                                          "format": "ggml", **meta},
+                                          "format": "ggml"},
                                 progress=progress)
             conn.close()
     return entry
