@@ -156,6 +156,7 @@ class SliceEngine {
         ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * kMaxTokens}, f32);
         ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
+        argmax_keys_ = torch::zeros({kMaxTokens}, dev.dtype(torch::kInt64));
     }
 
     void set_layer(int64_t li, torch::Tensor attn_norm,
@@ -265,12 +266,30 @@ class SliceEngine {
                          qb, kc, vc, pp, sp, ifr, E_, D_, ctx_, T);
             launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_, D_,
                              ctx_);
-            launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
-                          xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
+            // wo/w2 tile count (E/16) alone underfills 256 CUs — split K
+            // across grid.y with atomic partials, then rebuild the
+            // sumsq/xprep side channel with the (cheap) prep pass.
+            const bool split = (E_ / 16) < 512;
+            if (split) {
+                launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
+                              nullptr, nullptr, T, GM_ATOMIC);
+                launch_prep_x(s, xp, xprep, ssf + li * kMaxTokens, E_, T);
+            } else {
+                launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
+                              xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
+            }
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
                          ssf + li * kMaxTokens, eps_, gprep, T);
-            launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
-                          xprep, ssa + (li + 1) * kMaxTokens, T, GM_RES_SQ);
+            if (split) {
+                launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
+                              nullptr, nullptr, T, GM_ATOMIC);
+                launch_prep_x(s, xp, xprep, ssa + (li + 1) * kMaxTokens, E_,
+                              T);
+            } else {
+                launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
+                              xprep, ssa + (li + 1) * kMaxTokens, T,
+                              GM_RES_SQ);
+            }
         }
         return x;
     }
@@ -326,9 +345,12 @@ class SliceEngine {
         auto out = torch::empty(
             {T},
             torch::TensorOptions().device(torch::kCUDA).dtype(torch::kInt32));
+        TORCH_CHECK(T <= kMaxTokens, "argmax: too many rows");
         hipStream_t s = c10::hip::getCurrentHIPStream().stream();
-        launch_argmax(s, lg.data_ptr<float>(), out.data_ptr<int>(), T,
-                      (int)lg.size(1));
+        launch_argmax(
+            s, lg.data_ptr<float>(),
+            reinterpret_cast<unsigned long long*>(argmax_keys_.data_ptr()),
+            out.data_ptr<int>(), T, (int)lg.size(1));
         return out;
     }
 
@@ -370,6 +392,7 @@ class SliceEngine {
     torch::Tensor k_cache_, v_cache_, inv_freq_;
     torch::Tensor xn_, qb_, ab_, ffb_;
     torch::Tensor xprep_, aprep_, gprep_, ss_attn_, ss_ffn_, ss_tmp_;
+    torch::Tensor argmax_keys_;
     bool has_extra_ = false;
     bool out_mfma_ = false;
     DevMat tok_, out_;

@@ -29,8 +29,10 @@ struct WMat2 {
     int wtype;  // WType (W_F32 never appears here — legacy path)
 };
 
-// k_gemm16 fused-epilogue modes
-enum GemmMode { GM_PLAIN = 0, GM_RES_SQ = 1, GM_NORM_PLAIN = 2 };
+// k_gemm16 fused-epilogue modes. GM_ATOMIC = grid-level split-K partials
+// atomicAdd'ed into y (residual pre-loaded); follow with launch_prep_x.
+enum GemmMode { GM_PLAIN = 0, GM_RES_SQ = 1, GM_NORM_PLAIN = 2,
+                GM_ATOMIC = 3 };
 
 void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T);
@@ -78,5 +80,5 @@ void launch_ffn_gate(hipStream_t s, const WMat& w1, const WMat& w3,
 void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
                   float* out, int T, int E);
 
-void launch_argmax(hipStream_t s, const float* logits, int* out, int T,
-                   int V);
+void launch_argmax(hipStream_t s, const float* logits,
+                   unsigned long long* keys, int* out, int T, int V);

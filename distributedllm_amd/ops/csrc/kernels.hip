@@ -571,37 +571,51 @@ __global__ void k_embed(WMat tab, const int* __restrict__ tokens,
 // tensor_processor.cpp:1894-1908), on-device so greedy decode never copies
 // the [V] logits to host.
 
-__global__ void k_argmax(const float* __restrict__ logits,
-                         int* __restrict__ out, int V) {
+// Two-stage: grid (T, NPART) partial blocks each reduce a V/NPART chunk
+// and atomicMax a packed u64 key (ordered-f32 ‖ bit-inverted index — the
+// index inversion makes ties resolve to the SMALLEST index, matching the
+// reference's first-max scan); a tiny decode kernel unpacks. One block per
+// row (the previous form) used only T CUs of 256 and measured 40 µs at
+// T=4, V=32000.
+__device__ __forceinline__ uint32_t ordered_f32(float v) {
+    union { float f; uint32_t u; } c;
+    c.f = v;
+    return (c.u & 0x80000000u) ? ~c.u : (c.u | 0x80000000u);
+}
+
+__global__ void k_argmax_part(const float* __restrict__ logits,
+                              unsigned long long* __restrict__ keys, int V) {
     const int t = blockIdx.x;
     const float* lt = logits + (size_t)t * V;
+    const int chunk = (V + gridDim.y - 1) / gridDim.y;
+    const int i0 = blockIdx.y * chunk;
+    const int i1 = min(V, i0 + chunk);
     float best = -INFINITY;
-    int bi = 0;
-    for (int i = threadIdx.x; i < V; i += BLOCK) {
+    int bi = i0;
+    for (int i = i0 + threadIdx.x; i < i1; i += BLOCK) {
         const float v = lt[i];
         if (v > best || (v == best && i < bi)) {
             best = v;
             bi = i;
         }
     }
-    __shared__ float sv[BLOCK];
-    __shared__ int si[BLOCK];
-    sv[threadIdx.x] = best;
-    si[threadIdx.x] = bi;
+    unsigned long long key =
+        ((unsigned long long)ordered_f32(best) << 32) | (uint32_t)(~bi);
+    __shared__ unsigned long long sk[BLOCK];
+    sk[threadIdx.x] = key;
     __syncthreads();
     for (int stride = BLOCK / 2; stride > 0; stride >>= 1) {
-        if (threadIdx.x < stride) {
-            const float ov = sv[threadIdx.x + stride];
-            const int oi = si[threadIdx.x + stride];
-            if (ov > sv[threadIdx.x] ||
-                (ov == sv[threadIdx.x] && oi < si[threadIdx.x])) {
-                sv[threadIdx.x] = ov;
-                si[threadIdx.x] = oi;
-            }
-        }
+        if (threadIdx.x < stride)
+            sk[threadIdx.x] = max(sk[threadIdx.x], sk[threadIdx.x + stride]);
         __syncthreads();
     }
-    if (threadIdx.x == 0) out[t] = si[0];
+    if (threadIdx.x == 0) atomicMax(keys + t, sk[0]);
+}
+
+__global__ void k_argmax_finish(const unsigned long long* __restrict__ keys,
+                                int* __restrict__ out, int T) {
+    const int t = blockIdx.x * blockDim.x + threadIdx.x;
+    if (t < T) out[t] = (int)~(uint32_t)(keys[t] & 0xFFFFFFFFu);
 }
 
 // ======================================================================
@@ -655,14 +669,16 @@ __device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
 struct KLoop {
     int lane, i, ks;
     int kb0, kb1;  // this wave's K-block range
-    __device__ void init(int nb) {
+    __device__ void init(int nb) { init_range(0, nb); }
+    // split a [b0, b1) block range (grid-level split-K) across the waves
+    __device__ void init_range(int b0, int b1) {
         lane = threadIdx.x & (WAVE - 1);
         i = lane & 15;
         ks = lane >> 4;
         const int wid = threadIdx.x / WAVE;
-        const int per = (nb + 3) >> 2;
-        kb0 = wid * per;
-        kb1 = min(nb, kb0 + per);
+        const int per = (b1 - b0 + NWAVES - 1) / NWAVES;
+        kb0 = b0 + wid * per;
+        kb1 = min(b1, kb0 + per);
     }
 };
 
@@ -717,20 +733,29 @@ __device__ __forceinline__ void a_frag_q4_scaled(uint32_t q, uint32_t ab,
 
 // One wave's software-pipelined K loop over NM matrices sharing the B
 // panel (NM=2 for the FFN's w1/w3 — halves B traffic and doubles the MFMA
-// work per load batch). PF blocks' loads are issued together before any
-// compute so ~PF*NM*256B of weight stream stays in flight per wave.
+// work per load batch).
+//
+// The loop is a ROTATED two-deep pipeline: the loads of batch i+1 are
+// issued before any compute of batch i, so 2×PF×NM×256 B of weight stream
+// is in flight per wave while the MFMAs run (the decode GEMV recipe of
+// cdna_hip_programming.md §5: "load straight to VGPRs, deep unroll, late
+// vmcnt" — the previous load→compute→load serial structure exposed the
+// ~900-cycle HBM latency on every batch and measured 0.5-0.9 TB/s).
+// Guarded per-element loads are avoided (hipcc's per-element-branch
+// vmcnt(0) trap, §5.4): full PF batches roll through the pipeline, the
+// remainder runs unpipelined after it.
 // Accumulation is MFMA C-chained over two alternating accumulators per
 // matrix (covers the dependent-accumulator latency).
 // acc[n][jj] ends with rows (l>>4)*4 + jj, col l&15 of output n.
-template <int WT, bool NORM, int NM, int PF = 4>
+template <int WT, bool NORM, int NM, int PF = 2>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
     const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep, float scale,
-    float acc[NM][4]) {
+    float acc[NM][4], int b0, int b1) {
     KLoop kl;
+    kl.init_range(b0, b1);
     const int nb = ws[0]->cols >> 5;
-    kl.init(nb);
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
     f32x4 c0[NM], c1[NM];
 #pragma unroll
@@ -749,6 +774,40 @@ __device__ __forceinline__ void wave_tile_kloop(
                    ((size_t)tile_row * (ws[n]->cols >> 3)) * 128;
         abbase[n] = (const uint32_t*)ws[n]->scales + (size_t)tile_row * nb * 16;
     }
+
+    struct Batch {
+        uint32_t q[PF][NM], ab[PF][NM];
+        uint4 aw[PF][NM];
+        uint4 xb[PF], nbv[PF];
+    };
+    // Two NAMED buffers, never indexed by a runtime value: a runtime
+    // buf[i&1] select sends the whole array to scratch (measured 240-336
+    // B/lane ScratchSize and a 3x kernel slowdown).
+    Batch bufA, bufB;
+
+    auto load_batch = [&](Batch& bt, int g) {
+#pragma unroll
+        for (int u = 0; u < PF; ++u) {
+            const int kb = g + u;
+            const int kc = kb * 4 + kl.ks;
+            bt.xb[u] = *reinterpret_cast<const uint4*>(
+                xprep + ((size_t)kc * 16 + kl.i) * 8);
+            if (NORM)
+                bt.nbv[u] = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+#pragma unroll
+            for (int n = 0; n < NM; ++n) {
+                if (WT == W_F16) {
+                    bt.aw[u][n] = *reinterpret_cast<const uint4*>(
+                        tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
+                } else {
+                    bt.q[u][n] = __builtin_nontemporal_load(
+                        qbase[n] + (size_t)kb * 64 + kl.ks * 16 + kl.i);
+                    bt.ab[u][n] = __builtin_nontemporal_load(
+                        abbase[n] + (size_t)kb * 16 + kl.i);
+                }
+            }
+        }
+    };
 
     auto compute_one = [&](int parity, const uint32_t q[NM],
                            const uint32_t ab[NM], const uint4 aw[NM],
@@ -780,37 +839,27 @@ __device__ __forceinline__ void wave_tile_kloop(
         }
     };
 
-    int g = kl.kb0;
-    for (; g + PF <= kl.kb1; g += PF) {
-        uint32_t q[PF][NM], ab[PF][NM];
-        uint4 aw[PF][NM];
-        uint4 xb[PF], nbv[PF];
-#pragma unroll
-        for (int u = 0; u < PF; ++u) {
-            const int kb = g + u;
-            const int kc = kb * 4 + kl.ks;
-            xb[u] = *reinterpret_cast<const uint4*>(
-                xprep + ((size_t)kc * 16 + kl.i) * 8);
-            if (NORM)
-                nbv[u] = *reinterpret_cast<const uint4*>(normprep + kc * 8);
-#pragma unroll
-            for (int n = 0; n < NM; ++n) {
-                if (WT == W_F16) {
-                    aw[u][n] = *reinterpret_cast<const uint4*>(
-                        tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
-                } else {
-                    q[u][n] = __builtin_nontemporal_load(
-                        qbase[n] + (size_t)kb * 64 + kl.ks * 16 + kl.i);
-                    ab[u][n] = __builtin_nontemporal_load(
-                        abbase[n] + (size_t)kb * 16 + kl.i);
-                }
-            }
-        }
+    auto compute_batch = [&](Batch& bt) {
 #pragma unroll
         for (int u = 0; u < PF; ++u)
-            compute_one(u & 1, q[u], ab[u], aw[u], xb[u], nbv[u]);
+            compute_one(u & 1, bt.q[u], bt.ab[u], bt.aw[u], bt.xb[u],
+                        bt.nbv[u]);
+    };
+
+    const int nfull = (kl.kb1 - kl.kb0) / PF;
+    if (nfull > 0) {
+        load_batch(bufA, kl.kb0);
+        int it = 0;
+        while (true) {
+            if (it + 1 < nfull) load_batch(bufB, kl.kb0 + (it + 1) * PF);
+            compute_batch(bufA);
+            if (++it == nfull) break;
+            if (it + 1 < nfull) load_batch(bufA, kl.kb0 + (it + 1) * PF);
+            compute_batch(bufB);
+            if (++it == nfull) break;
+        }
     }
-    for (; g < kl.kb1; ++g) {
+    for (int g = kl.kb0 + nfull * PF; g < kl.kb1; ++g) {
         uint32_t q[NM], ab[NM];
         uint4 aw[NM];
         uint4 xb, nbv;
@@ -904,6 +953,11 @@ __global__ void k_prep_x(const float* __restrict__ x,
 }
 
 // ------------------------------------------------------------- k_gemm16
+// MODE GM_ATOMIC: grid-level split-K — gridDim.y blocks per row-tile each
+// cover nb/gridDim.y K-blocks and atomicAdd their partial tile into y
+// (which already holds the residual stream); a following k_prep_x pass
+// rebuilds the sumsq/xprep side-channels. Used when rows/16 alone cannot
+// fill 256 CUs (wo: E/16 = 200 blocks).
 template <int WT, int MODE>
 __global__ __launch_bounds__(BLOCK) void k_gemm16(
     WMat2 w, const unsigned short* __restrict__ bprep,
@@ -918,10 +972,27 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     if (NORM) scale = norm_scale(ss_in, j < T ? j : 0, w.cols, eps);
     float acc[1][4];
     const WMat2* ws[1] = {&w};
-    wave_tile_kloop<WT, NORM, 1>(ws, blockIdx.x, bprep, normprep, scale, acc);
+    const int nb = w.cols >> 5;
+    int b0 = 0, b1 = nb;
+    if (MODE == GM_ATOMIC) {
+        const int per = (nb + gridDim.y - 1) / gridDim.y;
+        b0 = blockIdx.y * per;
+        b1 = min(nb, b0 + per);
+    }
+    wave_tile_kloop<WT, NORM, 1>(ws, blockIdx.x, bprep, normprep, scale, acc,
+                                 b0, b1);
     __shared__ float lds[3 * 64 * 4];
     combine_acc<1>(acc, lds);
     if (threadIdx.x >= WAVE) return;
+    if (MODE == GM_ATOMIC) {
+        const int r0a = blockIdx.x * 16 + (lane >> 4) * 4;
+        if (j < T) {
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj)
+                atomicAdd(y + (size_t)j * w.rows + r0a + jj, acc[0][jj]);
+        }
+        return;
+    }
     // wave 0 epilogue: rows r = blockIdx.x*16 + (lane>>4)*4 + jj, col j.
     const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
     float sq = 0.0f;
@@ -974,7 +1045,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const float scale = norm_scale(ss_in, j < T ? j : 0, E, eps);
     float acc[1][4];
     const WMat2* ws[1] = {&w};
-    wave_tile_kloop<WT, true, 1>(ws, tile, xprep, normprep, scale, acc);
+    wave_tile_kloop<WT, true, 1>(ws, tile, xprep, normprep, scale, acc, 0,
+                                 E >> 5);
     __shared__ float lds[3 * 64 * 4];
     combine_acc<1>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;
@@ -1023,7 +1095,7 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     float acc[2][4];
     const WMat2* ws[2] = {&w1, &w3};
     wave_tile_kloop<WT, true, 2>(ws, blockIdx.x, xprep, normprep, scale,
-                                 acc);
+                                 acc, 0, w1.cols >> 5);
     __shared__ float lds[3 * 64 * 4 * 2];
     combine_acc<2>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;
@@ -1169,7 +1241,18 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
                    const unsigned short* normprep, const float* ss_in,
                    float eps, float* y, unsigned short* xprep_out,
                    float* ss_out, int T, int mode) {
-    const dim3 grid(w.rows / 16);
+    const int R = w.rows / 16;
+    if (mode == GM_ATOMIC) {
+        // pick the split so R*KS lands near 2-3 blocks/CU (256 CUs)
+        int ks = 1;
+        while (R * ks < 512 && ks < 8) ks <<= 1;
+        const dim3 grid(R, ks);
+        DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
+            (k_gemm16<WTc, GM_ATOMIC>), grid, dim3(BLOCK), 0, s, w, bprep,
+            normprep, ss_in, eps, y, xprep_out, ss_out, T));
+        return;
+    }
+    const dim3 grid(R);
     DISPATCH_WT2(w.wtype, {
         if (mode == GM_RES_SQ)
             hipLaunchKernelGGL((k_gemm16<WTc, GM_RES_SQ>), grid, dim3(BLOCK),
@@ -1255,7 +1338,12 @@ void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
     }
 }
 
-void launch_argmax(hipStream_t s, const float* logits, int* out, int T,
-                   int V) {
-    hipLaunchKernelGGL(k_argmax, dim3(T), dim3(BLOCK), 0, s, logits, out, V);
+void launch_argmax(hipStream_t s, const float* logits,
+                   unsigned long long* keys, int* out, int T, int V) {
+    hipMemsetAsync(keys, 0, sizeof(unsigned long long) * T, s);
+    const int npart = min(32, (V + BLOCK - 1) / BLOCK);
+    hipLaunchKernelGGL(k_argmax_part, dim3(T, npart), dim3(BLOCK), 0, s,
+                       logits, keys, V);
+    hipLaunchKernelGGL(k_argmax_finish, dim3(1), dim3(64), 0, s, keys, out,
+                       T);
 }
