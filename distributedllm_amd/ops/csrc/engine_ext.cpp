@@ -323,6 +323,7 @@ class SliceEngine {
             {T, V_},
             torch::TensorOptions().device(torch::kCUDA).dtype(torch::kFloat32));
         if (out_mfma_) {
+            hipMemsetAsync(ss_tmp_.data_ptr(), 0, sizeof(float) * T, s);
             launch_prep_x(s, xin.data_ptr<float>(), u16p(xprep_),
                           ss_tmp_.data_ptr<float>(), E_, T);
             launch_gemm16(s, mout_.w, u16p(xprep_), u16p(final_normprep_),

@@ -747,7 +747,7 @@ __device__ __forceinline__ void a_frag_q4_scaled(uint32_t q, uint32_t ab,
 // Accumulation is MFMA C-chained over two alternating accumulators per
 // matrix (covers the dependent-accumulator latency).
 // acc[n][jj] ends with rows (l>>4)*4 + jj, col l&15 of output n.
-template <int WT, bool NORM, int NM, int PF = 2>
+template <int WT, bool NORM, int NM, int PF = 4>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
     const unsigned short* __restrict__ xprep,
@@ -923,16 +923,22 @@ __device__ __forceinline__ float norm_scale(const float* ss, int j, int cols,
 
 // ------------------------------------------------------------- k_prep_x
 // sumsq + bf16 xprep of an incoming f32 activation block (forward entry,
-// logits entry). Grid: T blocks.
+// split-K gemm epilogue, logits entry). Grid: (T, C) — C column chunks per
+// row so the launch fills more than T CUs; ss[t] accumulates via atomics
+// and MUST be zeroed before the launch (a T-block single-chunk version
+// measured 4.6 µs — launch+latency bound on 16 CUs of 256).
 __global__ void k_prep_x(const float* __restrict__ x,
                          unsigned short* __restrict__ xprep,
                          float* __restrict__ ss, int cols) {
     const int t = blockIdx.x;
     const float* xt = x + (size_t)t * cols;
     __shared__ float red[NWAVES];
-    float sum = 0.0f;
     const int nkc = cols >> 3;
-    for (int kc = threadIdx.x; kc < nkc; kc += BLOCK) {
+    const int per = (nkc + gridDim.y - 1) / gridDim.y;
+    const int kc0 = blockIdx.y * per;
+    const int kc1 = min(nkc, kc0 + per);
+    float sum = 0.0f;
+    for (int kc = kc0 + threadIdx.x; kc < kc1; kc += BLOCK) {
         const float4 a = *reinterpret_cast<const float4*>(xt + kc * 8);
         const float4 b = *reinterpret_cast<const float4*>(xt + kc * 8 + 4);
         sum += a.x * a.x + a.y * a.y + a.z * a.z + a.w * a.w;
@@ -949,7 +955,7 @@ __global__ void k_prep_x(const float* __restrict__ x,
     if ((threadIdx.x & (WAVE - 1)) == 0) red[wid] = sum;
     __syncthreads();
     if (threadIdx.x == 0)
-        ss[t] = red[0] + red[1] + red[2] + red[3];
+        atomicAdd(ss + t, red[0] + red[1] + red[2] + red[3]);
 }
 
 // ------------------------------------------------------------- k_gemm16
@@ -1213,8 +1219,10 @@ void launch_attention(hipStream_t s, const float* q_buf,
 
 void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T) {
-    hipLaunchKernelGGL(k_prep_x, dim3(T), dim3(BLOCK), 0, s, x, xprep, ss,
-                       cols);
+    // chunk columns so the grid covers >T CUs; ss must be zeroed upstream
+    const int chunks = max(1, min(8, cols / 2048));
+    hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
+                       xprep, ss, cols);
 }
 
 #define DISPATCH_WT2(WTV, ...)                   \
