@@ -67,19 +67,21 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             qword |= n[:, :, :, j] << ((j % 2) * 16 + (j // 2) * 4)
         qs2 = np.ascontiguousarray(
             qword.reshape(R, 16, nb, 4).transpose(0, 2, 3, 1))
-        # per-(row, block) prescale pair: w = alpha*(n+128) + beta
-        # (q4_0: alpha=d, beta=-136d; q4_1: alpha=d, beta=m-128d)
+        # per-(row, block) (alpha, beta) f16 pair; the kernel computes
+        # w = alpha*((1024+n) + csub) + beta with csub = -1032 (q4_0,
+        # giving alpha*(n-8)) or -1024 (q4_1, giving alpha*n + beta) in
+        # exact packed-f16 arithmetic (kernels.hip a_frag_q4)
         bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
         a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
         if t.gtype == ggml.GGML_TYPE_Q4_0:
             d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
             d = d.reshape(rows, nb).astype(np.float32)
-            alpha, beta = d, -136.0 * d
+            alpha, beta = d, np.zeros_like(d)
         else:
             dm = np.ascontiguousarray(a[:, :, :4]).view(np.float16)
             dm = dm.reshape(rows, nb, 2).astype(np.float32)
             alpha = dm[:, :, 0]
-            beta = dm[:, :, 1] - 128.0 * dm[:, :, 0]
+            beta = dm[:, :, 1]
         ab = np.stack([alpha, beta], axis=-1).astype(np.float16)
         scales = np.ascontiguousarray(
             ab.reshape(R, 16, nb, 2).transpose(0, 2, 1, 3))
@@ -88,9 +90,8 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
         return data, sc, t.gtype
     if t.gtype == ggml.GGML_TYPE_F16:
         w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
-        bf = _f32_to_bf16_u16(w.astype(np.float32))
         tile = np.ascontiguousarray(
-            bf.reshape(R, 16, cols // 8, 8).transpose(0, 2, 1, 3))
+            w.reshape(R, 16, cols // 8, 8).transpose(0, 2, 1, 3))
         data = torch.from_numpy(tile.view(np.int16)).to(device)
         return data, torch.empty(0), ggml.GGML_TYPE_F16
     # f32: legacy scalar path, plain [rows, cols]
@@ -185,12 +186,14 @@ class HIPSliceEngine:
                                      generator=g)
                 alpha = ((torch.rand(R * nb * 16, 1, device="cuda",
                                      generator=g) * 0.5 + 0.75) * 0.003)
-                ab = torch.cat([alpha, -136.0 * alpha], dim=1)
+                beta = (torch.zeros_like(alpha) if wt == ggml.GGML_TYPE_Q4_0
+                        else alpha * 0.1)
+                ab = torch.cat([alpha, beta], dim=1)
                 return data, ab.to(torch.float16).contiguous(), wt
             if wt == ggml.GGML_TYPE_F16:
                 data = (torch.randn(rows * cols, device="cuda", generator=g,
                                     dtype=torch.float32) * 0.02)
-                return (data.to(torch.bfloat16).view(torch.int16),
+                return (data.to(torch.float16).view(torch.int16),
                         torch.empty(0), wt)
             data = (torch.randn(rows, cols, device="cuda", generator=g,
                                 dtype=torch.float32) * 0.02)

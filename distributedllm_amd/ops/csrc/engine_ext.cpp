@@ -87,8 +87,8 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
     } else {
         TORCH_CHECK(wtype == W_F16, "unsupported tiled wtype");
         TORCH_CHECK(data.scalar_type() == torch::kInt16 ||
-                        data.scalar_type() == torch::kBFloat16,
-                    "f16 weights must arrive as bf16 tiles");
+                        data.scalar_type() == torch::kHalf,
+                    "f16 weights must arrive as f16 tiles");
         TORCH_CHECK(data.numel() == rows * cols, "bf16 tile size mismatch");
         m.w.scales = nullptr;
     }
@@ -166,8 +166,8 @@ class SliceEngine {
         Layer& l = layers_[li];
         l.attn_norm = check_f32(attn_norm, "attn_norm");
         l.ffn_norm = check_f32(ffn_norm, "ffn_norm");
-        l.attn_normprep = attn_norm.to(torch::kBFloat16).contiguous();
-        l.ffn_normprep = ffn_norm.to(torch::kBFloat16).contiguous();
+        l.attn_normprep = attn_norm.to(torch::kHalf).contiguous();
+        l.ffn_normprep = ffn_norm.to(torch::kHalf).contiguous();
         const int64_t rows[7] = {E_, E_, E_, E_, F_, E_, F_};
         const int64_t cols[7] = {E_, E_, E_, E_, E_, F_, E_};
         auto first = mats[0].cast<py::tuple>();
@@ -198,7 +198,7 @@ class SliceEngine {
                    int64_t out_wtype, int64_t n_vocab) {
         V_ = (int)n_vocab;
         final_norm_ = check_f32(norm_w, "norm_w");
-        final_normprep_ = norm_w.to(torch::kBFloat16).contiguous();
+        final_normprep_ = norm_w.to(torch::kHalf).contiguous();
         // the embedding table always arrives in the legacy SoA layout
         // (gather kernel), the lm_head in the layout its path needs
         tok_ = DevMat{};

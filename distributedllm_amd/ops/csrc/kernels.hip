@@ -454,10 +454,10 @@ __global__ void k_attention(
         const int e = h * D + threadIdx.x;
         out[(size_t)t * E + e] = v;
         if (out_prep != nullptr) {
-            // bf16 B-layout side-channel for the wo MFMA consumer:
+            // f16 B-layout side-channel for the wo MFMA consumer:
             // element (k=e, j=t) at [(e>>3)*16 + t]*8 + (e&7)
-            union { __hip_bfloat16 b; unsigned short u; } c;
-            c.b = __float2bfloat16(v);
+            union { __half h; unsigned short u; } c;
+            c.h = __float2half(v);
             out_prep[((size_t)(e >> 3) * 16 + t) * 8 + (e & 7)] = c.u;
         }
     }
@@ -641,12 +641,12 @@ __global__ void k_argmax_finish(const unsigned long long* __restrict__ keys,
 // tile; partial accumulators combine through LDS; wave 0 runs the fused
 // epilogue (residual add + sumsq atomics + xprep/rope/cache/silu writes).
 
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
 
 union ABFrag {
     uint32_t u[4];
-    bf16x8 v;
+    f16x8 v;
 };
 
 __device__ __forceinline__ float bflo(uint32_t w) {
@@ -662,6 +662,23 @@ __device__ __forceinline__ float bfhi(uint32_t w) {
 __device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
     union { __hip_bfloat162 b; uint32_t u; } c;
     c.b = __hip_bfloat162(__float2bfloat16(lo), __float2bfloat16(hi));
+    return c.u;
+}
+
+// packed-f16 helpers for the MFMA side-channels / dequant
+__device__ __forceinline__ uint32_t pack_f16(float lo, float hi) {
+    union { __half2 h; uint32_t u; } c;
+    c.h = __floats2half2_rn(lo, hi);
+    return c.u;
+}
+__device__ __forceinline__ __half2 u2h2(uint32_t w) {
+    union { uint32_t u; __half2 h; } c;
+    c.u = w;
+    return c.h;
+}
+__device__ __forceinline__ uint32_t h22u(__half2 h) {
+    union { __half2 h; uint32_t u; } c;
+    c.h = h;
     return c.u;
 }
 
@@ -682,52 +699,33 @@ struct KLoop {
     }
 };
 
-// Load + (for the norm path) RMS-normalize the B fragment for block kb.
-template <bool NORM>
-__device__ __forceinline__ void b_frag(
-    const unsigned short* __restrict__ xprep,
-    const unsigned short* __restrict__ normprep, float scale, int kb,
-    int i /*col j = i*/, int ks, ABFrag& b) {
-    const int kc = kb * 4 + ks;
-    const uint4 xb = *reinterpret_cast<const uint4*>(xprep + (kc * 16 + i) * 8);
-    if (NORM) {
-        const uint4 nb_ = *reinterpret_cast<const uint4*>(normprep + kc * 8);
-        const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
-        const uint32_t nw[4] = {nb_.x, nb_.y, nb_.z, nb_.w};
-#pragma unroll
-        for (int w = 0; w < 4; ++w) {
-            const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
-            const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
-            b.u[w] = pack_bf16(f0, f1);
-        }
-    } else {
-        b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
-    }
-}
-
-// Dequantize one repacked nibble word into a prescaled bf16 A fragment:
-// w = alpha * (n+128) + beta, with the per-(row, block) (alpha, beta) f16
-// pair precomputed at repack time (q4_0: alpha=d, beta=-136d; q4_1:
-// alpha=d, beta=m-128d). Prescaling on the A side keeps the MFMA
-// C-chainable (AGPR accumulate, no per-block VALU fixup, no cross-lane
-// bias reduction) and unifies q4_0/q4_1 into one kernel path.
-__device__ __forceinline__ void a_frag_q4_scaled(uint32_t q, uint32_t ab,
-                                                 ABFrag& a) {
-    // ab = (beta_f16 << 16) | alpha_f16
-    union { __half2 h; uint32_t u; } c;
-    c.u = ab;
-    const float al = __half2float(__low2half(c.h));
-    const float be = __half2float(__high2half(c.h));
+// Dequantize one repacked nibble word into an f16 A fragment with pure
+// packed-f16 VALU (no unpack-to-f32 shift/and per half — the bf16 variant
+// of this dequant measured VALU-issue-bound):
+//   0x6400 | n is EXACTLY the f16 value 1024+n, so
+//   q4_0: w = alpha * ((1024+n) - 1032)        (n-8 exact in f16)
+//   q4_1: w = alpha * ((1024+n) - 1024) + beta (beta = m)
+// per word: and/shr/or unpack + one v_pk_add_f16 + one v_pk_fma_f16; the
+// result is the correctly-rounded f16 of the true q4 value (single
+// rounding). ab = (beta_f16 << 16) | alpha_f16 per (row, block).
+template <int WT>
+__device__ __forceinline__ void a_frag_q4(uint32_t q, uint32_t ab,
+                                          ABFrag& a) {
+    const __half2 abh = u2h2(ab);
+    const __half2 alpha2 = __half2half2(__low2half(abh));
+    const __half2 beta2 = __half2half2(__high2half(abh));
+    // f16 bit patterns: -1032 = 0xE408, -1024 = 0xE400
+    const uint32_t csub = (WT == W_Q4_0) ? 0xE408E408u : 0xE400E400u;
+    const __half2 c2 = u2h2(csub);
     uint32_t w[4];
-    w[0] = 0x43004300u | (q & 0x000F000Fu);
-    w[1] = 0x43004300u | ((q >> 4) & 0x000F000Fu);
-    w[2] = 0x43004300u | ((q >> 8) & 0x000F000Fu);
-    w[3] = 0x43004300u | ((q >> 12) & 0x000F000Fu);
+    w[0] = 0x64006400u | (q & 0x000F000Fu);
+    w[1] = 0x64006400u | ((q >> 4) & 0x000F000Fu);
+    w[2] = 0x64006400u | ((q >> 8) & 0x000F000Fu);
+    w[3] = 0x64006400u | ((q >> 12) & 0x000F000Fu);
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
-        const float f0 = fmaf(al, bflo(w[i]), be);
-        const float f1 = fmaf(al, bfhi(w[i]), be);
-        a.u[i] = pack_bf16(f0, f1);
+        const __half2 v = __hadd2(u2h2(w[i]), c2);
+        a.u[i] = h22u(__hfma2(v, alpha2, beta2));
     }
 }
 
@@ -764,16 +762,28 @@ __device__ __forceinline__ void wave_tile_kloop(
         c1[n] = zero;
     }
 
-    const uint32_t* qbase[NM];
-    const unsigned short* tbase[NM];
-    const uint32_t* abbase[NM];
+    // Running per-wave pointers, advanced by a constant stride per batch so
+    // every load in the unrolled batch has a CONSTANT immediate offset —
+    // per-load 64-bit address recomputation (v_lshl_add_u64 per load)
+    // dominated the issue stream otherwise (kernels measured VALU-issue-
+    // bound: 45% active / 31% issue-stall in SQ counters).
+    const uint32_t* qp[NM];
+    const uint32_t* abp[NM];
+    const unsigned short* tp[NM];
 #pragma unroll
     for (int n = 0; n < NM; ++n) {
-        qbase[n] = (const uint32_t*)ws[n]->data + ((size_t)tile_row * nb) * 64;
-        tbase[n] = (const unsigned short*)ws[n]->data +
-                   ((size_t)tile_row * (ws[n]->cols >> 3)) * 128;
-        abbase[n] = (const uint32_t*)ws[n]->scales + (size_t)tile_row * nb * 16;
+        qp[n] = (const uint32_t*)ws[n]->data +
+                ((size_t)tile_row * nb + kl.kb0) * 64 + kl.ks * 16 + kl.i;
+        abp[n] = (const uint32_t*)ws[n]->scales +
+                 ((size_t)tile_row * nb + kl.kb0) * 16 + kl.i;
+        tp[n] = (const unsigned short*)ws[n]->data +
+                ((size_t)tile_row * (ws[n]->cols >> 3)) * 128 +
+                ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
     }
+    const unsigned short* xp =
+        xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
+    const unsigned short* np =
+        normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
     struct Batch {
         uint32_t q[PF][NM], ab[PF][NM];
@@ -785,29 +795,39 @@ __device__ __forceinline__ void wave_tile_kloop(
     // B/lane ScratchSize and a 3x kernel slowdown).
     Batch bufA, bufB;
 
-    auto load_batch = [&](Batch& bt, int g) {
+    // loads one PF-batch at constant offsets, then advances the pointers
+    auto load_batch = [&](Batch& bt) {
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
-            const int kb = g + u;
-            const int kc = kb * 4 + kl.ks;
-            bt.xb[u] = *reinterpret_cast<const uint4*>(
-                xprep + ((size_t)kc * 16 + kl.i) * 8);
+            bt.xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
             if (NORM)
-                bt.nbv[u] = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+                bt.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
 #pragma unroll
             for (int n = 0; n < NM; ++n) {
                 if (WT == W_F16) {
-                    bt.aw[u][n] = *reinterpret_cast<const uint4*>(
-                        tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
+                    bt.aw[u][n] =
+                        *reinterpret_cast<const uint4*>(tp[n] + u * 512);
                 } else {
-                    bt.q[u][n] = __builtin_nontemporal_load(
-                        qbase[n] + (size_t)kb * 64 + kl.ks * 16 + kl.i);
-                    bt.ab[u][n] = __builtin_nontemporal_load(
-                        abbase[n] + (size_t)kb * 16 + kl.i);
+                    bt.q[u][n] = __builtin_nontemporal_load(qp[n] + u * 64);
+                    bt.ab[u][n] = __builtin_nontemporal_load(abp[n] + u * 16);
                 }
             }
         }
+        xp += PF * 512;
+        if (NORM) np += PF * 32;
+#pragma unroll
+        for (int n = 0; n < NM; ++n) {
+            if (WT == W_F16) {
+                tp[n] += PF * 512;
+            } else {
+                qp[n] += PF * 64;
+                abp[n] += PF * 16;
+            }
+        }
     };
+
+    // column-j norm scale as a packed f16 broadcast, hoisted out of the loop
+    const __half2 scale2 = __float2half2_rn(scale);
 
     auto compute_one = [&](int parity, const uint32_t q[NM],
                            const uint32_t ab[NM], const uint4 aw[NM],
@@ -818,9 +838,8 @@ __device__ __forceinline__ void wave_tile_kloop(
             const uint32_t nw[4] = {nbv.x, nbv.y, nbv.z, nbv.w};
 #pragma unroll
             for (int w = 0; w < 4; ++w) {
-                const float f0 = bflo(xw[w]) * bflo(nw[w]) * scale;
-                const float f1 = bfhi(xw[w]) * bfhi(nw[w]) * scale;
-                b.u[w] = pack_bf16(f0, f1);
+                const __half2 v = __hmul2(u2h2(xw[w]), u2h2(nw[w]));
+                b.u[w] = h22u(__hmul2(v, scale2));
             }
         } else {
             b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
@@ -832,10 +851,10 @@ __device__ __forceinline__ void wave_tile_kloop(
                 a.u[0] = aw[n].x; a.u[1] = aw[n].y;
                 a.u[2] = aw[n].z; a.u[3] = aw[n].w;
             } else {
-                a_frag_q4_scaled(q[n], ab[n], a);
+                a_frag_q4<WT>(q[n], ab[n], a);
             }
             f32x4& c = parity ? c1[n] : c0[n];
-            c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c, 0, 0, 0);
+            c = __builtin_amdgcn_mfma_f32_16x16x32_f16(a.v, b.v, c, 0, 0, 0);
         }
     };
 
@@ -848,13 +867,13 @@ __device__ __forceinline__ void wave_tile_kloop(
 
     const int nfull = (kl.kb1 - kl.kb0) / PF;
     if (nfull > 0) {
-        load_batch(bufA, kl.kb0);
+        load_batch(bufA);
         int it = 0;
         while (true) {
-            if (it + 1 < nfull) load_batch(bufB, kl.kb0 + (it + 1) * PF);
+            if (it + 1 < nfull) load_batch(bufB);
             compute_batch(bufA);
             if (++it == nfull) break;
-            if (it + 1 < nfull) load_batch(bufA, kl.kb0 + (it + 1) * PF);
+            if (it + 1 < nfull) load_batch(bufA);
             compute_batch(bufB);
             if (++it == nfull) break;
         }
@@ -863,21 +882,22 @@ __device__ __forceinline__ void wave_tile_kloop(
         uint32_t q[NM], ab[NM];
         uint4 aw[NM];
         uint4 xb, nbv;
-        const int kc = g * 4 + kl.ks;
-        xb = *reinterpret_cast<const uint4*>(
-            xprep + ((size_t)kc * 16 + kl.i) * 8);
-        if (NORM)
-            nbv = *reinterpret_cast<const uint4*>(normprep + kc * 8);
+        xb = *reinterpret_cast<const uint4*>(xp);
+        xp += 512;
+        if (NORM) {
+            nbv = *reinterpret_cast<const uint4*>(np);
+            np += 32;
+        }
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
             if (WT == W_F16) {
-                aw[n] = *reinterpret_cast<const uint4*>(
-                    tbase[n] + ((size_t)kc * 16 + kl.i) * 8);
+                aw[n] = *reinterpret_cast<const uint4*>(tp[n]);
+                tp[n] += 512;
             } else {
-                q[n] = __builtin_nontemporal_load(
-                    qbase[n] + (size_t)g * 64 + kl.ks * 16 + kl.i);
-                ab[n] = __builtin_nontemporal_load(
-                    abbase[n] + (size_t)g * 16 + kl.i);
+                q[n] = __builtin_nontemporal_load(qp[n]);
+                qp[n] += 64;
+                ab[n] = __builtin_nontemporal_load(abp[n]);
+                abp[n] += 16;
             }
         }
         compute_one(g & 1, q, ab, aw, xb, nbv);
@@ -944,10 +964,10 @@ __global__ void k_prep_x(const float* __restrict__ x,
         sum += a.x * a.x + a.y * a.y + a.z * a.z + a.w * a.w;
         sum += b.x * b.x + b.y * b.y + b.z * b.z + b.w * b.w;
         uint4 o;
-        o.x = pack_bf16(a.x, a.y);
-        o.y = pack_bf16(a.z, a.w);
-        o.z = pack_bf16(b.x, b.y);
-        o.w = pack_bf16(b.z, b.w);
+        o.x = pack_f16(a.x, a.y);
+        o.y = pack_f16(a.z, a.w);
+        o.z = pack_f16(b.x, b.y);
+        o.w = pack_f16(b.z, b.w);
         *reinterpret_cast<uint4*>(xprep + ((size_t)kc * 16 + t) * 8) = o;
     }
     sum = wave_reduce_sum(sum);
@@ -1020,8 +1040,8 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     if (MODE == GM_RES_SQ && xprep_out != nullptr && j < T) {
         // 4 consecutive rows -> one aligned 8 B bf16x4 chunk of xprep
         uint2 o;
-        o.x = pack_bf16(acc[0][0], acc[0][1]);
-        o.y = pack_bf16(acc[0][2], acc[0][3]);
+        o.x = pack_f16(acc[0][0], acc[0][1]);
+        o.y = pack_f16(acc[0][2], acc[0][3]);
         *reinterpret_cast<uint2*>(
             xprep_out + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
     }
@@ -1114,8 +1134,8 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
         g[jj] = silu * acc[1][jj];
     }
     uint2 o;
-    o.x = pack_bf16(g[0], g[1]);
-    o.y = pack_bf16(g[2], g[3]);
+    o.x = pack_f16(g[0], g[1]);
+    o.y = pack_f16(g[2], g[3]);
     *reinterpret_cast<uint2*>(
         gprep + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
 }
