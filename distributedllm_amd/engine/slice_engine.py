@@ -61,30 +61,37 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
     R = rows // 16
     if t.gtype in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
         nb = cols // 32
+        nbp = (nb + 3) & ~3  # padded to the dwordx4 load group of 4 blocks
         n = _nibbles(t).astype(np.uint32).reshape(rows, nb, 4, 8)
-        qword = np.zeros((rows, nb, 4), dtype=np.uint32)
+        qword = np.zeros((rows, nbp, 4), dtype=np.uint32)
         for j in range(8):
-            qword |= n[:, :, :, j] << ((j % 2) * 16 + (j // 2) * 4)
+            qword[:, :nb] |= n[:, :, :, j] << ((j % 2) * 16 + (j // 2) * 4)
+        # wide-load layout: u32[R][nbp/4][4 ws][16 i][4 kb-in-group] — one
+        # dwordx4 per lane covers 4 consecutive K-blocks (1 KiB per wave)
         qs2 = np.ascontiguousarray(
-            qword.reshape(R, 16, nb, 4).transpose(0, 2, 3, 1))
+            qword.reshape(R, 16, nbp // 4, 4, 4)       # [R][i][g4][kb][ws]
+            .transpose(0, 2, 4, 1, 3))                 # [R][g4][ws][i][kb]
         # per-(row, block) (alpha, beta) f16 pair; the kernel computes
         # w = alpha*((1024+n) + csub) + beta with csub = -1032 (q4_0,
         # giving alpha*(n-8)) or -1024 (q4_1, giving alpha*n + beta) in
-        # exact packed-f16 arithmetic (kernels.hip a_frag_q4)
+        # exact packed-f16 arithmetic (kernels.hip a_frag_q4); pad blocks
+        # carry alpha=beta=0 and contribute exact zeros.
         bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
         a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
+        alpha = np.zeros((rows, nbp), dtype=np.float32)
+        beta = np.zeros((rows, nbp), dtype=np.float32)
         if t.gtype == ggml.GGML_TYPE_Q4_0:
             d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
-            d = d.reshape(rows, nb).astype(np.float32)
-            alpha, beta = d, np.zeros_like(d)
+            alpha[:, :nb] = d.reshape(rows, nb).astype(np.float32)
         else:
             dm = np.ascontiguousarray(a[:, :, :4]).view(np.float16)
             dm = dm.reshape(rows, nb, 2).astype(np.float32)
-            alpha = dm[:, :, 0]
-            beta = dm[:, :, 1]
+            alpha[:, :nb] = dm[:, :, 0]
+            beta[:, :nb] = dm[:, :, 1]
         ab = np.stack([alpha, beta], axis=-1).astype(np.float16)
         scales = np.ascontiguousarray(
-            ab.reshape(R, 16, nb, 2).transpose(0, 2, 1, 3))
+            ab.reshape(R, 16, nbp // 4, 4, 2)          # [R][i][g4][kb][2]
+            .transpose(0, 2, 1, 3, 4))                 # [R][g4][i][kb][2]
         data = torch.from_numpy(qs2.view(np.int32)).to(device)
         sc = torch.from_numpy(scales).to(device)
         return data, sc, t.gtype
@@ -181,15 +188,22 @@ class HIPSliceEngine:
             # byte the same compute/HBM traffic as a real checkpoint
             if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
                 R, nb = rows // 16, cols // 32
-                data = torch.randint(-2**31, 2**31 - 1, (R * nb * 64,),
+                nbp = (nb + 3) & ~3
+                # any random bits are valid nibble words (the dequant masks
+                # them into finite f16 values), so the data stream needs no
+                # pad handling; pad scale blocks are (0, 0) => exact zeros
+                data = torch.randint(-2**31, 2**31 - 1, (R * nbp * 64,),
                                      dtype=torch.int32, device="cuda",
                                      generator=g)
-                alpha = ((torch.rand(R * nb * 16, 1, device="cuda",
+                alpha = ((torch.rand(rows, nbp, device="cuda",
                                      generator=g) * 0.5 + 0.75) * 0.003)
+                alpha[:, nb:] = 0.0
                 beta = (torch.zeros_like(alpha) if wt == ggml.GGML_TYPE_Q4_0
                         else alpha * 0.1)
-                ab = torch.cat([alpha, beta], dim=1)
-                return data, ab.to(torch.float16).contiguous(), wt
+                ab = torch.stack([alpha, beta], dim=-1).to(torch.float16)
+                ab = (ab.reshape(R, 16, nbp // 4, 4, 2)
+                      .permute(0, 2, 1, 3, 4).contiguous())
+                return data, ab, wt
             if wt == ggml.GGML_TYPE_F16:
                 data = (torch.randn(rows * cols, device="cuda", generator=g,
                                     dtype=torch.float32) * 0.02)

@@ -73,15 +73,16 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
     m.w.cols = (int)cols;
     m.w.wtype = (int)wtype;
     m.w.data = data.data_ptr();
+    const int64_t nbp = (nb + 3) & ~3;  // K-blocks padded to load groups
     if (wtype == W_Q4_0 || wtype == W_Q4_1) {
         TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
-                        data.numel() == R * nb * 64,
-                    "q4 tiled data must be u32[R][nb][4][16]");
+                        data.numel() == R * nbp * 64,
+                    "q4 tiled data must be u32[R][nbp/4][4][16][4]");
         TORCH_CHECK(scales.defined() && scales.is_cuda() &&
                         scales.is_contiguous() &&
                         scales.scalar_type() == torch::kFloat16 &&
-                        scales.numel() == R * nb * 16 * 2,
-                    "q4 tiled scales must be f16 (alpha,beta)[R][nb][16]");
+                        scales.numel() == R * nbp * 16 * 2,
+                    "q4 tiled scales must be f16 (a,b)[R][nbp/4][16][4]");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {
@@ -111,6 +112,15 @@ torch::Tensor check_i32(const torch::Tensor& t, const char* name) {
 
 unsigned short* u16p(torch::Tensor& t) {
     return reinterpret_cast<unsigned short*>(t.data_ptr());
+}
+
+// f16 norm-weight side channel, zero-padded to the q4 load-group multiple
+torch::Tensor pad_normprep(const torch::Tensor& w) {
+    auto h = w.to(torch::kHalf).contiguous();
+    const int64_t n = h.numel();
+    const int64_t padded = ((n / 32 + 3) & ~(int64_t)3) * 32;
+    if (padded == n) return h;
+    return torch::constant_pad_nd(h, {0, padded - n}, 0).contiguous();
 }
 
 class SliceEngine {
@@ -150,9 +160,15 @@ class SliceEngine {
         ab_ = torch::empty({kMaxTokens, E_}, f32);
         ffb_ = torch::empty({kMaxTokens, F_}, f32);
         // MFMA-path side channels
-        xprep_ = torch::empty({(int64_t)E_ * kMaxTokens}, u16);
-        aprep_ = torch::empty({(int64_t)E_ * kMaxTokens}, u16);
-        gprep_ = torch::empty({(int64_t)F_ * kMaxTokens}, u16);
+        // side channels sized for the PADDED K-block count (q4 wide-load
+        // groups of 4 blocks) and zero-filled: the pad region is consumed
+        // by alpha=0 weight blocks but must hold finite f16 values
+        auto side = [](int cols) {
+            return (int64_t)(((cols / 32 + 3) & ~3)) * 512;
+        };
+        xprep_ = torch::zeros({side(E_)}, u16);
+        aprep_ = torch::zeros({side(E_)}, u16);
+        gprep_ = torch::zeros({side(F_)}, u16);
         ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * kMaxTokens}, f32);
         ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
@@ -166,8 +182,8 @@ class SliceEngine {
         Layer& l = layers_[li];
         l.attn_norm = check_f32(attn_norm, "attn_norm");
         l.ffn_norm = check_f32(ffn_norm, "ffn_norm");
-        l.attn_normprep = attn_norm.to(torch::kHalf).contiguous();
-        l.ffn_normprep = ffn_norm.to(torch::kHalf).contiguous();
+        l.attn_normprep = pad_normprep(attn_norm);
+        l.ffn_normprep = pad_normprep(ffn_norm);
         const int64_t rows[7] = {E_, E_, E_, E_, F_, E_, F_};
         const int64_t cols[7] = {E_, E_, E_, E_, E_, F_, E_};
         auto first = mats[0].cast<py::tuple>();
@@ -198,7 +214,7 @@ class SliceEngine {
                    int64_t out_wtype, int64_t n_vocab) {
         V_ = (int)n_vocab;
         final_norm_ = check_f32(norm_w, "norm_w");
-        final_normprep_ = norm_w.to(torch::kHalf).contiguous();
+        final_normprep_ = pad_normprep(norm_w);
         // the embedding table always arrives in the legacy SoA layout
         // (gather kernel), the lm_head in the layout its path needs
         tok_ = DevMat{};

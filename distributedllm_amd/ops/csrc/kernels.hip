@@ -643,6 +643,7 @@ __global__ void k_argmax_finish(const unsigned long long* __restrict__ keys,
 
 typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
 typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 union ABFrag {
     uint32_t u[4];
@@ -683,18 +684,24 @@ __device__ __forceinline__ uint32_t h22u(__half2 h) {
 }
 
 // Per-wave state for the K loop: lane (i = l&15 row, ks = l>>4 k-span).
+// For q4 weights every wave range is aligned to the 4-K-block load group
+// (the repacked layout packs 4 consecutive K-blocks per dwordx4); ranges
+// live in the PADDED block count (repack zero-pads nb to a multiple of 4
+// with alpha=beta=0 blocks that contribute exact zeros).
 struct KLoop {
     int lane, i, ks;
     int kb0, kb1;  // this wave's K-block range
-    __device__ void init(int nb) { init_range(0, nb); }
-    // split a [b0, b1) block range (grid-level split-K) across the waves
-    __device__ void init_range(int b0, int b1) {
+    // split a [b0, b1) block range (grid-level split-K) across the waves;
+    // align=4 for q4 group loads, 1 for f16
+    __device__ void init_range(int b0, int b1, int align) {
         lane = threadIdx.x & (WAVE - 1);
         i = lane & 15;
         ks = lane >> 4;
         const int wid = threadIdx.x / WAVE;
-        const int per = (b1 - b0 + NWAVES - 1) / NWAVES;
-        kb0 = b0 + wid * per;
+        const int am = align - 1;
+        int per = (b1 - b0 + NWAVES - 1) / NWAVES;
+        per = (per + am) & ~am;
+        kb0 = min(b1, b0 + wid * per);
         kb1 = min(b1, kb0 + per);
     }
 };
@@ -752,8 +759,9 @@ __device__ __forceinline__ void wave_tile_kloop(
     const unsigned short* __restrict__ normprep, float scale,
     float acc[NM][4], int b0, int b1) {
     KLoop kl;
-    kl.init_range(b0, b1);
-    const int nb = ws[0]->cols >> 5;
+    kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4);
+    const int nb0 = ws[0]->cols >> 5;
+    const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
     f32x4 c0[NM], c1[NM];
 #pragma unroll
@@ -767,15 +775,23 @@ __device__ __forceinline__ void wave_tile_kloop(
     // per-load 64-bit address recomputation (v_lshl_add_u64 per load)
     // dominated the issue stream otherwise (kernels measured VALU-issue-
     // bound: 45% active / 31% issue-stall in SQ counters).
+    //
+    // Weight-stream loads are dwordx4: the q4 data layout groups 4
+    // consecutive K-blocks per lane (u32[R][nb/4][4 ks][16 i][4 kb]) so one
+    // nontemporal uint4 moves 1 KiB/wave of nibbles; the (alpha, beta)
+    // stream likewise ([R][nb/4][16 i][4 kb]). Narrow dword loads (256
+    // B/wave-instruction) left the kernels TA-issue-bound at 1.3-2.6 TB/s.
+    static_assert(PF % 4 == 0, "PF must be a multiple of the 4-kb q group");
     const uint32_t* qp[NM];
     const uint32_t* abp[NM];
     const unsigned short* tp[NM];
 #pragma unroll
     for (int n = 0; n < NM; ++n) {
         qp[n] = (const uint32_t*)ws[n]->data +
-                ((size_t)tile_row * nb + kl.kb0) * 64 + kl.ks * 16 + kl.i;
+                ((size_t)tile_row * nb + kl.kb0) * 64 +
+                (kl.ks * 16 + kl.i) * 4;
         abp[n] = (const uint32_t*)ws[n]->scales +
-                 ((size_t)tile_row * nb + kl.kb0) * 16 + kl.i;
+                 ((size_t)tile_row * nb + kl.kb0) * 16 + kl.i * 4;
         tp[n] = (const unsigned short*)ws[n]->data +
                 ((size_t)tile_row * (ws[n]->cols >> 3)) * 128 +
                 ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
@@ -786,7 +802,7 @@ __device__ __forceinline__ void wave_tile_kloop(
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
     struct Batch {
-        uint32_t q[PF][NM], ab[PF][NM];
+        u32x4 q[PF / 4][NM], ab[PF / 4][NM];
         uint4 aw[PF][NM];
         uint4 xb[PF], nbv[PF];
     };
@@ -802,14 +818,21 @@ __device__ __forceinline__ void wave_tile_kloop(
             bt.xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
             if (NORM)
                 bt.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
+        }
+#pragma unroll
+        for (int u4 = 0; u4 < PF / 4; ++u4) {
 #pragma unroll
             for (int n = 0; n < NM; ++n) {
                 if (WT == W_F16) {
-                    bt.aw[u][n] =
-                        *reinterpret_cast<const uint4*>(tp[n] + u * 512);
+#pragma unroll
+                    for (int v = 0; v < 4; ++v)
+                        bt.aw[u4 * 4 + v][n] = *reinterpret_cast<const uint4*>(
+                            tp[n] + (u4 * 4 + v) * 512);
                 } else {
-                    bt.q[u][n] = __builtin_nontemporal_load(qp[n] + u * 64);
-                    bt.ab[u][n] = __builtin_nontemporal_load(abp[n] + u * 16);
+                    bt.q[u4][n] = __builtin_nontemporal_load(
+                        reinterpret_cast<const u32x4*>(qp[n]) + u4 * 64);
+                    bt.ab[u4][n] = __builtin_nontemporal_load(
+                        reinterpret_cast<const u32x4*>(abp[n]) + u4 * 16);
                 }
             }
         }
@@ -860,9 +883,15 @@ __device__ __forceinline__ void wave_tile_kloop(
 
     auto compute_batch = [&](Batch& bt) {
 #pragma unroll
-        for (int u = 0; u < PF; ++u)
-            compute_one(u & 1, bt.q[u], bt.ab[u], bt.aw[u], bt.xb[u],
-                        bt.nbv[u]);
+        for (int u = 0; u < PF; ++u) {
+            uint32_t q[NM], ab[NM];
+#pragma unroll
+            for (int n = 0; n < NM; ++n) {
+                q[n] = bt.q[u / 4][n][u % 4];
+                ab[n] = bt.ab[u / 4][n][u % 4];
+            }
+            compute_one(u & 1, q, ab, bt.aw[u], bt.xb[u], bt.nbv[u]);
+        }
     };
 
     const int nfull = (kl.kb1 - kl.kb0) / PF;
@@ -998,12 +1027,15 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     if (NORM) scale = norm_scale(ss_in, j < T ? j : 0, w.cols, eps);
     float acc[1][4];
     const WMat2* ws[1] = {&w};
-    const int nb = w.cols >> 5;
-    int b0 = 0, b1 = nb;
+    const int nbt = w.cols >> 5;
+    const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);  // padded
+    int b0 = 0, b1 = nbk;
     if (MODE == GM_ATOMIC) {
-        const int per = (nb + gridDim.y - 1) / gridDim.y;
-        b0 = blockIdx.y * per;
-        b1 = min(nb, b0 + per);
+        // split in 4-aligned units so q4 group loads stay whole
+        int per = (nbk + gridDim.y - 1) / gridDim.y;
+        per = (per + 3) & ~3;
+        b0 = min(nbk, (int)blockIdx.y * per);
+        b1 = min(nbk, b0 + per);
     }
     wave_tile_kloop<WT, NORM, 1>(ws, blockIdx.x, bprep, normprep, scale, acc,
                                  b0, b1);
@@ -1071,8 +1103,9 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const float scale = norm_scale(ss_in, j < T ? j : 0, E, eps);
     float acc[1][4];
     const WMat2* ws[1] = {&w};
+    const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
     wave_tile_kloop<WT, true, 1>(ws, tile, xprep, normprep, scale, acc, 0,
-                                 E >> 5);
+                                 nbe);
     __shared__ float lds[3 * 64 * 4];
     combine_acc<1>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;
@@ -1120,8 +1153,10 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     const float scale = norm_scale(ss_in, j < T ? j : 0, w1.cols, eps);
     float acc[2][4];
     const WMat2* ws[2] = {&w1, &w3};
+    const int nbf = (WT == W_F16) ? (w1.cols >> 5)
+                                  : (((w1.cols >> 5) + 3) & ~3);
     wave_tile_kloop<WT, true, 2>(ws, blockIdx.x, xprep, normprep, scale,
-                                 acc, 0, w1.cols >> 5);
+                                 acc, 0, nbf);
     __shared__ float lds[3 * 64 * 4 * 2];
     combine_acc<2>(acc, lds);
     if (threadIdx.x >= WAVE || j >= T) return;
@@ -1239,8 +1274,10 @@ void launch_attention(hipStream_t s, const float* q_buf,
 
 void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T) {
-    // chunk columns so the grid covers >T CUs; ss must be zeroed upstream
-    const int chunks = max(1, min(8, cols / 2048));
+    // chunk columns so the launch spreads over ~256 CUs even at T=16
+    // (a T-block launch measured 4.6 us on 16 CUs); ss zeroed upstream
+    const int want = 256 / max(T, 1);
+    const int chunks = max(1, min(want, cols / 1024));
     hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
                        xprep, ss, cols);
 }
