@@ -92,14 +92,21 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
         scales = np.ascontiguousarray(
             ab.reshape(R, 16, nbp // 4, 4, 2)          # [R][i][g4][kb][2]
             .transpose(0, 2, 1, 3, 4))                 # [R][g4][i][kb][2]
-        data = torch.from_numpy(qs2.view(np.int32)).to(device)
-        sc = torch.from_numpy(scales).to(device)
+        # one-prefetch-batch tail slack (kernels overread one batch)
+        data = torch.from_numpy(
+            np.concatenate([qs2.view(np.int32).reshape(-1),
+                            np.zeros(256, np.int32)])).to(device)
+        sc = torch.from_numpy(
+            np.concatenate([scales.reshape(-1),
+                            np.zeros(128, np.float16)])).to(device)
         return data, sc, t.gtype
     if t.gtype == ggml.GGML_TYPE_F16:
         w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
         tile = np.ascontiguousarray(
             w.reshape(R, 16, cols // 8, 8).transpose(0, 2, 1, 3))
-        data = torch.from_numpy(tile.view(np.int16)).to(device)
+        data = torch.from_numpy(
+            np.concatenate([tile.view(np.int16).reshape(-1),
+                            np.zeros(2048, np.int16)])).to(device)
         return data, torch.empty(0), ggml.GGML_TYPE_F16
     # f32: legacy scalar path, plain [rows, cols]
     a = np.frombuffer(t.raw, np.float32).reshape(rows, cols)
@@ -192,7 +199,8 @@ class HIPSliceEngine:
                 # any random bits are valid nibble words (the dequant masks
                 # them into finite f16 values), so the data stream needs no
                 # pad handling; pad scale blocks are (0, 0) => exact zeros
-                data = torch.randint(-2**31, 2**31 - 1, (R * nbp * 64,),
+                data = torch.randint(-2**31, 2**31 - 1,
+                                     (R * nbp * 64 + 256,),
                                      dtype=torch.int32, device="cuda",
                                      generator=g)
                 alpha = ((torch.rand(rows, nbp, device="cuda",
@@ -202,10 +210,13 @@ class HIPSliceEngine:
                         else alpha * 0.1)
                 ab = torch.stack([alpha, beta], dim=-1).to(torch.float16)
                 ab = (ab.reshape(R, 16, nbp // 4, 4, 2)
-                      .permute(0, 2, 1, 3, 4).contiguous())
+                      .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
+                ab = torch.cat([ab, torch.zeros(128, dtype=torch.float16,
+                                                device="cuda")])
                 return data, ab, wt
             if wt == ggml.GGML_TYPE_F16:
-                data = (torch.randn(rows * cols, device="cuda", generator=g,
+                data = (torch.randn(rows * cols + 2048, device="cuda",
+                                    generator=g,
                                     dtype=torch.float32) * 0.02)
                 return (data.to(torch.float16).view(torch.int16),
                         torch.empty(0), wt)

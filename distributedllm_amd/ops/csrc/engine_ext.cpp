@@ -29,6 +29,13 @@ namespace {
 
 constexpr int kMaxTokens = 16;  // per-forward token-tile cap (MFMA N)
 
+// Tail slack appended to every MFMA-path allocation: the pipelined K loop
+// prefetches ONE load batch past its range (kernels.hip wave_tile_kloop).
+constexpr int64_t kTailSlackQ = 256;     // int32 elements (1 KiB)
+constexpr int64_t kTailSlackAB = 128;    // f16 elements (256 B)
+constexpr int64_t kTailSlackF16 = 2048;  // f16 elements (4 KiB)
+constexpr int64_t kTailSlackSide = 2048; // int16 elements (4 KiB)
+
 struct DevMat {  // legacy scalar-path matrix (W_F32 models)
     torch::Tensor data, scales;
     WMat w{};
@@ -76,13 +83,14 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
     const int64_t nbp = (nb + 3) & ~3;  // K-blocks padded to load groups
     if (wtype == W_Q4_0 || wtype == W_Q4_1) {
         TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
-                        data.numel() == R * nbp * 64,
-                    "q4 tiled data must be u32[R][nbp/4][4][16][4]");
+                        data.numel() == R * nbp * 64 + kTailSlackQ,
+                    "q4 tiled data must be u32[R][nbp/4][4][16][4] + slack");
         TORCH_CHECK(scales.defined() && scales.is_cuda() &&
                         scales.is_contiguous() &&
                         scales.scalar_type() == torch::kFloat16 &&
-                        scales.numel() == R * nbp * 16 * 2,
-                    "q4 tiled scales must be f16 (a,b)[R][nbp/4][16][4]");
+                        scales.numel() == R * nbp * 16 * 2 + kTailSlackAB,
+                    "q4 tiled scales must be f16 (a,b)[R][nbp/4][16][4] "
+                    "+ slack");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {
@@ -90,7 +98,8 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
         TORCH_CHECK(data.scalar_type() == torch::kInt16 ||
                         data.scalar_type() == torch::kHalf,
                     "f16 weights must arrive as f16 tiles");
-        TORCH_CHECK(data.numel() == rows * cols, "bf16 tile size mismatch");
+        TORCH_CHECK(data.numel() == rows * cols + kTailSlackF16,
+                    "f16 tile size mismatch (tail slack missing)");
         m.w.scales = nullptr;
     }
     return m;
@@ -118,8 +127,8 @@ unsigned short* u16p(torch::Tensor& t) {
 torch::Tensor pad_normprep(const torch::Tensor& w) {
     auto h = w.to(torch::kHalf).contiguous();
     const int64_t n = h.numel();
-    const int64_t padded = ((n / 32 + 3) & ~(int64_t)3) * 32;
-    if (padded == n) return h;
+    // group-pad plus one prefetch batch of slack (4 groups x 32 shorts)
+    const int64_t padded = ((n / 32 + 3) & ~(int64_t)3) * 32 + 128;
     return torch::constant_pad_nd(h, {0, padded - n}, 0).contiguous();
 }
 
@@ -164,7 +173,7 @@ class SliceEngine {
         // groups of 4 blocks) and zero-filled: the pad region is consumed
         // by alpha=0 weight blocks but must hold finite f16 values
         auto side = [](int cols) {
-            return (int64_t)(((cols / 32 + 3) & ~3)) * 512;
+            return (int64_t)(((cols / 32 + 3) & ~3)) * 512 + kTailSlackSide;
         };
         xprep_ = torch::zeros({side(E_)}, u16);
         aprep_ = torch::zeros({side(E_)}, u16);

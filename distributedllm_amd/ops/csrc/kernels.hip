@@ -701,8 +701,12 @@ struct KLoop {
         const int am = align - 1;
         int per = (b1 - b0 + NWAVES - 1) / NWAVES;
         per = (per + am) & ~am;
-        kb0 = min(b1, b0 + wid * per);
-        kb1 = min(b1, kb0 + per);
+        // readfirstlane: wid is wave-uniform but derived from threadIdx,
+        // so without this the compiler treats every K loop bounded by
+        // kb0/kb1 as lane-divergent and wraps each iteration in exec-mask
+        // bookkeeping (saveexec/or/andn2 + implicit-def accvgpr moves)
+        kb0 = __builtin_amdgcn_readfirstlane(min(b1, b0 + wid * per));
+        kb1 = __builtin_amdgcn_readfirstlane(min(b1, kb0 + per));
     }
 };
 
@@ -801,24 +805,21 @@ __device__ __forceinline__ void wave_tile_kloop(
     const unsigned short* np =
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
+    // Only the HBM weight stream is double-buffered (the B-panel side
+    // channels are small and L2-resident — buffering them too doubled the
+    // live state past the VGPR budget and the allocator bounced the idle
+    // buffer through AGPRs, ~6 wasted issue slots per MFMA).
     struct Batch {
         u32x4 q[PF / 4][NM], ab[PF / 4][NM];
         uint4 aw[PF][NM];
-        uint4 xb[PF], nbv[PF];
     };
     // Two NAMED buffers, never indexed by a runtime value: a runtime
     // buf[i&1] select sends the whole array to scratch (measured 240-336
     // B/lane ScratchSize and a 3x kernel slowdown).
     Batch bufA, bufB;
 
-    // loads one PF-batch at constant offsets, then advances the pointers
+    // loads one PF-batch of WEIGHTS at constant offsets, then advances
     auto load_batch = [&](Batch& bt) {
-#pragma unroll
-        for (int u = 0; u < PF; ++u) {
-            bt.xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
-            if (NORM)
-                bt.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
-        }
 #pragma unroll
         for (int u4 = 0; u4 < PF / 4; ++u4) {
 #pragma unroll
@@ -836,8 +837,6 @@ __device__ __forceinline__ void wave_tile_kloop(
                 }
             }
         }
-        xp += PF * 512;
-        if (NORM) np += PF * 32;
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
             if (WT == W_F16) {
@@ -882,6 +881,17 @@ __device__ __forceinline__ void wave_tile_kloop(
     };
 
     auto compute_batch = [&](Batch& bt) {
+        // B-panel loads issue here, all PF up front (L2-resident, short
+        // latency; single-buffered on purpose — see Batch comment)
+        uint4 xb[PF], nbv[PF];
+#pragma unroll
+        for (int u = 0; u < PF; ++u) {
+            xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
+            if (NORM)
+                nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
+        }
+        xp += PF * 512;
+        if (NORM) np += PF * 32;
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
             uint32_t q[NM], ab[NM];
@@ -890,19 +900,26 @@ __device__ __forceinline__ void wave_tile_kloop(
                 q[n] = bt.q[u / 4][n][u % 4];
                 ab[n] = bt.ab[u / 4][n][u % 4];
             }
-            compute_one(u & 1, q, ab, bt.aw[u], bt.xb[u], bt.nbv[u]);
+            compute_one(u & 1, q, ab, bt.aw[u], xb[u], nbv[u]);
         }
     };
 
+    // Steady-state loop with UNCONDITIONAL prefetch: a branch around the
+    // load_batch would leave the outstanding-load count unknowable at the
+    // next compute and hipcc then drains vmcnt toward 0 inside the body
+    // (observed: vmcnt(7)..(0) chains; with this form it emits one counted
+    // vmcnt(12) per batch). The final iteration prefetches ONE batch past
+    // kb1 — every weight/side-channel allocation carries a tail slack of
+    // one batch for this (engine_ext.cpp kTailSlack*).
     const int nfull = (kl.kb1 - kl.kb0) / PF;
     if (nfull > 0) {
         load_batch(bufA);
         int it = 0;
         while (true) {
-            if (it + 1 < nfull) load_batch(bufB);
+            load_batch(bufB);
             compute_batch(bufA);
             if (++it == nfull) break;
-            if (it + 1 < nfull) load_batch(bufA);
+            load_batch(bufA);
             compute_batch(bufB);
             if (++it == nfull) break;
         }
