@@ -805,21 +805,29 @@ __device__ __forceinline__ void wave_tile_kloop(
     const unsigned short* np =
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
-    // Only the HBM weight stream is double-buffered (the B-panel side
-    // channels are small and L2-resident — buffering them too doubled the
-    // live state past the VGPR budget and the allocator bounced the idle
-    // buffer through AGPRs, ~6 wasted issue slots per MFMA).
     struct Batch {
         u32x4 q[PF / 4][NM], ab[PF / 4][NM];
         uint4 aw[PF][NM];
+        uint4 xb[PF], nbv[PF];
     };
     // Two NAMED buffers, never indexed by a runtime value: a runtime
     // buf[i&1] select sends the whole array to scratch (measured 240-336
     // B/lane ScratchSize and a 3x kernel slowdown).
     Batch bufA, bufB;
 
-    // loads one PF-batch of WEIGHTS at constant offsets, then advances
+    // Loads one PF-batch at constant offsets, then advances the pointers.
+    // ORDER MATTERS: the (L2-resident) B-panel loads go FIRST — vmcnt
+    // retires loads in issue order, so an xb load issued after the next
+    // batch's HBM weight loads cannot satisfy its wait until those ~900-
+    // cycle loads return; xb-after-weights serialized every compute on
+    // the prefetch (observed as vmcnt draining to 0 inside the body).
     auto load_batch = [&](Batch& bt) {
+#pragma unroll
+        for (int u = 0; u < PF; ++u) {
+            bt.xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
+            if (NORM)
+                bt.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
+        }
 #pragma unroll
         for (int u4 = 0; u4 < PF / 4; ++u4) {
 #pragma unroll
@@ -837,6 +845,8 @@ __device__ __forceinline__ void wave_tile_kloop(
                 }
             }
         }
+        xp += PF * 512;
+        if (NORM) np += PF * 32;
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
             if (WT == W_F16) {
@@ -881,17 +891,6 @@ __device__ __forceinline__ void wave_tile_kloop(
     };
 
     auto compute_batch = [&](Batch& bt) {
-        // B-panel loads issue here, all PF up front (L2-resident, short
-        // latency; single-buffered on purpose — see Batch comment)
-        uint4 xb[PF], nbv[PF];
-#pragma unroll
-        for (int u = 0; u < PF; ++u) {
-            xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
-            if (NORM)
-                nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
-        }
-        xp += PF * 512;
-        if (NORM) np += PF * 32;
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
             uint32_t q[NM], ab[NM];
@@ -900,7 +899,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                 q[n] = bt.q[u / 4][n][u % 4];
                 ab[n] = bt.ab[u / 4][n][u % 4];
             }
-            compute_one(u & 1, q, ab, bt.aw[u], xb[u], nbv[u]);
+            compute_one(u & 1, q, ab, bt.aw[u], bt.xb[u], bt.nbv[u]);
         }
     };
 
