@@ -40,10 +40,11 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="open_llama_3b")
     ap.add_argument("--ftype", default="q4_0", choices=list(FTYPES))
-    ap.add_argument("--mbs", type=int, default=16,
-                    help="sequences per micro-batch (engine token-tile is "
-                         "16; decode is HBM-bound so tokens/s scales ~"
-                         "linearly with batch)")
+    ap.add_argument("--mbs", type=int, default=64,
+                    help="sequences per micro-batch (engine processes up "
+                         "to 64 tokens = 4 MFMA column tiles per kernel; "
+                         "decode is HBM-bound so tokens/s scales ~linearly "
+                         "with batch)")
     ap.add_argument("--ctx", type=int, default=2048)
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()

@@ -27,14 +27,14 @@
 
 namespace {
 
-constexpr int kMaxTokens = 16;  // per-forward token-tile cap (MFMA N)
+constexpr int kMaxTokens = 64;  // per-forward token cap (4 MFMA col tiles)
 
 // Tail slack appended to every MFMA-path allocation: the pipelined K loop
 // prefetches ONE load batch past its range (kernels.hip wave_tile_kloop).
 constexpr int64_t kTailSlackQ = 256;     // int32 elements (1 KiB)
 constexpr int64_t kTailSlackAB = 128;    // f16 elements (256 B)
 constexpr int64_t kTailSlackF16 = 2048;  // f16 elements (4 KiB)
-constexpr int64_t kTailSlackSide = 2048; // int16 elements (4 KiB)
+constexpr int64_t kTailSlackSide = 8192; // int16 elements (16 KiB)
 
 struct DevMat {  // legacy scalar-path matrix (W_F32 models)
     torch::Tensor data, scales;
@@ -172,8 +172,11 @@ class SliceEngine {
         // side channels sized for the PADDED K-block count (q4 wide-load
         // groups of 4 blocks) and zero-filled: the pad region is consumed
         // by alpha=0 weight blocks but must hold finite f16 values
+        // per K-chunk the side channel holds up to 4 column tiles of 16
+        // tokens (JT_MAX) — layout [kc][jt][16][8] f16
         auto side = [](int cols) {
-            return (int64_t)(((cols / 32 + 3) & ~3)) * 512 + kTailSlackSide;
+            return (int64_t)(((cols / 32 + 3) & ~3)) * 512 * 4 +
+                   kTailSlackSide;
         };
         xprep_ = torch::zeros({side(E_)}, u16);
         aprep_ = torch::zeros({side(E_)}, u16);

@@ -367,7 +367,7 @@ __global__ void k_attention(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
     const __half* __restrict__ v_cache, float* __restrict__ out,
     unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
-    const int* __restrict__ seq, int E, int D, int n_ctx) {
+    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw) {
     const int t = blockIdx.x;
     const int h = blockIdx.y;
     const int J = pos[t] + 1;
@@ -455,10 +455,11 @@ __global__ void k_attention(
         out[(size_t)t * E + e] = v;
         if (out_prep != nullptr) {
             // f16 B-layout side-channel for the wo MFMA consumer:
-            // element (k=e, j=t) at [(e>>3)*16 + t]*8 + (e&7)
+            // element (k=e, token t) at [((e>>3)*JT + t/16)*16 + t%16]*8
             union { __half h; unsigned short u; } c;
             c.h = __float2half(v);
-            out_prep[((size_t)(e >> 3) * 16 + t) * 8 + (e & 7)] = c.u;
+            out_prep[(((size_t)(e >> 3) * jtw + (t >> 4)) * 16 + (t & 15))
+                         * 8 + (e & 7)] = c.u;
         }
     }
 }
@@ -741,51 +742,56 @@ __device__ __forceinline__ void a_frag_q4(uint32_t q, uint32_t ab,
 }
 
 // One wave's software-pipelined K loop over NM matrices sharing the B
-// panel (NM=2 for the FFN's w1/w3 — halves B traffic and doubles the MFMA
-// work per load batch).
+// panel (NM=2 for the FFN's w1/w3) and JT 16-token column tiles sharing
+// the A (weight) stream. JT is THE batched-decode lever: the HBM weight
+// stream and the dequant VALU are paid once for JT*16 tokens, so decode
+// tokens/s scales nearly linearly in JT while the kernel wall barely
+// moves (memory-bound).
 //
-// The loop is a ROTATED two-deep pipeline: the loads of batch i+1 are
-// issued before any compute of batch i, so 2×PF×NM×256 B of weight stream
-// is in flight per wave while the MFMAs run (the decode GEMV recipe of
-// cdna_hip_programming.md §5: "load straight to VGPRs, deep unroll, late
-// vmcnt" — the previous load→compute→load serial structure exposed the
-// ~900-cycle HBM latency on every batch and measured 0.5-0.9 TB/s).
-// Guarded per-element loads are avoided (hipcc's per-element-branch
-// vmcnt(0) trap, §5.4): full PF batches roll through the pipeline, the
-// remainder runs unpipelined after it.
-// Accumulation is MFMA C-chained over two alternating accumulators per
-// matrix (covers the dependent-accumulator latency).
-// acc[n][jj] ends with rows (l>>4)*4 + jj, col l&15 of output n.
-template <int WT, bool NORM, int NM, int PF = 4>
+// Pipeline shape (all learned from SQ counters / emitted waits):
+//  * rotated two-deep UNCONDITIONAL prefetch of the weight batch — a
+//    branch around loads makes hipcc drain vmcnt inside the body; the
+//    last iteration overreads ONE batch (tail slack in every allocation),
+//  * the (L2-resident) B-panel loads of batch g issue BEFORE the HBM
+//    weight loads of batch g+1: vmcnt retires in issue order, so B loads
+//    issued after the prefetch could not complete before it,
+//  * per-batch pointers advance by constant strides (no per-load 64-bit
+//    address math),
+//  * wave-uniform loop bounds via readfirstlane (no exec-mask loop),
+//  * accumulation is MFMA C-chained over two alternating accumulator
+//    sets (covers dependent-accumulator latency).
+// acc[n][jt][jj] ends with rows (l>>4)*4+jj, col jt*16 + (l&15).
+template <int WT, bool NORM, int NM, int JT, int PF = 4>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
     const unsigned short* __restrict__ xprep,
-    const unsigned short* __restrict__ normprep, float scale,
-    float acc[NM][4], int b0, int b1) {
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps,
+    float acc[NM][JT][4], int b0, int b1) {
     KLoop kl;
     kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4);
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-    f32x4 c0[NM], c1[NM];
+    f32x4 c0[NM][JT], c1[NM][JT];
 #pragma unroll
-    for (int n = 0; n < NM; ++n) {
-        c0[n] = zero;
-        c1[n] = zero;
+    for (int n = 0; n < NM; ++n)
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt) {
+            c0[n][jt] = zero;
+            c1[n][jt] = zero;
+        }
+
+    // per-column-tile RMSNorm scale, hoisted (column j = lane&15 of tile jt)
+    __half2 scale2[JT];
+    if (NORM) {
+        const float inv_cols = 1.0f / (float)ws[0]->cols;
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt)
+            scale2[jt] = __float2half2_rn(
+                rsqrtf(ss_in[jt * 16 + kl.i] * inv_cols + eps));
     }
 
-    // Running per-wave pointers, advanced by a constant stride per batch so
-    // every load in the unrolled batch has a CONSTANT immediate offset —
-    // per-load 64-bit address recomputation (v_lshl_add_u64 per load)
-    // dominated the issue stream otherwise (kernels measured VALU-issue-
-    // bound: 45% active / 31% issue-stall in SQ counters).
-    //
-    // Weight-stream loads are dwordx4: the q4 data layout groups 4
-    // consecutive K-blocks per lane (u32[R][nb/4][4 ks][16 i][4 kb]) so one
-    // nontemporal uint4 moves 1 KiB/wave of nibbles; the (alpha, beta)
-    // stream likewise ([R][nb/4][16 i][4 kb]). Narrow dword loads (256
-    // B/wave-instruction) left the kernels TA-issue-bound at 1.3-2.6 TB/s.
-    static_assert(PF % 4 == 0, "PF must be a multiple of the 4-kb q group");
     const uint32_t* qp[NM];
     const uint32_t* abp[NM];
     const unsigned short* tp[NM];
@@ -800,34 +806,25 @@ __device__ __forceinline__ void wave_tile_kloop(
                 ((size_t)tile_row * (ws[n]->cols >> 3)) * 128 +
                 ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
     }
+    // xprep layout with JT: element (kc, jt, j, e) at ((kc*JT+jt)*16+j)*8+e
     const unsigned short* xp =
-        xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
+        xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * JT) * 128 + kl.i * 8;
     const unsigned short* np =
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
-    struct Batch {
+    struct Batch {  // weight stream only (HBM, nt, double-buffered)
         u32x4 q[PF / 4][NM], ab[PF / 4][NM];
         uint4 aw[PF][NM];
-        uint4 xb[PF], nbv[PF];
     };
-    // Two NAMED buffers, never indexed by a runtime value: a runtime
-    // buf[i&1] select sends the whole array to scratch (measured 240-336
-    // B/lane ScratchSize and a 3x kernel slowdown).
+    // NAMED buffers, never indexed by a runtime value (a runtime select
+    // sends the array to scratch: measured 240-336 B/lane and 3x slower).
     Batch bufA, bufB;
+    struct XPanel {  // B-panel side channel (L2-resident, per iteration)
+        uint4 xb[PF][JT];
+        uint4 nbv[PF];
+    };
 
-    // Loads one PF-batch at constant offsets, then advances the pointers.
-    // ORDER MATTERS: the (L2-resident) B-panel loads go FIRST — vmcnt
-    // retires loads in issue order, so an xb load issued after the next
-    // batch's HBM weight loads cannot satisfy its wait until those ~900-
-    // cycle loads return; xb-after-weights serialized every compute on
-    // the prefetch (observed as vmcnt draining to 0 inside the body).
-    auto load_batch = [&](Batch& bt) {
-#pragma unroll
-        for (int u = 0; u < PF; ++u) {
-            bt.xb[u] = *reinterpret_cast<const uint4*>(xp + u * 512);
-            if (NORM)
-                bt.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
-        }
+    auto load_w = [&](Batch& bt) {
 #pragma unroll
         for (int u4 = 0; u4 < PF / 4; ++u4) {
 #pragma unroll
@@ -845,8 +842,6 @@ __device__ __forceinline__ void wave_tile_kloop(
                 }
             }
         }
-        xp += PF * 512;
-        if (NORM) np += PF * 32;
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
             if (WT == W_F16) {
@@ -858,23 +853,41 @@ __device__ __forceinline__ void wave_tile_kloop(
         }
     };
 
-    // column-j norm scale as a packed f16 broadcast, hoisted out of the loop
-    const __half2 scale2 = __float2half2_rn(scale);
+    auto load_x = [&](XPanel& px) {
+#pragma unroll
+        for (int u = 0; u < PF; ++u) {
+#pragma unroll
+            for (int jt = 0; jt < JT; ++jt)
+                px.xb[u][jt] = *reinterpret_cast<const uint4*>(
+                    xp + (u * 4 * JT + jt) * 128);  // 4 kc-chunks per kb
+            if (NORM)
+                px.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
+        }
+        xp += PF * 4 * JT * 128;
+        if (NORM) np += PF * 32;
+    };
 
     auto compute_one = [&](int parity, const uint32_t q[NM],
                            const uint32_t ab[NM], const uint4 aw[NM],
-                           const uint4& xb, const uint4& nbv) {
-        ABFrag b;
-        if (NORM) {
-            const uint32_t xw[4] = {xb.x, xb.y, xb.z, xb.w};
-            const uint32_t nw[4] = {nbv.x, nbv.y, nbv.z, nbv.w};
+                           const uint4 xb[JT], const uint4& nbv) {
+        ABFrag b[JT];
 #pragma unroll
-            for (int w = 0; w < 4; ++w) {
-                const __half2 v = __hmul2(u2h2(xw[w]), u2h2(nw[w]));
-                b.u[w] = h22u(__hmul2(v, scale2));
+        for (int jt = 0; jt < JT; ++jt) {
+            if (NORM) {
+                const uint32_t xw[4] = {xb[jt].x, xb[jt].y, xb[jt].z,
+                                        xb[jt].w};
+                const uint32_t nw[4] = {nbv.x, nbv.y, nbv.z, nbv.w};
+#pragma unroll
+                for (int w = 0; w < 4; ++w) {
+                    const __half2 v = __hmul2(u2h2(xw[w]), u2h2(nw[w]));
+                    b[jt].u[w] = h22u(__hmul2(v, scale2[jt]));
+                }
+            } else {
+                b[jt].u[0] = xb[jt].x;
+                b[jt].u[1] = xb[jt].y;
+                b[jt].u[2] = xb[jt].z;
+                b[jt].u[3] = xb[jt].w;
             }
-        } else {
-            b.u[0] = xb.x; b.u[1] = xb.y; b.u[2] = xb.z; b.u[3] = xb.w;
         }
 #pragma unroll
         for (int n = 0; n < NM; ++n) {
@@ -885,12 +898,16 @@ __device__ __forceinline__ void wave_tile_kloop(
             } else {
                 a_frag_q4<WT>(q[n], ab[n], a);
             }
-            f32x4& c = parity ? c1[n] : c0[n];
-            c = __builtin_amdgcn_mfma_f32_16x16x32_f16(a.v, b.v, c, 0, 0, 0);
+#pragma unroll
+            for (int jt = 0; jt < JT; ++jt) {
+                f32x4& c = parity ? c1[n][jt] : c0[n][jt];
+                c = __builtin_amdgcn_mfma_f32_16x16x32_f16(a.v, b[jt].v, c,
+                                                           0, 0, 0);
+            }
         }
     };
 
-    auto compute_batch = [&](Batch& bt) {
+    auto compute_batch = [&](Batch& bt, XPanel& px) {
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
             uint32_t q[NM], ab[NM];
@@ -899,36 +916,39 @@ __device__ __forceinline__ void wave_tile_kloop(
                 q[n] = bt.q[u / 4][n][u % 4];
                 ab[n] = bt.ab[u / 4][n][u % 4];
             }
-            compute_one(u & 1, q, ab, bt.aw[u], bt.xb[u], bt.nbv[u]);
+            compute_one(u & 1, q, ab, bt.aw[u], px.xb[u], px.nbv[u]);
         }
     };
 
-    // Steady-state loop with UNCONDITIONAL prefetch: a branch around the
-    // load_batch would leave the outstanding-load count unknowable at the
-    // next compute and hipcc then drains vmcnt toward 0 inside the body
-    // (observed: vmcnt(7)..(0) chains; with this form it emits one counted
-    // vmcnt(12) per batch). The final iteration prefetches ONE batch past
-    // kb1 — every weight/side-channel allocation carries a tail slack of
-    // one batch for this (engine_ext.cpp kTailSlack*).
+    // Per iteration: [load B-panel (L2) -> prefetch next weights (HBM) ->
+    // compute]. The B-panel loads retire first (vmcnt is in-order and they
+    // are issued before the HBM batch), so compute's wait never drains the
+    // prefetch; one live XPanel keeps register pressure down.
     const int nfull = (kl.kb1 - kl.kb0) / PF;
     if (nfull > 0) {
-        load_batch(bufA);
+        load_w(bufA);
         int it = 0;
         while (true) {
-            load_batch(bufB);
-            compute_batch(bufA);
+            XPanel pxA;
+            load_x(pxA);
+            load_w(bufB);        // prefetch next weights (may overread 1)
+            compute_batch(bufA, pxA);
             if (++it == nfull) break;
-            load_batch(bufA);
-            compute_batch(bufB);
+            XPanel pxB;
+            load_x(pxB);
+            load_w(bufA);
+            compute_batch(bufB, pxB);
             if (++it == nfull) break;
         }
     }
     for (int g = kl.kb0 + nfull * PF; g < kl.kb1; ++g) {
         uint32_t q[NM], ab[NM];
         uint4 aw[NM];
-        uint4 xb, nbv;
-        xb = *reinterpret_cast<const uint4*>(xp);
-        xp += 512;
+        uint4 xb[JT], nbv;
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt)
+            xb[jt] = *reinterpret_cast<const uint4*>(xp + jt * 128);
+        xp += 4 * JT * 128;
         if (NORM) {
             nbv = *reinterpret_cast<const uint4*>(np);
             np += 32;
@@ -950,8 +970,10 @@ __device__ __forceinline__ void wave_tile_kloop(
 #pragma unroll
     for (int n = 0; n < NM; ++n)
 #pragma unroll
-        for (int jj = 0; jj < 4; ++jj)
-            acc[n][jj] = c0[n][jj] + c1[n][jj];
+        for (int jt = 0; jt < JT; ++jt)
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj)
+                acc[n][jt][jj] = c0[n][jt][jj] + c1[n][jt][jj];
 }
 
 // LDS combine of the 4 waves' partial accumulators; wave 0 ends with the
@@ -994,7 +1016,7 @@ __device__ __forceinline__ float norm_scale(const float* ss, int j, int cols,
 // measured 4.6 µs — launch+latency bound on 16 CUs of 256).
 __global__ void k_prep_x(const float* __restrict__ x,
                          unsigned short* __restrict__ xprep,
-                         float* __restrict__ ss, int cols) {
+                         float* __restrict__ ss, int cols, int jtw) {
     const int t = blockIdx.x;
     const float* xt = x + (size_t)t * cols;
     __shared__ float red[NWAVES];
@@ -1013,7 +1035,8 @@ __global__ void k_prep_x(const float* __restrict__ x,
         o.y = pack_f16(a.z, a.w);
         o.z = pack_f16(b.x, b.y);
         o.w = pack_f16(b.z, b.w);
-        *reinterpret_cast<uint4*>(xprep + ((size_t)kc * 16 + t) * 8) = o;
+        *reinterpret_cast<uint4*>(
+            xprep + (((size_t)kc * jtw + (t >> 4)) * 16 + (t & 15)) * 8) = o;
     }
     sum = wave_reduce_sum(sum);
     const int wid = threadIdx.x / WAVE;
@@ -1029,7 +1052,7 @@ __global__ void k_prep_x(const float* __restrict__ x,
 // (which already holds the residual stream); a following k_prep_x pass
 // rebuilds the sumsq/xprep side-channels. Used when rows/16 alone cannot
 // fill 256 CUs (wo: E/16 = 200 blocks).
-template <int WT, int MODE>
+template <int WT, int MODE, int JT>
 __global__ __launch_bounds__(BLOCK) void k_gemm16(
     WMat2 w, const unsigned short* __restrict__ bprep,
     const unsigned short* __restrict__ normprep,
@@ -1039,9 +1062,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     constexpr bool NORM = (MODE == GM_NORM_PLAIN);
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float scale = 1.0f;
-    if (NORM) scale = norm_scale(ss_in, j < T ? j : 0, w.cols, eps);
-    float acc[1][4];
+    float acc[1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbt = w.cols >> 5;
     const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);  // padded
@@ -1053,56 +1074,66 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
         b0 = min(nbk, (int)blockIdx.y * per);
         b1 = min(nbk, b0 + per);
     }
-    wave_tile_kloop<WT, NORM, 1>(ws, blockIdx.x, bprep, normprep, scale, acc,
-                                 b0, b1);
-    __shared__ float lds[3 * 64 * 4];
-    combine_acc<1>(acc, lds);
+    wave_tile_kloop<WT, NORM, 1, JT>(ws, blockIdx.x, bprep, normprep, ss_in,
+                                     eps, acc, b0, b1);
+    __shared__ float lds[3 * 64 * 4 * JT];
+    combine_acc<JT>(acc[0], lds);
     if (threadIdx.x >= WAVE) return;
+    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
     if (MODE == GM_ATOMIC) {
-        const int r0a = blockIdx.x * 16 + (lane >> 4) * 4;
-        if (j < T) {
 #pragma unroll
-            for (int jj = 0; jj < 4; ++jj)
-                atomicAdd(y + (size_t)j * w.rows + r0a + jj, acc[0][jj]);
+        for (int jt = 0; jt < JT; ++jt) {
+            const int j2 = jt * 16 + j;
+            if (j2 < T) {
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj)
+                    atomicAdd(y + (size_t)j2 * w.rows + r0 + jj,
+                              acc[0][jt][jj]);
+            }
         }
         return;
     }
-    // wave 0 epilogue: rows r = blockIdx.x*16 + (lane>>4)*4 + jj, col j.
-    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
-    float sq = 0.0f;
+    // wave-0 fused epilogue: rows r0..r0+3, col jt*16+j
 #pragma unroll
-    for (int jj = 0; jj < 4; ++jj) {
-        const int row = r0 + jj;
-        float v = acc[0][jj];
-        if (MODE == GM_RES_SQ) {
-            if (j < T) {
-                v += y[(size_t)j * w.rows + row];
-                y[(size_t)j * w.rows + row] = v;
-                sq += v * v;
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = jt * 16 + j;
+        float sq = 0.0f;
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            const int row = r0 + jj;
+            float v = acc[0][jt][jj];
+            if (MODE == GM_RES_SQ) {
+                if (j2 < T) {
+                    v += y[(size_t)j2 * w.rows + row];
+                    y[(size_t)j2 * w.rows + row] = v;
+                    sq += v * v;
+                }
+            } else {
+                if (j2 < T) y[(size_t)j2 * w.rows + row] = v;
             }
-        } else {
-            if (j < T) y[(size_t)j * w.rows + row] = v;
+            acc[0][jt][jj] = v;
         }
-        acc[0][jj] = v;
-    }
-    if (MODE == GM_RES_SQ && xprep_out != nullptr && j < T) {
-        // 4 consecutive rows -> one aligned 8 B bf16x4 chunk of xprep
-        uint2 o;
-        o.x = pack_f16(acc[0][0], acc[0][1]);
-        o.y = pack_f16(acc[0][2], acc[0][3]);
-        *reinterpret_cast<uint2*>(
-            xprep_out + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
-    }
-    if (MODE == GM_RES_SQ && ss_out != nullptr) {
-        sq += __shfl_xor(sq, 16);
-        sq += __shfl_xor(sq, 32);
-        if (lane < 16 && lane < T) atomicAdd(ss_out + lane, sq);
+        if (MODE == GM_RES_SQ && xprep_out != nullptr && j2 < T) {
+            // 4 consecutive rows -> one aligned 8 B f16x4 chunk of xprep
+            uint2 o;
+            o.x = pack_f16(acc[0][jt][0], acc[0][jt][1]);
+            o.y = pack_f16(acc[0][jt][2], acc[0][jt][3]);
+            *reinterpret_cast<uint2*>(
+                xprep_out + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
+                (r0 & 7)) = o;
+        }
+        if (MODE == GM_RES_SQ && ss_out != nullptr) {
+            float s2 = sq;
+            s2 += __shfl_xor(s2, 16);
+            s2 += __shfl_xor(s2, 32);
+            if (lane < 16 && j2 < T) atomicAdd(ss_out + j2, s2);
+        }
     }
 }
 
 // ------------------------------------------------------------- k_qkv16
 // QKV projections on MFMA + fused input RMSNorm + RoPE + KV append.
-template <int WT>
+template <int WT, int JT>
 __global__ __launch_bounds__(BLOCK) void k_qkv16(
     WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
@@ -1116,49 +1147,54 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    const float scale = norm_scale(ss_in, j < T ? j : 0, E, eps);
-    float acc[1][4];
+    float acc[1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 1>(ws, tile, xprep, normprep, scale, acc, 0,
-                                 nbe);
-    __shared__ float lds[3 * 64 * 4];
-    combine_acc<1>(acc, lds);
-    if (threadIdx.x >= WAVE || j >= T) return;
+    wave_tile_kloop<WT, true, 1, JT>(ws, tile, xprep, normprep, ss_in, eps,
+                                     acc, 0, nbe);
+    __shared__ float lds[3 * 64 * 4 * JT];
+    combine_acc<JT>(acc[0], lds);
+    if (threadIdx.x >= WAVE) return;
     const int r0 = tile * 16 + (lane >> 4) * 4;
-    const int p = pos[j];
-    if (mat == 2) {  // V rows: straight f16 cache append
-        __half* dst = v_cache + ((size_t)seq[j] * n_ctx + p) * E + r0;
 #pragma unroll
-        for (int jj = 0; jj < 4; ++jj) dst[jj] = __float2half(acc[0][jj]);
-        return;
-    }
-    // q/k: RoPE on in-lane pairs (rows r0+2q, r0+2q+1)
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = jt * 16 + j;
+        if (j2 >= T) continue;
+        const int p = pos[j2];
+        if (mat == 2) {  // V rows: straight f16 cache append
+            __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
 #pragma unroll
-    for (int q2 = 0; q2 < 2; ++q2) {
-        const int e = r0 + 2 * q2;
-        const int d = e % D;
-        const float theta = (float)p * inv_freq[d >> 1];
-        float sn, cs;
-        __sincosf(theta, &sn, &cs);
-        const float x0 = acc[0][2 * q2], x1 = acc[0][2 * q2 + 1];
-        const float o0 = x0 * cs - x1 * sn;
-        const float o1 = x0 * sn + x1 * cs;
-        if (mat == 0) {
-            q_buf[(size_t)j * E + e] = o0;
-            q_buf[(size_t)j * E + e + 1] = o1;
-        } else {
-            __half* dst = k_cache + ((size_t)seq[j] * n_ctx + p) * E + e;
-            dst[0] = __float2half(o0);
-            dst[1] = __float2half(o1);
+            for (int jj = 0; jj < 4; ++jj)
+                dst[jj] = __float2half(acc[0][jt][jj]);
+            continue;
+        }
+        // q/k: RoPE on in-lane pairs (rows r0+2q2, r0+2q2+1)
+#pragma unroll
+        for (int q2 = 0; q2 < 2; ++q2) {
+            const int e = r0 + 2 * q2;
+            const int d = e % D;
+            const float theta = (float)p * inv_freq[d >> 1];
+            float sn, cs;
+            __sincosf(theta, &sn, &cs);
+            const float x0 = acc[0][jt][2 * q2], x1 = acc[0][jt][2 * q2 + 1];
+            const float o0 = x0 * cs - x1 * sn;
+            const float o1 = x0 * sn + x1 * cs;
+            if (mat == 0) {
+                q_buf[(size_t)j2 * E + e] = o0;
+                q_buf[(size_t)j2 * E + e + 1] = o1;
+            } else {
+                __half* dst = k_cache + ((size_t)seq[j2] * n_ctx + p) * E + e;
+                dst[0] = __float2half(o0);
+                dst[1] = __float2half(o1);
+            }
         }
     }
 }
 
 // ------------------------------------------------------------- k_ffn16
 // w1 + w3 against the same B panel + fused input RMSNorm + SwiGLU; emits
-// the gate product straight into gprep (bf16 B-layout over F).
-template <int WT>
+// the gate product straight into gprep (f16 B-layout over F).
+template <int WT, int JT>
 __global__ __launch_bounds__(BLOCK) void k_ffn16(
     WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
@@ -1166,32 +1202,44 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     unsigned short* __restrict__ gprep, int T) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    const float scale = norm_scale(ss_in, j < T ? j : 0, w1.cols, eps);
-    float acc[2][4];
+    float acc[2][JT][4];
     const WMat2* ws[2] = {&w1, &w3};
     const int nbf = (WT == W_F16) ? (w1.cols >> 5)
                                   : (((w1.cols >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 2>(ws, blockIdx.x, xprep, normprep, scale,
-                                 acc, 0, nbf);
-    __shared__ float lds[3 * 64 * 4 * 2];
-    combine_acc<2>(acc, lds);
-    if (threadIdx.x >= WAVE || j >= T) return;
+    wave_tile_kloop<WT, true, 2, JT>(ws, blockIdx.x, xprep, normprep, ss_in,
+                                     eps, acc, 0, nbf);
+    __shared__ float lds[3 * 64 * 4 * 2 * JT];
+    combine_acc<2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
     const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
-    float g[4];
 #pragma unroll
-    for (int jj = 0; jj < 4; ++jj) {
-        const float v1 = acc[0][jj];
-        const float silu = v1 / (1.0f + __expf(-v1));
-        g[jj] = silu * acc[1][jj];
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = jt * 16 + j;
+        if (j2 >= T) continue;
+        float g[4];
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            const float v1 = acc[0][jt][jj];
+            const float silu = v1 / (1.0f + __expf(-v1));
+            g[jj] = silu * acc[1][jt][jj];
+        }
+        uint2 o;
+        o.x = pack_f16(g[0], g[1]);
+        o.y = pack_f16(g[2], g[3]);
+        *reinterpret_cast<uint2*>(
+            gprep + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
+            (r0 & 7)) = o;
     }
-    uint2 o;
-    o.x = pack_f16(g[0], g[1]);
-    o.y = pack_f16(g[2], g[3]);
-    *reinterpret_cast<uint2*>(
-        gprep + ((size_t)(r0 >> 3) * 16 + j) * 8 + (r0 & 7)) = o;
 }
 
 // ============================================================== launchers
+
+// column-tile count for a token batch T (1, 2 or 4 tiles of 16)
+static inline int pick_jt(int T) {
+    if (T <= 16) return 1;
+    if (T <= 32) return 2;
+    return 4;
+}
 
 static inline int pick_tmax(int T) {
     if (T <= 1) return 1;
@@ -1283,7 +1331,7 @@ void launch_attention(hipStream_t s, const float* q_buf,
     const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
     hipLaunchKernelGGL(k_attention, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos, seq,
-                       E, D, n_ctx);
+                       E, D, n_ctx, pick_jt(T));
 }
 
 // ------------------------------------------------- MFMA-path launchers
@@ -1295,8 +1343,27 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
     const int want = 256 / max(T, 1);
     const int chunks = max(1, min(want, cols / 1024));
     hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
-                       xprep, ss, cols);
+                       xprep, ss, cols, pick_jt(T));
 }
+
+#define DISPATCH_JT(JTV, ...)            \
+    switch (JTV) {                       \
+        case 1: {                        \
+            constexpr int JTc = 1;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        case 2: {                        \
+            constexpr int JTc = 2;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+        default: {                       \
+            constexpr int JTc = 4;       \
+            __VA_ARGS__;                 \
+            break;                       \
+        }                                \
+    }
 
 #define DISPATCH_WT2(WTV, ...)                   \
     switch (WTV) {                               \
@@ -1328,26 +1395,26 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
         int ks = 1;
         while (R * ks < 512 && ks < 8) ks <<= 1;
         const dim3 grid(R, ks);
-        DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
-            (k_gemm16<WTc, GM_ATOMIC>), grid, dim3(BLOCK), 0, s, w, bprep,
-            normprep, ss_in, eps, y, xprep_out, ss_out, T));
+        DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+            (k_gemm16<WTc, GM_ATOMIC, JTc>), grid, dim3(BLOCK), 0, s, w,
+            bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
         return;
     }
     const dim3 grid(R);
-    DISPATCH_WT2(w.wtype, {
+    DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), {
         if (mode == GM_RES_SQ)
-            hipLaunchKernelGGL((k_gemm16<WTc, GM_RES_SQ>), grid, dim3(BLOCK),
-                               0, s, w, bprep, normprep, ss_in, eps, y,
-                               xprep_out, ss_out, T);
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_RES_SQ, JTc>), grid,
+                               dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
+                               eps, y, xprep_out, ss_out, T);
         else if (mode == GM_NORM_PLAIN)
-            hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN>), grid,
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc>), grid,
                                dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
                                eps, y, xprep_out, ss_out, T);
         else
-            hipLaunchKernelGGL((k_gemm16<WTc, GM_PLAIN>), grid, dim3(BLOCK),
-                               0, s, w, bprep, normprep, ss_in, eps, y,
-                               xprep_out, ss_out, T);
-    });
+            hipLaunchKernelGGL((k_gemm16<WTc, GM_PLAIN, JTc>), grid,
+                               dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
+                               eps, y, xprep_out, ss_out, T);
+    }));
 }
 
 void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
@@ -1357,10 +1424,10 @@ void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                   __half* v_cache_layer, const int* pos, const int* seq,
                   const float* inv_freq, int E, int D, int n_ctx, int T) {
     const dim3 grid(3 * (E >> 4));
-    DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
-        (k_qkv16<WTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep, normprep,
-        ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq, inv_freq,
-        E, D, n_ctx, T));
+    DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+        (k_qkv16<WTc, JTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
+        normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq,
+        inv_freq, E, D, n_ctx, T)));
 }
 
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
@@ -1368,9 +1435,9 @@ void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
                   const unsigned short* normprep, const float* ss_in,
                   float eps, unsigned short* gprep, int T) {
     const dim3 grid(w1.rows / 16);
-    DISPATCH_WT2(w1.wtype, hipLaunchKernelGGL(
-        (k_ffn16<WTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep, normprep,
-        ss_in, eps, gprep, T));
+    DISPATCH_WT2(w1.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+        (k_ffn16<WTc, JTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep,
+        normprep, ss_in, eps, gprep, T)));
 }
 
 void launch_gemv(hipStream_t s, const WMat& w, const float* x,

@@ -100,11 +100,28 @@ def test_batched_decode_matches():
         _assert_close(y_gpu, y_cpu, label=f"batched step {t}")
 
 
-def test_prefill_tiling_over_max_tokens():
-    """T > kMaxTokens goes through host-side token tiling."""
-    f, hip, cpu = _engines(n_ctx=64)
+@pytest.mark.parametrize("T", [20, 40, 64])
+def test_multi_tile_token_batches(T):
+    """T > 16 runs the JT=2/4 multi-column-tile MFMA path (with a partial
+    last tile at T=20/40), including its logits/sampling stages."""
+    f, hip, cpu = _engines(n_ctx=96, max_batch=1)
     hp = f.hparams
-    T = 20
+    x = torch.randn(T, hp.n_embd) * 0.5
+    pos = torch.arange(T, dtype=torch.int32)
+    seq = torch.zeros(T, dtype=torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label=f"multi-tile T={T}")
+    lg_gpu = hip.logits(y_gpu.cuda().contiguous(), all_logits=True).cpu()
+    lg_cpu = cpu.logits(y_cpu, all_logits=True)
+    _assert_close(lg_gpu, lg_cpu, label=f"multi-tile logits T={T}")
+
+
+def test_prefill_tiling_over_max_tokens():
+    """T > kMaxTokens (64) goes through host-side token tiling."""
+    f, hip, cpu = _engines(n_ctx=96)
+    hp = f.hparams
+    T = 70
     x = torch.randn(T, hp.n_embd) * 0.5
     pos = torch.arange(T, dtype=torch.int32)
     seq = torch.zeros(T, dtype=torch.int32)
