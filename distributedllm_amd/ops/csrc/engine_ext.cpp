@@ -19,6 +19,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 
+#include <cstdlib>
 #include <memory>
 #include <string>
 #include <vector>
@@ -185,6 +186,9 @@ class SliceEngine {
         ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
         argmax_keys_ = torch::zeros({kMaxTokens}, dev.dtype(torch::kInt64));
+        // split-K partial slabs for wo/w2: f32[R][KS][kMaxTokens][16]
+        slab_ = torch::zeros(
+            {(int64_t)(E_ / 16) * gemm16_ks(E_) * kMaxTokens * 16}, f32);
     }
 
     void set_layer(int64_t li, torch::Tensor attn_norm,
@@ -295,13 +299,18 @@ class SliceEngine {
             launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_, D_,
                              ctx_);
             // wo/w2 tile count (E/16) alone underfills 256 CUs — split K
-            // across grid.y with atomic partials, then rebuild the
-            // sumsq/xprep side channel with the (cheap) prep pass.
-            const bool split = (E_ / 16) < 512;
+            // across grid.y into plain-stored slabs, then one fused
+            // reduce+residual+sumsq+xprep pass per matrix (no atomics;
+            // the kernel boundary provides slab visibility).
+            static const bool no_split = std::getenv("DLLM_NO_SPLITK");
+            const bool split = !no_split && (E_ / 16) < 512;
+            float* slab = slab_.data_ptr<float>();
+            const int ks = gemm16_ks(E_);
             if (split) {
-                launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
-                              nullptr, nullptr, T, GM_ATOMIC);
-                launch_prep_x(s, xp, xprep, ssf + li * kMaxTokens, E_, T);
+                launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_,
+                              slab, nullptr, nullptr, T, GM_SLAB);
+                launch_reduce_prep(s, xp, slab, ks, xprep,
+                                   ssf + li * kMaxTokens, E_, T);
             } else {
                 launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
                               xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
@@ -309,10 +318,10 @@ class SliceEngine {
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
                          ssf + li * kMaxTokens, eps_, gprep, T);
             if (split) {
-                launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
-                              nullptr, nullptr, T, GM_ATOMIC);
-                launch_prep_x(s, xp, xprep, ssa + (li + 1) * kMaxTokens, E_,
-                              T);
+                launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_,
+                              slab, nullptr, nullptr, T, GM_SLAB);
+                launch_reduce_prep(s, xp, slab, ks, xprep,
+                                   ssa + (li + 1) * kMaxTokens, E_, T);
             } else {
                 launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
                               xprep, ssa + (li + 1) * kMaxTokens, T,
@@ -421,7 +430,7 @@ class SliceEngine {
     torch::Tensor k_cache_, v_cache_, inv_freq_;
     torch::Tensor xn_, qb_, ab_, ffb_;
     torch::Tensor xprep_, aprep_, gprep_, ss_attn_, ss_ffn_, ss_tmp_;
-    torch::Tensor argmax_keys_;
+    torch::Tensor argmax_keys_, slab_;
     bool has_extra_ = false;
     bool out_mfma_ = false;
     DevMat tok_, out_;

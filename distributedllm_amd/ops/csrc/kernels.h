@@ -32,7 +32,14 @@ struct WMat2 {
 // k_gemm16 fused-epilogue modes. GM_ATOMIC = grid-level split-K partials
 // atomicAdd'ed into y (residual pre-loaded); follow with launch_prep_x.
 enum GemmMode { GM_PLAIN = 0, GM_RES_SQ = 1, GM_NORM_PLAIN = 2,
-                GM_ATOMIC = 3 };
+                GM_ATOMIC = 3, GM_SLAB = 4 };
+
+// split-K factor used by GM_ATOMIC/GM_SLAB (shared with the reducer)
+int gemm16_ks(int rows);
+
+// y[t] += sum of ks slab partials, then sumsq + f16 xprep of the result
+void launch_reduce_prep(hipStream_t s, float* y, const float* slab, int ks,
+                        unsigned short* xprep, float* ss, int cols, int T);
 
 void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T);

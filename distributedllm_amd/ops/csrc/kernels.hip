@@ -1046,6 +1046,60 @@ __global__ void k_prep_x(const float* __restrict__ x,
         atomicAdd(ss + t, red[0] + red[1] + red[2] + red[3]);
 }
 
+// ------------------------------------------------------- k_reduce_prep
+// Split-K reducer + side-channel prep in one pass: y[t] += sum of the KS
+// slab partials, then sumsq + f16 xprep of the result (replaces atomic
+// split-K partials + a separate k_prep_x — the 819K atomicAdd/kernel of
+// the atomic form dominated at JT=4). Grid (T, C); ss zeroed upstream.
+__global__ void k_reduce_prep(float* __restrict__ y,
+                              const float* __restrict__ slab, int ks,
+                              unsigned short* __restrict__ xprep,
+                              float* __restrict__ ss, int cols, int jtw) {
+    const int t = blockIdx.x;
+    float* yt = y + (size_t)t * cols;
+    __shared__ float red[NWAVES];
+    const int nkc = cols >> 3;
+    const int per = (nkc + gridDim.y - 1) / gridDim.y;
+    const int kc0 = blockIdx.y * per;
+    const int kc1 = min(nkc, kc0 + per);
+    float sum = 0.0f;
+    for (int kc = kc0 + threadIdx.x; kc < kc1; kc += BLOCK) {
+        float4 a = *reinterpret_cast<const float4*>(yt + kc * 8);
+        float4 b = *reinterpret_cast<const float4*>(yt + kc * 8 + 4);
+        const int e0 = kc * 8;            // 8 consecutive rows
+        const int tile = e0 >> 4;         // 16-row tile
+        const int i0 = e0 & 15;           // 0 or 8 within the tile
+        const float* sbase =
+            slab + ((size_t)tile * ks * 64 + t) * 16 + i0;
+        const size_t kstride = (size_t)64 * 16;
+        for (int k = 0; k < ks; ++k) {
+            const float4 p0 = *reinterpret_cast<const float4*>(
+                sbase + k * kstride);
+            const float4 p1 = *reinterpret_cast<const float4*>(
+                sbase + k * kstride + 4);
+            a.x += p0.x; a.y += p0.y; a.z += p0.z; a.w += p0.w;
+            b.x += p1.x; b.y += p1.y; b.z += p1.z; b.w += p1.w;
+        }
+        *reinterpret_cast<float4*>(yt + kc * 8) = a;
+        *reinterpret_cast<float4*>(yt + kc * 8 + 4) = b;
+        sum += a.x * a.x + a.y * a.y + a.z * a.z + a.w * a.w;
+        sum += b.x * b.x + b.y * b.y + b.z * b.z + b.w * b.w;
+        uint4 o;
+        o.x = pack_f16(a.x, a.y);
+        o.y = pack_f16(a.z, a.w);
+        o.z = pack_f16(b.x, b.y);
+        o.w = pack_f16(b.z, b.w);
+        *reinterpret_cast<uint4*>(
+            xprep + (((size_t)kc * jtw + (t >> 4)) * 16 + (t & 15)) * 8) = o;
+    }
+    sum = wave_reduce_sum(sum);
+    const int wid = threadIdx.x / WAVE;
+    if ((threadIdx.x & (WAVE - 1)) == 0) red[wid] = sum;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        atomicAdd(ss + t, red[0] + red[1] + red[2] + red[3]);
+}
+
 // ------------------------------------------------------------- k_gemm16
 // MODE GM_ATOMIC: grid-level split-K — gridDim.y blocks per row-tile each
 // cover nb/gridDim.y K-blocks and atomicAdd their partial tile into y
@@ -1067,7 +1121,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     const int nbt = w.cols >> 5;
     const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);  // padded
     int b0 = 0, b1 = nbk;
-    if (MODE == GM_ATOMIC) {
+    if (MODE == GM_ATOMIC || MODE == GM_SLAB) {
         // split in 4-aligned units so q4 group loads stay whole
         int per = (nbk + gridDim.y - 1) / gridDim.y;
         per = (per + 3) & ~3;
@@ -1090,6 +1144,26 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
                     atomicAdd(y + (size_t)j2 * w.rows + r0 + jj,
                               acc[0][jt][jj]);
             }
+        }
+        return;
+    }
+    if (MODE == GM_SLAB) {
+        // plain-store the partial tile; k_reduce_prep sums the KS slices
+        // at the next kernel boundary (no atomics, no pre-zeroing; the
+        // boundary provides visibility). Slab layout:
+        // f32[R][KS][kMaxTok][16 rows]; `y` is reused as the slab base.
+        float* slab = y +
+            ((size_t)blockIdx.x * gridDim.y + blockIdx.y) * 64 * 16;
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt) {
+            const int j2 = jt * 16 + j;
+            float4 v;
+            v.x = acc[0][jt][0];
+            v.y = acc[0][jt][1];
+            v.z = acc[0][jt][2];
+            v.w = acc[0][jt][3];
+            *reinterpret_cast<float4*>(
+                slab + ((size_t)j2 * 16) + (lane >> 4) * 4) = v;
         }
         return;
     }
@@ -1384,20 +1458,40 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
         }                                        \
     }
 
+int gemm16_ks(int rows) {
+    const int R = rows / 16;
+    int ks = 1;
+    while (R * ks < 512 && ks < 8) ks <<= 1;
+    return ks;
+}
+
+void launch_reduce_prep(hipStream_t s, float* y, const float* slab, int ks,
+                        unsigned short* xprep, float* ss, int cols, int T) {
+    const int want = 256 / max(T, 1);
+    const int chunks = max(1, min(want, cols / 1024));
+    hipLaunchKernelGGL(k_reduce_prep, dim3(T, chunks), dim3(BLOCK), 0, s, y,
+                       slab, ks, xprep, ss, cols, pick_jt(T));
+}
+
 void launch_gemm16(hipStream_t s, const WMat2& w,
                    const unsigned short* bprep,
                    const unsigned short* normprep, const float* ss_in,
                    float eps, float* y, unsigned short* xprep_out,
                    float* ss_out, int T, int mode) {
     const int R = w.rows / 16;
-    if (mode == GM_ATOMIC) {
-        // pick the split so R*KS lands near 2-3 blocks/CU (256 CUs)
-        int ks = 1;
-        while (R * ks < 512 && ks < 8) ks <<= 1;
+    if (mode == GM_ATOMIC || mode == GM_SLAB) {
+        // split K so R*KS lands near 2-3 blocks/CU (256 CUs)
+        const int ks = gemm16_ks(w.rows);
         const dim3 grid(R, ks);
-        DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
-            (k_gemm16<WTc, GM_ATOMIC, JTc>), grid, dim3(BLOCK), 0, s, w,
-            bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
+        if (mode == GM_SLAB) {
+            DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+                (k_gemm16<WTc, GM_SLAB, JTc>), grid, dim3(BLOCK), 0, s, w,
+                bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
+        } else {
+            DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+                (k_gemm16<WTc, GM_ATOMIC, JTc>), grid, dim3(BLOCK), 0, s, w,
+                bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
+        }
         return;
     }
     const dim3 grid(R);
