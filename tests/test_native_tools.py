@@ -1,0 +1,90 @@
+"""Native C++ tooling: byte-compatibility with the Python formats layer.
+
+The reference ships C++ `slice_model` and a vendored `quantize` binary
+(SURVEY §2.2 N2/N4); ours live in tools/ and must produce files that are
+byte-identical to the Python slicer/codec — the Python side is the tested
+canon, the C++ side must match it exactly.
+"""
+import subprocess
+from pathlib import Path
+
+import numpy as np
+import pytest
+
+from distributedllm_amd.formats import ggml, q4, slicer, synthetic
+
+TOOLS = Path(__file__).resolve().parent.parent / "tools"
+
+
+@pytest.fixture(scope="module")
+def bins(tmp_path_factory):
+    subprocess.run(["make", "-C", str(TOOLS)], check=True,
+                   capture_output=True)
+    return TOOLS / "bin"
+
+
+@pytest.fixture(scope="module")
+def model_file(tmp_path_factory):
+    d = tmp_path_factory.mktemp("native")
+    f = synthetic.build_model("tiny", ftype=ggml.FTYPE_MOSTLY_F16, seed=7)
+    path = d / "model_f16.bin"
+    f.save(str(path))
+    return f, path
+
+
+def test_slice_model_matches_python(bins, model_file, tmp_path):
+    f, path = model_file
+    out_cpp = tmp_path / "slice_cpp.bin"
+    subprocess.run([str(bins / "slice_model"), "slice", str(path), "1", "2",
+                    str(out_cpp)], check=True, capture_output=True)
+    out_py = tmp_path / "slice_py.bin"
+    slicer.make_slice(f, 1, 2).save(str(out_py))
+    assert out_cpp.read_bytes() == out_py.read_bytes()
+
+
+def test_extra_layers_matches_python(bins, model_file, tmp_path):
+    f, path = model_file
+    out_cpp = tmp_path / "extra_cpp.bin"
+    subprocess.run([str(bins / "slice_model"), "extra_layers", str(path),
+                    str(out_cpp)], check=True, capture_output=True)
+    out_py = tmp_path / "extra_py.bin"
+    slicer.make_extra_layers(f).save(str(out_py))
+    assert out_cpp.read_bytes() == out_py.read_bytes()
+
+
+def test_sliced_file_loads_and_runs(bins, model_file, tmp_path):
+    """A C++-produced slice loads through the normal engine path."""
+    _, path = model_file
+    out = tmp_path / "s.bin"
+    subprocess.run([str(bins / "slice_model"), "slice", str(path), "0", "1",
+                    str(out)], check=True, capture_output=True)
+    sf = ggml.GGMLFile.load(str(out), extended=True)
+    assert sf.hparams.first_layer == 0
+    assert sf.hparams.n_layer == 2
+    from distributedllm_amd.engine import TorchSliceEngine
+    import torch
+    eng = TorchSliceEngine.from_ggml(sf, n_ctx=16, max_batch=1)
+    x = torch.randn(2, sf.hparams.n_embd)
+    y = eng.forward(x, torch.tensor([0, 1], dtype=torch.int32),
+                    torch.zeros(2, dtype=torch.int32))
+    assert torch.isfinite(y).all()
+
+
+@pytest.mark.parametrize("target", ["q4_0", "q4_1"])
+def test_quantize_matches_python_codec(bins, model_file, tmp_path, target):
+    f, path = model_file
+    out_cpp = tmp_path / f"model_{target}.bin"
+    subprocess.run([str(bins / "quantize"), str(path), str(out_cpp), target],
+                   check=True, capture_output=True)
+    got = ggml.GGMLFile.load(str(out_cpp), extended=False)
+    gtype = (ggml.GGML_TYPE_Q4_0 if target == "q4_0"
+             else ggml.GGML_TYPE_Q4_1)
+    quant = q4.quantize_q4_0 if target == "q4_0" else q4.quantize_q4_1
+    assert got.hparams.ftype == gtype
+    for t in f.tensors:
+        g = got.tensor_map()[t.name]
+        if len(t.ne) == 1:
+            assert g.raw == t.raw  # 1-D stays f32
+            continue
+        want = quant(t.to_f32()).tobytes()
+        assert g.raw == want, f"{t.name}: q4 bytes differ from Python codec"

@@ -1,0 +1,120 @@
+// quantize — native CLI that requantizes a GGML model to q4_0/q4_1.
+//
+// Native counterpart of the vendored llama.cpp `quantize` binary the
+// reference's provisioning shells out to
+// (/root/reference/distllm/cli_api/provision.py:213-217), clean-room per
+// the block layouts in formats/q4.py: q4_0 blocks are
+// bit-identical to the Python codec (signed-amax scale, f16 d, nibble
+// pairs j/j+16). 1-D tensors stay f32, matching real checkpoints.
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "ggmlio.hpp"
+
+namespace {
+
+std::vector<float> to_f32(const ggmlio::Tensor& t) {
+    const size_t rows = t.rows(), cols = t.ne[0];
+    std::vector<float> out(rows * cols);
+    const uint8_t* p = t.raw.data();
+    switch (t.gtype) {
+        case ggmlio::F32:
+            std::memcpy(out.data(), p, out.size() * 4);
+            break;
+        case ggmlio::F16:
+            for (size_t i = 0; i < out.size(); ++i) {
+                uint16_t h;
+                std::memcpy(&h, p + i * 2, 2);
+                out[i] = ggmlio::f16_to_f32(h);
+            }
+            break;
+        case ggmlio::Q4_0:
+            for (size_t i = 0; i < out.size() / ggmlio::kQK; ++i)
+                ggmlio::dequantize_block_q4_0(p + i * ggmlio::kQ4_0Bytes,
+                                              out.data() + i * ggmlio::kQK);
+            break;
+        case ggmlio::Q4_1:
+            for (size_t i = 0; i < out.size() / ggmlio::kQK; ++i)
+                ggmlio::dequantize_block_q4_1(p + i * ggmlio::kQ4_1Bytes,
+                                              out.data() + i * ggmlio::kQK);
+            break;
+    }
+    return out;
+}
+
+ggmlio::Tensor from_f32(const ggmlio::Tensor& t, const std::vector<float>& x,
+                        ggmlio::GType target) {
+    ggmlio::Tensor out;
+    out.name = t.name;
+    out.ne = t.ne;
+    out.gtype = target;
+    out.raw.resize(out.nbytes());
+    uint8_t* p = out.raw.data();
+    switch (target) {
+        case ggmlio::F32:
+            std::memcpy(p, x.data(), x.size() * 4);
+            break;
+        case ggmlio::F16:
+            for (size_t i = 0; i < x.size(); ++i) {
+                const uint16_t h = ggmlio::f32_to_f16(x[i]);
+                std::memcpy(p + i * 2, &h, 2);
+            }
+            break;
+        case ggmlio::Q4_0:
+            for (size_t i = 0; i < x.size() / ggmlio::kQK; ++i)
+                ggmlio::quantize_block_q4_0(x.data() + i * ggmlio::kQK,
+                                            p + i * ggmlio::kQ4_0Bytes);
+            break;
+        case ggmlio::Q4_1:
+            for (size_t i = 0; i < x.size() / ggmlio::kQK; ++i)
+                ggmlio::quantize_block_q4_1(x.data() + i * ggmlio::kQK,
+                                            p + i * ggmlio::kQ4_1Bytes);
+            break;
+    }
+    return out;
+}
+
+}  // namespace
+
+int main(int argc, char** argv) {
+    if (argc < 4) {
+        std::fprintf(stderr,
+                     "usage: quantize <in.bin> <out.bin> <q4_0|q4_1|f16>\n");
+        return 2;
+    }
+    ggmlio::GType target;
+    const std::string t = argv[3];
+    if (t == "q4_0") target = ggmlio::Q4_0;
+    else if (t == "q4_1") target = ggmlio::Q4_1;
+    else if (t == "f16") target = ggmlio::F16;
+    else {
+        std::fprintf(stderr, "unknown target type %s\n", t.c_str());
+        return 2;
+    }
+    try {
+        ggmlio::Reader reader(argv[1]);
+        ggmlio::File in = reader.parse(/*extended=*/false);
+        ggmlio::File out;
+        out.hp = in.hp;
+        out.hp.ftype = (uint32_t)target;  // ftype ids match GType ids
+        out.vocab = in.vocab;
+        size_t quantized = 0;
+        for (const auto& ten : in.tensors) {
+            if (ten.ne.size() == 1 || ten.gtype == target) {
+                out.tensors.push_back(ten);  // 1-D stays f32, same-type copy
+                continue;
+            }
+            out.tensors.push_back(from_f32(ten, to_f32(ten), target));
+            ++quantized;
+        }
+        ggmlio::write_file(argv[2], out);
+        std::printf("quantized %zu of %zu tensors -> %s\n", quantized,
+                    in.tensors.size(), argv[2]);
+        return 0;
+    } catch (const std::exception& e) {
+        std::fprintf(stderr, "error: %s\n", e.what());
+        return 1;
+    }
+}
