@@ -19,6 +19,7 @@
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
 
+#include <algorithm>
 #include <cstdlib>
 #include <memory>
 #include <string>
@@ -186,9 +187,11 @@ class SliceEngine {
         ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
         argmax_keys_ = torch::zeros({kMaxTokens}, dev.dtype(torch::kInt64));
-        // split-K partial slabs for wo/w2: f32[R][KS][kMaxTokens][16]
-        slab_ = torch::zeros(
-            {(int64_t)(E_ / 16) * gemm16_ks(E_) * kMaxTokens * 16}, f32);
+        // split-K partial slabs: sized for the largest user — qkv
+        // (3E/16 tiles), ffn (2F/16), wo/w2 (E/16) — at the max split of 8
+        const int64_t slab_tiles =
+            std::max<int64_t>({3 * (E_ / 16), 2 * (F_ / 16), E_ / 16});
+        slab_ = torch::zeros({slab_tiles * 8 * kMaxTokens * 16}, f32);
     }
 
     void set_layer(int64_t li, torch::Tensor attn_norm,
@@ -295,7 +298,8 @@ class SliceEngine {
             __half* vc = vbase + (size_t)li * layer_stride;
             launch_qkv16(s, l.mq.w, l.mk.w, l.mv.w, xprep,
                          u16p(l.attn_normprep), ssa + li * kMaxTokens, eps_,
-                         qb, kc, vc, pp, sp, ifr, E_, D_, ctx_, T);
+                         qb, kc, vc, pp, sp, ifr, E_, D_, ctx_, T,
+                         slab_.data_ptr<float>());
             launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_, D_,
                              ctx_);
             // wo/w2 tile count (E/16) alone underfills 256 CUs — split K
@@ -315,8 +319,11 @@ class SliceEngine {
                 launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
                               xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
             }
+            // ffn keeps the fused RT=1 kernel: measured par with the
+            // slab+finish variant (its LDS/VGPR cost offsets the B saving)
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
-                         ssf + li * kMaxTokens, eps_, gprep, T);
+                         ssf + li * kMaxTokens, eps_, gprep, T,
+                         /*slab=*/nullptr);
             if (split) {
                 launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_,
                               slab, nullptr, nullptr, T, GM_SLAB);

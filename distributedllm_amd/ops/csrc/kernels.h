@@ -50,17 +50,19 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
                    float eps, float* y, unsigned short* xprep_out,
                    float* ss_out, int T, int mode);
 
+// slab != nullptr enables the RT=2 + split-K slab path for small models
 void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                   const WMat2& wv, const unsigned short* xprep,
                   const unsigned short* normprep, const float* ss_in,
                   float eps, float* q_buf, __half* k_cache_layer,
                   __half* v_cache_layer, const int* pos, const int* seq,
-                  const float* inv_freq, int E, int D, int n_ctx, int T);
+                  const float* inv_freq, int E, int D, int n_ctx, int T,
+                  float* slab);
 
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
                   const unsigned short* xprep,
                   const unsigned short* normprep, const float* ss_in,
-                  float eps, unsigned short* gprep, int T);
+                  float eps, unsigned short* gprep, int T, float* slab);
 
 void launch_rmsnorm(hipStream_t s, const float* x, const float* w, float* y,
                     int T, int E, float eps);

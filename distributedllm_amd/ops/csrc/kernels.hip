@@ -761,26 +761,35 @@ __device__ __forceinline__ void a_frag_q4(uint32_t q, uint32_t ab,
 //  * accumulation is MFMA C-chained over two alternating accumulator
 //    sets (covers dependent-accumulator latency).
 // acc[n][jt][jj] ends with rows (l>>4)*4+jj, col jt*16 + (l&15).
-template <int WT, bool NORM, int NM, int JT, int PF = 4>
+// RT = row tiles per wave: the wave's B fragments feed RT A-tile streams,
+// dividing the (L2-heavy) B-panel re-read traffic and the B-build VALU by
+// RT. acc[rt][n][jt][jj].
+template <int WT, bool NORM, int NM, int JT, int RT = 1, int PF = 4>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
     const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
     const float* __restrict__ ss_in, float eps,
-    float acc[NM][JT][4], int b0, int b1) {
+    float acc[RT][NM][JT][4], int b0, int b1) {
     KLoop kl;
     kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4);
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-    f32x4 c0[NM][JT], c1[NM][JT];
+    // RT>1 brings enough independent accumulators to cover the MFMA
+    // dependent latency without the dual c0/c1 set (which would double
+    // AGPR pressure); RT==1 keeps the alternating pair.
+    constexpr int NP = (RT > 1) ? 1 : 2;
+    f32x4 c0[RT][NM][JT], c1[NP == 2 ? RT : 1][NM][JT];
 #pragma unroll
-    for (int n = 0; n < NM; ++n)
+    for (int rt = 0; rt < RT; ++rt)
 #pragma unroll
-        for (int jt = 0; jt < JT; ++jt) {
-            c0[n][jt] = zero;
-            c1[n][jt] = zero;
-        }
+        for (int n = 0; n < NM; ++n)
+#pragma unroll
+            for (int jt = 0; jt < JT; ++jt) {
+                c0[rt][n][jt] = zero;
+                if (NP == 2) c1[rt][n][jt] = zero;
+            }
 
     // per-column-tile RMSNorm scale, hoisted (column j = lane&15 of tile jt)
     __half2 scale2[JT];
@@ -792,20 +801,23 @@ __device__ __forceinline__ void wave_tile_kloop(
                 rsqrtf(ss_in[jt * 16 + kl.i] * inv_cols + eps));
     }
 
-    const uint32_t* qp[NM];
-    const uint32_t* abp[NM];
-    const unsigned short* tp[NM];
+    const uint32_t* qp[RT][NM];
+    const uint32_t* abp[RT][NM];
+    const unsigned short* tp[RT][NM];
 #pragma unroll
-    for (int n = 0; n < NM; ++n) {
-        qp[n] = (const uint32_t*)ws[n]->data +
-                ((size_t)tile_row * nb + kl.kb0) * 64 +
-                (kl.ks * 16 + kl.i) * 4;
-        abp[n] = (const uint32_t*)ws[n]->scales +
-                 ((size_t)tile_row * nb + kl.kb0) * 16 + kl.i * 4;
-        tp[n] = (const unsigned short*)ws[n]->data +
-                ((size_t)tile_row * (ws[n]->cols >> 3)) * 128 +
-                ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
-    }
+    for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
+        for (int n = 0; n < NM; ++n) {
+            const int tr = tile_row * RT + rt;
+            qp[rt][n] = (const uint32_t*)ws[n]->data +
+                        ((size_t)tr * nb + kl.kb0) * 64 +
+                        (kl.ks * 16 + kl.i) * 4;
+            abp[rt][n] = (const uint32_t*)ws[n]->scales +
+                         ((size_t)tr * nb + kl.kb0) * 16 + kl.i * 4;
+            tp[rt][n] = (const unsigned short*)ws[n]->data +
+                        ((size_t)tr * (ws[n]->cols >> 3)) * 128 +
+                        ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
+        }
     // xprep layout with JT: element (kc, jt, j, e) at ((kc*JT+jt)*16+j)*8+e
     const unsigned short* xp =
         xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * JT) * 128 + kl.i * 8;
@@ -813,8 +825,8 @@ __device__ __forceinline__ void wave_tile_kloop(
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
     struct Batch {  // weight stream only (HBM, nt, double-buffered)
-        u32x4 q[PF / 4][NM], ab[PF / 4][NM];
-        uint4 aw[PF][NM];
+        u32x4 q[PF / 4][RT][NM], ab[PF / 4][RT][NM];
+        uint4 aw[PF][RT][NM];
     };
     // NAMED buffers, never indexed by a runtime value (a runtime select
     // sends the array to scratch: measured 240-336 B/lane and 3x slower).
@@ -828,29 +840,36 @@ __device__ __forceinline__ void wave_tile_kloop(
 #pragma unroll
         for (int u4 = 0; u4 < PF / 4; ++u4) {
 #pragma unroll
+            for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
+                for (int n = 0; n < NM; ++n) {
+                    if (WT == W_F16) {
+#pragma unroll
+                        for (int v = 0; v < 4; ++v)
+                            bt.aw[u4 * 4 + v][rt][n] =
+                                *reinterpret_cast<const uint4*>(
+                                    tp[rt][n] + (u4 * 4 + v) * 512);
+                    } else {
+                        bt.q[u4][rt][n] = __builtin_nontemporal_load(
+                            reinterpret_cast<const u32x4*>(qp[rt][n]) +
+                            u4 * 64);
+                        bt.ab[u4][rt][n] = __builtin_nontemporal_load(
+                            reinterpret_cast<const u32x4*>(abp[rt][n]) +
+                            u4 * 16);
+                    }
+                }
+        }
+#pragma unroll
+        for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
             for (int n = 0; n < NM; ++n) {
                 if (WT == W_F16) {
-#pragma unroll
-                    for (int v = 0; v < 4; ++v)
-                        bt.aw[u4 * 4 + v][n] = *reinterpret_cast<const uint4*>(
-                            tp[n] + (u4 * 4 + v) * 512);
+                    tp[rt][n] += PF * 512;
                 } else {
-                    bt.q[u4][n] = __builtin_nontemporal_load(
-                        reinterpret_cast<const u32x4*>(qp[n]) + u4 * 64);
-                    bt.ab[u4][n] = __builtin_nontemporal_load(
-                        reinterpret_cast<const u32x4*>(abp[n]) + u4 * 16);
+                    qp[rt][n] += PF * 64;
+                    abp[rt][n] += PF * 16;
                 }
             }
-        }
-#pragma unroll
-        for (int n = 0; n < NM; ++n) {
-            if (WT == W_F16) {
-                tp[n] += PF * 512;
-            } else {
-                qp[n] += PF * 64;
-                abp[n] += PF * 16;
-            }
-        }
     };
 
     auto load_x = [&](XPanel& px) {
@@ -867,9 +886,10 @@ __device__ __forceinline__ void wave_tile_kloop(
         if (NORM) np += PF * 32;
     };
 
-    auto compute_one = [&](int parity, const uint32_t q[NM],
-                           const uint32_t ab[NM], const uint4 aw[NM],
-                           const uint4 xb[JT], const uint4& nbv) {
+    auto compute_one = [&](int parity, const uint32_t q[RT][NM],
+                           const uint32_t ab[RT][NM],
+                           const uint4 aw[RT][NM], const uint4 xb[JT],
+                           const uint4& nbv) {
         ABFrag b[JT];
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt) {
@@ -890,32 +910,37 @@ __device__ __forceinline__ void wave_tile_kloop(
             }
         }
 #pragma unroll
-        for (int n = 0; n < NM; ++n) {
-            ABFrag a;
-            if (WT == W_F16) {
-                a.u[0] = aw[n].x; a.u[1] = aw[n].y;
-                a.u[2] = aw[n].z; a.u[3] = aw[n].w;
-            } else {
-                a_frag_q4<WT>(q[n], ab[n], a);
-            }
+        for (int rt = 0; rt < RT; ++rt)
 #pragma unroll
-            for (int jt = 0; jt < JT; ++jt) {
-                f32x4& c = parity ? c1[n][jt] : c0[n][jt];
-                c = __builtin_amdgcn_mfma_f32_16x16x32_f16(a.v, b[jt].v, c,
-                                                           0, 0, 0);
+            for (int n = 0; n < NM; ++n) {
+                ABFrag a;
+                if (WT == W_F16) {
+                    a.u[0] = aw[rt][n].x; a.u[1] = aw[rt][n].y;
+                    a.u[2] = aw[rt][n].z; a.u[3] = aw[rt][n].w;
+                } else {
+                    a_frag_q4<WT>(q[rt][n], ab[rt][n], a);
+                }
+#pragma unroll
+                for (int jt = 0; jt < JT; ++jt) {
+                    f32x4& c = (NP == 2 && parity) ? c1[rt][n][jt]
+                                                   : c0[rt][n][jt];
+                    c = __builtin_amdgcn_mfma_f32_16x16x32_f16(a.v, b[jt].v,
+                                                               c, 0, 0, 0);
+                }
             }
-        }
     };
 
     auto compute_batch = [&](Batch& bt, XPanel& px) {
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
-            uint32_t q[NM], ab[NM];
+            uint32_t q[RT][NM], ab[RT][NM];
 #pragma unroll
-            for (int n = 0; n < NM; ++n) {
-                q[n] = bt.q[u / 4][n][u % 4];
-                ab[n] = bt.ab[u / 4][n][u % 4];
-            }
+            for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
+                for (int n = 0; n < NM; ++n) {
+                    q[rt][n] = bt.q[u / 4][rt][n][u % 4];
+                    ab[rt][n] = bt.ab[u / 4][rt][n][u % 4];
+                }
             compute_one(u & 1, q, ab, bt.aw[u], px.xb[u], px.nbv[u]);
         }
     };
@@ -942,8 +967,8 @@ __device__ __forceinline__ void wave_tile_kloop(
         }
     }
     for (int g = kl.kb0 + nfull * PF; g < kl.kb1; ++g) {
-        uint32_t q[NM], ab[NM];
-        uint4 aw[NM];
+        uint32_t q[RT][NM], ab[RT][NM];
+        uint4 aw[RT][NM];
         uint4 xb[JT], nbv;
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt)
@@ -954,26 +979,32 @@ __device__ __forceinline__ void wave_tile_kloop(
             np += 32;
         }
 #pragma unroll
-        for (int n = 0; n < NM; ++n) {
-            if (WT == W_F16) {
-                aw[n] = *reinterpret_cast<const uint4*>(tp[n]);
-                tp[n] += 512;
-            } else {
-                q[n] = __builtin_nontemporal_load(qp[n]);
-                qp[n] += 64;
-                ab[n] = __builtin_nontemporal_load(abp[n]);
-                abp[n] += 16;
+        for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
+            for (int n = 0; n < NM; ++n) {
+                if (WT == W_F16) {
+                    aw[rt][n] = *reinterpret_cast<const uint4*>(tp[rt][n]);
+                    tp[rt][n] += 512;
+                } else {
+                    q[rt][n] = __builtin_nontemporal_load(qp[rt][n]);
+                    qp[rt][n] += 64;
+                    ab[rt][n] = __builtin_nontemporal_load(abp[rt][n]);
+                    abp[rt][n] += 16;
+                }
             }
-        }
         compute_one(g & 1, q, ab, aw, xb, nbv);
     }
 #pragma unroll
-    for (int n = 0; n < NM; ++n)
+    for (int rt = 0; rt < RT; ++rt)
 #pragma unroll
-        for (int jt = 0; jt < JT; ++jt)
+        for (int n = 0; n < NM; ++n)
 #pragma unroll
-            for (int jj = 0; jj < 4; ++jj)
-                acc[n][jt][jj] = c0[n][jt][jj] + c1[n][jt][jj];
+            for (int jt = 0; jt < JT; ++jt)
+#pragma unroll
+                for (int jj = 0; jj < 4; ++jj)
+                    acc[rt][n][jt][jj] =
+                        (NP == 2) ? c0[rt][n][jt][jj] + c1[rt][n][jt][jj]
+                                  : c0[rt][n][jt][jj];
 }
 
 // LDS combine of the 4 waves' partial accumulators; wave 0 ends with the
@@ -1116,7 +1147,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     constexpr bool NORM = (MODE == GM_NORM_PLAIN);
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][JT][4];
+    float acc[1][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbt = w.cols >> 5;
     const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);  // padded
@@ -1131,7 +1162,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     wave_tile_kloop<WT, NORM, 1, JT>(ws, blockIdx.x, bprep, normprep, ss_in,
                                      eps, acc, b0, b1);
     __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(acc[0], lds);
+    combine_acc<JT>(acc[0][0], lds);
     if (threadIdx.x >= WAVE) return;
     const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
     if (MODE == GM_ATOMIC) {
@@ -1142,7 +1173,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
 #pragma unroll
                 for (int jj = 0; jj < 4; ++jj)
                     atomicAdd(y + (size_t)j2 * w.rows + r0 + jj,
-                              acc[0][jt][jj]);
+                              acc[0][0][jt][jj]);
             }
         }
         return;
@@ -1158,10 +1189,10 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
         for (int jt = 0; jt < JT; ++jt) {
             const int j2 = jt * 16 + j;
             float4 v;
-            v.x = acc[0][jt][0];
-            v.y = acc[0][jt][1];
-            v.z = acc[0][jt][2];
-            v.w = acc[0][jt][3];
+            v.x = acc[0][0][jt][0];
+            v.y = acc[0][0][jt][1];
+            v.z = acc[0][0][jt][2];
+            v.w = acc[0][0][jt][3];
             *reinterpret_cast<float4*>(
                 slab + ((size_t)j2 * 16) + (lane >> 4) * 4) = v;
         }
@@ -1175,7 +1206,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
 #pragma unroll
         for (int jj = 0; jj < 4; ++jj) {
             const int row = r0 + jj;
-            float v = acc[0][jt][jj];
+            float v = acc[0][0][jt][jj];
             if (MODE == GM_RES_SQ) {
                 if (j2 < T) {
                     v += y[(size_t)j2 * w.rows + row];
@@ -1185,13 +1216,13 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
             } else {
                 if (j2 < T) y[(size_t)j2 * w.rows + row] = v;
             }
-            acc[0][jt][jj] = v;
+            acc[0][0][jt][jj] = v;
         }
         if (MODE == GM_RES_SQ && xprep_out != nullptr && j2 < T) {
             // 4 consecutive rows -> one aligned 8 B f16x4 chunk of xprep
             uint2 o;
-            o.x = pack_f16(acc[0][jt][0], acc[0][jt][1]);
-            o.y = pack_f16(acc[0][jt][2], acc[0][jt][3]);
+            o.x = pack_f16(acc[0][0][jt][0], acc[0][0][jt][1]);
+            o.y = pack_f16(acc[0][0][jt][2], acc[0][0][jt][3]);
             *reinterpret_cast<uint2*>(
                 xprep_out + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
                 (r0 & 7)) = o;
@@ -1221,13 +1252,13 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][JT][4];
+    float acc[1][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
     wave_tile_kloop<WT, true, 1, JT>(ws, tile, xprep, normprep, ss_in, eps,
                                      acc, 0, nbe);
     __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(acc[0], lds);
+    combine_acc<JT>(acc[0][0], lds);
     if (threadIdx.x >= WAVE) return;
     const int r0 = tile * 16 + (lane >> 4) * 4;
 #pragma unroll
@@ -1239,7 +1270,7 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
             __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
-                dst[jj] = __float2half(acc[0][jt][jj]);
+                dst[jj] = __float2half(acc[0][0][jt][jj]);
             continue;
         }
         // q/k: RoPE on in-lane pairs (rows r0+2q2, r0+2q2+1)
@@ -1250,7 +1281,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
             const float theta = (float)p * inv_freq[d >> 1];
             float sn, cs;
             __sincosf(theta, &sn, &cs);
-            const float x0 = acc[0][jt][2 * q2], x1 = acc[0][jt][2 * q2 + 1];
+            const float x0 = acc[0][0][jt][2 * q2];
+            const float x1 = acc[0][0][jt][2 * q2 + 1];
             const float o0 = x0 * cs - x1 * sn;
             const float o1 = x0 * sn + x1 * cs;
             if (mat == 0) {
@@ -1276,7 +1308,7 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     unsigned short* __restrict__ gprep, int T) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[2][JT][4];
+    float acc[1][2][JT][4];
     const WMat2* ws[2] = {&w1, &w3};
     const int nbf = (WT == W_F16) ? (w1.cols >> 5)
                                   : (((w1.cols >> 5) + 3) & ~3);
@@ -1293,9 +1325,9 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
         float g[4];
 #pragma unroll
         for (int jj = 0; jj < 4; ++jj) {
-            const float v1 = acc[0][jt][jj];
+            const float v1 = acc[0][0][jt][jj];
             const float silu = v1 / (1.0f + __expf(-v1));
-            g[jj] = silu * acc[1][jt][jj];
+            g[jj] = silu * acc[0][1][jt][jj];
         }
         uint2 o;
         o.x = pack_f16(g[0], g[1]);
@@ -1304,6 +1336,211 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
             gprep + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
             (r0 & 7)) = o;
     }
+}
+
+// ------------------------------------------------- slab qkv/ffn variants
+// For models whose tile counts underfill the chip (3B: qkv 600 tiles, ffn
+// 540), the fused kernels run at ~2.3 blocks/CU with the full B panel
+// re-read per tile. These variants use RT=2 row tiles per wave (halves
+// the B-panel L2 traffic and B-build VALU) plus grid-level split-K into
+// slabs; the finish kernels sum the slabs and run the rope/cache (qkv) or
+// SwiGLU (ffn) epilogue.
+
+template <int WT, int JT, int RT>
+__global__ __launch_bounds__(BLOCK) void k_qkv16_slab(
+    WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps, float* __restrict__ slab,
+    int E, int T) {
+    const int st_per_mat = (E >> 4) / RT;
+    const int mat = blockIdx.x / st_per_mat;
+    const int stile = blockIdx.x % st_per_mat;
+    const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float acc[RT][1][JT][4];
+    const WMat2* ws[1] = {&w};
+    const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
+    int per = (nbe + gridDim.y - 1) / gridDim.y;
+    per = (per + 3) & ~3;
+    const int b0 = min(nbe, (int)blockIdx.y * per);
+    const int b1 = min(nbe, b0 + per);
+    wave_tile_kloop<WT, true, 1, JT, RT>(ws, stile, xprep, normprep, ss_in,
+                                         eps, acc, b0, b1);
+    __shared__ float lds[3 * 64 * 4 * RT * JT];
+    combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+        const size_t gtile = (size_t)mat * (E >> 4) + stile * RT + rt;
+        float* sl = slab + ((gtile * gridDim.y + blockIdx.y) * 64) * 16;
+#pragma unroll
+        for (int jt = 0; jt < JT; ++jt) {
+            const int j2 = jt * 16 + j;
+            float4 v;
+            v.x = acc[rt][0][jt][0];
+            v.y = acc[rt][0][jt][1];
+            v.z = acc[rt][0][jt][2];
+            v.w = acc[rt][0][jt][3];
+            *reinterpret_cast<float4*>(sl + (size_t)j2 * 16 +
+                                       (lane >> 4) * 4) = v;
+        }
+    }
+}
+
+// Sum the qkv slabs, apply RoPE, write q_buf / KV caches. One thread per
+// (mat, 8-row chunk, token): vectorized float4 slab reads and uint4/f16x8
+// stores (scalar 2-4 B accesses made the first version a 5 us kernel).
+__global__ void k_qkv_finish(const float* __restrict__ slab, int ks,
+                             float* __restrict__ q_buf,
+                             __half* __restrict__ k_cache,
+                             __half* __restrict__ v_cache,
+                             const int* __restrict__ pos,
+                             const int* __restrict__ seq,
+                             const float* __restrict__ inv_freq, int E,
+                             int D, int n_ctx, int T) {
+    const int nchunks = 3 * E / 8;
+    const int idx = blockIdx.x * BLOCK + threadIdx.x;
+    if (idx >= nchunks * T) return;
+    const int chunk = idx % nchunks;
+    const int t = idx / nchunks;
+    const int mat = (8 * chunk) / E;
+    const int e = (8 * chunk) % E;   // 8-aligned
+    const size_t gt0 = ((size_t)mat * (E >> 4) + (e >> 4)) * ks;
+    float v[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) v[i] = 0.f;
+    for (int k = 0; k < ks; ++k) {
+        const float* sl = slab + ((gt0 + k) * 64 + t) * 16 + (e & 15);
+        const float4 a = *reinterpret_cast<const float4*>(sl);
+        const float4 b = *reinterpret_cast<const float4*>(sl + 4);
+        v[0] += a.x; v[1] += a.y; v[2] += a.z; v[3] += a.w;
+        v[4] += b.x; v[5] += b.y; v[6] += b.z; v[7] += b.w;
+    }
+    const int p = pos[t];
+    if (mat == 2) {
+        __half* dst = v_cache + ((size_t)seq[t] * n_ctx + p) * E + e;
+        uint4 o;
+        o.x = pack_f16(v[0], v[1]);
+        o.y = pack_f16(v[2], v[3]);
+        o.z = pack_f16(v[4], v[5]);
+        o.w = pack_f16(v[6], v[7]);
+        *reinterpret_cast<uint4*>(dst) = o;
+        return;
+    }
+    float o[8];
+#pragma unroll
+    for (int q2 = 0; q2 < 4; ++q2) {
+        const int d = (e + 2 * q2) % D;
+        const float theta = (float)p * inv_freq[d >> 1];
+        float sn, cs;
+        __sincosf(theta, &sn, &cs);
+        o[2 * q2] = v[2 * q2] * cs - v[2 * q2 + 1] * sn;
+        o[2 * q2 + 1] = v[2 * q2] * sn + v[2 * q2 + 1] * cs;
+    }
+    if (mat == 0) {
+        float* dst = q_buf + (size_t)t * E + e;
+        *reinterpret_cast<float4*>(dst) =
+            make_float4(o[0], o[1], o[2], o[3]);
+        *reinterpret_cast<float4*>(dst + 4) =
+            make_float4(o[4], o[5], o[6], o[7]);
+    } else {
+        __half* dst = k_cache + ((size_t)seq[t] * n_ctx + p) * E + e;
+        uint4 w;
+        w.x = pack_f16(o[0], o[1]);
+        w.y = pack_f16(o[2], o[3]);
+        w.z = pack_f16(o[4], o[5]);
+        w.w = pack_f16(o[6], o[7]);
+        *reinterpret_cast<uint4*>(dst) = w;
+    }
+}
+
+template <int WT, int JT, int RT>
+__global__ __launch_bounds__(BLOCK) void k_ffn16_slab(
+    WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps, float* __restrict__ slab,
+    int T) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float acc[RT][2][JT][4];
+    const WMat2* ws[2] = {&w1, &w3};
+    const int nbf = (WT == W_F16) ? (w1.cols >> 5)
+                                  : (((w1.cols >> 5) + 3) & ~3);
+    int per = (nbf + gridDim.y - 1) / gridDim.y;
+    per = (per + 3) & ~3;
+    const int b0 = min(nbf, (int)blockIdx.y * per);
+    const int b1 = min(nbf, b0 + per);
+    wave_tile_kloop<WT, true, 2, JT, RT>(ws, blockIdx.x, xprep, normprep,
+                                         ss_in, eps, acc, b0, b1);
+    __shared__ float lds[3 * 64 * 4 * RT * 2 * JT];
+    combine_acc<RT * 2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
+    const int Ftiles = w1.rows >> 4;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+#pragma unroll
+        for (int n = 0; n < 2; ++n) {
+            const size_t gtile = (size_t)n * Ftiles + blockIdx.x * RT + rt;
+            float* sl = slab + ((gtile * gridDim.y + blockIdx.y) * 64) * 16;
+#pragma unroll
+            for (int jt = 0; jt < JT; ++jt) {
+                const int j2 = jt * 16 + j;
+                float4 v;
+                v.x = acc[rt][n][jt][0];
+                v.y = acc[rt][n][jt][1];
+                v.z = acc[rt][n][jt][2];
+                v.w = acc[rt][n][jt][3];
+                *reinterpret_cast<float4*>(sl + (size_t)j2 * 16 +
+                                           (lane >> 4) * 4) = v;
+            }
+        }
+    }
+}
+
+// Sum the w1/w3 slabs, SwiGLU, emit gprep. One thread per (8-row chunk,
+// token): float4 slab reads, one uint4 (f16x8) gprep store.
+__global__ void k_ffn_finish(const float* __restrict__ slab, int ks,
+                             unsigned short* __restrict__ gprep, int F,
+                             int T, int jtw) {
+    const int nchunks = F / 8;
+    const int idx = blockIdx.x * BLOCK + threadIdx.x;
+    if (idx >= nchunks * T) return;
+    const int chunk = idx % nchunks;
+    const int t = idx / nchunks;
+    const int r = chunk * 8;
+    const int Ftiles = F >> 4;
+    const size_t t1 = ((size_t)(r >> 4)) * ks;
+    const size_t t3 = ((size_t)Ftiles + (r >> 4)) * ks;
+    float s1[8], s3[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) { s1[i] = 0.f; s3[i] = 0.f; }
+    for (int k = 0; k < ks; ++k) {
+        const float* p1 = slab + ((t1 + k) * 64 + t) * 16 + (r & 15);
+        const float* p3 = slab + ((t3 + k) * 64 + t) * 16 + (r & 15);
+        const float4 a1 = *reinterpret_cast<const float4*>(p1);
+        const float4 b1 = *reinterpret_cast<const float4*>(p1 + 4);
+        const float4 a3 = *reinterpret_cast<const float4*>(p3);
+        const float4 b3 = *reinterpret_cast<const float4*>(p3 + 4);
+        s1[0] += a1.x; s1[1] += a1.y; s1[2] += a1.z; s1[3] += a1.w;
+        s1[4] += b1.x; s1[5] += b1.y; s1[6] += b1.z; s1[7] += b1.w;
+        s3[0] += a3.x; s3[1] += a3.y; s3[2] += a3.z; s3[3] += a3.w;
+        s3[4] += b3.x; s3[5] += b3.y; s3[6] += b3.z; s3[7] += b3.w;
+    }
+    float g[8];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+        const float silu = s1[i] / (1.0f + __expf(-s1[i]));
+        g[i] = silu * s3[i];
+    }
+    uint4 o;
+    o.x = pack_f16(g[0], g[1]);
+    o.y = pack_f16(g[2], g[3]);
+    o.z = pack_f16(g[4], g[5]);
+    o.w = pack_f16(g[6], g[7]);
+    *reinterpret_cast<uint4*>(
+        gprep + (((size_t)(r >> 3) * jtw + (t >> 4)) * 16 + (t & 15)) * 8) =
+        o;
 }
 
 // ============================================================== launchers
@@ -1414,8 +1651,9 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T) {
     // chunk columns so the launch spreads over ~256 CUs even at T=16
     // (a T-block launch measured 4.6 us on 16 CUs); ss zeroed upstream
-    const int want = 256 / max(T, 1);
-    const int chunks = max(1, min(want, cols / 1024));
+    // target ~512 blocks with >=64 K-chunks of work per block
+    const int want = (512 + T - 1) / max(T, 1);
+    const int chunks = max(1, min(want, (cols >> 3) / 64));
     hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
                        xprep, ss, cols, pick_jt(T));
 }
@@ -1467,8 +1705,8 @@ int gemm16_ks(int rows) {
 
 void launch_reduce_prep(hipStream_t s, float* y, const float* slab, int ks,
                         unsigned short* xprep, float* ss, int cols, int T) {
-    const int want = 256 / max(T, 1);
-    const int chunks = max(1, min(want, cols / 1024));
+    const int want = (512 + T - 1) / max(T, 1);
+    const int chunks = max(1, min(want, (cols >> 3) / 64));
     hipLaunchKernelGGL(k_reduce_prep, dim3(T, chunks), dim3(BLOCK), 0, s, y,
                        slab, ks, xprep, ss, cols, pick_jt(T));
 }
@@ -1516,8 +1754,26 @@ void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                   const unsigned short* normprep, const float* ss_in,
                   float eps, float* q_buf, __half* k_cache_layer,
                   __half* v_cache_layer, const int* pos, const int* seq,
-                  const float* inv_freq, int E, int D, int n_ctx, int T) {
-    const dim3 grid(3 * (E >> 4));
+                  const float* inv_freq, int E, int D, int n_ctx, int T,
+                  float* slab) {
+    const int tiles3 = 3 * (E >> 4);
+    // slab split-K + RT=2 path when the fused grid underfills the chip
+    if (slab != nullptr && tiles3 < 1024 && ((E >> 4) % 2) == 0) {
+        int ks = 1;
+        while ((tiles3 / 2) * ks < 512 && ks < 8) ks <<= 1;
+        const dim3 grid(tiles3 / 2, ks);
+        DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+            (k_qkv16_slab<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk,
+            wv, xprep, normprep, ss_in, eps, slab, E, T)));
+        const int total = (3 * E / 8) * T;
+        hipLaunchKernelGGL(k_qkv_finish,
+                           dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK),
+                           0, s, slab, ks, q_buf, k_cache_layer,
+                           v_cache_layer, pos, seq, inv_freq, E, D, n_ctx,
+                           T);
+        return;
+    }
+    const dim3 grid(tiles3);
     DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_qkv16<WTc, JTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
         normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq,
@@ -1527,8 +1783,22 @@ void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
                   const unsigned short* xprep,
                   const unsigned short* normprep, const float* ss_in,
-                  float eps, unsigned short* gprep, int T) {
-    const dim3 grid(w1.rows / 16);
+                  float eps, unsigned short* gprep, int T, float* slab) {
+    const int tilesF = w1.rows / 16;
+    if (slab != nullptr && tilesF < 1024 && (tilesF % 2) == 0) {
+        int ks = 1;
+        while ((tilesF / 2) * ks < 512 && ks < 8) ks <<= 1;
+        const dim3 grid(tilesF / 2, ks);
+        DISPATCH_WT2(w1.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+            (k_ffn16_slab<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, w1, w3,
+            xprep, normprep, ss_in, eps, slab, T)));
+        const int total = (w1.rows / 8) * T;
+        hipLaunchKernelGGL(k_ffn_finish,
+                           dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK),
+                           0, s, slab, ks, gprep, w1.rows, T, pick_jt(T));
+        return;
+    }
+    const dim3 grid(tilesF);
     DISPATCH_WT2(w1.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_ffn16<WTc, JTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep,
         normprep, ss_in, eps, gprep, T)));
