@@ -180,6 +180,9 @@ class GenerateTextCommand(Command):
                                   greedy=args.greedy, seed=args.seed):
             print(piece, end="", flush=True)
         print()
+        r = llm.throughput.report()
+        print(f"[{r['count']:.0f} tokens in {r['seconds']:.2f}s = "
+              f"{r['per_second']:.2f} tok/s]", file=sys.stderr)
         return 0
 
 

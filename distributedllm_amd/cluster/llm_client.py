@@ -22,6 +22,7 @@ import torch
 
 from ..engine.sampler import Sampler
 from ..engine.tokenizer import Tokenizer
+from ..utils.metrics import StageTimer, ThroughputMeter
 from ..formats import ggml
 from ..models.llama import LlamaExtraRef, weights_from_ggml
 from .client import Connection, parse_address
@@ -37,6 +38,10 @@ class DistributedLLM:
         self.tokenizer = Tokenizer(extra.vocab)
         self.extra = LlamaExtraRef(weights_from_ggml(extra))
         self.n_past = 0
+        # built-in observability (the reference never reports
+        # its timing fields) — read after generate()/perplexity()
+        self.throughput = ThroughputMeter()
+        self.stages = StageTimer()
 
     # ------------------------------------------------------------- plumbing
 
@@ -48,8 +53,10 @@ class DistributedLLM:
     def propagate_tensor(self, x: np.ndarray) -> np.ndarray:
         """One hop through every node in layer order (the reference's
         client-mediated pipeline, common.py:148-154)."""
-        for conn, _, _ in self.nodes:
+        for i, (conn, _, _) in enumerate(self.nodes):
+            self.stages.start(f"hop{i}")
             x = conn.propagate_forward(x, start_pos=self.n_past)
+            self.stages.stop(f"hop{i}")
         self.n_past += x.shape[0]
         return x
 
@@ -67,6 +74,7 @@ class DistributedLLM:
                  seed: Optional[int] = None,
                  greedy: bool = False) -> Iterator[str]:
         self.clear_context()
+        self.throughput.reset()
         tokens = self.tokenizer.encode(prompt, bos=True)
         sampler = Sampler(temperature, repeat_penalty, seed=seed,
                           greedy=greedy)
@@ -76,6 +84,7 @@ class DistributedLLM:
             y = self.propagate_tensor(x)
             logits = self._logits(y, all_logits=False)[0]
             tid = sampler(logits)
+            self.throughput.tick()
             yield self.tokenizer.decode_token(tid)
             cur = [tid]
 

@@ -105,3 +105,16 @@ class TestSampler:
         logits = np.linspace(-1, 1, 32)
         assert [a(logits) for _ in range(10)] == \
                [b(logits) for _ in range(10)]
+
+
+def test_metrics_meters():
+    import time
+    from distributedllm_amd.utils.metrics import StageTimer, ThroughputMeter
+    m = ThroughputMeter()
+    m.tick(); time.sleep(0.01); m.tick(); m.tick(2)
+    r = m.report()
+    assert r["count"] == 4 and r["per_second"] > 0
+    t = StageTimer()
+    t.start("a"); time.sleep(0.005); t.stop("a")
+    rep = t.report()
+    assert rep["a"]["count"] == 1 and rep["a"]["seconds"] >= 0.004
