@@ -215,3 +215,22 @@ def test_find_n_mult_real_models():
         p = PRESETS[name]
         m = find_n_mult(p.n_embd, p.n_ff)
         assert ((2 * (4 * p.n_embd) // 3 + m - 1) // m) * m == p.n_ff
+
+
+def test_control_center(node, tmp_path):
+    """Cluster status model (reference ControlCenter capability)."""
+    from distributedllm_amd.cluster.control import ControlCenter, ModelSlice
+    addr = f"127.0.0.1:{node.port}"
+    cc = ControlCenter({addr: [0, 2]})
+    st = cc.get_status()[addr]
+    assert st.connectivity and not st.slice_loaded
+    ready, why = cc.pipeline_ready(n_layer=3)
+    assert not ready and "no slice" in why
+
+    cc.validate_slices(4, [ModelSlice("a", 0, 1), ModelSlice("b", 2, 3)])
+    with pytest.raises(ValueError):
+        cc.validate_slices(4, [ModelSlice("a", 0, 1), ModelSlice("b", 3, 3)])
+
+    # unreachable node
+    cc2 = ControlCenter({"127.0.0.1:1": [0, 0]})
+    assert not cc2.get_status()["127.0.0.1:1"].connectivity

@@ -61,6 +61,7 @@ def test_embed_matches(ftype):
 
 
 @pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_0,
+                                   ggml.FTYPE_MOSTLY_Q4_1,
                                    ggml.FTYPE_MOSTLY_F16])
 def test_forward_prefill_matches(ftype):
     f, hip, cpu = _engines(ftype=ftype)
@@ -173,3 +174,43 @@ def test_random_init_engine_runs():
     assert torch.isfinite(y).all()
     lg = eng.logits(y, all_logits=True)
     assert torch.isfinite(lg).all()
+
+
+def test_cli_end_to_end_on_gpu(tmp_path):
+    """The reference workflow (node + provision + generate + perplexity)
+    with the node's engine on the MI355X — proves the GGML slice path
+    loads into the HIP engine, not just the synthetic one."""
+    import json
+    import threading
+    from distributedllm_amd.cli import execute_command
+    from distributedllm_amd.cluster.node import NodeServer
+
+    srv = NodeServer("127.0.0.1", 0, str(tmp_path / "uploads"),
+                     device="cuda", n_ctx=128)
+    threading.Thread(target=srv.serve_forever, daemon=True).start()
+    try:
+        addr = f"127.0.0.1:{srv.port}"
+        root = tmp_path / "root"
+        root.mkdir()
+        cfg = {"model_id": "tiny_gpu", "location": "synthetic:tiny",
+               "nodes_map": {addr: [0, PRESETS["tiny"].n_layer - 1]},
+               "quantization": "q4_0",
+               "metadata": {"name": "tinygpu", "family": "llama_v1"}}
+        cfg_path = tmp_path / "cfg.json"
+        cfg_path.write_text(json.dumps(cfg))
+        assert execute_command(["provision", str(cfg_path),
+                                "--root", str(root)]) == 0
+        assert execute_command(["generate_text", str(cfg_path),
+                                "--prompt", "hello", "--num-tokens", "4",
+                                "--greedy", "--root", str(root)]) == 0
+        assert execute_command(["perplexity", str(cfg_path),
+                                "--prompt", "hello world words",
+                                "--root", str(root)]) == 0
+        # the node must actually be running the HIP engine
+        from distributedllm_amd.cluster.client import Connection
+        c = Connection("127.0.0.1", srv.port)
+        assert c.get_status().device == "cuda"
+        c.close()
+    finally:
+        srv.shutdown()
+        srv.server_close()
