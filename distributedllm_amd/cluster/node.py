@@ -68,6 +68,12 @@ class EngineSlice:
     def forward(self, x: np.ndarray, start_pos: int) -> np.ndarray:
         t = torch.from_numpy(np.ascontiguousarray(x, dtype=np.float32))
         T = t.shape[0]
+        if start_pos + T > self.engine.n_ctx:
+            # the reference never guards n_ctx overflow —
+            # here it would silently corrupt a neighboring sequence's KV
+            raise SliceError(
+                f"context overflow: start_pos {start_pos} + {T} tokens "
+                f"exceeds n_ctx {self.engine.n_ctx}")
         dev = self.device
         pos = torch.arange(start_pos, start_pos + T, dtype=torch.int32)
         seq = torch.zeros(T, dtype=torch.int32)

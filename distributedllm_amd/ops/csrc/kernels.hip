@@ -1137,7 +1137,7 @@ __global__ void k_reduce_prep(float* __restrict__ y,
 // (which already holds the residual stream); a following k_prep_x pass
 // rebuilds the sumsq/xprep side-channels. Used when rows/16 alone cannot
 // fill 256 CUs (wo: E/16 = 200 blocks).
-template <int WT, int MODE, int JT>
+template <int WT, int MODE, int JT, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_gemm16(
     WMat2 w, const unsigned short* __restrict__ bprep,
     const unsigned short* __restrict__ normprep,
@@ -1147,7 +1147,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
     constexpr bool NORM = (MODE == GM_NORM_PLAIN);
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][1][JT][4];
+    float acc[RT][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbt = w.cols >> 5;
     const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);  // padded
@@ -1159,12 +1159,12 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
         b0 = min(nbk, (int)blockIdx.y * per);
         b1 = min(nbk, b0 + per);
     }
-    wave_tile_kloop<WT, NORM, 1, JT>(ws, blockIdx.x, bprep, normprep, ss_in,
-                                     eps, acc, b0, b1);
-    __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(acc[0][0], lds);
+    wave_tile_kloop<WT, NORM, 1, JT, RT>(ws, blockIdx.x, bprep, normprep,
+                                         ss_in, eps, acc, b0, b1);
+    __shared__ float lds[3 * 64 * 4 * JT * RT];
+    combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
+    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;  // RT==1 modes
     if (MODE == GM_ATOMIC) {
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt) {
@@ -1198,7 +1198,10 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
         }
         return;
     }
-    // wave-0 fused epilogue: rows r0..r0+3, col jt*16+j
+    // wave-0 fused epilogue: rows r0..r0+3 per row tile, col jt*16+j
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+    const int r0 = (blockIdx.x * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j2 = jt * 16 + j;
@@ -1206,7 +1209,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
 #pragma unroll
         for (int jj = 0; jj < 4; ++jj) {
             const int row = r0 + jj;
-            float v = acc[0][0][jt][jj];
+            float v = acc[rt][0][jt][jj];
             if (MODE == GM_RES_SQ) {
                 if (j2 < T) {
                     v += y[(size_t)j2 * w.rows + row];
@@ -1216,13 +1219,13 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
             } else {
                 if (j2 < T) y[(size_t)j2 * w.rows + row] = v;
             }
-            acc[0][0][jt][jj] = v;
+            acc[rt][0][jt][jj] = v;
         }
         if (MODE == GM_RES_SQ && xprep_out != nullptr && j2 < T) {
             // 4 consecutive rows -> one aligned 8 B f16x4 chunk of xprep
             uint2 o;
-            o.x = pack_f16(acc[0][0][jt][0], acc[0][0][jt][1]);
-            o.y = pack_f16(acc[0][0][jt][2], acc[0][0][jt][3]);
+            o.x = pack_f16(acc[rt][0][jt][0], acc[rt][0][jt][1]);
+            o.y = pack_f16(acc[rt][0][jt][2], acc[rt][0][jt][3]);
             *reinterpret_cast<uint2*>(
                 xprep_out + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
                 (r0 & 7)) = o;
@@ -1233,6 +1236,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
             s2 += __shfl_xor(s2, 32);
             if (lane < 16 && j2 < T) atomicAdd(ss_out + j2, s2);
         }
+    }
     }
 }
 
@@ -1738,10 +1742,19 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
             hipLaunchKernelGGL((k_gemm16<WTc, GM_RES_SQ, JTc>), grid,
                                dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
                                eps, y, xprep_out, ss_out, T);
-        else if (mode == GM_NORM_PLAIN)
-            hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc>), grid,
-                               dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
-                               eps, y, xprep_out, ss_out, T);
+        else if (mode == GM_NORM_PLAIN) {
+            // lm_head: thousands of row tiles re-read the same B panel —
+            // RT=2 halves that L2 traffic (grid R/2 still fills the chip)
+            if (R % 2 == 0 && R >= 1024)
+                hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc, 2>),
+                                   dim3(R / 2), dim3(BLOCK), 0, s, w, bprep,
+                                   normprep, ss_in, eps, y, xprep_out,
+                                   ss_out, T);
+            else
+                hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc>), grid,
+                                   dim3(BLOCK), 0, s, w, bprep, normprep,
+                                   ss_in, eps, y, xprep_out, ss_out, T);
+        }
         else
             hipLaunchKernelGGL((k_gemm16<WTc, GM_PLAIN, JTc>), grid,
                                dim3(BLOCK), 0, s, w, bprep, normprep, ss_in,
