@@ -126,14 +126,11 @@ class DecodePipeline:
             self._graph_out.append(out)
 
     def _advance_mb(self, m: int) -> None:
-        """Push micro-batch m one token forward through this stage."""
+        """Push micro-batch m one token forward through this stage
+        (blocking comm; the overlapped path lives in run_steps)."""
         if not self.is_first:
             dist.recv(self.x_recv[m], src=self.rank - 1)
-        if self._graphs is not None:
-            self._graphs[m].replay()
-            out = self._graph_out[m]
-        else:
-            out = self._mb_compute(m)
+        out = self._compute(m)
         if not self.is_last:
             dist.send(out, dst=self.rank + 1)
         elif self.world > 1:
@@ -144,10 +141,66 @@ class DecodePipeline:
             # with the send above.
             dist.recv(self.tok[m], src=self.world - 1)
 
+    def _compute(self, m: int):
+        if self._graphs is not None:
+            self._graphs[m].replay()
+            return self._graph_out[m]
+        return self._mb_compute(m)
+
+    def _post_recv(self, m: int):
+        """Post the async receive of micro-batch m's NEXT input: mid/last
+        stages receive activations from the previous stage; rank 0 of a
+        multi-stage pipeline receives the sampled token ids from the last
+        stage."""
+        if not self.is_first:
+            return dist.irecv(self.x_recv[m], src=self.rank - 1)
+        if self.world > 1:
+            return dist.irecv(self.tok[m], src=self.world - 1)
+        return None
+
     def run_steps(self, steps: int) -> None:
-        for _ in range(steps):
-            for m in range(self.cfg.n_mb):
-                self._advance_mb(m)
+        """Overlapped pipeline driver (world > 1): every receive is posted
+        as soon as the buffer is free, so the activation hop of micro-batch
+        m rides xGMI while micro-batch m+1 computes (the reference pipeline
+        is fully serialized — SURVEY §2.3; the north-star design point is
+        comm/compute overlap across pipeline stages)."""
+        n_mb = self.cfg.n_mb
+        if self.world == 1 or steps == 0:
+            for _ in range(steps):
+                for m in range(n_mb):
+                    self._advance_mb(m)
+            return
+        recvs: List = [None] * n_mb
+        sends: List = [None] * n_mb
+        # prime: mid/last stages post receives for every micro-batch;
+        # rank 0 owns the initial tokens, so its first token receive is
+        # posted only after it has sent work downstream.
+        if not self.is_first:
+            for m in range(n_mb):
+                recvs[m] = self._post_recv(m)
+        for s in range(steps):
+            for m in range(n_mb):
+                if recvs[m] is not None:
+                    recvs[m].wait()
+                if sends[m] is not None:
+                    sends[m].wait()  # our out buffer is being re-written
+                out = self._compute(m)
+                dst = 0 if self.is_last else self.rank + 1
+                sends[m] = dist.isend(out, dst=dst)
+                # mid/last stages receive exactly `steps` activation
+                # blocks — do not post one past the end (it would never be
+                # matched); rank 0 must still drain the final token sends.
+                if self.is_first or s + 1 < steps:
+                    recvs[m] = self._post_recv(m)
+                else:
+                    recvs[m] = None
+        for m in range(n_mb):
+            if sends[m] is not None:
+                sends[m].wait()
+            # rank 0: drain the final token receives so the communicator
+            # is quiescent between run_steps calls
+            if recvs[m] is not None:
+                recvs[m].wait()
 
     def current_tokens(self) -> torch.Tensor:
         return torch.stack(self.tok)

@@ -26,20 +26,27 @@ def test_partition_layers():
     assert partition_layers(4, 2) == [(0, 2), (2, 2)]
 
 
-def _single_reference_tokens():
+SEEDS = [[5, 9], [11, 3]]  # starting tokens per micro-batch
+
+
+def _seed_tokens(pipe, n_mb):
+    for m in range(n_mb):
+        pipe.tok[m] = torch.tensor(SEEDS[m], dtype=torch.int32)
+
+
+def _single_reference_tokens(n_mb=1):
     f = synthetic.build_model("tiny", seed=0)
     ex = slicer.make_extra_layers(f)
-    eng = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=MBS)
+    eng = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=MBS * n_mb)
     eng.attach_extra(ex)
-    cfg = PipelineConfig(mbs=MBS, n_mb=1, device="cpu")
+    cfg = PipelineConfig(mbs=MBS, n_mb=n_mb, device="cpu")
     pipe = DecodePipeline(eng, cfg, rank=0, world=1)
-    # deterministic starting tokens
-    pipe.tok[0] = torch.tensor([5, 9], dtype=torch.int32)
+    _seed_tokens(pipe, n_mb)
     pipe.run_steps(STEPS)
-    return pipe.current_tokens()[0].tolist()
+    return pipe.current_tokens().tolist()
 
 
-def _rank_main(rank, world, port, q):
+def _rank_main(rank, world, port, n_mb, q):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -49,29 +56,39 @@ def _rank_main(rank, world, port, q):
     parts = partition_layers(f.hparams.n_layer, world)
     first, count = parts[rank]
     sl = slicer.make_slice(f, first, first + count - 1)
-    eng = TorchSliceEngine.from_ggml(sl, n_ctx=32, max_batch=MBS)
+    eng = TorchSliceEngine.from_ggml(sl, n_ctx=32, max_batch=MBS * n_mb)
     eng.attach_extra(ex)
-    cfg = PipelineConfig(mbs=MBS, n_mb=1, device="cpu")
+    cfg = PipelineConfig(mbs=MBS, n_mb=n_mb, device="cpu")
     pipe = DecodePipeline(eng, cfg, rank=rank, world=world)
-    pipe.tok[0] = torch.tensor([5, 9], dtype=torch.int32)
+    _seed_tokens(pipe, n_mb)
     pipe.run_steps(STEPS)
     if rank == 0:
-        q.put(pipe.current_tokens()[0].tolist())
+        q.put(pipe.current_tokens().tolist())
     dist.destroy_process_group()
 
 
-@pytest.mark.timeout(120)
-def test_two_stage_pipeline_matches_single():
-    ref = _single_reference_tokens()
+def _run_cluster(world, port, n_mb):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29531
-    procs = [ctx.Process(target=_rank_main, args=(r, 2, port, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_rank_main, args=(r, world, port, n_mb, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     got = q.get(timeout=100)
     for p in procs:
         p.join(timeout=30)
         assert p.exitcode == 0
-    assert got == ref
+    return got
+
+
+@pytest.mark.timeout(120)
+def test_two_stage_pipeline_matches_single():
+    assert _run_cluster(2, 29531, 1) == _single_reference_tokens(1)
+
+
+@pytest.mark.timeout(120)
+def test_two_stage_overlapped_micro_batches_match():
+    """n_mb=2 exercises the overlapped isend/irecv driver: stage 0
+    computes micro-batch 1 while micro-batch 0's activations are in
+    flight."""
+    assert _run_cluster(2, 29532, 2) == _single_reference_tokens(2)
