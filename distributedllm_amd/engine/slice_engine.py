@@ -28,13 +28,6 @@ from ..formats import ggml, q4
 from ..models.llama import RMS_EPS, ROPE_BASE, rms_norm, rope_interleaved
 
 
-def _f32_to_bf16_u16(a: np.ndarray) -> np.ndarray:
-    """Round-to-nearest-even f32 -> bf16 bit pattern (u16)."""
-    u = np.ascontiguousarray(a, dtype=np.float32).view(np.uint32)
-    rounded = u + 0x7FFF + ((u >> 16) & 1)
-    return (rounded >> 16).astype(np.uint16)
-
-
 def _nibbles(t: ggml.GGMLTensor) -> np.ndarray:
     """q4 tensor -> raw nibble values [rows, nb, 32] (u8, order = weights)."""
     rows, cols = t.shape_rows_cols
@@ -50,11 +43,14 @@ def _nibbles(t: ggml.GGMLTensor) -> np.ndarray:
 def repack_mfma(t: ggml.GGMLTensor, device: str):
     """On-disk tensor -> (data, scales, wtype) in the MFMA tile layout.
 
-    q4_0/q4_1: data u32[R][nb][4 words][16 rows] where word ws of a block
-    packs its 8 weights at bit positions (j%2)*16 + (j//2)*4 — the exact
-    arrangement the kernel's three OR/AND/SHR turn into a bf16x8 A-fragment
-    (kernels.hip a_frag_q4). scales f32[R][nb][16] (q4_1: (d,m) pairs).
-    f16: weights converted to bf16 tiles [R][cols/8][16 rows][8].
+    q4_0/q4_1: data u32[R][nbp/4][4 ws][16 rows][4 kb] — word ws of a
+    block packs its 8 weights at bit positions (j%2)*16 + (j//2)*4 (the
+    arrangement a_frag_q4 unpacks into an f16x8 A-fragment), with 4
+    consecutive K-blocks grouped per lane for dwordx4 loads and nb padded
+    to a multiple of 4 (zero-scaled pad blocks). scales are (alpha, beta)
+    f16 pairs in the matching grouped layout. f16 weights stay f16, tiled
+    [R][cols/8][16 rows][8]. Every tensor carries one prefetch batch of
+    zero tail slack (kernels overread one batch).
     """
     rows, cols = t.shape_rows_cols
     assert rows % 16 == 0 and cols % 32 == 0, (t.name, rows, cols)

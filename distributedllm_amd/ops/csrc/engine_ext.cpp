@@ -11,10 +11,12 @@
 //     device tensors, so the decode step is hipGraph-capturable and the KV
 //     "clear_context" is a host-side position reset.
 //
-// Decode layer = 5 fused kernels (QKV+RoPE+KV-append, attention, WO+res+
-// sumsq, FFN-gate+SwiGLU, W2+res+sumsq) — RMSNorm rides the sumsq
-// side-channel into the consumers' B-fragment builds. f32-weight models
-// (test configs) take a legacy scalar path with explicit RMSNorm kernels.
+// Decode layer = 7-9 kernels depending on model size: QKV (fused, or
+// RT=2 slab split-K + rope/cache finish), attention, WO (slab split-K),
+// fused reduce+residual+sumsq+xprep, FFN gate (fused SwiGLU), W2 (slab) +
+// reduce — RMSNorm rides the sumsq side-channel into the consumers'
+// B-fragment builds. f32-weight models (test configs) take a legacy
+// scalar path with explicit RMSNorm kernels.
 
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
@@ -50,7 +52,7 @@ struct DevMat2 {  // MFMA-tiled matrix
 
 struct Layer {
     torch::Tensor attn_norm, ffn_norm;           // f32 [E] (legacy path)
-    torch::Tensor attn_normprep, ffn_normprep;   // bf16 [E] (MFMA path)
+    torch::Tensor attn_normprep, ffn_normprep;   // f16 [E+pad] (MFMA path)
     // legacy (f32 models)
     DevMat wq, wk, wv, wo, w1, w2, w3;
     // MFMA path

@@ -3,22 +3,29 @@
 // Replaces the vendored ggml CPU kernels of the reference
 // (/root/reference/distllm/tensor_processor.cpp:474-809 builds the graph this
 // file executes natively; SURVEY.md §2.5 maps each op K1-K15 to a kernel
-// here). Decode (N small) is HBM-bandwidth-bound on the q4_0 weight stream,
-// so the GEMV family streams repacked (SoA) q4_0 blocks with coalesced
-// 16-byte nibble loads and fuses dequant + per-block f16 scale in-register;
-// RoPE + KV-cache append are fused into the QKV projection epilogue;
-// attention is a streaming online-softmax kernel (no [P+N,N,H] score tensor).
+// here). Design notes with the measurements that drove each choice:
+// docs/kernels.md.
 //
-// Weight layout in HBM (repacked at load, not the on-disk AoS layout):
-//   q4_0 : scales f16[rows][nb]  +  qs u8[rows][nb*16]   (nb = cols/32)
-//   q4_1 : scales f16[rows][nb*2] (d,m pairs) + qs as above
-//   f16  : data [rows][cols]
-//   f32  : data [rows][cols]
+// Production decode path (q4_0/q4_1/f16 weights): the MFMA dequant-GEMM
+// family around `wave_tile_kloop` — `mfma_f32_16x16x32_f16` tiles with the
+// 4-bit dequantization done in exact packed-f16 VALU (0x6400|n == 1024+n),
+// JT = 1/2/4 column tiles (up to 64 tokens) amortizing the HBM weight
+// stream, optional RT row tiles amortizing the L2 B-panel, software-
+// pipelined non-temporal dwordx4 weight loads, and RMSNorm fused into the
+// consumers' B-fragment builds via a sumsq side channel. Split-K variants
+// write plain f32 slabs combined by fused reduce/finish passes.
 //
-// Wavefront = 64 (CDNA); blocks of 256 threads = 4 waves; one wave per
-// output row in the GEMV family, lanes striding the row's q4 blocks so that
-// scale loads (2 B x 64 lanes) and nibble loads (16 B x 64 lanes) are fully
-// coalesced 128 B / 1 KiB wave transactions.
+// MFMA-path weight layout in HBM (repacked at load; python side in
+// engine/slice_engine.py::repack_mfma):
+//   q4   : data   u32[R][nbp/4][4 ws][16 i][4 kb]  (one dwordx4 per lane
+//          covers 4 consecutive K-blocks; nbp = nb padded to 4)
+//          scales f16 (alpha, beta)[R][nbp/4][16 i][4 kb]
+//   f16  : data   f16[R][cols/8][16 i][8]
+// plus one prefetch batch of tail slack on every allocation.
+//
+// A legacy scalar GEMV family (wave_row_dot) serves W_F32 test models.
+//
+// Wavefront = 64 (CDNA); blocks are 256 threads = 4 waves.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
@@ -650,22 +657,6 @@ union ABFrag {
     uint32_t u[4];
     f16x8 v;
 };
-
-__device__ __forceinline__ float bflo(uint32_t w) {
-    union { uint32_t u; float f; } c;
-    c.u = w << 16;
-    return c.f;
-}
-__device__ __forceinline__ float bfhi(uint32_t w) {
-    union { uint32_t u; float f; } c;
-    c.u = w & 0xFFFF0000u;
-    return c.f;
-}
-__device__ __forceinline__ uint32_t pack_bf16(float lo, float hi) {
-    union { __hip_bfloat162 b; uint32_t u; } c;
-    c.b = __hip_bfloat162(__float2bfloat16(lo), __float2bfloat16(hi));
-    return c.u;
-}
 
 // packed-f16 helpers for the MFMA side-channels / dequant
 __device__ __forceinline__ uint32_t pack_f16(float lo, float hi) {
