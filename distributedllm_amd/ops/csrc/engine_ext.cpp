@@ -321,8 +321,8 @@ class SliceEngine {
                 launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
                               xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
             }
-            // ffn keeps the fused RT=1 kernel: measured par with the
-            // slab+finish variant (its LDS/VGPR cost offsets the B saving)
+            // ffn keeps the fused RT=1 kernel: the slab+finish variant
+            // measured slower at both 512- and 1024-block split targets
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
                          ssf + li * kMaxTokens, eps_, gprep, T,
                          /*slab=*/nullptr);
