@@ -86,6 +86,10 @@ class NodeState:
     def __init__(self, uploads_dir: str, device: Optional[str] = None,
                  n_ctx: int = 2048):
         self.lock = threading.Lock()
+        # serializes forward passes: concurrent propagate_forward calls
+        # would interleave KV-cache writes of the shared engine (the
+        # reference leaves this unsynchronized — SURVEY §5.2)
+        self.fwd_lock = threading.Lock()
         self.uploads = UploadManager(uploads_dir)
         self.slice = None  # DummySlice | EngineSlice
         self.device = device
@@ -178,7 +182,8 @@ class NodeState:
                                    description="load a slice first")
         x = msg.values.reshape(msg.axis0, msg.axis1)
         try:
-            y = s.forward(x, msg.start_pos)
+            with self.fwd_lock:
+                y = s.forward(x, msg.start_pos)
         except Exception as e:  # noqa: BLE001
             return P.ResponseError(operation="propagate_forward",
                                    error="neural_computation_error",
