@@ -69,7 +69,10 @@ class DecodePipeline:
                     for _ in range(M)]
         self.seq = [torch.arange(m * mbs, (m + 1) * mbs, dtype=torch.int32,
                                  device=dev) for m in range(M)]
-        self.x_recv = [torch.empty(mbs, E, dtype=torch.float32, device=dev)
+        # zero-init: mid-stage graph-capture warmup runs before any real
+        # activations arrive; empty-buffer garbage could seed NaNs into
+        # the KV cache rows written during warmup
+        self.x_recv = [torch.zeros(mbs, E, dtype=torch.float32, device=dev)
                        for _ in range(M)]
         self.tok = [torch.randint(3, engine.hp.n_vocab, (mbs,),
                                   dtype=torch.int32, device=dev)
