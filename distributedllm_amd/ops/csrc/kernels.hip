@@ -1233,7 +1233,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
 
 // ------------------------------------------------------------- k_qkv16
 // QKV projections on MFMA + fused input RMSNorm + RoPE + KV append.
-template <int WT, int JT>
+template <int WT, int JT, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_qkv16(
     WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
@@ -1241,21 +1241,23 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     __half* __restrict__ k_cache, __half* __restrict__ v_cache,
     const int* __restrict__ pos, const int* __restrict__ seq,
     const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T) {
-    const int tiles_per_mat = E >> 4;
+    const int tiles_per_mat = (E >> 4) / RT;
     const int mat = blockIdx.x / tiles_per_mat;
     const int tile = blockIdx.x % tiles_per_mat;
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][1][JT][4];
+    float acc[RT][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 1, JT>(ws, tile, xprep, normprep, ss_in, eps,
-                                     acc, 0, nbe);
-    __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(acc[0][0], lds);
+    wave_tile_kloop<WT, true, 1, JT, RT>(ws, tile, xprep, normprep, ss_in,
+                                         eps, acc, 0, nbe);
+    __shared__ float lds[3 * 64 * 4 * RT * JT];
+    combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = tile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+    const int r0 = (tile * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j2 = jt * 16 + j;
@@ -1265,7 +1267,7 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
             __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
-                dst[jj] = __float2half(acc[0][0][jt][jj]);
+                dst[jj] = __float2half(acc[rt][0][jt][jj]);
             continue;
         }
         // q/k: RoPE on in-lane pairs (rows r0+2q2, r0+2q2+1)
@@ -1276,8 +1278,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
             const float theta = (float)p * inv_freq[d >> 1];
             float sn, cs;
             __sincosf(theta, &sn, &cs);
-            const float x0 = acc[0][0][jt][2 * q2];
-            const float x1 = acc[0][0][jt][2 * q2 + 1];
+            const float x0 = acc[rt][0][jt][2 * q2];
+            const float x1 = acc[rt][0][jt][2 * q2 + 1];
             const float o0 = x0 * cs - x1 * sn;
             const float o1 = x0 * sn + x1 * cs;
             if (mat == 0) {
@@ -1290,12 +1292,13 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
             }
         }
     }
+    }
 }
 
 // ------------------------------------------------------------- k_ffn16
 // w1 + w3 against the same B panel + fused input RMSNorm + SwiGLU; emits
 // the gate product straight into gprep (f16 B-layout over F).
-template <int WT, int JT>
+template <int WT, int JT, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_ffn16(
     WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
@@ -1303,33 +1306,36 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16(
     unsigned short* __restrict__ gprep, int T) {
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][2][JT][4];
+    float acc[RT][2][JT][4];
     const WMat2* ws[2] = {&w1, &w3};
     const int nbf = (WT == W_F16) ? (w1.cols >> 5)
                                   : (((w1.cols >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 2, JT>(ws, blockIdx.x, xprep, normprep, ss_in,
-                                     eps, acc, 0, nbf);
-    __shared__ float lds[3 * 64 * 4 * 2 * JT];
-    combine_acc<2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    wave_tile_kloop<WT, true, 2, JT, RT>(ws, blockIdx.x, xprep, normprep,
+                                         ss_in, eps, acc, 0, nbf);
+    __shared__ float lds[3 * 64 * 4 * RT * 2 * JT];
+    combine_acc<RT * 2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = blockIdx.x * 16 + (lane >> 4) * 4;
 #pragma unroll
-    for (int jt = 0; jt < JT; ++jt) {
-        const int j2 = jt * 16 + j;
-        if (j2 >= T) continue;
-        float g[4];
+    for (int rt = 0; rt < RT; ++rt) {
+        const int r0 = (blockIdx.x * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
-        for (int jj = 0; jj < 4; ++jj) {
-            const float v1 = acc[0][0][jt][jj];
-            const float silu = v1 / (1.0f + __expf(-v1));
-            g[jj] = silu * acc[0][1][jt][jj];
+        for (int jt = 0; jt < JT; ++jt) {
+            const int j2 = jt * 16 + j;
+            if (j2 >= T) continue;
+            float g[4];
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                const float v1 = acc[rt][0][jt][jj];
+                const float silu = v1 / (1.0f + __expf(-v1));
+                g[jj] = silu * acc[rt][1][jt][jj];
+            }
+            uint2 o;
+            o.x = pack_f16(g[0], g[1]);
+            o.y = pack_f16(g[2], g[3]);
+            *reinterpret_cast<uint2*>(
+                gprep + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
+                (r0 & 7)) = o;
         }
-        uint2 o;
-        o.x = pack_f16(g[0], g[1]);
-        o.y = pack_f16(g[2], g[3]);
-        *reinterpret_cast<uint2*>(
-            gprep + (((size_t)(r0 >> 3) * JT + jt) * 16 + j) * 8 +
-            (r0 & 7)) = o;
     }
 }
 
@@ -1777,6 +1783,16 @@ void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                            T);
         return;
     }
+    if (((E >> 4) % 2) == 0 && tiles3 / 2 >= 512) {
+        // big models: RT=2 fused — halves the B-panel re-read while the
+        // halved grid still fills the chip
+        const dim3 grid(tiles3 / 2);
+        DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+            (k_qkv16<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk, wv,
+            xprep, normprep, ss_in, eps, q_buf, k_cache_layer,
+            v_cache_layer, pos, seq, inv_freq, E, D, n_ctx, T)));
+        return;
+    }
     const dim3 grid(tiles3);
     DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_qkv16<WTc, JTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
@@ -1800,6 +1816,13 @@ void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
         hipLaunchKernelGGL(k_ffn_finish,
                            dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK),
                            0, s, slab, ks, gprep, w1.rows, T, pick_jt(T));
+        return;
+    }
+    if ((tilesF % 2) == 0 && tilesF / 2 >= 512) {
+        const dim3 grid(tilesF / 2);
+        DISPATCH_WT2(w1.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
+            (k_ffn16<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, w1, w3, xprep,
+            normprep, ss_in, eps, gprep, T)));
         return;
     }
     const dim3 grid(tilesF);
