@@ -15,6 +15,23 @@ import threading
 from .base import Command
 
 
+def _random_dataset_prompt(path: str, seed=None) -> str:
+    """Random prompt from a local HF dataset directory (the reference
+    samples from a hub dataset, perplexity.py:35-51; this environment has
+    no network, so the dataset must already be on disk)."""
+    import random
+    from datasets import load_from_disk
+    ds = load_from_disk(path)
+    if hasattr(ds, "keys") and not hasattr(ds, "features"):  # DatasetDict
+        ds = ds[sorted(ds.keys())[0]]
+    rng = random.Random(seed)
+    row = ds[rng.randrange(len(ds))]
+    for key in ("text", "content", "sentence", "document"):
+        if key in row and isinstance(row[key], str) and row[key].strip():
+            return row[key]
+    raise ValueError(f"no text-like column found in dataset at {path}")
+
+
 def _progress(label: str):
     def cb(sent: int, total: int) -> None:
         pct = 100.0 * sent / max(total, 1)
@@ -195,17 +212,26 @@ class PerplexityCommand(Command):
         p.add_argument("--prompt", default=None)
         p.add_argument("--file", default=None,
                        help="read the text from a file instead")
+        p.add_argument("--dataset", default=None,
+                       help="local HF dataset directory (datasets.load_"
+                            "from_disk); a random text row is sampled, "
+                            "like the reference's --dataset")
+        p.add_argument("--seed", type=int, default=None)
         p.add_argument("--root", default=".")
 
     def __call__(self, args) -> int:
-        if (args.prompt is None) == (args.file is None):
-            print("provide exactly one of --prompt / --file",
+        sources = [x for x in (args.prompt, args.file, args.dataset)
+                   if x is not None]
+        if len(sources) != 1:
+            print("provide exactly one of --prompt / --file / --dataset",
                   file=sys.stderr)
             return 2
         text = args.prompt
         if args.file:
             with open(args.file) as f:
                 text = f.read()
+        if args.dataset:
+            text = _random_dataset_prompt(args.dataset, args.seed)
         from ..cluster.llm_client import get_llm
         llm = get_llm(args.config, root=args.root)
         ppl = llm.perplexity(text)

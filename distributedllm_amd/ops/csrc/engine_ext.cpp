@@ -291,8 +291,8 @@ class SliceEngine {
         float* ssa = ss_attn_.data_ptr<float>();
         float* ssf = ss_ffn_.data_ptr<float>();
         // zero the atomic sumsq slots, then stage x into the side channel
-        hipMemsetAsync(ssa, 0, sizeof(float) * (L_ + 1) * kMaxTokens, s);
-        hipMemsetAsync(ssf, 0, sizeof(float) * L_ * kMaxTokens, s);
+        (void)hipMemsetAsync(ssa, 0, sizeof(float) * (L_ + 1) * kMaxTokens, s);
+        (void)hipMemsetAsync(ssf, 0, sizeof(float) * L_ * kMaxTokens, s);
         launch_prep_x(s, xp, xprep, ssa, E_, T);
         for (int li = 0; li < L_; ++li) {
             Layer& l = layers_[li];
@@ -369,7 +369,7 @@ class SliceEngine {
             {T, V_},
             torch::TensorOptions().device(torch::kCUDA).dtype(torch::kFloat32));
         if (out_mfma_) {
-            hipMemsetAsync(ss_tmp_.data_ptr(), 0, sizeof(float) * T, s);
+            (void)hipMemsetAsync(ss_tmp_.data_ptr(), 0, sizeof(float) * T, s);
             launch_prep_x(s, xin.data_ptr<float>(), u16p(xprep_),
                           ss_tmp_.data_ptr<float>(), E_, T);
             launch_gemm16(s, mout_.w, u16p(xprep_), u16p(final_normprep_),

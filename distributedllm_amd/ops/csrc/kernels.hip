@@ -1879,7 +1879,7 @@ void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
 
 void launch_argmax(hipStream_t s, const float* logits,
                    unsigned long long* keys, int* out, int T, int V) {
-    hipMemsetAsync(keys, 0, sizeof(unsigned long long) * T, s);
+    (void)hipMemsetAsync(keys, 0, sizeof(unsigned long long) * T, s);
     const int npart = min(32, (V + BLOCK - 1) / BLOCK);
     hipLaunchKernelGGL(k_argmax_part, dim3(T, npart), dim3(BLOCK), 0, s,
                        logits, keys, V);

@@ -234,3 +234,27 @@ def test_control_center(node, tmp_path):
     # unreachable node
     cc2 = ControlCenter({"127.0.0.1:1": [0, 0]})
     assert not cc2.get_status()["127.0.0.1:1"].connectivity
+
+
+def test_perplexity_dataset_source(node, tmp_path, capsys):
+    """--dataset samples a random prompt from a local HF dataset
+    (reference perplexity.py:35-51 capability, offline)."""
+    import datasets
+    d = datasets.Dataset.from_dict(
+        {"text": ["hello world of words and more words", "another text"]})
+    ds_path = tmp_path / "ds"
+    d.save_to_disk(str(ds_path))
+
+    addr = f"127.0.0.1:{node.port}"
+    root = tmp_path / "root"; root.mkdir()
+    cfg = {"model_id": "tiny_ds", "location": "synthetic:tiny",
+           "nodes_map": {addr: [0, PRESETS["tiny"].n_layer - 1]},
+           "quantization": "f16", "metadata": {"name": "tinyds"}}
+    cfg_path = tmp_path / "cfg.json"
+    cfg_path.write_text(json.dumps(cfg))
+    assert execute_command(["provision", str(cfg_path),
+                            "--root", str(root)]) == 0
+    assert execute_command(["perplexity", str(cfg_path),
+                            "--dataset", str(ds_path), "--seed", "0",
+                            "--root", str(root)]) == 0
+    assert "perplexity:" in capsys.readouterr().out
