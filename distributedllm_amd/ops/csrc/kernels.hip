@@ -767,10 +767,13 @@ __device__ __forceinline__ void wave_tile_kloop(
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
-    // RT>1 brings enough independent accumulators to cover the MFMA
-    // dependent latency without the dual c0/c1 set (which would double
-    // AGPR pressure); RT==1 keeps the alternating pair.
-    constexpr int NP = (RT > 1) ? 1 : 2;
+    // With >=4 independent MFMAs per K-block (JT/NM/RT product) the
+    // accumulator reuse distance already covers the MFMA dependent
+    // latency — and the runtime `parity ? c1 : c0` reference select is
+    // poison: the compiler materialized it as per-MFMA cndmask + AGPR
+    // read/write round-trips (338 of 445 inner-loop instructions in the
+    // JT=4 ffn kernel). Only the low-parallelism shapes keep the pair.
+    constexpr int NP = (NM * JT * RT >= 4) ? 1 : 2;
     f32x4 c0[RT][NM][JT], c1[NP == 2 ? RT : 1][NM][JT];
 #pragma unroll
     for (int rt = 0; rt < RT; ++rt)
