@@ -322,7 +322,7 @@ class SliceEngine {
                               xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
             }
             // ffn keeps the fused RT=1 kernel: the slab+finish variant
-            // measured slower at both 512- and 1024-block split targets
+            // measured slower at every split/occupancy combination tried
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
                          ssf + li * kMaxTokens, eps_, gprep, T,
                          /*slab=*/nullptr);
