@@ -286,4 +286,9 @@ def receive_message(sock: socket.socket) -> Message:
     payload = _recv_exact(sock, length)
     if hashlib.sha256(payload).digest() != digest:
         raise IntegrityError("payload sha256 mismatch")
-    return Message.decode(payload)
+    try:
+        return Message.decode(payload)
+    except ProtocolError:
+        raise
+    except Exception as e:  # malformed body -> typed error, never a crash
+        raise ProtocolError(f"malformed message: {type(e).__name__}: {e}")

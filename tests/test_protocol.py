@@ -101,3 +101,29 @@ def test_float_array_is_raw_le():
 def test_unknown_message_rejected():
     with pytest.raises(P.ProtocolError):
         P.Message.decode(b"\x03xyz\x00\x00")
+
+
+def test_garbage_bytes_rejected():
+    """Corrupt/hostile frames raise typed protocol errors, never crash:
+    bad magic lengths, truncation, checksum mismatch, unknown names."""
+    import random
+    rng = random.Random(0)
+    # oversized length field
+    sock = FakeSocket(chunk=64)
+    sock.buf += struct.pack("<I", P.MAX_PAYLOAD + 1) + b"\x00" * 32
+    with pytest.raises(P.ProtocolError):
+        P.receive_message(sock)
+    # random garbage payloads with VALID framing -> decode errors, typed
+    for _ in range(50):
+        payload = bytes(rng.randrange(256) for _ in range(rng.randrange(1, 60)))
+        sock = FakeSocket(chunk=7)
+        digest = hashlib.sha256(payload).digest()
+        sock.buf += struct.pack("<I", len(payload)) + digest + payload
+        with pytest.raises(P.ProtocolError):
+            P.receive_message(sock)
+    # checksum mismatch
+    msg = P.RequestStatus().encode()
+    sock = FakeSocket(chunk=16)
+    sock.buf += struct.pack("<I", len(msg)) + b"\x00" * 32 + msg
+    with pytest.raises(P.IntegrityError):
+        P.receive_message(sock)
