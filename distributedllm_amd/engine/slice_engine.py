@@ -284,11 +284,13 @@ class HIPSliceEngine:
         self.has_extra = True
 
     def forward(self, x: torch.Tensor, pos: torch.Tensor,
-                seq: torch.Tensor) -> torch.Tensor:
+                seq: torch.Tensor, decode: bool = False) -> torch.Tensor:
+        """decode=True asserts every token is a distinct sequence
+        (batched decode) — enables the qkv-slab + fused-attention path."""
         T = x.shape[0]
         mt = self._eng.max_tokens
         if T <= mt:
-            return self._eng.forward(x, pos, seq)
+            return self._eng.forward(x, pos, seq, decode=decode)
         # token-tile larger inputs (prefill); KV order is preserved because
         # tile i's cache rows are written before tile i+1 attends.
         outs = []
@@ -355,7 +357,7 @@ class TorchSliceEngine:
         self.has_extra = True
 
     def forward(self, x: torch.Tensor, pos: torch.Tensor,
-                seq: torch.Tensor) -> torch.Tensor:
+                seq: torch.Tensor, decode: bool = False) -> torch.Tensor:
         hp = self.hp
         T, E = x.shape
         h, d = hp.n_head, hp.head_dim

@@ -257,8 +257,10 @@ class SliceEngine {
     }
 
     // x: [T, E] f32 (modified in place and returned), pos/seq: [T] i32.
+    // decode=true asserts every token is a distinct sequence (batched
+    // decode), enabling the qkv-slab + fused-attention path.
     torch::Tensor forward(torch::Tensor x, torch::Tensor pos,
-                          torch::Tensor seq) {
+                          torch::Tensor seq, bool decode = false) {
         check_f32(x, "x");
         check_i32(pos, "pos");
         check_i32(seq, "seq");
@@ -298,12 +300,16 @@ class SliceEngine {
             Layer& l = layers_[li];
             __half* kc = kbase + (size_t)li * layer_stride;
             __half* vc = vbase + (size_t)li * layer_stride;
-            launch_qkv16(s, l.mq.w, l.mk.w, l.mv.w, xprep,
-                         u16p(l.attn_normprep), ssa + li * kMaxTokens, eps_,
-                         qb, kc, vc, pp, sp, ifr, E_, D_, ctx_, T,
-                         slab_.data_ptr<float>());
-            launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_, D_,
-                             ctx_);
+            const int slab_used = launch_qkv16(
+                s, l.mq.w, l.mk.w, l.mv.w, xprep, u16p(l.attn_normprep),
+                ssa + li * kMaxTokens, eps_, qb, kc, vc, pp, sp, ifr, E_,
+                D_, ctx_, T, slab_.data_ptr<float>(),
+                /*skip_finish=*/decode ? 1 : 0);
+            const bool fuse = decode && slab_used;
+            launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_,
+                             D_, ctx_,
+                             fuse ? slab_.data_ptr<float>() : nullptr,
+                             fuse ? qkv16_ks(E_) : 0, ifr);
             // wo/w2 tile count (E/16) alone underfills 256 CUs — split K
             // across grid.y into plain-stored slabs, then one fused
             // reduce+residual+sumsq+xprep pass per matrix (no atomics;
@@ -422,7 +428,7 @@ class SliceEngine {
             launch_qkv_rope_append(s, l.wq.w, l.wk.w, l.wv.w, xn, qb, kc, vc,
                                    pp, sp, ifr, E_, D_, ctx_, T);
             launch_attention(s, qb, kc, vc, ab, nullptr, pp, sp, T, H_, E_,
-                             D_, ctx_);
+                             D_, ctx_, nullptr, 0, ifr);
             launch_gemv(s, l.wo.w, ab, /*res=*/xp, xp, T);
             launch_rmsnorm(s, xp, l.ffn_norm.data_ptr<float>(), xn, T, E_,
                            eps_);
@@ -459,7 +465,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
              py::arg("eps"), py::arg("rope_base"))
         .def("set_layer", &SliceEngine::set_layer)
         .def("set_extra", &SliceEngine::set_extra)
-        .def("forward", &SliceEngine::forward)
+        .def("forward", &SliceEngine::forward, py::arg("x"),
+             py::arg("pos"), py::arg("seq"), py::arg("decode") = false)
         .def("embed", &SliceEngine::embed)
         .def("logits", &SliceEngine::logits, py::arg("x"),
              py::arg("all_logits") = false)

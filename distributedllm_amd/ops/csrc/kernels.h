@@ -50,14 +50,19 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
                    float eps, float* y, unsigned short* xprep_out,
                    float* ss_out, int T, int mode);
 
-// slab != nullptr enables the RT=2 + split-K slab path for small models
-void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
-                  const WMat2& wv, const unsigned short* xprep,
-                  const unsigned short* normprep, const float* ss_in,
-                  float eps, float* q_buf, __half* k_cache_layer,
-                  __half* v_cache_layer, const int* pos, const int* seq,
-                  const float* inv_freq, int E, int D, int n_ctx, int T,
-                  float* slab);
+// slab != nullptr enables the RT=2 + split-K slab path for small models;
+// returns 1 when the slab path ran. skip_finish=1 leaves the slabs un-
+// combined for a fused-attention consumer (decode only).
+int launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                 const WMat2& wv, const unsigned short* xprep,
+                 const unsigned short* normprep, const float* ss_in,
+                 float eps, float* q_buf, __half* k_cache_layer,
+                 __half* v_cache_layer, const int* pos, const int* seq,
+                 const float* inv_freq, int E, int D, int n_ctx, int T,
+                 float* slab, int skip_finish);
+
+// the qkv slab split factor (shared with the fused-attention consumer)
+int qkv16_ks(int E);
 
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
                   const unsigned short* xprep,
@@ -74,11 +79,14 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
                             const float* inv_freq, int E, int D, int n_ctx,
                             int T);
 
+// qkv_slab != nullptr: decode-fused variant — the attention prologue
+// sums the un-combined qkv slabs, applies RoPE and appends the KV row
+// (valid only when every token is its own sequence)
 void launch_attention(hipStream_t s, const float* q_buf,
-                      const __half* k_cache_layer,
-                      const __half* v_cache_layer, float* out,
-                      unsigned short* out_prep, const int* pos,
-                      const int* seq, int T, int H, int E, int D, int n_ctx);
+                      const __half* k_cache_layer, __half* v_cache_layer,
+                      float* out, unsigned short* out_prep, const int* pos,
+                      const int* seq, int T, int H, int E, int D, int n_ctx,
+                      const float* qkv_slab, int ks, const float* inv_freq);
 
 void launch_gemv(hipStream_t s, const WMat& w, const float* x,
                  const float* res, float* y, int T);

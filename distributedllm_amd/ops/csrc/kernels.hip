@@ -370,11 +370,19 @@ __global__ void k_qkv_rope_append(
 // entries [0, pos[t]] of sequence seq[t] — causality for prefill rows comes
 // from their per-row positions.
 
+// FUSEQKV (decode only — every token its own sequence, so no block needs
+// another token's new KV row): the prologue sums the qkv split-K slabs
+// for this (token, head) slice, applies RoPE, appends the K/V cache row
+// itself and stages q in LDS — replacing the separate k_qkv_finish pass
+// (each such small kernel costs ~5 us at this chain shape).
+template <bool FUSEQKV>
 __global__ void k_attention(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
-    const __half* __restrict__ v_cache, float* __restrict__ out,
+    __half* __restrict__ v_cache, float* __restrict__ out,
     unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
-    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw) {
+    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw,
+    const float* __restrict__ qkv_slab, int ks,
+    const float* __restrict__ inv_freq) {
     const int t = blockIdx.x;
     const int h = blockIdx.y;
     const int J = pos[t] + 1;
@@ -386,9 +394,52 @@ __global__ void k_attention(
     float* lds_p = smem + D;      // [BLOCK]
     float* lds_red = lds_p + BLOCK;  // [NWAVES]
 
-    for (int d = threadIdx.x; d < D; d += BLOCK)
-        lds_q[d] = q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d;
-    __syncthreads();
+    if (FUSEQKV) {
+        // qkv_slab layout: f32[3*(E/16) gtiles][ks][64 tok][16 rows];
+        // matrices q/k/v are consecutive gtile ranges.
+        const int p = J - 1;
+        const int Et = E >> 4;
+        const size_t matstride = (size_t)Et * ks * 64 * 16;
+        __half* kv_k = const_cast<__half*>(k_cache) + base + (size_t)p * E;
+        __half* kv_v = v_cache + base + (size_t)p * E;
+        for (int d = threadIdx.x; d < D; d += BLOCK) {
+            const int e = h * D + d;
+            float q = 0.f, k = 0.f, v = 0.f;
+            for (int c = 0; c < ks; ++c) {
+                const size_t off =
+                    (((size_t)(e >> 4) * ks + c) * 64 + t) * 16 + (e & 15);
+                q += qkv_slab[off];
+                k += qkv_slab[matstride + off];
+                v += qkv_slab[2 * matstride + off];
+            }
+            lds_q[d] = q;
+            lds_p[d] = k;  // staged for the pair rotation below
+            kv_v[d] = __float2half(v);
+        }
+        __syncthreads();
+        // RoPE on (even, odd) pairs; h*D is even so e % D == d
+        for (int d2 = threadIdx.x; d2 < D / 2; d2 += BLOCK) {
+            const int d0 = 2 * d2;
+            const float theta = (float)p * inv_freq[d2];
+            float sn, cs;
+            __sincosf(theta, &sn, &cs);
+            const float q0 = lds_q[d0], q1 = lds_q[d0 + 1];
+            const float k0 = lds_p[d0], k1 = lds_p[d0 + 1];
+            lds_q[d0] = (q0 * cs - q1 * sn) * inv_sqrt_d;
+            lds_q[d0 + 1] = (q0 * sn + q1 * cs) * inv_sqrt_d;
+            kv_k[d0] = __float2half(k0 * cs - k1 * sn);
+            kv_k[d0 + 1] = __float2half(k0 * sn + k1 * cs);
+        }
+        // same-block visibility of the new K/V row: stores above are this
+        // CU's own L1 write-through; __syncthreads orders them before the
+        // scan loop's loads (cross-CU coherence is not needed — decode
+        // tokens are distinct sequences, no other block reads this row)
+        __syncthreads();
+    } else {
+        for (int d = threadIdx.x; d < D; d += BLOCK)
+            lds_q[d] = q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d;
+        __syncthreads();
+    }
 
     float m = -INFINITY;  // running max (block-uniform)
     float l = 0.0f;       // running denom (block-uniform)
@@ -1639,15 +1690,23 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
 }
 
 void launch_attention(hipStream_t s, const float* q_buf,
-                      const __half* k_cache_layer,
-                      const __half* v_cache_layer, float* out,
-                      unsigned short* out_prep, const int* pos,
-                      const int* seq, int T, int H, int E, int D, int n_ctx) {
+                      const __half* k_cache_layer, __half* v_cache_layer,
+                      float* out, unsigned short* out_prep, const int* pos,
+                      const int* seq, int T, int H, int E, int D, int n_ctx,
+                      const float* qkv_slab, int ks,
+                      const float* inv_freq) {
     const dim3 grid(T, H);
     const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
-    hipLaunchKernelGGL(k_attention, grid, dim3(BLOCK), lds, s, q_buf,
+    if (qkv_slab != nullptr) {
+        hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
+                           q_buf, k_cache_layer, v_cache_layer, out,
+                           out_prep, pos, seq, E, D, n_ctx, pick_jt(T),
+                           qkv_slab, ks, inv_freq);
+        return;
+    }
+    hipLaunchKernelGGL(k_attention<false>, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos, seq,
-                       E, D, n_ctx, pick_jt(T));
+                       E, D, n_ctx, pick_jt(T), qkv_slab, ks, inv_freq);
 }
 
 // ------------------------------------------------- MFMA-path launchers
@@ -1700,6 +1759,13 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
             break;                               \
         }                                        \
     }
+
+int qkv16_ks(int E) {
+    const int tiles3 = 3 * (E >> 4);
+    int ks = 1;
+    while ((tiles3 / 2) * ks < 512 && ks < 8) ks <<= 1;
+    return ks;
+}
 
 int gemm16_ks(int rows) {
     const int R = rows / 16;
@@ -1763,29 +1829,30 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
     }));
 }
 
-void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
-                  const WMat2& wv, const unsigned short* xprep,
-                  const unsigned short* normprep, const float* ss_in,
-                  float eps, float* q_buf, __half* k_cache_layer,
-                  __half* v_cache_layer, const int* pos, const int* seq,
-                  const float* inv_freq, int E, int D, int n_ctx, int T,
-                  float* slab) {
+int launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                 const WMat2& wv, const unsigned short* xprep,
+                 const unsigned short* normprep, const float* ss_in,
+                 float eps, float* q_buf, __half* k_cache_layer,
+                 __half* v_cache_layer, const int* pos, const int* seq,
+                 const float* inv_freq, int E, int D, int n_ctx, int T,
+                 float* slab, int skip_finish) {
     const int tiles3 = 3 * (E >> 4);
     // slab split-K + RT=2 path when the fused grid underfills the chip
     if (slab != nullptr && tiles3 < 1024 && ((E >> 4) % 2) == 0) {
-        int ks = 1;
-        while ((tiles3 / 2) * ks < 512 && ks < 8) ks <<= 1;
+        const int ks = qkv16_ks(E);
         const dim3 grid(tiles3 / 2, ks);
         DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
             (k_qkv16_slab<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk,
             wv, xprep, normprep, ss_in, eps, slab, E, T)));
-        const int total = (3 * E / 8) * T;
-        hipLaunchKernelGGL(k_qkv_finish,
-                           dim3((total + BLOCK - 1) / BLOCK), dim3(BLOCK),
-                           0, s, slab, ks, q_buf, k_cache_layer,
-                           v_cache_layer, pos, seq, inv_freq, E, D, n_ctx,
-                           T);
-        return;
+        if (!skip_finish) {
+            const int total = (3 * E / 8) * T;
+            hipLaunchKernelGGL(k_qkv_finish,
+                               dim3((total + BLOCK - 1) / BLOCK),
+                               dim3(BLOCK), 0, s, slab, ks, q_buf,
+                               k_cache_layer, v_cache_layer, pos, seq,
+                               inv_freq, E, D, n_ctx, T);
+        }
+        return 1;
     }
     if (((E >> 4) % 2) == 0 && tiles3 / 2 >= 512) {
         // big models: RT=2 fused — halves the B-panel re-read while the
@@ -1795,13 +1862,14 @@ void launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
             (k_qkv16<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk, wv,
             xprep, normprep, ss_in, eps, q_buf, k_cache_layer,
             v_cache_layer, pos, seq, inv_freq, E, D, n_ctx, T)));
-        return;
+        return 0;
     }
     const dim3 grid(tiles3);
     DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_qkv16<WTc, JTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
         normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq,
         inv_freq, E, D, n_ctx, T)));
+    return 0;
 }
 
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,

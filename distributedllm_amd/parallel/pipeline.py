@@ -93,7 +93,9 @@ class DecodePipeline:
             x = eng.embed(self.tok[m])
         else:
             x = self.x_recv[m]
-        y = eng.forward(x, self.pos[m], self.seq[m])
+        # every sequence advances one token per step — the batched-decode
+        # contract the engine's fused qkv/attention path requires
+        y = eng.forward(x, self.pos[m], self.seq[m], decode=True)
         out = y
         if self.is_last:
             lg = eng.logits(y, all_logits=True)

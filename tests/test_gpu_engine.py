@@ -214,3 +214,21 @@ def test_cli_end_to_end_on_gpu(tmp_path):
     finally:
         srv.shutdown()
         srv.server_close()
+
+
+@pytest.mark.parametrize("T", [2, 40])
+def test_decode_fused_attention_matches(T):
+    """decode=True (distinct sequences) runs the qkv-slab + fused-
+    attention path; results must match the fp32 reference like the
+    unfused path does."""
+    f, hip, cpu = _engines(n_ctx=16, max_batch=64)
+    hp = f.hparams
+    torch.manual_seed(5)
+    for step in range(3):
+        x = torch.randn(T, hp.n_embd) * 0.5
+        pos = torch.full((T,), step, dtype=torch.int32)
+        seq = torch.arange(T, dtype=torch.int32)
+        y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda(),
+                            decode=True).cpu()
+        y_cpu = cpu.forward(x.clone(), pos, seq)
+        _assert_close(y_gpu, y_cpu, label=f"fused decode T={T} step {step}")
