@@ -726,6 +726,16 @@ __device__ __forceinline__ uint32_t h22u(__half2 h) {
     return c.u;
 }
 
+// write-through 16 B store (sc0 sc1): the split-K slab publish leaves no
+// dirty lines in the writer XCD's L2, so the consumer kernel's reads are
+// not cross-XCD dirty-line snoops (cdna_hip_programming.md §5 split-K
+// recipe: write-through slab stores beat plain stores + boundary flush)
+__device__ __forceinline__ void store_f4_wt(float* addr, float4 v) {
+    f32x4 w = {v.x, v.y, v.z, v.w};
+    asm volatile("global_store_dwordx4 %0, %1, off sc0 sc1\n\ts_nop 1"
+                 :: "v"(addr), "v"(w) : "memory");
+}
+
 // Per-wave state for the K loop: lane (i = l&15 row, ks = l>>4 k-span).
 // For q4 weights every wave range is aligned to the 4-K-block load group
 // (the repacked layout packs 4 consecutive K-blocks per dwordx4); ranges
@@ -1239,8 +1249,7 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
             v.y = acc[0][0][jt][1];
             v.z = acc[0][0][jt][2];
             v.w = acc[0][0][jt][3];
-            *reinterpret_cast<float4*>(
-                slab + ((size_t)j2 * 16) + (lane >> 4) * 4) = v;
+            store_f4_wt(slab + ((size_t)j2 * 16) + (lane >> 4) * 4, v);
         }
         return;
     }
@@ -1438,8 +1447,7 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_slab(
             v.y = acc[rt][0][jt][1];
             v.z = acc[rt][0][jt][2];
             v.w = acc[rt][0][jt][3];
-            *reinterpret_cast<float4*>(sl + (size_t)j2 * 16 +
-                                       (lane >> 4) * 4) = v;
+            store_f4_wt(sl + (size_t)j2 * 16 + (lane >> 4) * 4, v);
         }
     }
 }
@@ -1547,8 +1555,7 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16_slab(
                 v.y = acc[rt][n][jt][1];
                 v.z = acc[rt][n][jt][2];
                 v.w = acc[rt][n][jt][3];
-                *reinterpret_cast<float4*>(sl + (size_t)j2 * 16 +
-                                           (lane >> 4) * 4) = v;
+                store_f4_wt(sl + (size_t)j2 * 16 + (lane >> 4) * 4, v);
             }
         }
     }
