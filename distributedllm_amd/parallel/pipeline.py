@@ -210,6 +210,64 @@ class DecodePipeline:
     def current_tokens(self) -> torch.Tensor:
         return torch.stack(self.tok)
 
+    # ------------------------------------------------------- real serving
+
+    def prime(self, prompt_ids) -> None:
+        """Prefill every sequence with the same prompt (the batched-
+        serving demo semantics; per-sequence prompts are a scheduling
+        feature, not an engine one). Processes the prompt in <=64-token
+        tiles through the pipeline stages eagerly, leaving each
+        sequence's KV at position len(prompt) and tok at the last prompt
+        token, ready for run_steps/generate."""
+        dev = self.cfg.device
+        prompt = torch.as_tensor(prompt_ids, dtype=torch.int32, device=dev)
+        Tp = int(prompt.numel())
+        assert Tp >= 1, "prompt must be non-empty"
+        mbs = self.cfg.mbs
+        E = self.engine.hp.n_embd
+        tile = max(1, 64 // mbs)  # prompt positions per pipeline hop
+        for m in range(self.cfg.n_mb):
+            for p0 in range(0, Tp, tile):
+                p1 = min(Tp, p0 + tile)
+                n = p1 - p0
+                # token-major layout: [n positions x mbs sequences]
+                pos = (torch.arange(p0, p1, dtype=torch.int32, device=dev)
+                       .repeat_interleave(mbs))
+                seq = self.seq[m].repeat(n)
+                if self.is_first:
+                    toks = prompt[p0:p1].repeat_interleave(mbs)
+                    x = self.engine.embed(toks)
+                else:
+                    x = torch.empty(n * mbs, E, dtype=torch.float32,
+                                    device=dev)
+                    dist.recv(x, src=self.rank - 1)
+                y = self.engine.forward(x, pos, seq)
+                if not self.is_last:
+                    dist.send(y, dst=self.rank + 1)
+            self.pos[m].fill_(Tp)
+            self.tok[m].fill_(int(prompt[-1].item()))
+
+
+def pipeline_generate(pipe: DecodePipeline, prompt_ids, max_steps: int,
+                      greedy: bool = True):
+    """Prompt-conditioned generation on the RCCL pipeline: prefill then
+    greedy decode. Rank 0 returns the generated token ids
+    [global_batch, max_steps]; other ranks return None. (Temperature
+    sampling lives in the TCP client path — the pipeline's on-device
+    sampler is greedy, reference `sample_next_token` semantics.)"""
+    assert greedy, "pipeline sampling is greedy (device argmax)"
+    pipe.prime(prompt_ids)
+    out = []
+    for _ in range(max_steps):
+        pipe.run_steps(1)
+        if pipe.is_first:
+            out.append(pipe.current_tokens().clone())
+    if pipe.is_first:
+        # [steps][n_mb][mbs] -> [n_mb*mbs, steps]
+        t = torch.stack(out)
+        return t.permute(1, 2, 0).reshape(-1, len(out))
+    return None
+
 
 def timed_decode(pipe: DecodePipeline, steps: int, warmup: int,
                  device: str, use_graphs: bool = True) -> float:

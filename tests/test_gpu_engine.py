@@ -232,3 +232,27 @@ def test_decode_fused_attention_matches(T):
                             decode=True).cpu()
         y_cpu = cpu.forward(x.clone(), pos, seq)
         _assert_close(y_gpu, y_cpu, label=f"fused decode T={T} step {step}")
+
+
+def test_pipeline_generate_on_gpu():
+    """Prompt-conditioned prefill + greedy decode through the HIP engine
+    (single stage) produces finite, deterministic continuations."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.formats import slicer, synthetic
+    from distributedllm_amd.parallel.pipeline import (
+        DecodePipeline, PipelineConfig, pipeline_generate)
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    eng = HIPSliceEngine.from_ggml(f, n_ctx=64, max_batch=2)
+    eng.attach_extra(ex)
+    cfg = PipelineConfig(mbs=2, n_mb=1, device="cuda")
+    pipe = DecodePipeline(eng, cfg, rank=0, world=1)
+    toks = pipeline_generate(pipe, [5, 9, 3], max_steps=6)
+    assert toks.shape == (2, 6)
+    assert toks[0].tolist() == toks[1].tolist()  # same prompt -> same greedy
+    # a second pipeline over the same weights reproduces the continuation
+    eng2 = HIPSliceEngine.from_ggml(f, n_ctx=64, max_batch=2)
+    eng2.attach_extra(ex)
+    pipe2 = DecodePipeline(eng2, cfg, rank=0, world=1)
+    toks2 = pipeline_generate(pipe2, [5, 9, 3], max_steps=6)
+    assert toks2.tolist() == toks.tolist()
