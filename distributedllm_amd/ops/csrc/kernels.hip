@@ -1722,8 +1722,10 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
                    float* ss, int cols, int T) {
     // chunk columns so the launch spreads over ~256 CUs even at T=16
     // (a T-block launch measured 4.6 us on 16 CUs); ss zeroed upstream
-    // target ~512 blocks with >=64 K-chunks of work per block
-    const int want = (512 + T - 1) / max(T, 1);
+    // ~128 blocks total: a microbenchmark of this kernel shape (see
+    // profiles/microbench_small_kernel.hip) shows larger grids LOSE —
+    // launch/ramp overhead outweighs the parallelism for ~1 MB of work
+    const int want = (128 + T - 1) / max(T, 1);
     const int chunks = max(1, min(want, (cols >> 3) / 64));
     hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
                        xprep, ss, cols, pick_jt(T));
@@ -1783,7 +1785,7 @@ int gemm16_ks(int rows) {
 
 void launch_reduce_prep(hipStream_t s, float* y, const float* slab, int ks,
                         unsigned short* xprep, float* ss, int cols, int T) {
-    const int want = (512 + T - 1) / max(T, 1);
+    const int want = (128 + T - 1) / max(T, 1);  // see microbench note
     const int chunks = max(1, min(want, (cols >> 3) / 64));
     hipLaunchKernelGGL(k_reduce_prep, dim3(T, chunks), dim3(BLOCK), 0, s, y,
                        slab, ks, xprep, ss, cols, pick_jt(T));
