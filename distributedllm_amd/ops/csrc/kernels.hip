@@ -1235,21 +1235,26 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
         return;
     }
     if (MODE == GM_SLAB) {
-        // plain-store the partial tile; k_reduce_prep sums the KS slices
-        // at the next kernel boundary (no atomics, no pre-zeroing; the
-        // boundary provides visibility). Slab layout:
-        // f32[R][KS][kMaxTok][16 rows]; `y` is reused as the slab base.
-        float* slab = y +
-            ((size_t)blockIdx.x * gridDim.y + blockIdx.y) * 64 * 16;
+        // write-through-store the partial tile(s); k_reduce_prep sums the
+        // KS slices at the next kernel boundary (no atomics, no
+        // pre-zeroing). Slab layout: f32[R][KS][kMaxTok][16 rows]; `y` is
+        // reused as the slab base. With RT>1 a block covers RT row tiles
+        // sharing its B-panel reads.
 #pragma unroll
-        for (int jt = 0; jt < JT; ++jt) {
-            const int j2 = jt * 16 + j;
-            float4 v;
-            v.x = acc[0][0][jt][0];
-            v.y = acc[0][0][jt][1];
-            v.z = acc[0][0][jt][2];
-            v.w = acc[0][0][jt][3];
-            store_f4_wt(slab + ((size_t)j2 * 16) + (lane >> 4) * 4, v);
+        for (int rt = 0; rt < RT; ++rt) {
+            float* slab = y +
+                ((size_t)(blockIdx.x * RT + rt) * gridDim.y + blockIdx.y) *
+                    64 * 16;
+#pragma unroll
+            for (int jt = 0; jt < JT; ++jt) {
+                const int j2 = jt * 16 + j;
+                float4 v;
+                v.x = acc[rt][0][jt][0];
+                v.y = acc[rt][0][jt][1];
+                v.z = acc[rt][0][jt][2];
+                v.w = acc[rt][0][jt][3];
+                store_f4_wt(slab + ((size_t)j2 * 16) + (lane >> 4) * 4, v);
+            }
         }
         return;
     }
@@ -1800,12 +1805,24 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
     if (mode == GM_ATOMIC || mode == GM_SLAB) {
         // split K so R*KS lands near 2-3 blocks/CU (256 CUs)
         const int ks = gemm16_ks(w.rows);
-        const dim3 grid(R, ks);
         if (mode == GM_SLAB) {
-            DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
-                (k_gemm16<WTc, GM_SLAB, JTc>), grid, dim3(BLOCK), 0, s, w,
-                bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
+            if (R % 2 == 0) {  // RT=2: halve the B-panel re-read
+                const dim3 grid(R / 2, ks);
+                DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T),
+                    hipLaunchKernelGGL(
+                        (k_gemm16<WTc, GM_SLAB, JTc, 2>), grid, dim3(BLOCK),
+                        0, s, w, bprep, normprep, ss_in, eps, y, xprep_out,
+                        ss_out, T)));
+            } else {
+                const dim3 grid(R, ks);
+                DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T),
+                    hipLaunchKernelGGL(
+                        (k_gemm16<WTc, GM_SLAB, JTc>), grid, dim3(BLOCK), 0,
+                        s, w, bprep, normprep, ss_in, eps, y, xprep_out,
+                        ss_out, T)));
+            }
         } else {
+            const dim3 grid(R, ks);
             DISPATCH_WT2(w.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
                 (k_gemm16<WTc, GM_ATOMIC, JTc>), grid, dim3(BLOCK), 0, s, w,
                 bprep, normprep, ss_in, eps, y, xprep_out, ss_out, T)));
