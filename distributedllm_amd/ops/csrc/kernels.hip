@@ -1865,6 +1865,7 @@ int launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
     const int tiles3 = 3 * (E >> 4);
     // slab split-K + RT=2 path when the fused grid underfills the chip
     if (slab != nullptr && tiles3 < 1024 && ((E >> 4) % 2) == 0) {
+        // RT=2 measured best (RT=4 loses ~3%: fill drops below 2/CU)
         const int ks = qkv16_ks(E);
         const dim3 grid(tiles3 / 2, ks);
         DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
