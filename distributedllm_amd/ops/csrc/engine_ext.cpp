@@ -315,7 +315,8 @@ class SliceEngine {
             // reduce+residual+sumsq+xprep pass per matrix (no atomics;
             // the kernel boundary provides slab visibility).
             static const bool no_split = std::getenv("DLLM_NO_SPLITK");
-            const bool split = !no_split && (E_ / 16) < 512;
+            // slab split-K + RT=2 wo/w2 wins at every model size tested
+            const bool split = !no_split && (E_ / 16) % 2 == 0;
             float* slab = slab_.data_ptr<float>();
             const int ks = gemm16_ks(E_);
             if (split) {

@@ -1785,6 +1785,9 @@ int gemm16_ks(int rows) {
     const int R = rows / 16;
     int ks = 1;
     while (R * ks < 512 && ks < 8) ks <<= 1;
+    // the slab path runs RT=2 (half the blocks) — keep at least ks=2 so
+    // big-E models (65B: R=512) still land >=512 blocks
+    if (ks < 2) ks = 2;
     return ks;
 }
 
