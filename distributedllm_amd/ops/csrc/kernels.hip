@@ -746,13 +746,13 @@ struct KLoop {
     int kb0, kb1;  // this wave's K-block range
     // split a [b0, b1) block range (grid-level split-K) across the waves;
     // align=4 for q4 group loads, 1 for f16
-    __device__ void init_range(int b0, int b1, int align) {
+    __device__ void init_range(int b0, int b1, int align, int nwaves = NWAVES) {
         lane = threadIdx.x & (WAVE - 1);
         i = lane & 15;
         ks = lane >> 4;
         const int wid = threadIdx.x / WAVE;
         const int am = align - 1;
-        int per = (b1 - b0 + NWAVES - 1) / NWAVES;
+        int per = (b1 - b0 + nwaves - 1) / nwaves;
         per = (per + am) & ~am;
         // readfirstlane: wid is wave-uniform but derived from threadIdx,
         // so without this the compiler treats every K loop bounded by
@@ -816,7 +816,8 @@ __device__ __forceinline__ void a_frag_q4(uint32_t q, uint32_t ab,
 // RT = row tiles per wave: the wave's B fragments feed RT A-tile streams,
 // dividing the (L2-heavy) B-panel re-read traffic and the B-build VALU by
 // RT. acc[rt][n][jt][jj].
-template <int WT, bool NORM, int NM, int JT, int RT = 1, int PF = 4>
+template <int WT, bool NORM, int NM, int JT, int RT = 1, int PF = 4,
+          int NW = NWAVES>
 __device__ __forceinline__ void wave_tile_kloop(
     const WMat2* const* ws, int tile_row,
     const unsigned short* __restrict__ xprep,
@@ -824,7 +825,7 @@ __device__ __forceinline__ void wave_tile_kloop(
     const float* __restrict__ ss_in, float eps,
     float acc[RT][NM][JT][4], int b0, int b1) {
     KLoop kl;
-    kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4);
+    kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4, NW);
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
@@ -1063,11 +1064,11 @@ __device__ __forceinline__ void wave_tile_kloop(
                                   : c0[rt][n][jt][jj];
 }
 
-// LDS combine of the 4 waves' partial accumulators; wave 0 ends with the
-// full 16x16 tile (4 rows x 1 col per lane). NACC = accumulator sets.
-template <int NACC>
+// LDS combine of the NW waves' partial accumulators; wave 0 ends with
+// the full 16x16 tile (4 rows x 1 col per lane). NACC = accumulator sets.
+template <int NACC, int NW = NWAVES>
 __device__ __forceinline__ void combine_acc(float acc[NACC][4],
-                                            float* lds /* 3*64*4*NACC */) {
+                                            float* lds /* (NW-1)*64*4*NACC */) {
     const int wid = threadIdx.x / WAVE;
     const int lane = threadIdx.x & (WAVE - 1);
     if (wid > 0) {
@@ -1080,7 +1081,7 @@ __device__ __forceinline__ void combine_acc(float acc[NACC][4],
     __syncthreads();
     if (wid == 0) {
 #pragma unroll
-        for (int w = 0; w < 3; ++w) {
+        for (int w = 0; w < NW - 1; ++w) {
             const float* src = lds + ((w * 64 + lane) * 4) * NACC;
 #pragma unroll
             for (int n = 0; n < NACC; ++n)
@@ -1927,6 +1928,7 @@ void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
             normprep, ss_in, eps, gprep, T)));
         return;
     }
+
     const dim3 grid(tilesF);
     DISPATCH_WT2(w1.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_ffn16<WTc, JTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep,
