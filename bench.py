@@ -8,8 +8,8 @@ JSON line with the whole-job tokens/sec.
 
 Metric: tokens generated per second across the whole node (every sequence
 advances one token per step; tokens/step = global_batch). Weak scaling:
-global batch = 4 sequences per pipeline stage, so per-GPU work per step is
-constant as N grows (each stage holds ~L/N layers but serves N micro-
+global batch = 64 sequences per pipeline stage, so per-GPU work per step
+is constant as N grows (each stage holds ~L/N layers but serves N micro-
 batches per step). Synthetic setup: random-init weights of the real
 architecture generated directly on-device in the engine's q4_0 layout
 (identical compute + HBM traffic to a real checkpoint; no network for real
@@ -82,7 +82,7 @@ def main() -> int:
     else:  # CPU fallback so the contract is testable without a GPU
         from distributedllm_amd.engine import TorchSliceEngine
         from distributedllm_amd.formats import synthetic
-        from distributedllm_amd.models.llama import weights_from_ggml, LlamaPreset
+        from distributedllm_amd.models.llama import weights_from_ggml
         tiny = PRESETS["tiny"]
         hp = tiny.hparams(FTYPES[args.ftype])
         parts = partition_layers(tiny.n_layer, world)

@@ -11,7 +11,7 @@ from typing import List, Optional
 # note: import the submodule BEFORE the registry dict of the same name —
 # the dict would otherwise shadow the submodule for `from . import commands`
 from . import commands as _commands  # noqa: F401  (registers all commands)
-from .base import Command, commands
+from .base import commands
 
 
 def build_parser() -> argparse.ArgumentParser:

@@ -9,8 +9,8 @@ model's slice layout against the hparams, which is what the reference's
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from dataclasses import dataclass
+from typing import Dict, List, Tuple
 
 from .client import Connection, parse_address
 

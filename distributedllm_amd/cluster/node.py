@@ -20,7 +20,7 @@ import os
 import socket
 import socketserver
 import threading
-from typing import Dict, Optional
+from typing import Optional
 
 import numpy as np
 import torch

@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import json
 import os
-from typing import Dict, Optional
+from typing import Dict
 
 from ..formats import ggml, slicer
 from ..formats.synthetic import build_model

@@ -19,12 +19,12 @@ once at load on the host, uploads via torch pinned copies.
 from __future__ import annotations
 
 import math
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import numpy as np
 import torch
 
-from ..formats import ggml, q4
+from ..formats import ggml
 from ..models.llama import RMS_EPS, ROPE_BASE, rms_norm, rope_interleaved
 
 

@@ -27,7 +27,6 @@ from typing import Dict, List, Tuple
 
 import numpy as np
 
-from ..models.llama import LlamaPreset
 from . import ggml
 from .synthetic import synthetic_vocab
 
