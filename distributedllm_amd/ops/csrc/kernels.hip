@@ -1841,8 +1841,13 @@ void launch_gemm16(hipStream_t s, const WMat2& w,
                                eps, y, xprep_out, ss_out, T);
         else if (mode == GM_NORM_PLAIN) {
             // lm_head: thousands of row tiles re-read the same B panel —
-            // RT=2 halves that L2 traffic (grid R/2 still fills the chip)
-            if (R % 2 == 0 && R >= 1024)
+            // RT divides that L2 traffic while the grid still fills
+            if (R % 4 == 0 && R / 4 >= 448)
+                hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc, 4>),
+                                   dim3(R / 4), dim3(BLOCK), 0, s, w, bprep,
+                                   normprep, ss_in, eps, y, xprep_out,
+                                   ss_out, T);
+            else if (R % 2 == 0 && R >= 1024)
                 hipLaunchKernelGGL((k_gemm16<WTc, GM_NORM_PLAIN, JTc, 2>),
                                    dim3(R / 2), dim3(BLOCK), 0, s, w, bprep,
                                    normprep, ss_in, eps, y, xprep_out,
