@@ -109,12 +109,47 @@ class _NodeCommand(Command):
         return Connection(host, port)
 
 
-class StatusCommand(_NodeCommand):
+class StatusCommand(Command):
     name = "status"
-    help = "Report a node's status (loaded slice, device)"
+    help = ("Report a node's status (host:port), or a whole cluster's "
+            "pipeline readiness (--config)")
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("address", nargs="?", default=None,
+                       help="node address host:port")
+        p.add_argument("--config", default=None,
+                       help="cluster config: poll every nodes_map node")
+        p.add_argument("--n-layer", type=int, default=None,
+                       help="with --config: also check the loaded slices "
+                            "tile this many layers")
 
     def __call__(self, args) -> int:
-        conn = self._conn(args)
+        if (args.address is None) == (args.config is None):
+            print("provide exactly one of <address> / --config",
+                  file=sys.stderr)
+            return 2
+        if args.config:
+            from ..cluster.control import ControlCenter
+            with open(args.config) as f:
+                nodes_map = json.load(f)["nodes_map"]
+            cc = ControlCenter(nodes_map)
+            status = cc.get_status()
+            for addr, st in sorted(status.items()):
+                print(json.dumps({"node": addr,
+                                  "connectivity": st.connectivity,
+                                  "model": st.model,
+                                  "first_layer": st.first_layer,
+                                  "n_layers": st.n_layers,
+                                  "device": st.device,
+                                  "error": st.error}))
+            if args.n_layer:
+                ready, why = cc.pipeline_ready(args.n_layer)
+                print(f"pipeline_ready: {ready} ({why})")
+                return 0 if ready else 1
+            return 0
+        from ..cluster.client import Connection, parse_address
+        host, port = parse_address(args.address)
+        conn = Connection(host, port)
         s = conn.get_status()
         print(json.dumps({"status": s.status, "model": s.model,
                           "first_layer": s.first_layer,

@@ -258,3 +258,16 @@ def test_perplexity_dataset_source(node, tmp_path, capsys):
                             "--dataset", str(ds_path), "--seed", "0",
                             "--root", str(root)]) == 0
     assert "perplexity:" in capsys.readouterr().out
+
+
+def test_status_config_cluster(node, tmp_path, capsys):
+    addr = f"127.0.0.1:{node.port}"
+    cfg = {"model_id": "x", "location": "synthetic:tiny",
+           "nodes_map": {addr: [0, 2]}}
+    cfg_path = tmp_path / "c.json"
+    cfg_path.write_text(json.dumps(cfg))
+    # nodes up but no slice loaded -> not ready (exit 1 with --n-layer)
+    assert execute_command(["status", "--config", str(cfg_path),
+                            "--n-layer", "3"]) == 1
+    out = capsys.readouterr().out
+    assert '"connectivity": true' in out and "pipeline_ready: False" in out
