@@ -33,8 +33,13 @@ def _random_dataset_prompt(path: str, seed=None) -> str:
 
 
 def _progress(label: str):
+    state = {"last": -1.0}
+
     def cb(sent: int, total: int) -> None:
         pct = 100.0 * sent / max(total, 1)
+        if pct - state["last"] < 1.0 and sent < total:
+            return  # throttle to ~1% steps (piped output stays readable)
+        state["last"] = pct
         print(f"\r[{label}] {sent}/{total} bytes ({pct:5.1f}%)",
               end="", flush=True)
         if sent >= total:
