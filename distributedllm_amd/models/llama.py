@@ -177,21 +177,31 @@ class LlamaSliceRef:
 
 
 class LlamaExtraRef:
-    """Embedding lookup + final norm / lm_head from extra-layer weights."""
+    """Embedding lookup + final norm / lm_head from extra-layer weights.
 
-    def __init__(self, weights: Dict[str, torch.Tensor]):
-        self.tok = weights["tok_embeddings.weight"]   # [V, E]
-        self.norm = weights["norm.weight"]            # [E]
-        self.out = weights["output.weight"]           # [V, E]
+    On a machine with a GPU the lm_head matmul runs there (a [V, E] fp32
+    matmul per generated token is 20-30 ms on CPU at 3B scale and
+    dominated the TCP client's single-stream latency); results come back
+    as CPU tensors either way."""
+
+    def __init__(self, weights: Dict[str, torch.Tensor],
+                 device: Optional[str] = None):
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        self.device = device
+        self.tok = weights["tok_embeddings.weight"].to(device)   # [V, E]
+        self.norm = weights["norm.weight"].to(device)            # [E]
+        self.out = weights["output.weight"].to(device)           # [V, E]
 
     def embed(self, tokens: List[int]) -> torch.Tensor:
-        idx = torch.tensor(tokens, dtype=torch.long)
-        return self.tok[idx]
+        idx = torch.tensor(tokens, dtype=torch.long, device=self.device)
+        return self.tok[idx].cpu()
 
     def logits(self, x: torch.Tensor, all_logits: bool = False) -> torch.Tensor:
+        x = x.to(self.device)
         y = rms_norm(x) * self.norm
         lg = y @ self.out.T
-        return lg if all_logits else lg[-1:]
+        return (lg if all_logits else lg[-1:]).cpu()
 
 
 def weights_from_ggml(f: ggml.GGMLFile) -> Dict[str, torch.Tensor]:
