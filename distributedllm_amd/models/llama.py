@@ -72,6 +72,9 @@ PRESETS: Dict[str, LlamaPreset] = {
     # tiny configs for tests (head_dim 8 resp. 100-like non-pow2 = 20)
     "tiny": LlamaPreset("tiny", 256, 64, 32, 4, 3),
     "tiny_oddhead": LlamaPreset("tiny_oddhead", 256, 96, 32, 4, 2),
+    # mid-size test config: large enough that the split-K/RT kernel paths
+    # run with realistic grids (E/16=32 tiles, F=1408)
+    "small": LlamaPreset("small", 512, 512, 64, 8, 2),
 }
 
 

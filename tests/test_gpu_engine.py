@@ -256,3 +256,34 @@ def test_pipeline_generate_on_gpu():
     pipe2 = DecodePipeline(eng2, cfg, rank=0, world=1)
     toks2 = pipeline_generate(pipe2, [5, 9, 3], max_steps=6)
     assert toks2.tolist() == toks.tolist()
+
+
+@pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_0,
+                                   ggml.FTYPE_MOSTLY_F16])
+def test_midsize_parity(ftype):
+    """E=512 exercises the slab/RT kernel paths with realistic grids
+    (the tiny presets run them at degenerate sizes)."""
+    f, hip, cpu = _engines(preset="small", ftype=ftype, n_ctx=96,
+                           max_batch=1)
+    hp = f.hparams
+    torch.manual_seed(7)
+    T = 48
+    x = torch.randn(T, hp.n_embd) * 0.5
+    pos = torch.arange(T, dtype=torch.int32)
+    seq = torch.zeros(T, dtype=torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label=f"midsize prefill ftype={ftype}")
+    lg_gpu = hip.logits(y_gpu.cuda().contiguous(), all_logits=True).cpu()
+    lg_cpu = cpu.logits(y_cpu, all_logits=True)
+    _assert_close(lg_gpu, lg_cpu, label="midsize logits")
+    # decode continuation with distinct sequences (fused-attention path)
+    f2, hip2, cpu2 = _engines(preset="small", ftype=ftype, n_ctx=16,
+                              max_batch=24)
+    x = torch.randn(24, hp.n_embd) * 0.5
+    pos = torch.zeros(24, dtype=torch.int32)
+    seq = torch.arange(24, dtype=torch.int32)
+    y_gpu = hip2.forward(x.cuda(), pos.cuda(), seq.cuda(),
+                         decode=True).cpu()
+    y_cpu = cpu2.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label="midsize fused decode")
