@@ -225,9 +225,13 @@ class HIPSliceEngine:
                     0.01).contiguous()
 
         shapes = [(E, E), (E, E), (E, E), (E, E), (F, E), (E, F), (F, E)]
+        eng._layers_cache = []
         for li in range(n_layers):
             mats = [mat(r, c) for r, c in shapes]
-            eng._eng.set_layer(li, norm_w(E), norm_w(E), mats)
+            an, fn = norm_w(E), norm_w(E)
+            eng._layers_cache.append((an, fn, mats))
+            eng._eng.set_layer(li, an, fn, mats)
+        eng._extra_cache = None
         if with_extra:
             # the embedding table uses the legacy SoA layout (gather kernel)
             if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
@@ -249,10 +253,29 @@ class HIPSliceEngine:
                          0.02)
                 tok_s, tok_t = torch.empty(0), wt
             out_d, out_s, out_t = mat(V, E)
-            eng._eng.set_extra(tok_d, tok_s, tok_t, norm_w(E), out_d, out_s,
+            fin = norm_w(E)
+            eng._extra_cache = (tok_d, tok_s, tok_t, fin, out_d, out_s,
+                                out_t, V)
+            eng._eng.set_extra(tok_d, tok_s, tok_t, fin, out_d, out_s,
                                out_t, V)
             eng.has_extra = True
         return eng
+
+    def clone_shared(self) -> "HIPSliceEngine":
+        """A second engine over the SAME weight tensors (HBM shared) with
+        its own KV cache and side-channel buffers — lets independent
+        micro-batches run concurrently on separate HIP streams without
+        duplicating the model."""
+        assert getattr(self, "_layers_cache", None) is not None, \
+            "clone_shared requires an engine built by .random()"
+        twin = HIPSliceEngine(self.hp, self.n_layers, self.first_layer,
+                              self.n_ctx, self.max_batch)
+        for li, (an, fn, mats) in enumerate(self._layers_cache):
+            twin._eng.set_layer(li, an, fn, mats)
+        if self._extra_cache is not None:
+            twin._eng.set_extra(*self._extra_cache)
+            twin.has_extra = True
+        return twin
 
     def load_layers(self, f: ggml.GGMLFile) -> None:
         tm = f.tensor_map()

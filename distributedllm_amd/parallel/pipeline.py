@@ -55,8 +55,12 @@ class DecodePipeline:
     """
 
     def __init__(self, engine, cfg: PipelineConfig,
-                 rank: int = 0, world: int = 1):
+                 rank: int = 0, world: int = 1, engines=None):
+        """engines: optional per-micro-batch engine list (weight-sharing
+        clones) — single-GPU mode then runs each micro-batch on its own
+        HIP stream, overlapping the latency-bound kernel chains."""
         self.engine = engine
+        self.engines = engines  # None => self.engine for every mb
         self.cfg = cfg
         self.rank = rank
         self.world = world
@@ -67,8 +71,15 @@ class DecodePipeline:
         M, mbs = cfg.n_mb, cfg.mbs
         self.pos = [torch.zeros(mbs, dtype=torch.int32, device=dev)
                     for _ in range(M)]
-        self.seq = [torch.arange(m * mbs, (m + 1) * mbs, dtype=torch.int32,
-                                 device=dev) for m in range(M)]
+        # per-mb engines have their OWN KV caches of mbs slots -> local
+        # sequence ids; a shared engine partitions its slots across mbs
+        if engines is not None:
+            self.seq = [torch.arange(mbs, dtype=torch.int32, device=dev)
+                        for _ in range(M)]
+        else:
+            self.seq = [torch.arange(m * mbs, (m + 1) * mbs,
+                                     dtype=torch.int32, device=dev)
+                        for m in range(M)]
         # zero-init: mid-stage graph-capture warmup runs before any real
         # activations arrive; empty-buffer garbage could seed NaNs into
         # the KV cache rows written during warmup
@@ -79,8 +90,12 @@ class DecodePipeline:
                     for _ in range(M)]
         self._graphs = None      # per-mb captured hipGraphs
         self._graph_out = None   # per-mb tensors the graph writes
+        self._streams = None     # per-mb streams (single-GPU multi-mb)
 
     # ------------------------------------------------------ hipGraph mode
+
+    def _eng(self, m: int):
+        return self.engines[m] if self.engines is not None else self.engine
 
     def _mb_compute(self, m: int):
         """The capturable (comm-free) compute of micro-batch m.
@@ -88,7 +103,7 @@ class DecodePipeline:
         Returns the tensor the next stage needs: activations for mid
         stages, sampled token ids for the last stage (None for single-GPU,
         where tok[m] is updated in place)."""
-        eng = self.engine
+        eng = self._eng(m)
         if self.is_first:
             x = eng.embed(self.tok[m])
         else:
@@ -171,6 +186,23 @@ class DecodePipeline:
         comm/compute overlap across pipeline stages)."""
         n_mb = self.cfg.n_mb
         if self.world == 1 or steps == 0:
+            if (self.engines is not None and self.cfg.device == "cuda"
+                    and n_mb > 1):
+                # single-GPU multi-stream: each micro-batch's step chain
+                # lives on its own stream; chains are independent (per-mb
+                # engines/KV/tokens), so the whole schedule is enqueued
+                # with no cross-stream syncs and the streams fill each
+                # other's latency stalls.
+                if self._streams is None:
+                    self._streams = [torch.cuda.Stream()
+                                     for _ in range(n_mb)]
+                for _ in range(steps):
+                    for m in range(n_mb):
+                        with torch.cuda.stream(self._streams[m]):
+                            self._compute(m)
+                for st in self._streams:
+                    torch.cuda.current_stream().wait_stream(st)
+                return
             for _ in range(steps):
                 for m in range(n_mb):
                     self._advance_mb(m)
