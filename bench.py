@@ -79,12 +79,14 @@ def main() -> int:
     # Weak scaling: global batch grows with N, per-GPU work fixed.
     n_mb = world if world > 1 else args.single_gpu_mbs
     if world == 1 and n_mb > 1:
-        # cap concurrent engines by KV-cache memory (each clone owns a
-        # full [L, mbs, ctx, E] f16 K+V pair; keep total under ~200 GB)
+        # each clone owns a full [L, mbs, ctx, E] f16 K+V pair; 2 streams
+        # measured a net LOSS on every model (contention without enough
+        # overlap) while 3-4 won, so use multi-stream only when >=3 fit
+        # in ~200 GB of KV budget
         kv_bytes = (preset.n_layer * args.mbs * args.ctx *
                     preset.n_embd * 4)
-        while n_mb > 1 and kv_bytes * n_mb > 150e9:
-            n_mb -= 1
+        fit = max(1, int(200e9 // max(kv_bytes, 1)))
+        n_mb = min(n_mb, fit) if fit >= 3 else 1
     cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
 
     engines = None
