@@ -126,6 +126,29 @@ def _single_generate_reference():
     return pipeline_generate(pipe, [5, 9, 3], max_steps=4).tolist()
 
 
+def test_multi_engine_micro_batches_match_shared_engine():
+    """Per-micro-batch engine clones (bench.py's single-GPU multi-stream
+    mode) use LOCAL sequence ids into each clone's own KV cache; the
+    decoded tokens must be identical to one shared engine partitioning
+    its KV slots across micro-batches. On CPU this runs the same
+    DecodePipeline code minus the per-stream scheduling."""
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    n_mb = 2
+    engines = []
+    for _ in range(n_mb):
+        e = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=MBS)
+        e.attach_extra(ex)
+        engines.append(e)
+    cfg = PipelineConfig(mbs=MBS, n_mb=n_mb, device="cpu")
+    pipe = DecodePipeline(engines[0], cfg, rank=0, world=1,
+                          engines=engines)
+    assert pipe.seq[1].tolist() == [0, 1]  # local ids, not [2, 3]
+    _seed_tokens(pipe, n_mb)
+    pipe.run_steps(STEPS)
+    assert pipe.current_tokens().tolist() == _single_reference_tokens(n_mb)
+
+
 @pytest.mark.timeout(120)
 def test_pipeline_generate_prompt_conditioned():
     """Prefill + greedy decode through 2 pipeline stages must produce the
