@@ -92,8 +92,14 @@ def main() -> int:
     engines = None
     if device == "cuda":
         from distributedllm_amd.engine import HIPSliceEngine
+        # world>1 shares ONE engine across the n_mb in-flight micro-
+        # batches, so its KV cache must hold every micro-batch's slots
+        # (the pipeline assigns sequence ids [0, n_mb*mbs)); single-GPU
+        # multi-stream instead gives each micro-batch its own clone with
+        # LOCAL ids, so mbs slots per engine suffice.
+        slots = args.mbs * (n_mb if world > 1 else 1)
         eng = HIPSliceEngine.random(hp, n_layers=count, first_layer=first,
-                                    n_ctx=args.ctx, max_batch=args.mbs,
+                                    n_ctx=args.ctx, max_batch=slots,
                                     seed=args.seed, with_extra=True)
         if world == 1 and n_mb > 1:
             engines = [eng] + [eng.clone_shared() for _ in range(n_mb - 1)]

@@ -74,12 +74,24 @@ class DecodePipeline:
         # per-mb engines have their OWN KV caches of mbs slots -> local
         # sequence ids; a shared engine partitions its slots across mbs
         if engines is not None:
+            need = mbs
             self.seq = [torch.arange(mbs, dtype=torch.int32, device=dev)
                         for _ in range(M)]
         else:
+            need = M * mbs
             self.seq = [torch.arange(m * mbs, (m + 1) * mbs,
                                      dtype=torch.int32, device=dev)
                         for m in range(M)]
+        # the engine KV cache is indexed by sequence id with no device-
+        # side bounds check — catch undersized engines at construction
+        for e in (engines if engines is not None else [engine]):
+            cap = int(getattr(e, "max_batch", need))
+            if cap < need:
+                raise ValueError(
+                    f"engine KV holds {cap} sequence slots but the "
+                    f"pipeline assigns ids [0, {need}) "
+                    f"(n_mb={M} x mbs={mbs}); build the engine with "
+                    f"max_batch >= {need}")
         # zero-init: mid-stage graph-capture warmup runs before any real
         # activations arrive; empty-buffer garbage could seed NaNs into
         # the KV cache rows written during warmup
