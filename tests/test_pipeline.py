@@ -168,3 +168,29 @@ def test_pipeline_generate_prompt_conditioned():
     # both sequences of a micro-batch got identical prompts -> identical
     # greedy continuations; and they continue the prompt deterministically
     assert got[0] == got[1]
+
+
+@pytest.mark.timeout(180)
+def test_bench_driver_contract():
+    """`python bench.py --steps K --warmup W` must print exactly one JSON
+    line with the driver-contract fields (the round driver parses this)."""
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=150,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    r = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in r, key
+    assert r["steps"] == 2 and r["warmup"] == 1 and r["n_gpus"] == 1
+    assert r["unit"] == "tokens/s" and r["higher_is_better"] is True
+    assert r["scaling"] == "weak" and "synthetic" in r["data"]
+    assert r["value"] > 0
+    assert {"model", "global_batch", "seq_len",
+            "parallelism"} <= set(r["config"])
