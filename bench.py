@@ -79,13 +79,20 @@ def main() -> int:
     # Weak scaling: global batch grows with N, per-GPU work fixed.
     n_mb = world if world > 1 else args.single_gpu_mbs
     if world == 1 and n_mb > 1:
-        # each clone owns a full [L, mbs, ctx, E] f16 K+V pair; 2 streams
-        # measured a net LOSS on every model (contention without enough
-        # overlap) while 3-4 won, so use multi-stream only when >=3 fit
-        # in ~200 GB of KV budget
+        # each clone owns a full [L, mbs, ctx, E] f16 K+V pair (weights
+        # are shared); 2 streams measured a net LOSS on every model
+        # (contention without enough overlap) while 3-4 won, so use
+        # multi-stream only when >=3 KV sets fit in HBM next to the
+        # model weights (~260 GB usable of 288)
         kv_bytes = (preset.n_layer * args.mbs * args.ctx *
                     preset.n_embd * 4)
-        fit = max(1, int(200e9 // max(kv_bytes, 1)))
+        E, F, V = preset.n_embd, preset.n_ff, 32000
+        n_weights = (preset.n_layer * (4 * E * E + 3 * E * F) +
+                     2 * V * E)
+        bpw = {"q4_0": 0.5625, "q4_1": 0.625, "f16": 2.0,
+               "f32": 4.0}[args.ftype]
+        w_bytes = n_weights * bpw * 1.1  # repack padding/scales margin
+        fit = max(1, int((260e9 - w_bytes) // max(kv_bytes, 1)))
         n_mb = min(n_mb, fit) if fit >= 3 else 1
     cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
 

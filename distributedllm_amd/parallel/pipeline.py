@@ -103,6 +103,7 @@ class DecodePipeline:
         self._graphs = None      # per-mb captured hipGraphs
         self._graph_out = None   # per-mb tensors the graph writes
         self._streams = None     # per-mb streams (single-GPU multi-mb)
+        self.temperature = 0.0   # 0 = greedy argmax (graph-capturable)
 
     # ------------------------------------------------------ hipGraph mode
 
@@ -126,7 +127,16 @@ class DecodePipeline:
         out = y
         if self.is_last:
             lg = eng.logits(y, all_logits=True)
-            nxt = eng.argmax(lg)
+            if self.temperature > 0.0:
+                # device-side temperature sampling (eager path only —
+                # capture_graphs pins greedy; RNG state is not
+                # graph-stable). Repetition-penalty sampling lives in
+                # the TCP client path (engine/sampler.py parity).
+                probs = torch.softmax(lg.float() / self.temperature,
+                                      dim=-1)
+                nxt = torch.multinomial(probs, 1)[:, 0].to(torch.int32)
+            else:
+                nxt = eng.argmax(lg)
             if self.world == 1:
                 self.tok[m].copy_(nxt)
                 out = None
@@ -140,6 +150,8 @@ class DecodePipeline:
         steady state). Comm (RCCL send/recv) stays eager between replays.
         The warmup steps run for real (they advance positions/KV)."""
         assert self.cfg.device == "cuda"
+        assert self.temperature == 0.0, \
+            "sampled decode is eager-only (RNG is not graph-stable)"
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):
@@ -293,13 +305,17 @@ class DecodePipeline:
 
 
 def pipeline_generate(pipe: DecodePipeline, prompt_ids, max_steps: int,
-                      greedy: bool = True):
+                      greedy: bool = True, temperature: float = 0.0):
     """Prompt-conditioned generation on the RCCL pipeline: prefill then
-    greedy decode. Rank 0 returns the generated token ids
-    [global_batch, max_steps]; other ranks return None. (Temperature
-    sampling lives in the TCP client path — the pipeline's on-device
-    sampler is greedy, reference `sample_next_token` semantics.)"""
-    assert greedy, "pipeline sampling is greedy (device argmax)"
+    decode. Rank 0 returns the generated token ids
+    [global_batch, max_steps]; other ranks return None. Greedy device
+    argmax by default (reference `sample_next_token` semantics);
+    temperature > 0 switches the last rank to on-device softmax
+    sampling (repetition-penalty sampling lives in the TCP client path,
+    engine/sampler.py)."""
+    assert greedy or temperature > 0.0, \
+        "non-greedy decode needs temperature > 0"
+    pipe.temperature = float(temperature)  # > 0 wins over greedy
     pipe.prime(prompt_ids)
     out = []
     for _ in range(max_steps):
