@@ -287,3 +287,43 @@ def test_midsize_parity(ftype):
                          decode=True).cpu()
     y_cpu = cpu2.forward(x.clone(), pos, seq)
     _assert_close(y_gpu, y_cpu, label="midsize fused decode")
+
+
+@pytest.mark.gpu
+def test_multi_stream_clones_match_shared_engine_on_gpu():
+    """bench.py's single-GPU multi-stream mode: per-micro-batch
+    weight-sharing clones, each stepped on its own HIP stream, must
+    decode the same tokens as one shared engine holding every
+    micro-batch's KV slots (no cross-stream interference)."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.models.llama import PRESETS
+    from distributedllm_amd.parallel.pipeline import (
+        DecodePipeline, PipelineConfig)
+    hp = PRESETS["tiny"].hparams(ggml.FTYPE_MOSTLY_Q4_0)
+    mbs, n_mb, steps = 2, 3, 4
+
+    def seed_tokens(pipe):
+        for m in range(n_mb):
+            pipe.tok[m] = torch.tensor([5 + m, 9 + 2 * m],
+                                       dtype=torch.int32, device="cuda")
+
+    # reference: one engine, partitioned KV slots, sequential micro-batches
+    ref_eng = HIPSliceEngine.random(hp, n_layers=hp.n_layer, n_ctx=32,
+                                    max_batch=mbs * n_mb, seed=0,
+                                    with_extra=True)
+    cfg = PipelineConfig(mbs=mbs, n_mb=n_mb, device="cuda")
+    ref = DecodePipeline(ref_eng, cfg, rank=0, world=1)
+    seed_tokens(ref)
+    ref.run_steps(steps)
+    torch.cuda.synchronize()
+    want = ref.current_tokens().tolist()
+
+    # same weights (seed=0), clones + concurrent streams
+    eng = HIPSliceEngine.random(hp, n_layers=hp.n_layer, n_ctx=32,
+                                max_batch=mbs, seed=0, with_extra=True)
+    engines = [eng] + [eng.clone_shared() for _ in range(n_mb - 1)]
+    pipe = DecodePipeline(eng, cfg, rank=0, world=1, engines=engines)
+    seed_tokens(pipe)
+    pipe.run_steps(steps)
+    torch.cuda.synchronize()
+    assert pipe.current_tokens().tolist() == want
