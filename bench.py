@@ -46,11 +46,11 @@ def main() -> int:
                          "decode is HBM-bound so tokens/s scales ~linearly "
                          "with batch)")
     ap.add_argument("--ctx", type=int, default=2048)
-    ap.add_argument("--single-gpu-mbs", type=int, default=4,
+    ap.add_argument("--single-gpu-mbs", type=int, default=5,
                     help="micro-batches (= concurrent HIP streams) on a "
                          "single GPU; weights are shared between them "
-                         "(auto-capped by KV memory; 4 measured best on "
-                         "3B: +34%% over one stream)")
+                         "(auto-capped by HBM fit; 5 measured best on "
+                         "3B: +47%% over one stream, 6 regresses)")
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
 
