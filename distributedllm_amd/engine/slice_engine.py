@@ -159,6 +159,11 @@ class HIPSliceEngine:
             eps=RMS_EPS, rope_base=ROPE_BASE)
         self.device = "cuda"
         self.has_extra = False
+        # weight-tensor references for clone_shared (filled by .random()
+        # or load_layers/attach_extra; the C++ engine holds references to
+        # the same device tensors, so clones share HBM)
+        self._layers_cache = None
+        self._extra_cache = None
 
     @classmethod
     def from_ggml(cls, f: ggml.GGMLFile, n_ctx: int = 2048,
@@ -267,7 +272,7 @@ class HIPSliceEngine:
         micro-batches run concurrently on separate HIP streams without
         duplicating the model."""
         assert getattr(self, "_layers_cache", None) is not None, \
-            "clone_shared requires an engine built by .random()"
+            "clone_shared requires an engine built by .random()/.from_ggml"
         twin = HIPSliceEngine(self.hp, self.n_layers, self.first_layer,
                               self.n_ctx, self.max_batch)
         for li, (an, fn, mats) in enumerate(self._layers_cache):
@@ -279,6 +284,7 @@ class HIPSliceEngine:
 
     def load_layers(self, f: ggml.GGMLFile) -> None:
         tm = f.tensor_map()
+        self._layers_cache = []
         for li in range(self.n_layers):
             gi = li + self.first_layer
             pre = f"layers.{gi}."
@@ -292,6 +298,7 @@ class HIPSliceEngine:
                        "feed_forward.w1.weight", "feed_forward.w2.weight",
                        "feed_forward.w3.weight"):
                 mats.append(repack_mfma(tm[pre + nm], self.device))
+            self._layers_cache.append((attn_norm, ffn_norm, mats))
             self._eng.set_layer(li, attn_norm, ffn_norm, mats)
 
     def attach_extra(self, f: ggml.GGMLFile) -> None:
@@ -302,8 +309,9 @@ class HIPSliceEngine:
                                           self.device)
         out_d, out_s, out_t = repack_mfma(tm["output.weight"], self.device)
         norm = torch.from_numpy(tm["norm.weight"].to_f32()).to(self.device)
-        self._eng.set_extra(tok_d, tok_s, tok_t, norm, out_d, out_s, out_t,
-                            self.hp.n_vocab)
+        self._extra_cache = (tok_d, tok_s, tok_t, norm, out_d, out_s,
+                             out_t, self.hp.n_vocab)
+        self._eng.set_extra(*self._extra_cache)
         self.has_extra = True
 
     def forward(self, x: torch.Tensor, pos: torch.Tensor,

@@ -327,3 +327,26 @@ def test_multi_stream_clones_match_shared_engine_on_gpu():
     pipe.run_steps(steps)
     torch.cuda.synchronize()
     assert pipe.current_tokens().tolist() == want
+
+
+@pytest.mark.gpu
+def test_clone_shared_from_checkpoint():
+    """clone_shared also works for engines loaded from GGML bytes (real
+    multi-stream serving, not just synthetic benches): the clone shares
+    the weight tensors and reproduces the original's forward exactly."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.formats import slicer, synthetic
+    f = synthetic.build_model("tiny", seed=3)
+    ex = slicer.make_extra_layers(f)
+    eng = HIPSliceEngine.from_ggml(f, n_ctx=32, max_batch=2)
+    eng.attach_extra(ex)
+    twin = eng.clone_shared()
+    assert twin.has_extra
+    x = torch.randn(2, f.hparams.n_embd, device="cuda") * 0.5
+    pos = torch.zeros(2, dtype=torch.int32, device="cuda")
+    seq = torch.arange(2, dtype=torch.int32, device="cuda")
+    y0 = eng.forward(x.clone(), pos, seq, decode=True)
+    y1 = twin.forward(x.clone(), pos, seq, decode=True)
+    assert torch.equal(y0, y1)
+    assert torch.equal(eng.logits(y0, all_logits=True),
+                       twin.logits(y1, all_logits=True))
