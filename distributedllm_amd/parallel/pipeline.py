@@ -270,11 +270,14 @@ class DecodePipeline:
 
     def prime(self, prompt_ids) -> None:
         """Prefill every sequence with the same prompt (the batched-
-        serving demo semantics; per-sequence prompts are a scheduling
-        feature, not an engine one). Processes the prompt in <=64-token
-        tiles through the pipeline stages eagerly, leaving each
-        sequence's KV at position len(prompt) and tok at the last prompt
-        token, ready for run_steps/generate."""
+        serving demo semantics; per-sequence prompts live in
+        serving.ContinuousBatcher). Processes prompt[:-1] in <=64-token
+        tiles through the pipeline stages eagerly and leaves tok at the
+        LAST prompt token with pos = len(prompt)-1, so the first decode
+        step evaluates that token at its true position and samples the
+        true continuation (the TCP client's `generate` semantics —
+        feeding the last token again at a new position would condition
+        generation on a duplicated token)."""
         dev = self.cfg.device
         prompt = torch.as_tensor(prompt_ids, dtype=torch.int32, device=dev)
         Tp = int(prompt.numel())
@@ -283,8 +286,8 @@ class DecodePipeline:
         E = self.engine.hp.n_embd
         tile = max(1, 64 // mbs)  # prompt positions per pipeline hop
         for m in range(self.cfg.n_mb):
-            for p0 in range(0, Tp, tile):
-                p1 = min(Tp, p0 + tile)
+            for p0 in range(0, Tp - 1, tile):
+                p1 = min(Tp - 1, p0 + tile)
                 n = p1 - p0
                 # token-major layout: [n positions x mbs sequences]
                 pos = (torch.arange(p0, p1, dtype=torch.int32, device=dev)
@@ -300,7 +303,7 @@ class DecodePipeline:
                 y = self.engine.forward(x, pos, seq)
                 if not self.is_last:
                     dist.send(y, dst=self.rank + 1)
-            self.pos[m].fill_(Tp)
+            self.pos[m].fill_(Tp - 1)
             self.tok[m].fill_(int(prompt[-1].item()))
 
 

@@ -218,3 +218,37 @@ def test_pipeline_generate_temperature_sampling():
     assert all(0 <= t < V for row in a for t in row)
     assert a == run(0.8)              # same seed -> same sample
     assert run(1e-4) == run(0.0)      # temp -> 0 degenerates to greedy
+
+
+def test_prime_matches_reference_generation_semantics():
+    """pipeline_generate must equal the canonical loop (the TCP client's
+    semantics): prefill the prompt, sample from the LAST prompt
+    position's logits, then feed only sampled tokens — no duplicated
+    last-prompt-token position in KV."""
+    from distributedllm_amd.parallel.pipeline import pipeline_generate
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    prompt, steps = [5, 9, 3], 4
+
+    # canonical single-sequence loop
+    eng = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=1)
+    eng.attach_extra(ex)
+    want = []
+    cur, n_past = list(prompt), 0
+    for _ in range(steps):
+        toks = torch.tensor(cur, dtype=torch.int32)
+        pos = torch.arange(n_past, n_past + len(cur), dtype=torch.int32)
+        seq = torch.zeros(len(cur), dtype=torch.int32)
+        y = eng.forward(eng.embed(toks), pos, seq)
+        lg = eng.logits(y[-1:].contiguous(), all_logits=True)
+        tid = int(torch.argmax(lg[0]).item())
+        want.append(tid)
+        n_past += len(cur)
+        cur = [tid]
+
+    eng2 = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=MBS)
+    eng2.attach_extra(ex)
+    cfg = PipelineConfig(mbs=MBS, n_mb=1, device="cpu")
+    pipe = DecodePipeline(eng2, cfg, rank=0, world=1)
+    got = pipeline_generate(pipe, prompt, max_steps=steps).tolist()
+    assert got[0] == want and got[1] == want
