@@ -1,0 +1,162 @@
+"""Continuous batching over one slice engine.
+
+The reference serves exactly one request at a time (its `generate` loop
+holds the whole cluster, /root/reference/distllm/cli_api/common.py:94-111,
+and node forwards are serialized). This layer is what the MI355X engine's
+batched-decode design (JT column tiles, per-sequence KV slots, explicit
+pos/seq on every forward — engine_ext.cpp) was built for: many requests
+with different prompts and lengths share one decode step, new requests
+are admitted into free KV slots the moment one finishes, and every
+decode step advances every active request by one token.
+
+Semantics per request match the TCP client's `generate`
+(cluster/llm_client.py): prefill `prompt[:-1]`, then decode the last
+prompt token at its true position and sample the continuation — greedy
+through the engine's device argmax, or per-request host `Sampler`
+(reference parity, including repetition penalty over that request's
+sampled ids only).
+"""
+from __future__ import annotations
+
+from collections import deque
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional, Sequence
+
+import torch
+
+from ..engine.sampler import Sampler
+
+
+@dataclass
+class Request:
+    rid: int
+    prompt: List[int]
+    max_new: int
+    sampler: Optional[Sampler] = None  # None => greedy device argmax
+    eos_id: Optional[int] = None
+    out: List[int] = field(default_factory=list)
+    done: bool = False
+    slot: int = -1        # KV slot while active
+    _next_tok: int = -1   # token to feed at _pos on the next step
+    _pos: int = -1
+
+
+class ContinuousBatcher:
+    """Slot-based scheduler: submit() any time, step() advances every
+    active request one token and admits queued requests into free slots.
+
+    A finished request's KV slot is reused immediately — safe because
+    the engine's attention streams exactly rows [0, pos) of a slot, so
+    a new sequence starting at pos 0 never sees the old one's rows.
+    """
+
+    def __init__(self, engine, max_slots: Optional[int] = None,
+                 eos_id: Optional[int] = None):
+        self.engine = engine
+        slots = int(getattr(engine, "max_batch", 1))
+        if max_slots is not None:
+            slots = min(slots, max_slots)
+        # one decode step feeds one token per active request; keep the
+        # step within a single fused-decode kernel launch where possible
+        self.n_slots = max(1, slots)
+        self.eos_id = eos_id
+        self.free: List[int] = list(range(self.n_slots))
+        self.active: Dict[int, Request] = {}   # slot -> request
+        self.queue: deque[Request] = deque()
+        self._next_rid = 0
+        self._dev = getattr(engine, "device", "cpu")
+
+    # ------------------------------------------------------------- intake
+
+    def submit(self, prompt_ids: Sequence[int], max_new: int,
+               sampler: Optional[Sampler] = None,
+               eos_id: Optional[int] = None) -> Request:
+        assert len(prompt_ids) >= 1 and max_new >= 1
+        r = Request(rid=self._next_rid, prompt=list(map(int, prompt_ids)),
+                    max_new=max_new, sampler=sampler,
+                    eos_id=self.eos_id if eos_id is None else eos_id)
+        self._next_rid += 1
+        self.queue.append(r)
+        return r
+
+    def _admit(self) -> None:
+        eng = self.engine
+        while self.free and self.queue:
+            r = self.queue.popleft()
+            r.slot = self.free.pop()
+            n_ctx = getattr(eng, "n_ctx", 1 << 30)
+            assert len(r.prompt) + r.max_new <= n_ctx, \
+                f"request {r.rid}: prompt+max_new exceeds n_ctx={n_ctx}"
+            body = r.prompt[:-1]
+            if body:  # prefill all but the last prompt token
+                toks = torch.tensor(body, dtype=torch.int32,
+                                    device=self._dev)
+                pos = torch.arange(len(body), dtype=torch.int32,
+                                   device=self._dev)
+                seq = torch.full((len(body),), r.slot, dtype=torch.int32,
+                                 device=self._dev)
+                eng.forward(eng.embed(toks), pos, seq)
+            r._next_tok = r.prompt[-1]
+            r._pos = len(r.prompt) - 1
+            self.active[r.slot] = r
+
+    # --------------------------------------------------------------- step
+
+    def step(self) -> List[Request]:
+        """Admit what fits, advance every active request one token;
+        returns the requests that finished on this step."""
+        self._admit()
+        if not self.active:
+            return []
+        slots = sorted(self.active)
+        reqs = [self.active[s] for s in slots]
+        toks = torch.tensor([r._next_tok for r in reqs],
+                            dtype=torch.int32, device=self._dev)
+        pos = torch.tensor([r._pos for r in reqs], dtype=torch.int32,
+                           device=self._dev)
+        seq = torch.tensor(slots, dtype=torch.int32, device=self._dev)
+        eng = self.engine
+        y = eng.forward(eng.embed(toks), pos, seq, decode=True)
+        lg = eng.logits(y, all_logits=True)
+        greedy_ids = None
+        if any(r.sampler is None for r in reqs):
+            greedy_ids = eng.argmax(lg)
+            if greedy_ids.device.type != "cpu":
+                greedy_ids = greedy_ids.cpu()
+        lg_host = None
+        if any(r.sampler is not None for r in reqs):
+            lg_host = lg.float().cpu().numpy()
+
+        finished: List[Request] = []
+        for i, r in enumerate(reqs):
+            if r.sampler is None:
+                tid = int(greedy_ids[i])
+            else:
+                tid = r.sampler(lg_host[i])
+            r.out.append(tid)
+            r._next_tok = tid
+            r._pos += 1
+            if len(r.out) >= r.max_new or (r.eos_id is not None and
+                                           tid == r.eos_id):
+                r.done = True
+                del self.active[r.slot]
+                self.free.append(r.slot)
+                r.slot = -1
+                finished.append(r)
+        return finished
+
+    # --------------------------------------------------------- convenience
+
+    @property
+    def pending(self) -> int:
+        return len(self.queue) + len(self.active)
+
+    def run_all(self, max_steps: int = 1 << 20) -> List[Request]:
+        """Drive step() until every submitted request finishes."""
+        out: List[Request] = []
+        for _ in range(max_steps):
+            if not self.pending:
+                break
+            out.extend(self.step())
+        assert not self.pending, "run_all hit max_steps with work pending"
+        return out
