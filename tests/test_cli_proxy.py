@@ -39,9 +39,11 @@ def _write_dummy(tmp_path, k=2.0, b=1.0):
 
 def test_all_commands_registered():
     from distributedllm_amd.cli.base import commands
+    # the reference's 9 commands plus the batched-serving extension
     assert set(commands) == {
         "provision", "run_node", "status", "push_slice", "load_slice",
-        "list_slices", "generate_text", "perplexity", "run_proxy"}
+        "list_slices", "generate_text", "perplexity", "run_proxy",
+        "batch_generate"}
     build_parser()  # parser builds without error
 
 

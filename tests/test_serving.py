@@ -97,3 +97,23 @@ def test_context_overflow_rejected():
     bat.submit([1] * 30, 10)  # 40 > n_ctx=32
     with pytest.raises(AssertionError):
         bat.run_all(max_steps=5)
+
+
+def test_batch_generate_cli(tmp_path):
+    """End-to-end: the batch_generate command serves several prompts
+    concurrently from a model file and reports throughput."""
+    import subprocess
+    import sys
+    p = tmp_path / "tiny.bin"
+    synthetic.build_model("tiny", seed=0).save(str(p))
+    pf = tmp_path / "prompts.txt"
+    pf.write_text("hello\nworld\nagain\n")
+    out = subprocess.run(
+        [sys.executable, "manager.py", "batch_generate", str(p),
+         "--prompts-file", str(pf), "--num-tokens", "5", "--greedy",
+         "--ctx", "64", "--slots", "2"],
+        capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-1500:]
+    lines = out.stdout.strip().splitlines()
+    assert len(lines) == 3 and all("->" in ln for ln in lines)
+    assert "3 requests, 15 tokens" in out.stderr
