@@ -73,43 +73,46 @@ def main() -> int:
     parts = partition_layers(preset.n_layer, world)
     first, count = parts[rank]
 
-    # micro-batches in flight = number of stages (keeps every stage
-    # busy); on a single GPU, micro-batches run on separate HIP streams
-    # over SHARED weights, overlapping the latency-bound kernel chains.
+    # Concurrency structure:
+    #  * world > 1: one pipeline lane = `world` micro-batches in flight
+    #    (keeps every stage busy); k lanes run concurrently per rank on
+    #    separate HIP streams over SHARED weights (weight-sharing engine
+    #    clones, one per lane), so each GPU gets the same multi-stream
+    #    overlap the single-GPU config has.
+    #  * world == 1: k lanes of one micro-batch each.
     # Weak scaling: global batch grows with N, per-GPU work fixed.
-    n_mb = world if world > 1 else args.single_gpu_mbs
-    if world == 1 and n_mb > 1:
-        # each clone owns a full [L, mbs, ctx, E] f16 K+V pair (weights
-        # are shared); 2 streams measured a net LOSS on every model
-        # (contention without enough overlap) while 3-4 won, so use
-        # multi-stream only when >=3 KV sets fit in HBM next to the
-        # model weights (~260 GB usable of 288)
-        kv_bytes = (preset.n_layer * args.mbs * args.ctx *
-                    preset.n_embd * 4)
+    # Lane count: 2 streams measured a net LOSS on every model
+    # (contention without enough overlap) while 3-5 won, so multi-stream
+    # only when >=3 lanes' KV sets fit in HBM next to the model weights
+    # (~260 GB usable of 288). All ranks must agree on k: size with the
+    # largest per-rank slice (rank 0 holds the most layers).
+    n_lanes = args.single_gpu_mbs
+    if n_lanes > 1:
+        max_layers = parts[0][1]
+        kv_bytes = (max_layers * world * args.mbs * args.ctx *
+                    preset.n_embd * 4)  # per-lane clone, f16 K+V
         E, F, V = preset.n_embd, preset.n_ff, 32000
-        n_weights = (preset.n_layer * (4 * E * E + 3 * E * F) +
-                     2 * V * E)
+        n_weights = (max_layers * (4 * E * E + 3 * E * F) + 2 * V * E)
         bpw = {"q4_0": 0.5625, "q4_1": 0.625, "f16": 2.0,
                "f32": 4.0}[args.ftype]
         w_bytes = n_weights * bpw * 1.1  # repack padding/scales margin
         fit = max(1, int((260e9 - w_bytes) // max(kv_bytes, 1)))
-        n_mb = min(n_mb, fit) if fit >= 3 else 1
+        n_lanes = min(n_lanes, fit) if fit >= 3 else 1
+    n_mb = n_lanes * max(world, 1) if world > 1 else n_lanes
     cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
 
     engines = None
     if device == "cuda":
         from distributedllm_amd.engine import HIPSliceEngine
-        # world>1 shares ONE engine across the n_mb in-flight micro-
-        # batches, so its KV cache must hold every micro-batch's slots
-        # (the pipeline assigns sequence ids [0, n_mb*mbs)); single-GPU
-        # multi-stream instead gives each micro-batch its own clone with
-        # LOCAL ids, so mbs slots per engine suffice.
-        slots = args.mbs * (n_mb if world > 1 else 1)
+        # each lane clone holds KV for its lane's in-flight micro-
+        # batches: `world` of them per lane (1 on a single GPU)
+        slots = args.mbs * (n_mb // n_lanes)
         eng = HIPSliceEngine.random(hp, n_layers=count, first_layer=first,
                                     n_ctx=args.ctx, max_batch=slots,
                                     seed=args.seed, with_extra=True)
-        if world == 1 and n_mb > 1:
-            engines = [eng] + [eng.clone_shared() for _ in range(n_mb - 1)]
+        if n_lanes > 1:
+            engines = [eng] + [eng.clone_shared()
+                               for _ in range(n_lanes - 1)]
     else:  # CPU fallback so the contract is testable without a GPU
         from distributedllm_amd.engine import TorchSliceEngine
         from distributedllm_amd.formats import synthetic

@@ -56,9 +56,12 @@ class DecodePipeline:
 
     def __init__(self, engine, cfg: PipelineConfig,
                  rank: int = 0, world: int = 1, engines=None):
-        """engines: optional per-micro-batch engine list (weight-sharing
-        clones) — single-GPU mode then runs each micro-batch on its own
-        HIP stream, overlapping the latency-bound kernel chains."""
+        """engines: optional list of k <= n_mb weight-sharing engine
+        clones ("stream lanes"): micro-batch m runs on engines[m % k]
+        and, on CUDA, on that lane's own HIP stream — concurrent lanes
+        overlap the latency-bound kernel chains (single- AND multi-GPU;
+        micro-batches of one lane share scratch, so a lane is
+        stream-serialized by construction)."""
         self.engine = engine
         self.engines = engines  # None => self.engine for every mb
         self.cfg = cfg
@@ -69,14 +72,18 @@ class DecodePipeline:
         dev = cfg.device
         E = engine.hp.n_embd
         M, mbs = cfg.n_mb, cfg.mbs
+        self.n_lanes = len(engines) if engines is not None else 1
         self.pos = [torch.zeros(mbs, dtype=torch.int32, device=dev)
                     for _ in range(M)]
-        # per-mb engines have their OWN KV caches of mbs slots -> local
-        # sequence ids; a shared engine partitions its slots across mbs
+        # lane clones have their OWN KV caches: micro-batch m is the
+        # (m // k)-th of its lane, so its slot range is local to the
+        # clone; a single shared engine partitions slots across all mbs
         if engines is not None:
-            need = mbs
-            self.seq = [torch.arange(mbs, dtype=torch.int32, device=dev)
-                        for _ in range(M)]
+            k = self.n_lanes
+            need = ((M + k - 1) // k) * mbs
+            self.seq = [torch.arange((m // k) * mbs, (m // k + 1) * mbs,
+                                     dtype=torch.int32, device=dev)
+                        for m in range(M)]
         else:
             need = M * mbs
             self.seq = [torch.arange(m * mbs, (m + 1) * mbs,
@@ -89,9 +96,9 @@ class DecodePipeline:
             if cap < need:
                 raise ValueError(
                     f"engine KV holds {cap} sequence slots but the "
-                    f"pipeline assigns ids [0, {need}) "
-                    f"(n_mb={M} x mbs={mbs}); build the engine with "
-                    f"max_batch >= {need}")
+                    f"pipeline assigns ids [0, {need}) per lane "
+                    f"(n_mb={M} x mbs={mbs}, {self.n_lanes} lanes); "
+                    f"build the engine with max_batch >= {need}")
         # zero-init: mid-stage graph-capture warmup runs before any real
         # activations arrive; empty-buffer garbage could seed NaNs into
         # the KV cache rows written during warmup
@@ -108,7 +115,9 @@ class DecodePipeline:
     # ------------------------------------------------------ hipGraph mode
 
     def _eng(self, m: int):
-        return self.engines[m] if self.engines is not None else self.engine
+        if self.engines is None:
+            return self.engine
+        return self.engines[m % self.n_lanes]
 
     def _mb_compute(self, m: int):
         """The capturable (comm-free) compute of micro-batch m.
@@ -209,20 +218,22 @@ class DecodePipeline:
         is fully serialized — SURVEY §2.3; the north-star design point is
         comm/compute overlap across pipeline stages)."""
         n_mb = self.cfg.n_mb
+        use_streams = (self.engines is not None and
+                       self.cfg.device == "cuda" and self.n_lanes > 1)
+        if use_streams and self._streams is None:
+            self._streams = [torch.cuda.Stream()
+                             for _ in range(self.n_lanes)]
         if self.world == 1 or steps == 0:
-            if (self.engines is not None and self.cfg.device == "cuda"
-                    and n_mb > 1):
-                # single-GPU multi-stream: each micro-batch's step chain
-                # lives on its own stream; chains are independent (per-mb
+            if use_streams:
+                # single-GPU multi-stream: each lane's step chain lives
+                # on its own stream; lanes are independent (per-lane
                 # engines/KV/tokens), so the whole schedule is enqueued
                 # with no cross-stream syncs and the streams fill each
                 # other's latency stalls.
-                if self._streams is None:
-                    self._streams = [torch.cuda.Stream()
-                                     for _ in range(n_mb)]
                 for _ in range(steps):
                     for m in range(n_mb):
-                        with torch.cuda.stream(self._streams[m]):
+                        with torch.cuda.stream(
+                                self._streams[m % self.n_lanes]):
                             self._compute(m)
                 for st in self._streams:
                     torch.cuda.current_stream().wait_stream(st)
@@ -231,6 +242,20 @@ class DecodePipeline:
                 for m in range(n_mb):
                     self._advance_mb(m)
             return
+
+        # multi-GPU overlapped driver. With lanes, micro-batch m's
+        # compute AND its NCCL ops run under lane m%k's stream (the
+        # NCCL work is ordered against the current stream at call
+        # time), so k lanes overlap on the GPU while the host-side
+        # posting order keeps every (src, dst) send/recv sequence
+        # deterministic — torch P2P has no tags, matching is by order.
+        from contextlib import nullcontext
+
+        def lane_ctx(m):
+            if use_streams:
+                return torch.cuda.stream(self._streams[m % self.n_lanes])
+            return nullcontext()
+
         recvs: List = [None] * n_mb
         sends: List = [None] * n_mb
         # prime: mid/last stages post receives for every micro-batch;
@@ -238,23 +263,26 @@ class DecodePipeline:
         # posted only after it has sent work downstream.
         if not self.is_first:
             for m in range(n_mb):
-                recvs[m] = self._post_recv(m)
+                with lane_ctx(m):
+                    recvs[m] = self._post_recv(m)
         for s in range(steps):
             for m in range(n_mb):
-                if recvs[m] is not None:
-                    recvs[m].wait()
-                if sends[m] is not None:
-                    sends[m].wait()  # our out buffer is being re-written
-                out = self._compute(m)
-                dst = 0 if self.is_last else self.rank + 1
-                sends[m] = dist.isend(out, dst=dst)
-                # mid/last stages receive exactly `steps` activation
-                # blocks — do not post one past the end (it would never be
-                # matched); rank 0 must still drain the final token sends.
-                if self.is_first or s + 1 < steps:
-                    recvs[m] = self._post_recv(m)
-                else:
-                    recvs[m] = None
+                with lane_ctx(m):
+                    if recvs[m] is not None:
+                        recvs[m].wait()
+                    if sends[m] is not None:
+                        sends[m].wait()  # out buffer is being re-written
+                    out = self._compute(m)
+                    dst = 0 if self.is_last else self.rank + 1
+                    sends[m] = dist.isend(out, dst=dst)
+                    # mid/last stages receive exactly `steps` activation
+                    # blocks — do not post one past the end (it would
+                    # never be matched); rank 0 must still drain the
+                    # final token sends.
+                    if self.is_first or s + 1 < steps:
+                        recvs[m] = self._post_recv(m)
+                    else:
+                        recvs[m] = None
         for m in range(n_mb):
             if sends[m] is not None:
                 sends[m].wait()
@@ -262,6 +290,9 @@ class DecodePipeline:
             # is quiescent between run_steps calls
             if recvs[m] is not None:
                 recvs[m].wait()
+        if use_streams:
+            for st in self._streams:
+                torch.cuda.current_stream().wait_stream(st)
 
     def current_tokens(self) -> torch.Tensor:
         return torch.stack(self.tok)

@@ -26,7 +26,7 @@ def test_partition_layers():
     assert partition_layers(4, 2) == [(0, 2), (2, 2)]
 
 
-SEEDS = [[5, 9], [11, 3]]  # starting tokens per micro-batch
+SEEDS = [[5, 9], [11, 3], [4, 8], [13, 2]]  # start tokens per micro-batch
 
 
 def _seed_tokens(pipe, n_mb):
@@ -46,7 +46,7 @@ def _single_reference_tokens(n_mb=1):
     return pipe.current_tokens().tolist()
 
 
-def _rank_main(rank, world, port, n_mb, q):
+def _rank_main(rank, world, port, n_mb, q, n_lanes=1):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -56,10 +56,23 @@ def _rank_main(rank, world, port, n_mb, q):
     parts = partition_layers(f.hparams.n_layer, world)
     first, count = parts[rank]
     sl = slicer.make_slice(f, first, first + count - 1)
-    eng = TorchSliceEngine.from_ggml(sl, n_ctx=32, max_batch=MBS * n_mb)
-    eng.attach_extra(ex)
+    per_lane = MBS * ((n_mb + n_lanes - 1) // n_lanes)
+    engines = None
+    if n_lanes > 1:
+        engines = []
+        for _ in range(n_lanes):
+            e = TorchSliceEngine.from_ggml(sl, n_ctx=32,
+                                           max_batch=per_lane)
+            e.attach_extra(ex)
+            engines.append(e)
+        eng = engines[0]
+    else:
+        eng = TorchSliceEngine.from_ggml(sl, n_ctx=32,
+                                         max_batch=MBS * n_mb)
+        eng.attach_extra(ex)
     cfg = PipelineConfig(mbs=MBS, n_mb=n_mb, device="cpu")
-    pipe = DecodePipeline(eng, cfg, rank=rank, world=world)
+    pipe = DecodePipeline(eng, cfg, rank=rank, world=world,
+                          engines=engines)
     _seed_tokens(pipe, n_mb)
     pipe.run_steps(STEPS)
     if rank == 0:
@@ -67,10 +80,11 @@ def _rank_main(rank, world, port, n_mb, q):
     dist.destroy_process_group()
 
 
-def _run_cluster(world, port, n_mb):
+def _run_cluster(world, port, n_mb, n_lanes=1):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main, args=(r, world, port, n_mb, q))
+    procs = [ctx.Process(target=_rank_main,
+                         args=(r, world, port, n_mb, q, n_lanes))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -147,6 +161,15 @@ def test_multi_engine_micro_batches_match_shared_engine():
     _seed_tokens(pipe, n_mb)
     pipe.run_steps(STEPS)
     assert pipe.current_tokens().tolist() == _single_reference_tokens(n_mb)
+
+
+@pytest.mark.timeout(120)
+def test_two_stage_two_lane_pipeline_matches_single():
+    """2 ranks x 2 lanes (weight-sharing clones per rank, n_mb=4):
+    micro-batch m runs on lane m%2 with lane-local KV slots; tokens
+    must match one engine running all 4 micro-batches."""
+    assert _run_cluster(2, 29534, 4, n_lanes=2) == \
+        _single_reference_tokens(4)
 
 
 @pytest.mark.timeout(120)

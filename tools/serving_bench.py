@@ -4,7 +4,12 @@ prompts served concurrently through serving.ContinuousBatcher on one
 engine — measures end-to-end serving tokens/s including scheduling,
 logits and sampling (bench.py measures the raw decode pipeline)."""
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
 
 import torch
 
