@@ -19,6 +19,7 @@ sampled ids only).
 from __future__ import annotations
 
 from collections import deque
+from contextlib import nullcontext as _nullctx
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence
 
@@ -51,13 +52,20 @@ class ContinuousBatcher:
     """
 
     def __init__(self, engine, max_slots: Optional[int] = None,
-                 eos_id: Optional[int] = None):
+                 eos_id: Optional[int] = None, engines=None):
+        """engines: optional list of k weight-sharing clones ("lanes");
+        global slot s lives on lane s%k as that clone's local slot s//k,
+        and on CUDA each lane's forward runs on its own HIP stream —
+        the decode chain is latency-bound, so concurrent lanes overlap
+        (same mechanics as the pipeline's stream lanes)."""
         self.engine = engine
-        slots = int(getattr(engine, "max_batch", 1))
+        self.lanes = engines if engines else [engine]
+        k = len(self.lanes)
+        per_lane = min(int(getattr(e, "max_batch", 1))
+                       for e in self.lanes)
+        slots = per_lane * k
         if max_slots is not None:
             slots = min(slots, max_slots)
-        # one decode step feeds one token per active request; keep the
-        # step within a single fused-decode kernel launch where possible
         self.n_slots = max(1, slots)
         self.eos_id = eos_id
         self.free: List[int] = list(range(self.n_slots))
@@ -65,6 +73,13 @@ class ContinuousBatcher:
         self.queue: deque[Request] = deque()
         self._next_rid = 0
         self._dev = getattr(engine, "device", "cpu")
+        self._streams = None
+        if k > 1 and self._dev == "cuda":
+            self._streams = [torch.cuda.Stream() for _ in self.lanes]
+
+    def _lane(self, slot: int):
+        k = len(self.lanes)
+        return self.lanes[slot % k], slot // k
 
     # ------------------------------------------------------------- intake
 
@@ -80,7 +95,7 @@ class ContinuousBatcher:
         return r
 
     def _admit(self) -> None:
-        eng = self.engine
+        eng = self.engine  # n_ctx/bounds source; lanes share hparams
         while self.free and self.queue:
             r = self.queue.popleft()
             r.slot = self.free.pop()
@@ -89,13 +104,14 @@ class ContinuousBatcher:
                 f"request {r.rid}: prompt+max_new exceeds n_ctx={n_ctx}"
             body = r.prompt[:-1]
             if body:  # prefill all but the last prompt token
+                lane_eng, local = self._lane(r.slot)
                 toks = torch.tensor(body, dtype=torch.int32,
                                     device=self._dev)
                 pos = torch.arange(len(body), dtype=torch.int32,
                                    device=self._dev)
-                seq = torch.full((len(body),), r.slot, dtype=torch.int32,
+                seq = torch.full((len(body),), local, dtype=torch.int32,
                                  device=self._dev)
-                eng.forward(eng.embed(toks), pos, seq)
+                lane_eng.forward(lane_eng.embed(toks), pos, seq)
             r._next_tok = r.prompt[-1]
             r._pos = len(r.prompt) - 1
             self.active[r.slot] = r
@@ -108,26 +124,47 @@ class ContinuousBatcher:
         self._admit()
         if not self.active:
             return []
-        slots = sorted(self.active)
-        reqs = [self.active[s] for s in slots]
-        toks = torch.tensor([r._next_tok for r in reqs],
-                            dtype=torch.int32, device=self._dev)
-        pos = torch.tensor([r._pos for r in reqs], dtype=torch.int32,
-                           device=self._dev)
-        seq = torch.tensor(slots, dtype=torch.int32, device=self._dev)
-        eng = self.engine
-        y = eng.forward(eng.embed(toks), pos, seq, decode=True)
-        lg = eng.logits(y, all_logits=True)
-        greedy_ids = None
-        if any(r.sampler is None for r in reqs):
-            greedy_ids = eng.argmax(lg)
-            if greedy_ids.device.type != "cpu":
-                greedy_ids = greedy_ids.cpu()
-        lg_host = None
-        if any(r.sampler is not None for r in reqs):
-            lg_host = lg.float().cpu().numpy()
+        k = len(self.lanes)
+        # one decode launch per lane, each on its own stream (CUDA)
+        per_lane: List[List[int]] = [[] for _ in range(k)]
+        for s in sorted(self.active):
+            per_lane[s % k].append(s)
+        work = []  # (slots, reqs, greedy_ids, lg, lg_needed)
+        for j, lane_slots in enumerate(per_lane):
+            if not lane_slots:
+                continue
+            reqs = [self.active[s] for s in lane_slots]
+            eng = self.lanes[j]
+            ctx = (torch.cuda.stream(self._streams[j])
+                   if self._streams is not None else _nullctx())
+            with ctx:
+                toks = torch.tensor([r._next_tok for r in reqs],
+                                    dtype=torch.int32, device=self._dev)
+                pos = torch.tensor([r._pos for r in reqs],
+                                   dtype=torch.int32, device=self._dev)
+                seq = torch.tensor([s // k for s in lane_slots],
+                                   dtype=torch.int32, device=self._dev)
+                y = eng.forward(eng.embed(toks), pos, seq, decode=True)
+                lg = eng.logits(y, all_logits=True)
+                greedy_ids = None
+                if any(r.sampler is None for r in reqs):
+                    greedy_ids = eng.argmax(lg)
+            work.append((reqs, greedy_ids, lg))
+        if self._streams is not None:
+            for st in self._streams:
+                torch.cuda.current_stream().wait_stream(st)
 
         finished: List[Request] = []
+        for reqs, greedy_ids, lg in work:
+            if greedy_ids is not None and greedy_ids.device.type != "cpu":
+                greedy_ids = greedy_ids.cpu()
+            lg_host = None
+            if any(r.sampler is not None for r in reqs):
+                lg_host = lg.float().cpu().numpy()
+            self._advance(reqs, greedy_ids, lg_host, finished)
+        return finished
+
+    def _advance(self, reqs, greedy_ids, lg_host, finished) -> None:
         for i, r in enumerate(reqs):
             if r.sampler is None:
                 tid = int(greedy_ids[i])
@@ -143,7 +180,6 @@ class ContinuousBatcher:
                 self.free.append(r.slot)
                 r.slot = -1
                 finished.append(r)
-        return finished
 
     # --------------------------------------------------------- convenience
 

@@ -117,3 +117,22 @@ def test_batch_generate_cli(tmp_path):
     lines = out.stdout.strip().splitlines()
     assert len(lines) == 3 and all("->" in ln for ln in lines)
     assert "3 requests, 15 tokens" in out.stderr
+
+
+def test_multi_lane_batcher_matches_canonical():
+    """2 lanes (weight-sharing clone semantics; on CPU two engines over
+    the same GGML bytes): slot s lives on lane s%2, tokens must match
+    the one-at-a-time loop for every request."""
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    lanes = []
+    for _ in range(2):
+        e = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=2)
+        e.attach_extra(ex)
+        lanes.append(e)
+    bat = ContinuousBatcher(lanes[0], engines=lanes)
+    assert bat.n_slots == 4
+    reqs = [bat.submit(p, s) for p, s in zip(PROMPTS, STEPS)]
+    bat.run_all(max_steps=50)
+    for r, p, s in zip(reqs, PROMPTS, STEPS):
+        assert r.done and r.out == _canonical(p, s)

@@ -25,15 +25,23 @@ def main():
     ap.add_argument("--requests", type=int, default=256)
     ap.add_argument("--prompt-len", type=int, default=32)
     ap.add_argument("--num-tokens", type=int, default=64)
-    ap.add_argument("--slots", type=int, default=64)
+    ap.add_argument("--slots", type=int, default=64,
+                    help="KV slots per lane")
+    ap.add_argument("--lanes", type=int, default=5,
+                    help="concurrent stream lanes (weight-sharing "
+                         "clones); total slots = slots x lanes")
     ap.add_argument("--ctx", type=int, default=2048)
     args = ap.parse_args()
 
     hp = PRESETS[args.model].hparams(ggml.FTYPE_MOSTLY_Q4_0)
     eng = HIPSliceEngine.random(hp, n_layers=hp.n_layer, n_ctx=args.ctx,
                                 max_batch=args.slots, with_extra=True)
+    lanes = None
+    if args.lanes > 1:
+        lanes = [eng] + [eng.clone_shared()
+                         for _ in range(args.lanes - 1)]
     g = torch.Generator().manual_seed(1)
-    bat = ContinuousBatcher(eng, max_slots=args.slots)
+    bat = ContinuousBatcher(eng, engines=lanes)
     reqs = [bat.submit(torch.randint(3, hp.n_vocab, (args.prompt_len,),
                                      generator=g).tolist(),
                        args.num_tokens)
@@ -48,7 +56,8 @@ def main():
     dt = time.perf_counter() - t0
     total = sum(len(r.out) for r in reqs)
     print(f"{args.requests} requests x {args.num_tokens} new tokens "
-          f"(prompt {args.prompt_len}, {args.slots} slots): "
+          f"(prompt {args.prompt_len}, {bat.n_slots} slots on "
+          f"{len(bat.lanes)} lanes): "
           f"{total} tokens in {dt:.2f}s = {total/dt:.0f} tok/s, "
           f"{steps} decode steps")
 
