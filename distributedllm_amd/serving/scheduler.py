@@ -105,13 +105,20 @@ class ContinuousBatcher:
             body = r.prompt[:-1]
             if body:  # prefill all but the last prompt token
                 lane_eng, local = self._lane(r.slot)
-                toks = torch.tensor(body, dtype=torch.int32,
-                                    device=self._dev)
-                pos = torch.arange(len(body), dtype=torch.int32,
-                                   device=self._dev)
-                seq = torch.full((len(body),), local, dtype=torch.int32,
-                                 device=self._dev)
-                lane_eng.forward(lane_eng.embed(toks), pos, seq)
+                # prefill on the slot's lane stream: the lane's next
+                # decode launch must observe these KV writes, and
+                # same-stream ordering gives that without a sync
+                ctx = (torch.cuda.stream(
+                           self._streams[r.slot % len(self.lanes)])
+                       if self._streams is not None else _nullctx())
+                with ctx:
+                    toks = torch.tensor(body, dtype=torch.int32,
+                                        device=self._dev)
+                    pos = torch.arange(len(body), dtype=torch.int32,
+                                       device=self._dev)
+                    seq = torch.full((len(body),), local,
+                                     dtype=torch.int32, device=self._dev)
+                    lane_eng.forward(lane_eng.embed(toks), pos, seq)
             r._next_tok = r.prompt[-1]
             r._pos = len(r.prompt) - 1
             self.active[r.slot] = r
@@ -129,7 +136,7 @@ class ContinuousBatcher:
         per_lane: List[List[int]] = [[] for _ in range(k)]
         for s in sorted(self.active):
             per_lane[s % k].append(s)
-        work = []  # (slots, reqs, greedy_ids, lg, lg_needed)
+        work = []  # (reqs, greedy_ids, lg) per lane
         for j, lane_slots in enumerate(per_lane):
             if not lane_slots:
                 continue
