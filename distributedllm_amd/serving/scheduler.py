@@ -96,32 +96,47 @@ class ContinuousBatcher:
 
     def _admit(self) -> None:
         eng = self.engine  # n_ctx/bounds source; lanes share hparams
+        n_ctx = getattr(eng, "n_ctx", 1 << 30)
+        k = len(self.lanes)
+        # admissions grouped per lane, prefilled as ONE concatenated
+        # (token, pos, seq) stream per lane — the engine tiles any
+        # mixed-sequence stream internally, so several short prompts
+        # share each prefill launch instead of paying one kernel chain
+        # per request
+        per_lane: List[List[Request]] = [[] for _ in range(k)]
         while self.free and self.queue:
             r = self.queue.popleft()
             r.slot = self.free.pop()
-            n_ctx = getattr(eng, "n_ctx", 1 << 30)
             assert len(r.prompt) + r.max_new <= n_ctx, \
                 f"request {r.rid}: prompt+max_new exceeds n_ctx={n_ctx}"
-            body = r.prompt[:-1]
-            if body:  # prefill all but the last prompt token
-                lane_eng, local = self._lane(r.slot)
-                # prefill on the slot's lane stream: the lane's next
-                # decode launch must observe these KV writes, and
-                # same-stream ordering gives that without a sync
-                ctx = (torch.cuda.stream(
-                           self._streams[r.slot % len(self.lanes)])
-                       if self._streams is not None else _nullctx())
-                with ctx:
-                    toks = torch.tensor(body, dtype=torch.int32,
-                                        device=self._dev)
-                    pos = torch.arange(len(body), dtype=torch.int32,
-                                       device=self._dev)
-                    seq = torch.full((len(body),), local,
-                                     dtype=torch.int32, device=self._dev)
-                    lane_eng.forward(lane_eng.embed(toks), pos, seq)
             r._next_tok = r.prompt[-1]
             r._pos = len(r.prompt) - 1
             self.active[r.slot] = r
+            if len(r.prompt) > 1:
+                per_lane[r.slot % k].append(r)
+        for j, reqs in enumerate(per_lane):
+            if not reqs:
+                continue
+            toks, pos, seq = [], [], []
+            for r in reqs:
+                body = r.prompt[:-1]
+                toks += body
+                pos += list(range(len(body)))
+                seq += [r.slot // k] * len(body)
+            lane_eng = self.lanes[j]
+            # prefill on the lane's stream: the lane's next decode
+            # launch must observe these KV writes, and same-stream
+            # ordering gives that without a sync
+            ctx = (torch.cuda.stream(self._streams[j])
+                   if self._streams is not None else _nullctx())
+            with ctx:
+                t = torch.tensor(toks, dtype=torch.int32,
+                                 device=self._dev)
+                p = torch.tensor(pos, dtype=torch.int32,
+                                 device=self._dev)
+                q = torch.tensor(seq, dtype=torch.int32,
+                                 device=self._dev)
+                lane_eng.forward(lane_eng.embed(t), p, q)
 
     # --------------------------------------------------------------- step
 
