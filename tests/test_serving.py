@@ -136,3 +136,31 @@ def test_multi_lane_batcher_matches_canonical():
     bat.run_all(max_steps=50)
     for r, p, s in zip(reqs, PROMPTS, STEPS):
         assert r.done and r.out == _canonical(p, s)
+
+
+def test_random_arrivals_match_canonical():
+    """Randomized schedule stress: random prompts/lengths submitted in
+    waves against a 3-slot batcher must each decode exactly the
+    canonical continuation (slot reuse, queueing, mixed admission
+    prefills all exercised)."""
+    import random
+    rng = random.Random(0)
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    eng = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=3)
+    eng.attach_extra(ex)
+    bat = ContinuousBatcher(eng)
+    V = f.hparams.n_vocab
+    cases = []
+    for wave in range(3):
+        for _ in range(rng.randint(1, 3)):
+            prompt = [rng.randrange(3, V) for _ in
+                      range(rng.randint(1, 6))]
+            steps = rng.randint(1, 5)
+            cases.append((prompt, steps, bat.submit(prompt, steps)))
+        for _ in range(rng.randint(1, 4)):  # interleave decode steps
+            bat.step()
+    bat.run_all(max_steps=100)
+    for prompt, steps, r in cases:
+        assert r.done and r.out == _canonical(prompt, steps), \
+            (prompt, steps, r.out)
