@@ -8,9 +8,9 @@ JSON line with the whole-job tokens/sec.
 
 Metric: tokens generated per second across the whole node (every sequence
 advances one token per step; tokens/step = global_batch). Weak scaling:
-global batch = 64 sequences per pipeline stage, so per-GPU work per step
-is constant as N grows (each stage holds ~L/N layers but serves N micro-
-batches per step). Synthetic setup: random-init weights of the real
+global batch = 64 sequences per micro-batch x k stream lanes x N stages,
+so per-GPU work per step is constant as N grows (each stage holds ~L/N
+layers but serves N micro-batches per lane per step). Synthetic setup: random-init weights of the real
 architecture generated directly on-device in the engine's q4_0 layout
 (identical compute + HBM traffic to a real checkpoint; no network for real
 weights — BASELINE.md).
@@ -46,11 +46,13 @@ def main() -> int:
                          "decode is HBM-bound so tokens/s scales ~linearly "
                          "with batch)")
     ap.add_argument("--ctx", type=int, default=2048)
-    ap.add_argument("--single-gpu-mbs", type=int, default=5,
-                    help="micro-batches (= concurrent HIP streams) on a "
-                         "single GPU; weights are shared between them "
-                         "(auto-capped by HBM fit; 5 measured best on "
-                         "3B: +47%% over one stream, 6 regresses)")
+    ap.add_argument("--lanes", "--single-gpu-mbs", dest="lanes",
+                    type=int, default=5,
+                    help="concurrent stream lanes per GPU (weight-"
+                         "sharing engine clones, one HIP stream each; "
+                         "with N>1 GPUs each lane is a full pipeline). "
+                         "Auto-capped by HBM fit; 5 measured best on "
+                         "3B: +47%% over one stream, 6 regresses")
     ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
 
@@ -86,7 +88,7 @@ def main() -> int:
     # only when >=3 lanes' KV sets fit in HBM next to the model weights
     # (~260 GB usable of 288). All ranks must agree on k: size with the
     # largest per-rank slice (rank 0 holds the most layers).
-    n_lanes = args.single_gpu_mbs
+    n_lanes = args.lanes
     if n_lanes > 1:
         max_layers = parts[0][1]
         kv_bytes = (max_layers * world * args.mbs * args.ctx *
@@ -164,6 +166,7 @@ def main() -> int:
                 "n_ctx": args.ctx,
                 "parallelism": f"pp{n_gpus}",
                 "micro_batches": n_mb,
+                "lanes": n_lanes,
                 "mbs": args.mbs,
                 "device": device,
             },
