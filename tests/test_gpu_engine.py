@@ -390,3 +390,36 @@ def test_continuous_batcher_on_gpu():
     torch.cuda.synchronize()
     for r, p, s in zip(reqs, prompts, steps):
         assert r.done and r.out == canonical(p, s)
+
+
+def test_slice_chain_matches_monolith_on_gpu():
+    """Two chained slice engines (layers [0,0] and [1,1] of the tiny
+    model, activations handed between them exactly like pipeline ranks)
+    must produce the monolithic engine's logits bit-for-bit — the
+    single-GPU version of SURVEY §4's slice-vs-monolith parity."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.formats import slicer, synthetic
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+
+    mono = HIPSliceEngine.from_ggml(f, n_ctx=32, max_batch=2)
+    mono.attach_extra(ex)
+    s0 = HIPSliceEngine.from_ggml(slicer.make_slice(f, 0, 0), n_ctx=32,
+                                  max_batch=2)
+    s1 = HIPSliceEngine.from_ggml(slicer.make_slice(f, 1, 1), n_ctx=32,
+                                  max_batch=2)
+    s0.attach_extra(ex)  # embeds on the first stage
+    s1.attach_extra(ex)  # lm-head on the last stage
+
+    toks = torch.tensor([5, 9], dtype=torch.int32, device="cuda")
+    pos = torch.zeros(2, dtype=torch.int32, device="cuda")
+    seq = torch.arange(2, dtype=torch.int32, device="cuda")
+
+    y_mono = mono.forward(mono.embed(toks), pos, seq, decode=True)
+    lg_mono = mono.logits(y_mono, all_logits=True)
+
+    x = s0.forward(s0.embed(toks), pos.clone(), seq, decode=True)
+    y_chain = s1.forward(x, pos.clone(), seq, decode=True)
+    lg_chain = s1.logits(y_chain, all_logits=True)
+
+    assert torch.equal(lg_mono, lg_chain)
