@@ -395,8 +395,11 @@ def test_continuous_batcher_on_gpu():
 def test_slice_chain_matches_monolith_on_gpu():
     """Two chained slice engines (layers [0,0] and [1,1] of the tiny
     model, activations handed between them exactly like pipeline ranks)
-    must produce the monolithic engine's logits bit-for-bit — the
-    single-GPU version of SURVEY §4's slice-vs-monolith parity."""
+    must reproduce the monolithic engine's logits — the single-GPU
+    version of SURVEY §4's slice-vs-monolith parity. Not bit-exact: the
+    boundary regenerates the sumsq side-channel with a different
+    reduction order (k_prep_x vs the fused k_reduce_prep), a ~1-ulp
+    rsqrt difference; assert at a tolerance far below any layout bug."""
     from distributedllm_amd.engine import HIPSliceEngine
     from distributedllm_amd.formats import slicer, synthetic
     f = synthetic.build_model("tiny", seed=0)
@@ -422,4 +425,5 @@ def test_slice_chain_matches_monolith_on_gpu():
     y_chain = s1.forward(x, pos.clone(), seq, decode=True)
     lg_chain = s1.logits(y_chain, all_logits=True)
 
-    assert torch.equal(lg_mono, lg_chain)
+    _assert_close(lg_chain, lg_mono, rel_rms=1e-3, rel_max=1e-2,
+                  label="slice chain vs monolith")
