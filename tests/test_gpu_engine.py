@@ -393,13 +393,14 @@ def test_continuous_batcher_on_gpu():
 
 
 def test_slice_chain_matches_monolith_on_gpu():
-    """Two chained slice engines (layers [0,0] and [1,1] of the tiny
-    model, activations handed between them exactly like pipeline ranks)
-    must reproduce the monolithic engine's logits — the single-GPU
-    version of SURVEY §4's slice-vs-monolith parity. Not bit-exact: the
-    boundary regenerates the sumsq side-channel with a different
-    reduction order (k_prep_x vs the fused k_reduce_prep), a ~1-ulp
-    rsqrt difference; assert at a tolerance far below any layout bug."""
+    """Two chained slice engines (layers [0,0] and [1,2] of the
+    3-layer tiny model, activations handed between them exactly like
+    pipeline ranks) must reproduce the monolithic engine's logits —
+    the single-GPU version of SURVEY §4's slice-vs-monolith parity.
+    Not necessarily bit-exact: the boundary regenerates the sumsq
+    side-channel with a different reduction order (k_prep_x vs the
+    fused k_reduce_prep), a ~1-ulp rsqrt difference; assert at a
+    tolerance far below any layout bug."""
     from distributedllm_amd.engine import HIPSliceEngine
     from distributedllm_amd.formats import slicer, synthetic
     f = synthetic.build_model("tiny", seed=0)
@@ -409,8 +410,9 @@ def test_slice_chain_matches_monolith_on_gpu():
     mono.attach_extra(ex)
     s0 = HIPSliceEngine.from_ggml(slicer.make_slice(f, 0, 0), n_ctx=32,
                                   max_batch=2)
-    s1 = HIPSliceEngine.from_ggml(slicer.make_slice(f, 1, 1), n_ctx=32,
-                                  max_batch=2)
+    s1 = HIPSliceEngine.from_ggml(
+        slicer.make_slice(f, 1, f.hparams.n_layer - 1), n_ctx=32,
+        max_batch=2)
     s0.attach_extra(ex)  # embeds on the first stage
     s1.attach_extra(ex)  # lm-head on the last stage
 
