@@ -27,7 +27,10 @@ from .registry import ModelEntry, Registry, SliceEntry
 
 VALID_QUANT = {"f32": ggml.FTYPE_ALL_F32, "f16": ggml.FTYPE_MOSTLY_F16,
                "q4_0": ggml.FTYPE_MOSTLY_Q4_0,
-               "q4_1": ggml.FTYPE_MOSTLY_Q4_1}
+               "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
+               "q5_0": ggml.FTYPE_MOSTLY_Q5_0,
+               "q5_1": ggml.FTYPE_MOSTLY_Q5_1,
+               "q8_0": ggml.FTYPE_MOSTLY_Q8_0}
 VALID_FAMILY = {"llama_v1", "llama_v2"}
 
 

@@ -41,18 +41,27 @@ GGML_TYPE_F32 = 0
 GGML_TYPE_F16 = 1
 GGML_TYPE_Q4_0 = 2
 GGML_TYPE_Q4_1 = 3
+GGML_TYPE_Q5_0 = 6   # ids 4/5 were the removed q4_2/q4_3
+GGML_TYPE_Q5_1 = 7
+GGML_TYPE_Q8_0 = 8
 
-# model-level ftype
+# model-level ftype (llama_ftype ids; 4 = Q4_1_SOME_F16, 5/6 removed)
 FTYPE_ALL_F32 = 0
 FTYPE_MOSTLY_F16 = 1
 FTYPE_MOSTLY_Q4_0 = 2
 FTYPE_MOSTLY_Q4_1 = 3
+FTYPE_MOSTLY_Q8_0 = 7
+FTYPE_MOSTLY_Q5_0 = 8
+FTYPE_MOSTLY_Q5_1 = 9
 
 _FTYPE_TO_GGML = {
     FTYPE_ALL_F32: GGML_TYPE_F32,
     FTYPE_MOSTLY_F16: GGML_TYPE_F16,
     FTYPE_MOSTLY_Q4_0: GGML_TYPE_Q4_0,
     FTYPE_MOSTLY_Q4_1: GGML_TYPE_Q4_1,
+    FTYPE_MOSTLY_Q8_0: GGML_TYPE_Q8_0,
+    FTYPE_MOSTLY_Q5_0: GGML_TYPE_Q5_0,
+    FTYPE_MOSTLY_Q5_1: GGML_TYPE_Q5_1,
 }
 
 TYPE_NAMES = {
@@ -60,6 +69,23 @@ TYPE_NAMES = {
     GGML_TYPE_F16: "f16",
     GGML_TYPE_Q4_0: "q4_0",
     GGML_TYPE_Q4_1: "q4_1",
+    GGML_TYPE_Q5_0: "q5_0",
+    GGML_TYPE_Q5_1: "q5_1",
+    GGML_TYPE_Q8_0: "q8_0",
+}
+
+# block codecs for the quantized types: (block bytes, quantize, dequantize)
+_BLOCK_CODECS = {
+    GGML_TYPE_Q4_0: (q4.Q4_0_BLOCK_BYTES, q4.quantize_q4_0,
+                     q4.dequantize_q4_0),
+    GGML_TYPE_Q4_1: (q4.Q4_1_BLOCK_BYTES, q4.quantize_q4_1,
+                     q4.dequantize_q4_1),
+    GGML_TYPE_Q5_0: (q4.Q5_0_BLOCK_BYTES, q4.quantize_q5_0,
+                     q4.dequantize_q5_0),
+    GGML_TYPE_Q5_1: (q4.Q5_1_BLOCK_BYTES, q4.quantize_q5_1,
+                     q4.dequantize_q5_1),
+    GGML_TYPE_Q8_0: (q4.Q8_0_BLOCK_BYTES, q4.quantize_q8_0,
+                     q4.dequantize_q8_0),
 }
 
 
@@ -73,14 +99,11 @@ def tensor_nbytes(gtype: int, ne: Tuple[int, ...]) -> int:
         return 4 * n0 * rows
     if gtype == GGML_TYPE_F16:
         return 2 * n0 * rows
-    if gtype == GGML_TYPE_Q4_0:
+    if gtype in _BLOCK_CODECS:
         if n0 % q4.QK4:
-            raise ValueError(f"q4_0 row length {n0} not a multiple of 32")
-        return (n0 // q4.QK4) * q4.Q4_0_BLOCK_BYTES * rows
-    if gtype == GGML_TYPE_Q4_1:
-        if n0 % q4.QK4:
-            raise ValueError(f"q4_1 row length {n0} not a multiple of 32")
-        return (n0 // q4.QK4) * q4.Q4_1_BLOCK_BYTES * rows
+            raise ValueError(
+                f"{TYPE_NAMES[gtype]} row length {n0} not a multiple of 32")
+        return (n0 // q4.QK4) * _BLOCK_CODECS[gtype][0] * rows
     raise ValueError(f"unsupported ggml type {gtype}")
 
 
@@ -133,12 +156,9 @@ class GGMLTensor:
             a = np.frombuffer(self.raw, dtype=np.float32).astype(np.float32)
         elif self.gtype == GGML_TYPE_F16:
             a = np.frombuffer(self.raw, dtype=np.float16).astype(np.float32)
-        elif self.gtype == GGML_TYPE_Q4_0:
+        elif self.gtype in _BLOCK_CODECS:
             u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
-            a = q4.dequantize_q4_0(u, cols)
-        elif self.gtype == GGML_TYPE_Q4_1:
-            u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
-            a = q4.dequantize_q4_1(u, cols)
+            a = _BLOCK_CODECS[self.gtype][2](u, cols)
         else:
             raise ValueError(f"unsupported ggml type {self.gtype}")
         if len(self.ne) == 1:
@@ -159,10 +179,8 @@ class GGMLTensor:
             raw = a.astype(np.float32).tobytes()
         elif gtype == GGML_TYPE_F16:
             raw = a.astype(np.float16).tobytes()
-        elif gtype == GGML_TYPE_Q4_0:
-            raw = q4.quantize_q4_0(a).tobytes()
-        elif gtype == GGML_TYPE_Q4_1:
-            raw = q4.quantize_q4_1(a).tobytes()
+        elif gtype in _BLOCK_CODECS:
+            raw = _BLOCK_CODECS[gtype][1](a).tobytes()
         else:
             raise ValueError(f"unsupported ggml type {gtype}")
         return cls(name=name, ne=ne, gtype=gtype, raw=raw)

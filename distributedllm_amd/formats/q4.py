@@ -21,6 +21,8 @@ from __future__ import annotations
 
 import numpy as np
 
+from typing import Tuple
+
 QK4 = 32  # weights per block
 Q4_0_BLOCK_BYTES = 2 + 16
 Q4_1_BLOCK_BYTES = 4 + 16
@@ -115,3 +117,129 @@ def dequantize_q4_1(raw: np.ndarray, n: int) -> np.ndarray:
     hi = (qs >> 4).astype(np.float32)
     q = np.concatenate([lo, hi], axis=-1)
     return (q * d[..., None] + m[..., None]).reshape(lead + (n,))
+
+
+# ---------------------------------------------------------------- q5 / q8
+# The remaining GGJT-v3 "classic" block formats the reference engine loads
+# (tensor_processor.cpp:210-212 accepts GGML_TYPE_Q5_0/Q5_1/Q8_0):
+#
+# * q5_0: ``d`` (f16) ‖ ``qh`` (u32, the 5th bits) ‖ 16 nibble bytes.
+#   Weight j's low 4 bits sit in nibble j (same j/j+16 byte split as
+#   q4_0); bit j of ``qh`` is weight j's 5th bit (j+16 for the high
+#   nibbles). Dequant: ``x = d * (q5 - 16)``; quant: signed-amax
+#   ``d = m/-16``, ``q5 = clamp(round(x/d) + 16, 0, 31)``.
+# * q5_1: ``d`` (f16) ‖ ``m`` (f16) ‖ ``qh`` ‖ 16 nibble bytes.
+#   ``x = d*q5 + m``; ``d = (max-min)/31``, ``m = min``.
+# * q8_0: ``d`` (f16) ‖ 32 int8. ``x = d*q``; ``d = amax/127``.
+
+Q5_0_BLOCK_BYTES = 2 + 4 + 16
+Q5_1_BLOCK_BYTES = 4 + 4 + 16
+Q8_0_BLOCK_BYTES = 2 + 32
+
+
+def _pack_q5(q: np.ndarray) -> Tuple[np.ndarray, np.ndarray]:
+    """5-bit values [nb, 32] -> (16 nibble bytes, qh u32 as 4 bytes)."""
+    lo5, hi5 = q[..., :16], q[..., 16:]
+    packed = ((lo5 & 0xF) | ((hi5 & 0xF) << 4)).astype(np.uint8)
+    bits = np.concatenate([(lo5 >> 4) & 1, (hi5 >> 4) & 1],
+                          axis=-1).astype(np.uint32)
+    qh = (bits << np.arange(32, dtype=np.uint32)).sum(
+        axis=-1, dtype=np.uint32)
+    return packed, qh.astype("<u4")[..., None].view(np.uint8)
+
+
+def _unpack_q5(qs: np.ndarray, qh_bytes: np.ndarray) -> np.ndarray:
+    """(16 nibble bytes, 4 qh bytes) -> 5-bit values [..., 32]."""
+    qh = qh_bytes.copy().view("<u4")[..., 0]
+    bits = (qh[..., None] >> np.arange(32, dtype=np.uint32)) & 1
+    lo = (qs & 0x0F) | ((bits[..., :16] << 4).astype(np.uint8))
+    hi = (qs >> 4) | ((bits[..., 16:] << 4).astype(np.uint8))
+    return np.concatenate([lo, hi], axis=-1)
+
+
+def quantize_q5_0(x: np.ndarray) -> np.ndarray:
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    lead = x.shape[:-1]
+    nb = _check_shape(x.shape[-1])
+    b = x.reshape(-1, nb, QK4)
+    idx = np.argmax(np.abs(b), axis=-1)
+    m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
+    d = (m / -16.0).astype(np.float16)
+    df = d.astype(np.float32)
+    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    q = np.clip(np.rint(b * inv[..., None]) + 16, 0, 31).astype(np.uint8)
+    packed, qh = _pack_q5(q)
+    out = np.empty(b.shape[:2] + (Q5_0_BLOCK_BYTES,), dtype=np.uint8)
+    out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
+    out[..., 2:6] = qh
+    out[..., 6:] = packed
+    return out.reshape(lead + (nb * Q5_0_BLOCK_BYTES,))
+
+
+def dequantize_q5_0(raw: np.ndarray, n: int) -> np.ndarray:
+    raw = np.ascontiguousarray(raw, dtype=np.uint8)
+    nb = _check_shape(n)
+    lead = raw.shape[:-1]
+    b = raw.reshape(-1, nb, Q5_0_BLOCK_BYTES)
+    d = b[..., 0:2].copy().view(np.float16)[..., 0].astype(np.float32)
+    q = _unpack_q5(b[..., 6:], b[..., 2:6]).astype(np.float32) - 16.0
+    return (q * d[..., None]).reshape(lead + (n,))
+
+
+def quantize_q5_1(x: np.ndarray) -> np.ndarray:
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    lead = x.shape[:-1]
+    nb = _check_shape(x.shape[-1])
+    b = x.reshape(-1, nb, QK4)
+    mn = b.min(axis=-1)
+    mx = b.max(axis=-1)
+    d = ((mx - mn) / 31.0).astype(np.float16)
+    m = mn.astype(np.float16)
+    df = d.astype(np.float32)
+    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) *
+                        inv[..., None]), 0, 31).astype(np.uint8)
+    packed, qh = _pack_q5(q)
+    out = np.empty(b.shape[:2] + (Q5_1_BLOCK_BYTES,), dtype=np.uint8)
+    out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
+    out[..., 2:4] = m[..., None].view(np.uint8).reshape(m.shape + (2,))
+    out[..., 4:8] = qh
+    out[..., 8:] = packed
+    return out.reshape(lead + (nb * Q5_1_BLOCK_BYTES,))
+
+
+def dequantize_q5_1(raw: np.ndarray, n: int) -> np.ndarray:
+    raw = np.ascontiguousarray(raw, dtype=np.uint8)
+    nb = _check_shape(n)
+    lead = raw.shape[:-1]
+    b = raw.reshape(-1, nb, Q5_1_BLOCK_BYTES)
+    d = b[..., 0:2].copy().view(np.float16)[..., 0].astype(np.float32)
+    m = b[..., 2:4].copy().view(np.float16)[..., 0].astype(np.float32)
+    q = _unpack_q5(b[..., 8:], b[..., 4:8]).astype(np.float32)
+    return (q * d[..., None] + m[..., None]).reshape(lead + (n,))
+
+
+def quantize_q8_0(x: np.ndarray) -> np.ndarray:
+    x = np.ascontiguousarray(x, dtype=np.float32)
+    lead = x.shape[:-1]
+    nb = _check_shape(x.shape[-1])
+    b = x.reshape(-1, nb, QK4)
+    amax = np.abs(b).max(axis=-1)
+    d = (amax / 127.0).astype(np.float16)
+    df = d.astype(np.float32)
+    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    q = np.rint(b * inv[..., None]).astype(np.int8)
+    out = np.empty(b.shape[:2] + (Q8_0_BLOCK_BYTES,), dtype=np.uint8)
+    out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
+    out[..., 2:] = q.view(np.uint8)
+    return out.reshape(lead + (nb * Q8_0_BLOCK_BYTES,))
+
+
+def dequantize_q8_0(raw: np.ndarray, n: int) -> np.ndarray:
+    raw = np.ascontiguousarray(raw, dtype=np.uint8)
+    nb = _check_shape(n)
+    lead = raw.shape[:-1]
+    b = raw.reshape(-1, nb, Q8_0_BLOCK_BYTES)
+    d = b[..., 0:2].copy().view(np.float16)[..., 0].astype(np.float32)
+    q = b[..., 2:].copy().view(np.int8).astype(np.float32)
+    return (q * d[..., None]).reshape(lead + (n,))

@@ -70,17 +70,24 @@ def test_sliced_file_loads_and_runs(bins, model_file, tmp_path):
     assert torch.isfinite(y).all()
 
 
-@pytest.mark.parametrize("target", ["q4_0", "q4_1"])
+QUANTS = {
+    "q4_0": (ggml.FTYPE_MOSTLY_Q4_0, q4.quantize_q4_0),
+    "q4_1": (ggml.FTYPE_MOSTLY_Q4_1, q4.quantize_q4_1),
+    "q5_0": (ggml.FTYPE_MOSTLY_Q5_0, q4.quantize_q5_0),
+    "q5_1": (ggml.FTYPE_MOSTLY_Q5_1, q4.quantize_q5_1),
+    "q8_0": (ggml.FTYPE_MOSTLY_Q8_0, q4.quantize_q8_0),
+}
+
+
+@pytest.mark.parametrize("target", list(QUANTS))
 def test_quantize_matches_python_codec(bins, model_file, tmp_path, target):
     f, path = model_file
     out_cpp = tmp_path / f"model_{target}.bin"
     subprocess.run([str(bins / "quantize"), str(path), str(out_cpp), target],
                    check=True, capture_output=True)
     got = ggml.GGMLFile.load(str(out_cpp), extended=False)
-    gtype = (ggml.GGML_TYPE_Q4_0 if target == "q4_0"
-             else ggml.GGML_TYPE_Q4_1)
-    quant = q4.quantize_q4_0 if target == "q4_0" else q4.quantize_q4_1
-    assert got.hparams.ftype == gtype
+    ftype, quant = QUANTS[target]
+    assert got.hparams.ftype == ftype
     for t in f.tensors:
         g = got.tensor_map()[t.name]
         if len(t.ne) == 1:

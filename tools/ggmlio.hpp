@@ -33,10 +33,14 @@ constexpr uint32_t kMagic = 0x67676A74;  // "ggjt" little-endian
 constexpr uint32_t kVersion = 3;
 constexpr uint32_t kExtraLayersFirstLayer = 0xFFFFFFFFu;
 
-enum GType : uint32_t { F32 = 0, F16 = 1, Q4_0 = 2, Q4_1 = 3 };
-constexpr int kQK = 32;               // weights per q4 block
+enum GType : uint32_t { F32 = 0, F16 = 1, Q4_0 = 2, Q4_1 = 3,
+                        Q5_0 = 6, Q5_1 = 7, Q8_0 = 8 };
+constexpr int kQK = 32;               // weights per quant block
 constexpr int kQ4_0Bytes = 18;        // f16 d + 16 nibble bytes
 constexpr int kQ4_1Bytes = 20;        // f16 d + f16 m + 16 nibble bytes
+constexpr int kQ5_0Bytes = 22;        // f16 d + u32 qh + 16 nibble bytes
+constexpr int kQ5_1Bytes = 24;        // f16 d + f16 m + u32 qh + 16 B
+constexpr int kQ8_0Bytes = 34;        // f16 d + 32 int8
 
 // ------------------------------------------------------------- f16 codec
 // Round-to-nearest-even f32 -> f16 (bit-exact with numpy astype(float16)).
@@ -108,6 +112,15 @@ inline size_t row_bytes(GType t, uint32_t ne0) {
         case Q4_1:
             if (ne0 % kQK) throw std::runtime_error("q4_1 row not /32");
             return (size_t)(ne0 / kQK) * kQ4_1Bytes;
+        case Q5_0:
+            if (ne0 % kQK) throw std::runtime_error("q5_0 row not /32");
+            return (size_t)(ne0 / kQK) * kQ5_0Bytes;
+        case Q5_1:
+            if (ne0 % kQK) throw std::runtime_error("q5_1 row not /32");
+            return (size_t)(ne0 / kQK) * kQ5_1Bytes;
+        case Q8_0:
+            if (ne0 % kQK) throw std::runtime_error("q8_0 row not /32");
+            return (size_t)(ne0 / kQK) * kQ8_0Bytes;
     }
     throw std::runtime_error("unknown ggml type");
 }
@@ -174,6 +187,100 @@ inline void dequantize_block_q4_1(const uint8_t* in, float* x) {
         x[j] = d * (float)(in[2 + j] & 0x0F) + m;
         x[j + 16] = d * (float)(in[2 + j] >> 4) + m;
     }
+}
+
+inline void quantize_block_q5_0(const float* x, uint8_t* out) {
+    float amax = 0.0f, m = 0.0f;
+    for (int j = 0; j < kQK; ++j) {
+        if (std::fabs(x[j]) > amax) { amax = std::fabs(x[j]); m = x[j]; }
+    }
+    const uint16_t dh = f32_to_f16(m / -16.0f);
+    const float d = f16_to_f32(dh);
+    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    std::memcpy(out, &dh, 2);
+    uint32_t qh = 0;
+    for (int j = 0; j < 16; ++j) {
+        const int q0 = std::clamp((int)std::nearbyintf(x[j] * inv) + 16,
+                                  0, 31);
+        const int q1 = std::clamp(
+            (int)std::nearbyintf(x[j + 16] * inv) + 16, 0, 31);
+        out[6 + j] = (uint8_t)((q0 & 0xF) | ((q1 & 0xF) << 4));
+        qh |= (uint32_t)((q0 >> 4) & 1) << j;
+        qh |= (uint32_t)((q1 >> 4) & 1) << (j + 16);
+    }
+    std::memcpy(out + 2, &qh, 4);
+}
+
+inline void dequantize_block_q5_0(const uint8_t* in, float* x) {
+    uint16_t dh;
+    std::memcpy(&dh, in, 2);
+    uint32_t qh;
+    std::memcpy(&qh, in + 2, 4);
+    const float d = f16_to_f32(dh);
+    for (int j = 0; j < 16; ++j) {
+        const int q0 = (in[6 + j] & 0x0F) | (int)(((qh >> j) & 1) << 4);
+        const int q1 = (in[6 + j] >> 4) | (int)(((qh >> (j + 16)) & 1) << 4);
+        x[j] = d * (float)(q0 - 16);
+        x[j + 16] = d * (float)(q1 - 16);
+    }
+}
+
+inline void quantize_block_q5_1(const float* x, uint8_t* out) {
+    float mn = x[0], mx = x[0];
+    for (int j = 1; j < kQK; ++j) {
+        mn = std::min(mn, x[j]);
+        mx = std::max(mx, x[j]);
+    }
+    const uint16_t dh = f32_to_f16((mx - mn) / 31.0f);
+    const uint16_t mh = f32_to_f16(mn);
+    const float d = f16_to_f32(dh), m = f16_to_f32(mh);
+    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    std::memcpy(out, &dh, 2);
+    std::memcpy(out + 2, &mh, 2);
+    uint32_t qh = 0;
+    for (int j = 0; j < 16; ++j) {
+        const int q0 = std::clamp(
+            (int)std::nearbyintf((x[j] - m) * inv), 0, 31);
+        const int q1 = std::clamp(
+            (int)std::nearbyintf((x[j + 16] - m) * inv), 0, 31);
+        out[8 + j] = (uint8_t)((q0 & 0xF) | ((q1 & 0xF) << 4));
+        qh |= (uint32_t)((q0 >> 4) & 1) << j;
+        qh |= (uint32_t)((q1 >> 4) & 1) << (j + 16);
+    }
+    std::memcpy(out + 4, &qh, 4);
+}
+
+inline void dequantize_block_q5_1(const uint8_t* in, float* x) {
+    uint16_t dh, mh;
+    std::memcpy(&dh, in, 2);
+    std::memcpy(&mh, in + 2, 2);
+    uint32_t qh;
+    std::memcpy(&qh, in + 4, 4);
+    const float d = f16_to_f32(dh), m = f16_to_f32(mh);
+    for (int j = 0; j < 16; ++j) {
+        const int q0 = (in[8 + j] & 0x0F) | (int)(((qh >> j) & 1) << 4);
+        const int q1 = (in[8 + j] >> 4) | (int)(((qh >> (j + 16)) & 1) << 4);
+        x[j] = d * (float)q0 + m;
+        x[j + 16] = d * (float)q1 + m;
+    }
+}
+
+inline void quantize_block_q8_0(const float* x, uint8_t* out) {
+    float amax = 0.0f;
+    for (int j = 0; j < kQK; ++j) amax = std::max(amax, std::fabs(x[j]));
+    const uint16_t dh = f32_to_f16(amax / 127.0f);
+    const float d = f16_to_f32(dh);
+    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    std::memcpy(out, &dh, 2);
+    for (int j = 0; j < kQK; ++j)
+        out[2 + j] = (uint8_t)(int8_t)std::nearbyintf(x[j] * inv);
+}
+
+inline void dequantize_block_q8_0(const uint8_t* in, float* x) {
+    uint16_t dh;
+    std::memcpy(&dh, in, 2);
+    const float d = f16_to_f32(dh);
+    for (int j = 0; j < kQK; ++j) x[j] = d * (float)(int8_t)in[2 + j];
 }
 
 // --------------------------------------------------------------- file IO

@@ -1,4 +1,5 @@
-// quantize — native CLI that requantizes a GGML model to q4_0/q4_1.
+// quantize — native CLI that requantizes a GGML model to
+// q4_0/q4_1/q5_0/q5_1/q8_0.
 //
 // Native counterpart of the vendored llama.cpp `quantize` binary the
 // reference's provisioning shells out to
@@ -40,6 +41,21 @@ std::vector<float> to_f32(const ggmlio::Tensor& t) {
                 ggmlio::dequantize_block_q4_1(p + i * ggmlio::kQ4_1Bytes,
                                               out.data() + i * ggmlio::kQK);
             break;
+        case ggmlio::Q5_0:
+            for (size_t i = 0; i < out.size() / ggmlio::kQK; ++i)
+                ggmlio::dequantize_block_q5_0(p + i * ggmlio::kQ5_0Bytes,
+                                              out.data() + i * ggmlio::kQK);
+            break;
+        case ggmlio::Q5_1:
+            for (size_t i = 0; i < out.size() / ggmlio::kQK; ++i)
+                ggmlio::dequantize_block_q5_1(p + i * ggmlio::kQ5_1Bytes,
+                                              out.data() + i * ggmlio::kQK);
+            break;
+        case ggmlio::Q8_0:
+            for (size_t i = 0; i < out.size() / ggmlio::kQK; ++i)
+                ggmlio::dequantize_block_q8_0(p + i * ggmlio::kQ8_0Bytes,
+                                              out.data() + i * ggmlio::kQK);
+            break;
     }
     return out;
 }
@@ -72,6 +88,21 @@ ggmlio::Tensor from_f32(const ggmlio::Tensor& t, const std::vector<float>& x,
                 ggmlio::quantize_block_q4_1(x.data() + i * ggmlio::kQK,
                                             p + i * ggmlio::kQ4_1Bytes);
             break;
+        case ggmlio::Q5_0:
+            for (size_t i = 0; i < x.size() / ggmlio::kQK; ++i)
+                ggmlio::quantize_block_q5_0(x.data() + i * ggmlio::kQK,
+                                            p + i * ggmlio::kQ5_0Bytes);
+            break;
+        case ggmlio::Q5_1:
+            for (size_t i = 0; i < x.size() / ggmlio::kQK; ++i)
+                ggmlio::quantize_block_q5_1(x.data() + i * ggmlio::kQK,
+                                            p + i * ggmlio::kQ5_1Bytes);
+            break;
+        case ggmlio::Q8_0:
+            for (size_t i = 0; i < x.size() / ggmlio::kQK; ++i)
+                ggmlio::quantize_block_q8_0(x.data() + i * ggmlio::kQK,
+                                            p + i * ggmlio::kQ8_0Bytes);
+            break;
     }
     return out;
 }
@@ -81,14 +112,18 @@ ggmlio::Tensor from_f32(const ggmlio::Tensor& t, const std::vector<float>& x,
 int main(int argc, char** argv) {
     if (argc < 4) {
         std::fprintf(stderr,
-                     "usage: quantize <in.bin> <out.bin> <q4_0|q4_1|f16>\n");
+                     "usage: quantize <in.bin> <out.bin> <q4_0|q4_1|q5_0|q5_1|q8_0|f16>\n");
         return 2;
     }
     ggmlio::GType target;
+    uint32_t ftype;  // llama_ftype id (differs from the GType for q5/q8)
     const std::string t = argv[3];
-    if (t == "q4_0") target = ggmlio::Q4_0;
-    else if (t == "q4_1") target = ggmlio::Q4_1;
-    else if (t == "f16") target = ggmlio::F16;
+    if (t == "q4_0") { target = ggmlio::Q4_0; ftype = 2; }
+    else if (t == "q4_1") { target = ggmlio::Q4_1; ftype = 3; }
+    else if (t == "f16") { target = ggmlio::F16; ftype = 1; }
+    else if (t == "q8_0") { target = ggmlio::Q8_0; ftype = 7; }
+    else if (t == "q5_0") { target = ggmlio::Q5_0; ftype = 8; }
+    else if (t == "q5_1") { target = ggmlio::Q5_1; ftype = 9; }
     else {
         std::fprintf(stderr, "unknown target type %s\n", t.c_str());
         return 2;
@@ -98,7 +133,7 @@ int main(int argc, char** argv) {
         ggmlio::File in = reader.parse(/*extended=*/false);
         ggmlio::File out;
         out.hp = in.hp;
-        out.hp.ftype = (uint32_t)target;  // ftype ids match GType ids
+        out.hp.ftype = ftype;
         out.vocab = in.vocab;
         size_t quantized = 0;
         for (const auto& ten : in.tensors) {
