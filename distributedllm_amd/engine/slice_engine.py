@@ -40,6 +40,57 @@ def _nibbles(t: ggml.GGMLTensor) -> np.ndarray:
     return np.concatenate([lo, hi], axis=-1)  # weight j / j+16 layout
 
 
+_BYTE_GTYPES = (ggml.GGML_TYPE_Q5_0, ggml.GGML_TYPE_Q5_1,
+                ggml.GGML_TYPE_Q8_0)
+
+
+def _byte_values(t: ggml.GGMLTensor):
+    """q5_0/q5_1/q8_0 tensor -> (u8 values [rows, nb, 32] re-biased for
+    the kernel's unified csub=-1152, alpha f32, beta f32).
+
+    The kernel computes w = alpha*((1024+u) - 1152) + beta, so:
+      q8_0: u = q + 128          -> alpha=d, beta=0
+      q5_0: u = q5 + 112         -> alpha=d, beta=0   (q5-16 exact)
+      q5_1: u = q5 + 112         -> alpha=d, beta=m + 16*d
+    """
+    rows, cols = t.shape_rows_cols
+    nb = cols // 32
+    u8 = np.frombuffer(t.raw, np.uint8)
+    if t.gtype == ggml.GGML_TYPE_Q8_0:
+        a = u8.reshape(rows, nb, 34)
+        d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
+        alpha = d.reshape(rows, nb).astype(np.float32)
+        beta = np.zeros_like(alpha)
+        vals = (a[:, :, 2:] ^ 0x80).astype(np.uint8)  # int8 + 128
+        return vals, alpha, beta
+    bs = 22 if t.gtype == ggml.GGML_TYPE_Q5_0 else 24
+    hoff = 2 if t.gtype == ggml.GGML_TYPE_Q5_0 else 4
+    a = np.frombuffer(t.raw, np.uint8).reshape(rows, nb, bs)
+    q5 = q4._unpack_q5(a[:, :, hoff + 4:],
+                       np.ascontiguousarray(a[:, :, hoff:hoff + 4]))
+    vals = (q5 + 112).astype(np.uint8)
+    d = np.ascontiguousarray(a[:, :, :2]).view(np.float16)
+    alpha = d.reshape(rows, nb).astype(np.float32)
+    if t.gtype == ggml.GGML_TYPE_Q5_0:
+        beta = np.zeros_like(alpha)
+    else:
+        m = np.ascontiguousarray(a[:, :, 2:4]).view(np.float16)
+        beta = (m.reshape(rows, nb).astype(np.float32) + 16.0 * alpha)
+    return vals, alpha, beta
+
+
+def _pack_grouped_scales(alpha, beta, rows, nb, nbp, R):
+    """(alpha, beta) f32 [rows, nb] -> grouped f16 layout
+    [R][nbp/4][16][4][2] + zero-padded blocks, flat."""
+    al = np.zeros((rows, nbp), dtype=np.float32)
+    be = np.zeros((rows, nbp), dtype=np.float32)
+    al[:, :nb] = alpha
+    be[:, :nb] = beta
+    ab = np.stack([al, be], axis=-1).astype(np.float16)
+    return np.ascontiguousarray(
+        ab.reshape(R, 16, nbp // 4, 4, 2).transpose(0, 2, 1, 3, 4))
+
+
 def repack_mfma(t: ggml.GGMLTensor, device: str):
     """On-disk tensor -> (data, scales, wtype) in the MFMA tile layout.
 
@@ -96,6 +147,31 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             np.concatenate([scales.reshape(-1),
                             np.zeros(128, np.float16)])).to(device)
         return data, sc, t.gtype
+    if t.gtype in _BYTE_GTYPES:
+        nb = cols // 32
+        nbp = (nb + 3) & ~3
+        vals, alpha, beta = _byte_values(t)
+        # byte order per lane-block 8 weights: [w0,w2,w1,w3, w4,w6,w5,w7]
+        # so the kernel's (q & 0x00FF00FF) / (q>>8 & ..) masks yield the
+        # (even, odd) f16 pairs of the A fragment (a_frag_q8)
+        perm = [0, 2, 1, 3, 4, 6, 5, 7]
+        v = vals.reshape(rows, nb, 4, 8)[:, :, :, perm]
+        vp = np.zeros((rows, nbp, 4, 8), dtype=np.uint8)
+        vp[:, :nb] = v
+        # -> u32[R][nbp/4][4 ws][16 i][4 kb][2] with the block's two u32
+        # adjacent per lane (one dwordx4 pair covers a 4-block group)
+        q32 = vp.reshape(rows, nbp, 4, 2, 4).copy().view("<u4")[..., 0]
+        qs2 = np.ascontiguousarray(
+            q32.reshape(R, 16, nbp // 4, 4, 4, 2)   # [R][i][g4][kb][ws][2]
+            .transpose(0, 2, 4, 1, 3, 5))           # [R][g4][ws][i][kb][2]
+        scales = _pack_grouped_scales(alpha, beta, rows, nb, nbp, R)
+        data = torch.from_numpy(
+            np.concatenate([qs2.view(np.int32).reshape(-1),
+                            np.zeros(512, np.int32)])).to(device)
+        sc = torch.from_numpy(
+            np.concatenate([scales.reshape(-1),
+                            np.zeros(128, np.float16)])).to(device)
+        return data, sc, ggml.GGML_TYPE_Q8_0  # W_Q8B for all byte formats
     if t.gtype == ggml.GGML_TYPE_F16:
         w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
         tile = np.ascontiguousarray(
@@ -134,6 +210,12 @@ def _upload_mat(t: ggml.GGMLTensor, device: str):
         d = torch.from_numpy(qs).to(device)
         s = torch.from_numpy(scales.copy()).to(device)
         return d, s, t.gtype
+    if t.gtype in _BYTE_GTYPES:
+        # embedding gather for q5/q8 tables: dequantize once to f16 (the
+        # gather kernel has no byte path; exact dequant, 2 B/weight)
+        a = t.to_f32().astype(np.float16)
+        return (torch.from_numpy(a).to(device), torch.empty(0),
+                ggml.GGML_TYPE_F16)
     if t.gtype == ggml.GGML_TYPE_F16:
         a = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
         return torch.from_numpy(a.copy()).to(device), torch.empty(0), t.gtype
@@ -194,6 +276,24 @@ class HIPSliceEngine:
             # random weights directly in the MFMA tile layouts — any random
             # bit pattern is a valid q4 nibble word, so this is byte-for-
             # byte the same compute/HBM traffic as a real checkpoint
+            if wt in _BYTE_GTYPES:
+                R, nb = rows // 16, cols // 32
+                nbp = (nb + 3) & ~3
+                data = torch.randint(-2**31, 2**31 - 1,
+                                     (R * nbp * 128 + 512,),
+                                     dtype=torch.int32, device="cuda",
+                                     generator=g)
+                alpha = ((torch.rand(rows, nbp, device="cuda",
+                                     generator=g) * 0.5 + 0.75) * 0.003)
+                alpha[:, nb:] = 0.0
+                beta = (alpha * 0.1 if wt == ggml.GGML_TYPE_Q5_1
+                        else torch.zeros_like(alpha))
+                ab = torch.stack([alpha, beta], dim=-1).to(torch.float16)
+                ab = (ab.reshape(R, 16, nbp // 4, 4, 2)
+                      .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
+                ab = torch.cat([ab, torch.zeros(128, dtype=torch.float16,
+                                                device="cuda")])
+                return data, ab, ggml.GGML_TYPE_Q8_0
             if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
                 R, nb = rows // 16, cols // 32
                 nbp = (nb + 3) & ~3
@@ -239,7 +339,12 @@ class HIPSliceEngine:
         eng._extra_cache = None
         if with_extra:
             # the embedding table uses the legacy SoA layout (gather kernel)
-            if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+            if wt in _BYTE_GTYPES:
+                tok_d = (torch.randn(V, E, device="cuda", generator=g,
+                                     dtype=torch.float32) * 0.02).to(
+                    torch.float16)
+                tok_s, tok_t = torch.empty(0), ggml.GGML_TYPE_F16
+            elif wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
                 nb = E // 32
                 per = 2 if wt == ggml.GGML_TYPE_Q4_1 else 1
                 tok_d = torch.randint(0, 256, (V, nb * 16),

@@ -36,6 +36,7 @@ constexpr int kMaxTokens = 64;  // per-forward token cap (4 MFMA col tiles)
 // Tail slack appended to every MFMA-path allocation: the pipelined K loop
 // prefetches ONE load batch past its range (kernels.hip wave_tile_kloop).
 constexpr int64_t kTailSlackQ = 256;     // int32 elements (1 KiB)
+constexpr int64_t kTailSlackQ8 = 512;    // int32 elements (byte stream)
 constexpr int64_t kTailSlackAB = 128;    // f16 elements (256 B)
 constexpr int64_t kTailSlackF16 = 2048;  // f16 elements (4 KiB)
 constexpr int64_t kTailSlackSide = 8192; // int16 elements (16 KiB)
@@ -95,6 +96,19 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
                         scales.numel() == R * nbp * 16 * 2 + kTailSlackAB,
                     "q4 tiled scales must be f16 (a,b)[R][nbp/4][16][4] "
                     "+ slack");
+        m.scales = scales;
+        m.w.scales = scales.data_ptr();
+    } else if (wtype == W_Q8B) {
+        TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
+                        data.numel() == R * nbp * 128 + kTailSlackQ8,
+                    "byte-quant tiled data must be u32[R][nbp/4][4][16][8]"
+                    " + slack");
+        TORCH_CHECK(scales.defined() && scales.is_cuda() &&
+                        scales.is_contiguous() &&
+                        scales.scalar_type() == torch::kFloat16 &&
+                        scales.numel() == R * nbp * 16 * 2 + kTailSlackAB,
+                    "byte-quant scales must be f16 (a,b)[R][nbp/4][16][4]"
+                    " + slack");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {

@@ -5,7 +5,10 @@
 #include <hip/hip_fp16.h>
 #include <stdint.h>
 
-enum WType { W_F32 = 0, W_F16 = 1, W_Q4_0 = 2, W_Q4_1 = 3 };
+// W_Q8B is the BYTE-stream quant path: q8_0 weights re-biased to u8 at
+// repack, and q5_0/q5_1 expanded to re-biased bytes — all three share
+// one kernel form (w = alpha*((1024+u) - 1152) + beta in packed f16).
+enum WType { W_F32 = 0, W_F16 = 1, W_Q4_0 = 2, W_Q4_1 = 3, W_Q8B = 8 };
 
 // One weight matrix resident in HBM, repacked SoA (see kernels.hip header).
 struct WMat {

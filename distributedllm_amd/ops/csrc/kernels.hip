@@ -793,6 +793,30 @@ __device__ __forceinline__ void a_frag_q4(uint32_t q, uint32_t ab,
     }
 }
 
+// Byte-stream A fragment (W_Q8B): 8 re-biased u8 weights arrive as two
+// u32 with byte order [w0,w2,w1,w3] / [w4,w6,w5,w7], so the same mask
+// trick as q4 yields f16 pairs: 0x6400|b == f16(1024+b) exactly for
+// b in [0,255], then w = alpha*((1024+b) - 1152) + beta (csub -1152 =
+// 0xE480; repack re-biases q8_0 by +128 and q5_x by +112 so one csub
+// serves all three formats).
+__device__ __forceinline__ void a_frag_q8(uint32_t qA, uint32_t qB,
+                                          uint32_t ab, ABFrag& a) {
+    const __half2 abh = u2h2(ab);
+    const __half2 alpha2 = __half2half2(__low2half(abh));
+    const __half2 beta2 = __half2half2(__high2half(abh));
+    const __half2 c2 = u2h2(0xE480E480u);  // f16 -1152
+    uint32_t w[4];
+    w[0] = 0x64006400u | (qA & 0x00FF00FFu);
+    w[1] = 0x64006400u | ((qA >> 8) & 0x00FF00FFu);
+    w[2] = 0x64006400u | (qB & 0x00FF00FFu);
+    w[3] = 0x64006400u | ((qB >> 8) & 0x00FF00FFu);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+        const __half2 v = __hadd2(u2h2(w[i]), c2);
+        a.u[i] = h22u(__hfma2(v, alpha2, beta2));
+    }
+}
+
 // One wave's software-pipelined K loop over NM matrices sharing the B
 // panel (NM=2 for the FFN's w1/w3) and JT 16-token column tiles sharing
 // the A (weight) stream. JT is THE batched-decode lever: the HBM weight
@@ -828,6 +852,8 @@ __device__ __forceinline__ void wave_tile_kloop(
     kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4, NW);
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
+    // u32 per lane per K-block: 1 for q4 nibbles, 2 for the byte stream
+    constexpr int QW = (WT == W_Q8B) ? 2 : 1;
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
     // With >=4 independent MFMAs per K-block (JT/NM/RT product) the
     // accumulator reuse distance already covers the MFMA dependent
@@ -866,8 +892,8 @@ __device__ __forceinline__ void wave_tile_kloop(
         for (int n = 0; n < NM; ++n) {
             const int tr = tile_row * RT + rt;
             qp[rt][n] = (const uint32_t*)ws[n]->data +
-                        ((size_t)tr * nb + kl.kb0) * 64 +
-                        (kl.ks * 16 + kl.i) * 4;
+                        ((size_t)tr * nb + kl.kb0) * 64 * QW +
+                        (kl.ks * 16 + kl.i) * 4 * QW;
             abp[rt][n] = (const uint32_t*)ws[n]->scales +
                          ((size_t)tr * nb + kl.kb0) * 16 + kl.i * 4;
             tp[rt][n] = (const unsigned short*)ws[n]->data +
@@ -881,7 +907,7 @@ __device__ __forceinline__ void wave_tile_kloop(
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
     struct Batch {  // weight stream only (HBM, nt, double-buffered)
-        u32x4 q[PF / 4][RT][NM], ab[PF / 4][RT][NM];
+        u32x4 q[(PF / 4) * QW][RT][NM], ab[PF / 4][RT][NM];
         uint4 aw[PF][RT][NM];
     };
     // NAMED buffers, never indexed by a runtime value (a runtime select
@@ -906,9 +932,12 @@ __device__ __forceinline__ void wave_tile_kloop(
                                 *reinterpret_cast<const uint4*>(
                                     tp[rt][n] + (u4 * 4 + v) * 512);
                     } else {
-                        bt.q[u4][rt][n] = __builtin_nontemporal_load(
-                            reinterpret_cast<const u32x4*>(qp[rt][n]) +
-                            u4 * 64);
+#pragma unroll
+                        for (int qi = 0; qi < QW; ++qi)
+                            bt.q[u4 * QW + qi][rt][n] =
+                                __builtin_nontemporal_load(
+                                    reinterpret_cast<const u32x4*>(
+                                        qp[rt][n]) + u4 * 64 * QW + qi);
                         bt.ab[u4][rt][n] = __builtin_nontemporal_load(
                             reinterpret_cast<const u32x4*>(abp[rt][n]) +
                             u4 * 16);
@@ -922,7 +951,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                 if (WT == W_F16) {
                     tp[rt][n] += PF * 512;
                 } else {
-                    qp[rt][n] += PF * 64;
+                    qp[rt][n] += PF * 64 * QW;
                     abp[rt][n] += PF * 16;
                 }
             }
@@ -942,7 +971,7 @@ __device__ __forceinline__ void wave_tile_kloop(
         if (NORM) np += PF * 32;
     };
 
-    auto compute_one = [&](int parity, const uint32_t q[RT][NM],
+    auto compute_one = [&](int parity, const uint32_t q[RT][NM][QW],
                            const uint32_t ab[RT][NM],
                            const uint4 aw[RT][NM], const uint4 xb[JT],
                            const uint4& nbv) {
@@ -973,8 +1002,11 @@ __device__ __forceinline__ void wave_tile_kloop(
                 if (WT == W_F16) {
                     a.u[0] = aw[rt][n].x; a.u[1] = aw[rt][n].y;
                     a.u[2] = aw[rt][n].z; a.u[3] = aw[rt][n].w;
+                } else if (WT == W_Q8B) {
+                    a_frag_q8(q[rt][n][0], q[rt][n][QW - 1], ab[rt][n],
+                              a);
                 } else {
-                    a_frag_q4<WT>(q[rt][n], ab[rt][n], a);
+                    a_frag_q4<WT>(q[rt][n][0], ab[rt][n], a);
                 }
 #pragma unroll
                 for (int jt = 0; jt < JT; ++jt) {
@@ -989,12 +1021,19 @@ __device__ __forceinline__ void wave_tile_kloop(
     auto compute_batch = [&](Batch& bt, XPanel& px) {
 #pragma unroll
         for (int u = 0; u < PF; ++u) {
-            uint32_t q[RT][NM], ab[RT][NM];
+            uint32_t q[RT][NM][QW], ab[RT][NM];
 #pragma unroll
             for (int rt = 0; rt < RT; ++rt)
 #pragma unroll
                 for (int n = 0; n < NM; ++n) {
-                    q[rt][n] = bt.q[u / 4][rt][n][u % 4];
+                    if (WT == W_Q8B) {
+                        // lane's 8 u32 per group: [blk0.A, blk0.B, ...]
+                        q[rt][n][0] = bt.q[u / 2][rt][n][(u % 2) * 2];
+                        q[rt][n][QW - 1] =
+                            bt.q[u / 2][rt][n][(u % 2) * 2 + 1];
+                    } else {
+                        q[rt][n][0] = bt.q[u / 4][rt][n][u % 4];
+                    }
                     ab[rt][n] = bt.ab[u / 4][rt][n][u % 4];
                 }
             compute_one(u & 1, q, ab, bt.aw[u], px.xb[u], px.nbv[u]);
@@ -1024,7 +1063,7 @@ __device__ __forceinline__ void wave_tile_kloop(
         }
     }
     for (int g = kl.kb0 + nfull * PF; g < kl.kb1; ++g) {
-        uint32_t q[RT][NM], ab[RT][NM];
+        uint32_t q[RT][NM][QW], ab[RT][NM];
         uint4 aw[RT][NM];
         uint4 xb[JT], nbv;
 #pragma unroll
@@ -1043,8 +1082,11 @@ __device__ __forceinline__ void wave_tile_kloop(
                     aw[rt][n] = *reinterpret_cast<const uint4*>(tp[rt][n]);
                     tp[rt][n] += 512;
                 } else {
-                    q[rt][n] = __builtin_nontemporal_load(qp[rt][n]);
-                    qp[rt][n] += 64;
+#pragma unroll
+                    for (int qi = 0; qi < QW; ++qi)
+                        q[rt][n][qi] =
+                            __builtin_nontemporal_load(qp[rt][n] + qi);
+                    qp[rt][n] += 64 * QW;
                     ab[rt][n] = __builtin_nontemporal_load(abp[rt][n]);
                     abp[rt][n] += 16;
                 }
@@ -1765,6 +1807,11 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
         }                                        \
         case W_Q4_1: {                           \
             constexpr int WTc = W_Q4_1;          \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        case W_Q8B: {                            \
+            constexpr int WTc = W_Q8B;           \
             __VA_ARGS__;                         \
             break;                               \
         }                                        \

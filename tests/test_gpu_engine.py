@@ -51,18 +51,27 @@ def test_native_extension_is_loaded():
 @pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_0,
                                    ggml.FTYPE_MOSTLY_Q4_1,
                                    ggml.FTYPE_MOSTLY_F16,
-                                   ggml.FTYPE_ALL_F32])
+                                   ggml.FTYPE_ALL_F32,
+                                   ggml.FTYPE_MOSTLY_Q8_0,
+                                   ggml.FTYPE_MOSTLY_Q5_0,
+                                   ggml.FTYPE_MOSTLY_Q5_1])
 def test_embed_matches(ftype):
     f, hip, cpu = _engines(ftype=ftype)
     toks = torch.tensor([0, 5, 17, 255], dtype=torch.int32)
     a = hip.embed(toks.cuda()).cpu()
     b = cpu.embed(toks)
-    assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
+    # byte-format tables are dequantized to f16 at load (one rounding)
+    atol = 2e-3 if ftype in (ggml.FTYPE_MOSTLY_Q8_0, ggml.FTYPE_MOSTLY_Q5_0,
+                             ggml.FTYPE_MOSTLY_Q5_1) else 1e-6
+    assert torch.allclose(a, b, atol=atol), (a - b).abs().max()
 
 
 @pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_0,
                                    ggml.FTYPE_MOSTLY_Q4_1,
-                                   ggml.FTYPE_MOSTLY_F16])
+                                   ggml.FTYPE_MOSTLY_F16,
+                                   ggml.FTYPE_MOSTLY_Q8_0,
+                                   ggml.FTYPE_MOSTLY_Q5_0,
+                                   ggml.FTYPE_MOSTLY_Q5_1])
 def test_forward_prefill_matches(ftype):
     f, hip, cpu = _engines(ftype=ftype)
     hp = f.hparams
@@ -259,7 +268,8 @@ def test_pipeline_generate_on_gpu():
 
 
 @pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_0,
-                                   ggml.FTYPE_MOSTLY_F16])
+                                   ggml.FTYPE_MOSTLY_F16,
+                                   ggml.FTYPE_MOSTLY_Q8_0])
 def test_midsize_parity(ftype):
     """E=512 exercises the slab/RT kernel paths with realistic grids
     (the tiny presets run them at degenerate sizes)."""
