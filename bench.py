@@ -30,6 +30,8 @@ from distributedllm_amd.parallel.pipeline import (
     DecodePipeline, PipelineConfig, partition_layers, timed_decode)
 
 FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
+          "q5_0": ggml.FTYPE_MOSTLY_Q5_0, "q5_1": ggml.FTYPE_MOSTLY_Q5_1,
+          "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
           "f16": ggml.FTYPE_MOSTLY_F16, "f32": ggml.FTYPE_ALL_F32}
 
 
@@ -95,8 +97,9 @@ def main() -> int:
                     preset.n_embd * 4)  # per-lane clone, f16 K+V
         E, F, V = preset.n_embd, preset.n_ff, 32000
         n_weights = (max_layers * (4 * E * E + 3 * E * F) + 2 * V * E)
-        bpw = {"q4_0": 0.5625, "q4_1": 0.625, "f16": 2.0,
-               "f32": 4.0}[args.ftype]
+        bpw = {"q4_0": 0.5625, "q4_1": 0.625, "q5_0": 1.125,
+               "q5_1": 1.125, "q8_0": 1.125, "f16": 2.0,
+               "f32": 4.0}[args.ftype]  # byte formats: 1B + scales
         w_bytes = n_weights * bpw * 1.1  # repack padding/scales margin
         fit = max(1, int((260e9 - w_bytes) // max(kv_bytes, 1)))
         n_lanes = min(n_lanes, fit) if fit >= 3 else 1
