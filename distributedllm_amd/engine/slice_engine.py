@@ -24,7 +24,7 @@ from typing import Dict, Optional
 import numpy as np
 import torch
 
-from ..formats import ggml
+from ..formats import ggml, q4
 from ..models.llama import RMS_EPS, ROPE_BASE, rms_norm, rope_interleaved
 
 
