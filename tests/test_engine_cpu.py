@@ -83,3 +83,22 @@ class TestTorchEngineParity:
         y_ref = ref_s.forward(ref_e.embed([1, 5, 9]))
         lg_ref = ref_e.logits(y_ref)
         assert torch.allclose(lg, lg_ref, atol=1e-3)
+
+
+def test_byte_quant_repack_values_exact():
+    """The W_Q8B repack's re-biased bytes must reconstruct the codec's
+    dequantized weights exactly: w = alpha*(u - 128) + beta (the GPU
+    kernel computes the same quantity in packed f16)."""
+    import numpy as np
+    from distributedllm_amd.engine.slice_engine import _byte_values
+    from distributedllm_amd.formats import ggml, synthetic
+    for ft in (ggml.FTYPE_MOSTLY_Q5_0, ggml.FTYPE_MOSTLY_Q5_1,
+               ggml.FTYPE_MOSTLY_Q8_0):
+        f = synthetic.build_model("tiny", seed=0, ftype=ft)
+        t = next(x for x in f.tensors
+                 if x.name.endswith("attention.wq.weight"))
+        vals, alpha, beta = _byte_values(t)
+        w = (alpha[..., None] * (vals.astype(np.float32) - 128.0) +
+             beta[..., None])
+        ref = t.to_f32().reshape(w.shape[0], -1, 32)
+        assert np.abs(w - ref).max() == 0.0, ggml.TYPE_NAMES[t.gtype]
