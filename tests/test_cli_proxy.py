@@ -273,3 +273,29 @@ def test_status_config_cluster(node, tmp_path, capsys):
                             "--n-layer", "3"]) == 1
     out = capsys.readouterr().out
     assert '"connectivity": true' in out and "pipeline_ready: False" in out
+
+
+def test_cli_provision_q8_0_generate(node, tmp_path, capsys):
+    """Provisioning with a classic non-q4 quantization (q8_0) runs the
+    whole requantize -> slice -> push -> load -> generate flow (the
+    reference engine accepts q5/q8 GGJT files; so do our nodes)."""
+    addr = f"127.0.0.1:{node.port}"
+    root = tmp_path / "rootq8"
+    root.mkdir()
+    cfg = {"model_id": "tiny_q8",
+           "location": "synthetic:tiny",
+           "nodes_map": {addr: [0, PRESETS["tiny"].n_layer - 1]},
+           "quantization": "q8_0",
+           "metadata": {"name": "tinyq8", "family": "llama_v1"}}
+    cfg_path = tmp_path / "cfgq8.json"
+    cfg_path.write_text(json.dumps(cfg))
+    assert execute_command(["provision", str(cfg_path),
+                            "--root", str(root)]) == 0
+    from distributedllm_amd.formats import ggml
+    base = ggml.GGMLFile.load(
+        str(root / "models" / "tiny_q8" / "model_q8_0.bin"), extended=False)
+    assert base.hparams.ftype == ggml.FTYPE_MOSTLY_Q8_0
+    assert execute_command(["generate_text", str(cfg_path),
+                            "--prompt", "hi", "--num-tokens", "2",
+                            "--greedy", "--root", str(root)]) == 0
+    assert "hi" in capsys.readouterr().out
