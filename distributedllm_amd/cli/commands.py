@@ -224,6 +224,10 @@ class GenerateTextCommand(Command):
         p.add_argument("--rp", type=float, default=1.1,
                        help="repetition penalty")
         p.add_argument("--greedy", action="store_true")
+        p.add_argument("--top-k", type=int, default=0,
+                       help="sample from the k most likely tokens (0=off)")
+        p.add_argument("--top-p", type=float, default=1.0,
+                       help="nucleus sampling mass (1.0=off)")
         p.add_argument("--seed", type=int, default=None)
         p.add_argument("--root", default=".")
 
@@ -234,7 +238,8 @@ class GenerateTextCommand(Command):
         for piece in llm.generate(args.prompt, max_steps=args.num_tokens,
                                   temperature=args.temp,
                                   repeat_penalty=args.rp,
-                                  greedy=args.greedy, seed=args.seed):
+                                  greedy=args.greedy, seed=args.seed,
+                                  top_k=args.top_k, top_p=args.top_p):
             print(piece, end="", flush=True)
         print()
         r = llm.throughput.report()
@@ -298,6 +303,8 @@ class BatchGenerateCommand(Command):
         p.add_argument("--temp", type=float, default=0.7)
         p.add_argument("--rp", type=float, default=1.1)
         p.add_argument("--greedy", action="store_true")
+        p.add_argument("--top-k", type=int, default=0)
+        p.add_argument("--top-p", type=float, default=1.0)
         p.add_argument("--seed", type=int, default=None)
 
     def __call__(self, args) -> int:
@@ -329,7 +336,8 @@ class BatchGenerateCommand(Command):
         reqs = []
         for text in prompts:
             sampler = None if args.greedy else \
-                Sampler(args.temp, args.rp, seed=args.seed)
+                Sampler(args.temp, args.rp, seed=args.seed,
+                        top_k=args.top_k, top_p=args.top_p)
             reqs.append((text, bat.submit(tok.encode(text, bos=True),
                                           args.num_tokens,
                                           sampler=sampler)))

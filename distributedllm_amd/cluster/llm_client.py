@@ -71,13 +71,13 @@ class DistributedLLM:
 
     def generate(self, prompt: str, max_steps: int = 50,
                  temperature: float = 0.7, repeat_penalty: float = 1.1,
-                 seed: Optional[int] = None,
-                 greedy: bool = False) -> Iterator[str]:
+                 seed: Optional[int] = None, greedy: bool = False,
+                 top_k: int = 0, top_p: float = 1.0) -> Iterator[str]:
         self.clear_context()
         self.throughput.reset()
         tokens = self.tokenizer.encode(prompt, bos=True)
         sampler = Sampler(temperature, repeat_penalty, seed=seed,
-                          greedy=greedy)
+                          greedy=greedy, top_k=top_k, top_p=top_p)
         cur = tokens
         for _ in range(max_steps):
             x = self._embed(cur)

@@ -8,7 +8,9 @@
 * all logits divide by (temperature + 1e-5),
 * softmax then categorical draw (numpy RNG).
 
-A greedy mode mirrors the engine's device argmax semantics.
+A greedy mode mirrors the engine's device argmax semantics. Optional
+``top_k``/``top_p`` (nucleus) filtering extends the reference surface
+(off by default — defaults are exact reference behavior).
 """
 from __future__ import annotations
 
@@ -27,11 +29,14 @@ class Sampler:
     EPS = 1e-5
 
     def __init__(self, temperature: float = 0.7, repeat_penalty: float = 1.1,
-                 seed: Optional[int] = None, greedy: bool = False):
+                 seed: Optional[int] = None, greedy: bool = False,
+                 top_k: int = 0, top_p: float = 1.0):
         self.T = temperature
         self.penalty = repeat_penalty
         self.previous_ids: List[int] = []
         self.greedy = greedy
+        self.top_k = top_k      # 0 = off
+        self.top_p = top_p      # 1.0 = off
         self.rng = np.random.default_rng(seed)
 
     def __call__(self, logits) -> int:
@@ -44,6 +49,19 @@ class Sampler:
         mask = np.isin(np.arange(size), self.previous_ids)
         penalties = (mask * self.penalty + ~mask) * (self.T + self.EPS)
         probs = softmax(logits / penalties)
+        if self.top_k and self.top_k < size:
+            kth = np.partition(probs, -self.top_k)[-self.top_k]
+            probs = np.where(probs >= kth, probs, 0.0)
+        if self.top_p < 1.0:
+            order = np.argsort(-probs)
+            csum = np.cumsum(probs[order])
+            # keep the smallest prefix whose mass reaches top_p
+            cut = int(np.searchsorted(csum, self.top_p) + 1)
+            keep = np.zeros(size, dtype=bool)
+            keep[order[:cut]] = True
+            probs = np.where(keep, probs, 0.0)
+        if self.top_k or self.top_p < 1.0:
+            probs = probs / probs.sum()
         tid = int(self.rng.choice(size, p=probs))
         self.previous_ids.append(tid)
         return tid

@@ -118,3 +118,29 @@ def test_metrics_meters():
     t.start("a"); time.sleep(0.005); t.stop("a")
     rep = t.report()
     assert rep["a"]["count"] == 1 and rep["a"]["seconds"] >= 0.004
+
+
+def test_sampler_top_k_top_p():
+    """top_k/top_p are opt-in filters on the reference sampler: top_k=1
+    degenerates to greedy; top_p keeps only the nucleus; defaults leave
+    the distribution untouched (byte-compatible with the reference)."""
+    import numpy as np
+    from distributedllm_amd.engine.sampler import Sampler
+    logits = np.array([5.0, 4.0, 1.0, -2.0, 0.5])
+
+    s = Sampler(temperature=1.0, repeat_penalty=1.0, seed=0, top_k=1)
+    assert s(logits) == 0  # only the argmax survives
+
+    # top_p tight enough to keep just the two dominant tokens
+    draws = set()
+    s2 = Sampler(temperature=1.0, repeat_penalty=1.0, seed=1, top_p=0.9)
+    for _ in range(50):
+        s2.previous_ids = []  # isolate the filter from the penalty
+        draws.add(s2(logits))
+    assert draws <= {0, 1}
+
+    # defaults = reference behavior (identical draw stream)
+    a = Sampler(temperature=0.8, repeat_penalty=1.1, seed=7)
+    b = Sampler(temperature=0.8, repeat_penalty=1.1, seed=7,
+                top_k=0, top_p=1.0)
+    assert [a(logits) for _ in range(5)] == [b(logits) for _ in range(5)]
