@@ -356,6 +356,47 @@ class BatchGenerateCommand(Command):
         return 0
 
 
+class ServeHttpCommand(Command):
+    name = "serve_http"
+    help = ("HTTP serving front (FastAPI/uvicorn) over the continuous "
+            "batcher: POST /generate, GET /health")
+
+    def configure(self, p: argparse.ArgumentParser) -> None:
+        p.add_argument("model", help="GGML model file (full, not a slice)")
+        p.add_argument("--host", default="127.0.0.1")
+        p.add_argument("--port", type=int, default=8080)
+        p.add_argument("--slots", type=int, default=64)
+        p.add_argument("--lanes", type=int, default=1,
+                       help="stream lanes (weight-sharing clones; GPU)")
+        p.add_argument("--ctx", type=int, default=2048)
+
+    def __call__(self, args) -> int:
+        import uvicorn
+
+        from ..engine import engine_for_slice
+        from ..engine.tokenizer import Tokenizer
+        from ..formats import ggml, slicer
+        from ..serving import ContinuousBatcher, build_http_app
+
+        f = ggml.GGMLFile.load(args.model,
+                               extended=ggml.sniff_extended(args.model))
+        eng = engine_for_slice(f, n_ctx=args.ctx, max_batch=args.slots)
+        eng.attach_extra(slicer.make_extra_layers(f))
+        lanes = None
+        if args.lanes > 1 and hasattr(eng, "clone_shared"):
+            lanes = [eng] + [eng.clone_shared()
+                             for _ in range(args.lanes - 1)]
+        tok = Tokenizer(f.vocab)
+        bat = ContinuousBatcher(eng, engines=lanes)
+        app, worker = build_http_app(bat, tok)
+        try:
+            uvicorn.run(app, host=args.host, port=args.port,
+                        log_level="warning")
+        finally:
+            worker.stop()
+        return 0
+
+
 class RunProxyCommand(Command):
     name = "run_proxy"
     help = "Run the NAT-traversal proxy (bridges clients to a reverse node)"

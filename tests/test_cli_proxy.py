@@ -43,7 +43,7 @@ def test_all_commands_registered():
     assert set(commands) == {
         "provision", "run_node", "status", "push_slice", "load_slice",
         "list_slices", "generate_text", "perplexity", "run_proxy",
-        "batch_generate"}
+        "batch_generate", "serve_http"}
     build_parser()  # parser builds without error
 
 
