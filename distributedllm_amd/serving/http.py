@@ -96,8 +96,7 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
     def health():
         return {"status": "ok", "pending": batcher.pending}
 
-    @app.post("/generate")
-    def generate(r: GenerateRequest):
+    def _submit(r: GenerateRequest):
         if r.num_tokens < 1:
             raise HTTPException(422, "num_tokens must be >= 1")
         ids = tokenizer.encode(r.prompt, bos=True)
@@ -105,11 +104,52 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
         if r.temperature > 0.0:
             sampler = Sampler(r.temperature, r.repeat_penalty, seed=r.seed,
                               top_k=r.top_k, top_p=r.top_p)
+        return ids, sampler
+
+    @app.post("/generate")
+    def generate(r: GenerateRequest):
+        ids, sampler = _submit(r)
         try:
             req = worker.submit_and_wait(ids, r.num_tokens, sampler)
         except TimeoutError as e:
             raise HTTPException(504, str(e))
         return {"text": tokenizer.decode(req.out), "tokens": req.out}
+
+    @app.post("/generate_stream")
+    def generate_stream(r: GenerateRequest):
+        """Server-sent events: one `data:` line per decoded token as the
+        batcher produces it, then a final `done` event with the full
+        text — tokens stream while OTHER requests share the same decode
+        steps."""
+        import json as _json
+
+        from fastapi.responses import StreamingResponse
+
+        ids, sampler = _submit(r)
+        with worker.cond:
+            req = batcher.submit(ids, r.num_tokens, sampler=sampler)
+            worker.cond.notify_all()
+
+        def events():
+            sent = 0
+            while True:
+                with worker.cond:
+                    while len(req.out) == sent and not req.done:
+                        worker.cond.wait(timeout=1.0)
+                    chunk = list(req.out[sent:])
+                    sent = len(req.out)
+                    done = req.done
+                for tid in chunk:
+                    yield ("data: " + _json.dumps(
+                        {"token": tid,
+                         "piece": tokenizer.decode_token(tid)}) + "\n\n")
+                if done:
+                    yield ("event: done\ndata: " + _json.dumps(
+                        {"text": tokenizer.decode(req.out),
+                         "tokens": list(req.out)}) + "\n\n")
+                    return
+
+        return StreamingResponse(events(), media_type="text/event-stream")
 
     app.state.worker = worker
     return app, worker

@@ -66,3 +66,30 @@ def test_sampled_generation_and_validation(client_and_worker):
     assert r.status_code == 200 and len(r.json()["tokens"]) == 4
     assert c.post("/generate", json={"prompt": "x",
                                      "num_tokens": 0}).status_code == 422
+
+
+def test_streaming_generation(client_and_worker):
+    """SSE stream: one data event per token, final done event carries
+    the full text, and the stream equals the non-streaming result."""
+    import json
+    c, _ = client_and_worker
+    want = c.post("/generate", json={"prompt": "hello",
+                                     "num_tokens": 5}).json()
+    pieces, final = [], None
+    with c.stream("POST", "/generate_stream",
+                  json={"prompt": "hello", "num_tokens": 5}) as r:
+        assert r.status_code == 200
+        event = None
+        for line in r.iter_lines():
+            if line.startswith("event:"):
+                event = line.split(":", 1)[1].strip()
+            elif line.startswith("data:"):
+                d = json.loads(line.split(":", 1)[1])
+                if event == "done":
+                    final = d
+                else:
+                    pieces.append(d["token"])
+                    event = None
+    assert pieces == want["tokens"]
+    assert final is not None and final["tokens"] == want["tokens"]
+    assert final["text"] == want["text"]
