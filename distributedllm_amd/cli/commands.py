@@ -366,11 +366,13 @@ class ServeHttpCommand(Command):
         p.add_argument("--host", default="127.0.0.1")
         p.add_argument("--port", type=int, default=8080)
         p.add_argument("--slots", type=int, default=64)
-        p.add_argument("--lanes", type=int, default=1,
-                       help="stream lanes (weight-sharing clones; GPU)")
+        p.add_argument("--lanes", type=int, default=0,
+                       help="stream lanes (weight-sharing clones; GPU); "
+                            "0 = auto from HBM fit (>=3 or 1, max 5)")
         p.add_argument("--ctx", type=int, default=2048)
 
     def __call__(self, args) -> int:
+        import torch
         import uvicorn
 
         from ..engine import engine_for_slice
@@ -380,12 +382,21 @@ class ServeHttpCommand(Command):
 
         f = ggml.GGMLFile.load(args.model,
                                extended=ggml.sniff_extended(args.model))
+        n_lanes = args.lanes
+        if n_lanes == 0:
+            n_lanes = 1
+            if torch.cuda.is_available():
+                hp = f.hparams
+                kv = hp.n_layer * args.slots * args.ctx * hp.n_embd * 4
+                w_bytes = sum(len(t.raw) for t in f.tensors) * 2.2
+                fit = max(1, int((260e9 - w_bytes) // max(kv, 1)))
+                n_lanes = min(5, fit) if fit >= 3 else 1
         eng = engine_for_slice(f, n_ctx=args.ctx, max_batch=args.slots)
         eng.attach_extra(slicer.make_extra_layers(f))
         lanes = None
-        if args.lanes > 1 and hasattr(eng, "clone_shared"):
+        if n_lanes > 1 and hasattr(eng, "clone_shared"):
             lanes = [eng] + [eng.clone_shared()
-                             for _ in range(args.lanes - 1)]
+                             for _ in range(n_lanes - 1)]
         tok = Tokenizer(f.vocab)
         bat = ContinuousBatcher(eng, engines=lanes)
         app, worker = build_http_app(bat, tok)
