@@ -49,7 +49,7 @@ def quantize_q4_0(x: np.ndarray) -> np.ndarray:
     m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
     d = (m / -8.0).astype(np.float16)
     df = d.astype(np.float32)
-    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
     q = np.clip(np.rint(b * inv[..., None]) + 8, 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
     packed = (lo | (hi << 4)).astype(np.uint8)
@@ -89,7 +89,7 @@ def quantize_q4_1(x: np.ndarray) -> np.ndarray:
     d = ((mx - mn) / 15.0).astype(np.float16)
     m = mn.astype(np.float16)
     df = d.astype(np.float32)
-    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
     q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) * inv[..., None]),
                 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
@@ -166,7 +166,7 @@ def quantize_q5_0(x: np.ndarray) -> np.ndarray:
     m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
     d = (m / -16.0).astype(np.float16)
     df = d.astype(np.float32)
-    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
     q = np.clip(np.rint(b * inv[..., None]) + 16, 0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)
     out = np.empty(b.shape[:2] + (Q5_0_BLOCK_BYTES,), dtype=np.uint8)
@@ -196,7 +196,7 @@ def quantize_q5_1(x: np.ndarray) -> np.ndarray:
     d = ((mx - mn) / 31.0).astype(np.float16)
     m = mn.astype(np.float16)
     df = d.astype(np.float32)
-    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
     q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) *
                         inv[..., None]), 0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)
@@ -227,7 +227,7 @@ def quantize_q8_0(x: np.ndarray) -> np.ndarray:
     amax = np.abs(b).max(axis=-1)
     d = (amax / 127.0).astype(np.float16)
     df = d.astype(np.float32)
-    inv = np.where(df != 0.0, 1.0 / df, 0.0)
+    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
     q = np.rint(b * inv[..., None]).astype(np.int8)
     out = np.empty(b.shape[:2] + (Q8_0_BLOCK_BYTES,), dtype=np.uint8)
     out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))

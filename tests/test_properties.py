@@ -80,3 +80,45 @@ def test_protocol_roundtrip_arbitrary_values(name, blob, num, arr):
     got = P.receive_message(sock)
     assert got.axis0 == m2.axis0 and got.start_pos == m2.start_pos
     np.testing.assert_array_equal(got.values, m2.values)
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.floats(min_value=-1e4, max_value=1e4, width=32),
+                min_size=32, max_size=96).filter(lambda v: len(v) % 32 == 0))
+def test_q8_0_roundtrip_error_bound(vals):
+    x = np.array(vals, dtype=np.float32).reshape(1, -1)
+    deq = q4.dequantize_q8_0(q4.quantize_q8_0(x), x.shape[-1])
+    for b in range(x.shape[-1] // 32):
+        blk = x[0, b * 32:(b + 1) * 32]
+        step = max(abs(float(np.float16(np.max(np.abs(blk)) / 127.0))),
+                   1e-12)
+        err = np.max(np.abs(deq[0, b * 32:(b + 1) * 32] - blk))
+        # 0.5*step from rounding q, + d's own f16 rounding amplified by
+        # |q| <= 127: relative 2^-11 for normal d, up to half the min
+        # f16 subnormal (2.98e-8) absolute for tiny scales
+        assert err <= step * 0.57 + 127 * 3.0e-8 + 1e-6 * step
+
+
+@settings(max_examples=50, deadline=None)
+@given(st.lists(st.floats(min_value=-1e4, max_value=1e4, width=32),
+                min_size=32, max_size=96).filter(lambda v: len(v) % 32 == 0),
+       st.sampled_from(["q5_0", "q5_1"]))
+def test_q5_roundtrip_fixed_point_and_bits(vals, which):
+    """Quantization is a fixed point (requantizing the dequantized values
+    reproduces the bytes) and every decoded 5-bit value is in [0, 31]."""
+    x = np.array(vals, dtype=np.float32).reshape(1, -1)
+    n = x.shape[-1]
+    quant = q4.quantize_q5_0 if which == "q5_0" else q4.quantize_q5_1
+    deq = q4.dequantize_q5_0 if which == "q5_0" else q4.dequantize_q5_1
+    raw = quant(x)
+    y = deq(raw, n)
+    # byte-level idempotence can flip the sign of a ZERO scale in an
+    # all-zero block (d = -0.0 vs +0.0 — llama.cpp float math does the
+    # same); the VALUE fixed point is the real invariant
+    assert np.array_equal(deq(quant(y), n), y)
+    bs = q4.Q5_0_BLOCK_BYTES if which == "q5_0" else q4.Q5_1_BLOCK_BYTES
+    hoff = 2 if which == "q5_0" else 4
+    b = raw.reshape(1, -1, bs)
+    q5 = q4._unpack_q5(b[..., hoff + 4:],
+                       np.ascontiguousarray(b[..., hoff:hoff + 4]))
+    assert q5.max() <= 31
