@@ -49,17 +49,26 @@ def main():
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     steps = 0
+    done_at = {}
     while bat.pending:
-        bat.step()
+        fin = bat.step()
+        if fin:
+            torch.cuda.synchronize()
+            now = time.perf_counter() - t0
+            for r in fin:
+                done_at[r.rid] = now
         steps += 1
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     total = sum(len(r.out) for r in reqs)
+    lat = sorted(done_at.values())
+    p = lambda q: lat[min(len(lat) - 1, int(q * len(lat)))]
     print(f"{args.requests} requests x {args.num_tokens} new tokens "
           f"(prompt {args.prompt_len}, {bat.n_slots} slots on "
           f"{len(bat.lanes)} lanes): "
           f"{total} tokens in {dt:.2f}s = {total/dt:.0f} tok/s, "
-          f"{steps} decode steps")
+          f"{steps} decode steps; completion p50={p(0.5):.2f}s "
+          f"p95={p(0.95):.2f}s max={lat[-1]:.2f}s")
 
 
 if __name__ == "__main__":
