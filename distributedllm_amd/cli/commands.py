@@ -277,9 +277,14 @@ class PerplexityCommand(Command):
                 text = f.read()
         if args.dataset:
             text = _random_dataset_prompt(args.dataset, args.seed)
-        from ..cluster.llm_client import get_llm
-        llm = get_llm(args.config, root=args.root)
-        ppl = llm.perplexity(text)
+        if args.config.endswith(".bin"):
+            # local mode: evaluate directly on a GGML model file (no
+            # cluster needed; same NLL math as the distributed client)
+            ppl = _local_perplexity(args.config, text)
+        else:
+            from ..cluster.llm_client import get_llm
+            llm = get_llm(args.config, root=args.root)
+            ppl = llm.perplexity(text)
         print(f"perplexity: {ppl:.4f}")
         return 0
 
@@ -406,6 +411,35 @@ class ServeHttpCommand(Command):
         finally:
             worker.stop()
         return 0
+
+
+def _local_perplexity(model_path: str, text: str) -> float:
+    """exp(mean NLL) on a local engine (reference semantics,
+    common.py:113-141 — identical math to DistributedLLM.perplexity)."""
+    import numpy as np
+    import torch
+
+    from ..engine import engine_for_slice
+    from ..engine.tokenizer import Tokenizer
+    from ..formats import ggml, slicer
+
+    f = ggml.GGMLFile.load(model_path,
+                           extended=ggml.sniff_extended(model_path))
+    tok = Tokenizer(f.vocab)
+    tokens = tok.encode(text, bos=True)
+    if len(tokens) < 2:
+        raise ValueError("perplexity needs at least 2 tokens")
+    eng = engine_for_slice(f, n_ctx=max(len(tokens) + 1, 16), max_batch=1)
+    eng.attach_extra(slicer.make_extra_layers(f))
+    dev = getattr(eng, "device", "cpu")
+    ids = torch.tensor(tokens[:-1], dtype=torch.int32, device=dev)
+    pos = torch.arange(len(tokens) - 1, dtype=torch.int32, device=dev)
+    seq = torch.zeros(len(tokens) - 1, dtype=torch.int32, device=dev)
+    y = eng.forward(eng.embed(ids), pos, seq)
+    lg = eng.logits(y, all_logits=True)
+    logp = torch.log_softmax(lg.float().cpu(), dim=-1).numpy()
+    nll = [-logp[i, tokens[i + 1]] for i in range(len(tokens) - 1)]
+    return float(np.exp(np.mean(nll)))
 
 
 class RunProxyCommand(Command):

@@ -164,3 +164,16 @@ def test_random_arrivals_match_canonical():
     for prompt, steps, r in cases:
         assert r.done and r.out == _canonical(prompt, steps), \
             (prompt, steps, r.out)
+
+
+def test_local_perplexity_cli(tmp_path, capsys):
+    """perplexity accepts a local GGML file (no cluster): same NLL math
+    as the distributed client on the same bytes."""
+    from distributedllm_amd.cli import execute_command
+    p = tmp_path / "tiny.bin"
+    synthetic.build_model("tiny", seed=0).save(str(p))
+    assert execute_command(["perplexity", str(p),
+                            "--prompt", "hello world of words"]) == 0
+    out = capsys.readouterr().out
+    ppl = float(out.split("perplexity:")[1].strip())
+    assert ppl > 1.0
