@@ -91,10 +91,23 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
 
     app = FastAPI(title="distllm-mi355x")
     worker = BatcherWorker(batcher).start()
+    stats = {"requests": 0, "tokens": 0, "t0": time.monotonic()}
 
     @app.get("/health")
     def health():
         return {"status": "ok", "pending": batcher.pending}
+
+    @app.get("/metrics")
+    def metrics():
+        up = time.monotonic() - stats["t0"]
+        return {"requests_served": stats["requests"],
+                "tokens_generated": stats["tokens"],
+                "pending": batcher.pending,
+                "slots": batcher.n_slots,
+                "lanes": len(batcher.lanes),
+                "uptime_s": round(up, 1),
+                "tokens_per_s_lifetime": round(
+                    stats["tokens"] / max(up, 1e-9), 1)}
 
     def _submit(r: GenerateRequest):
         if r.num_tokens < 1:
@@ -113,6 +126,8 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
             req = worker.submit_and_wait(ids, r.num_tokens, sampler)
         except TimeoutError as e:
             raise HTTPException(504, str(e))
+        stats["requests"] += 1
+        stats["tokens"] += len(req.out)
         return {"text": tokenizer.decode(req.out), "tokens": req.out}
 
     @app.post("/generate_stream")

@@ -203,6 +203,24 @@ class ContinuousBatcher:
                 r.slot = -1
                 finished.append(r)
 
+    def cancel(self, req: Request) -> bool:
+        """Abort a request: drop it from the queue, or free its slot if
+        active (the slot's KV rows are dead — safe to reuse, attention
+        is pos-bounded). Returns False if it already finished."""
+        if req.done:
+            return False
+        if req.slot >= 0 and req.slot in self.active:
+            del self.active[req.slot]
+            self.free.append(req.slot)
+            req.slot = -1
+        else:
+            try:
+                self.queue.remove(req)
+            except ValueError:
+                return False
+        req.done = True
+        return True
+
     # --------------------------------------------------------- convenience
 
     @property

@@ -93,3 +93,11 @@ def test_streaming_generation(client_and_worker):
     assert pieces == want["tokens"]
     assert final is not None and final["tokens"] == want["tokens"]
     assert final["text"] == want["text"]
+
+
+def test_metrics_endpoint(client_and_worker):
+    c, _ = client_and_worker
+    c.post("/generate", json={"prompt": "m", "num_tokens": 3})
+    m = c.get("/metrics").json()
+    assert m["requests_served"] >= 1 and m["tokens_generated"] >= 3
+    assert m["slots"] >= 1 and m["lanes"] == 1

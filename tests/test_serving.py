@@ -177,3 +177,19 @@ def test_local_perplexity_cli(tmp_path, capsys):
     out = capsys.readouterr().out
     ppl = float(out.split("perplexity:")[1].strip())
     assert ppl > 1.0
+
+
+def test_cancel_queued_and_active():
+    bat = ContinuousBatcher(_engine(1))
+    r1 = bat.submit(PROMPTS[0], 10)
+    r2 = bat.submit(PROMPTS[1], 10)   # queued behind r1 (1 slot)
+    bat.step()
+    assert r1.slot >= 0 and len(bat.queue) == 1
+    assert bat.cancel(r2) and r2.done  # cancel while queued
+    assert bat.cancel(r1) and r1.done  # cancel while active, frees slot
+    assert bat.free == [0] and bat.pending == 0
+    # a new request reuses the freed slot and decodes correctly
+    r3 = bat.submit(PROMPTS[2], STEPS[2])
+    bat.run_all(max_steps=50)
+    assert r3.out == _canonical(PROMPTS[2], STEPS[2])
+    assert not bat.cancel(r3)  # already finished
