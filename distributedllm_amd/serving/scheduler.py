@@ -40,6 +40,7 @@ class Request:
     slot: int = -1        # KV slot while active
     _next_tok: int = -1   # token to feed at _pos on the next step
     _pos: int = -1
+    _pf_pos: int = 0      # prompt tokens already prefilled (chunked)
 
 
 class ContinuousBatcher:
@@ -52,12 +53,19 @@ class ContinuousBatcher:
     """
 
     def __init__(self, engine, max_slots: Optional[int] = None,
-                 eos_id: Optional[int] = None, engines=None):
+                 eos_id: Optional[int] = None, engines=None,
+                 prefill_chunk: Optional[int] = None):
         """engines: optional list of k weight-sharing clones ("lanes");
         global slot s lives on lane s%k as that clone's local slot s//k,
         and on CUDA each lane's forward runs on its own HIP stream —
         the decode chain is latency-bound, so concurrent lanes overlap
-        (same mechanics as the pipeline's stream lanes)."""
+        (same mechanics as the pipeline's stream lanes).
+
+        prefill_chunk: max prompt tokens prefilled per lane per step
+        (None = unbounded, maximizes throughput). A bound interleaves
+        long-prompt admission with decode steps, so in-flight requests
+        keep producing tokens while a long prompt loads (latency
+        fairness — Sarathi-style chunked prefill)."""
         self.engine = engine
         self.lanes = engines if engines else [engine]
         k = len(self.lanes)
@@ -73,6 +81,8 @@ class ContinuousBatcher:
         self.queue: deque[Request] = deque()
         self._next_rid = 0
         self._dev = getattr(engine, "device", "cpu")
+        self.prefill_chunk = prefill_chunk
+        self.prefilling: Dict[int, Request] = {}  # slot -> request
         self._streams = None
         if k > 1 and self._dev == "cuda":
             self._streams = [torch.cuda.Stream() for _ in self.lanes]
@@ -98,31 +108,46 @@ class ContinuousBatcher:
         eng = self.engine  # n_ctx/bounds source; lanes share hparams
         n_ctx = getattr(eng, "n_ctx", 1 << 30)
         k = len(self.lanes)
-        # admissions grouped per lane, prefilled as ONE concatenated
-        # (token, pos, seq) stream per lane — the engine tiles any
-        # mixed-sequence stream internally, so several short prompts
-        # share each prefill launch instead of paying one kernel chain
-        # per request
-        per_lane: List[List[Request]] = [[] for _ in range(k)]
+        # take queued requests into free slots; their prompt body
+        # prefills below (possibly across several steps when chunked)
         while self.free and self.queue:
             r = self.queue.popleft()
             r.slot = self.free.pop()
             assert len(r.prompt) + r.max_new <= n_ctx, \
                 f"request {r.rid}: prompt+max_new exceeds n_ctx={n_ctx}"
-            r._next_tok = r.prompt[-1]
-            r._pos = len(r.prompt) - 1
-            self.active[r.slot] = r
+            r._pf_pos = 0
             if len(r.prompt) > 1:
-                per_lane[r.slot % k].append(r)
+                self.prefilling[r.slot] = r
+            else:
+                self._activate(r)
+        # prefill as ONE concatenated (token, pos, seq) stream per lane
+        # — the engine tiles any mixed-sequence stream internally, so
+        # several prompts share each prefill launch; with prefill_chunk
+        # set, each lane advances at most that many prompt tokens per
+        # step and decode of OTHER requests interleaves
+        budget_full = self.prefill_chunk
+        per_lane: List[List[Request]] = [[] for _ in range(k)]
+        for slot in sorted(self.prefilling):
+            per_lane[slot % k].append(self.prefilling[slot])
         for j, reqs in enumerate(per_lane):
             if not reqs:
                 continue
+            budget = budget_full
             toks, pos, seq = [], [], []
             for r in reqs:
-                body = r.prompt[:-1]
-                toks += body
-                pos += list(range(len(body)))
-                seq += [r.slot // k] * len(body)
+                end = len(r.prompt) - 1
+                take = end - r._pf_pos
+                if budget is not None:
+                    take = min(take, budget)
+                    budget -= take
+                if take <= 0:
+                    continue
+                toks += r.prompt[r._pf_pos:r._pf_pos + take]
+                pos += list(range(r._pf_pos, r._pf_pos + take))
+                seq += [r.slot // k] * take
+                r._pf_pos += take
+            if not toks:
+                continue
             lane_eng = self.lanes[j]
             # prefill on the lane's stream: the lane's next decode
             # launch must observe these KV writes, and same-stream
@@ -137,6 +162,16 @@ class ContinuousBatcher:
                 q = torch.tensor(seq, dtype=torch.int32,
                                  device=self._dev)
                 lane_eng.forward(lane_eng.embed(t), p, q)
+        for slot in list(self.prefilling):
+            r = self.prefilling[slot]
+            if r._pf_pos >= len(r.prompt) - 1:
+                del self.prefilling[slot]
+                self._activate(r)
+
+    def _activate(self, r: Request) -> None:
+        r._next_tok = r.prompt[-1]
+        r._pos = len(r.prompt) - 1
+        self.active[r.slot] = r
 
     # --------------------------------------------------------------- step
 
@@ -209,8 +244,10 @@ class ContinuousBatcher:
         is pos-bounded). Returns False if it already finished."""
         if req.done:
             return False
-        if req.slot >= 0 and req.slot in self.active:
-            del self.active[req.slot]
+        if req.slot >= 0 and (req.slot in self.active or
+                              req.slot in self.prefilling):
+            self.active.pop(req.slot, None)
+            self.prefilling.pop(req.slot, None)
             self.free.append(req.slot)
             req.slot = -1
         else:
@@ -225,7 +262,7 @@ class ContinuousBatcher:
 
     @property
     def pending(self) -> int:
-        return len(self.queue) + len(self.active)
+        return len(self.queue) + len(self.active) + len(self.prefilling)
 
     def run_all(self, max_steps: int = 1 << 20) -> List[Request]:
         """Drive step() until every submitted request finishes."""

@@ -193,3 +193,25 @@ def test_cancel_queued_and_active():
     bat.run_all(max_steps=50)
     assert r3.out == _canonical(PROMPTS[2], STEPS[2])
     assert not bat.cancel(r3)  # already finished
+
+
+def test_chunked_prefill_interleaves_and_stays_exact():
+    """prefill_chunk bounds prompt tokens per lane per step: a long
+    prompt admits over several steps while an in-flight request keeps
+    decoding — and both requests still decode their canonical tokens."""
+    eng = _engine(2)
+    bat = ContinuousBatcher(eng, prefill_chunk=2)
+    long_prompt = [2, 11, 4, 6, 1, 7, 9]  # body of 6 -> 3 chunked steps
+    r_short = bat.submit([7], 6)
+    bat.step()  # r_short active and decoding
+    assert len(r_short.out) == 1
+    r_long = bat.submit(long_prompt, 3)
+    produced = []
+    while not r_long.done:
+        bat.step()
+        produced.append(len(r_short.out))
+    # the short request kept producing tokens during the long prefill
+    assert produced[0] > 1 and len(r_short.out) >= 4
+    bat.run_all(max_steps=30)
+    assert r_short.out == _canonical([7], 6)
+    assert r_long.out == _canonical(long_prompt, 3)
