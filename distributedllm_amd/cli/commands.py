@@ -375,6 +375,10 @@ class ServeHttpCommand(Command):
                        help="stream lanes (weight-sharing clones; GPU); "
                             "0 = auto from HBM fit (>=3 or 1, max 5)")
         p.add_argument("--ctx", type=int, default=2048)
+        p.add_argument("--prefill-chunk", type=int, default=64,
+                       help="max prompt tokens prefilled per lane per "
+                            "decode step (latency fairness; 0 = "
+                            "unbounded)")
 
     def __call__(self, args) -> int:
         import torch
@@ -403,7 +407,8 @@ class ServeHttpCommand(Command):
             lanes = [eng] + [eng.clone_shared()
                              for _ in range(n_lanes - 1)]
         tok = Tokenizer(f.vocab)
-        bat = ContinuousBatcher(eng, engines=lanes)
+        bat = ContinuousBatcher(eng, engines=lanes,
+                                prefill_chunk=args.prefill_chunk or None)
         app, worker = build_http_app(bat, tok)
         try:
             uvicorn.run(app, host=args.host, port=args.port,
