@@ -425,6 +425,11 @@ class SliceEngine {
     int64_t max_tokens() const { return kMaxTokens; }
     int64_t n_ctx() const { return ctx_; }
     int64_t max_batch() const { return B_; }
+    // the per-layer KV tensors ([L, B, ctx, E] f16); exposed so the
+    // Python prefill fast path (slice_engine.forward large-T branch)
+    // can append rows the kernel decode path then attends over
+    torch::Tensor k_cache() const { return k_cache_; }
+    torch::Tensor v_cache() const { return v_cache_; }
 
  private:
     void forward_legacy(hipStream_t s, float* xp, const int* pp,
@@ -488,5 +493,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         .def("argmax", &SliceEngine::argmax)
         .def_property_readonly("max_tokens", &SliceEngine::max_tokens)
         .def_property_readonly("n_ctx", &SliceEngine::n_ctx)
-        .def_property_readonly("max_batch", &SliceEngine::max_batch);
+        .def_property_readonly("max_batch", &SliceEngine::max_batch)
+        .def_property_readonly("k_cache", &SliceEngine::k_cache)
+        .def_property_readonly("v_cache", &SliceEngine::v_cache);
 }

@@ -439,3 +439,45 @@ def test_slice_chain_matches_monolith_on_gpu():
 
     _assert_close(lg_chain, lg_mono, rel_rms=1e-3, rel_max=1e-2,
                   label="slice chain vs monolith")
+
+
+def test_rocblas_prefill_path_matches_cpu():
+    """T >= PREFILL_MIN_T takes the rocBLAS prefill path (detiled f16
+    weights, library GEMMs, torch attention over the engine's KV) —
+    must match the fp32 CPU reference like the kernel path does, and
+    the kernel decode path must attend seamlessly over its KV rows."""
+    f, hip, cpu = _engines(preset="small", n_ctx=256, max_batch=2)
+    hp = f.hparams
+    assert hip._prefill_ok()
+    torch.manual_seed(11)
+    T = 160  # >= PREFILL_MIN_T -> prefill branch
+    x = torch.randn(T, hp.n_embd) * 0.5
+    pos = torch.arange(T, dtype=torch.int32)
+    seq = torch.zeros(T, dtype=torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label="rocblas prefill")
+    # decode the next token through the KERNEL path on the same KV
+    xd = torch.randn(1, hp.n_embd) * 0.5
+    pd = torch.tensor([T], dtype=torch.int32)
+    sd = torch.zeros(1, dtype=torch.int32)
+    y2_gpu = hip.forward(xd.cuda(), pd.cuda(), sd.cuda(),
+                         decode=True).cpu()
+    y2_cpu = cpu.forward(xd.clone(), pd, sd)
+    _assert_close(y2_gpu, y2_cpu, label="decode after rocblas prefill")
+
+
+def test_rocblas_prefill_mixed_stream():
+    """The batched-admission shape: two sequences' prompts concatenated
+    in one stream — span splitting + per-span causal masks must match
+    per-sequence kernel prefills."""
+    f, hip, cpu = _engines(preset="small", n_ctx=256, max_batch=2)
+    hp = f.hparams
+    torch.manual_seed(12)
+    n0, n1 = 96, 80
+    x = torch.randn(n0 + n1, hp.n_embd) * 0.5
+    pos = torch.cat([torch.arange(n0), torch.arange(n1)]).to(torch.int32)
+    seq = torch.cat([torch.zeros(n0), torch.ones(n1)]).to(torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label="mixed-stream rocblas prefill")
