@@ -246,6 +246,11 @@ class HIPSliceEngine:
         # the same device tensors, so clones share HBM)
         self._layers_cache = None
         self._extra_cache = None
+        # detiled-f16 weights for the rocBLAS prefill path, lazily built
+        # and SHARED with clones; gated by _prefill_cache_budget bytes
+        # (detile-per-call otherwise — correct, slower)
+        self._prefill_cache: dict = {}
+        self._prefill_cache_budget = 16e9
 
     @classmethod
     def from_ggml(cls, f: ggml.GGMLFile, n_ctx: int = 2048,
@@ -383,6 +388,7 @@ class HIPSliceEngine:
         for li, (an, fn, mats) in enumerate(self._layers_cache):
             twin._eng.set_layer(li, an, fn, mats)
         twin._layers_cache = self._layers_cache  # enables further clones
+        twin._prefill_cache = self._prefill_cache  # shared detiled f16
         if self._extra_cache is not None:
             twin._extra_cache = self._extra_cache
             twin._eng.set_extra(*self._extra_cache)
@@ -543,9 +549,17 @@ class HIPSliceEngine:
             o[..., 1::2] = t0 * sin[:, None, :] + t1 * cos[:, None, :]
             return o.reshape(T, E)
 
+        E2, F2 = hp.n_embd, hp.n_ff
+        cacheable = (len(self._layers_cache) *
+                     (4 * E2 * E2 + 3 * E2 * F2) * 2
+                     <= self._prefill_cache_budget)
         for li, (an, fn, mats) in enumerate(self._layers_cache):
-            ws = [self._detile_half(mats[i], *self._mat_shape(i))
-                  for i in range(7)]
+            ws = self._prefill_cache.get(li)
+            if ws is None:
+                ws = [self._detile_half(mats[i], *self._mat_shape(i))
+                      for i in range(7)]
+                if cacheable:
+                    self._prefill_cache[li] = ws
             a = (rms_norm(x) * an).to(torch.half)
             q = rope((a @ ws[0].t()).float())
             k = rope((a @ ws[1].t()).float())
