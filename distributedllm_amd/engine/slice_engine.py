@@ -427,11 +427,12 @@ class HIPSliceEngine:
         self._eng.set_extra(*self._extra_cache)
         self.has_extra = True
 
-    # prompt streams at least this long take the rocBLAS prefill path
-    # (dequantize weights to f16, library GEMMs) instead of 64-token
-    # tiles through the decode kernels — the decode-GEMM class is
-    # load-path-bound, so large-M library GEMMs win (docs/roadmap.md #2)
-    PREFILL_MIN_T = 128
+    # prompt spans at least this long take the rocBLAS prefill path
+    # (cached detiled-f16 weights, library GEMMs): measured on 3B q4_0,
+    # the library path wins 1.6x at span 512 and 2.5x at 1024 but LOSES
+    # below ~384 (per-call aten overhead) and on many-short-span mixed
+    # admission streams — so the gate is the LONGEST span, not total T
+    PREFILL_MIN_SPAN = 384
 
     def forward(self, x: torch.Tensor, pos: torch.Tensor,
                 seq: torch.Tensor, decode: bool = False) -> torch.Tensor:
@@ -441,8 +442,10 @@ class HIPSliceEngine:
         mt = self._eng.max_tokens
         if T <= mt:
             return self._eng.forward(x, pos, seq, decode=decode)
-        if T >= self.PREFILL_MIN_T and self._prefill_ok():
-            return self._prefill(x, pos, seq)
+        if T >= self.PREFILL_MIN_SPAN and self._prefill_ok():
+            spans = self._spans(pos.tolist(), seq.tolist())
+            if max(j - i for i, j in spans) >= self.PREFILL_MIN_SPAN:
+                return self._prefill(x, pos, seq, spans)
         # token-tile larger inputs (prefill); KV order is preserved because
         # tile i's cache rows are written before tile i+1 attends.
         outs = []
@@ -516,7 +519,7 @@ class HIPSliceEngine:
             i = j
         return out
 
-    def _prefill(self, x, pos, seq):
+    def _prefill(self, x, pos, seq, spans=None):
         """Large-T forward: detile each weight matrix to f16 and run the
         six GEMMs through rocBLAS (torch.matmul — plain library GEMMs,
         where M >= 128 belongs); RoPE/KV/attention keep the engine's
@@ -531,7 +534,8 @@ class HIPSliceEngine:
         kc, vc = self._eng.k_cache, self._eng.v_cache
         pos_l = pos.tolist()
         seq_l = seq.tolist()
-        spans = self._spans(pos_l, seq_l)
+        if spans is None:
+            spans = self._spans(pos_l, seq_l)
         dev = x.device
         inv = ROPE_BASE ** (-2.0 * torch.arange(D // 2, device=dev,
                                                 dtype=torch.float32) / D)

@@ -47,9 +47,11 @@ def quantize_q4_0(x: np.ndarray) -> np.ndarray:
     # signed amax: the element with the largest |value|, keeping its sign
     idx = np.argmax(np.abs(b), axis=-1)
     m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
-    d = (m / -8.0).astype(np.float16)
-    df = d.astype(np.float32)
-    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
+    # llama.cpp semantics: the STORED scale is f16 but the quantization
+    # divides by the unrounded f32 scale (quantize_row_q4_0_reference)
+    d32 = (m / -8.0).astype(np.float32)
+    d = d32.astype(np.float16)
+    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
     q = np.clip(np.rint(b * inv[..., None]) + 8, 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
     packed = (lo | (hi << 4)).astype(np.uint8)
@@ -86,11 +88,11 @@ def quantize_q4_1(x: np.ndarray) -> np.ndarray:
 
     mn = b.min(axis=-1)
     mx = b.max(axis=-1)
-    d = ((mx - mn) / 15.0).astype(np.float16)
+    d32 = ((mx - mn) / 15.0).astype(np.float32)
+    d = d32.astype(np.float16)
     m = mn.astype(np.float16)
-    df = d.astype(np.float32)
-    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
-    q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) * inv[..., None]),
+    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    q = np.clip(np.rint((b - mn[..., None]) * inv[..., None]),
                 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
     packed = (lo | (hi << 4)).astype(np.uint8)
@@ -164,9 +166,9 @@ def quantize_q5_0(x: np.ndarray) -> np.ndarray:
     b = x.reshape(-1, nb, QK4)
     idx = np.argmax(np.abs(b), axis=-1)
     m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
-    d = (m / -16.0).astype(np.float16)
-    df = d.astype(np.float32)
-    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
+    d32 = (m / -16.0).astype(np.float32)
+    d = d32.astype(np.float16)
+    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
     q = np.clip(np.rint(b * inv[..., None]) + 16, 0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)
     out = np.empty(b.shape[:2] + (Q5_0_BLOCK_BYTES,), dtype=np.uint8)
@@ -193,12 +195,12 @@ def quantize_q5_1(x: np.ndarray) -> np.ndarray:
     b = x.reshape(-1, nb, QK4)
     mn = b.min(axis=-1)
     mx = b.max(axis=-1)
-    d = ((mx - mn) / 31.0).astype(np.float16)
+    d32 = ((mx - mn) / 31.0).astype(np.float32)
+    d = d32.astype(np.float16)
     m = mn.astype(np.float16)
-    df = d.astype(np.float32)
-    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
-    q = np.clip(np.rint((b - m.astype(np.float32)[..., None]) *
-                        inv[..., None]), 0, 31).astype(np.uint8)
+    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    q = np.clip(np.rint((b - mn[..., None]) * inv[..., None]),
+                0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)
     out = np.empty(b.shape[:2] + (Q5_1_BLOCK_BYTES,), dtype=np.uint8)
     out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
@@ -225,9 +227,9 @@ def quantize_q8_0(x: np.ndarray) -> np.ndarray:
     nb = _check_shape(x.shape[-1])
     b = x.reshape(-1, nb, QK4)
     amax = np.abs(b).max(axis=-1)
-    d = (amax / 127.0).astype(np.float16)
-    df = d.astype(np.float32)
-    inv = np.divide(1.0, df, out=np.zeros_like(df), where=df != 0.0)
+    d32 = (amax / 127.0).astype(np.float32)
+    d = d32.astype(np.float16)
+    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
     q = np.rint(b * inv[..., None]).astype(np.int8)
     out = np.empty(b.shape[:2] + (Q8_0_BLOCK_BYTES,), dtype=np.uint8)
     out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))

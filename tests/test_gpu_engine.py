@@ -446,11 +446,11 @@ def test_rocblas_prefill_path_matches_cpu():
     weights, library GEMMs, torch attention over the engine's KV) —
     must match the fp32 CPU reference like the kernel path does, and
     the kernel decode path must attend seamlessly over its KV rows."""
-    f, hip, cpu = _engines(preset="small", n_ctx=256, max_batch=2)
+    f, hip, cpu = _engines(preset="small", n_ctx=512, max_batch=2)
     hp = f.hparams
     assert hip._prefill_ok()
     torch.manual_seed(11)
-    T = 160  # >= PREFILL_MIN_T -> prefill branch
+    T = 400  # span >= PREFILL_MIN_SPAN -> rocBLAS prefill branch
     x = torch.randn(T, hp.n_embd) * 0.5
     pos = torch.arange(T, dtype=torch.int32)
     seq = torch.zeros(T, dtype=torch.int32)
@@ -458,6 +458,7 @@ def test_rocblas_prefill_path_matches_cpu():
     y_cpu = cpu.forward(x.clone(), pos, seq)
     _assert_close(y_gpu, y_cpu, label="rocblas prefill")
     # decode the next token through the KERNEL path on the same KV
+    assert len(hip._prefill_cache) > 0  # the branch actually ran
     xd = torch.randn(1, hp.n_embd) * 0.5
     pd = torch.tensor([T], dtype=torch.int32)
     sd = torch.zeros(1, dtype=torch.int32)
@@ -471,10 +472,10 @@ def test_rocblas_prefill_mixed_stream():
     """The batched-admission shape: two sequences' prompts concatenated
     in one stream — span splitting + per-span causal masks must match
     per-sequence kernel prefills."""
-    f, hip, cpu = _engines(preset="small", n_ctx=256, max_batch=2)
+    f, hip, cpu = _engines(preset="small", n_ctx=512, max_batch=2)
     hp = f.hparams
     torch.manual_seed(12)
-    n0, n1 = 96, 80
+    n0, n1 = 400, 80  # long span triggers the path; short span rides it
     x = torch.randn(n0 + n1, hp.n_embd) * 0.5
     pos = torch.cat([torch.arange(n0), torch.arange(n1)]).to(torch.int32)
     seq = torch.cat([torch.zeros(n0), torch.ones(n1)]).to(torch.int32)

@@ -134,8 +134,8 @@ inline void quantize_block_q4_0(const float* x, uint8_t* out) {
             m = x[j];
         }
     }
-    const uint16_t dh = f32_to_f16(m / -8.0f);
-    const float d = f16_to_f32(dh);
+    const float d = m / -8.0f;  // llama.cpp: invert the UNROUNDED scale
+    const uint16_t dh = f32_to_f16(d);
     const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     for (int j = 0; j < 16; ++j) {
@@ -152,10 +152,10 @@ inline void quantize_block_q4_1(const float* x, uint8_t* out) {
         mn = std::min(mn, x[j]);
         mx = std::max(mx, x[j]);
     }
-    const uint16_t dh = f32_to_f16((mx - mn) / 15.0f);
+    const float d = (mx - mn) / 15.0f;
+    const uint16_t dh = f32_to_f16(d);
     const uint16_t mh = f32_to_f16(mn);
-    const float d = f16_to_f32(dh);
-    const float m = f16_to_f32(mh);
+    const float m = mn;
     const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     std::memcpy(out + 2, &mh, 2);
@@ -194,8 +194,8 @@ inline void quantize_block_q5_0(const float* x, uint8_t* out) {
     for (int j = 0; j < kQK; ++j) {
         if (std::fabs(x[j]) > amax) { amax = std::fabs(x[j]); m = x[j]; }
     }
-    const uint16_t dh = f32_to_f16(m / -16.0f);
-    const float d = f16_to_f32(dh);
+    const float d = m / -16.0f;
+    const uint16_t dh = f32_to_f16(d);
     const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     uint32_t qh = 0;
@@ -231,9 +231,10 @@ inline void quantize_block_q5_1(const float* x, uint8_t* out) {
         mn = std::min(mn, x[j]);
         mx = std::max(mx, x[j]);
     }
-    const uint16_t dh = f32_to_f16((mx - mn) / 31.0f);
+    const float d = (mx - mn) / 31.0f;
+    const uint16_t dh = f32_to_f16(d);
     const uint16_t mh = f32_to_f16(mn);
-    const float d = f16_to_f32(dh), m = f16_to_f32(mh);
+    const float m = mn;
     const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     std::memcpy(out + 2, &mh, 2);
@@ -268,8 +269,8 @@ inline void dequantize_block_q5_1(const uint8_t* in, float* x) {
 inline void quantize_block_q8_0(const float* x, uint8_t* out) {
     float amax = 0.0f;
     for (int j = 0; j < kQK; ++j) amax = std::max(amax, std::fabs(x[j]));
-    const uint16_t dh = f32_to_f16(amax / 127.0f);
-    const float d = f16_to_f32(dh);
+    const float d = amax / 127.0f;
+    const uint16_t dh = f32_to_f16(d);
     const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     for (int j = 0; j < kQK; ++j)
