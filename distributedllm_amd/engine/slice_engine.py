@@ -444,7 +444,10 @@ class HIPSliceEngine:
             return self._eng.forward(x, pos, seq, decode=decode)
         if T >= self.PREFILL_MIN_SPAN and self._prefill_ok():
             spans = self._spans(pos.tolist(), seq.tolist())
-            if max(j - i for i, j in spans) >= self.PREFILL_MIN_SPAN:
+            # single-span streams only: that is where the library path
+            # is measured to win (one long prompt); many-span admission
+            # streams pay per-span aten overhead x layers and lose badly
+            if len(spans) == 1:
                 return self._prefill(x, pos, seq, spans)
         # token-tile larger inputs (prefill); KV order is preserved because
         # tile i's cache rows are written before tile i+1 attends.
