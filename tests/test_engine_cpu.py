@@ -102,3 +102,30 @@ def test_byte_quant_repack_values_exact():
              beta[..., None])
         ref = t.to_f32().reshape(w.shape[0], -1, 32)
         assert np.abs(w - ref).max() == 0.0, ggml.TYPE_NAMES[t.gtype]
+
+
+def test_detile_inverts_repack():
+    """The prefill path's detile must exactly invert repack_mfma: for
+    every quant format, repack (to CPU tensors) then _detile_half
+    reproduces the codec's dequantized weights (f16-rounded)."""
+    import numpy as np
+    import torch
+    from distributedllm_amd.engine.slice_engine import (
+        HIPSliceEngine, repack_mfma)
+    from distributedllm_amd.formats import ggml, synthetic
+    for ft in (ggml.FTYPE_MOSTLY_Q4_0, ggml.FTYPE_MOSTLY_Q4_1,
+               ggml.FTYPE_MOSTLY_Q8_0, ggml.FTYPE_MOSTLY_Q5_0,
+               ggml.FTYPE_MOSTLY_Q5_1, ggml.FTYPE_MOSTLY_F16):
+        f = synthetic.build_model("tiny", seed=1, ftype=ft)
+        t = next(x for x in f.tensors
+                 if x.name.endswith("feed_forward.w1.weight"))
+        rows, cols = t.shape_rows_cols
+        mat = repack_mfma(t, "cpu")
+        got = HIPSliceEngine._detile_half(None, mat, rows, cols).float()
+        want = torch.from_numpy(t.to_f32())
+        # detile emits f16 (one rounding beyond the codec's own f16
+        # arithmetic); q4_1/q5_1 fold beta=m (+16d) with one more f16
+        # rounding at repack
+        tol = 4e-3 * want.abs().max().item() + 1e-6
+        err = (got - want).abs().max().item()
+        assert err <= tol, (ggml.TYPE_NAMES[ggml._FTYPE_TO_GGML[ft]], err)
