@@ -31,6 +31,13 @@ def main():
                     help="concurrent stream lanes (weight-sharing "
                          "clones); total slots = slots x lanes")
     ap.add_argument("--ctx", type=int, default=2048)
+    ap.add_argument("--prefill-chunk", type=int, default=0,
+                    help="chunked-prefill bound (0 = unbounded)")
+    ap.add_argument("--long-requests", type=int, default=0,
+                    help="long-prompt requests arriving mid-run (QoS "
+                         "experiment: how much do they stall in-flight "
+                         "decodes?)")
+    ap.add_argument("--long-prompt", type=int, default=512)
     args = ap.parse_args()
 
     hp = PRESETS[args.model].hparams(ggml.FTYPE_MOSTLY_Q4_0)
@@ -41,7 +48,8 @@ def main():
         lanes = [eng] + [eng.clone_shared()
                          for _ in range(args.lanes - 1)]
     g = torch.Generator().manual_seed(1)
-    bat = ContinuousBatcher(eng, engines=lanes)
+    bat = ContinuousBatcher(eng, engines=lanes,
+                            prefill_chunk=args.prefill_chunk or None)
     reqs = [bat.submit(torch.randint(3, hp.n_vocab, (args.prompt_len,),
                                      generator=g).tolist(),
                        args.num_tokens)
@@ -50,7 +58,10 @@ def main():
     t0 = time.perf_counter()
     steps = 0
     done_at = {}
+    long_at = max(1, args.num_tokens // 4)
+    stalls = []  # per-step wall time after the long arrivals
     while bat.pending:
+        ts = time.perf_counter()
         fin = bat.step()
         if fin:
             torch.cuda.synchronize()
@@ -58,6 +69,14 @@ def main():
             for r in fin:
                 done_at[r.rid] = now
         steps += 1
+        if args.long_requests and steps >= long_at:
+            torch.cuda.synchronize()
+            stalls.append(time.perf_counter() - ts)
+        if args.long_requests and steps == long_at:
+            for _ in range(args.long_requests):
+                reqs.append(bat.submit(
+                    torch.randint(3, hp.n_vocab, (args.long_prompt,),
+                                  generator=g).tolist(), args.num_tokens))
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     total = sum(len(r.out) for r in reqs)
@@ -69,6 +88,12 @@ def main():
           f"{total} tokens in {dt:.2f}s = {total/dt:.0f} tok/s, "
           f"{steps} decode steps; completion p50={p(0.5):.2f}s "
           f"p95={p(0.95):.2f}s max={lat[-1]:.2f}s")
+    if stalls:
+        ss = sorted(stalls)
+        print(f"per-step wall after long arrivals (stall = decode "
+              f"starvation): p50={ss[len(ss)//2]*1e3:.0f}ms "
+              f"max={ss[-1]*1e3:.0f}ms over {len(ss)} steps "
+              f"(chunk={args.prefill_chunk or 'unbounded'})")
 
 
 if __name__ == "__main__":
