@@ -129,3 +129,20 @@ def test_detile_inverts_repack():
         tol = 4e-3 * want.abs().max().item() + 1e-6
         err = (got - want).abs().max().item()
         assert err <= tol, (ggml.TYPE_NAMES[ggml._FTYPE_TO_GGML[ft]], err)
+
+
+def test_prefill_span_splitting():
+    """_spans finds contiguous same-sequence ascending-position runs —
+    the geometry the prefill path's causal masks depend on."""
+    from distributedllm_amd.engine.slice_engine import HIPSliceEngine
+    spans = HIPSliceEngine._spans
+    assert spans(None, [0, 1, 2], [0, 0, 0]) == [(0, 3)]
+    # sequence change splits
+    assert spans(None, [0, 1, 0, 1], [0, 0, 1, 1]) == [(0, 2), (2, 4)]
+    # position gap splits (restart of the same sequence)
+    assert spans(None, [0, 1, 5, 6], [0, 0, 0, 0]) == [(0, 2), (2, 4)]
+    # descending positions split to singletons
+    assert spans(None, [3, 2, 1], [0, 0, 0]) == [(0, 1), (1, 2), (2, 3)]
+    # the pipeline prime token-major layout: seq alternates every token
+    assert spans(None, [0, 0, 1, 1], [0, 1, 0, 1]) == \
+        [(0, 1), (1, 2), (2, 3), (3, 4)]
