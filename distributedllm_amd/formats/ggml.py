@@ -249,6 +249,9 @@ class GGMLFile:
 
         def u32() -> int:
             nonlocal off
+            if off + 4 > len(data):
+                raise ValueError(
+                    f"truncated GGJT file {path!r} (at byte {off})")
             (v,) = struct.unpack_from("<I", data, off)
             off += 4
             return v
@@ -314,6 +317,9 @@ def sniff_extended(path: str) -> bool:
     """
     with open(path, "rb") as f:
         head = f.read(4 * 10)
+    if len(head) < 4 * 10:
+        raise ValueError(f"truncated GGJT file {path!r} "
+                         f"({len(head)} bytes)")
     vals = struct.unpack_from("<10I", head, 0)
     if vals[0] != GGJT_MAGIC:
         raise ValueError("not a GGJT file")

@@ -151,3 +151,20 @@ class TestSlicer:
             slicer.make_slice(f, 2, 1)
         with pytest.raises(ValueError):
             slicer.make_slice(f, 0, 99)
+
+
+def test_malformed_files_raise_clean_errors(tmp_path):
+    """Truncated/garbage files raise ValueError with a clear message,
+    never raw struct errors (node load_slice maps these to typed
+    protocol errors)."""
+    import pytest
+    cases = {"empty.bin": b"",
+             "garbage.bin": b"nonsense",
+             "trunc.bin": b"tjgg\x03\x00\x00\x00\x01"}
+    for name, content in cases.items():
+        p = tmp_path / name
+        p.write_bytes(content)
+        with pytest.raises(ValueError):
+            ggml.GGMLFile.load(str(p), extended=False)
+        with pytest.raises(ValueError):
+            ggml.sniff_extended(str(p))
