@@ -30,4 +30,15 @@ def build_parser() -> argparse.ArgumentParser:
 
 def execute_command(argv: Optional[List[str]] = None) -> int:
     args = build_parser().parse_args(argv)
-    return args._cmd(args)
+    try:
+        return args._cmd(args)
+    except KeyboardInterrupt:
+        return 130
+    except (ConnectionError, OSError, ValueError, KeyError) as e:
+        # operational failures (node down, bad config, malformed file)
+        # get one clear line, not a traceback (reference CLI behavior:
+        # cli_api/__init__.py prints argparse errors but lets everything
+        # else crash)
+        import sys as _sys
+        print(f"error: {type(e).__name__}: {e}", file=_sys.stderr)
+        return 1

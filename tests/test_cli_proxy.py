@@ -299,3 +299,17 @@ def test_cli_provision_q8_0_generate(node, tmp_path, capsys):
                             "--prompt", "hi", "--num-tokens", "2",
                             "--greedy", "--root", str(root)]) == 0
     assert "hi" in capsys.readouterr().out
+
+
+def test_cli_clean_error_on_unreachable_node(tmp_path, capsys):
+    """Operational failures (node down) print one error line, exit 1 —
+    no traceback."""
+    cfg = {"model_id": "x", "location": "synthetic:tiny",
+           "nodes_map": {"127.0.0.1:1": [0, PRESETS["tiny"].n_layer - 1]},
+           "quantization": "f16", "metadata": {"name": "x"}}
+    p = tmp_path / "cfg.json"
+    p.write_text(json.dumps(cfg))
+    rc = execute_command(["provision", str(p), "--root", str(tmp_path)])
+    assert rc == 1
+    err = capsys.readouterr().err
+    assert "error:" in err and "Traceback" not in err
