@@ -28,13 +28,20 @@ def build_parser() -> argparse.ArgumentParser:
     return parser
 
 
+def _operational_errors():
+    from ..cluster.client import OperationFailedError  # noqa: F401
+    from ..cluster.uploads import UploadError  # noqa: F401
+    return (ConnectionError, OSError, ValueError, KeyError,
+            OperationFailedError, UploadError)
+
+
 def execute_command(argv: Optional[List[str]] = None) -> int:
     args = build_parser().parse_args(argv)
     try:
         return args._cmd(args)
     except KeyboardInterrupt:
         return 130
-    except (ConnectionError, OSError, ValueError, KeyError) as e:
+    except _operational_errors() as e:
         # operational failures (node down, bad config, malformed file)
         # get one clear line, not a traceback (reference CLI behavior:
         # cli_api/__init__.py prints argparse errors but lets everything
