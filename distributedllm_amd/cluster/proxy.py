@@ -86,6 +86,11 @@ class ProxyServer:
                 description="expected request_greeting"))
             sock.close()
             return
+        # a stalled-but-connected node must not hold the roundtrip lock
+        # forever: bound every node read/write so _roundtrip converts a
+        # stall into node_lost (socket.timeout is an OSError subclass,
+        # so the except path below already covers it)
+        sock.settimeout(120.0)
         with self._node_lock:
             old = self._node_sock
             self._node_sock = sock

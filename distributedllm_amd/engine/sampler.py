@@ -52,6 +52,11 @@ class Sampler:
         if self.top_k and self.top_k < size:
             kth = np.partition(probs, -self.top_k)[-self.top_k]
             probs = np.where(probs >= kth, probs, 0.0)
+            # renormalize before the nucleus cut: sequential-filter
+            # semantics (llama.cpp/HF renormalize between filters) —
+            # without this, top_p over the <1 surviving mass keeps a
+            # different set, and top_p > surviving mass is a no-op
+            probs = probs / probs.sum()
         if self.top_p < 1.0:
             order = np.argsort(-probs)
             csum = np.cumsum(probs[order])

@@ -229,8 +229,16 @@ def quantize_q8_0(x: np.ndarray) -> np.ndarray:
     amax = np.abs(b).max(axis=-1)
     d32 = (amax / 127.0).astype(np.float32)
     d = d32.astype(np.float16)
-    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
-    q = np.rint(b * inv[..., None]).astype(np.int8)
+    # subnormal d32 would overflow 1/d32 to inf and make the int8 cast
+    # undefined — treat sub-tiny-scale blocks as all-zero (their values
+    # are below f16 resolution anyway) and clamp the quantized range
+    tiny = np.finfo(np.float32).tiny
+    with np.errstate(over="ignore", invalid="ignore"):
+        inv = np.divide(1.0, d32, out=np.zeros_like(d32),
+                        where=d32 >= tiny)
+        q = np.clip(np.rint(b * inv[..., None]), -127.0, 127.0)
+    q = np.nan_to_num(q, nan=0.0, posinf=127.0, neginf=-127.0)
+    q = q.astype(np.int8)
     out = np.empty(b.shape[:2] + (Q8_0_BLOCK_BYTES,), dtype=np.uint8)
     out[..., 0:2] = d[..., None].view(np.uint8).reshape(d.shape + (2,))
     out[..., 2:] = q.view(np.uint8)

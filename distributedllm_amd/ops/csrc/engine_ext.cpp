@@ -284,6 +284,21 @@ class SliceEngine {
                     "forward handles at most ", kMaxTokens,
                     " tokens per call; tile larger batches host-side");
         TORCH_CHECK(pos.numel() == T && seq.numel() == T, "pos/seq size");
+        // Opt-in backstop against out-of-bounds KV writes (an oversized
+        // request that slipped past host-side validation). Costs a host
+        // sync per forward and is illegal under graph capture, so it is
+        // env-gated rather than always-on.
+        static const bool bounds_check = [] {
+            const char* v = std::getenv("DLLM_BOUNDS_CHECK");
+            return v && v[0] == '1';
+        }();
+        if (bounds_check) {
+            const int64_t pmax = pos.max().item<int64_t>();
+            const int64_t smax = seq.max().item<int64_t>();
+            TORCH_CHECK(pmax < ctx_, "position ", pmax, " >= n_ctx ", ctx_);
+            TORCH_CHECK(smax < B_, "sequence id ", smax, " >= max_batch ",
+                        B_);
+        }
         for (int li = 0; li < L_; ++li)
             TORCH_CHECK(loaded_[li], "layer ", li, " not loaded");
         hipStream_t s = c10::hip::getCurrentHIPStream().stream();

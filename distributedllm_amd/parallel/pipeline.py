@@ -317,6 +317,11 @@ class DecodePipeline:
         E = self.engine.hp.n_embd
         tile = max(1, 64 // mbs)  # prompt positions per pipeline hop
         for m in range(self.cfg.n_mb):
+            # lane engines hold separate KV caches: micro-batch m's
+            # prompt KV must land on ITS lane's engine (and lane-local
+            # seq ids, self.seq[m]) or decode reads empty rows
+            # (ADVICE r1 — prime() used self.engine for every mb)
+            eng = self._eng(m)
             for p0 in range(0, Tp - 1, tile):
                 p1 = min(Tp - 1, p0 + tile)
                 n = p1 - p0
@@ -326,12 +331,12 @@ class DecodePipeline:
                 seq = self.seq[m].repeat(n)
                 if self.is_first:
                     toks = prompt[p0:p1].repeat_interleave(mbs)
-                    x = self.engine.embed(toks)
+                    x = eng.embed(toks)
                 else:
                     x = torch.empty(n * mbs, E, dtype=torch.float32,
                                     device=dev)
                     dist.recv(x, src=self.rank - 1)
-                y = self.engine.forward(x, pos, seq)
+                y = eng.forward(x, pos, seq)
                 if not self.is_last:
                     dist.send(y, dst=self.rank + 1)
             self.pos[m].fill_(Tp - 1)

@@ -67,7 +67,17 @@ class BatcherWorker:
                 if not self.batcher.pending:
                     self.cond.wait(timeout=0.05)
                     continue
-                self.batcher.step()
+                try:
+                    self.batcher.step()
+                except Exception:  # one bad step must not kill serving
+                    import traceback
+                    traceback.print_exc()
+                    # fail every in-flight request rather than hang its
+                    # waiter; slots are reclaimed through cancel()
+                    for r in (list(self.batcher.active.values())
+                              + list(self.batcher.prefilling.values())
+                              + list(self.batcher.queue)):
+                        self.batcher.cancel(r)
                 self.cond.notify_all()
 
     def submit_and_wait(self, prompt_ids, max_new,
@@ -80,6 +90,9 @@ class BatcherWorker:
             while not req.done:
                 left = deadline - time.monotonic()
                 if left <= 0:
+                    # reclaim the KV slot — an abandoned request must
+                    # not keep generating
+                    self.batcher.cancel(req)
                     raise TimeoutError("generation timed out")
                 self.cond.wait(timeout=min(left, 1.0))
         return req
@@ -124,6 +137,8 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
         ids, sampler = _submit(r)
         try:
             req = worker.submit_and_wait(ids, r.num_tokens, sampler)
+        except ValueError as e:        # oversized prompt+num_tokens
+            raise HTTPException(422, str(e))
         except TimeoutError as e:
             raise HTTPException(504, str(e))
         stats["requests"] += 1
@@ -142,27 +157,39 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
 
         ids, sampler = _submit(r)
         with worker.cond:
-            req = batcher.submit(ids, r.num_tokens, sampler=sampler)
+            try:
+                req = batcher.submit(ids, r.num_tokens, sampler=sampler)
+            except ValueError as e:
+                raise HTTPException(422, str(e))
             worker.cond.notify_all()
 
         def events():
             sent = 0
-            while True:
-                with worker.cond:
-                    while len(req.out) == sent and not req.done:
-                        worker.cond.wait(timeout=1.0)
-                    chunk = list(req.out[sent:])
-                    sent = len(req.out)
-                    done = req.done
-                for tid in chunk:
-                    yield ("data: " + _json.dumps(
-                        {"token": tid,
-                         "piece": tokenizer.decode_token(tid)}) + "\n\n")
-                if done:
-                    yield ("event: done\ndata: " + _json.dumps(
-                        {"text": tokenizer.decode(req.out),
-                         "tokens": list(req.out)}) + "\n\n")
-                    return
+            try:
+                while True:
+                    with worker.cond:
+                        while len(req.out) == sent and not req.done:
+                            worker.cond.wait(timeout=1.0)
+                        chunk = list(req.out[sent:])
+                        sent = len(req.out)
+                        done = req.done
+                    for tid in chunk:
+                        yield ("data: " + _json.dumps(
+                            {"token": tid,
+                             "piece": tokenizer.decode_token(tid)})
+                            + "\n\n")
+                    if done:
+                        yield ("event: done\ndata: " + _json.dumps(
+                            {"text": tokenizer.decode(req.out),
+                             "tokens": list(req.out)}) + "\n\n")
+                        return
+            finally:
+                # client went away mid-stream (GeneratorExit) or any
+                # failure: free the KV slot instead of decoding to
+                # max_new for nobody
+                if not req.done:
+                    with worker.cond:
+                        batcher.cancel(req)
 
         return StreamingResponse(events(), media_type="text/event-stream")
 

@@ -96,7 +96,18 @@ class ContinuousBatcher:
     def submit(self, prompt_ids: Sequence[int], max_new: int,
                sampler: Optional[Sampler] = None,
                eos_id: Optional[int] = None) -> Request:
-        assert len(prompt_ids) >= 1 and max_new >= 1
+        """Raises ValueError (before any slot is taken) on an invalid or
+        oversized request — the caller can reject it (HTTP 422) without
+        the decode loop ever seeing it."""
+        if len(prompt_ids) < 1:
+            raise ValueError("prompt must contain at least one token")
+        if max_new < 1:
+            raise ValueError("max_new must be >= 1")
+        n_ctx = getattr(self.engine, "n_ctx", None)
+        if n_ctx is not None and len(prompt_ids) + max_new > n_ctx:
+            raise ValueError(
+                f"prompt ({len(prompt_ids)}) + max_new ({max_new}) "
+                f"exceeds n_ctx={n_ctx}")
         r = Request(rid=self._next_rid, prompt=list(map(int, prompt_ids)),
                     max_new=max_new, sampler=sampler,
                     eos_id=self.eos_id if eos_id is None else eos_id)
@@ -112,9 +123,13 @@ class ContinuousBatcher:
         # prefills below (possibly across several steps when chunked)
         while self.free and self.queue:
             r = self.queue.popleft()
+            # submit() already validated against n_ctx; this is a cheap
+            # backstop against an engine swap shrinking n_ctx after
+            # submission — drop the request rather than corrupt the KV
+            if len(r.prompt) + r.max_new > n_ctx:
+                r.done = True
+                continue
             r.slot = self.free.pop()
-            assert len(r.prompt) + r.max_new <= n_ctx, \
-                f"request {r.rid}: prompt+max_new exceeds n_ctx={n_ctx}"
             r._pf_pos = 0
             if len(r.prompt) > 1:
                 self.prefilling[r.slot] = r

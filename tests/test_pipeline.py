@@ -275,3 +275,35 @@ def test_prime_matches_reference_generation_semantics():
     pipe = DecodePipeline(eng2, cfg, rank=0, world=1)
     got = pipeline_generate(pipe, prompt, max_steps=steps).tolist()
     assert got[0] == want and got[1] == want
+
+
+def test_prime_with_lane_engines_matches_shared_engine():
+    """prime() must write each micro-batch's prompt KV into ITS lane's
+    engine (lane-local seq ids) — with 2 lane clones, prompt-conditioned
+    generation must equal the single shared-engine pipeline (ADVICE r1:
+    prime() previously sent every micro-batch through lane 0)."""
+    from distributedllm_amd.parallel.pipeline import pipeline_generate
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    prompt, steps, n_mb, n_lanes = [5, 9, 3], 4, 4, 2
+
+    def run(lanes):
+        if lanes:
+            per_lane = MBS * ((n_mb + n_lanes - 1) // n_lanes)
+            engines = []
+            for _ in range(n_lanes):
+                e = TorchSliceEngine.from_ggml(f, n_ctx=32,
+                                               max_batch=per_lane)
+                e.attach_extra(ex)
+                engines.append(e)
+            eng = engines[0]
+        else:
+            engines = None
+            eng = TorchSliceEngine.from_ggml(f, n_ctx=32,
+                                             max_batch=MBS * n_mb)
+            eng.attach_extra(ex)
+        cfg = PipelineConfig(mbs=MBS, n_mb=n_mb, device="cpu")
+        pipe = DecodePipeline(eng, cfg, rank=0, world=1, engines=engines)
+        return pipeline_generate(pipe, prompt, max_steps=steps).tolist()
+
+    assert run(lanes=True) == run(lanes=False)

@@ -93,10 +93,19 @@ def test_eos_stops_early():
 
 
 def test_context_overflow_rejected():
+    # rejected at submit() time, BEFORE any KV slot is taken, so one bad
+    # request can never kill the decode loop
     bat = ContinuousBatcher(_engine(1))
-    bat.submit([1] * 30, 10)  # 40 > n_ctx=32
-    with pytest.raises(AssertionError):
-        bat.run_all(max_steps=5)
+    with pytest.raises(ValueError):
+        bat.submit([1] * 30, 10)  # 40 > n_ctx=32
+    with pytest.raises(ValueError):
+        bat.submit([], 5)
+    with pytest.raises(ValueError):
+        bat.submit([1, 2], 0)
+    # a valid request right at the boundary is accepted and completes
+    bat.submit([1] * 16, 16)  # 32 == n_ctx
+    done = bat.run_all(max_steps=64)
+    assert len(done) == 1 and len(done[0].out) == 16
 
 
 def test_batch_generate_cli(tmp_path):

@@ -144,3 +144,23 @@ def test_sampler_top_k_top_p():
     b = Sampler(temperature=0.8, repeat_penalty=1.1, seed=7,
                 top_k=0, top_p=1.0)
     assert [a(logits) for _ in range(5)] == [b(logits) for _ in range(5)]
+
+
+def test_sampler_top_p_renormalizes_after_top_k():
+    """Sequential-filter semantics (llama.cpp/HF): after the top_k mask
+    the surviving probs are renormalized BEFORE the nucleus cut, so
+    top_p operates on conditional mass. With top_k=2 keeping ~86% of
+    the raw mass and top_p=0.9, the nucleus over the renormalized pair
+    keeps only the dominant token — pre-fix, top_p > surviving mass was
+    silently a no-op and token 1 kept appearing."""
+    import numpy as np
+    from distributedllm_amd.engine.sampler import Sampler
+    logits = np.array([3.0, 1.0, 0.5, 0.0, -1.0])
+    # raw softmax: p0≈0.78, p1≈0.11 → after top_k=2 renorm: 0.879/0.121
+    s = Sampler(temperature=1.0, repeat_penalty=1.0, seed=0,
+                top_k=2, top_p=0.85)
+    draws = set()
+    for _ in range(100):
+        s.previous_ids = []
+        draws.add(s(logits))
+    assert draws == {0}
