@@ -56,6 +56,12 @@ def main() -> int:
                          "Auto-capped by HBM fit; 5 measured best on "
                          "3B: +47%% over one stream, 6 regresses")
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--backend", default="auto",
+                    choices=["auto", "nccl", "gloo"],
+                    help="auto = nccl(RCCL) on GPU, gloo on CPU. "
+                         "gloo + GPU = host-staged hops: lets world>1 "
+                         "run on ONE physical GPU (RCCL refuses "
+                         "duplicate devices) for pipeline shakeout")
     ap.add_argument("--prefill-depth", type=int, default=512,
                     help="untimed decode steps that fill the KV before "
                          "timing, so the headline is a steady-state "
@@ -70,7 +76,8 @@ def main() -> int:
 
     import torch.distributed as dist
     device = "cuda" if torch.cuda.is_available() else "cpu"
-    backend = "nccl" if device == "cuda" else "gloo"
+    backend = ("nccl" if device == "cuda" else "gloo") \
+        if args.backend == "auto" else args.backend
     if world > 1:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         dist.init_process_group(backend=backend, rank=rank, world_size=world)
@@ -195,6 +202,7 @@ def main() -> int:
                 "lanes": n_lanes,
                 "mbs": args.mbs,
                 "device": device,
+                "backend": backend if world > 1 else None,
             },
         }
         print(json.dumps(result), flush=True)

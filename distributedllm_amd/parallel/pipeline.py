@@ -46,6 +46,18 @@ class PipelineConfig:
         return self.mbs * self.n_mb
 
 
+class _StagedWork:
+    """irecv handle for a host-staged hop: wait() completes the gloo
+    receive then uploads into the device-side target."""
+
+    def __init__(self, req, host: torch.Tensor, dev: torch.Tensor):
+        self.req, self.host, self.dev = req, host, dev
+
+    def wait(self) -> None:
+        self.req.wait()
+        self.dev.copy_(self.host)
+
+
 class DecodePipeline:
     """Synchronized batched decode across pipeline stages.
 
@@ -111,6 +123,55 @@ class DecodePipeline:
         self._graph_out = None   # per-mb tensors the graph writes
         self._streams = None     # per-mb streams (single-GPU multi-mb)
         self.temperature = 0.0   # 0 = greedy argmax (graph-capturable)
+        # Host-staged hops: gloo cannot move CUDA tensors, but a gloo
+        # pipeline with CUDA compute is how the multi-rank path is
+        # shaken out on a 1-GPU box (RCCL refuses two ranks on one
+        # device — "Duplicate GPU detected"). Device tensors bounce
+        # through pinned host buffers around each gloo send/recv.
+        self._staged = (dev == "cuda" and world > 1 and
+                        dist.is_initialized() and
+                        dist.get_backend() == "gloo")
+        self._hop_bufs = {}
+
+    # -------------------------------------------------- transport (hops)
+
+    def _hopbuf(self, key, like: torch.Tensor) -> torch.Tensor:
+        buf = self._hop_bufs.get(key)
+        if (buf is None or buf.shape != like.shape or
+                buf.dtype != like.dtype):
+            buf = torch.empty(like.shape, dtype=like.dtype, device="cpu",
+                              pin_memory=True)
+            self._hop_bufs[key] = buf
+        return buf
+
+    def _hop_send(self, t: torch.Tensor, dst: int, key) -> None:
+        if not self._staged:
+            dist.send(t, dst=dst)
+            return
+        h = self._hopbuf(("s",) + key, t)
+        h.copy_(t)  # D2H, synchronous w.r.t. host
+        dist.send(h, dst=dst)
+
+    def _hop_recv(self, t: torch.Tensor, src: int, key) -> None:
+        if not self._staged:
+            dist.recv(t, src=src)
+            return
+        h = self._hopbuf(("r",) + key, t)
+        dist.recv(h, src=src)
+        t.copy_(h)
+
+    def _hop_isend(self, t: torch.Tensor, dst: int, key):
+        if not self._staged:
+            return dist.isend(t, dst=dst)
+        h = self._hopbuf(("s",) + key, t)
+        h.copy_(t)
+        return dist.isend(h, dst=dst)
+
+    def _hop_irecv(self, t: torch.Tensor, src: int, key):
+        if not self._staged:
+            return dist.irecv(t, src=src)
+        h = self._hopbuf(("r",) + key, t)
+        return _StagedWork(dist.irecv(h, src=src), h, t)
 
     # ------------------------------------------------------ hipGraph mode
 
@@ -182,17 +243,17 @@ class DecodePipeline:
         """Push micro-batch m one token forward through this stage
         (blocking comm; the overlapped path lives in run_steps)."""
         if not self.is_first:
-            dist.recv(self.x_recv[m], src=self.rank - 1)
+            self._hop_recv(self.x_recv[m], self.rank - 1, ("x", m))
         out = self._compute(m)
         if not self.is_last:
-            dist.send(out, dst=self.rank + 1)
+            self._hop_send(out, self.rank + 1, ("x", m))
         elif self.world > 1:
-            dist.send(out, dst=0)
+            self._hop_send(out, 0, ("t", m))
         if self.is_first and self.world > 1:
             # token ids for this micro-batch's next step come from the
             # last stage; ordered FIFO per rank pair, so recv here pairs
             # with the send above.
-            dist.recv(self.tok[m], src=self.world - 1)
+            self._hop_recv(self.tok[m], self.world - 1, ("t", m))
 
     def _compute(self, m: int):
         if self._graphs is not None:
@@ -206,9 +267,9 @@ class DecodePipeline:
         multi-stage pipeline receives the sampled token ids from the last
         stage."""
         if not self.is_first:
-            return dist.irecv(self.x_recv[m], src=self.rank - 1)
+            return self._hop_irecv(self.x_recv[m], self.rank - 1, ("x", m))
         if self.world > 1:
-            return dist.irecv(self.tok[m], src=self.world - 1)
+            return self._hop_irecv(self.tok[m], self.world - 1, ("t", m))
         return None
 
     def run_steps(self, steps: int) -> None:
@@ -274,7 +335,8 @@ class DecodePipeline:
                         sends[m].wait()  # out buffer is being re-written
                     out = self._compute(m)
                     dst = 0 if self.is_last else self.rank + 1
-                    sends[m] = dist.isend(out, dst=dst)
+                    sends[m] = self._hop_isend(
+                        out, dst, ("t", m) if self.is_last else ("x", m))
                     # mid/last stages receive exactly `steps` activation
                     # blocks — do not post one past the end (it would
                     # never be matched); rank 0 must still drain the
@@ -335,10 +397,10 @@ class DecodePipeline:
                 else:
                     x = torch.empty(n * mbs, E, dtype=torch.float32,
                                     device=dev)
-                    dist.recv(x, src=self.rank - 1)
+                    self._hop_recv(x, self.rank - 1, ("pf", m))
                 y = eng.forward(x, pos, seq)
                 if not self.is_last:
-                    dist.send(y, dst=self.rank + 1)
+                    self._hop_send(y, self.rank + 1, ("pf", m))
             self.pos[m].fill_(Tp - 1)
             self.tok[m].fill_(int(prompt[-1].item()))
 
