@@ -223,11 +223,51 @@ def _upload_mat(t: ggml.GGMLTensor, device: str):
     return torch.from_numpy(a.copy()).to(device), torch.empty(0), t.gtype
 
 
+def detile_mfma(mat, rows: int, cols: int) -> torch.Tensor:
+    """Reference inverse of repack_mfma: tiled weights -> plain f16
+    [rows, cols]. Test/debug utility — documents the tile layouts the
+    kernels stream (the production prefill path reads tiles directly)."""
+    data, sc, wt = mat
+    if wt == ggml.GGML_TYPE_F16:
+        t = data[:rows * cols].view(torch.half)
+        return (t.view(rows // 16, cols // 8, 16, 8)
+                .permute(0, 2, 1, 3).reshape(rows, cols))
+    R, nb = rows // 16, cols // 32
+    nbp = (nb + 3) & ~3
+    ab = sc[:R * nbp * 32].view(R, nbp // 4, 16, 4, 2).float()
+    alpha = ab[..., 0][:, :, None, :, :, None]  # [R,g4,1,i,kb,1]
+    beta = ab[..., 1][:, :, None, :, :, None]
+    if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
+        qw = data[:R * nbp * 64].view(R, nbp // 4, 4, 16, 4)
+        shifts = torch.tensor(
+            [(j % 2) * 16 + (j // 2) * 4 for j in range(8)],
+            device=data.device, dtype=torch.int32)
+        n = ((qw.unsqueeze(-1) >> shifts) & 0xF).float()
+        if wt == ggml.GGML_TYPE_Q4_0:
+            w = alpha * (n - 8.0)
+        else:
+            w = alpha * n + beta
+        # [R,g4,ws,i,kb,8] -> [R,i,g4,kb,ws,8] -> [rows, nbp*32]
+        w = w.permute(0, 3, 1, 4, 2, 5).reshape(rows, nbp * 32)
+        return w[:, :cols].to(torch.half)
+    # W_Q8B byte stream: [R][g4][ws][i][kb][2 u32], u32 bytes
+    # [w0,w2,w1,w3] so weight jj sits at byte [0,2,1,3][jj]
+    qw = data[:R * nbp * 128].view(R, nbp // 4, 4, 16, 4, 2)
+    bshift = torch.tensor([0, 16, 8, 24], device=data.device,
+                          dtype=torch.int32)
+    u = ((qw.unsqueeze(-1) >> bshift) & 0xFF).float()
+    w = alpha.unsqueeze(-1) * (u - 128.0) + beta.unsqueeze(-1)
+    # [R,g4,ws,i,kb,2,4] -> [R,i,g4,kb,ws,2,4] -> [rows, nbp*32]
+    w = w.permute(0, 3, 1, 4, 2, 5, 6).reshape(rows, nbp * 32)
+    return w[:, :cols].to(torch.half)
+
+
 class HIPSliceEngine:
     """Production engine: CDNA4 kernels, weights resident in HBM3E."""
 
     def __init__(self, hp: ggml.Hparams, n_layers: int, first_layer: int,
-                 n_ctx: int = 2048, max_batch: int = 16):
+                 n_ctx: int = 2048, max_batch: int = 16,
+                 max_prefill: Optional[int] = None):
         from .. import ops
         core = ops.core()
         self.hp = hp
@@ -235,10 +275,14 @@ class HIPSliceEngine:
         self.n_layers = n_layers
         self.n_ctx = n_ctx
         self.max_batch = max_batch
+        # prefill token cap per forward call (side channels scale with it;
+        # larger prompts tile host-side in max_prefill chunks)
+        if max_prefill is None:
+            max_prefill = max(64, min(n_ctx, 2048))
         self._eng = core.SliceEngine(
             n_embd=hp.n_embd, n_head=hp.n_head, n_layers=n_layers,
             n_ff=hp.n_ff, n_ctx=n_ctx, max_batch=max_batch,
-            eps=RMS_EPS, rope_base=ROPE_BASE)
+            eps=RMS_EPS, rope_base=ROPE_BASE, max_prefill=max_prefill)
         self.device = "cuda"
         self.has_extra = False
         # weight-tensor references for clone_shared (filled by .random()
@@ -246,11 +290,6 @@ class HIPSliceEngine:
         # the same device tensors, so clones share HBM)
         self._layers_cache = None
         self._extra_cache = None
-        # detiled-f16 weights for the rocBLAS prefill path, lazily built
-        # and SHARED with clones; gated by _prefill_cache_budget bytes
-        # (detile-per-call otherwise — correct, slower)
-        self._prefill_cache: dict = {}
-        self._prefill_cache_budget = 16e9
 
     @classmethod
     def from_ggml(cls, f: ggml.GGMLFile, n_ctx: int = 2048,
@@ -388,7 +427,6 @@ class HIPSliceEngine:
         for li, (an, fn, mats) in enumerate(self._layers_cache):
             twin._eng.set_layer(li, an, fn, mats)
         twin._layers_cache = self._layers_cache  # enables further clones
-        twin._prefill_cache = self._prefill_cache  # shared detiled f16
         if self._extra_cache is not None:
             twin._extra_cache = self._extra_cache
             twin._eng.set_extra(*self._extra_cache)
@@ -427,176 +465,37 @@ class HIPSliceEngine:
         self._eng.set_extra(*self._extra_cache)
         self.has_extra = True
 
-    # prompt spans at least this long take the rocBLAS prefill path
-    # (cached detiled-f16 weights, library GEMMs): measured on 3B q4_0,
-    # the library path wins 1.6x at span 512 and 2.5x at 1024 but LOSES
-    # below ~384 (per-call aten overhead) and on many-short-span mixed
-    # admission streams — so the gate is the LONGEST span, not total T
-    PREFILL_MIN_SPAN = 384
-
     def forward(self, x: torch.Tensor, pos: torch.Tensor,
                 seq: torch.Tensor, decode: bool = False) -> torch.Tensor:
         """decode=True asserts every token is a distinct sequence
-        (batched decode) — enables the qkv-slab + fused-attention path."""
+        (batched decode) — enables the qkv-slab + fused-attention path.
+
+        T > 64 (prefill) runs the native large-M path: hand-written
+        XCD-grouped dequant-GEMM kernels over all T tokens, the layer
+        loop sequenced in C++ (replaced the round-1 rocBLAS-over-
+        detiled-f16 path — no f16 weight copy, mixed multi-span
+        admission streams handled natively)."""
         T = x.shape[0]
-        mt = self._eng.max_tokens
-        if T <= mt:
+        cap = self._eng.max_tokens if decode else self._eng.max_prefill
+        if not decode and not self._mfma_path():
+            cap = self._eng.max_tokens  # legacy f32 path: 64-token tiles
+        if T <= cap:
             return self._eng.forward(x, pos, seq, decode=decode)
-        if T >= self.PREFILL_MIN_SPAN and self._prefill_ok():
-            spans = self._spans(pos.tolist(), seq.tolist())
-            # single-span streams only: that is where the library path
-            # is measured to win (one long prompt); many-span admission
-            # streams pay per-span aten overhead x layers and lose badly
-            if len(spans) == 1:
-                return self._prefill(x, pos, seq, spans)
-        # token-tile larger inputs (prefill); KV order is preserved because
-        # tile i's cache rows are written before tile i+1 attends.
+        # token-tile larger inputs; KV order is preserved because tile
+        # i's cache rows are written before tile i+1 attends.
         outs = []
-        for t0 in range(0, T, mt):
-            t1 = min(T, t0 + mt)
+        for t0 in range(0, T, cap):
+            t1 = min(T, t0 + cap)
             outs.append(self._eng.forward(
                 x[t0:t1].contiguous(), pos[t0:t1].contiguous(),
-                seq[t0:t1].contiguous()))
+                seq[t0:t1].contiguous(), decode=decode))
         return torch.cat(outs, dim=0)
 
-    # ------------------------------------------------ rocBLAS prefill path
-
-    def _prefill_ok(self) -> bool:
+    def _mfma_path(self) -> bool:
         if not self._layers_cache:
             return False
         return all(m[2] != ggml.GGML_TYPE_F32
                    for (_, _, mats) in self._layers_cache for m in mats)
-
-    def _mat_shape(self, idx: int):
-        E, F = self.hp.n_embd, self.hp.n_ff
-        return [(E, E), (E, E), (E, E), (E, E),
-                (F, E), (E, F), (F, E)][idx]
-
-    def _detile_half(self, mat, rows: int, cols: int) -> torch.Tensor:
-        """Repacked MFMA-tile weights -> plain f16 [rows, cols] (inverse
-        of repack_mfma, pure torch ops on device)."""
-        data, sc, wt = mat
-        if wt == ggml.GGML_TYPE_F16:
-            t = data[:rows * cols].view(torch.half)
-            return (t.view(rows // 16, cols // 8, 16, 8)
-                    .permute(0, 2, 1, 3).reshape(rows, cols))
-        R, nb = rows // 16, cols // 32
-        nbp = (nb + 3) & ~3
-        ab = sc[:R * nbp * 32].view(R, nbp // 4, 16, 4, 2).float()
-        alpha = ab[..., 0][:, :, None, :, :, None]  # [R,g4,1,i,kb,1]
-        beta = ab[..., 1][:, :, None, :, :, None]
-        if wt in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
-            qw = data[:R * nbp * 64].view(R, nbp // 4, 4, 16, 4)
-            shifts = torch.tensor(
-                [(j % 2) * 16 + (j // 2) * 4 for j in range(8)],
-                device=data.device, dtype=torch.int32)
-            n = ((qw.unsqueeze(-1) >> shifts) & 0xF).float()
-            if wt == ggml.GGML_TYPE_Q4_0:
-                w = alpha * (n - 8.0)
-            else:
-                w = alpha * n + beta
-            # [R,g4,ws,i,kb,8] -> [R,i,g4,kb,ws,8] -> [rows, nbp*32]
-            w = w.permute(0, 3, 1, 4, 2, 5).reshape(rows, nbp * 32)
-            return w[:, :cols].to(torch.half)
-        # W_Q8B byte stream: [R][g4][ws][i][kb][2 u32], u32 bytes
-        # [w0,w2,w1,w3] so weight jj sits at byte [0,2,1,3][jj]
-        qw = data[:R * nbp * 128].view(R, nbp // 4, 4, 16, 4, 2)
-        bshift = torch.tensor([0, 16, 8, 24], device=data.device,
-                              dtype=torch.int32)
-        u = ((qw.unsqueeze(-1) >> bshift) & 0xFF).float()
-        w = alpha.unsqueeze(-1) * (u - 128.0) + beta.unsqueeze(-1)
-        # [R,g4,ws,i,kb,2,4] -> [R,i,g4,kb,ws,2,4] -> [rows, nbp*32]
-        w = w.permute(0, 3, 1, 4, 2, 5, 6).reshape(rows, nbp * 32)
-        return w[:, :cols].to(torch.half)
-
-    def _spans(self, pos_l, seq_l):
-        """Contiguous same-sequence ascending-position runs (one request
-        each in the batched-admission prefill stream)."""
-        out, i, T = [], 0, len(pos_l)
-        while i < T:
-            j = i + 1
-            while (j < T and seq_l[j] == seq_l[i] and
-                   pos_l[j] == pos_l[j - 1] + 1):
-                j += 1
-            out.append((i, j))
-            i = j
-        return out
-
-    def _prefill(self, x, pos, seq, spans=None):
-        """Large-T forward: detile each weight matrix to f16 and run the
-        six GEMMs through rocBLAS (torch.matmul — plain library GEMMs,
-        where M >= 128 belongs); RoPE/KV/attention keep the engine's
-        exact semantics (f16 KV rows, pos-bounded causal attention), so
-        the kernel decode path attends seamlessly over rows written
-        here. Returns the f32 residual stream like the kernel path (the
-        next forward call rebuilds its own side channels at entry)."""
-        hp = self.hp
-        E, H = hp.n_embd, hp.n_head
-        D = hp.head_dim
-        T = x.shape[0]
-        kc, vc = self._eng.k_cache, self._eng.v_cache
-        pos_l = pos.tolist()
-        seq_l = seq.tolist()
-        if spans is None:
-            spans = self._spans(pos_l, seq_l)
-        dev = x.device
-        inv = ROPE_BASE ** (-2.0 * torch.arange(D // 2, device=dev,
-                                                dtype=torch.float32) / D)
-        theta = pos.float()[:, None] * inv[None, :]
-        cos, sin = torch.cos(theta), torch.sin(theta)  # [T, D/2]
-        posi = pos.long()
-        seqi = seq.long()
-        scale = 1.0 / math.sqrt(D)
-
-        def rope(t):
-            t = t.view(T, H, D)
-            t0, t1 = t[..., 0::2], t[..., 1::2]
-            o = torch.empty_like(t)
-            o[..., 0::2] = t0 * cos[:, None, :] - t1 * sin[:, None, :]
-            o[..., 1::2] = t0 * sin[:, None, :] + t1 * cos[:, None, :]
-            return o.reshape(T, E)
-
-        E2, F2 = hp.n_embd, hp.n_ff
-        cacheable = (len(self._layers_cache) *
-                     (4 * E2 * E2 + 3 * E2 * F2) * 2
-                     <= self._prefill_cache_budget)
-        for li, (an, fn, mats) in enumerate(self._layers_cache):
-            ws = self._prefill_cache.get(li)
-            if ws is None:
-                ws = [self._detile_half(mats[i], *self._mat_shape(i))
-                      for i in range(7)]
-                if cacheable:
-                    self._prefill_cache[li] = ws
-            a = (rms_norm(x) * an).to(torch.half)
-            q = rope((a @ ws[0].t()).float())
-            k = rope((a @ ws[1].t()).float())
-            v = (a @ ws[2].t()).float()
-            kc[li, seqi, posi] = k.half()
-            vc[li, seqi, posi] = v.half()
-            o = torch.empty_like(x)
-            for (i0, i1) in spans:
-                s, p0 = seq_l[i0], pos_l[i0]
-                p1 = pos_l[i1 - 1]
-                n = i1 - i0
-                qs = q[i0:i1].view(n, H, D).transpose(0, 1)      # [H,n,D]
-                keys = (kc[li, s, :p1 + 1].float()
-                        .view(p1 + 1, H, D).permute(1, 2, 0))    # [H,D,P]
-                att = torch.bmm(qs, keys) * scale                # [H,n,P]
-                js = torch.arange(p1 + 1, device=dev)
-                lim = (p0 + torch.arange(n, device=dev))[:, None]
-                att = att.masked_fill(js[None, None, :] > lim[None],
-                                      float("-inf"))
-                prob = torch.softmax(att, dim=-1)
-                vals = (vc[li, s, :p1 + 1].float()
-                        .view(p1 + 1, H, D).permute(1, 0, 2))    # [H,P,D]
-                o[i0:i1] = (torch.bmm(prob, vals)
-                            .transpose(0, 1).reshape(n, E))
-            x = x + (o.half() @ ws[3].t()).float()
-            f = (rms_norm(x) * fn).to(torch.half)
-            g = torch.nn.functional.silu((f @ ws[4].t()).float())
-            u = (f @ ws[6].t()).float()
-            x = x + ((g * u).half() @ ws[5].t()).float()
-        return x
 
     def embed(self, tokens: torch.Tensor) -> torch.Tensor:
         return self._eng.embed(tokens.to(self.device, torch.int32))

@@ -154,7 +154,7 @@ class SliceEngine {
  public:
     SliceEngine(int64_t n_embd, int64_t n_head, int64_t n_layers,
                 int64_t n_ff, int64_t n_ctx, int64_t max_batch, double eps,
-                double rope_base)
+                double rope_base, int64_t max_prefill = 1024)
         : E_((int)n_embd),
           H_((int)n_head),
           D_((int)(n_embd / n_head)),
@@ -167,6 +167,10 @@ class SliceEngine {
         TORCH_CHECK(D_ % 2 == 0, "head_dim must be even for RoPE pairs");
         TORCH_CHECK(D_ <= 256, "head_dim > 256 unsupported");
         TORCH_CHECK(E_ % 16 == 0 && F_ % 16 == 0, "E/F must be 16-aligned");
+        // prefill token cap per forward call: side channels and q/attn
+        // buffers scale with it (a few hundred MB at 2048 on 65B shapes)
+        maxP_ = std::max((int)max_prefill, kMaxTokens);
+        ssw_ = maxP_;  // per-layer stride of the sumsq side channels
         layers_.resize(L_);
         loaded_.assign(L_, false);
         auto dev = torch::TensorOptions().device(torch::kCUDA);
@@ -183,24 +187,26 @@ class SliceEngine {
             torch::arange(0, D_ / 2, f32) * (-2.0f / (float)D_))
             .contiguous();
         xn_ = torch::empty({kMaxTokens, E_}, f32);
-        qb_ = torch::empty({kMaxTokens, E_}, f32);
-        ab_ = torch::empty({kMaxTokens, E_}, f32);
+        qb_ = torch::empty({maxP_, E_}, f32);
+        ab_ = torch::empty({maxP_, E_}, f32);
         ffb_ = torch::empty({kMaxTokens, F_}, f32);
         // MFMA-path side channels
         // side channels sized for the PADDED K-block count (q4 wide-load
         // groups of 4 blocks) and zero-filled: the pad region is consumed
-        // by alpha=0 weight blocks but must hold finite f16 values
-        // per K-chunk the side channel holds up to 4 column tiles of 16
-        // tokens (JT_MAX) — layout [kc][jt][16][8] f16
-        auto side = [](int cols) {
-            return (int64_t)(((cols / 32 + 3) & ~3)) * 512 * 4 +
+        // by alpha=0 weight blocks but must hold finite f16 values.
+        // Token-panel width = jt_width(maxP_) 16-token tiles — the decode
+        // path uses the first 4 tiles of the same buffers; layout
+        // [kc][jtw][16][8] f16
+        const int64_t jtwp = jt_width(maxP_);
+        auto side = [&](int cols) {
+            return (int64_t)(((cols / 32 + 3) & ~3)) * 512 * jtwp +
                    kTailSlackSide;
         };
         xprep_ = torch::zeros({side(E_)}, u16);
         aprep_ = torch::zeros({side(E_)}, u16);
         gprep_ = torch::zeros({side(F_)}, u16);
-        ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * kMaxTokens}, f32);
-        ss_ffn_ = torch::zeros({(int64_t)L_ * kMaxTokens}, f32);
+        ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * ssw_}, f32);
+        ss_ffn_ = torch::zeros({(int64_t)L_ * ssw_}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
         argmax_keys_ = torch::zeros({kMaxTokens}, dev.dtype(torch::kInt64));
         // split-K partial slabs: sized for the largest user — qkv
@@ -280,9 +286,13 @@ class SliceEngine {
         check_i32(seq, "seq");
         const int T = (int)x.size(0);
         TORCH_CHECK(x.dim() == 2 && x.size(1) == E_, "x must be [T, E]");
-        TORCH_CHECK(T >= 1 && T <= kMaxTokens,
-                    "forward handles at most ", kMaxTokens,
+        const bool mfma = layers_.empty() ? false : layers_[0].mfma;
+        const int tcap = mfma ? maxP_ : kMaxTokens;
+        TORCH_CHECK(T >= 1 && T <= tcap,
+                    "forward handles at most ", tcap,
                     " tokens per call; tile larger batches host-side");
+        TORCH_CHECK(!decode || T <= kMaxTokens,
+                    "decode batches are capped at ", kMaxTokens);
         TORCH_CHECK(pos.numel() == T && seq.numel() == T, "pos/seq size");
         // Opt-in backstop against out-of-bounds KV writes (an oversized
         // request that slipped past host-side validation). Costs a host
@@ -322,16 +332,41 @@ class SliceEngine {
         float* ssa = ss_attn_.data_ptr<float>();
         float* ssf = ss_ffn_.data_ptr<float>();
         // zero the atomic sumsq slots, then stage x into the side channel
-        (void)hipMemsetAsync(ssa, 0, sizeof(float) * (L_ + 1) * kMaxTokens, s);
-        (void)hipMemsetAsync(ssf, 0, sizeof(float) * L_ * kMaxTokens, s);
+        (void)hipMemsetAsync(ssa, 0, sizeof(float) * (L_ + 1) * ssw_, s);
+        (void)hipMemsetAsync(ssf, 0, sizeof(float) * L_ * ssw_, s);
         launch_prep_x(s, xp, xprep, ssa, E_, T);
+        if (T > kMaxTokens) {
+            // large-M prefill: the *_mt kernels cover all T tokens in one
+            // launch per op (XCD-grouped token tiles), the whole layer
+            // loop sequenced here in C++ — no host tiling, no library
+            // GEMMs, q4 tiles read directly
+            for (int li = 0; li < L_; ++li) {
+                Layer& l = layers_[li];
+                __half* kc = kbase + (size_t)li * layer_stride;
+                __half* vc = vbase + (size_t)li * layer_stride;
+                launch_qkv16_mt(s, l.mq.w, l.mk.w, l.mv.w, xprep,
+                                u16p(l.attn_normprep), ssa + li * ssw_,
+                                eps_, qb, kc, vc, pp, sp, ifr, E_, D_,
+                                ctx_, T);
+                launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_,
+                                 E_, D_, ctx_, nullptr, 0, ifr);
+                launch_gemm16_mt(s, l.mo.w, aprep, xp, xprep,
+                                 ssf + li * ssw_, T, /*res_sq=*/1);
+                launch_ffn16_mt(s, l.m1.w, l.m3.w, xprep,
+                                u16p(l.ffn_normprep), ssf + li * ssw_,
+                                eps_, gprep, T);
+                launch_gemm16_mt(s, l.m2.w, gprep, xp, xprep,
+                                 ssa + (li + 1) * ssw_, T, /*res_sq=*/1);
+            }
+            return x;
+        }
         for (int li = 0; li < L_; ++li) {
             Layer& l = layers_[li];
             __half* kc = kbase + (size_t)li * layer_stride;
             __half* vc = vbase + (size_t)li * layer_stride;
             const int slab_used = launch_qkv16(
                 s, l.mq.w, l.mk.w, l.mv.w, xprep, u16p(l.attn_normprep),
-                ssa + li * kMaxTokens, eps_, qb, kc, vc, pp, sp, ifr, E_,
+                ssa + li * ssw_, eps_, qb, kc, vc, pp, sp, ifr, E_,
                 D_, ctx_, T, slab_.data_ptr<float>(),
                 /*skip_finish=*/decode ? 1 : 0);
             const bool fuse = decode && slab_used;
@@ -352,24 +387,24 @@ class SliceEngine {
                 launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_,
                               slab, nullptr, nullptr, T, GM_SLAB);
                 launch_reduce_prep(s, xp, slab, ks, xprep,
-                                   ssf + li * kMaxTokens, E_, T);
+                                   ssf + li * ssw_, E_, T);
             } else {
                 launch_gemm16(s, l.mo.w, aprep, nullptr, nullptr, eps_, xp,
-                              xprep, ssf + li * kMaxTokens, T, GM_RES_SQ);
+                              xprep, ssf + li * ssw_, T, GM_RES_SQ);
             }
             // ffn keeps the fused RT=1 kernel: the slab+finish variant
             // measured slower at every split/occupancy combination tried
             launch_ffn16(s, l.m1.w, l.m3.w, xprep, u16p(l.ffn_normprep),
-                         ssf + li * kMaxTokens, eps_, gprep, T,
+                         ssf + li * ssw_, eps_, gprep, T,
                          /*slab=*/nullptr);
             if (split) {
                 launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_,
                               slab, nullptr, nullptr, T, GM_SLAB);
                 launch_reduce_prep(s, xp, slab, ks, xprep,
-                                   ssa + (li + 1) * kMaxTokens, E_, T);
+                                   ssa + (li + 1) * ssw_, E_, T);
             } else {
                 launch_gemm16(s, l.m2.w, gprep, nullptr, nullptr, eps_, xp,
-                              xprep, ssa + (li + 1) * kMaxTokens, T,
+                              xprep, ssa + (li + 1) * ssw_, T,
                               GM_RES_SQ);
             }
         }
@@ -438,6 +473,7 @@ class SliceEngine {
     }
 
     int64_t max_tokens() const { return kMaxTokens; }
+    int64_t max_prefill() const { return maxP_; }
     int64_t n_ctx() const { return ctx_; }
     int64_t max_batch() const { return B_; }
     // the per-layer KV tensors ([L, B, ctx, E] f16); exposed so the
@@ -473,6 +509,8 @@ class SliceEngine {
     }
 
     int E_, H_, D_, F_, L_, ctx_, B_;
+    int maxP_ = kMaxTokens;  // prefill token cap per forward call
+    int ssw_ = kMaxTokens;   // per-layer sumsq side-channel stride
     int V_ = 0;
     float eps_;
     std::vector<Layer> layers_;
@@ -494,10 +532,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "MI355X-native layer-slice inference engine (CDNA4 HIP kernels)";
     py::class_<SliceEngine, std::shared_ptr<SliceEngine>>(m, "SliceEngine")
         .def(py::init<int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
-                      double, double>(),
+                      double, double, int64_t>(),
              py::arg("n_embd"), py::arg("n_head"), py::arg("n_layers"),
              py::arg("n_ff"), py::arg("n_ctx"), py::arg("max_batch"),
-             py::arg("eps"), py::arg("rope_base"))
+             py::arg("eps"), py::arg("rope_base"),
+             py::arg("max_prefill") = 1024)
         .def("set_layer", &SliceEngine::set_layer)
         .def("set_extra", &SliceEngine::set_extra)
         .def("forward", &SliceEngine::forward, py::arg("x"),
@@ -507,6 +546,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
              py::arg("all_logits") = false)
         .def("argmax", &SliceEngine::argmax)
         .def_property_readonly("max_tokens", &SliceEngine::max_tokens)
+        .def_property_readonly("max_prefill", &SliceEngine::max_prefill)
         .def_property_readonly("n_ctx", &SliceEngine::n_ctx)
         .def_property_readonly("max_batch", &SliceEngine::max_batch)
         .def_property_readonly("k_cache", &SliceEngine::k_cache)

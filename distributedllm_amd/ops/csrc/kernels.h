@@ -102,3 +102,26 @@ void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
 
 void launch_argmax(hipStream_t s, const float* logits,
                    unsigned long long* keys, int* out, int T, int V);
+
+// ---------------------------------------------------------- prefill path
+// Large-M (T > 64) layer kernels: one launch covers all T tokens with an
+// XCD-aware (row-tile x 64-token-group) grid. Side channels use the
+// jt_width(T) token-panel stride.
+int jt_width(int T);
+
+void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                     const WMat2& wv, const unsigned short* xprep,
+                     const unsigned short* normprep, const float* ss_in,
+                     float eps, float* q_buf, __half* k_cache_layer,
+                     __half* v_cache_layer, const int* pos, const int* seq,
+                     const float* inv_freq, int E, int D, int n_ctx, int T);
+
+void launch_gemm16_mt(hipStream_t s, const WMat2& w,
+                      const unsigned short* bprep, float* y,
+                      unsigned short* xprep_out, float* ss_out, int T,
+                      int res_sq);
+
+void launch_ffn16_mt(hipStream_t s, const WMat2& w1, const WMat2& w3,
+                     const unsigned short* xprep,
+                     const unsigned short* normprep, const float* ss_in,
+                     float eps, unsigned short* gprep, int T);

@@ -840,6 +840,10 @@ __device__ __forceinline__ void a_frag_q8(uint32_t qA, uint32_t qB,
 // RT = row tiles per wave: the wave's B fragments feed RT A-tile streams,
 // dividing the (L2-heavy) B-panel re-read traffic and the B-build VALU by
 // RT. acc[rt][n][jt][jj].
+// jtw/jt0: token-panel stride and first 16-token tile of this block in the
+// xprep side channel (layout [kc][jtw][16][8]). Decode callers pass
+// (JT, 0); the prefill *_mt kernels tile a T-wide panel with jtw = ceil(T/16)
+// padded to the 4-tile group and jt0 = mtile*JT.
 template <int WT, bool NORM, int NM, int JT, int RT = 1, int PF = 4,
           int NW = NWAVES>
 __device__ __forceinline__ void wave_tile_kloop(
@@ -847,7 +851,7 @@ __device__ __forceinline__ void wave_tile_kloop(
     const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
     const float* __restrict__ ss_in, float eps,
-    float acc[RT][NM][JT][4], int b0, int b1) {
+    float acc[RT][NM][JT][4], int b0, int b1, int jtw = JT, int jt0 = 0) {
     KLoop kl;
     kl.init_range(b0, b1, (WT == W_F16) ? 1 : 4, NW);
     const int nb0 = ws[0]->cols >> 5;
@@ -880,7 +884,7 @@ __device__ __forceinline__ void wave_tile_kloop(
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt)
             scale2[jt] = __float2half2_rn(
-                rsqrtf(ss_in[jt * 16 + kl.i] * inv_cols + eps));
+                rsqrtf(ss_in[(jt0 + jt) * 16 + kl.i] * inv_cols + eps));
     }
 
     const uint32_t* qp[RT][NM];
@@ -900,9 +904,9 @@ __device__ __forceinline__ void wave_tile_kloop(
                         ((size_t)tr * (ws[n]->cols >> 3)) * 128 +
                         ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
         }
-    // xprep layout with JT: element (kc, jt, j, e) at ((kc*JT+jt)*16+j)*8+e
+    // xprep layout: element (kc, jt, j, e) at ((kc*jtw + jt)*16 + j)*8 + e
     const unsigned short* xp =
-        xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * JT) * 128 + kl.i * 8;
+        xprep + ((size_t)(kl.kb0 * 4 + kl.ks) * jtw + jt0) * 128 + kl.i * 8;
     const unsigned short* np =
         normprep + (NORM ? (size_t)(kl.kb0 * 4 + kl.ks) * 8 : 0);
 
@@ -963,11 +967,11 @@ __device__ __forceinline__ void wave_tile_kloop(
 #pragma unroll
             for (int jt = 0; jt < JT; ++jt)
                 px.xb[u][jt] = *reinterpret_cast<const uint4*>(
-                    xp + (u * 4 * JT + jt) * 128);  // 4 kc-chunks per kb
+                    xp + ((size_t)u * 4 * jtw + jt) * 128);  // 4 kc per kb
             if (NORM)
                 px.nbv[u] = *reinterpret_cast<const uint4*>(np + u * 32);
         }
-        xp += PF * 4 * JT * 128;
+        xp += (size_t)PF * 4 * jtw * 128;
         if (NORM) np += PF * 32;
     };
 
@@ -1069,7 +1073,7 @@ __device__ __forceinline__ void wave_tile_kloop(
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt)
             xb[jt] = *reinterpret_cast<const uint4*>(xp + jt * 128);
-        xp += 4 * JT * 128;
+        xp += (size_t)4 * jtw * 128;
         if (NORM) {
             nbv = *reinterpret_cast<const uint4*>(np);
             np += 32;
@@ -1654,6 +1658,194 @@ __global__ void k_ffn_finish(const float* __restrict__ slab, int ks,
         o;
 }
 
+// ================================================== prefill (large-M) path
+// Hand-written large-M dequant-GEMM: retires the rocBLAS-over-detiled-f16
+// prefill path (q4/byte/f16 tiles read DIRECTLY, no f16 weight copy) and
+// the 64-token host tiling. One launch covers all T tokens: the grid is
+// (gtile row-tiles) x (MT 64-token groups), flattened with an XCD-aware
+// decode so all MT token-groups of one row tile land on the SAME XCD
+// back-to-back — the row tile's weight stream is read from HBM once and
+// re-served to the other groups from that XCD's L2 (blockIdx -> XCD is
+// round-robin on dispatch order, 8 XCDs; MI355X_MICROARCH.md).
+//
+// Returns (gtile, mtile); gtile may be >= GT (pad block -> caller exits).
+__device__ __forceinline__ void xcd_decode(int GT, int MT, int& gtile,
+                                           int& mtile) {
+    const int xcd = blockIdx.x & 7;
+    const int q = blockIdx.x >> 3;
+    mtile = q % MT;
+    gtile = (q / MT) * 8 + xcd;
+    (void)GT;
+}
+
+static inline int xcd_grid(int GT, int MT) {
+    return ((GT + 7) & ~7) * MT;
+}
+
+// QKV projections for token group mtile + RoPE + KV append (the large-M
+// analog of k_qkv16; epilogue identical, tokens indexed globally).
+template <int WT, int JT = 4>
+__global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
+    WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps, float* __restrict__ q_buf,
+    __half* __restrict__ k_cache, __half* __restrict__ v_cache,
+    const int* __restrict__ pos, const int* __restrict__ seq,
+    const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T,
+    int MT, int jtw) {
+    int gtile, mtile;
+    const int GT = 3 * (E >> 4);
+    xcd_decode(GT, MT, gtile, mtile);
+    if (gtile >= GT) return;
+    const int mat = gtile / (E >> 4);
+    const int tile = gtile % (E >> 4);
+    const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float acc[1][1][JT][4];
+    const WMat2* ws[1] = {&w};
+    const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
+    wave_tile_kloop<WT, true, 1, JT>(ws, tile, xprep, normprep, ss_in, eps,
+                                     acc, 0, nbe, jtw, mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * JT];
+    combine_acc<JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
+    const int r0 = tile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = (mtile * JT + jt) * 16 + j;
+        if (j2 >= T) continue;
+        const int p = pos[j2];
+        if (mat == 2) {
+            __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj)
+                dst[jj] = __float2half(acc[0][0][jt][jj]);
+            continue;
+        }
+#pragma unroll
+        for (int q2 = 0; q2 < 2; ++q2) {
+            const int e = r0 + 2 * q2;
+            const int d = e % D;
+            const float theta = (float)p * inv_freq[d >> 1];
+            float sn, cs;
+            __sincosf(theta, &sn, &cs);
+            const float x0 = acc[0][0][jt][2 * q2];
+            const float x1 = acc[0][0][jt][2 * q2 + 1];
+            const float o0 = x0 * cs - x1 * sn;
+            const float o1 = x0 * sn + x1 * cs;
+            if (mat == 0) {
+                q_buf[(size_t)j2 * E + e] = o0;
+                q_buf[(size_t)j2 * E + e + 1] = o1;
+            } else {
+                __half* dst =
+                    k_cache + ((size_t)seq[j2] * n_ctx + p) * E + e;
+                dst[0] = __float2half(o0);
+                dst[1] = __float2half(o1);
+            }
+        }
+    }
+}
+
+// Large-M GEMM with the fused residual + sumsq + xprep epilogue (wo / w2
+// consumers) or plain store. RES_SQ semantics match k_gemm16<GM_RES_SQ>.
+template <int WT, bool RES_SQ, int JT = 4>
+__global__ __launch_bounds__(BLOCK) void k_gemm16_mt(
+    WMat2 w, const unsigned short* __restrict__ bprep,
+    float* __restrict__ y, unsigned short* __restrict__ xprep_out,
+    float* __restrict__ ss_out, int T, int MT, int jtw) {
+    int gtile, mtile;
+    const int GT = w.rows >> 4;
+    xcd_decode(GT, MT, gtile, mtile);
+    if (gtile >= GT) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float acc[1][1][JT][4];
+    const WMat2* ws[1] = {&w};
+    const int nbt = w.cols >> 5;
+    const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);
+    wave_tile_kloop<WT, false, 1, JT>(ws, gtile, bprep, nullptr, nullptr,
+                                      0.f, acc, 0, nbk, jtw, mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * JT];
+    combine_acc<JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
+    const int r0 = gtile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = (mtile * JT + jt) * 16 + j;
+        if (j2 >= T) continue;
+        float sq = 0.0f;
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            float v = acc[0][0][jt][jj];
+            if (RES_SQ) {
+                v += y[(size_t)j2 * w.rows + r0 + jj];
+                sq += v * v;
+            }
+            y[(size_t)j2 * w.rows + r0 + jj] = v;
+            acc[0][0][jt][jj] = v;
+        }
+        if (RES_SQ && xprep_out != nullptr) {
+            uint2 o;
+            o.x = pack_f16(acc[0][0][jt][0], acc[0][0][jt][1]);
+            o.y = pack_f16(acc[0][0][jt][2], acc[0][0][jt][3]);
+            *reinterpret_cast<uint2*>(
+                xprep_out +
+                (((size_t)(r0 >> 3) * jtw + mtile * JT + jt) * 16 + j) * 8 +
+                (r0 & 7)) = o;
+        }
+        if (RES_SQ && ss_out != nullptr) {
+            float s2 = sq;
+            s2 += __shfl_xor(s2, 16);
+            s2 += __shfl_xor(s2, 32);
+            if (lane < 16) atomicAdd(ss_out + j2, s2);
+        }
+    }
+}
+
+// Large-M w1/w3 + fused RMSNorm + SwiGLU -> gprep (analog of k_ffn16).
+template <int WT, int JT = 4>
+__global__ __launch_bounds__(BLOCK) void k_ffn16_mt(
+    WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
+    const unsigned short* __restrict__ normprep,
+    const float* __restrict__ ss_in, float eps,
+    unsigned short* __restrict__ gprep, int T, int MT, int jtw) {
+    int gtile, mtile;
+    const int GT = w1.rows >> 4;
+    xcd_decode(GT, MT, gtile, mtile);
+    if (gtile >= GT) return;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const int j = lane & 15;
+    float acc[1][2][JT][4];
+    const WMat2* ws[2] = {&w1, &w3};
+    const int nbf = (WT == W_F16) ? (w1.cols >> 5)
+                                  : (((w1.cols >> 5) + 3) & ~3);
+    wave_tile_kloop<WT, true, 2, JT>(ws, gtile, xprep, normprep, ss_in,
+                                     eps, acc, 0, nbf, jtw, mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * 2 * JT];
+    combine_acc<2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    if (threadIdx.x >= WAVE) return;
+    const int r0 = gtile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int jt = 0; jt < JT; ++jt) {
+        const int j2 = (mtile * JT + jt) * 16 + j;
+        if (j2 >= T) continue;
+        float g[4];
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+            const float v1 = acc[0][0][jt][jj];
+            const float silu = v1 / (1.0f + __expf(-v1));
+            g[jj] = silu * acc[0][1][jt][jj];
+        }
+        uint2 o;
+        o.x = pack_f16(g[0], g[1]);
+        o.y = pack_f16(g[2], g[3]);
+        *reinterpret_cast<uint2*>(
+            gprep + (((size_t)(r0 >> 3) * jtw + mtile * JT + jt) * 16 + j)
+                        * 8 + (r0 & 7)) = o;
+    }
+}
+
 // ============================================================== launchers
 
 // column-tile count for a token batch T (1, 2 or 4 tiles of 16)
@@ -1661,6 +1853,14 @@ static inline int pick_jt(int T) {
     if (T <= 16) return 1;
     if (T <= 32) return 2;
     return 4;
+}
+
+// xprep token-panel width in 16-token tiles: the decode path packs up to
+// 4 tiles; the prefill path packs ceil(T/16) padded to whole 4-tile
+// groups (JT=4 kernels read whole groups; pad tiles are dead columns)
+int jt_width(int T) {
+    if (T <= 64) return pick_jt(T);
+    return (((T + 15) >> 4) + 3) & ~3;
 }
 
 static inline int pick_tmax(int T) {
@@ -1755,13 +1955,13 @@ void launch_attention(hipStream_t s, const float* q_buf,
     if (qkv_slab != nullptr) {
         hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
-                           out_prep, pos, seq, E, D, n_ctx, pick_jt(T),
+                           out_prep, pos, seq, E, D, n_ctx, jt_width(T),
                            qkv_slab, ks, inv_freq);
         return;
     }
     hipLaunchKernelGGL(k_attention<false>, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos, seq,
-                       E, D, n_ctx, pick_jt(T), qkv_slab, ks, inv_freq);
+                       E, D, n_ctx, jt_width(T), qkv_slab, ks, inv_freq);
 }
 
 // ------------------------------------------------- MFMA-path launchers
@@ -1776,7 +1976,7 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
     const int want = (128 + T - 1) / max(T, 1);
     const int chunks = max(1, min(want, (cols >> 3) / 64));
     hipLaunchKernelGGL(k_prep_x, dim3(T, chunks), dim3(BLOCK), 0, s, x,
-                       xprep, ss, cols, pick_jt(T));
+                       xprep, ss, cols, jt_width(T));
 }
 
 #define DISPATCH_JT(JTV, ...)            \
@@ -2031,6 +2231,54 @@ void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
                                tab, tokens, out, E);
             break;
     }
+}
+
+// ------------------------------------------------- prefill-path launchers
+
+void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
+                     const WMat2& wv, const unsigned short* xprep,
+                     const unsigned short* normprep, const float* ss_in,
+                     float eps, float* q_buf, __half* k_cache_layer,
+                     __half* v_cache_layer, const int* pos, const int* seq,
+                     const float* inv_freq, int E, int D, int n_ctx,
+                     int T) {
+    const int MT = (T + 63) >> 6;
+    const int jtw = jt_width(T);
+    const dim3 grid(xcd_grid(3 * (E >> 4), MT));
+    DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
+        (k_qkv16_mt<WTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
+        normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos,
+        seq, inv_freq, E, D, n_ctx, T, MT, jtw));
+}
+
+void launch_gemm16_mt(hipStream_t s, const WMat2& w,
+                      const unsigned short* bprep, float* y,
+                      unsigned short* xprep_out, float* ss_out, int T,
+                      int res_sq) {
+    const int MT = (T + 63) >> 6;
+    const int jtw = jt_width(T);
+    const dim3 grid(xcd_grid(w.rows >> 4, MT));
+    if (res_sq) {
+        DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
+            (k_gemm16_mt<WTc, true>), grid, dim3(BLOCK), 0, s, w, bprep, y,
+            xprep_out, ss_out, T, MT, jtw));
+    } else {
+        DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
+            (k_gemm16_mt<WTc, false>), grid, dim3(BLOCK), 0, s, w, bprep, y,
+            xprep_out, ss_out, T, MT, jtw));
+    }
+}
+
+void launch_ffn16_mt(hipStream_t s, const WMat2& w1, const WMat2& w3,
+                     const unsigned short* xprep,
+                     const unsigned short* normprep, const float* ss_in,
+                     float eps, unsigned short* gprep, int T) {
+    const int MT = (T + 63) >> 6;
+    const int jtw = jt_width(T);
+    const dim3 grid(xcd_grid(w1.rows >> 4, MT));
+    DISPATCH_WT2(w1.wtype, hipLaunchKernelGGL(
+        (k_ffn16_mt<WTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep, normprep,
+        ss_in, eps, gprep, T, MT, jtw));
 }
 
 void launch_argmax(hipStream_t s, const float* logits,

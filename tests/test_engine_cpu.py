@@ -105,13 +105,15 @@ def test_byte_quant_repack_values_exact():
 
 
 def test_detile_inverts_repack():
-    """The prefill path's detile must exactly invert repack_mfma: for
-    every quant format, repack (to CPU tensors) then _detile_half
-    reproduces the codec's dequantized weights (f16-rounded)."""
+    """detile_mfma (the reference inverse of the kernel tile layouts)
+    must exactly invert repack_mfma: for every quant format, repack (to
+    CPU tensors) then detile reproduces the codec's dequantized weights
+    (f16-rounded) — this is the CPU-side proof that the layouts the
+    prefill/decode kernels stream carry the exact quantized values."""
     import numpy as np
     import torch
     from distributedllm_amd.engine.slice_engine import (
-        HIPSliceEngine, repack_mfma)
+        detile_mfma, repack_mfma)
     from distributedllm_amd.formats import ggml, synthetic
     for ft in (ggml.FTYPE_MOSTLY_Q4_0, ggml.FTYPE_MOSTLY_Q4_1,
                ggml.FTYPE_MOSTLY_Q8_0, ggml.FTYPE_MOSTLY_Q5_0,
@@ -121,7 +123,7 @@ def test_detile_inverts_repack():
                  if x.name.endswith("feed_forward.w1.weight"))
         rows, cols = t.shape_rows_cols
         mat = repack_mfma(t, "cpu")
-        got = HIPSliceEngine._detile_half(None, mat, rows, cols).float()
+        got = detile_mfma(mat, rows, cols).float()
         want = torch.from_numpy(t.to_f32())
         # detile emits f16 (one rounding beyond the codec's own f16
         # arithmetic); q4_1/q5_1 fold beta=m (+16d) with one more f16
@@ -129,20 +131,3 @@ def test_detile_inverts_repack():
         tol = 4e-3 * want.abs().max().item() + 1e-6
         err = (got - want).abs().max().item()
         assert err <= tol, (ggml.TYPE_NAMES[ggml._FTYPE_TO_GGML[ft]], err)
-
-
-def test_prefill_span_splitting():
-    """_spans finds contiguous same-sequence ascending-position runs —
-    the geometry the prefill path's causal masks depend on."""
-    from distributedllm_amd.engine.slice_engine import HIPSliceEngine
-    spans = HIPSliceEngine._spans
-    assert spans(None, [0, 1, 2], [0, 0, 0]) == [(0, 3)]
-    # sequence change splits
-    assert spans(None, [0, 1, 0, 1], [0, 0, 1, 1]) == [(0, 2), (2, 4)]
-    # position gap splits (restart of the same sequence)
-    assert spans(None, [0, 1, 5, 6], [0, 0, 0, 0]) == [(0, 2), (2, 4)]
-    # descending positions split to singletons
-    assert spans(None, [3, 2, 1], [0, 0, 0]) == [(0, 1), (1, 2), (2, 3)]
-    # the pipeline prime token-major layout: seq alternates every token
-    assert spans(None, [0, 0, 1, 1], [0, 1, 0, 1]) == \
-        [(0, 1), (1, 2), (2, 3), (3, 4)]

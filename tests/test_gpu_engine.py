@@ -441,55 +441,70 @@ def test_slice_chain_matches_monolith_on_gpu():
                   label="slice chain vs monolith")
 
 
-def test_rocblas_prefill_path_matches_cpu():
-    """T >= PREFILL_MIN_T takes the rocBLAS prefill path (detiled f16
-    weights, library GEMMs, torch attention over the engine's KV) —
-    must match the fp32 CPU reference like the kernel path does, and
-    the kernel decode path must attend seamlessly over its KV rows."""
+def test_native_prefill_path_matches_cpu():
+    """T > 64 takes the native large-M prefill path (k_qkv16_mt /
+    k_gemm16_mt / k_ffn16_mt — XCD-grouped token tiles, q4 read
+    directly, the layer loop sequenced in C++) — must match the fp32
+    CPU reference, and the kernel decode path must attend seamlessly
+    over the KV rows it wrote."""
     f, hip, cpu = _engines(preset="small", n_ctx=512, max_batch=2)
     hp = f.hparams
-    assert hip._prefill_ok()
+    assert hip._mfma_path()
     torch.manual_seed(11)
-    T = 400  # span >= PREFILL_MIN_SPAN -> rocBLAS prefill branch
+    T = 400  # > 64, not a multiple of 64 -> exercises the pad tiles
     x = torch.randn(T, hp.n_embd) * 0.5
     pos = torch.arange(T, dtype=torch.int32)
     seq = torch.zeros(T, dtype=torch.int32)
     y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
     y_cpu = cpu.forward(x.clone(), pos, seq)
-    _assert_close(y_gpu, y_cpu, label="rocblas prefill")
-    # decode the next token through the KERNEL path on the same KV
-    assert len(hip._prefill_cache) > 0  # the branch actually ran
+    _assert_close(y_gpu, y_cpu, label="native prefill")
+    # decode the next token through the decode path on the same KV
     xd = torch.randn(1, hp.n_embd) * 0.5
     pd = torch.tensor([T], dtype=torch.int32)
     sd = torch.zeros(1, dtype=torch.int32)
     y2_gpu = hip.forward(xd.cuda(), pd.cuda(), sd.cuda(),
                          decode=True).cpu()
     y2_cpu = cpu.forward(xd.clone(), pd, sd)
-    _assert_close(y2_gpu, y2_cpu, label="decode after rocblas prefill")
+    _assert_close(y2_gpu, y2_cpu, label="decode after native prefill")
 
 
-def test_rocblas_prefill_mixed_stream_falls_back():
-    """A many-span admission stream must NOT take the library path (it
-    pays per-span overhead x layers) — it falls back to 64-token tiles
-    and stays correct; a follow-up single-span call on another sequence
-    takes the library path against the same KV."""
+def test_native_prefill_mixed_stream_matches_cpu():
+    """A mixed multi-span admission stream (two prompts of different
+    sequences concatenated in one call) runs through the SAME native
+    prefill kernels — per-token pos/seq indexing, no span special-
+    casing (the round-1 rocBLAS path was single-span only)."""
     f, hip, cpu = _engines(preset="small", n_ctx=512, max_batch=2)
     hp = f.hparams
     torch.manual_seed(12)
-    n0, n1 = 400, 80
+    n0, n1 = 200, 80
     x = torch.randn(n0 + n1, hp.n_embd) * 0.5
     pos = torch.cat([torch.arange(n0), torch.arange(n1)]).to(torch.int32)
     seq = torch.cat([torch.zeros(n0), torch.ones(n1)]).to(torch.int32)
-    assert len(hip._spans(pos.tolist(), seq.tolist())) == 2
     y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
     y_cpu = cpu.forward(x.clone(), pos, seq)
-    _assert_close(y_gpu, y_cpu, label="mixed-stream tile fallback")
-    assert len(hip._prefill_cache) == 0  # library path did not run
-    # single-span continuation of sequence 1 through the library path
-    x2 = torch.randn(400, hp.n_embd) * 0.5
-    pos2 = torch.arange(n1, n1 + 400, dtype=torch.int32)
-    seq2 = torch.ones(400, dtype=torch.int32)
+    _assert_close(y_gpu, y_cpu, label="mixed-stream native prefill")
+    # single-span continuation of sequence 1 against the same KV
+    x2 = torch.randn(300, hp.n_embd) * 0.5
+    pos2 = torch.arange(n1, n1 + 300, dtype=torch.int32)
+    seq2 = torch.ones(300, dtype=torch.int32)
     y2_gpu = hip.forward(x2.cuda(), pos2.cuda(), seq2.cuda()).cpu()
     y2_cpu = cpu.forward(x2.clone(), pos2, seq2)
-    assert len(hip._prefill_cache) > 0
-    _assert_close(y2_gpu, y2_cpu, label="library continuation")
+    _assert_close(y2_gpu, y2_cpu, label="prefill continuation")
+
+
+@pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q8_0,
+                                   ggml.FTYPE_MOSTLY_F16])
+def test_native_prefill_other_wtypes(ftype):
+    """The byte-stream (W_Q8B) and f16 tile paths run the same _mt
+    prefill kernels."""
+    f, hip, cpu = _engines(preset="small", ftype=ftype, n_ctx=256,
+                           max_batch=1)
+    hp = f.hparams
+    torch.manual_seed(13)
+    T = 130
+    x = torch.randn(T, hp.n_embd) * 0.5
+    pos = torch.arange(T, dtype=torch.int32)
+    seq = torch.zeros(T, dtype=torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label=f"native prefill ftype={ftype}")
