@@ -1065,6 +1065,22 @@ __device__ __forceinline__ void wave_tile_kloop(
             compute_batch(bufB, pxB);
             if (++it == nfull) break;
         }
+        // load_w ran nfull+1 times (the trailing prefetch) but only
+        // nfull batches were consumed — rewind one batch so the tail
+        // loop reads the right K-blocks. (Latent until the prefill _mt
+        // kernels: 4-aligned q4 ranges never leave a tail, and the f16
+        // slab kernels' per-wave ranges are < PF so nfull == 0.)
+#pragma unroll
+        for (int rt = 0; rt < RT; ++rt)
+#pragma unroll
+            for (int n = 0; n < NM; ++n) {
+                if (WT == W_F16) {
+                    tp[rt][n] -= PF * 512;
+                } else {
+                    qp[rt][n] -= PF * 64 * QW;
+                    abp[rt][n] -= PF * 16;
+                }
+            }
     }
     for (int g = kl.kb0 + nfull * PF; g < kl.kb1; ++g) {
         uint32_t q[RT][NM][QW], ab[RT][NM];

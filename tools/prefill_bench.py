@@ -3,6 +3,9 @@
 tokens (the native large-M kernel path for S > 64). Round-1 baselines
 to beat (3B q4_0, rocBLAS-over-detiled-f16): 32.9k tok/s @ S=512,
 37.8k @ 1024; 64-token tile path ~20-25k."""
+import os
+import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import argparse
 import json
 import time
