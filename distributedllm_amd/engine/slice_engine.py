@@ -303,14 +303,16 @@ class HIPSliceEngine:
     @classmethod
     def random(cls, hp: ggml.Hparams, n_layers: int, first_layer: int = 0,
                n_ctx: int = 2048, max_batch: int = 16, seed: int = 0,
-               with_extra: bool = True) -> "HIPSliceEngine":
+               with_extra: bool = True,
+               max_prefill: Optional[int] = None) -> "HIPSliceEngine":
         """Random-init engine straight on the GPU (synthetic benchmarking).
 
         Generates weights directly in the repacked kernel layout — identical
         compute and HBM traffic to a real checkpoint of this architecture,
         without materializing a multi-GB GGML file on disk.
         """
-        eng = cls(hp, n_layers, first_layer, n_ctx, max_batch)
+        eng = cls(hp, n_layers, first_layer, n_ctx, max_batch,
+                  max_prefill=max_prefill)
         g = torch.Generator(device="cuda")
         g.manual_seed(seed)
         E, F, V = hp.n_embd, hp.n_ff, hp.n_vocab
