@@ -348,8 +348,8 @@ class SliceEngine {
                                 u16p(l.attn_normprep), ssa + li * ssw_,
                                 eps_, qb, kc, vc, pp, sp, ifr, E_, D_,
                                 ctx_, T);
-                launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_,
-                                 E_, D_, ctx_, nullptr, 0, ifr);
+                launch_attn_prefill(s, qb, kc, vc, ab, aprep, pp, sp, T,
+                                    H_, E_, D_, ctx_);
                 launch_gemm16_mt(s, l.mo.w, aprep, xp, xprep,
                                  ssf + li * ssw_, T, /*res_sq=*/1);
                 launch_ffn16_mt(s, l.m1.w, l.m3.w, xprep,

@@ -109,6 +109,15 @@ void launch_argmax(hipStream_t s, const float* logits,
 // jt_width(T) token-panel stride.
 int jt_width(int T);
 
+// Q-tiled (16 queries/block) streaming attention for prefill: each K/V
+// row is read once per 16 queries; mixed-seq tiles segment internally
+void launch_attn_prefill(hipStream_t s, const float* q_buf,
+                         const __half* k_cache_layer,
+                         const __half* v_cache_layer, float* out,
+                         unsigned short* out_prep, const int* pos,
+                         const int* seq, int T, int H, int E, int D,
+                         int n_ctx);
+
 void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
                      const WMat2& wv, const unsigned short* xprep,
                      const unsigned short* normprep, const float* ss_in,

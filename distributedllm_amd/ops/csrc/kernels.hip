@@ -522,6 +522,167 @@ __global__ void k_attention(
     }
 }
 
+// ------------------------------------------------- prefill attention (QT=16)
+// Q-tiled streaming attention for the large-M prefill path: one block
+// handles QT=16 query tokens of a (head, 16-token tile), so each K/V row
+// is read ONCE per 16 queries instead of once per query — 16x less HBM
+// traffic than k_attention's decode shape (which is exactly bandwidth-
+// bound at large T: 22 ms of a 48 ms span-1024 forward, profiles/).
+// Mixed admission streams: a tile may span several sequences; the block
+// processes each same-seq segment separately (causality per token via
+// its own pos, exactly like k_attention).
+__global__ void k_attn_prefill(
+    const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
+    const __half* __restrict__ v_cache, float* __restrict__ out,
+    unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
+    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw, int T) {
+    constexpr int QT = 16;
+    const int t0 = blockIdx.x * QT;
+    const int h = blockIdx.y;
+    const int nq = min(QT, T - t0);
+    const float inv_sqrt_d = rsqrtf((float)D);
+
+    extern __shared__ float smem[];
+    float* lds_q = smem;                  // [QT][D] prescaled q
+    float* lds_s = lds_q + QT * D;        // [QT][BLOCK] scores -> p
+    float* lds_mred = lds_s + QT * BLOCK; // [QT] chunk m_new
+    float* lds_pred = lds_mred + QT;      // [QT] chunk sum_p
+    __shared__ int lds_pos[QT];
+    __shared__ int lds_seq[QT];
+
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    if (tid < nq) {
+        lds_pos[tid] = pos[t0 + tid];
+        lds_seq[tid] = seq[t0 + tid];
+    }
+    for (int i = tid; i < nq * D; i += BLOCK) {
+        const int q = i / D, d = i % D;
+        lds_q[q * D + d] =
+            q_buf[(size_t)(t0 + q) * E + h * D + d] * inv_sqrt_d;
+    }
+    __syncthreads();
+
+    float o[QT];      // thread d accumulates output[d] for each q
+    float m[QT], l[QT];
+#pragma unroll
+    for (int q = 0; q < QT; ++q) {
+        o[q] = 0.0f;
+        m[q] = -INFINITY;
+        l[q] = 0.0f;
+    }
+
+    int a = 0;
+    while (a < nq) {  // same-sequence segments of the tile
+        const int sseq = lds_seq[a];
+        int b = a + 1;
+        while (b < nq && lds_seq[b] == sseq) ++b;
+        int jmax = 0;
+        for (int q = a; q < b; ++q) jmax = max(jmax, lds_pos[q]);
+        const int J = jmax + 1;
+        const size_t base = (size_t)sseq * n_ctx * E + h * D;
+
+        for (int j0 = 0; j0 < J; j0 += BLOCK) {
+            const int jj = j0 + tid;
+            // scores: thread jj reads K row once, dots all segment q's
+            if (jj < J) {
+                const __half* krow = k_cache + base + (size_t)jj * E;
+                const int nh2 = D >> 2;
+                const float2* k2 = reinterpret_cast<const float2*>(krow);
+                float acc[QT];
+#pragma unroll
+                for (int q = 0; q < QT; ++q) acc[q] = 0.0f;
+                for (int c = 0; c < nh2; ++c) {
+                    const float2 raw = k2[c];
+                    const __half2* hh =
+                        reinterpret_cast<const __half2*>(&raw);
+                    const float2 ka = __half22float2(hh[0]);
+                    const float2 kb = __half22float2(hh[1]);
+                    for (int q = a; q < b; ++q) {
+                        const float* qr = lds_q + q * D + c * 4;
+                        float s = fmaf(ka.x, qr[0], 0.0f);
+                        s = fmaf(ka.y, qr[1], s);
+                        s = fmaf(kb.x, qr[2], s);
+                        s = fmaf(kb.y, qr[3], s);
+                        acc[q] += s;
+                    }
+                }
+                for (int q = a; q < b; ++q)
+                    lds_s[q * BLOCK + tid] =
+                        (jj <= lds_pos[q]) ? acc[q] : -INFINITY;
+            } else {
+                for (int q = a; q < b; ++q)
+                    lds_s[q * BLOCK + tid] = -INFINITY;
+            }
+            __syncthreads();
+            // per-q online-softmax bookkeeping: wave w owns q = w (mod 4)
+            for (int q = a + wid; q < b; q += NWAVES) {
+                float* row = lds_s + q * BLOCK;
+                float wm = -INFINITY;
+#pragma unroll
+                for (int i = 0; i < BLOCK / WAVE; ++i)
+                    wm = fmaxf(wm, row[i * WAVE + lane]);
+                wm = wave_reduce_max(wm);
+                wm = __shfl(wm, 0);
+                const float m_new = fmaxf(m[q], wm);
+                float ws = 0.0f;
+#pragma unroll
+                for (int i = 0; i < BLOCK / WAVE; ++i) {
+                    const float p = __expf(row[i * WAVE + lane] - m_new);
+                    row[i * WAVE + lane] = p;
+                    ws += p;
+                }
+                ws = wave_reduce_sum(ws);
+                if (lane == 0) {
+                    lds_mred[q] = m_new;
+                    lds_pred[q] = ws;
+                }
+            }
+            __syncthreads();
+            // V accumulation: thread d sums p[q][j] * V[j][d]
+            if (tid < D) {
+                const int jlim = min(BLOCK, J - j0);
+                const __half* vcol = v_cache + base + (size_t)j0 * E + tid;
+                float alpha[QT];
+                for (int q = a; q < b; ++q) {
+                    const float m_new = lds_mred[q];
+                    alpha[q] = (m[q] == -INFINITY)
+                                   ? 0.0f
+                                   : __expf(m[q] - m_new);
+                    o[q] *= alpha[q];
+                    l[q] = l[q] * alpha[q] + lds_pred[q];
+                    m[q] = m_new;
+                }
+                for (int jc = 0; jc < jlim; ++jc) {
+                    const float v = __half2float(vcol[(size_t)jc * E]);
+                    for (int q = a; q < b; ++q)
+                        o[q] = fmaf(lds_s[q * BLOCK + jc], v, o[q]);
+                }
+            } else {
+                // keep m/l consistent on inactive threads (not used)
+                for (int q = a; q < b; ++q) m[q] = lds_mred[q];
+            }
+            __syncthreads();  // lds_s reuse next chunk
+        }
+        a = b;
+    }
+    if (tid < D) {
+        for (int q = 0; q < nq; ++q) {
+            const float v = o[q] / l[q];
+            const int e = h * D + tid;
+            const int t = t0 + q;
+            out[(size_t)t * E + e] = v;
+            if (out_prep != nullptr) {
+                union { __half h; unsigned short u; } c;
+                c.h = __float2half(v);
+                out_prep[(((size_t)(e >> 3) * jtw + (t >> 4)) * 16 +
+                          (t & 15)) * 8 + (e & 7)] = c.u;
+            }
+        }
+    }
+}
+
 // --------------------------------------------------------------- GEMV(+res)
 // K9/K10/K13/K14 of SURVEY §2.5: output projection / FFN down / lm_head,
 // with the residual add fused into the epilogue.
@@ -1700,7 +1861,8 @@ static inline int xcd_grid(int GT, int MT) {
 
 // QKV projections for token group mtile + RoPE + KV append (the large-M
 // analog of k_qkv16; epilogue identical, tokens indexed globally).
-template <int WT, int JT = 4>
+// RT row tiles per wave share the B panel (its load + norm-build cost).
+template <int WT, int JT = 4, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
     WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
@@ -1710,23 +1872,27 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
     const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T,
     int MT, int jtw) {
     int gtile, mtile;
-    const int GT = 3 * (E >> 4);
+    const int tiles_per_mat = (E >> 4) / RT;
+    const int GT = 3 * tiles_per_mat;
     xcd_decode(GT, MT, gtile, mtile);
     if (gtile >= GT) return;
-    const int mat = gtile / (E >> 4);
-    const int tile = gtile % (E >> 4);
+    const int mat = gtile / tiles_per_mat;
+    const int tile = gtile % tiles_per_mat;
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][1][JT][4];
+    float acc[RT][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbe = (WT == W_F16) ? (E >> 5) : (((E >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 1, JT>(ws, tile, xprep, normprep, ss_in, eps,
-                                     acc, 0, nbe, jtw, mtile * JT);
-    __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    wave_tile_kloop<WT, true, 1, JT, RT>(ws, tile, xprep, normprep, ss_in,
+                                         eps, acc, 0, nbe, jtw,
+                                         mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * JT * RT];
+    combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = tile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+    const int r0 = (tile * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j2 = (mtile * JT + jt) * 16 + j;
@@ -1736,7 +1902,7 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
             __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
-                dst[jj] = __float2half(acc[0][0][jt][jj]);
+                dst[jj] = __float2half(acc[rt][0][jt][jj]);
             continue;
         }
 #pragma unroll
@@ -1746,8 +1912,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
             const float theta = (float)p * inv_freq[d >> 1];
             float sn, cs;
             __sincosf(theta, &sn, &cs);
-            const float x0 = acc[0][0][jt][2 * q2];
-            const float x1 = acc[0][0][jt][2 * q2 + 1];
+            const float x0 = acc[rt][0][jt][2 * q2];
+            const float x1 = acc[rt][0][jt][2 * q2 + 1];
             const float o0 = x0 * cs - x1 * sn;
             const float o1 = x0 * sn + x1 * cs;
             if (mat == 0) {
@@ -1761,31 +1927,35 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
             }
         }
     }
+    }
 }
 
 // Large-M GEMM with the fused residual + sumsq + xprep epilogue (wo / w2
 // consumers) or plain store. RES_SQ semantics match k_gemm16<GM_RES_SQ>.
-template <int WT, bool RES_SQ, int JT = 4>
+template <int WT, bool RES_SQ, int JT = 4, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_gemm16_mt(
     WMat2 w, const unsigned short* __restrict__ bprep,
     float* __restrict__ y, unsigned short* __restrict__ xprep_out,
     float* __restrict__ ss_out, int T, int MT, int jtw) {
     int gtile, mtile;
-    const int GT = w.rows >> 4;
+    const int GT = (w.rows >> 4) / RT;
     xcd_decode(GT, MT, gtile, mtile);
     if (gtile >= GT) return;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][1][JT][4];
+    float acc[RT][1][JT][4];
     const WMat2* ws[1] = {&w};
     const int nbt = w.cols >> 5;
     const int nbk = (WT == W_F16) ? nbt : ((nbt + 3) & ~3);
-    wave_tile_kloop<WT, false, 1, JT>(ws, gtile, bprep, nullptr, nullptr,
-                                      0.f, acc, 0, nbk, jtw, mtile * JT);
-    __shared__ float lds[3 * 64 * 4 * JT];
-    combine_acc<JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    wave_tile_kloop<WT, false, 1, JT, RT>(ws, gtile, bprep, nullptr,
+                                          nullptr, 0.f, acc, 0, nbk, jtw,
+                                          mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * JT * RT];
+    combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = gtile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+    const int r0 = (gtile * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j2 = (mtile * JT + jt) * 16 + j;
@@ -1793,18 +1963,18 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16_mt(
         float sq = 0.0f;
 #pragma unroll
         for (int jj = 0; jj < 4; ++jj) {
-            float v = acc[0][0][jt][jj];
+            float v = acc[rt][0][jt][jj];
             if (RES_SQ) {
                 v += y[(size_t)j2 * w.rows + r0 + jj];
                 sq += v * v;
             }
             y[(size_t)j2 * w.rows + r0 + jj] = v;
-            acc[0][0][jt][jj] = v;
+            acc[rt][0][jt][jj] = v;
         }
         if (RES_SQ && xprep_out != nullptr) {
             uint2 o;
-            o.x = pack_f16(acc[0][0][jt][0], acc[0][0][jt][1]);
-            o.y = pack_f16(acc[0][0][jt][2], acc[0][0][jt][3]);
+            o.x = pack_f16(acc[rt][0][jt][0], acc[rt][0][jt][1]);
+            o.y = pack_f16(acc[rt][0][jt][2], acc[rt][0][jt][3]);
             *reinterpret_cast<uint2*>(
                 xprep_out +
                 (((size_t)(r0 >> 3) * jtw + mtile * JT + jt) * 16 + j) * 8 +
@@ -1817,31 +1987,35 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16_mt(
             if (lane < 16) atomicAdd(ss_out + j2, s2);
         }
     }
+    }
 }
 
 // Large-M w1/w3 + fused RMSNorm + SwiGLU -> gprep (analog of k_ffn16).
-template <int WT, int JT = 4>
+template <int WT, int JT = 4, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_ffn16_mt(
     WMat2 w1, WMat2 w3, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
     const float* __restrict__ ss_in, float eps,
     unsigned short* __restrict__ gprep, int T, int MT, int jtw) {
     int gtile, mtile;
-    const int GT = w1.rows >> 4;
+    const int GT = (w1.rows >> 4) / RT;
     xcd_decode(GT, MT, gtile, mtile);
     if (gtile >= GT) return;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
-    float acc[1][2][JT][4];
+    float acc[RT][2][JT][4];
     const WMat2* ws[2] = {&w1, &w3};
     const int nbf = (WT == W_F16) ? (w1.cols >> 5)
                                   : (((w1.cols >> 5) + 3) & ~3);
-    wave_tile_kloop<WT, true, 2, JT>(ws, gtile, xprep, normprep, ss_in,
-                                     eps, acc, 0, nbf, jtw, mtile * JT);
-    __shared__ float lds[3 * 64 * 4 * 2 * JT];
-    combine_acc<2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
+    wave_tile_kloop<WT, true, 2, JT, RT>(ws, gtile, xprep, normprep, ss_in,
+                                         eps, acc, 0, nbf, jtw,
+                                         mtile * JT);
+    __shared__ float lds[3 * 64 * 4 * 2 * JT * RT];
+    combine_acc<RT * 2 * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
-    const int r0 = gtile * 16 + (lane >> 4) * 4;
+#pragma unroll
+    for (int rt = 0; rt < RT; ++rt) {
+    const int r0 = (gtile * RT + rt) * 16 + (lane >> 4) * 4;
 #pragma unroll
     for (int jt = 0; jt < JT; ++jt) {
         const int j2 = (mtile * JT + jt) * 16 + j;
@@ -1849,9 +2023,9 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16_mt(
         float g[4];
 #pragma unroll
         for (int jj = 0; jj < 4; ++jj) {
-            const float v1 = acc[0][0][jt][jj];
+            const float v1 = acc[rt][0][jt][jj];
             const float silu = v1 / (1.0f + __expf(-v1));
-            g[jj] = silu * acc[0][1][jt][jj];
+            g[jj] = silu * acc[rt][1][jt][jj];
         }
         uint2 o;
         o.x = pack_f16(g[0], g[1]);
@@ -1859,6 +2033,7 @@ __global__ __launch_bounds__(BLOCK) void k_ffn16_mt(
         *reinterpret_cast<uint2*>(
             gprep + (((size_t)(r0 >> 3) * jtw + mtile * JT + jt) * 16 + j)
                         * 8 + (r0 & 7)) = o;
+    }
     }
 }
 
@@ -2251,6 +2426,19 @@ void launch_embed(hipStream_t s, const WMat& tab, const int* tokens,
 
 // ------------------------------------------------- prefill-path launchers
 
+void launch_attn_prefill(hipStream_t s, const float* q_buf,
+                         const __half* k_cache_layer,
+                         const __half* v_cache_layer, float* out,
+                         unsigned short* out_prep, const int* pos,
+                         const int* seq, int T, int H, int E, int D,
+                         int n_ctx) {
+    const dim3 grid((T + 15) / 16, H);
+    const size_t lds = (size_t)(16 * D + 16 * BLOCK + 32) * sizeof(float);
+    hipLaunchKernelGGL(k_attn_prefill, grid, dim3(BLOCK), lds, s, q_buf,
+                       k_cache_layer, v_cache_layer, out, out_prep, pos,
+                       seq, E, D, n_ctx, jt_width(T), T);
+}
+
 void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
                      const WMat2& wv, const unsigned short* xprep,
                      const unsigned short* normprep, const float* ss_in,
@@ -2260,6 +2448,16 @@ void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
                      int T) {
     const int MT = (T + 63) >> 6;
     const int jtw = jt_width(T);
+    // RT=2 halves the B-panel reads + norm-build VALU per MFMA as long
+    // as the halved grid still fills the chip
+    if ((E >> 4) % 2 == 0 && (3 * (E >> 4) / 2) * MT >= 512) {
+        const dim3 grid(xcd_grid(3 * (E >> 4) / 2, MT));
+        DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
+            (k_qkv16_mt<WTc, 4, 2>), grid, dim3(BLOCK), 0, s, wq, wk, wv,
+            xprep, normprep, ss_in, eps, q_buf, k_cache_layer,
+            v_cache_layer, pos, seq, inv_freq, E, D, n_ctx, T, MT, jtw));
+        return;
+    }
     const dim3 grid(xcd_grid(3 * (E >> 4), MT));
     DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
         (k_qkv16_mt<WTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
@@ -2273,7 +2471,21 @@ void launch_gemm16_mt(hipStream_t s, const WMat2& w,
                       int res_sq) {
     const int MT = (T + 63) >> 6;
     const int jtw = jt_width(T);
-    const dim3 grid(xcd_grid(w.rows >> 4, MT));
+    const int R = w.rows >> 4;
+    if (R % 2 == 0 && (R / 2) * MT >= 512) {
+        const dim3 grid(xcd_grid(R / 2, MT));
+        if (res_sq) {
+            DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
+                (k_gemm16_mt<WTc, true, 4, 2>), grid, dim3(BLOCK), 0, s, w,
+                bprep, y, xprep_out, ss_out, T, MT, jtw));
+        } else {
+            DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
+                (k_gemm16_mt<WTc, false, 4, 2>), grid, dim3(BLOCK), 0, s,
+                w, bprep, y, xprep_out, ss_out, T, MT, jtw));
+        }
+        return;
+    }
+    const dim3 grid(xcd_grid(R, MT));
     if (res_sq) {
         DISPATCH_WT2(w.wtype, hipLaunchKernelGGL(
             (k_gemm16_mt<WTc, true>), grid, dim3(BLOCK), 0, s, w, bprep, y,
@@ -2291,7 +2503,15 @@ void launch_ffn16_mt(hipStream_t s, const WMat2& w1, const WMat2& w3,
                      float eps, unsigned short* gprep, int T) {
     const int MT = (T + 63) >> 6;
     const int jtw = jt_width(T);
-    const dim3 grid(xcd_grid(w1.rows >> 4, MT));
+    const int R = w1.rows >> 4;
+    if (R % 2 == 0 && (R / 2) * MT >= 512) {
+        const dim3 grid(xcd_grid(R / 2, MT));
+        DISPATCH_WT2(w1.wtype, hipLaunchKernelGGL(
+            (k_ffn16_mt<WTc, 4, 2>), grid, dim3(BLOCK), 0, s, w1, w3,
+            xprep, normprep, ss_in, eps, gprep, T, MT, jtw));
+        return;
+    }
+    const dim3 grid(xcd_grid(R, MT));
     DISPATCH_WT2(w1.wtype, hipLaunchKernelGGL(
         (k_ffn16_mt<WTc>), grid, dim3(BLOCK), 0, s, w1, w3, xprep, normprep,
         ss_in, eps, gprep, T, MT, jtw));
