@@ -557,6 +557,10 @@ __global__ void k_attn_prefill(
         lds_pos[tid] = pos[t0 + tid];
         lds_seq[tid] = seq[t0 + tid];
     }
+    // lds_mred carries each q's running max ACROSS chunks (the reducing
+    // wave's threads don't all run the PV stage, so the register copy
+    // m[q] is only authoritative on threads < D)
+    if (tid < QT) lds_mred[tid] = -INFINITY;
     for (int i = tid; i < nq * D; i += BLOCK) {
         const int q = i / D, d = i % D;
         lds_q[q * D + d] =
@@ -573,6 +577,10 @@ __global__ void k_attn_prefill(
         l[q] = 0.0f;
     }
 
+    // Every per-q register array below is indexed ONLY by the fully
+    // unrolled compile-time q (runtime indexing would spill o/m/l to
+    // scratch — the first version did exactly that and ran 1.7x SLOWER
+    // than the untiled kernel at 455 ms/span-1024-run).
     int a = 0;
     while (a < nq) {  // same-sequence segments of the tile
         const int sseq = lds_seq[a];
@@ -581,6 +589,8 @@ __global__ void k_attn_prefill(
         int jmax = 0;
         for (int q = a; q < b; ++q) jmax = max(jmax, lds_pos[q]);
         const int J = jmax + 1;
+        const int au = __builtin_amdgcn_readfirstlane(a);
+        const int bu = __builtin_amdgcn_readfirstlane(b);
         const size_t base = (size_t)sseq * n_ctx * E + h * D;
 
         for (int j0 = 0; j0 < J; j0 += BLOCK) {
@@ -599,7 +609,9 @@ __global__ void k_attn_prefill(
                         reinterpret_cast<const __half2*>(&raw);
                     const float2 ka = __half22float2(hh[0]);
                     const float2 kb = __half22float2(hh[1]);
-                    for (int q = a; q < b; ++q) {
+#pragma unroll
+                    for (int q = 0; q < QT; ++q) {
+                        if (q < au || q >= bu) continue;
                         const float* qr = lds_q + q * D + c * 4;
                         float s = fmaf(ka.x, qr[0], 0.0f);
                         s = fmaf(ka.y, qr[1], s);
@@ -608,16 +620,22 @@ __global__ void k_attn_prefill(
                         acc[q] += s;
                     }
                 }
-                for (int q = a; q < b; ++q)
+#pragma unroll
+                for (int q = 0; q < QT; ++q) {
+                    if (q < au || q >= bu) continue;
                     lds_s[q * BLOCK + tid] =
                         (jj <= lds_pos[q]) ? acc[q] : -INFINITY;
+                }
             } else {
-                for (int q = a; q < b; ++q)
+#pragma unroll
+                for (int q = 0; q < QT; ++q) {
+                    if (q < au || q >= bu) continue;
                     lds_s[q * BLOCK + tid] = -INFINITY;
+                }
             }
             __syncthreads();
             // per-q online-softmax bookkeeping: wave w owns q = w (mod 4)
-            for (int q = a + wid; q < b; q += NWAVES) {
+            for (int q = au + wid; q < bu; q += NWAVES) {
                 float* row = lds_s + q * BLOCK;
                 float wm = -INFINITY;
 #pragma unroll
@@ -625,7 +643,8 @@ __global__ void k_attn_prefill(
                     wm = fmaxf(wm, row[i * WAVE + lane]);
                 wm = wave_reduce_max(wm);
                 wm = __shfl(wm, 0);
-                const float m_new = fmaxf(m[q], wm);
+                const float m_old = lds_mred[q];
+                const float m_new = fmaxf(m_old, wm);
                 float ws = 0.0f;
 #pragma unroll
                 for (int i = 0; i < BLOCK / WAVE; ++i) {
@@ -644,31 +663,33 @@ __global__ void k_attn_prefill(
             if (tid < D) {
                 const int jlim = min(BLOCK, J - j0);
                 const __half* vcol = v_cache + base + (size_t)j0 * E + tid;
-                float alpha[QT];
-                for (int q = a; q < b; ++q) {
+#pragma unroll
+                for (int q = 0; q < QT; ++q) {
+                    if (q < au || q >= bu) continue;
                     const float m_new = lds_mred[q];
-                    alpha[q] = (m[q] == -INFINITY)
-                                   ? 0.0f
-                                   : __expf(m[q] - m_new);
-                    o[q] *= alpha[q];
-                    l[q] = l[q] * alpha[q] + lds_pred[q];
+                    const float alpha =
+                        (m[q] == -INFINITY) ? 0.0f : __expf(m[q] - m_new);
+                    o[q] *= alpha;
+                    l[q] = l[q] * alpha + lds_pred[q];
                     m[q] = m_new;
                 }
                 for (int jc = 0; jc < jlim; ++jc) {
                     const float v = __half2float(vcol[(size_t)jc * E]);
-                    for (int q = a; q < b; ++q)
+#pragma unroll
+                    for (int q = 0; q < QT; ++q) {
+                        if (q < au || q >= bu) continue;
                         o[q] = fmaf(lds_s[q * BLOCK + jc], v, o[q]);
+                    }
                 }
-            } else {
-                // keep m/l consistent on inactive threads (not used)
-                for (int q = a; q < b; ++q) m[q] = lds_mred[q];
             }
             __syncthreads();  // lds_s reuse next chunk
         }
         a = b;
     }
     if (tid < D) {
-        for (int q = 0; q < nq; ++q) {
+#pragma unroll
+        for (int q = 0; q < QT; ++q) {
+            if (q >= nq) break;
             const float v = o[q] / l[q];
             const int e = h * D + tid;
             const int t = t0 + q;
