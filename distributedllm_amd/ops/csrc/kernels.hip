@@ -39,6 +39,34 @@
 #define NWAVES 4
 #define BLOCK 256
 
+// ----------------------------------------------------- MFMA fragment types
+
+typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+union ABFrag {
+    uint32_t u[4];
+    f16x8 v;
+};
+
+// packed-f16 helpers for the MFMA side-channels / dequant
+__device__ __forceinline__ uint32_t pack_f16(float lo, float hi) {
+    union { __half2 h; uint32_t u; } c;
+    c.h = __floats2half2_rn(lo, hi);
+    return c.u;
+}
+__device__ __forceinline__ __half2 u2h2(uint32_t w) {
+    union { uint32_t u; __half2 h; } c;
+    c.u = w;
+    return c.h;
+}
+__device__ __forceinline__ uint32_t h22u(__half2 h) {
+    union { __half2 h; uint32_t u; } c;
+    c.h = h;
+    return c.u;
+}
+
 // ---------------------------------------------------------------- reductions
 
 __device__ __forceinline__ float wave_reduce_sum(float v) {
@@ -704,6 +732,234 @@ __global__ void k_attn_prefill(
     }
 }
 
+// --------------------------------------- prefill flash attention (MFMA)
+// Matrix-core flash attention for the large-M prefill path (D <= 128):
+// one block = one (16-query tile, head); each of the 4 waves streams a
+// strided subset of the 16-row KV tiles FULLY IN REGISTERS — scores via
+// mfma_f32_16x16x32_f16 (A = K tile, B = the staged Q tile, C rows = j,
+// cols = q), online softmax on the 4 score rows each lane holds, then
+// PV via mfma_f32_16x16x16_f16 where the score C-layout (lane: q =
+// l&15, j = (l>>4)*4+jj) IS the B-operand layout (cols q, k-dim j) —
+// no transpose, no LDS, no syncs in the inner loop. Partial (m, l, O)
+// merge across waves once at the end. The VALU fallback k_attn_prefill
+// (below) covers D > 128 and measured ~25x off issue-roofline — this
+// kernel exists because 16x16 tiles of dot products belong on MFMA.
+typedef __attribute__((ext_vector_type(4))) _Float16 f16x4;
+
+union BFrag4 {
+    uint32_t u[2];
+    f16x4 v;
+};
+
+// 8 halves [d0, d0+8) of `row`, zero-padded past D
+__device__ __forceinline__ f16x8 load_kfrag8(const __half* __restrict__ row,
+                                             int d0, int D) {
+    ABFrag a;
+    if (d0 + 8 <= D) {
+        const uint4 u = *reinterpret_cast<const uint4*>(row + d0);
+        a.u[0] = u.x; a.u[1] = u.y; a.u[2] = u.z; a.u[3] = u.w;
+    } else {
+        __half tmp[8];
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            tmp[e] = (d0 + e < D) ? row[d0 + e] : __half(0.0f);
+        const uint4 u = *reinterpret_cast<const uint4*>(tmp);
+        a.u[0] = u.x; a.u[1] = u.y; a.u[2] = u.z; a.u[3] = u.w;
+    }
+    return a.v;
+}
+
+__global__ __launch_bounds__(BLOCK) void k_attn_prefill_mfma(
+    const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
+    const __half* __restrict__ v_cache, float* __restrict__ out,
+    unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
+    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw, int T) {
+    constexpr int QT = 16;
+    constexpr int DCMAX = 8;            // D <= 128 = 8 chunks of 16
+    const int t0 = blockIdx.x * QT;
+    const int h = blockIdx.y;
+    const int nq = min(QT, T - t0);
+    const int tid = threadIdx.x;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
+    const int lq = lane & 15;           // this lane's query column
+    const int lg = lane >> 4;           // lane group (j/k sub-span)
+    const int ndc = (D + 15) >> 4;      // 16-wide d chunks (PV / output)
+    const float inv_sqrt_d = rsqrtf((float)D);
+
+    __shared__ int lds_pos[QT];
+    __shared__ int lds_seq[QT];
+    // per-wave partials for the final merge
+    __shared__ float lds_m[NWAVES][QT];
+    __shared__ float lds_l[NWAVES][QT];
+    __shared__ float lds_o[NWAVES][QT][DCMAX * 16];
+
+    if (tid < QT) {
+        lds_pos[tid] = (t0 + tid < T) ? pos[t0 + tid] : 0;
+        lds_seq[tid] = (t0 + tid < T) ? seq[t0 + tid] : -1;
+    }
+    __syncthreads();
+
+    // stage Q as score-MFMA B fragments: lane (col q=lq, k-span lg*8
+    // within each 32-d chunk), prescaled by 1/sqrt(D), zero past D/nq
+    ABFrag qf[4];
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+        __half tmp[8];
+        const int d0 = c * 32 + lg * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+            float v = 0.0f;
+            if (t0 + lq < T && d0 + e < D)
+                v = q_buf[(size_t)(t0 + lq) * E + h * D + d0 + e] *
+                    inv_sqrt_d;
+            tmp[e] = __float2half(v);
+        }
+        const uint4 u = *reinterpret_cast<const uint4*>(tmp);
+        qf[c].u[0] = u.x; qf[c].u[1] = u.y;
+        qf[c].u[2] = u.z; qf[c].u[3] = u.w;
+    }
+
+    float m = -INFINITY, l = 0.0f;      // running state for column lq
+    f32x4 oacc[DCMAX];
+#pragma unroll
+    for (int dc = 0; dc < DCMAX; ++dc)
+        oacc[dc] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+    const int pos_q = lds_pos[min(lq, nq - 1)];
+    int a = 0;
+    while (a < nq) {  // same-sequence segments of the query tile
+        const int sseq = lds_seq[a];
+        int b = a + 1;
+        while (b < nq && lds_seq[b] == sseq) ++b;
+        int jmax = 0;
+        for (int q = a; q < b; ++q) jmax = max(jmax, lds_pos[q]);
+        const int J = jmax + 1;
+        const bool active = (lq >= a && lq < b);
+        const size_t base = (size_t)sseq * n_ctx * E + h * D;
+
+        // wave w streams tiles w, w+4, w+8, ... of this segment
+        for (int tj = wid * 16; tj < J; tj += NWAVES * 16) {
+            // scores: A = K rows (lane: row j=lq of the tile, k-span lg)
+            const bool jrow_ok = (tj + lq) < J;  // tile tail: no OOB read
+            const __half* krow =
+                k_cache + base + (size_t)(jrow_ok ? tj + lq : 0) * E;
+            f32x4 sc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+            for (int c = 0; c < 4; ++c) {
+                if (c * 32 >= D) break;
+                f16x8 kf = load_kfrag8(krow, c * 32 + lg * 8, D);
+                if (!jrow_ok) kf = f16x8{0, 0, 0, 0, 0, 0, 0, 0};
+                sc = __builtin_amdgcn_mfma_f32_16x16x32_f16(kf, qf[c].v,
+                                                            sc, 0, 0, 0);
+            }
+            // lane now holds s[j = tj + lg*4 + jj][q = lq]; mask + max
+            float s[4], tm = -INFINITY;
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                const int j = tj + lg * 4 + jj;
+                s[jj] = (active && j < J && j <= pos_q) ? sc[jj]
+                                                        : -INFINITY;
+                tm = fmaxf(tm, s[jj]);
+            }
+            tm = fmaxf(tm, __shfl_xor(tm, 16));
+            tm = fmaxf(tm, __shfl_xor(tm, 32));
+            // wave-uniform skip only (divergent MFMA is illegal); lanes
+            // whose q saw nothing yet keep (m=-inf, l=0, o=0) via
+            // dead-lane handling below — never compute -inf - -inf
+            if (!__any(tm != -INFINITY)) continue;
+            const float m_new = fmaxf(m, tm);
+            const bool dead = (m_new == -INFINITY);
+            const float alpha =
+                (dead || m == -INFINITY) ? (dead ? 1.0f : 0.0f)
+                                         : __expf(m - m_new);
+            BFrag4 p;
+            float ts = 0.0f;
+#pragma unroll
+            for (int jj = 0; jj < 4; ++jj) {
+                const float pv = dead ? 0.0f : __expf(s[jj] - m_new);
+                ts += pv;
+                reinterpret_cast<__half*>(p.u)[jj] = __float2half(pv);
+            }
+            ts += __shfl_xor(ts, 16);
+            ts += __shfl_xor(ts, 32);
+            l = l * alpha + ts;
+            m = m_new;
+#pragma unroll
+            for (int dc = 0; dc < DCMAX; ++dc) {
+                if (dc >= ndc) break;
+                oacc[dc][0] *= alpha; oacc[dc][1] *= alpha;
+                oacc[dc][2] *= alpha; oacc[dc][3] *= alpha;
+            }
+            // PV: A = V^T (lane: row d = dc*16+lq, k-span j = lg*4+e);
+            // B = p (exactly the score C layout). Masked j rows carry
+            // p = 0; KV memory is zero-initialized so V rows past J
+            // are finite.
+#pragma unroll
+            for (int dc = 0; dc < DCMAX; ++dc) {
+                if (dc >= ndc) break;
+                const int d = dc * 16 + lq;
+                BFrag4 vt;
+#pragma unroll
+                for (int e = 0; e < 4; ++e) {
+                    const int j = tj + lg * 4 + e;
+                    __half v(0.0f);
+                    if (d < D && j < J)
+                        v = v_cache[base + (size_t)j * E + d];
+                    reinterpret_cast<__half*>(vt.u)[e] = v;
+                }
+                oacc[dc] = __builtin_amdgcn_mfma_f32_16x16x16f16(
+                    vt.v, p.v, oacc[dc], 0, 0, 0);
+            }
+        }
+        a = b;
+    }
+
+    // merge the 4 waves' partials (each lane group holds redundant
+    // copies of (m, l) for its q; group 0 writes)
+    if (lg == 0) {
+        lds_m[wid][lq] = m;
+        lds_l[wid][lq] = l;
+    }
+#pragma unroll
+    for (int dc = 0; dc < DCMAX; ++dc) {
+        if (dc >= ndc) break;
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj)
+            lds_o[wid][lq][dc * 16 + lg * 4 + jj] = oacc[dc][jj];
+    }
+    __syncthreads();
+    // thread (q, d) recombines: 256 threads cover 16 q x 16 d per pass
+    const int qq = tid >> 4;
+    for (int d = tid & 15; d < D; d += 16) {
+        if (qq >= nq) break;
+        float mstar = -INFINITY;
+#pragma unroll
+        for (int w = 0; w < NWAVES; ++w)
+            mstar = fmaxf(mstar, lds_m[w][qq]);
+        if (mstar == -INFINITY) mstar = 0.0f;  // no unmasked rows
+        float lstar = 0.0f, ostar = 0.0f;
+#pragma unroll
+        for (int w = 0; w < NWAVES; ++w) {
+            const float mw = lds_m[w][qq];
+            const float sc = (mw == -INFINITY) ? 0.0f
+                                               : __expf(mw - mstar);
+            lstar += lds_l[w][qq] * sc;
+            ostar += lds_o[w][qq][d] * sc;
+        }
+        const float v = ostar / fmaxf(lstar, 1e-20f);
+        const int e = h * D + d;
+        const int t = t0 + qq;
+        out[(size_t)t * E + e] = v;
+        if (out_prep != nullptr) {
+            union { __half h; unsigned short u; } cvt;
+            cvt.h = __float2half(v);
+            out_prep[(((size_t)(e >> 3) * jtw + (t >> 4)) * 16 +
+                      (t & 15)) * 8 + (e & 7)] = cvt.u;
+        }
+    }
+}
+
 // --------------------------------------------------------------- GEMV(+res)
 // K9/K10/K13/K14 of SURVEY §2.5: output projection / FFN down / lm_head,
 // with the residual add fused into the epilogue.
@@ -882,31 +1138,8 @@ __global__ void k_argmax_finish(const unsigned long long* __restrict__ keys,
 // tile; partial accumulators combine through LDS; wave 0 runs the fused
 // epilogue (residual add + sumsq atomics + xprep/rope/cache/silu writes).
 
-typedef __attribute__((ext_vector_type(8))) _Float16 f16x8;
-typedef __attribute__((ext_vector_type(4))) float f32x4;
-typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
-
-union ABFrag {
-    uint32_t u[4];
-    f16x8 v;
-};
-
-// packed-f16 helpers for the MFMA side-channels / dequant
-__device__ __forceinline__ uint32_t pack_f16(float lo, float hi) {
-    union { __half2 h; uint32_t u; } c;
-    c.h = __floats2half2_rn(lo, hi);
-    return c.u;
-}
-__device__ __forceinline__ __half2 u2h2(uint32_t w) {
-    union { uint32_t u; __half2 h; } c;
-    c.u = w;
-    return c.h;
-}
-__device__ __forceinline__ uint32_t h22u(__half2 h) {
-    union { __half2 h; uint32_t u; } c;
-    c.h = h;
-    return c.u;
-}
+// (f16x8 / f32x4 / u32x4 / ABFrag / pack_f16 helpers are defined near
+// the top of the file — shared with the prefill attention kernels)
 
 // write-through 16 B store (sc0 sc1): the split-K slab publish leaves no
 // dirty lines in the writer XCD's L2, so the consumer kernel's reads are
@@ -2454,6 +2687,13 @@ void launch_attn_prefill(hipStream_t s, const float* q_buf,
                          const int* seq, int T, int H, int E, int D,
                          int n_ctx) {
     const dim3 grid((T + 15) / 16, H);
+    if (D <= 128) {  // every LLaMA head dim; matrix-core path
+        hipLaunchKernelGGL(k_attn_prefill_mfma, grid, dim3(BLOCK), 0, s,
+                           q_buf, k_cache_layer, v_cache_layer, out,
+                           out_prep, pos, seq, E, D, n_ctx, jt_width(T),
+                           T);
+        return;
+    }
     const size_t lds = (size_t)(16 * D + 16 * BLOCK + 32) * sizeof(float);
     hipLaunchKernelGGL(k_attn_prefill, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos,
