@@ -282,7 +282,8 @@ class HIPSliceEngine:
         self._eng = core.SliceEngine(
             n_embd=hp.n_embd, n_head=hp.n_head, n_layers=n_layers,
             n_ff=hp.n_ff, n_ctx=n_ctx, max_batch=max_batch,
-            eps=RMS_EPS, rope_base=ROPE_BASE, max_prefill=max_prefill)
+            eps=RMS_EPS, rope_base=ROPE_BASE, max_prefill=max_prefill,
+            n_head_kv=hp.kv_heads if hp.is_gqa else 0)
         self.device = "cuda"
         self.has_extra = False
         # weight-tensor references for clone_shared (filled by .random()
@@ -375,7 +376,9 @@ class HIPSliceEngine:
             return (1.0 + torch.randn(n, device="cuda", generator=g) *
                     0.01).contiguous()
 
-        shapes = [(E, E), (E, E), (E, E), (E, E), (F, E), (E, F), (F, E)]
+        Ekv = hp.n_embd_kv  # K/V projection rows (GQA: Hkv*D < E)
+        shapes = [(E, E), (Ekv, E), (Ekv, E), (E, E), (F, E), (E, F),
+                  (F, E)]
         eng._layers_cache = []
         for li in range(n_layers):
             mats = [mat(r, c) for r, c in shapes]
@@ -533,9 +536,10 @@ class TorchSliceEngine:
         self.n_ctx = n_ctx
         self.max_batch = max_batch
         self.device = device
-        h, d = hp.n_head, hp.head_dim
-        self.k_cache = torch.zeros(n_layers, max_batch, n_ctx, h, d)
-        self.v_cache = torch.zeros(n_layers, max_batch, n_ctx, h, d)
+        d = hp.head_dim
+        hkv = hp.kv_heads
+        self.k_cache = torch.zeros(n_layers, max_batch, n_ctx, hkv, d)
+        self.v_cache = torch.zeros(n_layers, max_batch, n_ctx, hkv, d)
         self.has_extra = all(
             k in weights for k in
             ("tok_embeddings.weight", "norm.weight", "output.weight"))
@@ -559,6 +563,8 @@ class TorchSliceEngine:
         hp = self.hp
         T, E = x.shape
         h, d = hp.n_head, hp.head_dim
+        hkv = hp.kv_heads
+        kv_map = torch.arange(h) // (h // hkv)  # GQA head sharing
         pos_l = pos.tolist()
         seq_l = seq.tolist()
         for li in range(self.n_layers):
@@ -572,11 +578,11 @@ class TorchSliceEngine:
             for t in range(T):  # per-row positions (prefill/decode unified)
                 p, b = pos_l[t], seq_l[t]
                 qt = rope_interleaved(q[t:t + 1].view(1, h, d), p)[0]
-                kt = rope_interleaved(k[t:t + 1].view(1, h, d), p)[0]
+                kt = rope_interleaved(k[t:t + 1].view(1, hkv, d), p)[0]
                 self.k_cache[li, b, p] = kt
-                self.v_cache[li, b, p] = v[t].view(h, d)
-                keys = self.k_cache[li, b, :p + 1]
-                vals = self.v_cache[li, b, :p + 1]
+                self.v_cache[li, b, p] = v[t].view(hkv, d)
+                keys = self.k_cache[li, b, :p + 1][:, kv_map]
+                vals = self.v_cache[li, b, :p + 1][:, kv_map]
                 att = torch.einsum("hd,jhd->hj", qt, keys) / math.sqrt(d)
                 prob = torch.softmax(att, dim=-1)
                 o[t] = torch.einsum("hj,jhd->hd", prob, vals).reshape(E)

@@ -33,6 +33,12 @@ from . import q4
 
 GGJT_MAGIC = 0x67676A74  # bytes 'tjgg' little-endian == "ggjt"
 GGJT_VERSION = 3
+# This framework's GQA extension: the GGJT v3 header cannot express
+# n_head_kv (llama.cpp of the reference's era passed it on the command
+# line, `-gqa 8`). Version 4 inserts ``n_head_kv:u32`` directly after
+# ``n_head``; it is written ONLY when n_head_kv != n_head, so every MHA
+# file stays byte-identical to the reference formats.
+GGJT_VERSION_GQA = 4
 
 EXTRA_LAYERS_FIRST_LAYER = 0xFFFFFFFF
 
@@ -117,6 +123,7 @@ class Hparams:
     n_rot: int
     ftype: int
     first_layer: Optional[int] = None  # present only in the extended format
+    n_head_kv: Optional[int] = None    # GQA (v4 header); None == n_head
 
     @property
     def n_ff(self) -> int:
@@ -127,6 +134,19 @@ class Hparams:
     @property
     def head_dim(self) -> int:
         return self.n_embd // self.n_head
+
+    @property
+    def kv_heads(self) -> int:
+        return self.n_head if self.n_head_kv is None else self.n_head_kv
+
+    @property
+    def n_embd_kv(self) -> int:
+        """Width of the K/V projections (= n_embd for MHA)."""
+        return self.kv_heads * self.head_dim
+
+    @property
+    def is_gqa(self) -> bool:
+        return self.kv_heads != self.n_head
 
 
 @dataclass
@@ -207,9 +227,13 @@ class GGMLFile:
 
     def _write(self, f: BinaryIO) -> None:
         hp = self.hparams
-        f.write(struct.pack("<II", GGJT_MAGIC, GGJT_VERSION))
-        fields = [hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head, hp.n_layer,
-                  hp.n_rot]
+        gqa = hp.is_gqa
+        f.write(struct.pack("<II", GGJT_MAGIC,
+                            GGJT_VERSION_GQA if gqa else GGJT_VERSION))
+        fields = [hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head]
+        if gqa:
+            fields.append(hp.kv_heads)
+        fields += [hp.n_layer, hp.n_rot]
         if hp.first_layer is not None:
             fields.append(hp.first_layer)
         fields.append(hp.ftype)
@@ -260,19 +284,21 @@ class GGMLFile:
         if magic != GGJT_MAGIC:
             raise ValueError(f"bad magic 0x{magic:08x}; not a GGJT file")
         version = u32()
-        if version != GGJT_VERSION:
+        if version not in (GGJT_VERSION, GGJT_VERSION_GQA):
             raise ValueError(f"unsupported GGJT version {version}")
         n_vocab = u32()
         n_embd = u32()
         n_mult = u32()
         n_head = u32()
+        n_head_kv = u32() if version == GGJT_VERSION_GQA else None
         n_layer = u32()
         n_rot = u32()
         first_layer = u32() if extended else None
         ftype = u32()
         hp = Hparams(n_vocab=n_vocab, n_embd=n_embd, n_mult=n_mult,
                      n_head=n_head, n_layer=n_layer, n_rot=n_rot,
-                     ftype=ftype, first_layer=first_layer)
+                     ftype=ftype, first_layer=first_layer,
+                     n_head_kv=n_head_kv)
 
         vocab: List[Tuple[bytes, float]] = []
         for _ in range(n_vocab):
@@ -316,21 +342,22 @@ def sniff_extended(path: str) -> bool:
     when unambiguous.
     """
     with open(path, "rb") as f:
-        head = f.read(4 * 10)
-    if len(head) < 4 * 10:
+        head = f.read(4 * 11)
+    if len(head) < 4 * 11:
         raise ValueError(f"truncated GGJT file {path!r} "
                          f"({len(head)} bytes)")
-    vals = struct.unpack_from("<10I", head, 0)
+    vals = struct.unpack_from("<11I", head, 0)
     if vals[0] != GGJT_MAGIC:
         raise ValueError("not a GGJT file")
-    # vals[2:8] = 7-field hparams; vals[8] would be first tensor field.
-    f7_ftype = vals[8 - 1]    # 7-field: ftype at index 8-1=7? careful below
-    # indices: 0 magic, 1 version, 2 n_vocab, 3 n_embd, 4 n_mult, 5 n_head,
-    # 6 n_layer, 7 n_rot, then [8]=ftype (7-field) or first_layer (8-field),
-    # [9]=ftype (8-field).
-    seven_ok = vals[8] in _FTYPE_TO_GGML
-    eight_ok = vals[9] in _FTYPE_TO_GGML and (
-        vals[8] == EXTRA_LAYERS_FIRST_LAYER or vals[8] < 4096)
+    # v4 (GQA) shifts every index after n_head by one
+    shift = 1 if vals[1] == GGJT_VERSION_GQA else 0
+    # indices (v3): 0 magic, 1 version, 2 n_vocab, 3 n_embd, 4 n_mult,
+    # 5 n_head, 6 n_layer, 7 n_rot, then [8]=ftype (7-field) or
+    # first_layer (8-field), [9]=ftype (8-field).
+    seven_ok = vals[8 + shift] in _FTYPE_TO_GGML
+    eight_ok = vals[9 + shift] in _FTYPE_TO_GGML and (
+        vals[8 + shift] == EXTRA_LAYERS_FIRST_LAYER or
+        vals[8 + shift] < 4096)
     if eight_ok and not seven_ok:
         return True
     if seven_ok and not eight_ok:

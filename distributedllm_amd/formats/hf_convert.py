@@ -117,9 +117,12 @@ def convert_hf_dir(hf_dir: str) -> ggml.GGMLFile:
     L = cfg["num_hidden_layers"]
     V = cfg["vocab_size"]
     F = cfg["intermediate_size"]
+    # GQA (llama-2 70B class): k/v projections carry num_key_value_heads
+    HKV = cfg.get("num_key_value_heads", H)
     hp = ggml.Hparams(n_vocab=V, n_embd=E, n_mult=find_n_mult(E, F),
                       n_head=H, n_layer=L, n_rot=E // H,
-                      ftype=ggml.FTYPE_MOSTLY_F16)
+                      ftype=ggml.FTYPE_MOSTLY_F16,
+                      n_head_kv=HKV if HKV != H else None)
 
     sd = _load_state_dict(hf_dir)
     tensors: List[ggml.GGMLTensor] = []
@@ -138,9 +141,10 @@ def convert_hf_dir(hf_dir: str) -> ggml.GGMLFile:
             if k not in sd:
                 raise KeyError(f"missing tensor {k!r} in {hf_dir}")
             w = sd[k]
-            if hf_suffix in ("self_attn.q_proj.weight",
-                             "self_attn.k_proj.weight"):
+            if hf_suffix == "self_attn.q_proj.weight":
                 w = permute_rotary(w, H)
+            elif hf_suffix == "self_attn.k_proj.weight":
+                w = permute_rotary(w, HKV)  # Hkv rows under GQA
             emit(f"layers.{i}.{g_suffix}", w)
 
     return ggml.GGMLFile(hparams=hp, vocab=load_hf_vocab(hf_dir, V),

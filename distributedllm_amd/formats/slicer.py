@@ -38,7 +38,8 @@ def make_slice(src: ggml.GGMLFile, idx_from: int, idx_to: int) -> ggml.GGMLFile:
     new_hp = ggml.Hparams(n_vocab=hp.n_vocab, n_embd=hp.n_embd,
                           n_mult=hp.n_mult, n_head=hp.n_head,
                           n_layer=idx_to - idx_from + 1, n_rot=hp.n_rot,
-                          ftype=hp.ftype, first_layer=idx_from)
+                          ftype=hp.ftype, first_layer=idx_from,
+                          n_head_kv=hp.n_head_kv)
     tensors = [t for t in src.tensors
                if (li := layer_index(t.name)) is not None
                and idx_from <= li <= idx_to]
@@ -51,7 +52,8 @@ def make_extra_layers(src: ggml.GGMLFile) -> ggml.GGMLFile:
     new_hp = ggml.Hparams(n_vocab=hp.n_vocab, n_embd=hp.n_embd,
                           n_mult=hp.n_mult, n_head=hp.n_head, n_layer=0,
                           n_rot=hp.n_rot, ftype=hp.ftype,
-                          first_layer=ggml.EXTRA_LAYERS_FIRST_LAYER)
+                          first_layer=ggml.EXTRA_LAYERS_FIRST_LAYER,
+                          n_head_kv=hp.n_head_kv)
     tensors = [t for t in src.tensors
                if t.name.startswith(("norm", "output", "tok_embeddings"))]
     return ggml.GGMLFile(hparams=new_hp, vocab=list(src.vocab),

@@ -55,10 +55,12 @@ def build_model(preset: str | LlamaPreset, ftype: int = ggml.FTYPE_MOSTLY_Q4_0,
     p = PRESETS[preset] if isinstance(preset, str) else preset
     L = p.n_layer if n_layer is None else n_layer
     hp = ggml.Hparams(n_vocab=p.n_vocab, n_embd=p.n_embd, n_mult=p.n_mult,
-                      n_head=p.n_head, n_layer=L, n_rot=p.n_rot, ftype=ftype)
+                      n_head=p.n_head, n_layer=L, n_rot=p.n_rot, ftype=ftype,
+                      n_head_kv=p.n_head_kv)
     wt = ggml._FTYPE_TO_GGML[ftype]
     rng = np.random.default_rng(seed)
     E, F, V = p.n_embd, p.n_ff, p.n_vocab
+    Ekv = p.n_embd_kv  # K/V projection width (= E for MHA)
 
     def mat(rows: int, cols: int) -> np.ndarray:
         return rng.standard_normal((rows, cols), dtype=np.float32) * scale
@@ -77,8 +79,10 @@ def build_model(preset: str | LlamaPreset, ftype: int = ggml.FTYPE_MOSTLY_Q4_0,
             ggml.GGMLTensor.from_f32(pre + "attention_norm.weight",
                                      norm_w(E), ggml.GGML_TYPE_F32),
             ggml.GGMLTensor.from_f32(pre + "attention.wq.weight", mat(E, E), wt),
-            ggml.GGMLTensor.from_f32(pre + "attention.wk.weight", mat(E, E), wt),
-            ggml.GGMLTensor.from_f32(pre + "attention.wv.weight", mat(E, E), wt),
+            ggml.GGMLTensor.from_f32(pre + "attention.wk.weight",
+                                     mat(Ekv, E), wt),
+            ggml.GGMLTensor.from_f32(pre + "attention.wv.weight",
+                                     mat(Ekv, E), wt),
             ggml.GGMLTensor.from_f32(pre + "attention.wo.weight", mat(E, E), wt),
             ggml.GGMLTensor.from_f32(pre + "ffn_norm.weight",
                                      norm_w(E), ggml.GGML_TYPE_F32),

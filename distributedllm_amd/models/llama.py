@@ -44,10 +44,19 @@ class LlamaPreset:
     n_mult: int
     n_head: int
     n_layer: int
+    n_head_kv: Optional[int] = None  # GQA (llama-v2 70B class); None = MHA
 
     @property
     def n_rot(self) -> int:
         return self.n_embd // self.n_head
+
+    @property
+    def kv_heads(self) -> int:
+        return self.n_head if self.n_head_kv is None else self.n_head_kv
+
+    @property
+    def n_embd_kv(self) -> int:
+        return self.kv_heads * (self.n_embd // self.n_head)
 
     @property
     def n_ff(self) -> int:
@@ -59,19 +68,27 @@ class LlamaPreset:
         return ggml.Hparams(n_vocab=self.n_vocab, n_embd=self.n_embd,
                             n_mult=self.n_mult, n_head=self.n_head,
                             n_layer=self.n_layer, n_rot=self.n_rot,
-                            ftype=ftype, first_layer=first_layer)
+                            ftype=ftype, first_layer=first_layer,
+                            n_head_kv=self.n_head_kv)
 
 
-# n_ff sanity: 3B=8640, 7B=11008, 13B=13824, 30B=17920, 65B=22016
+# n_ff sanity: 3B=8640, 7B=11008, 13B=13824, 30B=17920, 65B=22016;
+# llama2_70b: n_mult 28672 makes the era formula land on 28672 exactly
 PRESETS: Dict[str, LlamaPreset] = {
     "open_llama_3b": LlamaPreset("open_llama_3b", 32000, 3200, 216, 32, 26),
     "llama_7b": LlamaPreset("llama_7b", 32000, 4096, 256, 32, 32),
     "llama_13b": LlamaPreset("llama_13b", 32000, 5120, 256, 40, 40),
     "llama_30b": LlamaPreset("llama_30b", 32000, 6656, 256, 52, 60),
     "llama_65b": LlamaPreset("llama_65b", 32000, 8192, 256, 64, 80),
+    # llama-v2 family (7b/13b are MHA with the same dims as v1)
+    "llama2_70b": LlamaPreset("llama2_70b", 32000, 8192, 28672, 64, 80,
+                              n_head_kv=8),
     # tiny configs for tests (head_dim 8 resp. 100-like non-pow2 = 20)
     "tiny": LlamaPreset("tiny", 256, 64, 32, 4, 3),
     "tiny_oddhead": LlamaPreset("tiny_oddhead", 256, 96, 32, 4, 2),
+    # GQA test config: 8 query heads sharing 2 kv heads (group 4)
+    "tiny_gqa": LlamaPreset("tiny_gqa", 256, 128, 32, 8, 3, n_head_kv=2),
+    "small_gqa": LlamaPreset("small_gqa", 512, 512, 64, 8, 2, n_head_kv=2),
     # mid-size test config: large enough that the split-K/RT kernel paths
     # run with realistic grids (E/16=32 tiles, F=1408)
     "small": LlamaPreset("small", 512, 512, 64, 8, 2),
@@ -132,9 +149,10 @@ class LlamaSliceRef:
         self.n_layers = n_layers
         self.n_ctx = n_ctx
         self.n_past = 0
-        h, d = hp.n_head, hp.head_dim
-        self.k_cache = torch.zeros(n_layers, n_ctx, h, d)
-        self.v_cache = torch.zeros(n_layers, n_ctx, h, d)
+        d = hp.head_dim
+        hkv = hp.kv_heads
+        self.k_cache = torch.zeros(n_layers, n_ctx, hkv, d)
+        self.v_cache = torch.zeros(n_layers, n_ctx, hkv, d)
 
     def clear_context(self) -> None:
         self.n_past = 0
@@ -155,13 +173,16 @@ class LlamaSliceRef:
             q = a @ self.w[pre + "attention.wq.weight"].T
             k = a @ self.w[pre + "attention.wk.weight"].T
             v = a @ self.w[pre + "attention.wv.weight"].T
+            hkv = hp.kv_heads
             q = rope_interleaved(q.view(n, h, d), p)
-            k = rope_interleaved(k.view(n, h, d), p)
-            v = v.view(n, h, d)
+            k = rope_interleaved(k.view(n, hkv, d), p)
+            v = v.view(n, hkv, d)
             self.k_cache[li, p:p + n] = k
             self.v_cache[li, p:p + n] = v
-            keys = self.k_cache[li, :p + n]      # [P+N, H, D]
-            vals = self.v_cache[li, :p + n]
+            # GQA: q head i attends kv head i // (H/Hkv)
+            kv_map = torch.arange(h) // (h // hkv)
+            keys = self.k_cache[li, :p + n][:, kv_map]   # [P+N, H, D]
+            vals = self.v_cache[li, :p + n][:, kv_map]
             # scores[t, j] over j<=p+t
             att = torch.einsum("nhd,jhd->hnj", q, keys) / math.sqrt(d)
             mask = torch.arange(p + n)[None, :] > (

@@ -154,7 +154,8 @@ class SliceEngine {
  public:
     SliceEngine(int64_t n_embd, int64_t n_head, int64_t n_layers,
                 int64_t n_ff, int64_t n_ctx, int64_t max_batch, double eps,
-                double rope_base, int64_t max_prefill = 1024)
+                double rope_base, int64_t max_prefill = 1024,
+                int64_t n_head_kv = 0)
         : E_((int)n_embd),
           H_((int)n_head),
           D_((int)(n_embd / n_head)),
@@ -167,6 +168,11 @@ class SliceEngine {
         TORCH_CHECK(D_ % 2 == 0, "head_dim must be even for RoPE pairs");
         TORCH_CHECK(D_ <= 256, "head_dim > 256 unsupported");
         TORCH_CHECK(E_ % 16 == 0 && F_ % 16 == 0, "E/F must be 16-aligned");
+        // GQA: n_head_kv query-head groups share each KV head
+        HK_ = (n_head_kv > 0) ? (int)n_head_kv : H_;
+        TORCH_CHECK(H_ % HK_ == 0, "n_head not divisible by n_head_kv");
+        EK_ = HK_ * D_;
+        TORCH_CHECK(EK_ % 16 == 0, "kv width must be 16-aligned");
         // prefill token cap per forward call: side channels and q/attn
         // buffers scale with it (a few hundred MB at 2048 on 65B shapes)
         maxP_ = std::max((int)max_prefill, kMaxTokens);
@@ -177,10 +183,11 @@ class SliceEngine {
         auto f32 = dev.dtype(torch::kFloat32);
         auto f16 = dev.dtype(torch::kFloat16);
         auto u16 = dev.dtype(torch::kInt16);
-        // KV cache: [L, B, ctx, E] f16 — the HBM3E-resident analog of the
-        // reference's kv_cache_init (tensor_processor.cpp:1089-1132).
-        k_cache_ = torch::zeros({L_, B_, ctx_, E_}, f16);
-        v_cache_ = torch::zeros({L_, B_, ctx_, E_}, f16);
+        // KV cache: [L, B, ctx, Ekv] f16 — the HBM3E-resident analog of
+        // the reference's kv_cache_init (tensor_processor.cpp:1089-1132);
+        // Ekv = Hkv*D < E under GQA.
+        k_cache_ = torch::zeros({L_, B_, ctx_, EK_}, f16);
+        v_cache_ = torch::zeros({L_, B_, ctx_, EK_}, f16);
         // RoPE pair frequencies: theta_i = pos * base^(-2i/D)
         inv_freq_ = torch::pow(
             (float)rope_base,
@@ -225,11 +232,13 @@ class SliceEngine {
         l.ffn_norm = check_f32(ffn_norm, "ffn_norm");
         l.attn_normprep = pad_normprep(attn_norm);
         l.ffn_normprep = pad_normprep(ffn_norm);
-        const int64_t rows[7] = {E_, E_, E_, E_, F_, E_, F_};
+        const int64_t rows[7] = {E_, EK_, EK_, E_, F_, E_, F_};
         const int64_t cols[7] = {E_, E_, E_, E_, E_, F_, E_};
         auto first = mats[0].cast<py::tuple>();
         const int64_t wt0 = first[2].cast<int64_t>();
         l.mfma = (wt0 != W_F32);
+        TORCH_CHECK(l.mfma || EK_ == E_,
+                    "the legacy f32 path does not support GQA");
         DevMat* slots[7] = {&l.wq, &l.wk, &l.wv, &l.wo, &l.w1, &l.w2, &l.w3};
         DevMat2* slots2[7] = {&l.mq, &l.mk, &l.mv, &l.mo, &l.m1, &l.m2,
                               &l.m3};
@@ -315,7 +324,7 @@ class SliceEngine {
         float* xp = x.data_ptr<float>();
         const int* pp = pos.data_ptr<int>();
         const int* sp = seq.data_ptr<int>();
-        const size_t layer_stride = (size_t)B_ * ctx_ * E_;
+        const size_t layer_stride = (size_t)B_ * ctx_ * EK_;
         __half* kbase = reinterpret_cast<__half*>(k_cache_.data_ptr());
         __half* vbase = reinterpret_cast<__half*>(v_cache_.data_ptr());
         const float* ifr = inv_freq_.data_ptr<float>();
@@ -346,10 +355,10 @@ class SliceEngine {
                 __half* vc = vbase + (size_t)li * layer_stride;
                 launch_qkv16_mt(s, l.mq.w, l.mk.w, l.mv.w, xprep,
                                 u16p(l.attn_normprep), ssa + li * ssw_,
-                                eps_, qb, kc, vc, pp, sp, ifr, E_, D_,
-                                ctx_, T);
+                                eps_, qb, kc, vc, pp, sp, ifr, E_, EK_,
+                                D_, ctx_, T);
                 launch_attn_prefill(s, qb, kc, vc, ab, aprep, pp, sp, T,
-                                    H_, E_, D_, ctx_);
+                                    H_, E_, EK_, D_, ctx_);
                 launch_gemm16_mt(s, l.mo.w, aprep, xp, xprep,
                                  ssf + li * ssw_, T, /*res_sq=*/1);
                 launch_ffn16_mt(s, l.m1.w, l.m3.w, xprep,
@@ -366,14 +375,14 @@ class SliceEngine {
             __half* vc = vbase + (size_t)li * layer_stride;
             const int slab_used = launch_qkv16(
                 s, l.mq.w, l.mk.w, l.mv.w, xprep, u16p(l.attn_normprep),
-                ssa + li * ssw_, eps_, qb, kc, vc, pp, sp, ifr, E_,
+                ssa + li * ssw_, eps_, qb, kc, vc, pp, sp, ifr, E_, EK_,
                 D_, ctx_, T, slab_.data_ptr<float>(),
                 /*skip_finish=*/decode ? 1 : 0);
             const bool fuse = decode && slab_used;
             launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T, H_, E_,
-                             D_, ctx_,
+                             EK_, D_, ctx_,
                              fuse ? slab_.data_ptr<float>() : nullptr,
-                             fuse ? qkv16_ks(E_) : 0, ifr);
+                             fuse ? qkv16_ks(E_, EK_) : 0, ifr);
             // wo/w2 tile count (E/16) alone underfills 256 CUs — split K
             // across grid.y into plain-stored slabs, then one fused
             // reduce+residual+sumsq+xprep pass per matrix (no atomics;
@@ -499,7 +508,7 @@ class SliceEngine {
             launch_qkv_rope_append(s, l.wq.w, l.wk.w, l.wv.w, xn, qb, kc, vc,
                                    pp, sp, ifr, E_, D_, ctx_, T);
             launch_attention(s, qb, kc, vc, ab, nullptr, pp, sp, T, H_, E_,
-                             D_, ctx_, nullptr, 0, ifr);
+                             EK_, D_, ctx_, nullptr, 0, ifr);
             launch_gemv(s, l.wo.w, ab, /*res=*/xp, xp, T);
             launch_rmsnorm(s, xp, l.ffn_norm.data_ptr<float>(), xn, T, E_,
                            eps_);
@@ -509,6 +518,7 @@ class SliceEngine {
     }
 
     int E_, H_, D_, F_, L_, ctx_, B_;
+    int HK_ = 0, EK_ = 0;    // GQA kv heads / kv projection width
     int maxP_ = kMaxTokens;  // prefill token cap per forward call
     int ssw_ = kMaxTokens;   // per-layer sumsq side-channel stride
     int V_ = 0;
@@ -532,11 +542,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.doc() = "MI355X-native layer-slice inference engine (CDNA4 HIP kernels)";
     py::class_<SliceEngine, std::shared_ptr<SliceEngine>>(m, "SliceEngine")
         .def(py::init<int64_t, int64_t, int64_t, int64_t, int64_t, int64_t,
-                      double, double, int64_t>(),
+                      double, double, int64_t, int64_t>(),
              py::arg("n_embd"), py::arg("n_head"), py::arg("n_layers"),
              py::arg("n_ff"), py::arg("n_ctx"), py::arg("max_batch"),
              py::arg("eps"), py::arg("rope_base"),
-             py::arg("max_prefill") = 1024)
+             py::arg("max_prefill") = 1024, py::arg("n_head_kv") = 0)
         .def("set_layer", &SliceEngine::set_layer)
         .def("set_extra", &SliceEngine::set_extra)
         .def("forward", &SliceEngine::forward, py::arg("x"),

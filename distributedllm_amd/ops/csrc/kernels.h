@@ -61,11 +61,11 @@ int launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                  const unsigned short* normprep, const float* ss_in,
                  float eps, float* q_buf, __half* k_cache_layer,
                  __half* v_cache_layer, const int* pos, const int* seq,
-                 const float* inv_freq, int E, int D, int n_ctx, int T,
-                 float* slab, int skip_finish);
+                 const float* inv_freq, int E, int Ekv, int D, int n_ctx,
+                 int T, float* slab, int skip_finish);
 
 // the qkv slab split factor (shared with the fused-attention consumer)
-int qkv16_ks(int E);
+int qkv16_ks(int E, int Ekv);
 
 void launch_ffn16(hipStream_t s, const WMat2& w1, const WMat2& w3,
                   const unsigned short* xprep,
@@ -88,8 +88,9 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
 void launch_attention(hipStream_t s, const float* q_buf,
                       const __half* k_cache_layer, __half* v_cache_layer,
                       float* out, unsigned short* out_prep, const int* pos,
-                      const int* seq, int T, int H, int E, int D, int n_ctx,
-                      const float* qkv_slab, int ks, const float* inv_freq);
+                      const int* seq, int T, int H, int E, int Ekv, int D,
+                      int n_ctx, const float* qkv_slab, int ks,
+                      const float* inv_freq);
 
 void launch_gemv(hipStream_t s, const WMat& w, const float* x,
                  const float* res, float* y, int T);
@@ -115,15 +116,16 @@ void launch_attn_prefill(hipStream_t s, const float* q_buf,
                          const __half* k_cache_layer,
                          const __half* v_cache_layer, float* out,
                          unsigned short* out_prep, const int* pos,
-                         const int* seq, int T, int H, int E, int D,
-                         int n_ctx);
+                         const int* seq, int T, int H, int E, int Ekv,
+                         int D, int n_ctx);
 
 void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
                      const WMat2& wv, const unsigned short* xprep,
                      const unsigned short* normprep, const float* ss_in,
                      float eps, float* q_buf, __half* k_cache_layer,
                      __half* v_cache_layer, const int* pos, const int* seq,
-                     const float* inv_freq, int E, int D, int n_ctx, int T);
+                     const float* inv_freq, int E, int Ekv, int D,
+                     int n_ctx, int T);
 
 void launch_gemm16_mt(hipStream_t s, const WMat2& w,
                       const unsigned short* bprep, float* y,

@@ -408,13 +408,16 @@ __global__ void k_attention(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
     __half* __restrict__ v_cache, float* __restrict__ out,
     unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
-    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw,
-    const float* __restrict__ qkv_slab, int ks,
+    const int* __restrict__ seq, int E, int Ekv, int D, int n_ctx,
+    int jtw, const float* __restrict__ qkv_slab, int ks,
     const float* __restrict__ inv_freq) {
     const int t = blockIdx.x;
     const int h = blockIdx.y;
     const int J = pos[t] + 1;
-    const size_t base = (size_t)seq[t] * n_ctx * E + h * D;
+    // GQA: q head h reads kv head h / (H / Hkv); KV rows are Ekv wide
+    const int grp = (int)gridDim.y / (Ekv / D);
+    const int hk = h / grp;
+    const size_t base = (size_t)seq[t] * n_ctx * Ekv + hk * D;
     const float inv_sqrt_d = rsqrtf((float)D);
 
     extern __shared__ float smem[];
@@ -423,22 +426,28 @@ __global__ void k_attention(
     float* lds_red = lds_p + BLOCK;  // [NWAVES]
 
     if (FUSEQKV) {
-        // qkv_slab layout: f32[3*(E/16) gtiles][ks][64 tok][16 rows];
-        // matrices q/k/v are consecutive gtile ranges.
+        // qkv_slab layout: f32[(E + 2*Ekv)/16 gtiles][ks][64 tok][16
+        // rows]; q rows then k rows then v rows. With GQA the `grp`
+        // blocks sharing a kv head all append the SAME row bytes —
+        // redundant but benign.
         const int p = J - 1;
-        const int Et = E >> 4;
-        const size_t matstride = (size_t)Et * ks * 64 * 16;
-        __half* kv_k = const_cast<__half*>(k_cache) + base + (size_t)p * E;
-        __half* kv_v = v_cache + base + (size_t)p * E;
+        const size_t koff = (size_t)(E >> 4) * ks * 64 * 16;
+        const size_t voff = koff + (size_t)(Ekv >> 4) * ks * 64 * 16;
+        __half* kv_k =
+            const_cast<__half*>(k_cache) + base + (size_t)p * Ekv;
+        __half* kv_v = v_cache + base + (size_t)p * Ekv;
         for (int d = threadIdx.x; d < D; d += BLOCK) {
-            const int e = h * D + d;
+            const int e = h * D + d;        // row in q space
+            const int ek = hk * D + d;      // row in kv space
             float q = 0.f, k = 0.f, v = 0.f;
             for (int c = 0; c < ks; ++c) {
-                const size_t off =
-                    (((size_t)(e >> 4) * ks + c) * 64 + t) * 16 + (e & 15);
-                q += qkv_slab[off];
-                k += qkv_slab[matstride + off];
-                v += qkv_slab[2 * matstride + off];
+                q += qkv_slab[(((size_t)(e >> 4) * ks + c) * 64 + t) * 16 +
+                              (e & 15)];
+                const size_t offk =
+                    (((size_t)(ek >> 4) * ks + c) * 64 + t) * 16 +
+                    (ek & 15);
+                k += qkv_slab[koff + offk];
+                v += qkv_slab[voff + offk];
             }
             lds_q[d] = q;
             lds_p[d] = k;  // staged for the pair rotation below
@@ -477,7 +486,7 @@ __global__ void k_attention(
         const int jj = j0 + threadIdx.x;
         float s = -INFINITY;
         if (jj < J) {
-            const __half* krow = k_cache + base + (size_t)jj * E;
+            const __half* krow = k_cache + base + (size_t)jj * Ekv;
             float acc = 0.0f;
             const int nh2 = D >> 2;  // 4 halves per float2 load
             const float2* k2 = reinterpret_cast<const float2*>(krow);
@@ -516,22 +525,24 @@ __global__ void k_attention(
         if (threadIdx.x < D) {
             o *= alpha;
             const int jmax = min(BLOCK, J - j0);
-            const __half* vcol = v_cache + base + (size_t)j0 * E + threadIdx.x;
+            const __half* vcol =
+                v_cache + base + (size_t)j0 * Ekv + threadIdx.x;
             int jc = 0;
             for (; jc + 4 <= jmax; jc += 4) {
                 const float p0 = lds_p[jc], p1 = lds_p[jc + 1];
                 const float p2 = lds_p[jc + 2], p3 = lds_p[jc + 3];
-                const float v0 = __half2float(vcol[(size_t)jc * E]);
-                const float v1 = __half2float(vcol[(size_t)(jc + 1) * E]);
-                const float v2 = __half2float(vcol[(size_t)(jc + 2) * E]);
-                const float v3 = __half2float(vcol[(size_t)(jc + 3) * E]);
+                const float v0 = __half2float(vcol[(size_t)jc * Ekv]);
+                const float v1 = __half2float(vcol[(size_t)(jc + 1) * Ekv]);
+                const float v2 = __half2float(vcol[(size_t)(jc + 2) * Ekv]);
+                const float v3 = __half2float(vcol[(size_t)(jc + 3) * Ekv]);
                 o = fmaf(p0, v0, o);
                 o = fmaf(p1, v1, o);
                 o = fmaf(p2, v2, o);
                 o = fmaf(p3, v3, o);
             }
             for (; jc < jmax; ++jc)
-                o = fmaf(lds_p[jc], __half2float(vcol[(size_t)jc * E]), o);
+                o = fmaf(lds_p[jc],
+                         __half2float(vcol[(size_t)jc * Ekv]), o);
         }
         __syncthreads();
     }
@@ -563,7 +574,8 @@ __global__ void k_attn_prefill(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
     const __half* __restrict__ v_cache, float* __restrict__ out,
     unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
-    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw, int T) {
+    const int* __restrict__ seq, int E, int Ekv, int D, int n_ctx,
+    int jtw, int T) {
     constexpr int QT = 16;
     const int t0 = blockIdx.x * QT;
     const int h = blockIdx.y;
@@ -690,7 +702,8 @@ __global__ void k_attn_prefill(
             // V accumulation: thread d sums p[q][j] * V[j][d]
             if (tid < D) {
                 const int jlim = min(BLOCK, J - j0);
-                const __half* vcol = v_cache + base + (size_t)j0 * E + tid;
+                const __half* vcol =
+                    v_cache + base + (size_t)j0 * Ekv + tid;
 #pragma unroll
                 for (int q = 0; q < QT; ++q) {
                     if (q < au || q >= bu) continue;
@@ -702,7 +715,7 @@ __global__ void k_attn_prefill(
                     m[q] = m_new;
                 }
                 for (int jc = 0; jc < jlim; ++jc) {
-                    const float v = __half2float(vcol[(size_t)jc * E]);
+                    const float v = __half2float(vcol[(size_t)jc * Ekv]);
 #pragma unroll
                     for (int q = 0; q < QT; ++q) {
                         if (q < au || q >= bu) continue;
@@ -773,7 +786,8 @@ __global__ __launch_bounds__(BLOCK) void k_attn_prefill_mfma(
     const float* __restrict__ q_buf, const __half* __restrict__ k_cache,
     const __half* __restrict__ v_cache, float* __restrict__ out,
     unsigned short* __restrict__ out_prep, const int* __restrict__ pos,
-    const int* __restrict__ seq, int E, int D, int n_ctx, int jtw, int T) {
+    const int* __restrict__ seq, int E, int Ekv, int D, int n_ctx,
+    int jtw, int T) {
     constexpr int QT = 16;
     constexpr int DCMAX = 8;            // D <= 128 = 8 chunks of 16
     const int t0 = blockIdx.x * QT;
@@ -836,14 +850,15 @@ __global__ __launch_bounds__(BLOCK) void k_attn_prefill_mfma(
         for (int q = a; q < b; ++q) jmax = max(jmax, lds_pos[q]);
         const int J = jmax + 1;
         const bool active = (lq >= a && lq < b);
-        const size_t base = (size_t)sseq * n_ctx * E + h * D;
+        const int hk = h / ((int)gridDim.y / (Ekv / D));  // GQA kv head
+        const size_t base = (size_t)sseq * n_ctx * Ekv + hk * D;
 
         // wave w streams tiles w, w+4, w+8, ... of this segment
         for (int tj = wid * 16; tj < J; tj += NWAVES * 16) {
             // scores: A = K rows (lane: row j=lq of the tile, k-span lg)
             const bool jrow_ok = (tj + lq) < J;  // tile tail: no OOB read
             const __half* krow =
-                k_cache + base + (size_t)(jrow_ok ? tj + lq : 0) * E;
+                k_cache + base + (size_t)(jrow_ok ? tj + lq : 0) * Ekv;
             f32x4 sc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
             for (int c = 0; c < 4; ++c) {
@@ -905,7 +920,7 @@ __global__ __launch_bounds__(BLOCK) void k_attn_prefill_mfma(
                     const int j = tj + lg * 4 + e;
                     __half v(0.0f);
                     if (d < D && j < J)
-                        v = v_cache[base + (size_t)j * E + d];
+                        v = v_cache[base + (size_t)j * Ekv + d];
                     reinterpret_cast<__half*>(vt.u)[e] = v;
                 }
                 oacc[dc] = __builtin_amdgcn_mfma_f32_16x16x16f16(
@@ -1780,6 +1795,8 @@ __global__ __launch_bounds__(BLOCK) void k_gemm16(
 
 // ------------------------------------------------------------- k_qkv16
 // QKV projections on MFMA + fused input RMSNorm + RoPE + KV append.
+// GQA: wk/wv have Ekv = Hkv*D rows (< E); KV cache rows are Ekv wide.
+// Tile space is [0, E/16) for wq then [.., +Ekv/16) each for wk/wv.
 template <int WT, int JT, int RT = 1>
 __global__ __launch_bounds__(BLOCK) void k_qkv16(
     WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
@@ -1787,10 +1804,14 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
     const float* __restrict__ ss_in, float eps, float* __restrict__ q_buf,
     __half* __restrict__ k_cache, __half* __restrict__ v_cache,
     const int* __restrict__ pos, const int* __restrict__ seq,
-    const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T) {
-    const int tiles_per_mat = (E >> 4) / RT;
-    const int mat = blockIdx.x / tiles_per_mat;
-    const int tile = blockIdx.x % tiles_per_mat;
+    const float* __restrict__ inv_freq, int E, int Ekv, int D, int n_ctx,
+    int T) {
+    const int tq = (E >> 4) / RT;
+    const int tk = (Ekv >> 4) / RT;
+    const int mat = (blockIdx.x < tq) ? 0
+                    : (blockIdx.x < tq + tk) ? 1 : 2;
+    const int tile = blockIdx.x - ((mat == 0) ? 0 : (mat == 1) ? tq
+                                                               : tq + tk);
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
@@ -1811,7 +1832,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
         if (j2 >= T) continue;
         const int p = pos[j2];
         if (mat == 2) {  // V rows: straight f16 cache append
-            __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
+            __half* dst =
+                v_cache + ((size_t)seq[j2] * n_ctx + p) * Ekv + r0;
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
                 dst[jj] = __float2half(acc[rt][0][jt][jj]);
@@ -1833,7 +1855,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16(
                 q_buf[(size_t)j2 * E + e] = o0;
                 q_buf[(size_t)j2 * E + e + 1] = o1;
             } else {
-                __half* dst = k_cache + ((size_t)seq[j2] * n_ctx + p) * E + e;
+                __half* dst =
+                    k_cache + ((size_t)seq[j2] * n_ctx + p) * Ekv + e;
                 dst[0] = __float2half(o0);
                 dst[1] = __float2half(o1);
             }
@@ -1899,10 +1922,13 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_slab(
     WMat2 wq, WMat2 wk, WMat2 wv, const unsigned short* __restrict__ xprep,
     const unsigned short* __restrict__ normprep,
     const float* __restrict__ ss_in, float eps, float* __restrict__ slab,
-    int E, int T) {
-    const int st_per_mat = (E >> 4) / RT;
-    const int mat = blockIdx.x / st_per_mat;
-    const int stile = blockIdx.x % st_per_mat;
+    int E, int Ekv, int T) {
+    const int tq = (E >> 4) / RT;
+    const int tk = (Ekv >> 4) / RT;
+    const int mat = (blockIdx.x < tq) ? 0
+                    : (blockIdx.x < tq + tk) ? 1 : 2;
+    const int stile = blockIdx.x - ((mat == 0) ? 0 : (mat == 1) ? tq
+                                                                : tq + tk);
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
@@ -1918,9 +1944,13 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_slab(
     __shared__ float lds[3 * 64 * 4 * RT * JT];
     combine_acc<RT * JT>(reinterpret_cast<float(*)[4]>(acc), lds);
     if (threadIdx.x >= WAVE) return;
+    // slab gtile space: q rows [0, E/16) then k, v rows [.., +Ekv/16)
+    const int gbase = (mat == 0) ? 0
+                      : (mat == 1) ? (E >> 4)
+                                   : (E >> 4) + (Ekv >> 4);
 #pragma unroll
     for (int rt = 0; rt < RT; ++rt) {
-        const size_t gtile = (size_t)mat * (E >> 4) + stile * RT + rt;
+        const size_t gtile = (size_t)gbase + stile * RT + rt;
         float* sl = slab + ((gtile * gridDim.y + blockIdx.y) * 64) * 16;
 #pragma unroll
         for (int jt = 0; jt < JT; ++jt) {
@@ -1945,15 +1975,16 @@ __global__ void k_qkv_finish(const float* __restrict__ slab, int ks,
                              const int* __restrict__ pos,
                              const int* __restrict__ seq,
                              const float* __restrict__ inv_freq, int E,
-                             int D, int n_ctx, int T) {
-    const int nchunks = 3 * E / 8;
+                             int Ekv, int D, int n_ctx, int T) {
+    const int nchunks = (E + 2 * Ekv) / 8;
     const int idx = blockIdx.x * BLOCK + threadIdx.x;
     if (idx >= nchunks * T) return;
     const int chunk = idx % nchunks;
     const int t = idx / nchunks;
-    const int mat = (8 * chunk) / E;
-    const int e = (8 * chunk) % E;   // 8-aligned
-    const size_t gt0 = ((size_t)mat * (E >> 4) + (e >> 4)) * ks;
+    const int r = 8 * chunk;  // global row in the [E | Ekv | Ekv] space
+    const int mat = (r < E) ? 0 : (r < E + Ekv) ? 1 : 2;
+    const int e = (mat == 0) ? r : (mat == 1) ? r - E : r - E - Ekv;
+    const size_t gt0 = ((size_t)(r >> 4)) * ks;  // 8-aligned within mat
     float v[8];
 #pragma unroll
     for (int i = 0; i < 8; ++i) v[i] = 0.f;
@@ -1966,7 +1997,7 @@ __global__ void k_qkv_finish(const float* __restrict__ slab, int ks,
     }
     const int p = pos[t];
     if (mat == 2) {
-        __half* dst = v_cache + ((size_t)seq[t] * n_ctx + p) * E + e;
+        __half* dst = v_cache + ((size_t)seq[t] * n_ctx + p) * Ekv + e;
         uint4 o;
         o.x = pack_f16(v[0], v[1]);
         o.y = pack_f16(v[2], v[3]);
@@ -1992,7 +2023,7 @@ __global__ void k_qkv_finish(const float* __restrict__ slab, int ks,
         *reinterpret_cast<float4*>(dst + 4) =
             make_float4(o[4], o[5], o[6], o[7]);
     } else {
-        __half* dst = k_cache + ((size_t)seq[t] * n_ctx + p) * E + e;
+        __half* dst = k_cache + ((size_t)seq[t] * n_ctx + p) * Ekv + e;
         uint4 w;
         w.x = pack_f16(o[0], o[1]);
         w.y = pack_f16(o[2], o[3]);
@@ -2123,15 +2154,16 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
     const float* __restrict__ ss_in, float eps, float* __restrict__ q_buf,
     __half* __restrict__ k_cache, __half* __restrict__ v_cache,
     const int* __restrict__ pos, const int* __restrict__ seq,
-    const float* __restrict__ inv_freq, int E, int D, int n_ctx, int T,
-    int MT, int jtw) {
+    const float* __restrict__ inv_freq, int E, int Ekv, int D, int n_ctx,
+    int T, int MT, int jtw) {
     int gtile, mtile;
-    const int tiles_per_mat = (E >> 4) / RT;
-    const int GT = 3 * tiles_per_mat;
+    const int tq = (E >> 4) / RT;
+    const int tk = (Ekv >> 4) / RT;
+    const int GT = tq + 2 * tk;
     xcd_decode(GT, MT, gtile, mtile);
     if (gtile >= GT) return;
-    const int mat = gtile / tiles_per_mat;
-    const int tile = gtile % tiles_per_mat;
+    const int mat = (gtile < tq) ? 0 : (gtile < tq + tk) ? 1 : 2;
+    const int tile = gtile - ((mat == 0) ? 0 : (mat == 1) ? tq : tq + tk);
     const WMat2& w = (mat == 0) ? wq : (mat == 1) ? wk : wv;
     const int lane = threadIdx.x & (WAVE - 1);
     const int j = lane & 15;
@@ -2153,7 +2185,8 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
         if (j2 >= T) continue;
         const int p = pos[j2];
         if (mat == 2) {
-            __half* dst = v_cache + ((size_t)seq[j2] * n_ctx + p) * E + r0;
+            __half* dst =
+                v_cache + ((size_t)seq[j2] * n_ctx + p) * Ekv + r0;
 #pragma unroll
             for (int jj = 0; jj < 4; ++jj)
                 dst[jj] = __float2half(acc[rt][0][jt][jj]);
@@ -2175,7 +2208,7 @@ __global__ __launch_bounds__(BLOCK) void k_qkv16_mt(
                 q_buf[(size_t)j2 * E + e + 1] = o1;
             } else {
                 __half* dst =
-                    k_cache + ((size_t)seq[j2] * n_ctx + p) * E + e;
+                    k_cache + ((size_t)seq[j2] * n_ctx + p) * Ekv + e;
                 dst[0] = __float2half(o0);
                 dst[1] = __float2half(o1);
             }
@@ -2392,21 +2425,22 @@ void launch_qkv_rope_append(hipStream_t s, const WMat& wq, const WMat& wk,
 void launch_attention(hipStream_t s, const float* q_buf,
                       const __half* k_cache_layer, __half* v_cache_layer,
                       float* out, unsigned short* out_prep, const int* pos,
-                      const int* seq, int T, int H, int E, int D, int n_ctx,
-                      const float* qkv_slab, int ks,
+                      const int* seq, int T, int H, int E, int Ekv, int D,
+                      int n_ctx, const float* qkv_slab, int ks,
                       const float* inv_freq) {
     const dim3 grid(T, H);
     const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
     if (qkv_slab != nullptr) {
         hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
-                           out_prep, pos, seq, E, D, n_ctx, jt_width(T),
-                           qkv_slab, ks, inv_freq);
+                           out_prep, pos, seq, E, Ekv, D, n_ctx,
+                           jt_width(T), qkv_slab, ks, inv_freq);
         return;
     }
     hipLaunchKernelGGL(k_attention<false>, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos, seq,
-                       E, D, n_ctx, jt_width(T), qkv_slab, ks, inv_freq);
+                       E, Ekv, D, n_ctx, jt_width(T), qkv_slab, ks,
+                       inv_freq);
 }
 
 // ------------------------------------------------- MFMA-path launchers
@@ -2467,8 +2501,8 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
         }                                        \
     }
 
-int qkv16_ks(int E) {
-    const int tiles3 = 3 * (E >> 4);
+int qkv16_ks(int E, int Ekv) {
+    const int tiles3 = (E + 2 * Ekv) >> 4;
     int ks = 1;
     while ((tiles3 / 2) * ks < 512 && ks < 8) ks <<= 1;
     return ks;
@@ -2561,42 +2595,43 @@ int launch_qkv16(hipStream_t s, const WMat2& wq, const WMat2& wk,
                  const unsigned short* normprep, const float* ss_in,
                  float eps, float* q_buf, __half* k_cache_layer,
                  __half* v_cache_layer, const int* pos, const int* seq,
-                 const float* inv_freq, int E, int D, int n_ctx, int T,
-                 float* slab, int skip_finish) {
-    const int tiles3 = 3 * (E >> 4);
+                 const float* inv_freq, int E, int Ekv, int D, int n_ctx,
+                 int T, float* slab, int skip_finish) {
+    const int tiles3 = (E + 2 * Ekv) >> 4;
+    const bool rt2_ok = ((E >> 4) % 2 == 0) && ((Ekv >> 4) % 2 == 0);
     // slab split-K + RT=2 path when the fused grid underfills the chip
-    if (slab != nullptr && tiles3 < 1024 && ((E >> 4) % 2) == 0) {
+    if (slab != nullptr && tiles3 < 1024 && rt2_ok) {
         // RT=2 measured best (RT=4 loses ~3%: fill drops below 2/CU)
-        const int ks = qkv16_ks(E);
+        const int ks = qkv16_ks(E, Ekv);
         const dim3 grid(tiles3 / 2, ks);
         DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
             (k_qkv16_slab<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk,
-            wv, xprep, normprep, ss_in, eps, slab, E, T)));
+            wv, xprep, normprep, ss_in, eps, slab, E, Ekv, T)));
         if (!skip_finish) {
-            const int total = (3 * E / 8) * T;
+            const int total = ((E + 2 * Ekv) / 8) * T;
             hipLaunchKernelGGL(k_qkv_finish,
                                dim3((total + BLOCK - 1) / BLOCK),
                                dim3(BLOCK), 0, s, slab, ks, q_buf,
                                k_cache_layer, v_cache_layer, pos, seq,
-                               inv_freq, E, D, n_ctx, T);
+                               inv_freq, E, Ekv, D, n_ctx, T);
         }
         return 1;
     }
-    if (((E >> 4) % 2) == 0 && tiles3 / 2 >= 512) {
+    if (rt2_ok && tiles3 / 2 >= 512) {
         // big models: RT=2 fused — halves the B-panel re-read while the
         // halved grid still fills the chip
         const dim3 grid(tiles3 / 2);
         DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
             (k_qkv16<WTc, JTc, 2>), grid, dim3(BLOCK), 0, s, wq, wk, wv,
             xprep, normprep, ss_in, eps, q_buf, k_cache_layer,
-            v_cache_layer, pos, seq, inv_freq, E, D, n_ctx, T)));
+            v_cache_layer, pos, seq, inv_freq, E, Ekv, D, n_ctx, T)));
         return 0;
     }
     const dim3 grid(tiles3);
     DISPATCH_WT2(wq.wtype, DISPATCH_JT(pick_jt(T), hipLaunchKernelGGL(
         (k_qkv16<WTc, JTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
         normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos, seq,
-        inv_freq, E, D, n_ctx, T)));
+        inv_freq, E, Ekv, D, n_ctx, T)));
     return 0;
 }
 
@@ -2684,20 +2719,20 @@ void launch_attn_prefill(hipStream_t s, const float* q_buf,
                          const __half* k_cache_layer,
                          const __half* v_cache_layer, float* out,
                          unsigned short* out_prep, const int* pos,
-                         const int* seq, int T, int H, int E, int D,
-                         int n_ctx) {
+                         const int* seq, int T, int H, int E, int Ekv,
+                         int D, int n_ctx) {
     const dim3 grid((T + 15) / 16, H);
     if (D <= 128) {  // every LLaMA head dim; matrix-core path
         hipLaunchKernelGGL(k_attn_prefill_mfma, grid, dim3(BLOCK), 0, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
-                           out_prep, pos, seq, E, D, n_ctx, jt_width(T),
-                           T);
+                           out_prep, pos, seq, E, Ekv, D, n_ctx,
+                           jt_width(T), T);
         return;
     }
     const size_t lds = (size_t)(16 * D + 16 * BLOCK + 32) * sizeof(float);
     hipLaunchKernelGGL(k_attn_prefill, grid, dim3(BLOCK), lds, s, q_buf,
                        k_cache_layer, v_cache_layer, out, out_prep, pos,
-                       seq, E, D, n_ctx, jt_width(T), T);
+                       seq, E, Ekv, D, n_ctx, jt_width(T), T);
 }
 
 void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
@@ -2705,25 +2740,28 @@ void launch_qkv16_mt(hipStream_t s, const WMat2& wq, const WMat2& wk,
                      const unsigned short* normprep, const float* ss_in,
                      float eps, float* q_buf, __half* k_cache_layer,
                      __half* v_cache_layer, const int* pos, const int* seq,
-                     const float* inv_freq, int E, int D, int n_ctx,
-                     int T) {
+                     const float* inv_freq, int E, int Ekv, int D,
+                     int n_ctx, int T) {
     const int MT = (T + 63) >> 6;
     const int jtw = jt_width(T);
+    const int tiles3 = (E + 2 * Ekv) >> 4;
+    const bool rt2_ok = ((E >> 4) % 2 == 0) && ((Ekv >> 4) % 2 == 0);
     // RT=2 halves the B-panel reads + norm-build VALU per MFMA as long
     // as the halved grid still fills the chip
-    if ((E >> 4) % 2 == 0 && (3 * (E >> 4) / 2) * MT >= 512) {
-        const dim3 grid(xcd_grid(3 * (E >> 4) / 2, MT));
+    if (rt2_ok && (tiles3 / 2) * MT >= 512) {
+        const dim3 grid(xcd_grid(tiles3 / 2, MT));
         DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
             (k_qkv16_mt<WTc, 4, 2>), grid, dim3(BLOCK), 0, s, wq, wk, wv,
             xprep, normprep, ss_in, eps, q_buf, k_cache_layer,
-            v_cache_layer, pos, seq, inv_freq, E, D, n_ctx, T, MT, jtw));
+            v_cache_layer, pos, seq, inv_freq, E, Ekv, D, n_ctx, T, MT,
+            jtw));
         return;
     }
-    const dim3 grid(xcd_grid(3 * (E >> 4), MT));
+    const dim3 grid(xcd_grid(tiles3, MT));
     DISPATCH_WT2(wq.wtype, hipLaunchKernelGGL(
         (k_qkv16_mt<WTc>), grid, dim3(BLOCK), 0, s, wq, wk, wv, xprep,
         normprep, ss_in, eps, q_buf, k_cache_layer, v_cache_layer, pos,
-        seq, inv_freq, E, D, n_ctx, T, MT, jtw));
+        seq, inv_freq, E, Ekv, D, n_ctx, T, MT, jtw));
 }
 
 void launch_gemm16_mt(hipStream_t s, const WMat2& w,

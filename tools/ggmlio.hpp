@@ -31,6 +31,7 @@ namespace ggmlio {
 
 constexpr uint32_t kMagic = 0x67676A74;  // "ggjt" little-endian
 constexpr uint32_t kVersion = 3;
+constexpr uint32_t kVersionGqa = 4;  // framework GQA extension
 constexpr uint32_t kExtraLayersFirstLayer = 0xFFFFFFFFu;
 
 enum GType : uint32_t { F32 = 0, F16 = 1, Q4_0 = 2, Q4_1 = 3,
@@ -291,6 +292,10 @@ struct Hparams {
              n_rot = 0, ftype = 0;
     bool extended = false;       // 8-field header
     uint32_t first_layer = 0;    // valid when extended
+    // GQA extension (GGJT version 4): n_head_kv right after n_head;
+    // written only when gqa (MHA files stay byte-identical v3)
+    bool gqa = false;
+    uint32_t n_head_kv = 0;      // valid when gqa
 };
 
 struct Tensor {
@@ -332,12 +337,16 @@ class Reader {
         off_ = 0;
         File out;
         if (u32() != kMagic) throw std::runtime_error("bad GGJT magic");
-        if (u32() != kVersion) throw std::runtime_error("bad GGJT version");
+        const uint32_t version = u32();
+        if (version != kVersion && version != kVersionGqa)
+            throw std::runtime_error("bad GGJT version");
         Hparams& hp = out.hp;
         hp.n_vocab = u32();
         hp.n_embd = u32();
         hp.n_mult = u32();
         hp.n_head = u32();
+        hp.gqa = (version == kVersionGqa);
+        if (hp.gqa) hp.n_head_kv = u32();
         hp.n_layer = u32();
         hp.n_rot = u32();
         hp.extended = extended;
@@ -399,13 +408,14 @@ inline void write_file(const std::string& path, const File& file) {
         pos += n;
     };
     auto put32 = [&](uint32_t v) { put(&v, 4); };
-    put32(kMagic);
-    put32(kVersion);
     const Hparams& hp = file.hp;
+    put32(kMagic);
+    put32(hp.gqa ? kVersionGqa : kVersion);
     put32(hp.n_vocab);
     put32(hp.n_embd);
     put32(hp.n_mult);
     put32(hp.n_head);
+    if (hp.gqa) put32(hp.n_head_kv);
     put32(hp.n_layer);
     put32(hp.n_rot);
     if (hp.extended) put32(hp.first_layer);
