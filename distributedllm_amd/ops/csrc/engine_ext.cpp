@@ -166,7 +166,8 @@ class SliceEngine {
           eps_((float)eps) {
         TORCH_CHECK(E_ % H_ == 0, "n_embd not divisible by n_head");
         TORCH_CHECK(D_ % 2 == 0, "head_dim must be even for RoPE pairs");
-        TORCH_CHECK(D_ <= 256, "head_dim > 256 unsupported");
+        TORCH_CHECK(D_ <= 128, "head_dim > 128 unsupported (the "
+                    "attention kernels cover d < 128 per lane pair)");
         TORCH_CHECK(E_ % 16 == 0 && F_ % 16 == 0, "E/F must be 16-aligned");
         // GQA: n_head_kv query-head groups share each KV head
         HK_ = (n_head_kv > 0) ? (int)n_head_kv : H_;

@@ -480,7 +480,9 @@ __global__ void k_attention(
 
     float m = -INFINITY;  // running max (block-uniform)
     float l = 0.0f;       // running denom (block-uniform)
-    float o = 0.0f;       // thread d accumulator (threads >= D idle)
+    // per-(wave, lane) V partials: lane covers d = lane and lane + 64
+    float o = 0.0f, o2 = 0.0f;
+    const int wid = threadIdx.x / WAVE;
 
     for (int j0 = 0; j0 < J; j0 += BLOCK) {
         const int jj = j0 + threadIdx.x;
@@ -504,7 +506,6 @@ __global__ void k_attention(
         }
         // block max
         float wm = wave_reduce_max(s);
-        const int wid = threadIdx.x / WAVE;
         if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = wm;
         __syncthreads();
         float cm = fmaxf(fmaxf(lds_red[0], lds_red[1]),
@@ -521,42 +522,57 @@ __global__ void k_attention(
         const float sum_p = lds_red[0] + lds_red[1] + lds_red[2] + lds_red[3];
         l = l * alpha + sum_p;
         m = m_new;
-        // V accumulation: thread d sums p_j * V[j][d]
-        if (threadIdx.x < D) {
-            o *= alpha;
+        // V accumulation: all 4 waves split the chunk's rows (the old
+        // thread-d-only form idled waves 2-3 and serialized a 256-fma
+        // dependency chain; this was HALF the deep-decode step —
+        // profiles/). Wave w covers rows [w*64, +64); lane covers d =
+        // lane and lane+64 (D <= 128); per-wave partials merge at the
+        // very end (alpha scaling is block-uniform, so partials scale
+        // consistently every chunk).
+        {
             const int jmax = min(BLOCK, J - j0);
-            const __half* vcol =
-                v_cache + base + (size_t)j0 * Ekv + threadIdx.x;
-            int jc = 0;
-            for (; jc + 4 <= jmax; jc += 4) {
-                const float p0 = lds_p[jc], p1 = lds_p[jc + 1];
-                const float p2 = lds_p[jc + 2], p3 = lds_p[jc + 3];
-                const float v0 = __half2float(vcol[(size_t)jc * Ekv]);
-                const float v1 = __half2float(vcol[(size_t)(jc + 1) * Ekv]);
-                const float v2 = __half2float(vcol[(size_t)(jc + 2) * Ekv]);
-                const float v3 = __half2float(vcol[(size_t)(jc + 3) * Ekv]);
-                o = fmaf(p0, v0, o);
-                o = fmaf(p1, v1, o);
-                o = fmaf(p2, v2, o);
-                o = fmaf(p3, v3, o);
+            const int jq0 = wid * (BLOCK / NWAVES);
+            const int jq1 = min(jmax, jq0 + BLOCK / NWAVES);
+            const int lane = threadIdx.x & (WAVE - 1);
+            o *= alpha;
+            o2 *= alpha;
+            const __half* vrow0 =
+                v_cache + base + (size_t)(j0 + jq0) * Ekv;
+            const bool d2ok = (lane + WAVE) < D;
+            for (int jc = jq0; jc < jq1; ++jc) {
+                const float pj = lds_p[jc];
+                const __half* vr = vrow0 + (size_t)(jc - jq0) * Ekv;
+                if (lane < D) o = fmaf(pj, __half2float(vr[lane]), o);
+                if (d2ok)
+                    o2 = fmaf(pj, __half2float(vr[lane + WAVE]), o2);
             }
-            for (; jc < jmax; ++jc)
-                o = fmaf(lds_p[jc],
-                         __half2float(vcol[(size_t)jc * Ekv]), o);
         }
         __syncthreads();
     }
-    if (threadIdx.x < D) {
-        const float v = o / l;
-        const int e = h * D + threadIdx.x;
-        out[(size_t)t * E + e] = v;
-        if (out_prep != nullptr) {
-            // f16 B-layout side-channel for the wo MFMA consumer:
-            // element (k=e, token t) at [((e>>3)*JT + t/16)*16 + t%16]*8
-            union { __half h; unsigned short u; } c;
-            c.h = __float2half(v);
-            out_prep[(((size_t)(e >> 3) * jtw + (t >> 4)) * 16 + (t & 15))
-                         * 8 + (e & 7)] = c.u;
+    // merge the 4 waves' o partials: wave w writes its (d, d+64) pair
+    {
+        const int lane = threadIdx.x & (WAVE - 1);
+        float* lds_o = lds_red + NWAVES;  // [2][BLOCK] merge area
+        lds_o[wid * WAVE + lane] = o;
+        lds_o[BLOCK + wid * WAVE + lane] = o2;
+        __syncthreads();
+        if (threadIdx.x < D) {
+            const int d = threadIdx.x;
+            float v = 0.0f;
+#pragma unroll
+            for (int w = 0; w < NWAVES; ++w)
+                v += (d < WAVE) ? lds_o[w * WAVE + d]
+                                : lds_o[BLOCK + w * WAVE + (d - WAVE)];
+            v /= l;
+            const int e = h * D + d;
+            out[(size_t)t * E + e] = v;
+            if (out_prep != nullptr) {
+                // f16 B-layout side-channel for the wo MFMA consumer
+                union { __half h; unsigned short u; } c;
+                c.h = __float2half(v);
+                out_prep[(((size_t)(e >> 3) * jtw + (t >> 4)) * 16 +
+                          (t & 15)) * 8 + (e & 7)] = c.u;
+            }
         }
     }
 }
@@ -2429,7 +2445,7 @@ void launch_attention(hipStream_t s, const float* q_buf,
                       int n_ctx, const float* qkv_slab, int ks,
                       const float* inv_freq) {
     const dim3 grid(T, H);
-    const size_t lds = (D + BLOCK + NWAVES) * sizeof(float);
+    const size_t lds = (D + BLOCK + NWAVES + 2 * BLOCK) * sizeof(float);
     if (qkv_slab != nullptr) {
         hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
