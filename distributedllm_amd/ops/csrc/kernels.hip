@@ -398,6 +398,26 @@ __global__ void k_qkv_rope_append(
 // entries [0, pos[t]] of sequence seq[t] — causality for prefill rows comes
 // from their per-row positions.
 
+// 8 halves [d0, d0+8) of `row` as floats, zero-padded past D (guarded
+// tail so the last row of the cache never reads out of bounds)
+__device__ __forceinline__ void load_voct(const __half* __restrict__ row,
+                                          int d0, int D, float f[8]) {
+    if (d0 + 8 <= D) {
+        const uint4 u = *reinterpret_cast<const uint4*>(row + d0);
+        const __half2* h2 = reinterpret_cast<const __half2*>(&u);
+        const float2 a = __half22float2(h2[0]);
+        const float2 b = __half22float2(h2[1]);
+        const float2 c = __half22float2(h2[2]);
+        const float2 d = __half22float2(h2[3]);
+        f[0] = a.x; f[1] = a.y; f[2] = b.x; f[3] = b.y;
+        f[4] = c.x; f[5] = c.y; f[6] = d.x; f[7] = d.y;
+    } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+            f[e] = (d0 + e < D) ? __half2float(row[d0 + e]) : 0.0f;
+    }
+}
+
 // FUSEQKV (decode only — every token its own sequence, so no block needs
 // another token's new KV row): the prologue sums the qkv split-K slabs
 // for this (token, head) slice, applies RoPE, appends the K/V cache row
@@ -480,8 +500,14 @@ __global__ void k_attention(
 
     float m = -INFINITY;  // running max (block-uniform)
     float l = 0.0f;       // running denom (block-uniform)
-    // per-(wave, lane) V partials: lane covers d = lane and lane + 64
-    float o = 0.0f, o2 = 0.0f;
+    // V partials: lane covers the d-octet (lane&15)*8 of rows
+    // jc ≡ (lane>>4) (mod 4) in its wave's quarter — one 16 B load per
+    // row-octet (2 B scalar loads were the deep-decode ceiling:
+    // ~2 B/cyc/CU issue-bound, half the step time; MI355X_MICROARCH.md
+    // per-instruction table)
+    float o8[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o8[e] = 0.0f;
     const int wid = threadIdx.x / WAVE;
 
     for (int j0 = 0; j0 < J; j0 += BLOCK) {
@@ -490,17 +516,13 @@ __global__ void k_attention(
         if (jj < J) {
             const __half* krow = k_cache + base + (size_t)jj * Ekv;
             float acc = 0.0f;
-            const int nh2 = D >> 2;  // 4 halves per float2 load
-            const float2* k2 = reinterpret_cast<const float2*>(krow);
-            for (int c = 0; c < nh2; ++c) {
-                const float2 raw = k2[c];
-                const __half2* hh = reinterpret_cast<const __half2*>(&raw);
-                const float2 a = __half22float2(hh[0]);
-                const float2 b = __half22float2(hh[1]);
-                acc = fmaf(a.x, lds_q[c * 4 + 0], acc);
-                acc = fmaf(a.y, lds_q[c * 4 + 1], acc);
-                acc = fmaf(b.x, lds_q[c * 4 + 2], acc);
-                acc = fmaf(b.y, lds_q[c * 4 + 3], acc);
+            for (int c = 0; c < D; c += 8) {  // 16 B per load
+                float f[8];
+                load_voct(krow, c, D, f);
+#pragma unroll
+                for (int e = 0; e < 8; ++e)
+                    acc = fmaf(f[e], (c + e < D) ? lds_q[c + e] : 0.0f,
+                               acc);
             }
             s = acc;
         }
@@ -522,47 +544,49 @@ __global__ void k_attention(
         const float sum_p = lds_red[0] + lds_red[1] + lds_red[2] + lds_red[3];
         l = l * alpha + sum_p;
         m = m_new;
-        // V accumulation: all 4 waves split the chunk's rows (the old
-        // thread-d-only form idled waves 2-3 and serialized a 256-fma
-        // dependency chain; this was HALF the deep-decode step —
-        // profiles/). Wave w covers rows [w*64, +64); lane covers d =
-        // lane and lane+64 (D <= 128); per-wave partials merge at the
-        // very end (alpha scaling is block-uniform, so partials scale
-        // consistently every chunk).
+        // V accumulation: wave w covers chunk rows [w*64, +64); within
+        // the wave, lane sub-group (lane>>4) takes every 4th row and
+        // the lane's d-octet rides ONE 16 B load per row. Partials per
+        // (wave, sub, octet) merge once at the kernel end; alpha is
+        // block-uniform so partials scale consistently each chunk.
         {
             const int jmax = min(BLOCK, J - j0);
             const int jq0 = wid * (BLOCK / NWAVES);
             const int jq1 = min(jmax, jq0 + BLOCK / NWAVES);
             const int lane = threadIdx.x & (WAVE - 1);
-            o *= alpha;
-            o2 *= alpha;
-            const __half* vrow0 =
-                v_cache + base + (size_t)(j0 + jq0) * Ekv;
-            const bool d2ok = (lane + WAVE) < D;
-            for (int jc = jq0; jc < jq1; ++jc) {
-                const float pj = lds_p[jc];
-                const __half* vr = vrow0 + (size_t)(jc - jq0) * Ekv;
-                if (lane < D) o = fmaf(pj, __half2float(vr[lane]), o);
-                if (d2ok)
-                    o2 = fmaf(pj, __half2float(vr[lane + WAVE]), o2);
+            const int sub = lane >> 4;
+            const int d0 = (lane & 15) * 8;
+#pragma unroll
+            for (int e = 0; e < 8; ++e) o8[e] *= alpha;
+            if (d0 < D) {
+                for (int jc = jq0 + sub; jc < jq1; jc += 4) {
+                    const float pj = lds_p[jc];
+                    float f[8];
+                    load_voct(v_cache + base + (size_t)(j0 + jc) * Ekv,
+                              d0, D, f);
+#pragma unroll
+                    for (int e = 0; e < 8; ++e)
+                        o8[e] = fmaf(pj, f[e], o8[e]);
+                }
             }
         }
         __syncthreads();
     }
-    // merge the 4 waves' o partials: wave w writes its (d, d+64) pair
+    // merge the 16 (wave, sub) partial groups
     {
         const int lane = threadIdx.x & (WAVE - 1);
-        float* lds_o = lds_red + NWAVES;  // [2][BLOCK] merge area
-        lds_o[wid * WAVE + lane] = o;
-        lds_o[BLOCK + wid * WAVE + lane] = o2;
+        float* lds_o = lds_red + NWAVES;  // [16][128] merge area
+        const int g = wid * 4 + (lane >> 4);
+        const int d0 = (lane & 15) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) lds_o[g * 128 + d0 + e] = o8[e];
         __syncthreads();
         if (threadIdx.x < D) {
             const int d = threadIdx.x;
             float v = 0.0f;
 #pragma unroll
-            for (int w = 0; w < NWAVES; ++w)
-                v += (d < WAVE) ? lds_o[w * WAVE + d]
-                                : lds_o[BLOCK + w * WAVE + (d - WAVE)];
+            for (int g2 = 0; g2 < 4 * NWAVES; ++g2)
+                v += lds_o[g2 * 128 + d];
             v /= l;
             const int e = h * D + d;
             out[(size_t)t * E + e] = v;
@@ -2445,7 +2469,8 @@ void launch_attention(hipStream_t s, const float* q_buf,
                       int n_ctx, const float* qkv_slab, int ks,
                       const float* inv_freq) {
     const dim3 grid(T, H);
-    const size_t lds = (D + BLOCK + NWAVES + 2 * BLOCK) * sizeof(float);
+    const size_t lds =
+        (D + BLOCK + NWAVES + 16 * 128) * sizeof(float);
     if (qkv_slab != nullptr) {
         hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
