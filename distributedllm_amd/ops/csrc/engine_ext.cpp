@@ -186,9 +186,16 @@ class SliceEngine {
         auto u16 = dev.dtype(torch::kInt16);
         // KV cache: [L, B, ctx, Ekv] f16 — the HBM3E-resident analog of
         // the reference's kv_cache_init (tensor_processor.cpp:1089-1132);
-        // Ekv = Hkv*D < E under GQA.
-        k_cache_ = torch::zeros({L_, B_, ctx_, EK_}, f16);
-        v_cache_ = torch::zeros({L_, B_, ctx_, EK_}, f16);
+        // Ekv = Hkv*D < E under GQA. 16 halves of tail pad let the
+        // attention kernels read whole 16 B octets unconditionally at
+        // the last row (no divergent tail-load branches).
+        const int64_t kv_elems = (int64_t)L_ * B_ * ctx_ * EK_;
+        kv_store_k_ = torch::zeros({kv_elems + 16}, f16);
+        kv_store_v_ = torch::zeros({kv_elems + 16}, f16);
+        k_cache_ = kv_store_k_.narrow(0, 0, kv_elems)
+                       .view({L_, B_, ctx_, EK_});
+        v_cache_ = kv_store_v_.narrow(0, 0, kv_elems)
+                       .view({L_, B_, ctx_, EK_});
         // RoPE pair frequencies: theta_i = pos * base^(-2i/D)
         inv_freq_ = torch::pow(
             (float)rope_base,
@@ -526,6 +533,7 @@ class SliceEngine {
     float eps_;
     std::vector<Layer> layers_;
     std::vector<bool> loaded_;
+    torch::Tensor kv_store_k_, kv_store_v_;
     torch::Tensor k_cache_, v_cache_, inv_freq_;
     torch::Tensor xn_, qb_, ab_, ffb_;
     torch::Tensor xprep_, aprep_, gprep_, ss_attn_, ss_ffn_, ss_tmp_;

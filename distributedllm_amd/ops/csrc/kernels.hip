@@ -398,24 +398,23 @@ __global__ void k_qkv_rope_append(
 // entries [0, pos[t]] of sequence seq[t] — causality for prefill rows comes
 // from their per-row positions.
 
-// 8 halves [d0, d0+8) of `row` as floats, zero-padded past D (guarded
-// tail so the last row of the cache never reads out of bounds)
+// 8 halves [d0, d0+8) of `row` as floats — UNCONDITIONAL 16 B load
+// (the KV allocation carries 16 halves of tail pad, and a lane-varying
+// tail branch would execute BOTH paths with exec masking — the guarded
+// first version added 8 masked scalar loads to every wide load and
+// measured slower than 2 B loads). Elements past D return whatever the
+// next row holds: always finite (the cache is zero-initialized and only
+// ever written with real values); callers discard d >= D.
 __device__ __forceinline__ void load_voct(const __half* __restrict__ row,
-                                          int d0, int D, float f[8]) {
-    if (d0 + 8 <= D) {
-        const uint4 u = *reinterpret_cast<const uint4*>(row + d0);
-        const __half2* h2 = reinterpret_cast<const __half2*>(&u);
-        const float2 a = __half22float2(h2[0]);
-        const float2 b = __half22float2(h2[1]);
-        const float2 c = __half22float2(h2[2]);
-        const float2 d = __half22float2(h2[3]);
-        f[0] = a.x; f[1] = a.y; f[2] = b.x; f[3] = b.y;
-        f[4] = c.x; f[5] = c.y; f[6] = d.x; f[7] = d.y;
-    } else {
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-            f[e] = (d0 + e < D) ? __half2float(row[d0 + e]) : 0.0f;
-    }
+                                          int d0, float f[8]) {
+    const uint4 u = *reinterpret_cast<const uint4*>(row + d0);
+    const __half2* h2 = reinterpret_cast<const __half2*>(&u);
+    const float2 a = __half22float2(h2[0]);
+    const float2 b = __half22float2(h2[1]);
+    const float2 c = __half22float2(h2[2]);
+    const float2 d = __half22float2(h2[3]);
+    f[0] = a.x; f[1] = a.y; f[2] = b.x; f[3] = b.y;
+    f[4] = c.x; f[5] = c.y; f[6] = d.x; f[7] = d.y;
 }
 
 // FUSEQKV (decode only — every token its own sequence, so no block needs
@@ -487,14 +486,20 @@ __global__ void k_attention(
             kv_k[d0] = __float2half(k0 * cs - k1 * sn);
             kv_k[d0 + 1] = __float2half(k0 * sn + k1 * cs);
         }
+        // zero-pad lds_q to the score loop's octet width
+        for (int d = D + (int)threadIdx.x; d < ((D + 7) & ~7); d += BLOCK)
+            lds_q[d] = 0.0f;
         // same-block visibility of the new K/V row: stores above are this
         // CU's own L1 write-through; __syncthreads orders them before the
         // scan loop's loads (cross-CU coherence is not needed — decode
         // tokens are distinct sequences, no other block reads this row)
         __syncthreads();
     } else {
-        for (int d = threadIdx.x; d < D; d += BLOCK)
-            lds_q[d] = q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d;
+        const int Dp = (D + 7) & ~7;  // scores read whole octets
+        for (int d = threadIdx.x; d < Dp; d += BLOCK)
+            lds_q[d] = (d < D)
+                ? q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d
+                : 0.0f;
         __syncthreads();
     }
 
@@ -518,11 +523,10 @@ __global__ void k_attention(
             float acc = 0.0f;
             for (int c = 0; c < D; c += 8) {  // 16 B per load
                 float f[8];
-                load_voct(krow, c, D, f);
+                load_voct(krow, c, f);
 #pragma unroll
                 for (int e = 0; e < 8; ++e)
-                    acc = fmaf(f[e], (c + e < D) ? lds_q[c + e] : 0.0f,
-                               acc);
+                    acc = fmaf(f[e], lds_q[c + e], acc);  // lds_q 0-padded
             }
             s = acc;
         }
@@ -563,7 +567,7 @@ __global__ void k_attention(
                     const float pj = lds_p[jc];
                     float f[8];
                     load_voct(v_cache + base + (size_t)(j0 + jc) * Ekv,
-                              d0, D, f);
+                              d0, f);
 #pragma unroll
                     for (int e = 0; e < 8; ++e)
                         o8[e] = fmaf(pj, f[e], o8[e]);
