@@ -515,21 +515,45 @@ __global__ void k_attention(
     for (int e = 0; e < 8; ++e) o8[e] = 0.0f;
     const int wid = threadIdx.x / WAVE;
 
+    const int lane0 = threadIdx.x & (WAVE - 1);
+    const int sub0 = lane0 >> 4;        // row within the 4-row group
+    const int oct0 = lane0 & 15;        // d-octet within the row
+    const int Dp8 = (D + 7) & ~7;
     for (int j0 = 0; j0 < J; j0 += BLOCK) {
-        const int jj = j0 + threadIdx.x;
-        float s = -INFINITY;
-        if (jj < J) {
-            const __half* krow = k_cache + base + (size_t)jj * Ekv;
+        // scores, row-major: each wave iteration reads 4 K rows with 16
+        // lanes per row (one contiguous 16 B octet each — the whole row
+        // is fetched by ONE instruction, so cache lines are fully used
+        // the moment they arrive; the old thread-per-row form touched 64
+        // scattered lines per instruction and thrashed L1). Partial dots
+        // reduce across the row's 16 lanes with 4 shuffles.
+        const int jmax0 = min(BLOCK, J - j0);
+        const int jw0 = wid * (BLOCK / NWAVES);
+        const int jw1 = min(jmax0, jw0 + BLOCK / NWAVES);
+        // uniform trip count: every lane runs the same iterations (the
+        // 16-lane shuffles below must stay converged)
+        const int ngrp = (jw1 - jw0 + 3) >> 2;
+        for (int gi = 0; gi < ngrp; ++gi) {
+            const int jc = jw0 + gi * 4 + sub0;
             float acc = 0.0f;
-            for (int c = 0; c < D; c += 8) {  // 16 B per load
+            if (jc < jw1 && oct0 * 8 < Dp8) {
+                const __half* krow =
+                    k_cache + base + (size_t)(j0 + jc) * Ekv;
                 float f[8];
-                load_voct(krow, c, f);
+                load_voct(krow, oct0 * 8, f);
 #pragma unroll
                 for (int e = 0; e < 8; ++e)
-                    acc = fmaf(f[e], lds_q[c + e], acc);  // lds_q 0-padded
+                    acc = fmaf(f[e], lds_q[oct0 * 8 + e], acc);
             }
-            s = acc;
+            // reduce over the 16 octet lanes of this row
+            acc += __shfl_xor(acc, 1);
+            acc += __shfl_xor(acc, 2);
+            acc += __shfl_xor(acc, 4);
+            acc += __shfl_xor(acc, 8);
+            if (oct0 == 0 && jc < jw1) lds_p[jc] = acc;
         }
+        __syncthreads();
+        const int jj = j0 + threadIdx.x;
+        float s = (threadIdx.x < jmax0) ? lds_p[threadIdx.x] : -INFINITY;
         // block max
         float wm = wave_reduce_max(s);
         if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = wm;
