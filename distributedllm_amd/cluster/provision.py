@@ -30,7 +30,14 @@ VALID_QUANT = {"f32": ggml.FTYPE_ALL_F32, "f16": ggml.FTYPE_MOSTLY_F16,
                "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
                "q5_0": ggml.FTYPE_MOSTLY_Q5_0,
                "q5_1": ggml.FTYPE_MOSTLY_Q5_1,
-               "q8_0": ggml.FTYPE_MOSTLY_Q8_0}
+               "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
+               # k-quant super-block formats (256-weight blocks);
+               # requires n_embd/n_ff multiples of 256
+               "q2_K": ggml.FTYPE_MOSTLY_Q2_K,
+               "q3_K": ggml.FTYPE_MOSTLY_Q3_K_M,
+               "q4_K": ggml.FTYPE_MOSTLY_Q4_K_M,
+               "q5_K": ggml.FTYPE_MOSTLY_Q5_K_M,
+               "q6_K": ggml.FTYPE_MOSTLY_Q6_K}
 VALID_FAMILY = {"llama_v1", "llama_v2"}
 
 
@@ -45,8 +52,20 @@ def validate_metadata(meta: dict) -> None:
             raise ValueError(f"metadata field {key}={v!r} is not a safe id")
 
 
+# upstream-style fallback for tensors whose row length is not a
+# multiple of QK_K=256 under a k-quant target (e.g. OpenLLaMA-3B
+# E=3200): sub-5-bit targets fall back to q5_0, the rest to q8_0
+_KQUANT_FALLBACK = {
+    ggml.GGML_TYPE_Q2_K: ggml.GGML_TYPE_Q5_0,
+    ggml.GGML_TYPE_Q3_K: ggml.GGML_TYPE_Q5_0,
+    ggml.GGML_TYPE_Q4_K: ggml.GGML_TYPE_Q5_0,
+    ggml.GGML_TYPE_Q5_K: ggml.GGML_TYPE_Q5_0,
+    ggml.GGML_TYPE_Q6_K: ggml.GGML_TYPE_Q8_0,
+}
+
+
 def requantize(f: ggml.GGMLFile, ftype: int) -> ggml.GGMLFile:
-    """f16/f32 -> q4_0/q4_1 (or dtype change) with the in-repo codecs;
+    """f16/f32 -> quantized (or dtype change) with the in-repo codecs;
     1-D tensors stay f32 as in real checkpoints."""
     target = ggml._FTYPE_TO_GGML[ftype]
     tensors = []
@@ -54,11 +73,15 @@ def requantize(f: ggml.GGMLFile, ftype: int) -> ggml.GGMLFile:
         if len(t.ne) == 1 or t.gtype == target:
             tensors.append(t)
         else:
+            gt = target
+            if gt in _KQUANT_FALLBACK and t.ne[0] % 256 != 0:
+                gt = _KQUANT_FALLBACK[gt]
             tensors.append(
-                ggml.GGMLTensor.from_f32(t.name, t.to_f32(), target))
+                ggml.GGMLTensor.from_f32(t.name, t.to_f32(), gt))
     hp = f.hparams
     new_hp = ggml.Hparams(hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head,
-                          hp.n_layer, hp.n_rot, ftype, hp.first_layer)
+                          hp.n_layer, hp.n_rot, ftype, hp.first_layer,
+                          n_head_kv=hp.n_head_kv)
     return ggml.GGMLFile(hparams=new_hp, vocab=list(f.vocab),
                          tensors=tensors)
 

@@ -42,6 +42,113 @@ def _nibbles(t: ggml.GGMLTensor) -> np.ndarray:
 
 _BYTE_GTYPES = (ggml.GGML_TYPE_Q5_0, ggml.GGML_TYPE_Q5_1,
                 ggml.GGML_TYPE_Q8_0)
+# k-quants with per-32 affine sub-blocks ride W_Q8B; per-16 formats
+# (q2_K/q3_K/q6_K) ride W_Q8B16 (two scale planes)
+_K32_GTYPES = (ggml.GGML_TYPE_Q4_K, ggml.GGML_TYPE_Q5_K)
+_K16_GTYPES = (ggml.GGML_TYPE_Q2_K, ggml.GGML_TYPE_Q3_K,
+               ggml.GGML_TYPE_Q6_K)
+W_Q8B16 = 9  # kernels.h WType
+
+
+def _kquant_byte_values(t: ggml.GGMLTensor):
+    """k-quant tensor -> (u8 values [rows, nb32, 32] for the byte
+    kernel's w = alpha*(u-128) + beta form, alpha, beta).
+
+    Per-32 formats return alpha/beta [rows, nb32]; per-16 formats
+    return [rows, nb32, 2] (two 16-weight half-planes).
+    """
+    from ..formats import kquants as KQ
+    rows, cols = t.shape_rows_cols
+    nsb = cols // KQ.QK_K
+    u8 = np.frombuffer(t.raw, np.uint8)
+    if t.gtype == ggml.GGML_TYPE_Q4_K:
+        b = u8.reshape(rows, nsb, KQ.Q4_K_BLOCK_BYTES)
+        d = np.ascontiguousarray(b[:, :, 0:2]).view(np.float16)
+        dmin = np.ascontiguousarray(b[:, :, 2:4]).view(np.float16)
+        d = d.reshape(rows, nsb).astype(np.float32)
+        dmin = dmin.reshape(rows, nsb).astype(np.float32)
+        sc, mn = KQ._unpack_scales_k4(b[:, :, 4:16])
+        qs = b[:, :, 16:144].reshape(rows, nsb, 4, 32)
+        q = np.empty((rows, nsb, 8, 32), dtype=np.uint8)
+        q[:, :, 0::2] = qs & 0xF
+        q[:, :, 1::2] = qs >> 4
+        alpha = (d[..., None] * sc).reshape(rows, nsb * 8)
+        beta = (-(dmin[..., None] * mn)).reshape(rows, nsb * 8)
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        return vals, alpha.astype(np.float32), beta.astype(np.float32)
+    if t.gtype == ggml.GGML_TYPE_Q5_K:
+        b = u8.reshape(rows, nsb, KQ.Q5_K_BLOCK_BYTES)
+        d = np.ascontiguousarray(b[:, :, 0:2]).view(np.float16)
+        dmin = np.ascontiguousarray(b[:, :, 2:4]).view(np.float16)
+        d = d.reshape(rows, nsb).astype(np.float32)
+        dmin = dmin.reshape(rows, nsb).astype(np.float32)
+        sc, mn = KQ._unpack_scales_k4(b[:, :, 4:16])
+        qh = b[:, :, 16:48]
+        ql = b[:, :, 48:176].reshape(rows, nsb, 4, 32)
+        q = np.empty((rows, nsb, 8, 32), dtype=np.uint8)
+        for j in range(4):
+            q[:, :, 2 * j] = (ql[:, :, j] & 0xF) | \
+                (((qh >> (2 * j)) & 1) << 4)
+            q[:, :, 2 * j + 1] = (ql[:, :, j] >> 4) | \
+                (((qh >> (2 * j + 1)) & 1) << 4)
+        alpha = (d[..., None] * sc).reshape(rows, nsb * 8)
+        beta = (-(dmin[..., None] * mn)).reshape(rows, nsb * 8)
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        return vals, alpha.astype(np.float32), beta.astype(np.float32)
+    if t.gtype == ggml.GGML_TYPE_Q6_K:
+        b = u8.reshape(rows, nsb, KQ.Q6_K_BLOCK_BYTES)
+        ql = b[:, :, 0:128].reshape(rows, nsb, 2, 2, 32)
+        qh = b[:, :, 128:192].reshape(rows, nsb, 2, 32)
+        sc = np.ascontiguousarray(b[:, :, 192:208]).view(np.int8)
+        sc = sc.reshape(rows, nsb, 16).astype(np.float32)
+        d = np.ascontiguousarray(b[:, :, 208:210]).view(np.float16)
+        d = d.reshape(rows, nsb).astype(np.float32)
+        q = np.empty((rows, nsb, 2, 4, 32), dtype=np.int16)
+        q[:, :, :, 0] = (ql[:, :, :, 0] & 0xF) | (((qh >> 0) & 3) << 4)
+        q[:, :, :, 1] = (ql[:, :, :, 1] & 0xF) | (((qh >> 2) & 3) << 4)
+        q[:, :, :, 2] = (ql[:, :, :, 0] >> 4) | (((qh >> 4) & 3) << 4)
+        q[:, :, :, 3] = (ql[:, :, :, 1] >> 4) | (((qh >> 6) & 3) << 4)
+        # u = (q - 32) + 128; alpha16 = d*sc, beta16 = 0
+        vals = (q + 96).astype(np.uint8).reshape(rows, nsb * 8, 32)
+        alpha16 = (d[..., None] * sc).reshape(rows, nsb, 2, 4, 2)
+        alpha16 = alpha16.transpose(0, 1, 2, 3, 4).reshape(
+            rows, nsb * 8, 2)
+        beta16 = np.zeros_like(alpha16)
+        return vals, alpha16.astype(np.float32), beta16.astype(np.float32)
+    if t.gtype == ggml.GGML_TYPE_Q2_K:
+        b = u8.reshape(rows, nsb, KQ.Q2_K_BLOCK_BYTES)
+        sc = (b[:, :, 0:16] & 0xF).astype(np.float32)
+        mn = (b[:, :, 0:16] >> 4).astype(np.float32)
+        qs = b[:, :, 16:80].reshape(rows, nsb, 2, 32)
+        d = np.ascontiguousarray(b[:, :, 80:82]).view(np.float16)
+        dmin = np.ascontiguousarray(b[:, :, 82:84]).view(np.float16)
+        d = d.reshape(rows, nsb).astype(np.float32)
+        dmin = dmin.reshape(rows, nsb).astype(np.float32)
+        q = np.empty((rows, nsb, 2, 4, 32), dtype=np.uint8)
+        for j in range(4):
+            q[:, :, :, j] = (qs >> (2 * j)) & 3
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        alpha16 = (d[..., None] * sc).reshape(rows, nsb * 8, 2)
+        beta16 = (-(dmin[..., None] * mn)).reshape(rows, nsb * 8, 2)
+        return vals, alpha16.astype(np.float32), beta16.astype(np.float32)
+    assert t.gtype == ggml.GGML_TYPE_Q3_K, t.gtype
+    b = u8.reshape(rows, nsb, KQ.Q3_K_BLOCK_BYTES)
+    hm = b[:, :, 0:32]
+    qs = b[:, :, 32:96].reshape(rows, nsb, 2, 32)
+    from ..formats.kquants import _unpack_scales_q3
+    sc = _unpack_scales_q3(b[:, :, 96:108]).astype(np.float32)
+    d = np.ascontiguousarray(b[:, :, 108:110]).view(np.float16)
+    d = d.reshape(rows, nsb).astype(np.float32)
+    q = np.empty((rows, nsb, 2, 4, 32), dtype=np.int16)
+    for half in range(2):
+        for j in range(4):
+            low = ((qs[:, :, half] >> (2 * j)) & 3).astype(np.int16)
+            hi = ((hm >> (half * 4 + j)) & 1).astype(np.int16)
+            q[:, :, half, j] = low - np.where(hi != 0, 0, 4)
+    vals = (q + 128).astype(np.uint8).reshape(rows, nsb * 8, 32)
+    alpha16 = (d[..., None] * sc).reshape(rows, nsb * 8, 2)
+    beta16 = np.zeros_like(alpha16)
+    return vals, alpha16.astype(np.float32), beta16.astype(np.float32)
 
 
 def _byte_values(t: ggml.GGMLTensor):
@@ -147,10 +254,15 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             np.concatenate([scales.reshape(-1),
                             np.zeros(128, np.float16)])).to(device)
         return data, sc, t.gtype
-    if t.gtype in _BYTE_GTYPES:
+    if t.gtype in _BYTE_GTYPES or t.gtype in _K32_GTYPES or \
+            t.gtype in _K16_GTYPES:
         nb = cols // 32
         nbp = (nb + 3) & ~3
-        vals, alpha, beta = _byte_values(t)
+        per16 = t.gtype in _K16_GTYPES
+        if t.gtype in _BYTE_GTYPES:
+            vals, alpha, beta = _byte_values(t)
+        else:
+            vals, alpha, beta = _kquant_byte_values(t)
         # byte order per lane-block 8 weights: [w0,w2,w1,w3, w4,w6,w5,w7]
         # so the kernel's (q & 0x00FF00FF) / (q>>8 & ..) masks yield the
         # (even, odd) f16 pairs of the A fragment (a_frag_q8)
@@ -164,14 +276,28 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
         qs2 = np.ascontiguousarray(
             q32.reshape(R, 16, nbp // 4, 4, 4, 2)   # [R][i][g4][kb][ws][2]
             .transpose(0, 2, 4, 1, 3, 5))           # [R][g4][ws][i][kb][2]
-        scales = _pack_grouped_scales(alpha, beta, rows, nb, nbp, R)
+        if per16:
+            # two (alpha, beta) half-planes per block:
+            # [R][g4][2 hp][16 i][4 kb][2 f16]
+            al = np.zeros((rows, nbp, 2), dtype=np.float32)
+            be = np.zeros((rows, nbp, 2), dtype=np.float32)
+            al[:, :nb] = alpha
+            be[:, :nb] = beta
+            ab = np.stack([al, be], axis=-1).astype(np.float16)
+            scales = np.ascontiguousarray(
+                ab.reshape(R, 16, nbp // 4, 4, 2, 2)
+                .transpose(0, 2, 4, 1, 3, 5))
+            wt_out = W_Q8B16
+        else:
+            scales = _pack_grouped_scales(alpha, beta, rows, nb, nbp, R)
+            wt_out = ggml.GGML_TYPE_Q8_0  # = W_Q8B
         data = torch.from_numpy(
             np.concatenate([qs2.view(np.int32).reshape(-1),
                             np.zeros(512, np.int32)])).to(device)
         sc = torch.from_numpy(
             np.concatenate([scales.reshape(-1),
                             np.zeros(128, np.float16)])).to(device)
-        return data, sc, ggml.GGML_TYPE_Q8_0  # W_Q8B for all byte formats
+        return data, sc, wt_out
     if t.gtype == ggml.GGML_TYPE_F16:
         w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
         tile = np.ascontiguousarray(
@@ -210,9 +336,10 @@ def _upload_mat(t: ggml.GGMLTensor, device: str):
         d = torch.from_numpy(qs).to(device)
         s = torch.from_numpy(scales.copy()).to(device)
         return d, s, t.gtype
-    if t.gtype in _BYTE_GTYPES:
-        # embedding gather for q5/q8 tables: dequantize once to f16 (the
-        # gather kernel has no byte path; exact dequant, 2 B/weight)
+    if t.gtype in _BYTE_GTYPES or t.gtype in _K32_GTYPES or \
+            t.gtype in _K16_GTYPES:
+        # embedding gather for q5/q8/k-quant tables: dequantize once to
+        # f16 (the gather kernel has no byte path; 2 B/weight)
         a = t.to_f32().astype(np.float16)
         return (torch.from_numpy(a).to(device), torch.empty(0),
                 ggml.GGML_TYPE_F16)
@@ -323,7 +450,25 @@ class HIPSliceEngine:
             # random weights directly in the MFMA tile layouts — any random
             # bit pattern is a valid q4 nibble word, so this is byte-for-
             # byte the same compute/HBM traffic as a real checkpoint
-            if wt in _BYTE_GTYPES:
+            if wt in _K16_GTYPES:
+                # byte stream + per-16 (alpha, beta) planes (W_Q8B16)
+                R, nb = rows // 16, cols // 32
+                nbp = (nb + 3) & ~3
+                data = torch.randint(-2**31, 2**31 - 1,
+                                     (R * nbp * 128 + 512,),
+                                     dtype=torch.int32, device="cuda",
+                                     generator=g)
+                alpha = ((torch.rand(rows, nbp, 2, device="cuda",
+                                     generator=g) * 0.5 + 0.75) * 0.003)
+                alpha[:, nb:] = 0.0
+                beta = torch.zeros_like(alpha)
+                ab = torch.stack([alpha, beta], dim=-1).to(torch.float16)
+                ab = (ab.reshape(R, 16, nbp // 4, 4, 2, 2)
+                      .permute(0, 2, 4, 1, 3, 5).contiguous().reshape(-1))
+                ab = torch.cat([ab, torch.zeros(128, dtype=torch.float16,
+                                                device="cuda")])
+                return data, ab, W_Q8B16
+            if wt in _BYTE_GTYPES or wt in _K32_GTYPES:
                 R, nb = rows // 16, cols // 32
                 nbp = (nb + 3) & ~3
                 data = torch.randint(-2**31, 2**31 - 1,

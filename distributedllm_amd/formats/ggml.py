@@ -29,7 +29,7 @@ from typing import BinaryIO, Dict, List, Optional, Tuple
 
 import numpy as np
 
-from . import q4
+from . import kquants, q4
 
 GGJT_MAGIC = 0x67676A74  # bytes 'tjgg' little-endian == "ggjt"
 GGJT_VERSION = 3
@@ -50,6 +50,12 @@ GGML_TYPE_Q4_1 = 3
 GGML_TYPE_Q5_0 = 6   # ids 4/5 were the removed q4_2/q4_3
 GGML_TYPE_Q5_1 = 7
 GGML_TYPE_Q8_0 = 8
+# k-quant super-block types (256-weight blocks; ids of the reference era)
+GGML_TYPE_Q2_K = 10
+GGML_TYPE_Q3_K = 11
+GGML_TYPE_Q4_K = 12
+GGML_TYPE_Q5_K = 13
+GGML_TYPE_Q6_K = 14
 
 # model-level ftype (llama_ftype ids; 4 = Q4_1_SOME_F16, 5/6 removed)
 FTYPE_ALL_F32 = 0
@@ -59,6 +65,18 @@ FTYPE_MOSTLY_Q4_1 = 3
 FTYPE_MOSTLY_Q8_0 = 7
 FTYPE_MOSTLY_Q5_0 = 8
 FTYPE_MOSTLY_Q5_1 = 9
+# k-quant ftypes (recognized by the reference's loader era,
+# tensor_processor.cpp:846-855); the S/M/L variants differ only in which
+# tensors upstream mixes — we map each to its base k-quant type
+FTYPE_MOSTLY_Q2_K = 10
+FTYPE_MOSTLY_Q3_K_S = 11
+FTYPE_MOSTLY_Q3_K_M = 12
+FTYPE_MOSTLY_Q3_K_L = 13
+FTYPE_MOSTLY_Q4_K_S = 14
+FTYPE_MOSTLY_Q4_K_M = 15
+FTYPE_MOSTLY_Q5_K_S = 16
+FTYPE_MOSTLY_Q5_K_M = 17
+FTYPE_MOSTLY_Q6_K = 18
 
 _FTYPE_TO_GGML = {
     FTYPE_ALL_F32: GGML_TYPE_F32,
@@ -68,6 +86,15 @@ _FTYPE_TO_GGML = {
     FTYPE_MOSTLY_Q8_0: GGML_TYPE_Q8_0,
     FTYPE_MOSTLY_Q5_0: GGML_TYPE_Q5_0,
     FTYPE_MOSTLY_Q5_1: GGML_TYPE_Q5_1,
+    FTYPE_MOSTLY_Q2_K: GGML_TYPE_Q2_K,
+    FTYPE_MOSTLY_Q3_K_S: GGML_TYPE_Q3_K,
+    FTYPE_MOSTLY_Q3_K_M: GGML_TYPE_Q3_K,
+    FTYPE_MOSTLY_Q3_K_L: GGML_TYPE_Q3_K,
+    FTYPE_MOSTLY_Q4_K_S: GGML_TYPE_Q4_K,
+    FTYPE_MOSTLY_Q4_K_M: GGML_TYPE_Q4_K,
+    FTYPE_MOSTLY_Q5_K_S: GGML_TYPE_Q5_K,
+    FTYPE_MOSTLY_Q5_K_M: GGML_TYPE_Q5_K,
+    FTYPE_MOSTLY_Q6_K: GGML_TYPE_Q6_K,
 }
 
 TYPE_NAMES = {
@@ -78,6 +105,11 @@ TYPE_NAMES = {
     GGML_TYPE_Q5_0: "q5_0",
     GGML_TYPE_Q5_1: "q5_1",
     GGML_TYPE_Q8_0: "q8_0",
+    GGML_TYPE_Q2_K: "q2_K",
+    GGML_TYPE_Q3_K: "q3_K",
+    GGML_TYPE_Q4_K: "q4_K",
+    GGML_TYPE_Q5_K: "q5_K",
+    GGML_TYPE_Q6_K: "q6_K",
 }
 
 # block codecs for the quantized types: (block bytes, quantize, dequantize)
@@ -92,6 +124,15 @@ _BLOCK_CODECS = {
                      q4.dequantize_q5_1),
     GGML_TYPE_Q8_0: (q4.Q8_0_BLOCK_BYTES, q4.quantize_q8_0,
                      q4.dequantize_q8_0),
+}
+
+# k-quant codecs use 256-weight super-blocks
+_KBLOCK_CODECS = {
+    GGML_TYPE_Q2_K: kquants.CODECS["q2_K"],
+    GGML_TYPE_Q3_K: kquants.CODECS["q3_K"],
+    GGML_TYPE_Q4_K: kquants.CODECS["q4_K"],
+    GGML_TYPE_Q5_K: kquants.CODECS["q5_K"],
+    GGML_TYPE_Q6_K: kquants.CODECS["q6_K"],
 }
 
 
@@ -110,6 +151,12 @@ def tensor_nbytes(gtype: int, ne: Tuple[int, ...]) -> int:
             raise ValueError(
                 f"{TYPE_NAMES[gtype]} row length {n0} not a multiple of 32")
         return (n0 // q4.QK4) * _BLOCK_CODECS[gtype][0] * rows
+    if gtype in _KBLOCK_CODECS:
+        if n0 % kquants.QK_K:
+            raise ValueError(
+                f"{TYPE_NAMES[gtype]} row length {n0} not a multiple "
+                f"of {kquants.QK_K}")
+        return (n0 // kquants.QK_K) * _KBLOCK_CODECS[gtype][0] * rows
     raise ValueError(f"unsupported ggml type {gtype}")
 
 
@@ -179,6 +226,9 @@ class GGMLTensor:
         elif self.gtype in _BLOCK_CODECS:
             u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
             a = _BLOCK_CODECS[self.gtype][2](u, cols)
+        elif self.gtype in _KBLOCK_CODECS:
+            u = np.frombuffer(self.raw, dtype=np.uint8).reshape(rows, -1)
+            a = _KBLOCK_CODECS[self.gtype][2](u, cols)
         else:
             raise ValueError(f"unsupported ggml type {self.gtype}")
         if len(self.ne) == 1:
@@ -201,6 +251,8 @@ class GGMLTensor:
             raw = a.astype(np.float16).tobytes()
         elif gtype in _BLOCK_CODECS:
             raw = _BLOCK_CODECS[gtype][1](a).tobytes()
+        elif gtype in _KBLOCK_CODECS:
+            raw = _KBLOCK_CODECS[gtype][1](a).tobytes()
         else:
             raise ValueError(f"unsupported ggml type {gtype}")
         return cls(name=name, ne=ne, gtype=gtype, raw=raw)

@@ -92,6 +92,8 @@ PRESETS: Dict[str, LlamaPreset] = {
     # mid-size test config: large enough that the split-K/RT kernel paths
     # run with realistic grids (E/16=32 tiles, F=1408)
     "small": LlamaPreset("small", 512, 512, 64, 8, 2),
+    # k-quant test config: E and F both multiples of QK_K=256 (F=1536)
+    "small_k": LlamaPreset("small_k", 512, 512, 512, 8, 2),
 }
 
 

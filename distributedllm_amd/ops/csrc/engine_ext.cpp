@@ -98,7 +98,8 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
                     "+ slack");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
-    } else if (wtype == W_Q8B) {
+    } else if (wtype == W_Q8B || wtype == W_Q8B16) {
+        const int64_t abw = (wtype == W_Q8B16) ? 2 : 1;  // scale planes
         TORCH_CHECK(data.scalar_type() == torch::kInt32 &&
                         data.numel() == R * nbp * 128 + kTailSlackQ8,
                     "byte-quant tiled data must be u32[R][nbp/4][4][16][8]"
@@ -106,9 +107,10 @@ DevMat2 make_devmat2(torch::Tensor data, torch::Tensor scales, int64_t wtype,
         TORCH_CHECK(scales.defined() && scales.is_cuda() &&
                         scales.is_contiguous() &&
                         scales.scalar_type() == torch::kFloat16 &&
-                        scales.numel() == R * nbp * 16 * 2 + kTailSlackAB,
-                    "byte-quant scales must be f16 (a,b)[R][nbp/4][16][4]"
-                    " + slack");
+                        scales.numel() ==
+                            R * nbp * 16 * 2 * abw + kTailSlackAB,
+                    "byte-quant scales must be f16 (a,b)[R][nbp/4]"
+                    "[planes][16][4] + slack");
         m.scales = scales;
         m.w.scales = scales.data_ptr();
     } else {

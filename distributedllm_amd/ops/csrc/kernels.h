@@ -6,9 +6,13 @@
 #include <stdint.h>
 
 // W_Q8B is the BYTE-stream quant path: q8_0 weights re-biased to u8 at
-// repack, and q5_0/q5_1 expanded to re-biased bytes — all three share
-// one kernel form (w = alpha*((1024+u) - 1152) + beta in packed f16).
-enum WType { W_F32 = 0, W_F16 = 1, W_Q4_0 = 2, W_Q4_1 = 3, W_Q8B = 8 };
+// repack, and q5_0/q5_1/q4_K/q5_K expanded to re-biased bytes — all
+// share one kernel form (w = alpha*((1024+u) - 1152) + beta in packed
+// f16, (alpha, beta) per 32-weight block). W_Q8B16 is the same byte
+// stream with per-16-weight (alpha, beta) planes — the k-quant formats
+// whose sub-block scales have 16-weight granularity (q2_K/q3_K/q6_K).
+enum WType { W_F32 = 0, W_F16 = 1, W_Q4_0 = 2, W_Q4_1 = 3, W_Q8B = 8,
+             W_Q8B16 = 9 };
 
 // One weight matrix resident in HBM, repacked SoA (see kernels.hip header).
 struct WMat {

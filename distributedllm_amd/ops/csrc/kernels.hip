@@ -1355,7 +1355,10 @@ __device__ __forceinline__ void wave_tile_kloop(
     const int nb0 = ws[0]->cols >> 5;
     const int nb = (WT == W_F16) ? nb0 : ((nb0 + 3) & ~3);  // padded count
     // u32 per lane per K-block: 1 for q4 nibbles, 2 for the byte stream
-    constexpr int QW = (WT == W_Q8B) ? 2 : 1;
+    constexpr int QW = (WT == W_Q8B || WT == W_Q8B16) ? 2 : 1;
+    // scale planes: W_Q8B16 carries one (alpha, beta) pair per 16
+    // weights — lane k-span ks selects its half-plane (ks >> 1)
+    constexpr int ABW = (WT == W_Q8B16) ? 2 : 1;
     const f32x4 zero = {0.f, 0.f, 0.f, 0.f};
     // With >=4 independent MFMAs per K-block (JT/NM/RT product) the
     // accumulator reuse distance already covers the MFMA dependent
@@ -1397,7 +1400,8 @@ __device__ __forceinline__ void wave_tile_kloop(
                         ((size_t)tr * nb + kl.kb0) * 64 * QW +
                         (kl.ks * 16 + kl.i) * 4 * QW;
             abp[rt][n] = (const uint32_t*)ws[n]->scales +
-                         ((size_t)tr * nb + kl.kb0) * 16 + kl.i * 4;
+                         ((size_t)tr * nb + kl.kb0) * 16 * ABW +
+                         (ABW == 2 ? (kl.ks >> 1) * 64 : 0) + kl.i * 4;
             tp[rt][n] = (const unsigned short*)ws[n]->data +
                         ((size_t)tr * (ws[n]->cols >> 3)) * 128 +
                         ((size_t)(kl.kb0 * 4 + kl.ks) * 16 + kl.i) * 8;
@@ -1442,7 +1446,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                                         qp[rt][n]) + u4 * 64 * QW + qi);
                         bt.ab[u4][rt][n] = __builtin_nontemporal_load(
                             reinterpret_cast<const u32x4*>(abp[rt][n]) +
-                            u4 * 16);
+                            u4 * 16 * ABW);
                     }
                 }
         }
@@ -1454,7 +1458,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                     tp[rt][n] += PF * 512;
                 } else {
                     qp[rt][n] += PF * 64 * QW;
-                    abp[rt][n] += PF * 16;
+                    abp[rt][n] += PF * 16 * ABW;
                 }
             }
     };
@@ -1504,7 +1508,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                 if (WT == W_F16) {
                     a.u[0] = aw[rt][n].x; a.u[1] = aw[rt][n].y;
                     a.u[2] = aw[rt][n].z; a.u[3] = aw[rt][n].w;
-                } else if (WT == W_Q8B) {
+                } else if (WT == W_Q8B || WT == W_Q8B16) {
                     a_frag_q8(q[rt][n][0], q[rt][n][QW - 1], ab[rt][n],
                               a);
                 } else {
@@ -1528,7 +1532,7 @@ __device__ __forceinline__ void wave_tile_kloop(
             for (int rt = 0; rt < RT; ++rt)
 #pragma unroll
                 for (int n = 0; n < NM; ++n) {
-                    if (WT == W_Q8B) {
+                    if (WT == W_Q8B || WT == W_Q8B16) {
                         // lane's 8 u32 per group: [blk0.A, blk0.B, ...]
                         q[rt][n][0] = bt.q[u / 2][rt][n][(u % 2) * 2];
                         q[rt][n][QW - 1] =
@@ -1576,7 +1580,7 @@ __device__ __forceinline__ void wave_tile_kloop(
                     tp[rt][n] -= PF * 512;
                 } else {
                     qp[rt][n] -= PF * 64 * QW;
-                    abp[rt][n] -= PF * 16;
+                    abp[rt][n] -= PF * 16 * ABW;
                 }
             }
     }
@@ -2560,6 +2564,11 @@ void launch_prep_x(hipStream_t s, const float* x, unsigned short* xprep,
         }                                        \
         case W_Q8B: {                            \
             constexpr int WTc = W_Q8B;           \
+            __VA_ARGS__;                         \
+            break;                               \
+        }                                        \
+        case W_Q8B16: {                          \
+            constexpr int WTc = W_Q8B16;         \
             __VA_ARGS__;                         \
             break;                               \
         }                                        \

@@ -561,3 +561,42 @@ def test_gqa_random_engine_runs():
     y = eng.forward(x, pos, seq, decode=True)
     lg = eng.logits(y, all_logits=True)
     assert torch.isfinite(y).all() and torch.isfinite(lg).all()
+
+
+@pytest.mark.parametrize("ftype", [ggml.FTYPE_MOSTLY_Q4_K_M,
+                                   ggml.FTYPE_MOSTLY_Q5_K_M,
+                                   ggml.FTYPE_MOSTLY_Q6_K,
+                                   ggml.FTYPE_MOSTLY_Q3_K_M,
+                                   ggml.FTYPE_MOSTLY_Q2_K])
+def test_kquant_hip_engine_matches_cpu(ftype):
+    """k-quant weights through the MFMA byte path: q4_K/q5_K expand to
+    per-32 (alpha, beta) byte blocks (W_Q8B); q2_K/q3_K/q6_K to the
+    per-16 two-plane form (W_Q8B16). Decode, fused decode and the
+    native large-T prefill all must match the fp32 reference of the
+    same dequantized weights."""
+    f, hip, cpu = _engines(preset="small_k", ftype=ftype, n_ctx=256,
+                           max_batch=2)
+    hp = f.hparams
+    torch.manual_seed(31)
+    T = 24
+    x = torch.randn(T, hp.n_embd) * 0.5
+    pos = torch.arange(T, dtype=torch.int32)
+    seq = torch.zeros(T, dtype=torch.int32)
+    y_gpu = hip.forward(x.cuda(), pos.cuda(), seq.cuda()).cpu()
+    y_cpu = cpu.forward(x.clone(), pos, seq)
+    _assert_close(y_gpu, y_cpu, label=f"kquant prefill ftype={ftype}")
+    # fused batched decode
+    xd = torch.randn(2, hp.n_embd) * 0.5
+    pd = torch.tensor([T, 0], dtype=torch.int32)
+    sd = torch.arange(2, dtype=torch.int32)
+    y2 = hip.forward(xd.cuda(), pd.cuda(), sd.cuda(), decode=True).cpu()
+    y2c = cpu.forward(xd.clone(), pd, sd)
+    _assert_close(y2, y2c, label=f"kquant decode ftype={ftype}")
+    # native large-T prefill on sequence 1
+    T2 = 100
+    x3 = torch.randn(T2, hp.n_embd) * 0.5
+    p3 = torch.arange(1, 1 + T2, dtype=torch.int32)
+    s3 = torch.ones(T2, dtype=torch.int32)
+    y3 = hip.forward(x3.cuda(), p3.cuda(), s3.cuda()).cpu()
+    y3c = cpu.forward(x3.clone(), p3, s3)
+    _assert_close(y3, y3c, label=f"kquant native prefill ftype={ftype}")

@@ -35,7 +35,13 @@ constexpr uint32_t kVersionGqa = 4;  // framework GQA extension
 constexpr uint32_t kExtraLayersFirstLayer = 0xFFFFFFFFu;
 
 enum GType : uint32_t { F32 = 0, F16 = 1, Q4_0 = 2, Q4_1 = 3,
-                        Q5_0 = 6, Q5_1 = 7, Q8_0 = 8 };
+                        Q5_0 = 6, Q5_1 = 7, Q8_0 = 8,
+                        // k-quant super-block types (QK_K = 256); the
+                        // tools pass their bytes through (codec:
+                        // formats/kquants.py)
+                        Q2_K = 10, Q3_K = 11, Q4_K = 12, Q5_K = 13,
+                        Q6_K = 14 };
+constexpr uint32_t kQK_K = 256;
 constexpr int kQK = 32;               // weights per quant block
 constexpr int kQ4_0Bytes = 18;        // f16 d + 16 nibble bytes
 constexpr int kQ4_1Bytes = 20;        // f16 d + f16 m + 16 nibble bytes
@@ -122,6 +128,21 @@ inline size_t row_bytes(GType t, uint32_t ne0) {
         case Q8_0:
             if (ne0 % kQK) throw std::runtime_error("q8_0 row not /32");
             return (size_t)(ne0 / kQK) * kQ8_0Bytes;
+        case Q2_K:
+            if (ne0 % kQK_K) throw std::runtime_error("q2_K row not /256");
+            return (size_t)(ne0 / kQK_K) * 84;
+        case Q3_K:
+            if (ne0 % kQK_K) throw std::runtime_error("q3_K row not /256");
+            return (size_t)(ne0 / kQK_K) * 110;
+        case Q4_K:
+            if (ne0 % kQK_K) throw std::runtime_error("q4_K row not /256");
+            return (size_t)(ne0 / kQK_K) * 144;
+        case Q5_K:
+            if (ne0 % kQK_K) throw std::runtime_error("q5_K row not /256");
+            return (size_t)(ne0 / kQK_K) * 176;
+        case Q6_K:
+            if (ne0 % kQK_K) throw std::runtime_error("q6_K row not /256");
+            return (size_t)(ne0 / kQK_K) * 210;
     }
     throw std::runtime_error("unknown ggml type");
 }
