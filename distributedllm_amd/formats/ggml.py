@@ -201,7 +201,8 @@ class GGMLTensor:
     name: str
     ne: Tuple[int, ...]          # ne[0] = contiguous/input dim
     gtype: int
-    raw: bytes                   # on-disk bytes
+    raw: bytes                   # on-disk bytes (or a zero-copy
+                                 # memoryview into an mmap'd file)
 
     @property
     def nbytes(self) -> int:
@@ -278,49 +279,47 @@ class GGMLFile:
             self._write(f)
 
     def _write(self, f: BinaryIO) -> None:
-        hp = self.hparams
-        gqa = hp.is_gqa
-        f.write(struct.pack("<II", GGJT_MAGIC,
-                            GGJT_VERSION_GQA if gqa else GGJT_VERSION))
-        fields = [hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head]
-        if gqa:
-            fields.append(hp.kv_heads)
-        fields += [hp.n_layer, hp.n_rot]
-        if hp.first_layer is not None:
-            fields.append(hp.first_layer)
-        fields.append(hp.ftype)
-        f.write(struct.pack("<%dI" % len(fields), *fields))
-        if len(self.vocab) != hp.n_vocab:
-            raise ValueError(
-                f"vocab has {len(self.vocab)} entries, hparams say {hp.n_vocab}")
-        for word, score in self.vocab:
-            f.write(struct.pack("<I", len(word)))
-            f.write(word)
-            f.write(struct.pack("<f", score))
+        write_header(f, self.hparams, self.vocab)
         for t in self.tensors:
-            name_b = t.name.encode("utf-8")
-            f.write(struct.pack("<III", len(t.ne), len(name_b), t.gtype))
-            f.write(struct.pack("<%dI" % len(t.ne), *t.ne))
-            f.write(name_b)
-            pad = -f.tell() & 31
-            f.write(b"\x00" * pad)
-            if len(t.raw) != t.nbytes:
-                raise ValueError(
-                    f"tensor {t.name}: raw {len(t.raw)} B != expected {t.nbytes} B")
-            f.write(t.raw)
+            write_tensor_record(f, t)
 
     # ---------------- reading ----------------
 
     @classmethod
-    def load(cls, path: str, extended: bool,
-             with_data: bool = True) -> "GGMLFile":
-        """Read a GGJT v3 file.
+    def load(cls, path: str, extended: bool, with_data: bool = True,
+             use_mmap: bool = True) -> "GGMLFile":
+        """Read a GGJT v3/v4 file.
 
         ``extended=True`` for slice/extra_layers files (8-field hparams),
         ``False`` for original model files (7-field).
+
+        ``use_mmap`` (default): tensor bytes are zero-copy views into a
+        file mapping — the analog of the reference's mmap load
+        (tensor_processor.cpp:996-1074). Host RSS stays bounded by the
+        working set (the OS pages tensor data in on first touch and may
+        evict it), which is what makes 100 GB-class checkpoints loadable
+        through the per-tensor streaming repack. The mapping lives as
+        long as the GGMLFile (tensors hold memoryviews into it).
         """
+        if use_mmap and with_data:
+            import mmap as _mmap
+            f = open(path, "rb")
+            try:
+                mm = _mmap.mmap(f.fileno(), 0, access=_mmap.ACCESS_READ)
+            except ValueError:  # empty file
+                f.close()
+                raise ValueError(f"truncated GGJT file {path!r} (empty)")
+            out = cls._parse(path, memoryview(mm), extended, True)
+            out._mmap = mm      # keep the mapping alive
+            out._mmap_file = f
+            return out
         with open(path, "rb") as f:
             data = f.read()
+        return cls._parse(path, data, extended, with_data)
+
+    @classmethod
+    def _parse(cls, path: str, data, extended: bool,
+               with_data: bool) -> "GGMLFile":
         off = 0
 
         def u32() -> int:
@@ -355,7 +354,7 @@ class GGMLFile:
         vocab: List[Tuple[bytes, float]] = []
         for _ in range(n_vocab):
             ln = u32()
-            word = data[off:off + ln]
+            word = bytes(data[off:off + ln])
             off += ln
             (score,) = struct.unpack_from("<f", data, off)
             off += 4
@@ -371,7 +370,7 @@ class GGMLFile:
                 raise ValueError(f"tensor with {n_dims} dims")
             ne = struct.unpack_from("<%dI" % n_dims, data, off)
             off += 4 * n_dims
-            name = data[off:off + name_len].decode("utf-8")
+            name = bytes(data[off:off + name_len]).decode("utf-8")
             off += name_len
             off += -off & 31
             size = tensor_nbytes(gtype, ne)
@@ -379,9 +378,72 @@ class GGMLFile:
             if with_data and len(raw) != size:
                 raise ValueError(f"truncated tensor data for {name}")
             off += size
+            # mmap mode: raw stays a zero-copy view into the mapping;
+            # eager mode: a bytes copy as before
+            if not isinstance(data, memoryview):
+                raw = bytes(raw)
             tensors.append(GGMLTensor(name=name, ne=tuple(ne), gtype=gtype,
-                                      raw=bytes(raw)))
+                                      raw=raw))
         return cls(hparams=hp, vocab=vocab, tensors=tensors)
+
+
+def write_header(f: BinaryIO, hp: Hparams,
+                 vocab: List[Tuple[bytes, float]]) -> None:
+    gqa = hp.is_gqa
+    f.write(struct.pack("<II", GGJT_MAGIC,
+                        GGJT_VERSION_GQA if gqa else GGJT_VERSION))
+    fields = [hp.n_vocab, hp.n_embd, hp.n_mult, hp.n_head]
+    if gqa:
+        fields.append(hp.kv_heads)
+    fields += [hp.n_layer, hp.n_rot]
+    if hp.first_layer is not None:
+        fields.append(hp.first_layer)
+    fields.append(hp.ftype)
+    f.write(struct.pack("<%dI" % len(fields), *fields))
+    if len(vocab) != hp.n_vocab:
+        raise ValueError(
+            f"vocab has {len(vocab)} entries, hparams say {hp.n_vocab}")
+    for word, score in vocab:
+        f.write(struct.pack("<I", len(word)))
+        f.write(word)
+        f.write(struct.pack("<f", score))
+
+
+def write_tensor_record(f: BinaryIO, t: GGMLTensor) -> None:
+    name_b = t.name.encode("utf-8")
+    f.write(struct.pack("<III", len(t.ne), len(name_b), t.gtype))
+    f.write(struct.pack("<%dI" % len(t.ne), *t.ne))
+    f.write(name_b)
+    pad = -f.tell() & 31
+    f.write(b"\x00" * pad)
+    if len(t.raw) != t.nbytes:
+        raise ValueError(
+            f"tensor {t.name}: raw {len(t.raw)} B != expected "
+            f"{t.nbytes} B")
+    f.write(t.raw)
+
+
+class GGMLWriter:
+    """Streaming writer: header first, then one tensor record at a time
+    — builds 100 GB-class synthetic checkpoints without materializing
+    them in RAM (GGMLFile.save needs every tensor resident)."""
+
+    def __init__(self, path: str, hp: Hparams,
+                 vocab: List[Tuple[bytes, float]]):
+        self._f = open(path, "wb")
+        write_header(self._f, hp, vocab)
+
+    def add(self, t: GGMLTensor) -> None:
+        write_tensor_record(self._f, t)
+
+    def close(self) -> None:
+        self._f.close()
+
+    def __enter__(self) -> "GGMLWriter":
+        return self
+
+    def __exit__(self, *exc) -> None:
+        self.close()
 
 
 def sniff_extended(path: str) -> bool:
