@@ -379,6 +379,16 @@ class ServeHttpCommand(Command):
                        help="max prompt tokens prefilled per lane per "
                             "decode step (latency fairness; 0 = "
                             "unbounded)")
+        p.add_argument("--pipeline", action="store_true",
+                       help="serve across torch.distributed pipeline "
+                            "ranks (launch with torchrun, one rank per "
+                            "GPU; rank 0 runs the HTTP front + batcher, "
+                            "followers execute their layer slice)")
+        p.add_argument("--backend", default="auto",
+                       choices=["auto", "nccl", "gloo"],
+                       help="pipeline backend: auto = nccl(RCCL) on "
+                            "GPU; gloo supports CPU and one-GPU "
+                            "multi-rank shakeout")
 
     def __call__(self, args) -> int:
         import torch
@@ -389,6 +399,8 @@ class ServeHttpCommand(Command):
         from ..formats import ggml, slicer
         from ..serving import ContinuousBatcher, build_http_app
 
+        if args.pipeline:
+            return self._pipeline_main(args)
         f = ggml.GGMLFile.load(args.model,
                                extended=ggml.sniff_extended(args.model))
         n_lanes = args.lanes
@@ -415,6 +427,62 @@ class ServeHttpCommand(Command):
                         log_level="warning")
         finally:
             worker.stop()
+        return 0
+
+    def _pipeline_main(self, args) -> int:
+        """Multi-GPU serving: rank 0 = HTTP + batcher over the
+        PipelineEngine facade; follower ranks execute their slice."""
+        import os
+
+        import torch
+        import torch.distributed as dist
+        import uvicorn
+
+        from ..engine import engine_for_slice
+        from ..engine.tokenizer import Tokenizer
+        from ..formats import ggml, slicer
+        from ..parallel.pipeline import partition_layers
+        from ..serving import ContinuousBatcher, build_http_app
+        from ..serving.pipeline_server import PipelineEngine, serve_forever
+
+        world = int(os.environ.get("WORLD_SIZE", "1"))
+        rank = int(os.environ.get("RANK", "0"))
+        device = "cuda" if torch.cuda.is_available() else "cpu"
+        backend = ("nccl" if device == "cuda" else "gloo") \
+            if args.backend == "auto" else args.backend
+        if world > 1:
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            dist.init_process_group(backend, rank=rank, world_size=world)
+            if device == "cuda":
+                torch.cuda.set_device(
+                    int(os.environ.get("LOCAL_RANK", "0"))
+                    % torch.cuda.device_count())
+        f = ggml.GGMLFile.load(args.model,
+                               extended=ggml.sniff_extended(args.model))
+        parts = partition_layers(f.hparams.n_layer, world)
+        first, count = parts[rank]
+        sl = slicer.make_slice(f, first, first + count - 1) \
+            if world > 1 else f
+        ex = slicer.make_extra_layers(f)
+        eng = engine_for_slice(sl, n_ctx=args.ctx, max_batch=args.slots)
+        eng.attach_extra(ex)  # rank 0: embed; last rank: lm head
+        if rank > 0:
+            serve_forever(eng, rank, world)
+            dist.destroy_process_group()
+            return 0
+        facade = PipelineEngine(eng, rank, world) if world > 1 else eng
+        tok = Tokenizer(f.vocab)
+        bat = ContinuousBatcher(facade,
+                                prefill_chunk=args.prefill_chunk or None)
+        app, worker = build_http_app(bat, tok)
+        try:
+            uvicorn.run(app, host=args.host, port=args.port,
+                        log_level="warning")
+        finally:
+            worker.stop()
+            if world > 1:
+                facade.shutdown()
+                dist.destroy_process_group()
         return 0
 
 
