@@ -86,3 +86,61 @@ def test_two_rank_staged_pipeline_token_exact_on_gpu():
     assert len(got) == 2 * len(want)
     for row in got:
         assert row == want[0]
+
+
+def _serving_rank(rank, world, port, q):
+    import torch.distributed as dist
+
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.serving import ContinuousBatcher
+    from distributedllm_amd.serving.pipeline_server import (
+        PipelineEngine, serve_forever)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    f = synthetic.build_model("tiny", ftype=ggml.FTYPE_ALL_F32, seed=0)
+    ex = slicer.make_extra_layers(f)
+    parts = partition_layers(f.hparams.n_layer, world)
+    first, count = parts[rank]
+    sl = slicer.make_slice(f, first, first + count - 1)
+    eng = HIPSliceEngine.from_ggml(sl, n_ctx=32, max_batch=4)
+    eng.attach_extra(ex)
+    if rank > 0:
+        serve_forever(eng, rank, world)
+        dist.destroy_process_group()
+        return
+    facade = PipelineEngine(eng, rank, world)
+    bat = ContinuousBatcher(facade)
+    reqs = [bat.submit(p, m) for p, m in
+            zip([[5, 9, 3], [7], [11, 2, 8, 4, 1]], [4, 5, 3])]
+    bat.run_all(max_steps=64)
+    facade.shutdown()
+    q.put([r.out for r in reqs])
+    dist.destroy_process_group()
+
+
+def test_pipeline_serving_on_gpu_matches_single():
+    """Continuous batching across 2 HIP-engine ranks (staged-gloo hops
+    on one GPU) == the single HIP engine batcher, token-exact (f32
+    slices)."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.serving import ContinuousBatcher
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_serving_rank, args=(r, 2, 29731, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = q.get(timeout=240)
+    for p in procs:
+        p.join(timeout=240)
+        assert p.exitcode == 0
+    f = synthetic.build_model("tiny", ftype=ggml.FTYPE_ALL_F32, seed=0)
+    ex = slicer.make_extra_layers(f)
+    eng = HIPSliceEngine.from_ggml(f, n_ctx=32, max_batch=4)
+    eng.attach_extra(ex)
+    bat = ContinuousBatcher(eng)
+    reqs = [bat.submit(p, m) for p, m in
+            zip([[5, 9, 3], [7], [11, 2, 8, 4, 1]], [4, 5, 3])]
+    bat.run_all(max_steps=64)
+    assert got == [r.out for r in reqs]
