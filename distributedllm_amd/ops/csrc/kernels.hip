@@ -503,119 +503,130 @@ __global__ void k_attention(
         __syncthreads();
     }
 
-    float m = -INFINITY;  // running max (block-uniform)
-    float l = 0.0f;       // running denom (block-uniform)
-    // V partials: lane covers the d-octet (lane&15)*8 of rows
-    // jc ≡ (lane>>4) (mod 4) in its wave's quarter — one 16 B load per
-    // row-octet (2 B scalar loads were the deep-decode ceiling:
-    // ~2 B/cyc/CU issue-bound, half the step time; MI355X_MICROARCH.md
-    // per-instruction table)
-    float o8[8];
-#pragma unroll
-    for (int e = 0; e < 8; ++e) o8[e] = 0.0f;
+    // Flash-style per-wave streaming (zero __syncthreads in the loop):
+    // wave w owns 4-row groups g ≡ w (mod 4); within the wave, lane
+    // (sub = lane>>4, oct = lane&15) covers row (g*4 + sub)'s d-octet
+    // oct*8 — one 16 B load each for K and V per row, the row's dot
+    // reduced by a 4-step in-16 shuffle butterfly, q held in registers.
+    // Each wave keeps private (m, l, o8) partials; one flash merge at
+    // the end. (The earlier block-synchronized form paid 3 barriers +
+    // 2 cross-wave reductions per 256-row chunk and scalar-width V
+    // loads before that — the deep-decode ladder in
+    // profiles/r2_deep_decode_ladder.md tracks the steps.)
     const int wid = threadIdx.x / WAVE;
-
     const int lane0 = threadIdx.x & (WAVE - 1);
     const int sub0 = lane0 >> 4;        // row within the 4-row group
     const int oct0 = lane0 & 15;        // d-octet within the row
-    const int Dp8 = (D + 7) & ~7;
-    for (int j0 = 0; j0 < J; j0 += BLOCK) {
-        // scores, row-major: each wave iteration reads 4 K rows with 16
-        // lanes per row (one contiguous 16 B octet each — the whole row
-        // is fetched by ONE instruction, so cache lines are fully used
-        // the moment they arrive; the old thread-per-row form touched 64
-        // scattered lines per instruction and thrashed L1). Partial dots
-        // reduce across the row's 16 lanes with 4 shuffles.
-        const int jmax0 = min(BLOCK, J - j0);
-        const int jw0 = wid * (BLOCK / NWAVES);
-        const int jw1 = min(jmax0, jw0 + BLOCK / NWAVES);
-        // uniform trip count: every lane runs the same iterations (the
-        // 16-lane shuffles below must stay converged)
-        const int ngrp = (jw1 - jw0 + 3) >> 2;
-        for (int gi = 0; gi < ngrp; ++gi) {
-            const int jc = jw0 + gi * 4 + sub0;
-            float acc = 0.0f;
-            if (jc < jw1 && oct0 * 8 < Dp8) {
-                const __half* krow =
-                    k_cache + base + (size_t)(j0 + jc) * Ekv;
-                float f[8];
-                load_voct(krow, oct0 * 8, f);
+    const int d0 = oct0 * 8;
+    float q8[8];
 #pragma unroll
-                for (int e = 0; e < 8; ++e)
-                    acc = fmaf(f[e], lds_q[oct0 * 8 + e], acc);
-            }
-            // reduce over the 16 octet lanes of this row
-            acc += __shfl_xor(acc, 1);
-            acc += __shfl_xor(acc, 2);
-            acc += __shfl_xor(acc, 4);
-            acc += __shfl_xor(acc, 8);
-            if (oct0 == 0 && jc < jw1) lds_p[jc] = acc;
+    for (int e = 0; e < 8; ++e) {
+        const int d = d0 + e;
+        if (FUSEQKV) {
+            q8[e] = (d < D) ? lds_q[d] : 0.0f;  // prescaled in prologue
+        } else {
+            q8[e] = (d < D)
+                ? q_buf[(size_t)t * E + h * D + d] * inv_sqrt_d
+                : 0.0f;
         }
-        __syncthreads();
-        const int jj = j0 + threadIdx.x;
-        float s = (threadIdx.x < jmax0) ? lds_p[threadIdx.x] : -INFINITY;
-        // block max
-        float wm = wave_reduce_max(s);
-        if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = wm;
-        __syncthreads();
-        float cm = fmaxf(fmaxf(lds_red[0], lds_red[1]),
-                         fmaxf(lds_red[2], lds_red[3]));
-        const float m_new = fmaxf(m, cm);
-        const float alpha = (m == -INFINITY) ? 0.0f : __expf(m - m_new);
-        const float p = (jj < J) ? __expf(s - m_new) : 0.0f;
-        __syncthreads();  // lds_p reuse across chunks
-        lds_p[threadIdx.x] = p;
-        // block sum of p
-        float ws = wave_reduce_sum(p);
-        if ((threadIdx.x & (WAVE - 1)) == 0) lds_red[wid] = ws;
-        __syncthreads();
-        const float sum_p = lds_red[0] + lds_red[1] + lds_red[2] + lds_red[3];
-        l = l * alpha + sum_p;
-        m = m_new;
-        // V accumulation: wave w covers chunk rows [w*64, +64); within
-        // the wave, lane sub-group (lane>>4) takes every 4th row and
-        // the lane's d-octet rides ONE 16 B load per row. Partials per
-        // (wave, sub, octet) merge once at the kernel end; alpha is
-        // block-uniform so partials scale consistently each chunk.
-        {
-            const int jmax = min(BLOCK, J - j0);
-            const int jq0 = wid * (BLOCK / NWAVES);
-            const int jq1 = min(jmax, jq0 + BLOCK / NWAVES);
-            const int lane = threadIdx.x & (WAVE - 1);
-            const int sub = lane >> 4;
-            const int d0 = (lane & 15) * 8;
+    }
+    float m = -INFINITY, l = 0.0f;
+    float o8[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) o8[e] = 0.0f;
+
+    // two 4-row groups per iteration: both K loads issue before either
+    // dot needs them, and the 8 rows share one max/rescale pass
+    for (int g = wid; g * 4 < J; g += 2 * NWAVES) {
+        const int j1 = g * 4 + sub0;
+        const int g2i = g + NWAVES;
+        const int j2 = g2i * 4 + sub0;
+        const bool ok1 = j1 < J;
+        const bool ok2 = (g2i * 4 < J) && (j2 < J);
+        float s1 = 0.0f, s2 = 0.0f;
+        if (d0 < D) {
+            float f1[8], f2[8];
+            load_voct(k_cache + base + (size_t)(ok1 ? j1 : 0) * Ekv, d0,
+                      f1);
+            load_voct(k_cache + base + (size_t)(ok2 ? j2 : 0) * Ekv, d0,
+                      f2);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+                s1 = fmaf(f1[e], q8[e], s1);
+                s2 = fmaf(f2[e], q8[e], s2);
+            }
+        }
+        // full dots: butterfly over the 16 octet lanes
+#pragma unroll
+        for (int sh = 1; sh <= 8; sh <<= 1) {
+            s1 += __shfl_xor(s1, sh);
+            s2 += __shfl_xor(s2, sh);
+        }
+        if (!ok1) s1 = -INFINITY;
+        if (!ok2) s2 = -INFINITY;
+        // max over all 8 rows (sub groups at lane bits 4-5)
+        float gm = fmaxf(s1, s2);
+        gm = fmaxf(gm, __shfl_xor(gm, 16));
+        gm = fmaxf(gm, __shfl_xor(gm, 32));
+        if (gm > m) {  // rescale partials (wave-uniform branch)
+            const float alpha = (m == -INFINITY) ? 0.0f : __expf(m - gm);
 #pragma unroll
             for (int e = 0; e < 8; ++e) o8[e] *= alpha;
-            if (d0 < D) {
-                for (int jc = jq0 + sub; jc < jq1; jc += 4) {
-                    const float pj = lds_p[jc];
-                    float f[8];
-                    load_voct(v_cache + base + (size_t)(j0 + jc) * Ekv,
-                              d0, f);
+            l *= alpha;
+            m = gm;
+        }
+        const float p1 = ok1 ? __expf(s1 - m) : 0.0f;
+        const float p2 = ok2 ? __expf(s2 - m) : 0.0f;
+        float psum = (p1 + p2) + __shfl_xor(p1 + p2, 16);
+        psum += __shfl_xor(psum, 32);
+        l += psum;
+        if (d0 < D) {
+            float f1[8], f2[8];
+            // masked rows contribute p = 0 (V row 0 is a safe address)
+            load_voct(v_cache + base + (size_t)(ok1 ? j1 : 0) * Ekv, d0,
+                      f1);
+            load_voct(v_cache + base + (size_t)(ok2 ? j2 : 0) * Ekv, d0,
+                      f2);
 #pragma unroll
-                    for (int e = 0; e < 8; ++e)
-                        o8[e] = fmaf(pj, f[e], o8[e]);
-                }
+            for (int e = 0; e < 8; ++e) {
+                o8[e] = fmaf(p1, f1[e], o8[e]);
+                o8[e] = fmaf(p2, f2[e], o8[e]);
             }
         }
-        __syncthreads();
     }
-    // merge the 16 (wave, sub) partial groups
+
+    // flash merge of the 4 waves' (m, l) x 16 (wave, sub) o8 partials
     {
-        const int lane = threadIdx.x & (WAVE - 1);
-        float* lds_o = lds_red + NWAVES;  // [16][128] merge area
-        const int g = wid * 4 + (lane >> 4);
-        const int d0 = (lane & 15) * 8;
+        float* lds_m = lds_red;               // [NWAVES]
+        float* lds_l = lds_red + NWAVES;      // [NWAVES]
+        float* lds_o = lds_l + NWAVES;        // [16][128]
+        const int gidx = wid * 4 + sub0;
 #pragma unroll
-        for (int e = 0; e < 8; ++e) lds_o[g * 128 + d0 + e] = o8[e];
+        for (int e = 0; e < 8; ++e) lds_o[gidx * 128 + d0 + e] = o8[e];
+        if (lane0 == 0) {
+            lds_m[wid] = m;
+            lds_l[wid] = l;
+        }
         __syncthreads();
         if (threadIdx.x < D) {
             const int d = threadIdx.x;
-            float v = 0.0f;
+            float mstar = -INFINITY;
 #pragma unroll
-            for (int g2 = 0; g2 < 4 * NWAVES; ++g2)
-                v += lds_o[g2 * 128 + d];
-            v /= l;
+            for (int w = 0; w < NWAVES; ++w)
+                mstar = fmaxf(mstar, lds_m[w]);
+            float lstar = 0.0f, num = 0.0f;
+#pragma unroll
+            for (int w = 0; w < NWAVES; ++w) {
+                const float mw = lds_m[w];
+                const float sc =
+                    (mw == -INFINITY) ? 0.0f : __expf(mw - mstar);
+                lstar += lds_l[w] * sc;
+                num += sc * (lds_o[(w * 4 + 0) * 128 + d] +
+                             lds_o[(w * 4 + 1) * 128 + d] +
+                             lds_o[(w * 4 + 2) * 128 + d] +
+                             lds_o[(w * 4 + 3) * 128 + d]);
+            }
+            const float v = num / lstar;
             const int e = h * D + d;
             out[(size_t)t * E + e] = v;
             if (out_prep != nullptr) {
@@ -2501,8 +2512,10 @@ void launch_attention(hipStream_t s, const float* q_buf,
                       int n_ctx, const float* qkv_slab, int ks,
                       const float* inv_freq) {
     const dim3 grid(T, H);
+    // lds_q[D] + prologue stage[BLOCK] + lds_m/lds_l[2*NWAVES] +
+    // merge area [16][128]
     const size_t lds =
-        (D + BLOCK + NWAVES + 16 * 128) * sizeof(float);
+        (D + BLOCK + 2 * NWAVES + 16 * 128) * sizeof(float);
     if (qkv_slab != nullptr) {
         hipLaunchKernelGGL(k_attention<true>, grid, dim3(BLOCK), lds, s,
                            q_buf, k_cache_layer, v_cache_layer, out,
