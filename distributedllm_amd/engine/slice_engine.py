@@ -198,6 +198,49 @@ def _pack_grouped_scales(alpha, beta, rows, nb, nbp, R):
         ab.reshape(R, 16, nbp // 4, 4, 2).transpose(0, 2, 1, 3, 4))
 
 
+def _repack_q4_torch(t: ggml.GGMLTensor, device: str):
+    """q4_0/q4_1 -> MFMA tile layout with the bit work done by torch
+    integer ops ON THE GPU: only the compressed bytes cross PCIe and
+    the ~8 expansion passes run at HBM rate (the numpy path measured
+    0.2 GB/s on a 30B q4_0 load — BASELINE.md)."""
+    rows, cols = t.shape_rows_cols
+    R, nb = rows // 16, cols // 32
+    nbp = (nb + 3) & ~3
+    bs = 18 if t.gtype == ggml.GGML_TYPE_Q4_0 else 20
+    host = np.frombuffer(t.raw, np.uint8)
+    raw = torch.tensor(host, device=device)  # one host copy + H2D
+    a = raw.view(rows, nb, bs)
+    qs = a[:, :, bs - 16:]
+    # weight j sits at nibble j (low) / j+16 (high) of the 16 qs bytes
+    n = torch.cat([(qs & 0xF), (qs >> 4)], dim=-1).to(torch.int32)
+    n = n.view(rows, nb, 4, 8)  # [rows][block][ws][j]
+    shifts = torch.tensor([(j % 2) * 16 + (j // 2) * 4 for j in range(8)],
+                          device=device, dtype=torch.int32)
+    # disjoint 4-bit fields -> sum == OR
+    qword = (n << shifts).sum(dim=-1, dtype=torch.int64).to(torch.int32)
+    qp = torch.zeros(rows, nbp, 4, dtype=torch.int32, device=device)
+    qp[:, :nb] = qword
+    qs2 = (qp.view(R, 16, nbp // 4, 4, 4)
+           .permute(0, 2, 4, 1, 3).contiguous())
+    data = torch.cat([qs2.reshape(-1),
+                      torch.zeros(256, dtype=torch.int32, device=device)])
+    alpha = torch.zeros(rows, nbp, dtype=torch.float32, device=device)
+    beta = torch.zeros_like(alpha)
+    if t.gtype == ggml.GGML_TYPE_Q4_0:
+        d = a[:, :, :2].contiguous().view(torch.half)
+        alpha[:, :nb] = d.view(rows, nb).float()
+    else:
+        dm = a[:, :, :4].contiguous().view(torch.half).view(rows, nb, 2)
+        alpha[:, :nb] = dm[:, :, 0].float()
+        beta[:, :nb] = dm[:, :, 1].float()
+    ab = torch.stack([alpha, beta], dim=-1).to(torch.float16)
+    sc = (ab.view(R, 16, nbp // 4, 4, 2)
+          .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
+    sc = torch.cat([sc, torch.zeros(128, dtype=torch.float16,
+                                    device=device)])
+    return data, sc, t.gtype
+
+
 def repack_mfma(t: ggml.GGMLTensor, device: str):
     """On-disk tensor -> (data, scales, wtype) in the MFMA tile layout.
 
@@ -213,6 +256,9 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
     rows, cols = t.shape_rows_cols
     assert rows % 16 == 0 and cols % 32 == 0, (t.name, rows, cols)
     R = rows // 16
+    if t.gtype in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1) and \
+            device != "cpu":
+        return _repack_q4_torch(t, device)
     if t.gtype in (ggml.GGML_TYPE_Q4_0, ggml.GGML_TYPE_Q4_1):
         nb = cols // 32
         nbp = (nb + 3) & ~3  # padded to the dwordx4 load group of 4 blocks

@@ -131,3 +131,22 @@ def test_detile_inverts_repack():
         tol = 4e-3 * want.abs().max().item() + 1e-6
         err = (got - want).abs().max().item()
         assert err <= tol, (ggml.TYPE_NAMES[ggml._FTYPE_TO_GGML[ft]], err)
+
+
+def test_torch_q4_repack_matches_numpy():
+    """The GPU-side torch q4 repack (one H2D of compressed bytes, bit
+    work on device) must produce bit-identical tiles/scales to the
+    numpy reference path."""
+    import numpy as np
+    import torch
+    from distributedllm_amd.engine import slice_engine as SE
+    from distributedllm_amd.formats import ggml, synthetic
+    for ft in (ggml.FTYPE_MOSTLY_Q4_0, ggml.FTYPE_MOSTLY_Q4_1):
+        f = synthetic.build_model("small", seed=3, ftype=ft)
+        t = next(x for x in f.tensors
+                 if x.name.endswith("attention.wq.weight"))
+        d_np, s_np, wt_np = SE.repack_mfma(t, "cpu")       # numpy path
+        d_th, s_th, wt_th = SE._repack_q4_torch(t, "cpu")  # torch ops
+        assert wt_np == wt_th
+        assert torch.equal(d_np, d_th)
+        assert torch.equal(s_np.view(torch.int16), s_th.view(torch.int16))

@@ -22,8 +22,15 @@ FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
           "f16": ggml.FTYPE_MOSTLY_F16}
 
 
-def build_streaming(path: str, preset, ftype: int, seed: int = 0) -> float:
-    """Write the synthetic model file tensor by tensor (bounded RAM)."""
+def build_streaming(path: str, preset, ftype: int, seed: int = 0,
+                    fast: bool = False) -> float:
+    """Write the synthetic model file tensor by tensor (bounded RAM).
+
+    fast=True writes RANDOM RAW BYTES for the 2-D tensors (any bit
+    pattern is a valid quantized block; f16 scale bytes may be inf/nan
+    patterns, fine for a LOAD benchmark) — the f32->quantize path costs
+    minutes of CPU at 30B+ scale and measures the codec, not the
+    loader."""
     t0 = time.perf_counter()
     hp = preset.hparams(ftype)
     wt = ggml._FTYPE_TO_GGML[ftype]
@@ -32,6 +39,10 @@ def build_streaming(path: str, preset, ftype: int, seed: int = 0) -> float:
     Ekv = preset.n_embd_kv
 
     def mat(name, rows, cols):
+        if fast:
+            nb = ggml.tensor_nbytes(wt, (cols, rows))
+            return ggml.GGMLTensor(name=name, ne=(cols, rows), gtype=wt,
+                                   raw=rng.bytes(nb))
         a = rng.standard_normal((rows, cols), dtype=np.float32) * 0.02
         return ggml.GGMLTensor.from_f32(name, a, wt)
 
@@ -65,13 +76,16 @@ def main():
     ap.add_argument("--path", default="/tmp/load_bench_model.bin")
     ap.add_argument("--ctx", type=int, default=512)
     ap.add_argument("--keep", action="store_true")
+    ap.add_argument("--fast-build", action="store_true",
+                    help="random raw bytes instead of f32+quantize")
     args = ap.parse_args()
     p = PRESETS[args.model]
     ftype = FTYPES[args.ftype]
 
     t_build = None
     if not os.path.exists(args.path):
-        t_build = build_streaming(args.path, p, ftype)
+        t_build = build_streaming(args.path, p, ftype,
+                                  fast=args.fast_build)
     fsize = os.path.getsize(args.path)
 
     rss0 = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss * 1024
