@@ -672,9 +672,10 @@ class HIPSliceEngine:
         detiled-f16 path — no f16 weight copy, mixed multi-span
         admission streams handled natively)."""
         T = x.shape[0]
-        cap = self._eng.max_tokens if decode else self._eng.max_prefill
-        if not decode and not self._mfma_path():
-            cap = self._eng.max_tokens  # legacy f32 path: 64-token tiles
+        # the large-M path serves decode too (wide batched decode pays
+        # the weight stream + dequant once for the whole batch)
+        cap = self._eng.max_prefill if self._mfma_path() \
+            else self._eng.max_tokens
         if T <= cap:
             return self._eng.forward(x, pos, seq, decode=decode)
         # token-tile larger inputs; KV order is preserved because tile

@@ -310,8 +310,6 @@ class SliceEngine {
         TORCH_CHECK(T >= 1 && T <= tcap,
                     "forward handles at most ", tcap,
                     " tokens per call; tile larger batches host-side");
-        TORCH_CHECK(!decode || T <= kMaxTokens,
-                    "decode batches are capped at ", kMaxTokens);
         TORCH_CHECK(pos.numel() == T && seq.numel() == T, "pos/seq size");
         // Opt-in backstop against out-of-bounds KV writes (an oversized
         // request that slipped past host-side validation). Costs a host
@@ -355,10 +353,16 @@ class SliceEngine {
         (void)hipMemsetAsync(ssf, 0, sizeof(float) * L_ * ssw_, s);
         launch_prep_x(s, xp, xprep, ssa, E_, T);
         if (T > kMaxTokens) {
-            // large-M prefill: the *_mt kernels cover all T tokens in one
+            // large-M path: the *_mt kernels cover all T tokens in one
             // launch per op (XCD-grouped token tiles), the whole layer
             // loop sequenced here in C++ — no host tiling, no library
-            // GEMMs, q4 tiles read directly
+            // GEMMs, q4 tiles read directly. Serves BOTH prompt prefill
+            // and wide batched decode (decode=true: every token its own
+            // sequence — weights and dequant are paid ONCE for the whole
+            // batch instead of once per 64-token lane): only the
+            // attention kernel differs — 16-query tiles are spans for
+            // prefill, per-token streaming for the distinct-sequence
+            // decode shape.
             for (int li = 0; li < L_; ++li) {
                 Layer& l = layers_[li];
                 __half* kc = kbase + (size_t)li * layer_stride;
@@ -367,8 +371,13 @@ class SliceEngine {
                                 u16p(l.attn_normprep), ssa + li * ssw_,
                                 eps_, qb, kc, vc, pp, sp, ifr, E_, EK_,
                                 D_, ctx_, T);
-                launch_attn_prefill(s, qb, kc, vc, ab, aprep, pp, sp, T,
-                                    H_, E_, EK_, D_, ctx_);
+                if (decode)
+                    launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T,
+                                     H_, E_, EK_, D_, ctx_, nullptr, 0,
+                                     ifr);
+                else
+                    launch_attn_prefill(s, qb, kc, vc, ab, aprep, pp, sp,
+                                        T, H_, E_, EK_, D_, ctx_);
                 launch_gemm16_mt(s, l.mo.w, aprep, xp, xprep,
                                  ssf + li * ssw_, T, /*res_sq=*/1);
                 launch_ffn16_mt(s, l.m1.w, l.m3.w, xprep,
