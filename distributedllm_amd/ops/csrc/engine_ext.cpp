@@ -225,7 +225,7 @@ class SliceEngine {
         ss_attn_ = torch::zeros({(int64_t)(L_ + 1) * ssw_}, f32);
         ss_ffn_ = torch::zeros({(int64_t)L_ * ssw_}, f32);
         ss_tmp_ = torch::zeros({kMaxTokens}, f32);
-        argmax_keys_ = torch::zeros({kMaxTokens}, dev.dtype(torch::kInt64));
+        argmax_keys_ = torch::zeros({maxP_}, dev.dtype(torch::kInt64));
         // split-K partial slabs: sized for the largest user — qkv
         // (3E/16 tiles), ffn (2F/16), wo/w2 (E/16) — at the max split of 8
         const int64_t slab_tiles =
@@ -491,7 +491,7 @@ class SliceEngine {
         auto out = torch::empty(
             {T},
             torch::TensorOptions().device(torch::kCUDA).dtype(torch::kInt32));
-        TORCH_CHECK(T <= kMaxTokens, "argmax: too many rows");
+        TORCH_CHECK(T <= maxP_, "argmax: too many rows");
         hipStream_t s = c10::hip::getCurrentHIPStream().stream();
         launch_argmax(
             s, lg.data_ptr<float>(),
