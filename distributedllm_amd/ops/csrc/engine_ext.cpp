@@ -363,6 +363,16 @@ class SliceEngine {
             // attention kernel differs — 16-query tiles are spans for
             // prefill, per-token streaming for the distinct-sequence
             // decode shape.
+            static const bool dbg_sync = [] {
+                const char* v = std::getenv("DLLM_DEBUG_SYNC");
+                return v && v[0] == '1';
+            }();
+            auto ck = [&](const char* what, int li2) {
+                if (!dbg_sync) return;
+                hipError_t e = hipStreamSynchronize(s);
+                fprintf(stderr, "[dbg] L%d %s: %s\n", li2, what,
+                        hipGetErrorString(e));
+            };
             for (int li = 0; li < L_; ++li) {
                 Layer& l = layers_[li];
                 __half* kc = kbase + (size_t)li * layer_stride;
@@ -371,6 +381,7 @@ class SliceEngine {
                                 u16p(l.attn_normprep), ssa + li * ssw_,
                                 eps_, qb, kc, vc, pp, sp, ifr, E_, EK_,
                                 D_, ctx_, T);
+                ck("qkv_mt", li);
                 if (decode)
                     launch_attention(s, qb, kc, vc, ab, aprep, pp, sp, T,
                                      H_, E_, EK_, D_, ctx_, nullptr, 0,
@@ -378,13 +389,17 @@ class SliceEngine {
                 else
                     launch_attn_prefill(s, qb, kc, vc, ab, aprep, pp, sp,
                                         T, H_, E_, EK_, D_, ctx_);
+                ck("attn", li);
                 launch_gemm16_mt(s, l.mo.w, aprep, xp, xprep,
                                  ssf + li * ssw_, T, /*res_sq=*/1);
+                ck("wo", li);
                 launch_ffn16_mt(s, l.m1.w, l.m3.w, xprep,
                                 u16p(l.ffn_normprep), ssf + li * ssw_,
                                 eps_, gprep, T);
+                ck("ffn", li);
                 launch_gemm16_mt(s, l.m2.w, gprep, xp, xprep,
                                  ssa + (li + 1) * ssw_, T, /*res_sq=*/1);
+                ck("w2", li);
             }
             return x;
         }

@@ -2913,6 +2913,6 @@ void launch_argmax(hipStream_t s, const float* logits,
     const int npart = min(32, (V + BLOCK - 1) / BLOCK);
     hipLaunchKernelGGL(k_argmax_part, dim3(T, npart), dim3(BLOCK), 0, s,
                        logits, keys, V);
-    hipLaunchKernelGGL(k_argmax_finish, dim3(1), dim3(64), 0, s, keys, out,
-                       T);
+    hipLaunchKernelGGL(k_argmax_finish, dim3((T + 63) / 64), dim3(64), 0,
+                       s, keys, out, T);
 }
