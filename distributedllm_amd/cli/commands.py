@@ -371,9 +371,13 @@ class ServeHttpCommand(Command):
         p.add_argument("--host", default="127.0.0.1")
         p.add_argument("--port", type=int, default=8080)
         p.add_argument("--slots", type=int, default=64)
-        p.add_argument("--lanes", type=int, default=0,
-                       help="stream lanes (weight-sharing clones; GPU); "
-                            "0 = auto from HBM fit (>=3 or 1, max 5)")
+        p.add_argument("--lanes", type=int, default=1,
+                       help="stream lanes (weight-sharing clones; GPU). "
+                            "1 (default) serves all slots in ONE wide "
+                            "decode batch — weights/dequant paid once "
+                            "per step (26.2k vs 18.4k tok/s measured at "
+                            "256 requests on 3B); 0 = auto lane fit "
+                            "(the round-1 multi-stream mode)")
         p.add_argument("--ctx", type=int, default=2048)
         p.add_argument("--prefill-chunk", type=int, default=64,
                        help="max prompt tokens prefilled per lane per "
