@@ -600,3 +600,45 @@ def test_kquant_hip_engine_matches_cpu(ftype):
     y3 = hip.forward(x3.cuda(), p3.cuda(), s3.cuda()).cpu()
     y3c = cpu.forward(x3.clone(), p3, s3)
     _assert_close(y3, y3c, label=f"kquant native prefill ftype={ftype}")
+
+
+def test_wide_decode_matches_tiled():
+    """decode with T > 64 (the wide batched-decode path through the _mt
+    kernels) must match the same step run as two <=64-token decode
+    calls on identical weights."""
+    f = synthetic.build_model("small", ftype=ggml.FTYPE_MOSTLY_Q4_0,
+                              seed=5)
+    ex = slicer.make_extra_layers(f)
+    from distributedllm_amd.engine import HIPSliceEngine
+    a = HIPSliceEngine.from_ggml(f, n_ctx=64, max_batch=96)
+    b = HIPSliceEngine.from_ggml(f, n_ctx=64, max_batch=96)
+    a.attach_extra(ex)
+    b.attach_extra(ex)
+    hp = f.hparams
+    torch.manual_seed(41)
+    T = 96
+    x = (torch.randn(T, hp.n_embd) * 0.5).cuda()
+    pos = torch.zeros(T, dtype=torch.int32, device="cuda")
+    seq = torch.arange(T, dtype=torch.int32, device="cuda")
+    y_wide = a.forward(x.clone(), pos, seq, decode=True)
+    outs = [b.forward(x[i:i + 48].clone(), pos[i:i + 48],
+                      seq[i:i + 48].contiguous(), decode=True)
+            for i in (0, 48)]
+    _assert_close(y_wide.cpu(), torch.cat(outs).cpu(),
+                  label="wide vs tiled decode")
+
+
+def test_argmax_wide_rows():
+    """Device argmax must cover EVERY row for T > 64 (rows >= 64 were
+    uninitialized before the wide-serving fix — embed of the garbage
+    ids faulted the GPU; tools/debug_wide.py is the repro)."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    p = PRESETS["small"]
+    hp = p.hparams(ggml.FTYPE_MOSTLY_Q4_0)
+    eng = HIPSliceEngine.random(hp, n_layers=1, n_ctx=16, max_batch=1,
+                                seed=0)
+    torch.manual_seed(3)
+    lg = torch.randn(200, hp.n_vocab, device="cuda")
+    ids = eng.argmax(lg)
+    assert torch.equal(ids.cpu(),
+                       lg.argmax(dim=-1).to(torch.int32).cpu())
