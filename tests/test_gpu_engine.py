@@ -636,7 +636,7 @@ def test_argmax_wide_rows():
     p = PRESETS["small"]
     hp = p.hparams(ggml.FTYPE_MOSTLY_Q4_0)
     eng = HIPSliceEngine.random(hp, n_layers=1, n_ctx=16, max_batch=1,
-                                seed=0)
+                                seed=0, max_prefill=256)
     torch.manual_seed(3)
     lg = torch.randn(200, hp.n_vocab, device="cuda")
     ids = eng.argmax(lg)
