@@ -26,7 +26,8 @@ def test_partition_layers():
     assert partition_layers(4, 2) == [(0, 2), (2, 2)]
 
 
-SEEDS = [[5, 9], [11, 3], [4, 8], [13, 2]]  # start tokens per micro-batch
+SEEDS = [[5, 9], [11, 3], [4, 8], [13, 2],
+         [6, 1], [9, 7], [2, 12], [10, 5]]  # start tokens per micro-batch
 
 
 def _seed_tokens(pipe, n_mb):
@@ -34,8 +35,8 @@ def _seed_tokens(pipe, n_mb):
         pipe.tok[m] = torch.tensor(SEEDS[m], dtype=torch.int32)
 
 
-def _single_reference_tokens(n_mb=1):
-    f = synthetic.build_model("tiny", seed=0)
+def _single_reference_tokens(n_mb=1, n_layer=None):
+    f = synthetic.build_model("tiny", seed=0, n_layer=n_layer)
     ex = slicer.make_extra_layers(f)
     eng = TorchSliceEngine.from_ggml(f, n_ctx=32, max_batch=MBS * n_mb)
     eng.attach_extra(ex)
@@ -46,12 +47,12 @@ def _single_reference_tokens(n_mb=1):
     return pipe.current_tokens().tolist()
 
 
-def _rank_main(rank, world, port, n_mb, q, n_lanes=1):
+def _rank_main(rank, world, port, n_mb, q, n_lanes=1, n_layer=None):
     import torch.distributed as dist
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
-    f = synthetic.build_model("tiny", seed=0)
+    f = synthetic.build_model("tiny", seed=0, n_layer=n_layer)
     ex = slicer.make_extra_layers(f)
     parts = partition_layers(f.hparams.n_layer, world)
     first, count = parts[rank]
@@ -80,11 +81,11 @@ def _rank_main(rank, world, port, n_mb, q, n_lanes=1):
     dist.destroy_process_group()
 
 
-def _run_cluster(world, port, n_mb, n_lanes=1):
+def _run_cluster(world, port, n_mb, n_lanes=1, n_layer=None):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_rank_main,
-                         args=(r, world, port, n_mb, q, n_lanes))
+                         args=(r, world, port, n_mb, q, n_lanes, n_layer))
              for r in range(world)]
     for p in procs:
         p.start()
@@ -307,3 +308,12 @@ def test_prime_with_lane_engines_matches_shared_engine():
         return pipeline_generate(pipe, prompt, max_steps=steps).tolist()
 
     assert run(lanes=True) == run(lanes=False)
+
+
+def test_four_stage_two_lane_pipeline_matches_single():
+    """4 ranks x 2 lanes x 8 micro-batches: the deepest posting-order
+    stress the CPU rig can give the isend/irecv schedule before the
+    driver's first 8-GPU run (per-pair FIFO matching must hold at any
+    rank count)."""
+    assert _run_cluster(4, 29544, 8, n_lanes=2, n_layer=4) == \
+        _single_reference_tokens(8, n_layer=4)
