@@ -27,7 +27,7 @@ def main():
     ap.add_argument("--num-tokens", type=int, default=64)
     ap.add_argument("--slots", type=int, default=64,
                     help="KV slots per lane")
-    ap.add_argument("--lanes", type=int, default=5,
+    ap.add_argument("--lanes", type=int, default=1,
                     help="concurrent stream lanes (weight-sharing "
                          "clones); total slots = slots x lanes")
     ap.add_argument("--ctx", type=int, default=2048)
