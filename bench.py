@@ -116,7 +116,13 @@ def main() -> int:
                "f32": 4.0}[args.ftype]  # byte formats: 1B + scales
         w_bytes = n_weights * bpw * 1.1  # repack padding/scales margin
         fit = max(1, int((260e9 - w_bytes) // max(kv_bytes, 1)))
-        n_lanes = min(n_lanes, fit) if fit >= 3 else 1
+        if fit >= n_lanes:
+            pass                      # explicit/default count fits
+        elif fit >= 3:
+            n_lanes = fit             # auto-shrink (3-5 measured best)
+        else:
+            n_lanes = 1               # 2 lanes always lost at mbs 64;
+            #                           wide single-lane takes over
     n_mb = n_lanes * max(world, 1) if world > 1 else n_lanes
     cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
 
