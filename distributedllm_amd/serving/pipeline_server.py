@@ -14,6 +14,11 @@ Control flow per engine call: rank 0 broadcasts a small command header
 slice and hands activations to the next over the same P2P transport the
 bench pipeline uses (RCCL on GPUs, host-staged gloo on one GPU / CPU).
 Follower ranks sit in `serve_forever()` until a shutdown command.
+Failure domain: the facade's calls are collective — if rank 0 dies
+mid-broadcast, followers block in the next collective until the
+launcher (torchrun) tears the job down; per-request errors never reach
+the collectives (the batcher validates at submit() and the worker
+catches step() failures before any broadcast is cut short).
 
 The reference has no equivalent — its node serves ONE request at a time
 over TCP; this is the scheduling layer the north star's

@@ -5,8 +5,6 @@ compare the residual stream AND the KV caches layer by layer."""
 import os
 import sys
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
-import sys
-
 import torch
 
 from distributedllm_amd.engine import HIPSliceEngine, TorchSliceEngine
