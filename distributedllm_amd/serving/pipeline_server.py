@@ -36,6 +36,10 @@ OP_FORWARD = 1   # prefill: forward only, KV append
 OP_DECODE = 2    # forward + logits (+ greedy ids) back to rank 0
 
 
+def _nccl() -> bool:
+    return dist.is_initialized() and dist.get_backend() == "nccl"
+
+
 class _Staged:
     """Host-staged P2P for gloo-with-CUDA (one-GPU shakeout) — same
     trick as DecodePipeline._hop_*."""
@@ -75,6 +79,9 @@ class PipelineEngine:
         self.max_batch = engine.max_batch
         self.device = engine.device
         self._staged = _Staged(self.device)
+        # control-message device: RCCL collectives move CUDA tensors,
+        # gloo moves CPU tensors
+        self._cdev = self.device if _nccl() else "cpu"
         self._tokens: Optional[torch.Tensor] = None
         self._lg: Optional[torch.Tensor] = None
         self._ids: Optional[torch.Tensor] = None
@@ -93,11 +100,11 @@ class PipelineEngine:
         if self.world > 1:
             hdr = torch.tensor(
                 [OP_DECODE if decode else OP_FORWARD, T],
-                dtype=torch.int64)
+                dtype=torch.int64, device=self._cdev)
             dist.broadcast(hdr, src=0)
-            body = torch.stack([toks.cpu().to(torch.int64),
-                                pos.cpu().to(torch.int64),
-                                seq.cpu().to(torch.int64)])
+            body = torch.stack([toks.to(self._cdev, torch.int64),
+                                pos.to(self._cdev, torch.int64),
+                                seq.to(self._cdev, torch.int64)])
             dist.broadcast(body, src=0)
         y = self.engine.forward(
             self.engine.embed(toks),
@@ -129,7 +136,8 @@ class PipelineEngine:
     def shutdown(self) -> None:
         if self.world > 1:
             dist.broadcast(torch.tensor([OP_SHUTDOWN, 0],
-                                        dtype=torch.int64), src=0)
+                                        dtype=torch.int64,
+                                        device=self._cdev), src=0)
 
 
 def serve_forever(engine, rank: int, world: int) -> None:
@@ -138,14 +146,15 @@ def serve_forever(engine, rank: int, world: int) -> None:
     sends both to rank 0 (sampling semantics stay on rank 0)."""
     dev = engine.device
     staged = _Staged(dev)
+    cdev = dev if _nccl() else "cpu"
     E = engine.hp.n_embd
     while True:
-        hdr = torch.zeros(2, dtype=torch.int64)
+        hdr = torch.zeros(2, dtype=torch.int64, device=cdev)
         dist.broadcast(hdr, src=0)
         op, T = int(hdr[0]), int(hdr[1])
         if op == OP_SHUTDOWN:
             return
-        body = torch.zeros(3, T, dtype=torch.int64)
+        body = torch.zeros(3, T, dtype=torch.int64, device=cdev)
         dist.broadcast(body, src=0)
         pos = body[1].to(dev, torch.int32)
         seq = body[2].to(dev, torch.int32)
