@@ -95,3 +95,86 @@ def test_quantize_matches_python_codec(bins, model_file, tmp_path, target):
             continue
         want = quant(t.to_f32()).tobytes()
         assert g.raw == want, f"{t.name}: q4 bytes differ from Python codec"
+
+
+KQUANTS = {
+    "q2_K": (ggml.FTYPE_MOSTLY_Q2_K, "q2_K"),
+    "q3_K": (ggml.FTYPE_MOSTLY_Q3_K_M, "q3_K"),
+    "q4_K": (ggml.FTYPE_MOSTLY_Q4_K_M, "q4_K"),
+    "q5_K": (ggml.FTYPE_MOSTLY_Q5_K_M, "q5_K"),
+    "q6_K": (ggml.FTYPE_MOSTLY_Q6_K, "q6_K"),
+}
+
+
+@pytest.fixture(scope="module")
+def model_file_k(tmp_path_factory):
+    """f16 model whose E and F are multiples of QK_K=256 (small_k)."""
+    d = tmp_path_factory.mktemp("native_k")
+    f = synthetic.build_model("small_k", ftype=ggml.FTYPE_MOSTLY_F16, seed=9)
+    path = d / "model_k_f16.bin"
+    f.save(str(path))
+    return f, path
+
+
+@pytest.mark.parametrize("target", list(KQUANTS))
+def test_quantize_kquant_matches_python_codec(bins, model_file_k, tmp_path,
+                                              target):
+    from distributedllm_amd.formats import kquants
+    f, path = model_file_k
+    out_cpp = tmp_path / f"model_{target}.bin"
+    subprocess.run([str(bins / "quantize"), str(path), str(out_cpp), target],
+                   check=True, capture_output=True)
+    got = ggml.GGMLFile.load(str(out_cpp), extended=False)
+    ftype, name = KQUANTS[target]
+    assert got.hparams.ftype == ftype
+    _, quant, _ = kquants.CODECS[name]
+    for t in f.tensors:
+        g = got.tensor_map()[t.name]
+        if len(t.ne) == 1:
+            assert g.raw == t.raw
+            continue
+        want = quant(t.to_f32().reshape(-1, t.ne[0])).tobytes()
+        assert g.raw == want, f"{t.name}: {target} bytes != Python codec"
+
+
+def test_quantize_kquant_fallback_rows(bins, model_file, tmp_path):
+    """Rows not divisible by 256 fall back per-tensor like the Python
+    provisioner (q4_K -> q5_0, q6_K -> q8_0); tiny has n_embd=64."""
+    f, path = model_file  # tiny: every 2-D tensor has cols % 256 != 0
+    for target, fb_quant, fb_gt in [
+            ("q4_K", q4.quantize_q5_0, ggml.GGML_TYPE_Q5_0),
+            ("q6_K", q4.quantize_q8_0, ggml.GGML_TYPE_Q8_0)]:
+        out_cpp = tmp_path / f"model_fb_{target}.bin"
+        subprocess.run([str(bins / "quantize"), str(path), str(out_cpp),
+                        target], check=True, capture_output=True)
+        got = ggml.GGMLFile.load(str(out_cpp), extended=False)
+        for t in f.tensors:
+            g = got.tensor_map()[t.name]
+            if len(t.ne) == 1:
+                assert g.raw == t.raw
+                continue
+            assert g.gtype == fb_gt
+            assert g.raw == fb_quant(t.to_f32()).tobytes()
+
+
+def test_cpp_kquant_dequant_roundtrip(bins, model_file_k, tmp_path):
+    """C++ to_f32 of a k-quant file: requantizing q4_K -> q8_0 through
+    the native tool matches Python dequant -> Python q8_0."""
+    from distributedllm_amd.formats import kquants
+    f, path = model_file_k
+    kfile = tmp_path / "m_q4K.bin"
+    subprocess.run([str(bins / "quantize"), str(path), str(kfile), "q4_K"],
+                   check=True, capture_output=True)
+    out = tmp_path / "m_q8.bin"
+    subprocess.run([str(bins / "quantize"), str(kfile), str(out), "q8_0"],
+                   check=True, capture_output=True)
+    got = ggml.GGMLFile.load(str(out), extended=False)
+    kf = ggml.GGMLFile.load(str(kfile), extended=False)
+    for t in kf.tensors:
+        g = got.tensor_map()[t.name]
+        if len(t.ne) == 1:
+            continue
+        deq = kquants.dequantize_q4_K(
+            np.frombuffer(t.raw, dtype=np.uint8).reshape(t.shape_rows_cols[0], -1),
+            t.ne[0]).astype(np.float32)
+        assert g.raw == q4.quantize_q8_0(deq).tobytes()
