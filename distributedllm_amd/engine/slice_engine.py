@@ -241,6 +241,78 @@ def _repack_q4_torch(t: ggml.GGMLTensor, device: str):
     return data, sc, t.gtype
 
 
+def _repack_byte_torch(t: ggml.GGMLTensor, device: str):
+    """q5_0/q5_1/q8_0 -> W_Q8B byte-stream layout with the expansion done
+    by torch integer ops ON THE GPU (bit-identical to the numpy path;
+    same rationale as _repack_q4_torch: only compressed bytes cross
+    PCIe)."""
+    rows, cols = t.shape_rows_cols
+    R, nb = rows // 16, cols // 32
+    nbp = (nb + 3) & ~3
+    raw = torch.tensor(np.frombuffer(t.raw, np.uint8), device=device)
+    if t.gtype == ggml.GGML_TYPE_Q8_0:
+        a = raw.view(rows, nb, 34)
+        alpha = a[:, :, :2].contiguous().view(torch.half).view(rows, nb
+                                                               ).float()
+        beta = torch.zeros_like(alpha)
+        vals = a[:, :, 2:] ^ 0x80                     # int8 + 128
+    else:
+        bs = 22 if t.gtype == ggml.GGML_TYPE_Q5_0 else 24
+        hoff = 2 if t.gtype == ggml.GGML_TYPE_Q5_0 else 4
+        a = raw.view(rows, nb, bs)
+        qs = a[:, :, hoff + 4:]
+        qh = a[:, :, hoff:hoff + 4].contiguous().view(torch.int32
+                                                      ).view(rows, nb)
+        # 5th bits: arithmetic >> keeps bit k at position 0 after & 1
+        bits = ((qh.unsqueeze(-1) >>
+                 torch.arange(32, device=device, dtype=torch.int32)) & 1
+                ).to(torch.uint8)
+        lo = (qs & 0xF) | (bits[:, :, :16] << 4)
+        hi = (qs >> 4) | (bits[:, :, 16:] << 4)
+        vals = torch.cat([lo, hi], dim=-1) + 112      # q5 + 112 (<= 143)
+        alpha = a[:, :, :2].contiguous().view(torch.half).view(rows, nb
+                                                               ).float()
+        if t.gtype == ggml.GGML_TYPE_Q5_0:
+            beta = torch.zeros_like(alpha)
+        else:
+            m = a[:, :, 2:4].contiguous().view(torch.half).view(rows, nb)
+            beta = m.float() + 16.0 * alpha
+    perm = torch.tensor([0, 2, 1, 3, 4, 6, 5, 7], device=device)
+    v = vals.view(rows, nb, 4, 8)[:, :, :, perm]
+    vp = torch.zeros(rows, nbp, 4, 8, dtype=torch.uint8, device=device)
+    vp[:, :nb] = v
+    shifts = torch.tensor([0, 8, 16, 24], device=device,
+                          dtype=torch.int32)
+    q32 = (vp.view(rows, nbp, 4, 2, 4).to(torch.int32) << shifts
+           ).sum(dim=-1, dtype=torch.int64).to(torch.int32)
+    qs2 = (q32.view(R, 16, nbp // 4, 4, 4, 2)     # [R][i][g4][kb][ws][2]
+           .permute(0, 2, 4, 1, 3, 5).contiguous())
+    data = torch.cat([qs2.reshape(-1),
+                      torch.zeros(512, dtype=torch.int32, device=device)])
+    al = torch.zeros(rows, nbp, dtype=torch.float32, device=device)
+    be = torch.zeros_like(al)
+    al[:, :nb] = alpha
+    be[:, :nb] = beta
+    ab = torch.stack([al, be], dim=-1).to(torch.float16)
+    sc = (ab.view(R, 16, nbp // 4, 4, 2)
+          .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
+    sc = torch.cat([sc, torch.zeros(128, dtype=torch.float16,
+                                    device=device)])
+    return data, sc, ggml.GGML_TYPE_Q8_0             # = W_Q8B
+
+
+def _repack_f16_torch(t: ggml.GGMLTensor, device: str):
+    """f16 -> [R][cols/8][16][8] tile layout, transposed on the GPU."""
+    rows, cols = t.shape_rows_cols
+    R = rows // 16
+    w = torch.tensor(np.frombuffer(t.raw, np.int16), device=device)
+    tile = (w.view(R, 16, cols // 8, 8).permute(0, 2, 1, 3)
+            .contiguous().reshape(-1))
+    data = torch.cat([tile, torch.zeros(2048, dtype=torch.int16,
+                                        device=device)])
+    return data, torch.empty(0), ggml.GGML_TYPE_F16
+
+
 def repack_mfma(t: ggml.GGMLTensor, device: str):
     """On-disk tensor -> (data, scales, wtype) in the MFMA tile layout.
 
@@ -300,6 +372,8 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             np.concatenate([scales.reshape(-1),
                             np.zeros(128, np.float16)])).to(device)
         return data, sc, t.gtype
+    if t.gtype in _BYTE_GTYPES and device != "cpu":
+        return _repack_byte_torch(t, device)
     if t.gtype in _BYTE_GTYPES or t.gtype in _K32_GTYPES or \
             t.gtype in _K16_GTYPES:
         nb = cols // 32
@@ -344,6 +418,8 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
             np.concatenate([scales.reshape(-1),
                             np.zeros(128, np.float16)])).to(device)
         return data, sc, wt_out
+    if t.gtype == ggml.GGML_TYPE_F16 and device != "cpu":
+        return _repack_f16_torch(t, device)
     if t.gtype == ggml.GGML_TYPE_F16:
         w = np.frombuffer(t.raw, np.float16).reshape(rows, cols)
         tile = np.ascontiguousarray(

@@ -150,3 +150,34 @@ def test_torch_q4_repack_matches_numpy():
         assert wt_np == wt_th
         assert torch.equal(d_np, d_th)
         assert torch.equal(s_np.view(torch.int16), s_th.view(torch.int16))
+
+
+def test_torch_byte_repack_matches_numpy():
+    """Torch q5_0/q5_1/q8_0 byte-stream repack == numpy path, bit for
+    bit (same contract as the q4 port)."""
+    import torch
+    from distributedllm_amd.engine import slice_engine as SE
+    from distributedllm_amd.formats import ggml, synthetic
+    for ft in (ggml.FTYPE_MOSTLY_Q5_0, ggml.FTYPE_MOSTLY_Q5_1,
+               ggml.FTYPE_MOSTLY_Q8_0):
+        f = synthetic.build_model("small", seed=5, ftype=ft)
+        t = next(x for x in f.tensors
+                 if x.name.endswith("attention.wq.weight"))
+        d_np, s_np, wt_np = SE.repack_mfma(t, "cpu")        # numpy path
+        d_th, s_th, wt_th = SE._repack_byte_torch(t, "cpu")
+        assert wt_np == wt_th
+        assert torch.equal(d_np, d_th), ggml.TYPE_NAMES[t.gtype]
+        assert torch.equal(s_np.view(torch.int16), s_th.view(torch.int16))
+
+
+def test_torch_f16_repack_matches_numpy():
+    import torch
+    from distributedllm_amd.engine import slice_engine as SE
+    from distributedllm_amd.formats import ggml, synthetic
+    f = synthetic.build_model("small", seed=6, ftype=ggml.FTYPE_MOSTLY_F16)
+    t = next(x for x in f.tensors
+             if x.name.endswith("feed_forward.w1.weight"))
+    d_np, _, wt_np = SE.repack_mfma(t, "cpu")
+    d_th, _, wt_th = SE._repack_f16_torch(t, "cpu")
+    assert wt_np == wt_th
+    assert torch.equal(d_np.view(torch.int16), d_th.view(torch.int16))
