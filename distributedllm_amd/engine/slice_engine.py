@@ -241,6 +241,47 @@ def _repack_q4_torch(t: ggml.GGMLTensor, device: str):
     return data, sc, t.gtype
 
 
+def _pack_bytes_torch(vals, alpha, beta, rows, cols, per16, device):
+    """(u8 values [rows, nb, 32], alpha, beta) -> (data, scales, wtype)
+    in the W_Q8B/W_Q8B16 tile layout, all torch ops (device-agnostic,
+    bit-identical to the numpy packing in repack_mfma)."""
+    R, nb = rows // 16, cols // 32
+    nbp = (nb + 3) & ~3
+    perm = torch.tensor([0, 2, 1, 3, 4, 6, 5, 7], device=device)
+    v = vals.reshape(rows, nb, 4, 8)[:, :, :, perm]
+    vp = torch.zeros(rows, nbp, 4, 8, dtype=torch.uint8, device=device)
+    vp[:, :nb] = v
+    shifts = torch.tensor([0, 8, 16, 24], device=device,
+                          dtype=torch.int32)
+    q32 = (vp.view(rows, nbp, 4, 2, 4).to(torch.int32) << shifts
+           ).sum(dim=-1, dtype=torch.int64).to(torch.int32)
+    qs2 = (q32.view(R, 16, nbp // 4, 4, 4, 2)     # [R][i][g4][kb][ws][2]
+           .permute(0, 2, 4, 1, 3, 5).contiguous())
+    data = torch.cat([qs2.reshape(-1),
+                      torch.zeros(512, dtype=torch.int32, device=device)])
+    if per16:
+        al = torch.zeros(rows, nbp, 2, dtype=torch.float32, device=device)
+        be = torch.zeros_like(al)
+        al[:, :nb] = alpha
+        be[:, :nb] = beta
+        ab = torch.stack([al, be], dim=-1).to(torch.float16)
+        sc = (ab.view(R, 16, nbp // 4, 4, 2, 2)
+              .permute(0, 2, 4, 1, 3, 5).contiguous().reshape(-1))
+        wt_out = W_Q8B16
+    else:
+        al = torch.zeros(rows, nbp, dtype=torch.float32, device=device)
+        be = torch.zeros_like(al)
+        al[:, :nb] = alpha
+        be[:, :nb] = beta
+        ab = torch.stack([al, be], dim=-1).to(torch.float16)
+        sc = (ab.view(R, 16, nbp // 4, 4, 2)
+              .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
+        wt_out = ggml.GGML_TYPE_Q8_0              # = W_Q8B
+    sc = torch.cat([sc, torch.zeros(128, dtype=torch.float16,
+                                    device=device)])
+    return data, sc, wt_out
+
+
 def _repack_byte_torch(t: ggml.GGMLTensor, device: str):
     """q5_0/q5_1/q8_0 -> W_Q8B byte-stream layout with the expansion done
     by torch integer ops ON THE GPU (bit-identical to the numpy path;
@@ -277,28 +318,129 @@ def _repack_byte_torch(t: ggml.GGMLTensor, device: str):
         else:
             m = a[:, :, 2:4].contiguous().view(torch.half).view(rows, nb)
             beta = m.float() + 16.0 * alpha
-    perm = torch.tensor([0, 2, 1, 3, 4, 6, 5, 7], device=device)
-    v = vals.view(rows, nb, 4, 8)[:, :, :, perm]
-    vp = torch.zeros(rows, nbp, 4, 8, dtype=torch.uint8, device=device)
-    vp[:, :nb] = v
-    shifts = torch.tensor([0, 8, 16, 24], device=device,
-                          dtype=torch.int32)
-    q32 = (vp.view(rows, nbp, 4, 2, 4).to(torch.int32) << shifts
-           ).sum(dim=-1, dtype=torch.int64).to(torch.int32)
-    qs2 = (q32.view(R, 16, nbp // 4, 4, 4, 2)     # [R][i][g4][kb][ws][2]
-           .permute(0, 2, 4, 1, 3, 5).contiguous())
-    data = torch.cat([qs2.reshape(-1),
-                      torch.zeros(512, dtype=torch.int32, device=device)])
-    al = torch.zeros(rows, nbp, dtype=torch.float32, device=device)
-    be = torch.zeros_like(al)
-    al[:, :nb] = alpha
-    be[:, :nb] = beta
-    ab = torch.stack([al, be], dim=-1).to(torch.float16)
-    sc = (ab.view(R, 16, nbp // 4, 4, 2)
-          .permute(0, 2, 1, 3, 4).contiguous().reshape(-1))
-    sc = torch.cat([sc, torch.zeros(128, dtype=torch.float16,
-                                    device=device)])
-    return data, sc, ggml.GGML_TYPE_Q8_0             # = W_Q8B
+    return _pack_bytes_torch(vals, alpha, beta, rows, cols, False, device)
+
+
+def _unpack_scales_k4_torch(p):
+    """torch port of kquants._unpack_scales_k4: [..., 12] u8 ->
+    (sc, mn) u8 [..., 8]."""
+    sc = torch.empty(p.shape[:-1] + (8,), dtype=torch.uint8,
+                     device=p.device)
+    mn = torch.empty_like(sc)
+    sc[..., 0:4] = p[..., 0:4] & 63
+    mn[..., 0:4] = p[..., 4:8] & 63
+    sc[..., 4:8] = (p[..., 8:12] & 0xF) | ((p[..., 0:4] >> 6) << 4)
+    mn[..., 4:8] = (p[..., 8:12] >> 4) | ((p[..., 4:8] >> 6) << 4)
+    return sc, mn
+
+
+def _repack_kquant_torch(t: ggml.GGMLTensor, device: str):
+    """k-quant (q2_K..q6_K) -> W_Q8B/W_Q8B16 byte-stream layout with the
+    super-block expansion done by torch integer ops ON THE GPU —
+    bit-identical to the numpy _kquant_byte_values + packing path."""
+    from ..formats import kquants as KQ
+    rows, cols = t.shape_rows_cols
+    nsb = cols // KQ.QK_K
+    raw = torch.tensor(np.frombuffer(t.raw, np.uint8), device=device)
+
+    def f16col(b, lo, hi):
+        return (b[:, :, lo:hi].contiguous().view(torch.half)
+                .view(rows, nsb).float())
+
+    if t.gtype == ggml.GGML_TYPE_Q4_K:
+        b = raw.view(rows, nsb, KQ.Q4_K_BLOCK_BYTES)
+        d, dmin = f16col(b, 0, 2), f16col(b, 2, 4)
+        sc, mn = _unpack_scales_k4_torch(b[:, :, 4:16])
+        qs = b[:, :, 16:144].reshape(rows, nsb, 4, 32)
+        q = torch.empty(rows, nsb, 8, 32, dtype=torch.uint8,
+                        device=device)
+        q[:, :, 0::2] = qs & 0xF
+        q[:, :, 1::2] = qs >> 4
+        alpha = (d.unsqueeze(-1) * sc.float()).reshape(rows, nsb * 8)
+        beta = (-(dmin.unsqueeze(-1) * mn.float())).reshape(rows, nsb * 8)
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        return _pack_bytes_torch(vals, alpha, beta, rows, cols, False,
+                                 device)
+    if t.gtype == ggml.GGML_TYPE_Q5_K:
+        b = raw.view(rows, nsb, KQ.Q5_K_BLOCK_BYTES)
+        d, dmin = f16col(b, 0, 2), f16col(b, 2, 4)
+        sc, mn = _unpack_scales_k4_torch(b[:, :, 4:16])
+        qh = b[:, :, 16:48]
+        ql = b[:, :, 48:176].reshape(rows, nsb, 4, 32)
+        q = torch.empty(rows, nsb, 8, 32, dtype=torch.uint8,
+                        device=device)
+        for j in range(4):
+            q[:, :, 2 * j] = (ql[:, :, j] & 0xF) | \
+                (((qh >> (2 * j)) & 1) << 4)
+            q[:, :, 2 * j + 1] = (ql[:, :, j] >> 4) | \
+                (((qh >> (2 * j + 1)) & 1) << 4)
+        alpha = (d.unsqueeze(-1) * sc.float()).reshape(rows, nsb * 8)
+        beta = (-(dmin.unsqueeze(-1) * mn.float())).reshape(rows, nsb * 8)
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        return _pack_bytes_torch(vals, alpha, beta, rows, cols, False,
+                                 device)
+    if t.gtype == ggml.GGML_TYPE_Q6_K:
+        b = raw.view(rows, nsb, KQ.Q6_K_BLOCK_BYTES)
+        ql = b[:, :, 0:128].reshape(rows, nsb, 2, 2, 32)
+        qh = b[:, :, 128:192].reshape(rows, nsb, 2, 32)
+        sc = (b[:, :, 192:208].contiguous().view(torch.int8)
+              .view(rows, nsb, 16).float())
+        d = f16col(b, 208, 210)
+        q = torch.empty(rows, nsb, 2, 4, 32, dtype=torch.int16,
+                        device=device)
+        q[:, :, :, 0] = ((ql[:, :, :, 0] & 0xF) |
+                         (((qh >> 0) & 3) << 4)).to(torch.int16)
+        q[:, :, :, 1] = ((ql[:, :, :, 1] & 0xF) |
+                         (((qh >> 2) & 3) << 4)).to(torch.int16)
+        q[:, :, :, 2] = ((ql[:, :, :, 0] >> 4) |
+                         (((qh >> 4) & 3) << 4)).to(torch.int16)
+        q[:, :, :, 3] = ((ql[:, :, :, 1] >> 4) |
+                         (((qh >> 6) & 3) << 4)).to(torch.int16)
+        vals = (q + 96).to(torch.uint8).reshape(rows, nsb * 8, 32)
+        alpha16 = (d.unsqueeze(-1) * sc).reshape(rows, nsb * 8, 2)
+        beta16 = torch.zeros_like(alpha16)
+        return _pack_bytes_torch(vals, alpha16, beta16, rows, cols, True,
+                                 device)
+    if t.gtype == ggml.GGML_TYPE_Q2_K:
+        b = raw.view(rows, nsb, KQ.Q2_K_BLOCK_BYTES)
+        sc = (b[:, :, 0:16] & 0xF).float()
+        mn = (b[:, :, 0:16] >> 4).float()
+        qs = b[:, :, 16:80].reshape(rows, nsb, 2, 32)
+        d, dmin = f16col(b, 80, 82), f16col(b, 82, 84)
+        q = torch.empty(rows, nsb, 2, 4, 32, dtype=torch.uint8,
+                        device=device)
+        for j in range(4):
+            q[:, :, :, j] = (qs >> (2 * j)) & 3
+        vals = (q + 128).reshape(rows, nsb * 8, 32)
+        alpha16 = (d.unsqueeze(-1) * sc).reshape(rows, nsb * 8, 2)
+        beta16 = (-(dmin.unsqueeze(-1) * mn)).reshape(rows, nsb * 8, 2)
+        return _pack_bytes_torch(vals, alpha16, beta16, rows, cols, True,
+                                 device)
+    assert t.gtype == ggml.GGML_TYPE_Q3_K, t.gtype
+    b = raw.view(rows, nsb, KQ.Q3_K_BLOCK_BYTES)
+    hm = b[:, :, 0:32]
+    qs = b[:, :, 32:96].reshape(rows, nsb, 2, 32)
+    p = b[:, :, 96:108]
+    sc = torch.empty(rows, nsb, 16, dtype=torch.int16, device=device)
+    a0, a1, tt = p[..., 0:4], p[..., 4:8], p[..., 8:12]
+    sc[..., 0:4] = ((a0 & 0xF) | (((tt >> 0) & 3) << 4)).to(torch.int16)
+    sc[..., 4:8] = ((a1 & 0xF) | (((tt >> 2) & 3) << 4)).to(torch.int16)
+    sc[..., 8:12] = ((a0 >> 4) | (((tt >> 4) & 3) << 4)).to(torch.int16)
+    sc[..., 12:16] = ((a1 >> 4) | (((tt >> 6) & 3) << 4)).to(torch.int16)
+    sc = (sc - 32).float()
+    d = f16col(b, 108, 110)
+    q = torch.empty(rows, nsb, 2, 4, 32, dtype=torch.int16, device=device)
+    for half in range(2):
+        for j in range(4):
+            low = ((qs[:, :, half] >> (2 * j)) & 3).to(torch.int16)
+            hi = ((hm >> (half * 4 + j)) & 1).to(torch.int16)
+            q[:, :, half, j] = low - torch.where(
+                hi != 0, torch.zeros_like(hi), torch.full_like(hi, 4))
+    vals = (q + 128).to(torch.uint8).reshape(rows, nsb * 8, 32)
+    alpha16 = (d.unsqueeze(-1) * sc).reshape(rows, nsb * 8, 2)
+    beta16 = torch.zeros_like(alpha16)
+    return _pack_bytes_torch(vals, alpha16, beta16, rows, cols, True,
+                             device)
 
 
 def _repack_f16_torch(t: ggml.GGMLTensor, device: str):
@@ -374,6 +516,9 @@ def repack_mfma(t: ggml.GGMLTensor, device: str):
         return data, sc, t.gtype
     if t.gtype in _BYTE_GTYPES and device != "cpu":
         return _repack_byte_torch(t, device)
+    if (t.gtype in _K32_GTYPES or t.gtype in _K16_GTYPES) and \
+            device != "cpu":
+        return _repack_kquant_torch(t, device)
     if t.gtype in _BYTE_GTYPES or t.gtype in _K32_GTYPES or \
             t.gtype in _K16_GTYPES:
         nb = cols // 32

@@ -181,3 +181,23 @@ def test_torch_f16_repack_matches_numpy():
     d_th, _, wt_th = SE._repack_f16_torch(t, "cpu")
     assert wt_np == wt_th
     assert torch.equal(d_np.view(torch.int16), d_th.view(torch.int16))
+
+
+def test_torch_kquant_repack_matches_numpy():
+    """Torch q2_K..q6_K super-block repack == numpy path, bit for bit."""
+    import torch
+    from distributedllm_amd.engine import slice_engine as SE
+    from distributedllm_amd.formats import ggml, synthetic
+    for ft in (ggml.FTYPE_MOSTLY_Q2_K, ggml.FTYPE_MOSTLY_Q3_K_M,
+               ggml.FTYPE_MOSTLY_Q4_K_M, ggml.FTYPE_MOSTLY_Q5_K_M,
+               ggml.FTYPE_MOSTLY_Q6_K):
+        f = synthetic.build_model("small_k", seed=8, ftype=ft)
+        for tname in ("attention.wq.weight", "feed_forward.w1.weight"):
+            t = next(x for x in f.tensors if x.name.endswith(tname))
+            d_np, s_np, wt_np = SE.repack_mfma(t, "cpu")    # numpy path
+            d_th, s_th, wt_th = SE._repack_kquant_torch(t, "cpu")
+            assert wt_np == wt_th
+            assert torch.equal(d_np, d_th), ggml.TYPE_NAMES[t.gtype]
+            assert torch.equal(s_np.view(torch.int16),
+                               s_th.view(torch.int16)), \
+                ggml.TYPE_NAMES[t.gtype]

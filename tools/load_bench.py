@@ -19,7 +19,8 @@ from distributedllm_amd.models.llama import (
     PRESETS, layer_tensor_names)
 
 FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
-          "f16": ggml.FTYPE_MOSTLY_F16}
+          "f16": ggml.FTYPE_MOSTLY_F16, "q4_K": ggml.FTYPE_MOSTLY_Q4_K_M,
+          "q6_K": ggml.FTYPE_MOSTLY_Q6_K}
 
 
 def build_streaming(path: str, preset, ftype: int, seed: int = 0,
