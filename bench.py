@@ -167,8 +167,13 @@ def main() -> int:
     # the prefill steps run as extra (graph-replayed) warmup: real decode
     # work filling the KV with real activations — nothing in the timed
     # region is skipped or cached, it just starts at a stated depth
-    elapsed = timed_decode(pipe, args.steps,
-                           args.warmup + args.prefill_depth, device)
+    # DLLM_NO_GRAPH=1: eager launches instead of hipGraph replay —
+    # needed under rocprofv3 --pmc (per-dispatch counter instrumentation
+    # crashes on graph-replayed kernels); timing loses the launch-bound
+    # overlap, so only use for counter collection, not headline numbers
+    elapsed = timed_decode(
+        pipe, args.steps, args.warmup + args.prefill_depth, device,
+        use_graphs=os.environ.get("DLLM_NO_GRAPH", "0") != "1")
 
     # MAX over ranks
     t = torch.tensor([elapsed], dtype=torch.float64,
