@@ -288,8 +288,7 @@ def _repack_byte_torch(t: ggml.GGMLTensor, device: str):
     same rationale as _repack_q4_torch: only compressed bytes cross
     PCIe)."""
     rows, cols = t.shape_rows_cols
-    R, nb = rows // 16, cols // 32
-    nbp = (nb + 3) & ~3
+    nb = cols // 32
     raw = torch.tensor(np.frombuffer(t.raw, np.uint8), device=device)
     if t.gtype == ggml.GGML_TYPE_Q8_0:
         a = raw.view(rows, nb, 34)
