@@ -178,3 +178,74 @@ def test_cpp_kquant_dequant_roundtrip(bins, model_file_k, tmp_path):
             np.frombuffer(t.raw, dtype=np.uint8).reshape(t.shape_rows_cols[0], -1),
             t.ne[0]).astype(np.float32)
         assert g.raw == q4.quantize_q8_0(deq).tobytes()
+
+
+def test_quantize_edge_case_blocks_match_python(bins, tmp_path):
+    """Degenerate block contents (all-zero, constant, subnormal-amax,
+    near-f16-max) must produce identical bytes from the C++ tool and the
+    Python codecs — these hit the safe_inv/zero-scale paths."""
+    from distributedllm_amd.formats import kquants, synthetic
+    rows, cols = 16, 512          # k-quant compatible (cols % 256 == 0)
+    rng = np.random.default_rng(11)
+    w = (rng.standard_normal((rows, cols)) * 0.02).astype(np.float32)
+    w[0, :] = 0.0                               # all-zero rows
+    w[1, :] = 1.0                               # constant row
+    w[2, :] = -1.0
+    w[3, :256] = 1e-42                          # subnormal amax
+    w[4, :] = 3.0e4                             # near f16 max after /q
+    w[5, ::2] = -2.5e4
+    w[6, :32] = 0.0                             # zero block inside a row
+    f = synthetic.build_model("tiny", ftype=ggml.FTYPE_MOSTLY_F16, seed=1)
+    t = ggml.GGMLTensor.from_f32("edge.weight", w, ggml.GGML_TYPE_F16)
+    f.tensors.append(t)
+    path = tmp_path / "edge_f16.bin"
+    f.save(str(path))
+    w = w.astype(np.float16).astype(np.float32)  # what the file stores
+    classic = {"q4_0": q4.quantize_q4_0, "q4_1": q4.quantize_q4_1,
+               "q5_0": q4.quantize_q5_0, "q5_1": q4.quantize_q5_1,
+               "q8_0": q4.quantize_q8_0}
+    for target, quant in classic.items():
+        out = tmp_path / f"edge_{target}.bin"
+        subprocess.run([str(bins / "quantize"), str(path), str(out),
+                        target], check=True, capture_output=True)
+        got = ggml.GGMLFile.load(str(out), extended=False)
+        g = got.tensor_map()["edge.weight"]
+        assert g.raw == quant(w).tobytes(), target
+    for target in ("q2_K", "q3_K", "q4_K", "q5_K", "q6_K"):
+        out = tmp_path / f"edge_{target}.bin"
+        subprocess.run([str(bins / "quantize"), str(path), str(out),
+                        target], check=True, capture_output=True)
+        got = ggml.GGMLFile.load(str(out), extended=False)
+        g = got.tensor_map()["edge.weight"]
+        _, quant, _ = kquants.CODECS[target]
+        assert g.raw == quant(w).tobytes(), target
+
+
+def test_torch_repack_edge_case_bits(tmp_path):
+    """Same degenerate contents through the torch repack vs numpy repack
+    (inf/nan f16 scale patterns from random-byte files are also covered
+    by the GPU load bench; here we assert bit equality on the codec-
+    produced layouts)."""
+    import torch
+    from distributedllm_amd.engine import slice_engine as SE
+    from distributedllm_amd.formats import kquants
+    rows, cols = 16, 512
+    rng = np.random.default_rng(12)
+    w = (rng.standard_normal((rows, cols)) * 0.02).astype(np.float32)
+    w[0, :] = 0.0
+    w[3, :256] = 1e-42
+    w[4, :] = 3.0e4
+    for gt, quant in [
+            (ggml.GGML_TYPE_Q5_0, q4.quantize_q5_0),
+            (ggml.GGML_TYPE_Q8_0, q4.quantize_q8_0),
+            (ggml.GGML_TYPE_Q4_K, kquants.quantize_q4_K),
+            (ggml.GGML_TYPE_Q6_K, kquants.quantize_q6_K)]:
+        t = ggml.GGMLTensor(name="e", ne=(cols, rows), gtype=gt,
+                            raw=quant(w).tobytes())
+        d_np, s_np, wt_np = SE.repack_mfma(t, "cpu")
+        fn = (SE._repack_byte_torch
+              if gt in (ggml.GGML_TYPE_Q5_0, ggml.GGML_TYPE_Q8_0)
+              else SE._repack_kquant_torch)
+        d_th, s_th, wt_th = fn(t, "cpu")
+        assert wt_np == wt_th and torch.equal(d_np, d_th)
+        assert torch.equal(s_np.view(torch.int16), s_th.view(torch.int16))

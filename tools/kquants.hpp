@@ -35,6 +35,11 @@ inline float clipf(float v, float lo, float hi) {
     return v < lo ? lo : (v > hi ? hi : v);
 }
 
+// np.maximum semantics: ties (incl. +0 vs -0) return the SECOND arg —
+// matters only for the sign of a zero scale in all-zero groups, but the
+// codecs are byte-parity-tested against the numpy reference
+inline float npmax(float a, float b) { return a > b ? a : b; }
+
 inline void put_f16(uint8_t* p, float v) {
     const uint16_t h = ggmlio::f32_to_f16(v);
     p[0] = (uint8_t)(h & 0xFF);
@@ -191,10 +196,10 @@ inline void quantize_block_q6_K(const float* x, uint8_t* out) {
             gmin = std::min(gmin, b[i]);
         }
         // asymmetric range: q-32 in [-32, 31], neither side clips
-        gscale[g] = std::max(gmax / 31.0f, gmin / -32.0f);
+        gscale[g] = npmax(gmax / 31.0f, gmin / -32.0f);
     }
     float smax = gscale[0];
-    for (int g = 1; g < 16; ++g) smax = std::max(smax, gscale[g]);
+    for (int g = 1; g < 16; ++g) smax = npmax(smax, gscale[g]);
     const float d = f16rt(smax / 127.0f);
     const float id = safe_inv(d);
     int8_t sc[16];
@@ -333,10 +338,10 @@ inline void quantize_block_q3_K(const float* x, uint8_t* out) {
             gmin = std::min(gmin, b[i]);
         }
         // asymmetric range: q-4 in [-4, 3]
-        gscale[g] = std::max(gmax / 3.0f, gmin / -4.0f);
+        gscale[g] = npmax(gmax / 3.0f, gmin / -4.0f);
     }
     float smax = gscale[0];
-    for (int g = 1; g < 16; ++g) smax = std::max(smax, gscale[g]);
+    for (int g = 1; g < 16; ++g) smax = npmax(smax, gscale[g]);
     const float d = f16rt(smax / 31.0f);
     const float id = safe_inv(d);
     int8_t sc[16];
