@@ -32,6 +32,12 @@ from distributedllm_amd.parallel.pipeline import (
 FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
           "q5_0": ggml.FTYPE_MOSTLY_Q5_0, "q5_1": ggml.FTYPE_MOSTLY_Q5_1,
           "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
+          # k-quants ride the W_Q8B/W_Q8B16 byte streams in HBM (1 B
+          # per weight + scale planes), so their decode throughput is
+          # the byte-path number regardless of the smaller disk size
+          "q2_K": ggml.FTYPE_MOSTLY_Q2_K, "q3_K": ggml.FTYPE_MOSTLY_Q3_K_M,
+          "q4_K": ggml.FTYPE_MOSTLY_Q4_K_M,
+          "q5_K": ggml.FTYPE_MOSTLY_Q5_K_M, "q6_K": ggml.FTYPE_MOSTLY_Q6_K,
           "f16": ggml.FTYPE_MOSTLY_F16, "f32": ggml.FTYPE_ALL_F32}
 
 
@@ -112,8 +118,11 @@ def main() -> int:
         E, F, V = preset.n_embd, preset.n_ff, 32000
         n_weights = (max_layers * (4 * E * E + 3 * E * F) + 2 * V * E)
         bpw = {"q4_0": 0.5625, "q4_1": 0.625, "q5_0": 1.125,
-               "q5_1": 1.125, "q8_0": 1.125, "f16": 2.0,
-               "f32": 4.0}[args.ftype]  # byte formats: 1B + scales
+               "q5_1": 1.125, "q8_0": 1.125,
+               "q2_K": 1.25, "q3_K": 1.25, "q4_K": 1.125, "q5_K": 1.125,
+               "q6_K": 1.25, "f16": 2.0,
+               "f32": 4.0}[args.ftype]  # HBM-resident bytes/weight
+        #         (byte formats incl. k-quants: 1B + scale planes)
         w_bytes = n_weights * bpw * 1.1  # repack padding/scales margin
         fit = max(1, int((260e9 - w_bytes) // max(kv_bytes, 1)))
         if fit >= n_lanes:
