@@ -317,3 +317,30 @@ def test_four_stage_two_lane_pipeline_matches_single():
     rank count)."""
     assert _run_cluster(4, 29544, 8, n_lanes=2, n_layer=4) == \
         _single_reference_tokens(8, n_layer=4)
+
+
+def test_bench_driver_contract_world2():
+    """The driver's N>1 launch shape (torch.distributed.run, one rank
+    per GPU) must work end-to-end: rendezvous on 127.0.0.1, gloo on
+    CPU, per-rank layer partition, MAX-over-ranks timing, ONE JSON line
+    from rank 0 with the world-2 aggregate."""
+    import subprocess
+    import sys
+    port = str(29500 + os.getpid() % 500)
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", port, "bench.py", "--gpus", "2",
+         "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout
+    r = json.loads(lines[0])
+    assert r["n_gpus"] == 2 and r["value"] > 0
+    assert r["config"]["parallelism"] == "pp2"
+    assert r["config"]["backend"] == "gloo"
+    # weak scaling: per-rank work fixed, global batch grows with N
+    assert r["config"]["global_batch"] == \
+        r["config"]["mbs"] * r["config"]["micro_batches"]
