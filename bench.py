@@ -166,11 +166,17 @@ def main() -> int:
         # the CPU path exists to keep the driver contract testable; a
         # deep steady-state KV is a GPU-measurement concern
         args.prefill_depth = min(args.prefill_depth, 8)
-    # KV-depth guard: prefill + warmup + steps all advance positions
-    depth_end = args.prefill_depth + args.warmup + args.steps
+    # KV-depth guard: prefill + warmup + steps all advance positions —
+    # and so do capture_graphs' 2 eager warmup steps on the CUDA path
+    # (at a clamped-to-ctx config, forgetting them writes the last KV
+    # rows at pos >= n_ctx: out of bounds)
+    cap_steps = 2 if device == "cuda" else 0
+    depth_end = cap_steps + args.prefill_depth + args.warmup + args.steps
     if depth_end > args.ctx:
-        args.prefill_depth = max(0, args.ctx - args.warmup - args.steps)
-        depth_end = args.prefill_depth + args.warmup + args.steps
+        args.prefill_depth = max(0, args.ctx - args.warmup - args.steps
+                                 - cap_steps)
+        depth_end = (cap_steps + args.prefill_depth + args.warmup +
+                     args.steps)
     pipe = DecodePipeline(eng, cfg, rank=rank, world=world,
                           engines=engines)
     # the prefill steps run as extra (graph-replayed) warmup: real decode
@@ -214,8 +220,7 @@ def main() -> int:
                 "global_batch": cfg.global_batch,
                 "seq_len": depth_end,
                 "prefill_depth": args.prefill_depth,
-                "kv_depth_timed": [args.prefill_depth + args.warmup,
-                                   depth_end],
+                "kv_depth_timed": [depth_end - args.steps, depth_end],
                 "n_ctx": args.ctx,
                 "parallelism": f"pp{n_gpus}",
                 "micro_batches": n_mb,
