@@ -416,6 +416,12 @@ def pipeline_generate(pipe: DecodePipeline, prompt_ids, max_steps: int,
     engine/sampler.py)."""
     assert greedy or temperature > 0.0, \
         "non-greedy decode needs temperature > 0"
+    n_prompt = len(prompt_ids)
+    n_ctx = int(getattr(pipe.engine, "n_ctx", n_prompt + max_steps))
+    if n_prompt + max_steps > n_ctx:
+        raise ValueError(
+            f"prompt ({n_prompt}) + max_steps ({max_steps}) exceeds the "
+            f"engine context {n_ctx} — KV positions would go out of range")
     pipe.temperature = float(temperature)  # > 0 wins over greedy
     pipe.prime(prompt_ids)
     out = []

@@ -344,3 +344,14 @@ def test_bench_driver_contract_world2():
     # weak scaling: per-rank work fixed, global batch grows with N
     assert r["config"]["global_batch"] == \
         r["config"]["mbs"] * r["config"]["micro_batches"]
+
+
+def test_pipeline_generate_context_overflow_rejected():
+    from distributedllm_amd.parallel.pipeline import pipeline_generate
+    f = synthetic.build_model("tiny", seed=0)
+    ex = slicer.make_extra_layers(f)
+    eng = TorchSliceEngine.from_ggml(f, n_ctx=16, max_batch=2)
+    eng.attach_extra(ex)
+    pipe = DecodePipeline(eng, PipelineConfig(mbs=2, n_mb=1, device="cpu"))
+    with pytest.raises(ValueError, match="context"):
+        pipeline_generate(pipe, list(range(3, 13)), max_steps=10)
