@@ -48,11 +48,14 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=16)
     ap.add_argument("--model", default="open_llama_3b")
     ap.add_argument("--ftype", default="q4_0", choices=list(FTYPES))
-    ap.add_argument("--mbs", type=int, default=64,
-                    help="sequences per micro-batch (engine processes up "
-                         "to 64 tokens = 4 MFMA column tiles per kernel; "
-                         "decode is HBM-bound so tokens/s scales ~linearly "
-                         "with batch)")
+    ap.add_argument("--mbs", type=int, default=None,
+                    help="sequences per micro-batch (default 64 = one "
+                         "engine decode tile; decode is HBM-bound so "
+                         "tokens/s scales ~linearly with batch). When "
+                         "the KV fit forces a single lane and --mbs is "
+                         "not given, the lane auto-widens (WIDE decode: "
+                         "one T-token forward pays the weight stream "
+                         "once for the whole batch)")
     ap.add_argument("--ctx", type=int, default=2048)
     ap.add_argument("--lanes", "--single-gpu-mbs", dest="lanes",
                     type=int, default=5,
@@ -110,6 +113,9 @@ def main() -> int:
     # only when >=3 lanes' KV sets fit in HBM next to the model weights
     # (~260 GB usable of 288). All ranks must agree on k: size with the
     # largest per-rank slice (rank 0 holds the most layers).
+    mbs_explicit = args.mbs is not None
+    if args.mbs is None:
+        args.mbs = 64
     n_lanes = args.lanes
     if n_lanes > 1:
         max_layers = parts[0][1]
@@ -130,8 +136,15 @@ def main() -> int:
         elif fit >= 3:
             n_lanes = fit             # auto-shrink (3-5 measured best)
         else:
-            n_lanes = 1               # 2 lanes always lost at mbs 64;
-            #                           wide single-lane takes over
+            n_lanes = 1               # 2 lanes always lost at mbs 64
+            if device == "cuda" and not mbs_explicit:
+                # WIDE single-lane decode (BASELINE.md WIDE rows): use
+                # the KV room for one wide batch instead of stream
+                # lanes — measured +14-45% on the 7B-70B ladder
+                per_seq = kv_bytes // max(args.mbs, 1)
+                wide = int((260e9 - w_bytes) // max(per_seq, 1))
+                args.mbs = max(args.mbs,
+                               min(320, (wide // 64) * 64))
     n_mb = n_lanes * max(world, 1) if world > 1 else n_lanes
     cfg = PipelineConfig(mbs=args.mbs, n_mb=n_mb, device=device)
 
