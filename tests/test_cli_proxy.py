@@ -313,3 +313,24 @@ def test_cli_clean_error_on_unreachable_node(tmp_path, capsys):
     assert rc == 1
     err = capsys.readouterr().err
     assert "error:" in err and "Traceback" not in err
+
+
+def test_tcp_cluster_bench_tool(tmp_path):
+    """tools/tcp_cluster_bench.py (BASELINE config 1 driver) runs the
+    whole provision -> generate flow against real local nodes."""
+    import json as _json
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools",
+                                      "tcp_cluster_bench.py"),
+         "--nodes", "2", "--model", "tiny", "--tokens", "4",
+         "--base-port", str(21870 + os.getpid() % 500),
+         "--workdir", str(tmp_path)],
+        capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines()
+            if l.startswith("{") and "tok_s_warm" in l][-1]
+    r = _json.loads(line)
+    assert r["tokens"] == 4 and r["tok_s_warm"] > 0
