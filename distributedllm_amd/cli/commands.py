@@ -311,6 +311,12 @@ class BatchGenerateCommand(Command):
         p.add_argument("--top-k", type=int, default=0)
         p.add_argument("--top-p", type=float, default=1.0)
         p.add_argument("--seed", type=int, default=None)
+        p.add_argument("--speculative", action="store_true",
+                       help="prompt-lookup speculative decoding (greedy "
+                            "single-prompt only): multiple tokens per "
+                            "forward when the continuation repeats "
+                            "earlier text; output is token-exact with "
+                            "plain greedy")
 
     def __call__(self, args) -> int:
         import time
@@ -336,6 +342,25 @@ class BatchGenerateCommand(Command):
         eng = engine_for_slice(f, n_ctx=args.ctx, max_batch=n_slots)
         eng.attach_extra(slicer.make_extra_layers(f))
         tok = Tokenizer(f.vocab)
+
+        if args.speculative:
+            if len(prompts) != 1 or not args.greedy:
+                print("--speculative needs exactly one prompt and "
+                      "--greedy", file=sys.stderr)
+                return 2
+            from ..serving.speculative import SpecStats, pld_generate
+            st = SpecStats()
+            t0 = time.perf_counter()
+            out = pld_generate(eng, tok.encode(prompts[0], bos=True),
+                               args.num_tokens, stats=st)
+            dt = time.perf_counter() - t0
+            print(f"[0] {prompts[0]!r} -> {tok.decode(out)!r}")
+            print(f"[{len(out)} tokens in {dt:.2f}s = "
+                  f"{len(out) / max(dt, 1e-9):.1f} tok/s, "
+                  f"{st.forwards} forwards = "
+                  f"{len(out) / max(st.forwards, 1):.2f} tok/forward]",
+                  file=sys.stderr)
+            return 0
 
         bat = ContinuousBatcher(eng, max_slots=n_slots)
         reqs = []

@@ -642,3 +642,45 @@ def test_argmax_wide_rows():
     ids = eng.argmax(lg)
     assert torch.equal(ids.cpu(),
                        lg.argmax(dim=-1).to(torch.int32).cpu())
+
+
+def test_speculative_pld_exact_on_hip():
+    """Prompt-lookup speculative decoding on the HIP engine must emit
+    exactly the tokens sequential greedy emits (the drafts ride the
+    mixed-admission prefill path; rejections leave stale KV rows that
+    must never be read — serving/speculative.py)."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.formats import slicer, synthetic
+    from distributedllm_amd.serving.speculative import (SpecStats,
+                                                        pld_generate)
+
+    def eng():
+        f = synthetic.build_model("tiny", seed=0)
+        e = HIPSliceEngine.from_ggml(f, n_ctx=128, max_batch=1)
+        e.attach_extra(slicer.make_extra_layers(f))
+        return e
+
+    def greedy(e, prompt, n):
+        ids = list(prompt)
+        toks = torch.tensor(ids[:-1], dtype=torch.int32, device="cuda")
+        pos = torch.arange(len(ids) - 1, dtype=torch.int32, device="cuda")
+        e.forward(e.embed(toks), pos, torch.zeros_like(pos))
+        out, cur, p = [], ids[-1], len(ids) - 1
+        for _ in range(n):
+            y = e.forward(
+                e.embed(torch.tensor([cur], dtype=torch.int32,
+                                     device="cuda")),
+                torch.tensor([p], dtype=torch.int32, device="cuda"),
+                torch.zeros(1, dtype=torch.int32, device="cuda"),
+                decode=True)
+            cur = int(e.argmax(e.logits(y, all_logits=True))[0])
+            out.append(cur)
+            p += 1
+        return out
+
+    for prompt in ([7, 7, 7, 7], [5, 9, 3], [4, 8, 2, 4, 8, 2, 4, 8]):
+        want = greedy(eng(), prompt, 24)
+        st = SpecStats()
+        got = pld_generate(eng(), prompt, 24, ngram=2, k=6, stats=st)
+        assert got == want, (prompt, got, want)
+        assert st.forwards <= 24
