@@ -14,7 +14,9 @@ Endpoints:
       body: {"prompt": str, "num_tokens": int = 50,
              "temperature": float = 0.0 (0 = greedy),
              "repeat_penalty": float = 1.1, "seed": int | None,
-             "top_k": int = 0, "top_p": float = 1.0}
+             "top_k": int = 0, "top_p": float = 1.0,
+             "stop": [str, ...] | None}  # end at first stop string
+  POST /generate_stream   -> SSE, one data: line per token piece
 """
 from __future__ import annotations
 
@@ -35,6 +37,28 @@ class GenerateRequest(BaseModel):
     seed: Optional[int] = None
     top_k: int = 0
     top_p: float = 1.0
+    stop: Optional[list] = None    # stop strings: generation ends when
+    #                                the decoded continuation contains
+    #                                one; the stop text is not returned
+
+
+def trim_at_stop(tokenizer, ids, stops):
+    """(ids, text) trimmed at the FIRST occurrence of any stop string,
+    stop text excluded (OpenAI `stop` semantics). Token-granular: the
+    cut keeps the shortest token prefix whose decoded text contains no
+    stop — the returned text is cut mid-token-piece when a stop lands
+    inside one."""
+    text = tokenizer.decode(ids)
+    cut = min((p for p in (text.find(s) for s in stops) if p >= 0),
+              default=-1)
+    if cut < 0:
+        return list(ids), text, False
+    keep = []
+    for i, t in enumerate(ids):
+        if len(tokenizer.decode(ids[:i + 1])) > cut:
+            break
+        keep.append(t)
+    return keep, text[:cut], True
 
 
 class BatcherWorker:
@@ -82,12 +106,22 @@ class BatcherWorker:
 
     def submit_and_wait(self, prompt_ids, max_new,
                         sampler: Optional[Sampler],
-                        timeout: float = 300.0):
+                        timeout: float = 300.0,
+                        stop_check=None):
+        """stop_check(req) -> bool: called on every step wake; True
+        cancels the request (used for stop-string termination — the
+        caller trims the output afterwards)."""
         with self.cond:
             req = self.batcher.submit(prompt_ids, max_new, sampler=sampler)
             self.cond.notify_all()
             deadline = time.monotonic() + timeout
+            seen = 0
             while not req.done:
+                if stop_check is not None and len(req.out) > seen:
+                    seen = len(req.out)
+                    if stop_check(req):
+                        self.batcher.cancel(req)
+                        break
                 left = deadline - time.monotonic()
                 if left <= 0:
                     # reclaim the KV slot — an abandoned request must
@@ -135,15 +169,24 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
     @app.post("/generate")
     def generate(r: GenerateRequest):
         ids, sampler = _submit(r)
+        stops = [s for s in (r.stop or []) if s]
+        check = None
+        if stops:
+            def check(req):
+                return any(s in tokenizer.decode(req.out) for s in stops)
         try:
-            req = worker.submit_and_wait(ids, r.num_tokens, sampler)
+            req = worker.submit_and_wait(ids, r.num_tokens, sampler,
+                                         stop_check=check)
         except ValueError as e:        # oversized prompt+num_tokens
             raise HTTPException(422, str(e))
         except TimeoutError as e:
             raise HTTPException(504, str(e))
+        toks, text = list(req.out), tokenizer.decode(req.out)
+        if stops:
+            toks, text, _ = trim_at_stop(tokenizer, req.out, stops)
         stats["requests"] += 1
-        stats["tokens"] += len(req.out)
-        return {"text": tokenizer.decode(req.out), "tokens": req.out}
+        stats["tokens"] += len(toks)
+        return {"text": text, "tokens": toks}
 
     @app.post("/generate_stream")
     def generate_stream(r: GenerateRequest):
@@ -156,6 +199,7 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
         from fastapi.responses import StreamingResponse
 
         ids, sampler = _submit(r)
+        stops = [s for s in (r.stop or []) if s]
         with worker.cond:
             try:
                 req = batcher.submit(ids, r.num_tokens, sampler=sampler)
@@ -165,6 +209,7 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
 
         def events():
             sent = 0
+            text = ""
             try:
                 while True:
                     with worker.cond:
@@ -174,10 +219,33 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
                         sent = len(req.out)
                         done = req.done
                     for tid in chunk:
+                        piece = tokenizer.decode_token(tid)
+                        if stops:
+                            cand = text + piece
+                            cut = min((p for p in (cand.find(x)
+                                                   for x in stops)
+                                       if p >= 0), default=-1)
+                            if cut >= 0:
+                                # stop found: emit only up to the cut,
+                                # free the slot, finish the stream (a
+                                # stop spanning pieces may have partly
+                                # streamed — the done event carries the
+                                # trimmed text)
+                                tail = cand[len(text):cut]
+                                if tail:
+                                    yield ("data: " + _json.dumps(
+                                        {"token": tid, "piece": tail})
+                                        + "\n\n")
+                                with worker.cond:
+                                    batcher.cancel(req)
+                                yield ("event: done\ndata: " +
+                                       _json.dumps({"text": cand[:cut],
+                                                    "stopped": True})
+                                       + "\n\n")
+                                return
+                            text = cand
                         yield ("data: " + _json.dumps(
-                            {"token": tid,
-                             "piece": tokenizer.decode_token(tid)})
-                            + "\n\n")
+                            {"token": tid, "piece": piece}) + "\n\n")
                     if done:
                         yield ("event: done\ndata: " + _json.dumps(
                             {"text": tokenizer.decode(req.out),

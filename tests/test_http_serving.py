@@ -101,3 +101,56 @@ def test_metrics_endpoint(client_and_worker):
     m = c.get("/metrics").json()
     assert m["requests_served"] >= 1 and m["tokens_generated"] >= 3
     assert m["slots"] >= 1 and m["lanes"] == 1
+
+
+def test_stop_strings(client_and_worker):
+    """`stop` ends generation at the first occurrence of a stop string,
+    excluded from the returned text (OpenAI semantics)."""
+    c, _ = client_and_worker
+    full = c.post("/generate",
+                  json={"prompt": "hello", "num_tokens": 12}).json()
+    assert len(full["tokens"]) == 12
+    # use a piece from the middle of the greedy continuation as stop
+    mid = len(full["text"]) // 2
+    stop = full["text"][mid:mid + 2]
+    assert stop and stop in full["text"]
+    r = c.post("/generate", json={"prompt": "hello", "num_tokens": 12,
+                                  "stop": [stop]}).json()
+    assert stop not in r["text"]
+    assert full["text"].startswith(r["text"])
+    assert len(r["tokens"]) < 12
+    # a stop string that never appears changes nothing
+    r2 = c.post("/generate", json={"prompt": "hello", "num_tokens": 12,
+                                   "stop": [" never "]}).json()
+    assert r2["tokens"] == full["tokens"]
+
+
+def test_stop_strings_stream(client_and_worker):
+    import json as _json
+    c, _ = client_and_worker
+    full = c.post("/generate",
+                  json={"prompt": "hello", "num_tokens": 12}).json()
+    mid = len(full["text"]) // 2
+    stop = full["text"][mid:mid + 2]
+    with c.stream("POST", "/generate_stream",
+                  json={"prompt": "hello", "num_tokens": 12,
+                        "stop": [stop]}) as r:
+        lines = [ln for ln in r.iter_lines() if ln]
+    done = _json.loads(lines[-1].removeprefix("data: "))
+    assert done.get("stopped") is True
+    assert stop not in done["text"]
+    assert full["text"].startswith(done["text"])
+
+
+def test_trim_at_stop_unit():
+    from distributedllm_amd.serving.http import trim_at_stop
+
+    class Tok:  # 1 char per token
+        def decode(self, ids):
+            return "".join(chr(i) for i in ids)
+
+    ids = [ord(c) for c in "abcXYdef"]
+    keep, text, hit = trim_at_stop(Tok(), ids, ["XY"])
+    assert hit and text == "abc" and keep == [ord(c) for c in "abc"]
+    keep, text, hit = trim_at_stop(Tok(), ids, ["zz"])
+    assert not hit and text == "abcXYdef" and len(keep) == len(ids)
