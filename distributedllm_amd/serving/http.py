@@ -261,5 +261,51 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
 
         return StreamingResponse(events(), media_type="text/event-stream")
 
+    @app.post("/v1/completions")
+    def v1_completions(body: dict):
+        """OpenAI-compatible completions endpoint (non-streaming): lets
+        standard clients point at this server unchanged. Maps onto the
+        same batcher path as /generate."""
+        prompt = body.get("prompt", "")
+        if isinstance(prompt, list):        # the API allows a list
+            if len(prompt) != 1:
+                raise HTTPException(
+                    400, "only a single prompt per request is supported")
+            prompt = prompt[0]
+        if not isinstance(prompt, str) or not prompt:
+            raise HTTPException(400, "prompt must be a non-empty string")
+        stop = body.get("stop")
+        if isinstance(stop, str):
+            stop = [stop]
+        r = GenerateRequest(
+            prompt=prompt,
+            num_tokens=int(body.get("max_tokens", 16)),
+            temperature=float(body.get("temperature", 0.0)),
+            top_p=float(body.get("top_p", 1.0)),
+            seed=body.get("seed"),
+            stop=stop)
+        out = generate(r)
+        stats["v1_id"] = stats.get("v1_id", 0) + 1
+        return {
+            "id": f"cmpl-{stats['v1_id']}",
+            "object": "text_completion",
+            "model": body.get("model", "distllm-mi355x"),
+            "choices": [{
+                "text": out["text"],
+                "index": 0,
+                "logprobs": None,
+                "finish_reason":
+                    "length" if len(out["tokens"]) >= r.num_tokens
+                    else "stop",
+            }],
+            "usage": {
+                "prompt_tokens": len(tokenizer.encode(prompt, bos=True)),
+                "completion_tokens": len(out["tokens"]),
+                "total_tokens":
+                    len(tokenizer.encode(prompt, bos=True)) +
+                    len(out["tokens"]),
+            },
+        }
+
     app.state.worker = worker
     return app, worker

@@ -154,3 +154,35 @@ def test_trim_at_stop_unit():
     assert hit and text == "abc" and keep == [ord(c) for c in "abc"]
     keep, text, hit = trim_at_stop(Tok(), ids, ["zz"])
     assert not hit and text == "abcXYdef" and len(keep) == len(ids)
+
+
+def test_openai_completions_endpoint(client_and_worker):
+    """OpenAI-compatible /v1/completions maps onto the same batcher
+    path: same greedy tokens as /generate, response in the standard
+    shape (choices/usage/finish_reason)."""
+    c, _ = client_and_worker
+    want = c.post("/generate",
+                  json={"prompt": "hello", "num_tokens": 6}).json()
+    r = c.post("/v1/completions",
+               json={"prompt": "hello", "max_tokens": 6}).json()
+    assert r["object"] == "text_completion"
+    ch = r["choices"][0]
+    assert ch["text"] == want["text"]
+    assert ch["finish_reason"] == "length"
+    assert r["usage"]["completion_tokens"] == 6
+    assert r["usage"]["total_tokens"] == (r["usage"]["prompt_tokens"] + 6)
+    # stop string -> finish_reason "stop"
+    mid = len(want["text"]) // 2
+    stop = want["text"][mid:mid + 2]
+    r2 = c.post("/v1/completions",
+                json={"prompt": "hello", "max_tokens": 6,
+                      "stop": stop}).json()
+    assert r2["choices"][0]["finish_reason"] == "stop"
+    assert stop not in r2["choices"][0]["text"]
+    # list prompt (single) accepted; multi rejected
+    r3 = c.post("/v1/completions",
+                json={"prompt": ["hello"], "max_tokens": 2})
+    assert r3.status_code == 200
+    r4 = c.post("/v1/completions",
+                json={"prompt": ["a", "b"], "max_tokens": 2})
+    assert r4.status_code == 400
