@@ -408,6 +408,11 @@ class ServeHttpCommand(Command):
                        help="max prompt tokens prefilled per lane per "
                             "decode step (latency fairness; 0 = "
                             "unbounded)")
+        p.add_argument("--speculate", type=int, default=0, metavar="K",
+                       help="prompt-lookup speculation inside the "
+                            "shared decode step: up to K draft tokens "
+                            "per greedy request verified per forward "
+                            "(token-exact; 0 = off)")
         p.add_argument("--pipeline", action="store_true",
                        help="serve across torch.distributed pipeline "
                             "ranks (launch with torchrun, one rank per "
@@ -449,7 +454,9 @@ class ServeHttpCommand(Command):
                              for _ in range(n_lanes - 1)]
         tok = Tokenizer(f.vocab)
         bat = ContinuousBatcher(eng, engines=lanes,
-                                prefill_chunk=args.prefill_chunk or None)
+                                prefill_chunk=args.prefill_chunk or None,
+                                spec_ngram=3 if args.speculate else 0,
+                                spec_k=args.speculate)
         app, worker = build_http_app(bat, tok)
         try:
             uvicorn.run(app, host=args.host, port=args.port,

@@ -54,7 +54,8 @@ class ContinuousBatcher:
 
     def __init__(self, engine, max_slots: Optional[int] = None,
                  eos_id: Optional[int] = None, engines=None,
-                 prefill_chunk: Optional[int] = None):
+                 prefill_chunk: Optional[int] = None,
+                 spec_ngram: int = 0, spec_k: int = 0):
         """engines: optional list of k weight-sharing clones ("lanes");
         global slot s lives on lane s%k as that clone's local slot s//k,
         and on CUDA each lane's forward runs on its own HIP stream —
@@ -65,7 +66,19 @@ class ContinuousBatcher:
         (None = unbounded, maximizes throughput). A bound interleaves
         long-prompt admission with decode steps, so in-flight requests
         keep producing tokens while a long prompt loads (latency
-        fairness — Sarathi-style chunked prefill)."""
+        fairness — Sarathi-style chunked prefill).
+
+        spec_ngram/spec_k: optional prompt-lookup speculation INSIDE the
+        shared decode step (both > 0 to enable). Greedy requests whose
+        trailing spec_ngram tokens recur earlier in their sequence get
+        up to spec_k draft tokens verified in the same forward — the
+        drafts are just more (token, pos, seq) rows in the mixed stream
+        the engine already tiles (serving/speculative.py has the
+        single-stream form and the KV-staleness argument; per-slot
+        positions stay strictly sequential, so the same exactness
+        guarantee holds per request). Sampled requests always advance
+        one token. Token-exact with spec off — asserted in
+        tests/test_serving.py."""
         self.engine = engine
         self.lanes = engines if engines else [engine]
         k = len(self.lanes)
@@ -82,6 +95,8 @@ class ContinuousBatcher:
         self._next_rid = 0
         self._dev = getattr(engine, "device", "cpu")
         self.prefill_chunk = prefill_chunk
+        self.spec_ngram = spec_ngram
+        self.spec_k = spec_k
         self.prefilling: Dict[int, Request] = {}  # slot -> request
         self._streams = None
         if k > 1 and self._dev == "cuda":
@@ -201,57 +216,94 @@ class ContinuousBatcher:
         per_lane: List[List[int]] = [[] for _ in range(k)]
         for s in sorted(self.active):
             per_lane[s % k].append(s)
-        work = []  # (reqs, greedy_ids, lg) per lane
+        work = []  # (reqs, drafts, greedy_ids, lg) per lane
         for j, lane_slots in enumerate(per_lane):
             if not lane_slots:
                 continue
             reqs = [self.active[s] for s in lane_slots]
+            drafts = [self._draft(r) for r in reqs]
             eng = self.lanes[j]
             ctx = (torch.cuda.stream(self._streams[j])
                    if self._streams is not None else _nullctx())
             with ctx:
-                toks = torch.tensor([r._next_tok for r in reqs],
-                                    dtype=torch.int32, device=self._dev)
-                pos = torch.tensor([r._pos for r in reqs],
-                                   dtype=torch.int32, device=self._dev)
-                seq = torch.tensor([s // k for s in lane_slots],
-                                   dtype=torch.int32, device=self._dev)
-                y = eng.forward(eng.embed(toks), pos, seq, decode=True)
+                if not any(drafts):
+                    toks = [r._next_tok for r in reqs]
+                    pos = [r._pos for r in reqs]
+                    seq = [s // k for s in lane_slots]
+                    decode = True   # one row per request, distinct seqs
+                else:
+                    toks, pos, seq = [], [], []
+                    for r, s_, d in zip(reqs, lane_slots, drafts):
+                        toks += [r._next_tok] + d
+                        pos += list(range(r._pos, r._pos + 1 + len(d)))
+                        seq += [s_ // k] * (1 + len(d))
+                    decode = False  # same-seq multi-pos rows (mixed
+                    #                 admission stream semantics)
+                t = torch.tensor(toks, dtype=torch.int32,
+                                 device=self._dev)
+                p = torch.tensor(pos, dtype=torch.int32,
+                                 device=self._dev)
+                q = torch.tensor(seq, dtype=torch.int32,
+                                 device=self._dev)
+                y = eng.forward(eng.embed(t), p, q, decode=decode)
                 lg = eng.logits(y, all_logits=True)
                 greedy_ids = None
                 if any(r.sampler is None for r in reqs):
                     greedy_ids = eng.argmax(lg)
-            work.append((reqs, greedy_ids, lg))
+            work.append((reqs, drafts, greedy_ids, lg))
         if self._streams is not None:
             for st in self._streams:
                 torch.cuda.current_stream().wait_stream(st)
 
         finished: List[Request] = []
-        for reqs, greedy_ids, lg in work:
+        for reqs, drafts, greedy_ids, lg in work:
             if greedy_ids is not None and greedy_ids.device.type != "cpu":
                 greedy_ids = greedy_ids.cpu()
             lg_host = None
             if any(r.sampler is not None for r in reqs):
                 lg_host = lg.float().cpu().numpy()
-            self._advance(reqs, greedy_ids, lg_host, finished)
+            self._advance(reqs, drafts, greedy_ids, lg_host, finished)
         return finished
 
-    def _advance(self, reqs, greedy_ids, lg_host, finished) -> None:
-        for i, r in enumerate(reqs):
-            if r.sampler is None:
-                tid = int(greedy_ids[i])
+    def _draft(self, r: Request) -> List[int]:
+        """Prompt-lookup draft for a greedy request (possibly empty)."""
+        if (self.spec_k <= 0 or self.spec_ngram <= 0 or
+                r.sampler is not None):
+            return []
+        n_ctx = getattr(self.engine, "n_ctx", 1 << 30)
+        k = min(self.spec_k, r.max_new - len(r.out) - 1,
+                n_ctx - r._pos - 2)
+        if k <= 0:
+            return []
+        from .speculative import lookup_draft
+        return lookup_draft(r.prompt + r.out, self.spec_ngram, k)
+
+    def _advance(self, reqs, drafts, greedy_ids, lg_host,
+                 finished) -> None:
+        row = 0
+        for r, draft in zip(reqs, drafts):
+            nrows = 1 + len(draft)
+            if r.sampler is not None:
+                emit = [r.sampler(lg_host[row])]
             else:
-                tid = r.sampler(lg_host[i])
-            r.out.append(tid)
-            r._next_tok = tid
-            r._pos += 1
-            if len(r.out) >= r.max_new or (r.eos_id is not None and
-                                           tid == r.eos_id):
-                r.done = True
-                del self.active[r.slot]
-                self.free.append(r.slot)
-                r.slot = -1
-                finished.append(r)
+                nxt = [int(greedy_ids[row + i]) for i in range(nrows)]
+                acc = 0
+                while acc < len(draft) and draft[acc] == nxt[acc]:
+                    acc += 1
+                emit = nxt[:acc + 1]   # verified greedy continuations
+            row += nrows
+            for tid in emit:
+                r.out.append(tid)
+                r._next_tok = tid
+                r._pos += 1
+                if len(r.out) >= r.max_new or (r.eos_id is not None and
+                                               tid == r.eos_id):
+                    r.done = True
+                    del self.active[r.slot]
+                    self.free.append(r.slot)
+                    r.slot = -1
+                    finished.append(r)
+                    break
 
     def cancel(self, req: Request) -> bool:
         """Abort a request: drop it from the queue, or free its slot if

@@ -224,3 +224,62 @@ def test_chunked_prefill_interleaves_and_stays_exact():
     bat.run_all(max_steps=30)
     assert r_short.out == _canonical([7], 6)
     assert r_long.out == _canonical(long_prompt, 3)
+
+
+def test_speculative_batcher_matches_plain():
+    """spec_ngram/spec_k on: per-request outputs must be TOKEN-EXACT
+    with the non-speculative batcher (greedy), across mixed prompt
+    lengths — including repetitive prompts where drafts accept."""
+    prompts = [[7, 7, 7, 7], [5, 9, 3], [4, 8, 2, 4, 8, 2, 4, 8]]
+    steps = [12, 8, 10]
+    plain = ContinuousBatcher(_engine(3))
+    p_reqs = [plain.submit(p, s) for p, s in zip(prompts, steps)]
+    plain.run_all(max_steps=100)
+    spec = ContinuousBatcher(_engine(3), spec_ngram=2, spec_k=4)
+    s_reqs = [spec.submit(p, s) for p, s in zip(prompts, steps)]
+    n_steps = 0
+    while spec.pending:
+        spec.step()
+        n_steps += 1
+        assert n_steps < 100
+    for pr, sr in zip(p_reqs, s_reqs):
+        assert sr.done and sr.out == pr.out
+    # the repetitive request must have ridden accepted drafts: the
+    # whole batch finishes in fewer steps than the longest request's
+    # token count would need without speculation
+    assert n_steps < max(steps)
+
+
+def test_speculative_batcher_mixed_sampled():
+    """Sampled requests ride along one token per step and stay exact
+    (same seeded sampler => same logits sequence => same tokens)."""
+    from distributedllm_amd.engine.sampler import Sampler
+    prompts = [[7, 7, 7, 7], [5, 9, 3]]
+    plain = ContinuousBatcher(_engine(2))
+    pa = plain.submit(prompts[0], 10)
+    pb = plain.submit(prompts[1], 6, sampler=Sampler(0.8, 1.1, seed=3))
+    plain.run_all(max_steps=50)
+    spec = ContinuousBatcher(_engine(2), spec_ngram=2, spec_k=4)
+    sa = spec.submit(prompts[0], 10)
+    sb = spec.submit(prompts[1], 6, sampler=Sampler(0.8, 1.1, seed=3))
+    spec.run_all(max_steps=50)
+    assert sa.out == pa.out
+    assert sb.out == pb.out
+
+
+def test_speculative_batcher_eos_and_slot_reuse():
+    """EOS hit inside an accepted draft ends the request mid-emit and
+    frees the slot for the queue."""
+    plain = ContinuousBatcher(_engine(1))
+    full = plain.submit([7, 7, 7, 7], 16)
+    plain.run_all(max_steps=50)
+    if len(set(full.out)) < 2:
+        return  # degenerate continuation; nothing to cut on
+    eos = full.out[len(full.out) // 2]
+    want = full.out[:full.out.index(eos) + 1]
+    spec = ContinuousBatcher(_engine(1), spec_ngram=2, spec_k=4)
+    r1 = spec.submit([7, 7, 7, 7], 16, eos_id=eos)
+    r2 = spec.submit([5, 9, 3], 3)   # queued until r1's slot frees
+    spec.run_all(max_steps=60)
+    assert r1.out == want
+    assert r2.done and len(r2.out) == 3
