@@ -684,3 +684,34 @@ def test_speculative_pld_exact_on_hip():
         got = pld_generate(eng(), prompt, 24, ngram=2, k=6, stats=st)
         assert got == want, (prompt, got, want)
         assert st.forwards <= 24
+
+
+def test_speculative_batcher_exact_on_hip():
+    """Batched speculation on the HIP engine: per-request outputs must
+    equal the plain batcher's (drafts ride the mixed-admission stream;
+    serving/scheduler.py spec_ngram/spec_k)."""
+    from distributedllm_amd.engine import HIPSliceEngine
+    from distributedllm_amd.formats import slicer, synthetic
+    from distributedllm_amd.serving import ContinuousBatcher
+
+    def eng():
+        f = synthetic.build_model("tiny", seed=0)
+        e = HIPSliceEngine.from_ggml(f, n_ctx=64, max_batch=3)
+        e.attach_extra(slicer.make_extra_layers(f))
+        return e
+
+    prompts = [[7, 7, 7, 7], [5, 9, 3], [4, 8, 2, 4, 8, 2, 4, 8]]
+    steps = [12, 8, 10]
+    plain = ContinuousBatcher(eng())
+    p_reqs = [plain.submit(p, s) for p, s in zip(prompts, steps)]
+    plain.run_all(max_steps=100)
+    spec = ContinuousBatcher(eng(), spec_ngram=2, spec_k=4)
+    s_reqs = [spec.submit(p, s) for p, s in zip(prompts, steps)]
+    n = 0
+    while spec.pending:
+        spec.step()
+        n += 1
+        assert n < 100
+    for pr, sr in zip(p_reqs, s_reqs):
+        assert sr.done and sr.out == pr.out
+    assert n < max(steps)  # drafts accepted on the repetitive request
