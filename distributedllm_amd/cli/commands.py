@@ -508,8 +508,15 @@ class ServeHttpCommand(Command):
             return 0
         facade = PipelineEngine(eng, rank, world) if world > 1 else eng
         tok = Tokenizer(f.vocab)
+        if getattr(args, "speculate", 0) and world > 1:
+            # the pipeline facade returns logits only on decode steps;
+            # draft verification needs logits for non-decode rows
+            print("[serve] --speculate is single-engine only; "
+                  "ignored under --pipeline", file=sys.stderr)
+        spec = getattr(args, "speculate", 0) if world == 1 else 0
         bat = ContinuousBatcher(facade,
-                                prefill_chunk=args.prefill_chunk or None)
+                                prefill_chunk=args.prefill_chunk or None,
+                                spec_ngram=3 if spec else 0, spec_k=spec)
         app, worker = build_http_app(bat, tok)
         try:
             uvicorn.run(app, host=args.host, port=args.port,
