@@ -283,3 +283,28 @@ def test_speculative_batcher_eos_and_slot_reuse():
     spec.run_all(max_steps=60)
     assert r1.out == want
     assert r2.done and len(r2.out) == 3
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_speculative_batcher_fuzz_matches_plain(seed):
+    """Randomized request mix under slot pressure (queueing + slot
+    reuse): speculation must stay token-exact per request."""
+    import random
+    rng = random.Random(seed)
+    jobs = []
+    for _ in range(7):
+        plen = rng.randrange(1, 7)
+        prompt = [rng.randrange(3, 30) for _ in range(plen)]
+        if rng.random() < 0.5:           # bias toward repetition
+            prompt = prompt[:2] * 3
+        jobs.append((prompt, rng.randrange(2, 9)))
+
+    def run(spec):
+        bat = ContinuousBatcher(_engine(2), max_slots=2,
+                                spec_ngram=2 if spec else 0,
+                                spec_k=4 if spec else 0)
+        reqs = [bat.submit(p, s) for p, s in jobs]
+        bat.run_all(max_steps=500)
+        return [r.out for r in reqs]
+
+    assert run(False) == run(True)
