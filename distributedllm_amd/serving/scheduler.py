@@ -26,6 +26,7 @@ from typing import Dict, List, Optional, Sequence
 import torch
 
 from ..engine.sampler import Sampler
+from .speculative import lookup_draft
 
 
 @dataclass
@@ -275,7 +276,6 @@ class ContinuousBatcher:
                 n_ctx - r._pos - 2)
         if k <= 0:
             return []
-        from .speculative import lookup_draft
         return lookup_draft(r.prompt + r.out, self.spec_ngram, k)
 
     def _advance(self, reqs, drafts, greedy_ids, lg_host,
