@@ -249,3 +249,30 @@ def test_torch_repack_edge_case_bits(tmp_path):
         d_th, s_th, wt_th = fn(t, "cpu")
         assert wt_np == wt_th and torch.equal(d_np, d_th)
         assert torch.equal(s_np.view(torch.int16), s_th.view(torch.int16))
+
+
+def test_pmc_summary_tool(tmp_path):
+    """tools/pmc_summary.py aggregates rocprofv3 counter CSVs (the
+    format used for profiles/r2_pmc_decode.md)."""
+    import sys
+    d = tmp_path / "pmc" / "run"
+    d.mkdir(parents=True)
+    csv = d / "1_counter_collection.csv"
+    csv.write_text(
+        '"Dispatch_Id","Kernel_Name","Counter_Name","Counter_Value"\n'
+        '1,"void k_attention<false>(float const*)","SQ_WAVE_CYCLES",100\n'
+        '1,"void k_attention<false>(float const*)","FETCH_SIZE",7\n'
+        '2,"void k_attention<false>(float const*)","SQ_WAVE_CYCLES",50\n'
+        '3,"k_ffn16","SQ_WAVE_CYCLES",30\n')
+    out = subprocess.run(
+        [sys.executable, str(TOOLS / "pmc_summary.py"),
+         str(tmp_path / "pmc")],
+        capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    lines = out.stdout.splitlines()
+    assert lines[0].split()[:2] == ["kernel", "calls"]
+    attn = next(ln for ln in lines if ln.startswith("k_attention"))
+    assert attn.split()[1] == "2"              # distinct dispatch ids
+    assert "1.500e+02" in attn                 # WAVE_CYCLES summed
+    total = next(ln for ln in lines if ln.startswith("TOTAL"))
+    assert "1.800e+02" in total
