@@ -51,7 +51,11 @@ def quantize_q4_0(x: np.ndarray) -> np.ndarray:
     # divides by the unrounded f32 scale (quantize_row_q4_0_reference)
     d32 = (m / -8.0).astype(np.float32)
     d = d32.astype(np.float16)
-    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    # sub-tiny |d32| would overflow 1/d32 to inf (int cast of the
+    # products is then undefined); such blocks quantize to zeros
+    with np.errstate(over="ignore"):
+        inv = np.divide(1.0, d32, out=np.zeros_like(d32),
+                        where=np.abs(d32) >= np.finfo(np.float32).tiny)
     q = np.clip(np.rint(b * inv[..., None]) + 8, 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
     packed = (lo | (hi << 4)).astype(np.uint8)
@@ -91,7 +95,11 @@ def quantize_q4_1(x: np.ndarray) -> np.ndarray:
     d32 = ((mx - mn) / 15.0).astype(np.float32)
     d = d32.astype(np.float16)
     m = mn.astype(np.float16)
-    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    # sub-tiny |d32| would overflow 1/d32 to inf (int cast of the
+    # products is then undefined); such blocks quantize to zeros
+    with np.errstate(over="ignore"):
+        inv = np.divide(1.0, d32, out=np.zeros_like(d32),
+                        where=np.abs(d32) >= np.finfo(np.float32).tiny)
     q = np.clip(np.rint((b - mn[..., None]) * inv[..., None]),
                 0, 15).astype(np.uint8)
     lo, hi = q[..., :16], q[..., 16:]
@@ -168,7 +176,11 @@ def quantize_q5_0(x: np.ndarray) -> np.ndarray:
     m = np.take_along_axis(b, idx[..., None], axis=-1)[..., 0]
     d32 = (m / -16.0).astype(np.float32)
     d = d32.astype(np.float16)
-    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    # sub-tiny |d32| would overflow 1/d32 to inf (int cast of the
+    # products is then undefined); such blocks quantize to zeros
+    with np.errstate(over="ignore"):
+        inv = np.divide(1.0, d32, out=np.zeros_like(d32),
+                        where=np.abs(d32) >= np.finfo(np.float32).tiny)
     q = np.clip(np.rint(b * inv[..., None]) + 16, 0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)
     out = np.empty(b.shape[:2] + (Q5_0_BLOCK_BYTES,), dtype=np.uint8)
@@ -198,7 +210,11 @@ def quantize_q5_1(x: np.ndarray) -> np.ndarray:
     d32 = ((mx - mn) / 31.0).astype(np.float32)
     d = d32.astype(np.float16)
     m = mn.astype(np.float16)
-    inv = np.divide(1.0, d32, out=np.zeros_like(d32), where=d32 != 0.0)
+    # sub-tiny |d32| would overflow 1/d32 to inf (int cast of the
+    # products is then undefined); such blocks quantize to zeros
+    with np.errstate(over="ignore"):
+        inv = np.divide(1.0, d32, out=np.zeros_like(d32),
+                        where=np.abs(d32) >= np.finfo(np.float32).tiny)
     q = np.clip(np.rint((b - mn[..., None]) * inv[..., None]),
                 0, 31).astype(np.uint8)
     packed, qh = _pack_q5(q)

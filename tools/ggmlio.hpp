@@ -19,6 +19,7 @@
 #pragma once
 
 #include <algorithm>
+#include <cfloat>
 #include <cmath>
 #include <cstdint>
 #include <cstdio>
@@ -158,7 +159,7 @@ inline void quantize_block_q4_0(const float* x, uint8_t* out) {
     }
     const float d = m / -8.0f;  // llama.cpp: invert the UNROUNDED scale
     const uint16_t dh = f32_to_f16(d);
-    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    const float inv = (std::fabs(d) >= FLT_MIN) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     for (int j = 0; j < 16; ++j) {
         const float v0 = x[j] * inv, v1 = x[j + 16] * inv;
@@ -178,7 +179,7 @@ inline void quantize_block_q4_1(const float* x, uint8_t* out) {
     const uint16_t dh = f32_to_f16(d);
     const uint16_t mh = f32_to_f16(mn);
     const float m = mn;
-    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    const float inv = (std::fabs(d) >= FLT_MIN) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     std::memcpy(out + 2, &mh, 2);
     for (int j = 0; j < 16; ++j) {
@@ -218,7 +219,7 @@ inline void quantize_block_q5_0(const float* x, uint8_t* out) {
     }
     const float d = m / -16.0f;
     const uint16_t dh = f32_to_f16(d);
-    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    const float inv = (std::fabs(d) >= FLT_MIN) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     uint32_t qh = 0;
     for (int j = 0; j < 16; ++j) {
@@ -257,7 +258,7 @@ inline void quantize_block_q5_1(const float* x, uint8_t* out) {
     const uint16_t dh = f32_to_f16(d);
     const uint16_t mh = f32_to_f16(mn);
     const float m = mn;
-    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    const float inv = (std::fabs(d) >= FLT_MIN) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     std::memcpy(out + 2, &mh, 2);
     uint32_t qh = 0;
@@ -293,10 +294,11 @@ inline void quantize_block_q8_0(const float* x, uint8_t* out) {
     for (int j = 0; j < kQK; ++j) amax = std::max(amax, std::fabs(x[j]));
     const float d = amax / 127.0f;
     const uint16_t dh = f32_to_f16(d);
-    const float inv = (d != 0.0f) ? 1.0f / d : 0.0f;
+    const float inv = (std::fabs(d) >= FLT_MIN) ? 1.0f / d : 0.0f;
     std::memcpy(out, &dh, 2);
     for (int j = 0; j < kQK; ++j)
-        out[2 + j] = (uint8_t)(int8_t)std::nearbyintf(x[j] * inv);
+        out[2 + j] = (uint8_t)(int8_t)std::clamp(
+            (int)std::nearbyintf(x[j] * inv), -127, 127);
 }
 
 inline void dequantize_block_q8_0(const uint8_t* in, float* x) {
