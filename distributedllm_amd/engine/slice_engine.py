@@ -839,7 +839,8 @@ class HIPSliceEngine:
         assert getattr(self, "_layers_cache", None) is not None, \
             "clone_shared requires an engine built by .random()/.from_ggml"
         twin = HIPSliceEngine(self.hp, self.n_layers, self.first_layer,
-                              self.n_ctx, self.max_batch)
+                              self.n_ctx, self.max_batch,
+                              max_prefill=int(self._eng.max_prefill))
         for li, (an, fn, mats) in enumerate(self._layers_cache):
             twin._eng.set_layer(li, an, fn, mats)
         twin._layers_cache = self._layers_cache  # enables further clones
