@@ -1,4 +1,5 @@
 from .scheduler import ContinuousBatcher, Request  # noqa: F401
+from .speculative import SpecStats, pld_generate  # noqa: F401
 
 
 def build_http_app(batcher, tokenizer, eos_id=None):

@@ -63,8 +63,7 @@ def pld_generate(engine, prompt_ids, max_new: int, *, ngram: int = 3,
     Returns the generated token ids (length <= max_new; stops at
     eos_id). Token-exact with sequential greedy decode.
     """
-    dev = engine.device if isinstance(getattr(engine, "device", None),
-                                      str) else "cpu"
+    dev = getattr(engine, "device", "cpu")
     ids = [int(t) for t in prompt_ids]
     assert len(ids) >= 1, "prompt must be non-empty"
     n_ctx = int(engine.n_ctx)
