@@ -229,6 +229,10 @@ class GenerateTextCommand(Command):
         p.add_argument("--top-p", type=float, default=1.0,
                        help="nucleus sampling mass (1.0=off)")
         p.add_argument("--seed", type=int, default=None)
+        p.add_argument("--speculate", type=int, default=0, metavar="K",
+                       help="with --greedy: verify up to K prompt-lookup "
+                            "draft tokens per pipeline hop (token-exact; "
+                            "amortizes the per-token TCP round-trip)")
         p.add_argument("--root", default=".")
 
     def __call__(self, args) -> int:
@@ -239,7 +243,9 @@ class GenerateTextCommand(Command):
                                   temperature=args.temp,
                                   repeat_penalty=args.rp,
                                   greedy=args.greedy, seed=args.seed,
-                                  top_k=args.top_k, top_p=args.top_p):
+                                  top_k=args.top_k, top_p=args.top_p,
+                                  speculative=getattr(args, "speculate",
+                                                      0)):
             print(piece, end="", flush=True)
         print()
         r = llm.throughput.report()

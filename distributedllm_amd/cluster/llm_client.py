@@ -72,10 +72,21 @@ class DistributedLLM:
     def generate(self, prompt: str, max_steps: int = 50,
                  temperature: float = 0.7, repeat_penalty: float = 1.1,
                  seed: Optional[int] = None, greedy: bool = False,
-                 top_k: int = 0, top_p: float = 1.0) -> Iterator[str]:
+                 top_k: int = 0, top_p: float = 1.0,
+                 speculative: int = 0) -> Iterator[str]:
+        """speculative=K (greedy only): prompt-lookup drafts of up to K
+        tokens are verified per pipeline hop — the per-token TCP
+        round-trip (THE latency bound of this plane) is paid once per
+        accepted run instead of once per token. Token-exact with plain
+        greedy (serving/speculative.py has the argument; the nodes are
+        stateless over explicit start_pos, so a rejection simply rewinds
+        the client's n_past and the stale KV rows are rewritten)."""
         self.clear_context()
         self.throughput.reset()
         tokens = self.tokenizer.encode(prompt, bos=True)
+        if speculative > 0 and greedy:
+            yield from self._generate_spec(tokens, max_steps, speculative)
+            return
         sampler = Sampler(temperature, repeat_penalty, seed=seed,
                           greedy=greedy, top_k=top_k, top_p=top_p)
         cur = tokens
@@ -87,6 +98,34 @@ class DistributedLLM:
             self.throughput.tick()
             yield self.tokenizer.decode_token(tid)
             cur = [tid]
+
+    def _generate_spec(self, tokens: List[int], max_steps: int,
+                       k: int) -> Iterator[str]:
+        from ..serving.speculative import lookup_draft
+        if len(tokens) > 1:
+            self.propagate_tensor(self._embed(tokens[:-1]))
+        cur = tokens[-1]
+        p = len(tokens) - 1
+        emitted = 0
+        while emitted < max_steps:
+            draft = lookup_draft(tokens, 3,
+                                 min(k, max_steps - emitted - 1))
+            self.n_past = p   # rewind over any rejected rows
+            y = self.propagate_tensor(self._embed([cur] + draft))
+            lg = self._logits(np.ascontiguousarray(y), all_logits=True)
+            nxt = lg.argmax(axis=-1)
+            acc = 0
+            while acc < len(draft) and draft[acc] == int(nxt[acc]):
+                acc += 1
+            for tid in nxt[:acc + 1]:
+                tokens.append(int(tid))
+                emitted += 1
+                self.throughput.tick()
+                yield self.tokenizer.decode_token(int(tid))
+                if emitted >= max_steps:
+                    return
+            cur = tokens[-1]
+            p += acc + 1
 
     def perplexity(self, text: str) -> float:
         """exp(mean NLL) of each next token given its prefix

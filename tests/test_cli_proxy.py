@@ -334,3 +334,28 @@ def test_tcp_cluster_bench_tool(tmp_path):
             if l.startswith("{") and "tok_s_warm" in l][-1]
     r = _json.loads(line)
     assert r["tokens"] == 4 and r["tok_s_warm"] > 0
+
+
+def test_tcp_generate_speculative_exact(node, tmp_path):
+    """Speculative generation over the TCP pipeline must emit exactly
+    the plain-greedy continuation (drafts verified per hop; rejections
+    rewind the client's n_past against the stateless node)."""
+    from distributedllm_amd.cluster.llm_client import get_llm
+    addr = f"127.0.0.1:{node.port}"
+    root = tmp_path / "root"
+    root.mkdir()
+    cfg = {"model_id": "tiny_spec",
+           "location": "synthetic:tiny",
+           "nodes_map": {addr: [0, PRESETS["tiny"].n_layer - 1]},
+           "quantization": "f16",
+           "metadata": {"name": "tiny", "family": "llama_v1"}}
+    cfg_path = tmp_path / "cfg.json"
+    cfg_path.write_text(json.dumps(cfg))
+    assert execute_command(["provision", str(cfg_path),
+                            "--root", str(root)]) == 0
+    llm = get_llm(str(cfg_path), root=str(root))
+    for prompt in ("aaaa aaaa aaaa", "hello"):
+        plain = list(llm.generate(prompt, max_steps=16, greedy=True))
+        spec = list(llm.generate(prompt, max_steps=16, greedy=True,
+                                 speculative=6))
+        assert spec == plain, prompt
