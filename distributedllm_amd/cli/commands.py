@@ -319,10 +319,11 @@ class BatchGenerateCommand(Command):
         p.add_argument("--seed", type=int, default=None)
         p.add_argument("--speculative", action="store_true",
                        help="prompt-lookup speculative decoding (greedy "
-                            "single-prompt only): multiple tokens per "
-                            "forward when the continuation repeats "
-                            "earlier text; output is token-exact with "
-                            "plain greedy")
+                            "only): multiple tokens per forward when "
+                            "the continuation repeats earlier text; "
+                            "token-exact. One prompt uses the single-"
+                            "stream fast path, several enable in-"
+                            "batcher speculation (spec_k=8)")
 
     def __call__(self, args) -> int:
         import time
@@ -350,10 +351,10 @@ class BatchGenerateCommand(Command):
         tok = Tokenizer(f.vocab)
 
         if args.speculative:
-            if len(prompts) != 1 or not args.greedy:
-                print("--speculative needs exactly one prompt and "
-                      "--greedy", file=sys.stderr)
+            if not args.greedy:
+                print("--speculative needs --greedy", file=sys.stderr)
                 return 2
+        if args.speculative and len(prompts) == 1:
             from ..serving.speculative import SpecStats, pld_generate
             st = SpecStats()
             t0 = time.perf_counter()
@@ -368,7 +369,10 @@ class BatchGenerateCommand(Command):
                   file=sys.stderr)
             return 0
 
-        bat = ContinuousBatcher(eng, max_slots=n_slots)
+        spec = bool(args.speculative)
+        bat = ContinuousBatcher(eng, max_slots=n_slots,
+                                spec_ngram=3 if spec else 0,
+                                spec_k=8 if spec else 0)
         reqs = []
         for text in prompts:
             sampler = None if args.greedy else \
