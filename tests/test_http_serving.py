@@ -186,3 +186,27 @@ def test_openai_completions_endpoint(client_and_worker):
     r4 = c.post("/v1/completions",
                 json={"prompt": ["a", "b"], "max_tokens": 2})
     assert r4.status_code == 400
+
+
+def test_http_with_batcher_speculation():
+    """serve_http --speculate path: an app over a speculating batcher
+    returns exactly what the plain app returns."""
+    from fastapi.testclient import TestClient
+    f = synthetic.build_model("tiny", seed=0)
+
+    def app_for(spec):
+        eng = TorchSliceEngine.from_ggml(f, n_ctx=64, max_batch=2)
+        eng.attach_extra(slicer.make_extra_layers(f))
+        bat = ContinuousBatcher(eng, spec_ngram=3 if spec else 0,
+                                spec_k=8 if spec else 0)
+        return build_http_app(bat, Tokenizer(f.vocab))
+
+    app_p, w_p = app_for(False)
+    app_s, w_s = app_for(True)
+    body = {"prompt": "aa aa aa", "num_tokens": 12}
+    with TestClient(app_p) as cp, TestClient(app_s) as cs:
+        a = cp.post("/generate", json=body).json()
+        b = cs.post("/generate", json=body).json()
+    w_p.stop()
+    w_s.stop()
+    assert a["tokens"] == b["tokens"]
