@@ -42,6 +42,12 @@ def main() -> int:
     ap.add_argument("--model", default="open_llama_3b")
     ap.add_argument("--quant", default="q4_0")
     ap.add_argument("--tokens", type=int, default=16)
+    ap.add_argument("--speculate", type=int, default=0,
+                    help="also time a speculative run (greedy, K drafts "
+                         "per hop); synthetic random-init continuations "
+                         "repeat unrealistically often, so treat the "
+                         "acceptance as a mechanism demo, not a real-"
+                         "text estimate")
     ap.add_argument("--base-port", type=int, default=19870)
     ap.add_argument("--workdir", default="/tmp/tcp_cluster_bench")
     args = ap.parse_args()
@@ -96,10 +102,28 @@ def main() -> int:
                 timeout=3600)
             return time.time() - t0, out
 
+        def gen_spec():
+            t0 = time.time()
+            out = subprocess.run(
+                [sys.executable, os.path.join(REPO, "manager.py"),
+                 "generate_text", cfgp, "--prompt", "Once upon a time",
+                 "--num-tokens", str(args.tokens), "--greedy",
+                 "--speculate", str(args.speculate),
+                 "--root", args.workdir],
+                check=True, cwd=REPO, capture_output=True, text=True,
+                timeout=3600)
+            return time.time() - t0, out
+
         # cold: includes each node dequantizing + loading its slice on
         # first use; warm: slices stay resident in the node processes
         t_cold, out = gen()
         t_warm, out = gen()
+        t_spec = None
+        if args.speculate:
+            t_spec, out_s = gen_spec()
+            plain_txt = out.stdout.splitlines()[0]
+            spec_txt = out_s.stdout.splitlines()[0]
+            assert spec_txt == plain_txt, "speculative output diverged!"
         print(out.stdout.strip()[-400:])
         print(json.dumps({
             "config": f"{args.model} {args.quant}, {args.nodes} "
@@ -109,6 +133,9 @@ def main() -> int:
             "warm_generate_s": round(t_warm, 2),
             "tokens": args.tokens,
             "tok_s_warm": round(args.tokens / t_warm, 2),
+            "spec_generate_s": round(t_spec, 2) if t_spec else None,
+            "tok_s_spec": round(args.tokens / t_spec, 2) if t_spec
+            else None,
             "note": "end-to-end incl. per-token TCP round-trips through "
                     "every node (reference-workflow serialization)"}),
             flush=True)
