@@ -152,6 +152,9 @@ def build_app(batcher, tokenizer, eos_id: Optional[int] = None):
                 "pending": batcher.pending,
                 "slots": batcher.n_slots,
                 "lanes": len(batcher.lanes),
+                "decode_steps": batcher.steps_run,
+                "tokens_per_step": round(
+                    batcher.tokens_out / max(batcher.steps_run, 1), 3),
                 "uptime_s": round(up, 1),
                 "tokens_per_s_lifetime": round(
                     stats["tokens"] / max(up, 1e-9), 1)}

@@ -99,6 +99,8 @@ class ContinuousBatcher:
         self.spec_ngram = spec_ngram
         self.spec_k = spec_k
         self.prefilling: Dict[int, Request] = {}  # slot -> request
+        self.steps_run = 0      # decode steps executed
+        self.tokens_out = 0     # tokens emitted (== steps when spec off)
         self._streams = None
         if k > 1 and self._dev == "cuda":
             self._streams = [torch.cuda.Stream() for _ in self.lanes]
@@ -212,6 +214,7 @@ class ContinuousBatcher:
         self._admit()
         if not self.active:
             return []
+        self.steps_run += 1
         k = len(self.lanes)
         # one decode launch per lane, each on its own stream (CUDA)
         per_lane: List[List[int]] = [[] for _ in range(k)]
@@ -293,6 +296,7 @@ class ContinuousBatcher:
                 emit = nxt[:acc + 1]   # verified greedy continuations
             row += nrows
             for tid in emit:
+                self.tokens_out += 1
                 r.out.append(tid)
                 r._next_tok = tid
                 r._pos += 1
