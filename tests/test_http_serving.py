@@ -210,3 +210,50 @@ def test_http_with_batcher_speculation():
     w_p.stop()
     w_s.stop()
     assert a["tokens"] == b["tokens"]
+
+
+def test_concurrent_speculative_chunked_soak():
+    """Concurrency soak: threads hammer /generate over a speculating
+    batcher with chunked prefill; every response must be complete and
+    per-prompt deterministic (scheduling must never leak into
+    outputs)."""
+    import random
+    import threading
+    from fastapi.testclient import TestClient
+    f = synthetic.build_model("tiny", seed=0)
+    eng = TorchSliceEngine.from_ggml(f, n_ctx=64, max_batch=4)
+    eng.attach_extra(slicer.make_extra_layers(f))
+    bat = ContinuousBatcher(eng, prefill_chunk=8, spec_ngram=3, spec_k=6)
+    app, worker = build_http_app(bat, Tokenizer(f.vocab))
+    results, errors = {}, []
+    lock = threading.Lock()
+
+    def client_thread(tid):
+        rng = random.Random(tid)
+        with TestClient(app) as c:
+            for i in range(6):
+                words = " ".join(rng.choice(["aa", "bb", "cc"])
+                                 for _ in range(rng.randrange(1, 8)))
+                n = rng.randrange(1, 10)
+                r = c.post("/generate", json={"prompt": words,
+                                              "num_tokens": n})
+                if r.status_code != 200:
+                    errors.append((tid, i, r.status_code))
+                    return
+                toks = tuple(r.json()["tokens"])
+                with lock:
+                    key = (words, n)
+                    if len(toks) != n or results.get(key, toks) != toks:
+                        errors.append((tid, i, key, toks))
+                        return
+                    results[key] = toks
+
+    threads = [threading.Thread(target=client_thread, args=(t,))
+               for t in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=300)
+    worker.stop()
+    assert not errors, errors[:3]
+    assert bat.tokens_out >= bat.steps_run  # spec never reduces tokens
