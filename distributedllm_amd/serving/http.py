@@ -17,6 +17,8 @@ Endpoints:
              "top_k": int = 0, "top_p": float = 1.0,
              "stop": [str, ...] | None}  # end at first stop string
   POST /generate_stream   -> SSE, one data: line per token piece
+  POST /v1/completions    -> OpenAI-compatible completions shape
+  GET  /metrics           -> lifetime counters incl. tokens_per_step
 """
 from __future__ import annotations
 
