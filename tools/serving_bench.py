@@ -38,6 +38,12 @@ def main():
                          "experiment: how much do they stall in-flight "
                          "decodes?)")
     ap.add_argument("--long-prompt", type=int, default=512)
+    ap.add_argument("--speculate", type=int, default=0,
+                    help="in-batcher prompt-lookup speculation (greedy "
+                         "requests, K drafts; token-exact). NOTE: "
+                         "random-init synthetic text repeats "
+                         "unrealistically often — acceptance rates here "
+                         "overstate real-text gains")
     args = ap.parse_args()
 
     hp = PRESETS[args.model].hparams(ggml.FTYPE_MOSTLY_Q4_0)
@@ -49,7 +55,9 @@ def main():
                          for _ in range(args.lanes - 1)]
     g = torch.Generator().manual_seed(1)
     bat = ContinuousBatcher(eng, engines=lanes,
-                            prefill_chunk=args.prefill_chunk or None)
+                            prefill_chunk=args.prefill_chunk or None,
+                            spec_ngram=3 if args.speculate else 0,
+                            spec_k=args.speculate)
     reqs = [bat.submit(torch.randint(3, hp.n_vocab, (args.prompt_len,),
                                      generator=g).tolist(),
                        args.num_tokens)
