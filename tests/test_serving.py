@@ -308,3 +308,12 @@ def test_speculative_batcher_fuzz_matches_plain(seed):
         return [r.out for r in reqs]
 
     assert run(False) == run(True)
+
+
+def test_batch_generate_speculative_needs_greedy(tmp_path, capsys):
+    from distributedllm_amd.cli import execute_command
+    f = synthetic.build_model("tiny", seed=0)
+    path = tmp_path / "m.bin"
+    f.save(str(path))
+    assert execute_command(["batch_generate", str(path), "--prompt", "a",
+                            "--speculative"]) == 2
