@@ -97,3 +97,20 @@ def test_pld_eos_stops():
         eos = full[len(full) // 2]
         got = pld_generate(eng, [7, 7, 7], 16, eos_id=eos)
         assert got == full[:full.index(eos) + 1]
+
+
+def test_pld_exact_on_gqa_model():
+    """Speculation over a GQA engine (grouped KV heads) stays exact."""
+    from distributedllm_amd.formats import ggml
+
+    def eng():
+        f = synthetic.build_model("tiny_gqa", seed=2,
+                                  ftype=ggml.FTYPE_MOSTLY_F16)
+        e = TorchSliceEngine.from_ggml(f, n_ctx=96, max_batch=1)
+        e.attach_extra(slicer.make_extra_layers(f))
+        return e
+
+    for prompt in ([9, 9, 9, 9], [4, 8, 2, 4, 8, 2]):
+        want = _greedy(eng(), prompt, 20)
+        got = pld_generate(eng(), prompt, 20, ngram=2, k=5)
+        assert got == want, prompt
