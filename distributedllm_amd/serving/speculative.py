@@ -65,7 +65,8 @@ def pld_generate(engine, prompt_ids, max_new: int, *, ngram: int = 3,
     """
     dev = getattr(engine, "device", "cpu")
     ids = [int(t) for t in prompt_ids]
-    assert len(ids) >= 1, "prompt must be non-empty"
+    if not ids:
+        raise ValueError("prompt must be non-empty")
     n_ctx = int(engine.n_ctx)
     if len(ids) + max_new > n_ctx:
         raise ValueError(
