@@ -18,8 +18,12 @@ from distributedllm_amd.formats import ggml, synthetic
 from distributedllm_amd.models.llama import (
     PRESETS, layer_tensor_names)
 
-FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q8_0": ggml.FTYPE_MOSTLY_Q8_0,
-          "f16": ggml.FTYPE_MOSTLY_F16, "q4_K": ggml.FTYPE_MOSTLY_Q4_K_M,
+FTYPES = {"q4_0": ggml.FTYPE_MOSTLY_Q4_0, "q4_1": ggml.FTYPE_MOSTLY_Q4_1,
+          "q5_0": ggml.FTYPE_MOSTLY_Q5_0, "q5_1": ggml.FTYPE_MOSTLY_Q5_1,
+          "q8_0": ggml.FTYPE_MOSTLY_Q8_0, "f16": ggml.FTYPE_MOSTLY_F16,
+          "q2_K": ggml.FTYPE_MOSTLY_Q2_K, "q3_K": ggml.FTYPE_MOSTLY_Q3_K_M,
+          "q4_K": ggml.FTYPE_MOSTLY_Q4_K_M,
+          "q5_K": ggml.FTYPE_MOSTLY_Q5_K_M,
           "q6_K": ggml.FTYPE_MOSTLY_Q6_K}
 
 
