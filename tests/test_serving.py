@@ -300,9 +300,12 @@ def test_speculative_batcher_fuzz_matches_plain(seed):
         jobs.append((prompt, rng.randrange(2, 9)))
 
     def run(spec):
+        # chunked-prefill setting drawn independently per run: outputs
+        # must not depend on scheduling at all
         bat = ContinuousBatcher(_engine(2), max_slots=2,
                                 spec_ngram=2 if spec else 0,
-                                spec_k=4 if spec else 0)
+                                spec_k=4 if spec else 0,
+                                prefill_chunk=rng.choice([None, 4]))
         reqs = [bat.submit(p, s) for p, s in jobs]
         bat.run_all(max_steps=500)
         return [r.out for r in reqs]
